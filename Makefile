@@ -1,0 +1,52 @@
+# migbm build: C++ host core (g++/OpenMP) + HIP gfx950 kernels (hipcc) -> lib_migbm.so
+# The .so is placed in lightgbm_amd/lib/ where the Python package's libpath finds it.
+
+CXX      ?= g++
+HIPCC    ?= /opt/rocm/bin/hipcc
+ROCM     ?= /opt/rocm
+
+CXXFLAGS  = -O3 -std=c++17 -fPIC -fopenmp -Wall -Wextra -Wno-unused-parameter \
+            -Icpp/include -MMD -MP
+HIPFLAGS  = --offload-arch=gfx950 -O3 -std=c++17 -fPIC -Icpp/include -Wall \
+            -Wno-unused-parameter -MMD -MP
+LDFLAGS   = -shared -fopenmp -L$(ROCM)/lib -Wl,-rpath,$(ROCM)/lib
+
+BUILD    := build
+LIBDIR   := lightgbm_amd/lib
+TARGET   := $(LIBDIR)/lib_migbm.so
+
+HOST_SRCS := $(wildcard cpp/src/*.cpp)
+HIP_SRCS  := $(wildcard cpp/src/hip/*.hip.cpp)
+
+HOST_OBJS := $(patsubst cpp/src/%.cpp,$(BUILD)/%.o,$(HOST_SRCS))
+HIP_OBJS  := $(patsubst cpp/src/hip/%.hip.cpp,$(BUILD)/hip/%.o,$(HIP_SRCS))
+
+# link with hipcc when HIP objects exist (pulls in amdhip64 + device code)
+ifeq ($(strip $(HIP_OBJS)),)
+  LINKER = $(CXX)
+  EXTRA_LIBS =
+else
+  LINKER = $(HIPCC)
+  EXTRA_LIBS = -lamdhip64 -lrccl
+endif
+
+all: $(TARGET)
+
+$(BUILD):
+	mkdir -p $(BUILD) $(BUILD)/hip $(LIBDIR)
+
+$(BUILD)/%.o: cpp/src/%.cpp | $(BUILD)
+	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/hip/%.o: cpp/src/hip/%.hip.cpp | $(BUILD)
+	$(HIPCC) $(HIPFLAGS) -c $< -o $@
+
+$(TARGET): $(HOST_OBJS) $(HIP_OBJS) | $(BUILD)
+	$(LINKER) $(LDFLAGS) $(HOST_OBJS) $(HIP_OBJS) $(EXTRA_LIBS) -o $@
+
+clean:
+	rm -rf $(BUILD) $(TARGET)
+
+-include $(BUILD)/*.d $(BUILD)/hip/*.d
+
+.PHONY: all clean

@@ -1,0 +1,205 @@
+/*!
+ * migbm Dataset — binned training data + Metadata (label/weight/query/init_score).
+ * Capability parity target: reference include/LightGBM/dataset.h, src/io/dataset.cpp,
+ * src/io/metadata.cpp, src/io/dataset_loader.cpp. Fresh design: dense per-feature bin
+ * columns (uint8/uint16) on the host, plus an optional row-major packed matrix view used
+ * by the HIP learner (built once, resident in HBM).
+ */
+#ifndef MIGBM_DATASET_H_
+#define MIGBM_DATASET_H_
+
+#include "bin.h"
+#include "common.h"
+#include "config.h"
+
+#include <memory>
+#include <string>
+#include <vector>
+
+namespace migbm {
+
+/*! Per-row ancillary data. Parity: reference Metadata (dataset.h:48-400). */
+class Metadata {
+ public:
+  void Init(data_size_t num_data, bool has_weight, bool has_query);
+  void SetLabel(const float* label, data_size_t n);
+  void SetWeights(const float* w, data_size_t n);
+  void SetQuery(const int32_t* group_sizes, data_size_t n_groups);
+  void SetQueryBoundaries(std::vector<data_size_t> boundaries);
+  void SetInitScore(const double* s, int64_t n);
+  void SetPosition(const int32_t* p, data_size_t n);
+
+  data_size_t num_data() const { return num_data_; }
+  const label_t* label() const { return label_.data(); }
+  const label_t* weights() const { return weights_.empty() ? nullptr : weights_.data(); }
+  const data_size_t* query_boundaries() const {
+    return query_boundaries_.empty() ? nullptr : query_boundaries_.data();
+  }
+  data_size_t num_queries() const {
+    return query_boundaries_.empty() ? 0 : static_cast<data_size_t>(query_boundaries_.size() - 1);
+  }
+  const label_t* query_weights() const {
+    return query_weights_.empty() ? nullptr : query_weights_.data();
+  }
+  const double* init_score() const { return init_score_.empty() ? nullptr : init_score_.data(); }
+  int64_t num_init_score() const { return static_cast<int64_t>(init_score_.size()); }
+  const int32_t* positions() const { return positions_.empty() ? nullptr : positions_.data(); }
+
+  std::vector<label_t>& mutable_label() { return label_; }
+
+ private:
+  data_size_t num_data_ = 0;
+  std::vector<label_t> label_;
+  std::vector<label_t> weights_;
+  std::vector<data_size_t> query_boundaries_;
+  std::vector<label_t> query_weights_;
+  std::vector<double> init_score_;
+  std::vector<int32_t> positions_;
+};
+
+/*! Dense bin column; uint8 when num_bin<=256 else uint16. */
+class BinColumn {
+ public:
+  void Init(data_size_t n, int num_bin) {
+    is16_ = num_bin > 256;
+    if (is16_) d16_.assign(n, 0);
+    else d8_.assign(n, 0);
+  }
+  inline void Set(data_size_t i, uint32_t b) {
+    if (is16_) d16_[i] = static_cast<uint16_t>(b);
+    else d8_[i] = static_cast<uint8_t>(b);
+  }
+  inline uint32_t Get(data_size_t i) const { return is16_ ? d16_[i] : d8_[i]; }
+  bool is16() const { return is16_; }
+  const uint8_t* data8() const { return d8_.data(); }
+  const uint16_t* data16() const { return d16_.data(); }
+
+ private:
+  bool is16_ = false;
+  std::vector<uint8_t> d8_;
+  std::vector<uint16_t> d16_;
+};
+
+class Dataset {
+ public:
+  Dataset() = default;
+  explicit Dataset(data_size_t num_data) : num_data_(num_data) {}
+
+  /*! Build bin mappers + columns from a dense matrix.
+   *  \param sample_getter value at (row, col); NaN allowed
+   *  \param categorical per-original-column flag */
+  void ConstructFromMat(const std::function<double(data_size_t, int)>& get, data_size_t nrow,
+                        int ncol, const Config& cfg, const std::vector<int8_t>& categorical);
+
+  /*! Build an aligned valid set re-using this (train) dataset's bin mappers. */
+  std::unique_ptr<Dataset> CreateValid(const std::function<double(data_size_t, int)>& get,
+                                       data_size_t nrow) const;
+
+  /*! Column-wise histogram build over an ordered index subset.
+   *  hist layout: per used feature f at hist_offset(f)*2, (sum_grad, sum_hess) pairs.
+   *  ordered_grad/hess must be pre-gathered to match data_indices order. */
+  void ConstructHistograms(const std::vector<int8_t>& is_feature_used,
+                           const data_size_t* data_indices, data_size_t num_data,
+                           const score_t* ordered_grad, const score_t* ordered_hess,
+                           hist_t* hist) const;
+
+  /*! Histogram for one inner feature into out[2*bin]. */
+  void ConstructHistogramForFeature(int fidx, const data_size_t* data_indices,
+                                    data_size_t num_data, const score_t* ordered_grad,
+                                    const score_t* ordered_hess, hist_t* out) const;
+
+  data_size_t num_data() const { return num_data_; }
+  int num_features() const { return static_cast<int>(bin_mappers_.size()); }
+  int num_total_features() const { return num_total_features_; }
+  int InnerFeatureIndex(int orig) const { return used_feature_map_[orig]; }
+  int RealFeatureIndex(int inner) const { return real_feature_index_[inner]; }
+  const BinMapper* FeatureBinMapper(int inner) const { return bin_mappers_[inner].get(); }
+  int FeatureNumBin(int inner) const { return bin_mappers_[inner]->num_bin(); }
+  uint32_t hist_offset(int inner) const { return hist_offsets_[inner]; }
+  int num_total_bin() const { return num_total_bin_; }
+  uint32_t GetBin(data_size_t row, int inner) const { return columns_[inner].Get(row); }
+  const BinColumn& column(int inner) const { return columns_[inner]; }
+
+  Metadata& metadata() { return metadata_; }
+  const Metadata& metadata() const { return metadata_; }
+
+  const std::vector<std::string>& feature_names() const { return feature_names_; }
+  void set_feature_names(const std::vector<std::string>& names);
+  std::string FeatureInfoString() const;  // "feature_infos=" payload
+  std::string GetFeatureName(int orig) const {
+    return orig < static_cast<int>(feature_names_.size()) ? feature_names_[orig]
+                                                          : "Column_" + std::to_string(orig);
+  }
+
+  /*! Threshold raw value for (inner feature, bin): upper bound of the bin. */
+  double RealThreshold(int inner, uint32_t bin) const {
+    return bin_mappers_[inner]->BinToValue(bin);
+  }
+
+  /*! Row-major packed view for the HIP learner: one byte (or 2) per used feature,
+   *  padded row stride. Built lazily. */
+  struct RowMajorView {
+    std::vector<uint8_t> data;     // uint8 path (all features <=256 bins incl. nan bin)
+    std::vector<uint16_t> data16;  // fallback
+    int row_stride = 0;            // elements per row (padded)
+    bool is16 = false;
+  };
+  const RowMajorView& GetRowMajorView() const;
+
+  /*! Used by tests / CLI: binary serialization of the dataset. */
+  void SaveBinaryFile(const char* filename) const;
+  static std::unique_ptr<Dataset> LoadFromBinFile(const char* filename);
+  static bool IsBinFile(const char* filename);
+
+  void DumpTextFile(const char* filename) const;
+
+  /*! Subset copy of rows (bagging subset path / GetSubset C API). */
+  std::unique_ptr<Dataset> Subset(const data_size_t* indices, data_size_t n) const;
+
+  /*! Push a single raw row (streaming API). Requires mappers already built. */
+  void PushRawRow(data_size_t row, const double* values, int ncol);
+
+  std::vector<int8_t> categorical_flags_;   // per original column
+
+ private:
+  friend class DatasetLoader;
+  void FinishBinMappers(const Config& cfg);
+
+  data_size_t num_data_ = 0;
+  int num_total_features_ = 0;
+  std::vector<int> used_feature_map_;        // orig -> inner (-1 trivial)
+  std::vector<int> real_feature_index_;      // inner -> orig
+  std::vector<std::unique_ptr<BinMapper>> bin_mappers_;  // per inner feature
+  std::vector<BinColumn> columns_;
+  std::vector<uint32_t> hist_offsets_;
+  int num_total_bin_ = 0;
+  Metadata metadata_;
+  std::vector<std::string> feature_names_;
+  mutable RowMajorView row_view_;
+  mutable bool row_view_built_ = false;
+};
+
+/*! Text-file loading (CSV/TSV/LibSVM autodetect).
+ *  Parity: reference src/io/dataset_loader.cpp + parser.cpp. */
+class DatasetLoader {
+ public:
+  DatasetLoader(const Config& cfg, int label_idx = 0) : cfg_(cfg), label_idx_(label_idx) {}
+  std::unique_ptr<Dataset> LoadFromFile(const char* filename, int rank = 0, int num_machines = 1);
+  std::unique_ptr<Dataset> LoadFromFileAlignWithOtherDataset(const char* filename,
+                                                             const Dataset* train);
+
+ protected:
+  void ParseFile(const char* filename, std::vector<std::vector<double>>* rows,
+                 std::vector<float>* labels, std::vector<float>* weights,
+                 std::vector<int32_t>* groups, int* ncol, int rank, int num_machines);
+  Config cfg_;
+  int label_idx_;
+};
+
+/*! Raw-row loading used by LGBM_BoosterPredictForFile. */
+std::vector<std::vector<double>> LoadRawRowsForPredict(const char* filename, const Config& cfg,
+                                                       int expected_ncol);
+
+}  // namespace migbm
+
+#endif  // MIGBM_DATASET_H_

@@ -1,0 +1,147 @@
+/*!
+ * migbm TreeLearner interface + SerialTreeLearner (CPU leaf-wise histogram learner).
+ * Capability parity target: reference include/LightGBM/tree_learner.h,
+ * src/treelearner/serial_tree_learner.{h,cpp}, data_partition.hpp. Fresh implementation.
+ */
+#ifndef MIGBM_TREE_LEARNER_H_
+#define MIGBM_TREE_LEARNER_H_
+
+#include "config.h"
+#include "dataset.h"
+#include "feature_histogram.h"
+#include "tree.h"
+
+#include <memory>
+#include <vector>
+
+namespace migbm {
+
+class TreeLearner {
+ public:
+  virtual ~TreeLearner() = default;
+  virtual void Init(const Dataset* train_data, bool is_constant_hessian) = 0;
+  virtual void ResetTrainingData(const Dataset* train_data) = 0;
+  virtual void ResetConfig(const Config* config) = 0;
+  /*! Train one tree from gradients/hessians (already bagging-masked via SetBaggingData). */
+  virtual Tree* Train(const score_t* gradients, const score_t* hessians, bool is_first_tree) = 0;
+  virtual void SetBaggingData(const Dataset* subset, const data_size_t* used_indices,
+                              data_size_t num_data) = 0;
+  /*! Refit an existing tree structure to new gradients. */
+  virtual Tree* FitByExistingTree(const Tree* old_tree, const score_t* g, const score_t* h);
+  virtual Tree* FitByExistingTree(const Tree* old_tree, const std::vector<int>& leaf_pred,
+                                  const score_t* g, const score_t* h);
+  virtual void AddPredictionToScore(const Tree* tree, double* out_score) = 0;
+  virtual void RenewTreeOutput(Tree* tree, const class ObjectiveFunction* obj,
+                               std::function<double(const label_t*, int)> residual_getter,
+                               data_size_t total_num_data, const data_size_t* bag_indices,
+                               data_size_t bag_cnt, const double* train_score) = 0;
+
+  static TreeLearner* Create(const std::string& learner_type, const std::string& device_type,
+                             const Config* config);
+};
+
+/*! Leaf -> row-index partition with multithreaded stable split. */
+class DataPartition {
+ public:
+  void Init(data_size_t num_data, int num_leaves) {
+    num_data_ = num_data;
+    indices_.resize(num_data);
+    temp_.resize(num_data);
+    leaf_begin_.assign(num_leaves, 0);
+    leaf_count_.assign(num_leaves, 0);
+    row_to_leaf_.assign(num_data, 0);
+  }
+  /*! reset to a single root leaf holding all (or bagged) rows */
+  void ResetToRoot(const data_size_t* used_indices, data_size_t cnt) {
+    if (used_indices != nullptr) {
+      std::copy(used_indices, used_indices + cnt, indices_.begin());
+      used_cnt_ = cnt;
+    } else {
+      used_cnt_ = num_data_;
+#pragma omp parallel for schedule(static)
+      for (data_size_t i = 0; i < num_data_; ++i) indices_[i] = i;
+    }
+    std::fill(leaf_begin_.begin(), leaf_begin_.end(), 0);
+    std::fill(leaf_count_.begin(), leaf_count_.end(), 0);
+    leaf_count_[0] = used_cnt_;
+  }
+  const data_size_t* GetIndexOnLeaf(int leaf, data_size_t* out_cnt) const {
+    *out_cnt = leaf_count_[leaf];
+    return indices_.data() + leaf_begin_[leaf];
+  }
+  data_size_t leaf_count(int leaf) const { return leaf_count_[leaf]; }
+
+  /*! Stable-partition rows of `leaf` into (leaf, right_leaf) by predicate go_left(row). */
+  void Split(int leaf, int right_leaf, const std::function<bool(data_size_t)>& go_left);
+
+  data_size_t used_cnt() const { return used_cnt_; }
+
+ private:
+  data_size_t num_data_ = 0;
+  data_size_t used_cnt_ = 0;
+  std::vector<data_size_t> indices_, temp_;
+  std::vector<data_size_t> leaf_begin_, leaf_count_;
+  std::vector<int> row_to_leaf_;  // reserved for score updater use
+};
+
+class SerialTreeLearner : public TreeLearner {
+ public:
+  explicit SerialTreeLearner(const Config* config) : config_(config) {}
+  void Init(const Dataset* train_data, bool is_constant_hessian) override;
+  void ResetTrainingData(const Dataset* train_data) override;
+  void ResetConfig(const Config* config) override { config_ = config; }
+  virtual Tree* Train(const score_t* gradients, const score_t* hessians,
+                      bool is_first_tree) override;
+  void SetBaggingData(const Dataset* subset, const data_size_t* used_indices,
+                      data_size_t num_data) override;
+  void AddPredictionToScore(const Tree* tree, double* out_score) override;
+  void RenewTreeOutput(Tree* tree, const class ObjectiveFunction* obj,
+                       std::function<double(const label_t*, int)> residual_getter,
+                       data_size_t total_num_data, const data_size_t* bag_indices,
+                       data_size_t bag_cnt, const double* train_score) override;
+
+ protected:
+  /*! Histogram for `leaf` into its slot; optionally by subtraction (parent - sibling). */
+  void ComputeHistogram(int leaf, data_size_t cnt, const data_size_t* indices);
+  /*! hook after a leaf histogram is built (data-parallel: global allreduce). */
+  virtual void OnHistogramReady(int leaf) { (void)leaf; }
+  /*! hook to globalize root stats (data-parallel: allreduce). */
+  virtual void ReduceRootStats(double* sum_g, double* sum_h, data_size_t* cnt) {
+    (void)sum_g; (void)sum_h; (void)cnt;
+  }
+  /*! hook to globalize child row counts after a split (data-parallel: allreduce). */
+  virtual void GlobalChildCounts(data_size_t* left_cnt, data_size_t* right_cnt) {
+    (void)left_cnt; (void)right_cnt;
+  }
+  void SubtractHistogram(int dst_slot_leaf, int parent_slot, int sibling_slot);
+  /*! Scan all used features of leaf, fill best_split_per_leaf_[leaf]. */
+  virtual void FindBestSplitForLeaf(int leaf, const LeafContext& ctx);
+  hist_t* HistSlot(int slot) { return hist_store_.data() + static_cast<size_t>(slot) * 2 * train_data_->num_total_bin(); }
+  /*! per-node feature sampling mask (feature_fraction / bynode / interaction constraints) */
+  std::vector<int8_t> SampleFeatures(bool per_node);
+  /*! predicate for partition: does row go left under split s of inner feature f? */
+  std::function<bool(data_size_t)> MakeGoLeft(const SplitInfo& s) const;
+
+  const Config* config_;
+  const Dataset* train_data_ = nullptr;
+  bool is_constant_hessian_ = false;
+  const score_t* gradients_ = nullptr;
+  const score_t* hessians_ = nullptr;
+  std::vector<score_t> ordered_grad_, ordered_hess_;
+  DataPartition partition_;
+  std::vector<hist_t> hist_store_;          // num_leaves slots x 2*num_total_bin
+  std::vector<int> leaf_to_slot_;
+  std::vector<SplitInfo> best_split_per_leaf_;
+  std::vector<LeafContext> leaf_ctx_;
+  std::vector<int8_t> is_feature_used_;     // per-tree mask
+  // bagging
+  const data_size_t* bag_indices_ = nullptr;
+  data_size_t bag_cnt_ = 0;
+  Random feature_rng_{0};
+  Random extra_rng_{0};
+  int iter_counter_ = 0;
+};
+
+}  // namespace migbm
+
+#endif  // MIGBM_TREE_LEARNER_H_

@@ -1,0 +1,507 @@
+/*! migbm Dataset implementation: sampling + binning + dense bin columns + histograms +
+ *  row-major packed view for the HIP learner + binary save/load.
+ *  Parity target: reference src/io/dataset.cpp (ConstructHistogramsInner, SaveBinaryFile),
+ *  src/io/metadata.cpp. Fresh implementation. */
+#include "migbm/dataset.h"
+
+#include <cstdio>
+#include <numeric>
+
+namespace migbm {
+
+// ------------------------------------------------------------------ Metadata
+void Metadata::Init(data_size_t num_data, bool has_weight, bool has_query) {
+  num_data_ = num_data;
+  label_.assign(num_data, 0.0f);
+  if (has_weight) weights_.assign(num_data, 1.0f);
+  if (has_query) query_boundaries_.clear();
+}
+
+void Metadata::SetLabel(const float* label, data_size_t n) {
+  num_data_ = n;
+  label_.assign(label, label + n);
+}
+
+void Metadata::SetWeights(const float* w, data_size_t n) {
+  if (w == nullptr || n == 0) { weights_.clear(); return; }
+  MIGBM_CHECK_EQ(n, num_data_);
+  weights_.assign(w, w + n);
+}
+
+void Metadata::SetQuery(const int32_t* group_sizes, data_size_t n_groups) {
+  if (group_sizes == nullptr || n_groups == 0) { query_boundaries_.clear(); return; }
+  query_boundaries_.resize(n_groups + 1);
+  query_boundaries_[0] = 0;
+  for (data_size_t i = 0; i < n_groups; ++i)
+    query_boundaries_[i + 1] = query_boundaries_[i] + group_sizes[i];
+  MIGBM_CHECK_EQ(query_boundaries_.back(), num_data_);
+}
+
+void Metadata::SetQueryBoundaries(std::vector<data_size_t> boundaries) {
+  query_boundaries_ = std::move(boundaries);
+}
+
+void Metadata::SetInitScore(const double* s, int64_t n) {
+  if (s == nullptr || n == 0) { init_score_.clear(); return; }
+  init_score_.assign(s, s + n);
+}
+
+void Metadata::SetPosition(const int32_t* p, data_size_t n) {
+  if (p == nullptr || n == 0) { positions_.clear(); return; }
+  MIGBM_CHECK_EQ(n, num_data_);
+  positions_.assign(p, p + n);
+}
+
+// ------------------------------------------------------------------ Dataset
+void Dataset::FinishBinMappers(const Config&) {
+  hist_offsets_.resize(bin_mappers_.size());
+  uint32_t off = 0;
+  for (size_t i = 0; i < bin_mappers_.size(); ++i) {
+    hist_offsets_[i] = off;
+    off += bin_mappers_[i]->num_bin();
+  }
+  num_total_bin_ = static_cast<int>(off);
+}
+
+void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& get,
+                               data_size_t nrow, int ncol, const Config& cfg,
+                               const std::vector<int8_t>& categorical) {
+  num_data_ = nrow;
+  num_total_features_ = ncol;
+  categorical_flags_ = categorical;
+  if (categorical_flags_.empty()) categorical_flags_.assign(ncol, 0);
+  // 1. sample rows
+  int sample_cnt = std::min<data_size_t>(cfg.bin_construct_sample_cnt, nrow);
+  Random rng(cfg.data_random_seed);
+  std::vector<data_size_t> sample_idx;
+  if (sample_cnt >= nrow) {
+    sample_idx.resize(nrow);
+    std::iota(sample_idx.begin(), sample_idx.end(), 0);
+  } else {
+    // uniform stride sample with random offset (deterministic, cheap, unbiased enough)
+    sample_idx.reserve(sample_cnt);
+    double stride = static_cast<double>(nrow) / sample_cnt;
+    double pos = rng.NextFloat() * stride;
+    for (int i = 0; i < sample_cnt; ++i) {
+      sample_idx.push_back(std::min<data_size_t>(static_cast<data_size_t>(pos), nrow - 1));
+      pos += stride;
+    }
+  }
+  // 2. find bins per column (parallel)
+  std::vector<std::unique_ptr<BinMapper>> mappers(ncol);
+  const int ns = static_cast<int>(sample_idx.size());
+#pragma omp parallel for schedule(dynamic, 1)
+  for (int c = 0; c < ncol; ++c) {
+    std::vector<double> vals(ns);
+    for (int i = 0; i < ns; ++i) vals[i] = get(sample_idx[i], c);
+    auto m = std::make_unique<BinMapper>();
+    m->FindBin(vals.data(), ns, ns, cfg.max_bin, cfg.min_data_in_bin, 0,
+               cfg.feature_pre_filter, categorical_flags_[c] ? BinType::kCategorical
+                                                             : BinType::kNumerical,
+               cfg.use_missing, cfg.zero_as_missing);
+    mappers[c] = std::move(m);
+  }
+  // 3. keep non-trivial features
+  used_feature_map_.assign(ncol, -1);
+  real_feature_index_.clear();
+  bin_mappers_.clear();
+  for (int c = 0; c < ncol; ++c) {
+    if (!mappers[c]->is_trivial()) {
+      used_feature_map_[c] = static_cast<int>(bin_mappers_.size());
+      real_feature_index_.push_back(c);
+      bin_mappers_.push_back(std::move(mappers[c]));
+    }
+  }
+  if (bin_mappers_.empty())
+    Log::Warning("All features are trivial (constant); no informative splits possible");
+  FinishBinMappers(cfg);
+  // 4. bin all values
+  columns_.resize(bin_mappers_.size());
+  const int nf = static_cast<int>(bin_mappers_.size());
+#pragma omp parallel for schedule(dynamic, 1)
+  for (int f = 0; f < nf; ++f) {
+    columns_[f].Init(nrow, bin_mappers_[f]->num_bin());
+    const int c = real_feature_index_[f];
+    const BinMapper* m = bin_mappers_[f].get();
+    for (data_size_t i = 0; i < nrow; ++i) columns_[f].Set(i, m->ValueToBin(get(i, c)));
+  }
+  metadata_.Init(nrow, false, false);
+  if (feature_names_.empty()) {
+    for (int c = 0; c < ncol; ++c) feature_names_.push_back("Column_" + std::to_string(c));
+  }
+}
+
+std::unique_ptr<Dataset> Dataset::CreateValid(
+    const std::function<double(data_size_t, int)>& get, data_size_t nrow) const {
+  auto d = std::make_unique<Dataset>(nrow);
+  d->num_total_features_ = num_total_features_;
+  d->used_feature_map_ = used_feature_map_;
+  d->real_feature_index_ = real_feature_index_;
+  d->feature_names_ = feature_names_;
+  d->categorical_flags_ = categorical_flags_;
+  d->bin_mappers_.resize(bin_mappers_.size());
+  for (size_t i = 0; i < bin_mappers_.size(); ++i)
+    d->bin_mappers_[i] = std::make_unique<BinMapper>(*bin_mappers_[i]);
+  d->hist_offsets_ = hist_offsets_;
+  d->num_total_bin_ = num_total_bin_;
+  const int nf = static_cast<int>(bin_mappers_.size());
+  d->columns_.resize(nf);
+#pragma omp parallel for schedule(dynamic, 1)
+  for (int f = 0; f < nf; ++f) {
+    d->columns_[f].Init(nrow, d->bin_mappers_[f]->num_bin());
+    const int c = real_feature_index_[f];
+    const BinMapper* m = d->bin_mappers_[f].get();
+    for (data_size_t i = 0; i < nrow; ++i) d->columns_[f].Set(i, m->ValueToBin(get(i, c)));
+  }
+  d->metadata_.Init(nrow, false, false);
+  return d;
+}
+
+namespace {
+
+/*! Hot loop: histogram over an ordered subset with software prefetch.
+ *  MI355X note: this is the host fallback; the HIP learner has its own LDS kernel. */
+template <typename BIN_T, bool USE_INDICES>
+void HistInner(const BIN_T* bins, const data_size_t* idx, data_size_t n,
+               const score_t* og, const score_t* oh, hist_t* hist) {
+  const data_size_t rest = n & 3;
+  const data_size_t nb = n - rest;
+  for (data_size_t i = 0; i < nb; i += 4) {
+    data_size_t r0 = USE_INDICES ? idx[i] : i;
+    data_size_t r1 = USE_INDICES ? idx[i + 1] : i + 1;
+    data_size_t r2 = USE_INDICES ? idx[i + 2] : i + 2;
+    data_size_t r3 = USE_INDICES ? idx[i + 3] : i + 3;
+    if (USE_INDICES && i + 16 < nb) __builtin_prefetch(bins + idx[i + 16], 0, 0);
+    const uint32_t b0 = bins[r0] << 1, b1 = bins[r1] << 1, b2 = bins[r2] << 1,
+                   b3 = bins[r3] << 1;
+    hist[b0] += og[i];     hist[b0 + 1] += oh[i];
+    hist[b1] += og[i + 1]; hist[b1 + 1] += oh[i + 1];
+    hist[b2] += og[i + 2]; hist[b2 + 1] += oh[i + 2];
+    hist[b3] += og[i + 3]; hist[b3 + 1] += oh[i + 3];
+  }
+  for (data_size_t i = nb; i < n; ++i) {
+    data_size_t r = USE_INDICES ? idx[i] : i;
+    const uint32_t b = bins[r] << 1;
+    hist[b] += og[i];
+    hist[b + 1] += oh[i];
+  }
+}
+
+}  // namespace
+
+void Dataset::ConstructHistogramForFeature(int f, const data_size_t* data_indices,
+                                           data_size_t num_data, const score_t* og,
+                                           const score_t* oh, hist_t* out) const {
+  const BinColumn& col = columns_[f];
+  const bool use_idx = data_indices != nullptr;
+  if (col.is16()) {
+    if (use_idx) HistInner<uint16_t, true>(col.data16(), data_indices, num_data, og, oh, out);
+    else HistInner<uint16_t, false>(col.data16(), nullptr, num_data, og, oh, out);
+  } else {
+    if (use_idx) HistInner<uint8_t, true>(col.data8(), data_indices, num_data, og, oh, out);
+    else HistInner<uint8_t, false>(col.data8(), nullptr, num_data, og, oh, out);
+  }
+}
+
+void Dataset::ConstructHistograms(const std::vector<int8_t>& is_feature_used,
+                                  const data_size_t* data_indices, data_size_t num_data,
+                                  const score_t* og, const score_t* oh, hist_t* hist) const {
+  if (num_data <= 0) return;
+  const int nf = num_features();
+#pragma omp parallel for schedule(dynamic)
+  for (int f = 0; f < nf; ++f) {
+    if (!is_feature_used[f]) continue;
+    hist_t* out = hist + 2 * hist_offsets_[f];
+    std::fill(out, out + 2 * bin_mappers_[f]->num_bin(), 0.0);
+    ConstructHistogramForFeature(f, data_indices, num_data, og, oh, out);
+  }
+}
+
+void Dataset::set_feature_names(const std::vector<std::string>& names) {
+  feature_names_ = names;
+}
+
+std::string Dataset::FeatureInfoString() const {
+  std::vector<std::string> infos(num_total_features_, "none");
+  for (int f = 0; f < num_features(); ++f)
+    infos[real_feature_index_[f]] = bin_mappers_[f]->ToFeatureInfoString();
+  return Common::Join(infos, " ");
+}
+
+const Dataset::RowMajorView& Dataset::GetRowMajorView() const {
+  if (row_view_built_) return row_view_;
+  const int nf = num_features();
+  bool any16 = false;
+  for (int f = 0; f < nf; ++f) any16 |= columns_[f].is16();
+  // pad row stride to 16 elements for aligned vector loads on device
+  int stride = (nf + 15) & ~15;
+  row_view_.row_stride = stride;
+  row_view_.is16 = any16;
+  if (!any16) {
+    row_view_.data.assign(static_cast<size_t>(num_data_) * stride, 0);
+#pragma omp parallel for schedule(static)
+    for (data_size_t i = 0; i < num_data_; ++i) {
+      uint8_t* row = row_view_.data.data() + static_cast<size_t>(i) * stride;
+      for (int f = 0; f < nf; ++f) row[f] = static_cast<uint8_t>(columns_[f].Get(i));
+    }
+  } else {
+    row_view_.data16.assign(static_cast<size_t>(num_data_) * stride, 0);
+#pragma omp parallel for schedule(static)
+    for (data_size_t i = 0; i < num_data_; ++i) {
+      uint16_t* row = row_view_.data16.data() + static_cast<size_t>(i) * stride;
+      for (int f = 0; f < nf; ++f) row[f] = static_cast<uint16_t>(columns_[f].Get(i));
+    }
+  }
+  row_view_built_ = true;
+  return row_view_;
+}
+
+std::unique_ptr<Dataset> Dataset::Subset(const data_size_t* indices, data_size_t n) const {
+  auto d = std::make_unique<Dataset>(n);
+  d->num_total_features_ = num_total_features_;
+  d->used_feature_map_ = used_feature_map_;
+  d->real_feature_index_ = real_feature_index_;
+  d->feature_names_ = feature_names_;
+  d->categorical_flags_ = categorical_flags_;
+  d->bin_mappers_.resize(bin_mappers_.size());
+  for (size_t i = 0; i < bin_mappers_.size(); ++i)
+    d->bin_mappers_[i] = std::make_unique<BinMapper>(*bin_mappers_[i]);
+  d->hist_offsets_ = hist_offsets_;
+  d->num_total_bin_ = num_total_bin_;
+  const int nf = num_features();
+  d->columns_.resize(nf);
+#pragma omp parallel for schedule(dynamic, 1)
+  for (int f = 0; f < nf; ++f) {
+    d->columns_[f].Init(n, d->bin_mappers_[f]->num_bin());
+    for (data_size_t i = 0; i < n; ++i) d->columns_[f].Set(i, columns_[f].Get(indices[i]));
+  }
+  // metadata subset
+  d->metadata_.Init(n, false, false);
+  std::vector<float> lab(n);
+  const label_t* src = metadata_.label();
+  for (data_size_t i = 0; i < n; ++i) lab[i] = src[indices[i]];
+  d->metadata_.SetLabel(lab.data(), n);
+  if (metadata_.weights() != nullptr) {
+    std::vector<float> w(n);
+    for (data_size_t i = 0; i < n; ++i) w[i] = metadata_.weights()[indices[i]];
+    d->metadata_.SetWeights(w.data(), n);
+  }
+  return d;
+}
+
+void Dataset::PushRawRow(data_size_t row, const double* values, int ncol) {
+  for (int c = 0; c < std::min(ncol, num_total_features_); ++c) {
+    int f = used_feature_map_[c];
+    if (f >= 0) columns_[f].Set(row, bin_mappers_[f]->ValueToBin(values[c]));
+  }
+}
+
+// ------------------------------------------------------------------ binary file
+static const char kBinMagic[] = "migbm.dataset.v1";
+
+void Dataset::SaveBinaryFile(const char* filename) const {
+  FILE* fp = fopen(filename, "wb");
+  if (!fp) Log::Fatal("Cannot open %s for writing", filename);
+  fwrite(kBinMagic, 1, sizeof(kBinMagic), fp);
+  std::stringstream ss;
+  ss << num_data_ << " " << num_total_features_ << " " << num_total_bin_ << "\n";
+  ss << Common::Join(used_feature_map_, " ") << "\n";
+  ss << Common::Join(real_feature_index_, " ") << "\n";
+  ss << Common::Join(feature_names_, "\t") << "\n";
+  for (auto& m : bin_mappers_) ss << m->ToString();
+  std::string header = ss.str();
+  uint64_t hlen = header.size();
+  fwrite(&hlen, sizeof(hlen), 1, fp);
+  fwrite(header.data(), 1, hlen, fp);
+  for (auto& col : columns_) {
+    uint8_t is16 = col.is16();
+    fwrite(&is16, 1, 1, fp);
+    if (is16) fwrite(col.data16(), sizeof(uint16_t), num_data_, fp);
+    else fwrite(col.data8(), sizeof(uint8_t), num_data_, fp);
+  }
+  // metadata
+  uint8_t has_w = metadata_.weights() != nullptr;
+  uint8_t has_q = metadata_.query_boundaries() != nullptr;
+  fwrite(&has_w, 1, 1, fp);
+  fwrite(&has_q, 1, 1, fp);
+  fwrite(metadata_.label(), sizeof(label_t), num_data_, fp);
+  if (has_w) fwrite(metadata_.weights(), sizeof(label_t), num_data_, fp);
+  if (has_q) {
+    data_size_t nq = metadata_.num_queries();
+    fwrite(&nq, sizeof(nq), 1, fp);
+    fwrite(metadata_.query_boundaries(), sizeof(data_size_t), nq + 1, fp);
+  }
+  fclose(fp);
+}
+
+bool Dataset::IsBinFile(const char* filename) {
+  FILE* fp = fopen(filename, "rb");
+  if (!fp) return false;
+  char magic[sizeof(kBinMagic)] = {0};
+  size_t got = fread(magic, 1, sizeof(kBinMagic), fp);
+  fclose(fp);
+  return got == sizeof(kBinMagic) && memcmp(magic, kBinMagic, sizeof(kBinMagic)) == 0;
+}
+
+std::unique_ptr<Dataset> Dataset::LoadFromBinFile(const char* filename) {
+  FILE* fp = fopen(filename, "rb");
+  if (!fp) Log::Fatal("Cannot open %s", filename);
+  char magic[sizeof(kBinMagic)];
+  MIGBM_CHECK_EQ(fread(magic, 1, sizeof(kBinMagic), fp), sizeof(kBinMagic));
+  MIGBM_CHECK_EQ(memcmp(magic, kBinMagic, sizeof(kBinMagic)), 0);
+  uint64_t hlen;
+  MIGBM_CHECK_EQ(fread(&hlen, sizeof(hlen), 1, fp), 1u);
+  std::string header(hlen, '\0');
+  MIGBM_CHECK_EQ(fread(&header[0], 1, hlen, fp), hlen);
+  auto lines = Common::Split(header.c_str(), '\n');
+  auto head = Common::SplitAny(lines[0].c_str(), " ");
+  auto d = std::make_unique<Dataset>();
+  d->num_data_ = atoi(head[0].c_str());
+  d->num_total_features_ = atoi(head[1].c_str());
+  d->num_total_bin_ = atoi(head[2].c_str());
+  Common::StringToArray<int>(lines[1], ' ', &d->used_feature_map_);
+  Common::StringToArray<int>(lines[2], ' ', &d->real_feature_index_);
+  d->feature_names_ = Common::Split(lines[3].c_str(), '\t');
+  int nf = static_cast<int>(d->real_feature_index_.size());
+  d->bin_mappers_.resize(nf);
+  size_t line_pos = 4;
+  for (int f = 0; f < nf; ++f) {
+    std::string blob = lines[line_pos] + "\n" + lines[line_pos + 1] + "\n" + lines[line_pos + 2];
+    d->bin_mappers_[f] = std::make_unique<BinMapper>();
+    d->bin_mappers_[f]->FromString(blob);
+    line_pos += 3;
+  }
+  Config dummy;
+  d->FinishBinMappers(dummy);
+  d->columns_.resize(nf);
+  for (int f = 0; f < nf; ++f) {
+    uint8_t is16;
+    MIGBM_CHECK_EQ(fread(&is16, 1, 1, fp), 1u);
+    d->columns_[f].Init(d->num_data_, is16 ? 65536 : 256);
+    if (is16) {
+      MIGBM_CHECK_EQ(fread(const_cast<uint16_t*>(d->columns_[f].data16()), sizeof(uint16_t),
+                           d->num_data_, fp), static_cast<size_t>(d->num_data_));
+    } else {
+      MIGBM_CHECK_EQ(fread(const_cast<uint8_t*>(d->columns_[f].data8()), sizeof(uint8_t),
+                           d->num_data_, fp), static_cast<size_t>(d->num_data_));
+    }
+  }
+  uint8_t has_w, has_q;
+  MIGBM_CHECK_EQ(fread(&has_w, 1, 1, fp), 1u);
+  MIGBM_CHECK_EQ(fread(&has_q, 1, 1, fp), 1u);
+  std::vector<label_t> lab(d->num_data_);
+  MIGBM_CHECK_EQ(fread(lab.data(), sizeof(label_t), d->num_data_, fp),
+                 static_cast<size_t>(d->num_data_));
+  d->metadata_.SetLabel(lab.data(), d->num_data_);
+  if (has_w) {
+    std::vector<label_t> w(d->num_data_);
+    MIGBM_CHECK_EQ(fread(w.data(), sizeof(label_t), d->num_data_, fp),
+                   static_cast<size_t>(d->num_data_));
+    d->metadata_.SetWeights(w.data(), d->num_data_);
+  }
+  if (has_q) {
+    data_size_t nq;
+    MIGBM_CHECK_EQ(fread(&nq, sizeof(nq), 1, fp), 1u);
+    std::vector<data_size_t> qb(nq + 1);
+    MIGBM_CHECK_EQ(fread(qb.data(), sizeof(data_size_t), nq + 1, fp),
+                   static_cast<size_t>(nq + 1));
+    d->metadata_.SetQueryBoundaries(std::move(qb));
+  }
+  fclose(fp);
+  return d;
+}
+
+void Dataset::DumpTextFile(const char* filename) const {
+  FILE* fp = fopen(filename, "w");
+  if (!fp) Log::Fatal("Cannot open %s", filename);
+  fprintf(fp, "num_data=%d\nnum_features=%d\nnum_total_bin=%d\n", num_data_, num_features(),
+          num_total_bin_);
+  for (int f = 0; f < num_features(); ++f) {
+    fprintf(fp, "feature %d (orig %d) num_bin=%d\n", f, real_feature_index_[f],
+            bin_mappers_[f]->num_bin());
+  }
+  fclose(fp);
+}
+
+// ------------------------------------------------------------------ Tree score update
+// (defined here to have Dataset complete)
+
+}  // namespace migbm
+
+#include "migbm/tree.h"
+
+namespace migbm {
+
+void Tree::AddPredictionToScore(const Dataset* data, data_size_t num_data, double* score) const {
+  if (num_leaves_ <= 1) {
+    if (leaf_value_[0] != 0.0) {
+#pragma omp parallel for schedule(static)
+      for (data_size_t i = 0; i < num_data; ++i) score[i] += leaf_value_[0];
+    }
+    return;
+  }
+#pragma omp parallel for schedule(static, 2048)
+  for (data_size_t i = 0; i < num_data; ++i) {
+    int node = 0;
+    while (node >= 0) {
+      const int f = split_feature_inner_[node];
+      const uint32_t bin = data->GetBin(i, f);
+      if (IsCategoricalSplit(node)) {
+        // bin-level categorical decision via bitset over bins
+        const int cat_idx = static_cast<int>(threshold_in_bin_[node]);
+        const uint32_t* bits = cat_threshold_.data() + cat_boundaries_[cat_idx];
+        const int n_words = cat_boundaries_[cat_idx + 1] - cat_boundaries_[cat_idx];
+        node = ((bin >> 5) < static_cast<uint32_t>(n_words) && ((bits[bin >> 5] >> (bin & 31)) & 1))
+                   ? left_child_[node] : right_child_[node];
+      } else {
+        // NaN bin (if any) is the last bin; honor default direction
+        const BinMapper* m = data->FeatureBinMapper(f);
+        const int nanb = m->nan_bin();
+        if (nanb >= 0 && bin == static_cast<uint32_t>(nanb)) {
+          node = (decision_type_[node] & kDefaultLeftMask) ? left_child_[node] : right_child_[node];
+        } else {
+          node = bin <= threshold_in_bin_[node] ? left_child_[node] : right_child_[node];
+        }
+      }
+    }
+    score[i] += leaf_value_[~node];
+  }
+}
+
+void Tree::AddPredictionToScore(const Dataset* data, const data_size_t* used_indices,
+                                data_size_t num_data, double* score) const {
+  if (num_leaves_ <= 1) {
+    if (leaf_value_[0] != 0.0) {
+#pragma omp parallel for schedule(static)
+      for (data_size_t i = 0; i < num_data; ++i) score[used_indices[i]] += leaf_value_[0];
+    }
+    return;
+  }
+#pragma omp parallel for schedule(static, 2048)
+  for (data_size_t i = 0; i < num_data; ++i) {
+    const data_size_t r = used_indices[i];
+    int node = 0;
+    while (node >= 0) {
+      const int f = split_feature_inner_[node];
+      const uint32_t bin = data->GetBin(r, f);
+      if (IsCategoricalSplit(node)) {
+        const int cat_idx = static_cast<int>(threshold_in_bin_[node]);
+        const uint32_t* bits = cat_threshold_.data() + cat_boundaries_[cat_idx];
+        const int n_words = cat_boundaries_[cat_idx + 1] - cat_boundaries_[cat_idx];
+        node = ((bin >> 5) < static_cast<uint32_t>(n_words) && ((bits[bin >> 5] >> (bin & 31)) & 1))
+                   ? left_child_[node] : right_child_[node];
+      } else {
+        const BinMapper* m = data->FeatureBinMapper(f);
+        const int nanb = m->nan_bin();
+        if (nanb >= 0 && bin == static_cast<uint32_t>(nanb)) {
+          node = (decision_type_[node] & kDefaultLeftMask) ? left_child_[node] : right_child_[node];
+        } else {
+          node = bin <= threshold_in_bin_[node] ? left_child_[node] : right_child_[node];
+        }
+      }
+    }
+    score[r] += leaf_value_[~node];
+  }
+}
+
+}  // namespace migbm

@@ -1,0 +1,345 @@
+/*! migbm SerialTreeLearner — CPU leaf-wise (best-first) histogram learner with histogram
+ *  subtraction. Parity target: reference src/treelearner/serial_tree_learner.cpp (Train,
+ *  FindBestSplits, Split) — algorithm re-implemented fresh. */
+#include "migbm/tree_learner.h"
+#include "migbm/objective.h"
+
+#include <algorithm>
+#include <numeric>
+
+namespace migbm {
+
+// ------------------------------------------------------------------ DataPartition
+void DataPartition::Split(int leaf, int right_leaf,
+                          const std::function<bool(data_size_t)>& go_left) {
+  const data_size_t begin = leaf_begin_[leaf];
+  const data_size_t cnt = leaf_count_[leaf];
+  data_size_t* idx = indices_.data() + begin;
+  data_size_t* tmp = temp_.data() + begin;
+
+  const int max_threads = omp_get_max_threads();
+  int nblock;
+  data_size_t bsize;
+  Threading::BlockInfo<data_size_t>(max_threads, cnt, 1024, &nblock, &bsize);
+  std::vector<data_size_t> left_cnts(nblock, 0);
+#pragma omp parallel for schedule(static, 1)
+  for (int b = 0; b < nblock; ++b) {
+    data_size_t s = b * bsize, e = std::min(cnt, s + bsize);
+    data_size_t lc = 0;
+    for (data_size_t i = s; i < e; ++i) lc += go_left(idx[i]) ? 1 : 0;
+    left_cnts[b] = lc;
+  }
+  std::vector<data_size_t> loff(nblock + 1, 0), roff(nblock + 1, 0);
+  for (int b = 0; b < nblock; ++b) loff[b + 1] = loff[b] + left_cnts[b];
+  const data_size_t total_left = loff[nblock];
+  roff[0] = total_left;
+  for (int b = 0; b < nblock; ++b) {
+    data_size_t s = b * bsize, e = std::min(cnt, s + bsize);
+    roff[b + 1] = roff[b] + (e - s) - left_cnts[b];
+  }
+#pragma omp parallel for schedule(static, 1)
+  for (int b = 0; b < nblock; ++b) {
+    data_size_t s = b * bsize, e = std::min(cnt, s + bsize);
+    data_size_t lp = loff[b], rp = roff[b];
+    for (data_size_t i = s; i < e; ++i) {
+      if (go_left(idx[i])) tmp[lp++] = idx[i];
+      else tmp[rp++] = idx[i];
+    }
+  }
+  std::copy(tmp, tmp + cnt, idx);
+  leaf_count_[leaf] = total_left;
+  leaf_begin_[right_leaf] = begin + total_left;
+  leaf_count_[right_leaf] = cnt - total_left;
+}
+
+// ------------------------------------------------------------------ SerialTreeLearner
+void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
+  train_data_ = train_data;
+  is_constant_hessian_ = is_constant_hessian;
+  const data_size_t n = train_data_->num_data();
+  partition_.Init(n, config_->num_leaves);
+  ordered_grad_.resize(n);
+  ordered_hess_.resize(n);
+  hist_store_.resize(static_cast<size_t>(config_->num_leaves) * 2 *
+                     train_data_->num_total_bin());
+  leaf_to_slot_.resize(config_->num_leaves);
+  best_split_per_leaf_.resize(config_->num_leaves);
+  leaf_ctx_.resize(config_->num_leaves);
+  feature_rng_ = Random(config_->feature_fraction_seed);
+  extra_rng_ = Random(config_->extra_seed);
+}
+
+void SerialTreeLearner::ResetTrainingData(const Dataset* train_data) {
+  Init(train_data, is_constant_hessian_);
+}
+
+void SerialTreeLearner::SetBaggingData(const Dataset* subset, const data_size_t* used_indices,
+                                       data_size_t num_data) {
+  if (subset != nullptr) {
+    ResetTrainingData(subset);
+    bag_indices_ = nullptr;
+    bag_cnt_ = 0;
+  } else {
+    bag_indices_ = used_indices;
+    bag_cnt_ = num_data;
+  }
+}
+
+std::vector<int8_t> SerialTreeLearner::SampleFeatures(bool per_node) {
+  const int nf = train_data_->num_features();
+  std::vector<int8_t> used(nf, 1);
+  double frac = per_node ? config_->feature_fraction_bynode : config_->feature_fraction;
+  if (frac >= 1.0) return used;
+  int k = std::max(1, static_cast<int>(nf * frac));
+  std::fill(used.begin(), used.end(), 0);
+  auto sel = feature_rng_.Sample(nf, k);
+  for (int f : sel) used[f] = 1;
+  return used;
+}
+
+void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
+                                         const data_size_t* indices) {
+  // gather ordered gradients
+#pragma omp parallel for schedule(static)
+  for (data_size_t i = 0; i < cnt; ++i) {
+    ordered_grad_[i] = gradients_[indices[i]];
+    ordered_hess_[i] = hessians_[indices[i]];
+  }
+  hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
+  std::fill(hist, hist + 2 * train_data_->num_total_bin(), 0.0);
+  train_data_->ConstructHistograms(is_feature_used_, indices, cnt, ordered_grad_.data(),
+                                   ordered_hess_.data(), hist);
+}
+
+void SerialTreeLearner::SubtractHistogram(int /*dst_leaf*/, int parent_slot, int sibling_slot) {
+  hist_t* parent = HistSlot(parent_slot);
+  const hist_t* sib = HistSlot(sibling_slot);
+  const size_t n = 2 * static_cast<size_t>(train_data_->num_total_bin());
+#pragma omp parallel for schedule(static)
+  for (int64_t i = 0; i < static_cast<int64_t>(n); ++i) parent[i] -= sib[i];
+}
+
+void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
+  SplitInfo& best = best_split_per_leaf_[leaf];
+  best.Reset();
+  if (config_->max_depth > 0 && ctx.depth >= config_->max_depth) return;
+  if (ctx.num_data < 2 * config_->min_data_in_leaf) return;
+  const hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
+  const int nf = train_data_->num_features();
+  std::vector<int8_t> node_mask;
+  const std::vector<int8_t>* mask = &is_feature_used_;
+  if (config_->feature_fraction_bynode < 1.0) {
+    node_mask = SampleFeatures(true);
+    for (int f = 0; f < nf; ++f) node_mask[f] &= is_feature_used_[f];
+    mask = &node_mask;
+  }
+  std::vector<SplitInfo> cand(nf);
+#pragma omp parallel for schedule(static)
+  for (int f = 0; f < nf; ++f) {
+    cand[f].Reset();
+    if (!(*mask)[f]) continue;
+    const BinMapper* m = train_data_->FeatureBinMapper(f);
+    const hist_t* fh = hist + 2 * train_data_->hist_offset(f);
+    if (m->bin_type() == BinType::kCategorical) {
+      FindBestThresholdCategorical(fh, m->num_bin(), ctx, *config_, &cand[f]);
+    } else {
+      int rand_t = -1;
+      if (config_->extra_trees) {
+        // extra-trees: a single random threshold per feature per node
+        rand_t = extra_rng_.NextInt(0, std::max(1, m->num_numeric_bin() - 1));
+      }
+      int8_t mono = 0;
+      if (!config_->monotone_constraints.empty()) {
+        int orig = train_data_->RealFeatureIndex(f);
+        if (orig < static_cast<int>(config_->monotone_constraints.size()))
+          mono = static_cast<int8_t>(config_->monotone_constraints[orig]);
+      }
+      FindBestThresholdNumerical(fh, m->num_bin(), m->num_numeric_bin(), m->nan_bin(), ctx,
+                                 *config_, mono, rand_t, &cand[f]);
+    }
+    cand[f].feature = f;
+  }
+  for (int f = 0; f < nf; ++f) {
+    if (cand[f].IsValid() && cand[f] > best) best = cand[f];
+  }
+}
+
+std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& s) const {
+  const int f = s.feature;
+  const BinColumn& col = train_data_->column(f);
+  const BinMapper* m = train_data_->FeatureBinMapper(f);
+  if (!s.cat_bitset_inner.empty()) {
+    // categorical
+    std::vector<uint32_t> bits = s.cat_bitset_inner;
+    const int nwords = static_cast<int>(bits.size());
+    return [&col, bits, nwords](data_size_t row) {
+      uint32_t b = col.Get(row);
+      return (b >> 5) < static_cast<uint32_t>(nwords) && ((bits[b >> 5] >> (b & 31)) & 1);
+    };
+  }
+  const uint32_t thr = s.threshold;
+  const int nanb = m->nan_bin();
+  const bool default_left = s.default_left;
+  if (nanb < 0) {
+    return [&col, thr](data_size_t row) { return col.Get(row) <= thr; };
+  }
+  const uint32_t nb = static_cast<uint32_t>(nanb);
+  return [&col, thr, nb, default_left](data_size_t row) {
+    uint32_t b = col.Get(row);
+    if (b == nb) return default_left;
+    return b <= thr;
+  };
+}
+
+Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians,
+                               bool /*is_first_tree*/) {
+  gradients_ = gradients;
+  hessians_ = hessians;
+  ++iter_counter_;
+  is_feature_used_ = SampleFeatures(false);
+
+  auto tree = std::make_unique<Tree>(config_->num_leaves);
+  partition_.ResetToRoot(bag_indices_, bag_cnt_);
+  for (int i = 0; i < config_->num_leaves; ++i) leaf_to_slot_[i] = i;
+
+  // root stats
+  data_size_t root_cnt;
+  const data_size_t* root_idx = partition_.GetIndexOnLeaf(0, &root_cnt);
+  double sum_g = 0.0, sum_h = 0.0;
+#pragma omp parallel for schedule(static) reduction(+ : sum_g, sum_h)
+  for (data_size_t i = 0; i < root_cnt; ++i) {
+    sum_g += gradients_[root_idx[i]];
+    sum_h += hessians_[root_idx[i]];
+  }
+  data_size_t global_root_cnt = root_cnt;
+  ReduceRootStats(&sum_g, &sum_h, &global_root_cnt);
+  leaf_ctx_[0] = {sum_g, sum_h, global_root_cnt,
+                  GainMath::CalculateSplittedLeafOutput(sum_g, sum_h, config_->lambda_l1,
+                                                        config_->lambda_l2,
+                                                        config_->max_delta_step), 0};
+  ComputeHistogram(0, root_cnt, root_idx);
+  OnHistogramReady(0);
+  FindBestSplitForLeaf(0, leaf_ctx_[0]);
+
+  int num_leaves = 1;
+  for (int split_i = 0; split_i < config_->num_leaves - 1; ++split_i) {
+    // pick best leaf
+    int best_leaf = -1;
+    double best_gain = 0.0;
+    for (int l = 0; l < num_leaves; ++l) {
+      if (best_split_per_leaf_[l].IsValid() && best_split_per_leaf_[l].gain > best_gain) {
+        best_gain = best_split_per_leaf_[l].gain;
+        best_leaf = l;
+      }
+    }
+    if (best_leaf < 0) break;
+    SplitInfo& s = best_split_per_leaf_[best_leaf];
+    const int f = s.feature;
+    const int orig_f = train_data_->RealFeatureIndex(f);
+    const BinMapper* m = train_data_->FeatureBinMapper(f);
+    const int right_leaf = num_leaves;
+
+    // tree structure update
+    if (!s.cat_bitset_inner.empty()) {
+      // map bin-level bitset to category-value bitset for prediction on raw values
+      std::vector<uint32_t> cat_bits;
+      for (int w = 0; w < static_cast<int>(s.cat_bitset_inner.size()); ++w) {
+        uint32_t word = s.cat_bitset_inner[w];
+        while (word) {
+          int bit = __builtin_ctz(word);
+          word &= word - 1;
+          int bin = w * 32 + bit;
+          int cat = static_cast<int>(m->BinToValue(bin));
+          if (cat >= 0) {
+            if ((cat >> 5) >= static_cast<int>(cat_bits.size())) cat_bits.resize((cat >> 5) + 1, 0);
+            cat_bits[cat >> 5] |= 1u << (cat & 31);
+          }
+        }
+      }
+      tree->SplitCategorical(best_leaf, f, orig_f, cat_bits.data(),
+                             static_cast<int>(cat_bits.size()), s.left_output, s.right_output,
+                             s.left_count, s.right_count, s.left_sum_hessian,
+                             s.right_sum_hessian, static_cast<float>(s.gain),
+                             m->missing_type());
+      // store bin-level bitset on the tree node for training-time partition? partition uses s directly
+    } else {
+      tree->Split(best_leaf, f, orig_f, s.threshold,
+                  train_data_->RealThreshold(f, s.threshold), s.left_output, s.right_output,
+                  s.left_count, s.right_count, s.left_sum_hessian, s.right_sum_hessian,
+                  static_cast<float>(s.gain), m->missing_type(), s.default_left);
+    }
+
+    // partition rows
+    partition_.Split(best_leaf, right_leaf, MakeGoLeft(s));
+    data_size_t left_cnt_actual = partition_.leaf_count(best_leaf);
+    data_size_t right_cnt_actual = partition_.leaf_count(right_leaf);
+    GlobalChildCounts(&left_cnt_actual, &right_cnt_actual);
+
+    // child contexts
+    const double parent_out = (s.left_output * s.left_sum_hessian +
+                               s.right_output * s.right_sum_hessian) /
+                              std::max(s.left_sum_hessian + s.right_sum_hessian, kEpsilon);
+    int parent_depth = leaf_ctx_[best_leaf].depth;
+    leaf_ctx_[best_leaf] = {s.left_sum_gradient, s.left_sum_hessian, left_cnt_actual,
+                            parent_out, parent_depth + 1};
+    leaf_ctx_[right_leaf] = {s.right_sum_gradient, s.right_sum_hessian, right_cnt_actual,
+                             parent_out, parent_depth + 1};
+    ++num_leaves;
+
+    // histograms: build smaller child, subtract for larger.
+    // parent hist currently lives in slot leaf_to_slot_[best_leaf].
+    int parent_slot = leaf_to_slot_[best_leaf];
+    int spare_slot = right_leaf;  // unused slot id == right leaf index (fresh)
+    bool left_smaller = left_cnt_actual <= right_cnt_actual;
+    int small_leaf = left_smaller ? best_leaf : right_leaf;
+    int large_leaf = left_smaller ? right_leaf : best_leaf;
+    leaf_to_slot_[small_leaf] = spare_slot;
+    leaf_to_slot_[large_leaf] = parent_slot;
+    data_size_t small_cnt;
+    const data_size_t* small_idx = partition_.GetIndexOnLeaf(small_leaf, &small_cnt);
+    ComputeHistogram(small_leaf, small_cnt, small_idx);
+    OnHistogramReady(small_leaf);
+    // in-place: parent_slot -= small_slot -> becomes large hist
+    SubtractHistogram(large_leaf, parent_slot, spare_slot);
+
+    FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
+    FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);
+  }
+  return tree.release();
+}
+
+void SerialTreeLearner::AddPredictionToScore(const Tree* tree, double* out_score) {
+  if (bag_indices_ != nullptr) {
+    tree->AddPredictionToScore(train_data_, bag_indices_, bag_cnt_, out_score);
+  } else {
+    tree->AddPredictionToScore(train_data_, train_data_->num_data(), out_score);
+  }
+}
+
+void SerialTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
+                                        std::function<double(const label_t*, int)>,
+                                        data_size_t, const data_size_t*, data_size_t,
+                                        const double* train_score) {
+  if (obj == nullptr || !obj->NeedRenewTreeOutput()) return;
+  // generic renewal: per leaf, objective-specific output from the rows in the leaf
+  const int nl = tree->num_leaves();
+  data_size_t cnt;
+  for (int l = 0; l < nl; ++l) {
+    const data_size_t* idx = partition_.GetIndexOnLeaf(l, &cnt);
+    if (cnt == 0) continue;
+    double new_out = obj->RenewTreeOutput(tree->LeafOutput(l), idx, cnt, train_score);
+    tree->SetLeafOutput(l, new_out);
+  }
+}
+
+// ------------------------------------------------------------------ base fallbacks
+Tree* TreeLearner::FitByExistingTree(const Tree*, const score_t*, const score_t*) {
+  Log::Fatal("FitByExistingTree not supported by this learner");
+  return nullptr;
+}
+Tree* TreeLearner::FitByExistingTree(const Tree* t, const std::vector<int>&, const score_t* g,
+                                     const score_t* h) {
+  return FitByExistingTree(t, g, h);
+}
+
+}  // namespace migbm

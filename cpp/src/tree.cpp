@@ -1,0 +1,266 @@
+/*! migbm Tree implementation: split recording, model-text v4 serialization, JSON dump,
+ *  parsing. Format parity target: reference src/io/tree.cpp:343-500 (field names/order). */
+#include "migbm/tree.h"
+#include "migbm/dataset.h"
+
+#include <map>
+
+namespace migbm {
+
+Tree::Tree(int max_leaves, bool /*track_branch_features*/, bool is_linear)
+    : max_leaves_(max_leaves), num_leaves_(1), is_linear_(is_linear) {
+  const int m = max_leaves_;
+  left_child_.resize(m - 1);
+  right_child_.resize(m - 1);
+  split_feature_inner_.resize(m - 1);
+  split_feature_.resize(m - 1);
+  threshold_in_bin_.resize(m - 1);
+  threshold_.resize(m - 1);
+  decision_type_.assign(m - 1, 0);
+  split_gain_.resize(m - 1);
+  internal_value_.assign(m - 1, 0.0);
+  internal_weight_.assign(m - 1, 0.0);
+  internal_count_.assign(m - 1, 0);
+  leaf_value_.assign(m, 0.0);
+  leaf_weight_.assign(m, 0.0);
+  leaf_count_.assign(m, 0);
+  leaf_depth_.assign(m, 0);
+  leaf_parent_.assign(m, -1);
+  cat_boundaries_.push_back(0);
+}
+
+void Tree::RecordSplit(int leaf, int new_node, int feature, int real_feature,
+                       double left_value, double right_value, int left_cnt, int right_cnt,
+                       double left_weight, double right_weight, float gain) {
+  split_feature_inner_[new_node] = feature;
+  split_feature_[new_node] = real_feature;
+  split_gain_[new_node] = gain;
+  // hook up parent
+  int parent = leaf_parent_[leaf];
+  if (parent >= 0) {
+    if (left_child_[parent] == ~leaf) left_child_[parent] = new_node;
+    else right_child_[parent] = new_node;
+  }
+  // internal stats = combination of children
+  internal_value_[new_node] = (left_weight * left_value + right_weight * right_value) /
+                              std::max(left_weight + right_weight, kEpsilon);
+  internal_weight_[new_node] = left_weight + right_weight;
+  internal_count_[new_node] = left_cnt + right_cnt;
+  left_child_[new_node] = ~leaf;
+  right_child_[new_node] = ~num_leaves_;
+  leaf_parent_[leaf] = new_node;
+  leaf_parent_[num_leaves_] = new_node;
+  leaf_value_[leaf] = std::isnan(left_value) ? 0.0 : left_value;
+  leaf_value_[num_leaves_] = std::isnan(right_value) ? 0.0 : right_value;
+  leaf_weight_[leaf] = left_weight;
+  leaf_weight_[num_leaves_] = right_weight;
+  leaf_count_[leaf] = left_cnt;
+  leaf_count_[num_leaves_] = right_cnt;
+  leaf_depth_[num_leaves_] = leaf_depth_[leaf] + 1;
+  leaf_depth_[leaf] += 1;
+}
+
+int Tree::Split(int leaf, int feature, int real_feature, uint32_t threshold_bin,
+                double threshold_double, double left_value, double right_value, int left_cnt,
+                int right_cnt, double left_weight, double right_weight, float gain,
+                MissingType missing_type, bool default_left) {
+  const int new_node = num_leaves_ - 1;
+  decision_type_[new_node] = 0;
+  if (default_left) decision_type_[new_node] |= kDefaultLeftMask;
+  decision_type_[new_node] |= static_cast<int8_t>(static_cast<int>(missing_type) << 2);
+  threshold_in_bin_[new_node] = threshold_bin;
+  threshold_[new_node] = threshold_double;
+  RecordSplit(leaf, new_node, feature, real_feature, left_value, right_value, left_cnt,
+              right_cnt, left_weight, right_weight, gain);
+  ++num_leaves_;
+  return num_leaves_ - 1;
+}
+
+int Tree::SplitCategorical(int leaf, int feature, int real_feature,
+                           const uint32_t* threshold_bitset, int n_words, double left_value,
+                           double right_value, int left_cnt, int right_cnt, double left_weight,
+                           double right_weight, float gain, MissingType missing_type) {
+  const int new_node = num_leaves_ - 1;
+  decision_type_[new_node] = kCategoricalMask;
+  decision_type_[new_node] |= static_cast<int8_t>(static_cast<int>(missing_type) << 2);
+  threshold_in_bin_[new_node] = static_cast<uint32_t>(num_cat_);
+  threshold_[new_node] = static_cast<double>(num_cat_);
+  cat_boundaries_.push_back(cat_boundaries_.back() + n_words);
+  for (int i = 0; i < n_words; ++i) cat_threshold_.push_back(threshold_bitset[i]);
+  ++num_cat_;
+  RecordSplit(leaf, new_node, feature, real_feature, left_value, right_value, left_cnt,
+              right_cnt, left_weight, right_weight, gain);
+  ++num_leaves_;
+  return num_leaves_ - 1;
+}
+
+double Tree::GetUpperBoundValue() const {
+  double mx = leaf_value_[0];
+  for (int i = 1; i < num_leaves_; ++i) mx = std::max(mx, leaf_value_[i]);
+  return mx;
+}
+double Tree::GetLowerBoundValue() const {
+  double mn = leaf_value_[0];
+  for (int i = 1; i < num_leaves_; ++i) mn = std::min(mn, leaf_value_[i]);
+  return mn;
+}
+
+std::string Tree::ToString() const {
+  std::stringstream ss;
+  ss.precision(17);
+  const int ni = num_leaves_ - 1;
+  ss << "num_leaves=" << num_leaves_ << '\n';
+  ss << "num_cat=" << num_cat_ << '\n';
+  ss << "split_feature=" << Common::ArrayToString(split_feature_.data(), ni) << '\n';
+  ss << "split_gain=" << Common::ArrayToString(split_gain_.data(), ni) << '\n';
+  ss << "threshold=" << Common::ArrayToString(threshold_.data(), ni) << '\n';
+  {
+    // decision_type serialized as ints
+    std::vector<int> dt(ni);
+    for (int i = 0; i < ni; ++i) dt[i] = decision_type_[i];
+    ss << "decision_type=" << Common::ArrayToString(dt.data(), ni) << '\n';
+  }
+  ss << "left_child=" << Common::ArrayToString(left_child_.data(), ni) << '\n';
+  ss << "right_child=" << Common::ArrayToString(right_child_.data(), ni) << '\n';
+  ss << "leaf_value=" << Common::ArrayToString(leaf_value_.data(), num_leaves_) << '\n';
+  ss << "leaf_weight=" << Common::ArrayToString(leaf_weight_.data(), num_leaves_) << '\n';
+  ss << "leaf_count=" << Common::ArrayToString(leaf_count_.data(), num_leaves_) << '\n';
+  ss << "internal_value=" << Common::ArrayToString(internal_value_.data(), ni) << '\n';
+  ss << "internal_weight=" << Common::ArrayToString(internal_weight_.data(), ni) << '\n';
+  ss << "internal_count=" << Common::ArrayToString(internal_count_.data(), ni) << '\n';
+  if (num_cat_ > 0) {
+    ss << "cat_boundaries=" << Common::ArrayToString(cat_boundaries_.data(), cat_boundaries_.size()) << '\n';
+    ss << "cat_threshold=" << Common::ArrayToString(cat_threshold_.data(), cat_threshold_.size()) << '\n';
+  }
+  ss << "is_linear=" << (is_linear_ ? 1 : 0) << '\n';
+  ss << "shrinkage=" << Common::DoubleToStr(shrinkage_) << '\n';
+  return ss.str();
+}
+
+namespace {
+std::map<std::string, std::string> ParseKV(const char* str, size_t* used_len) {
+  std::map<std::string, std::string> kv;
+  const char* p = str;
+  size_t consumed = 0;
+  while (*p) {
+    const char* eol = strchr(p, '\n');
+    size_t len = eol ? static_cast<size_t>(eol - p) : strlen(p);
+    std::string line(p, len);
+    line = Common::Trim(line);
+    if (line.empty()) {
+      consumed = (eol ? (eol - str) + 1 : strlen(str));
+      if (!kv.empty()) break;  // blank line ends the tree block (after content started)
+      p = eol ? eol + 1 : p + len;
+      continue;
+    }
+    if (Common::StartsWith(line, "Tree=")) {
+      if (!kv.empty()) break;
+      p = eol ? eol + 1 : p + len;
+      consumed = p - str;
+      continue;
+    }
+    if (Common::StartsWith(line, "end of trees")) break;
+    auto eq = line.find('=');
+    if (eq != std::string::npos) kv[line.substr(0, eq)] = line.substr(eq + 1);
+    p = eol ? eol + 1 : p + len;
+    consumed = p - str;
+    if (!eol) break;
+  }
+  if (used_len) *used_len = consumed;
+  return kv;
+}
+}  // namespace
+
+Tree::Tree(const char* str, size_t* used_len) {
+  auto kv = ParseKV(str, used_len);
+  auto get = [&](const char* k) -> const std::string& {
+    static const std::string empty;
+    auto it = kv.find(k);
+    return it == kv.end() ? empty : it->second;
+  };
+  num_leaves_ = atoi(get("num_leaves").c_str());
+  num_cat_ = atoi(get("num_cat").c_str());
+  is_linear_ = atoi(get("is_linear").c_str()) != 0;
+  shrinkage_ = Common::Atof(get("shrinkage").c_str());
+  if (shrinkage_ == 0.0) shrinkage_ = 1.0;
+  max_leaves_ = std::max(num_leaves_, 1);
+  leaf_value_.assign(max_leaves_, 0.0);
+  Common::StringToArray<double>(get("leaf_value"), ' ', &leaf_value_);
+  if (num_leaves_ <= 1) { cat_boundaries_.push_back(0); return; }
+  Common::StringToArray<int>(get("split_feature"), ' ', &split_feature_);
+  split_feature_inner_ = split_feature_;  // standalone model: inner == real
+  Common::StringToArray<float>(get("split_gain"), ' ', &split_gain_);
+  Common::StringToArray<double>(get("threshold"), ' ', &threshold_);
+  {
+    std::vector<int> dt;
+    Common::StringToArray<int>(get("decision_type"), ' ', &dt);
+    decision_type_.resize(dt.size());
+    for (size_t i = 0; i < dt.size(); ++i) decision_type_[i] = static_cast<int8_t>(dt[i]);
+  }
+  Common::StringToArray<int>(get("left_child"), ' ', &left_child_);
+  Common::StringToArray<int>(get("right_child"), ' ', &right_child_);
+  Common::StringToArray<double>(get("leaf_weight"), ' ', &leaf_weight_);
+  Common::StringToArray<int>(get("leaf_count"), ' ', &leaf_count_);
+  Common::StringToArray<double>(get("internal_value"), ' ', &internal_value_);
+  Common::StringToArray<double>(get("internal_weight"), ' ', &internal_weight_);
+  Common::StringToArray<int>(get("internal_count"), ' ', &internal_count_);
+  if (num_cat_ > 0) {
+    Common::StringToArray<int>(get("cat_boundaries"), ' ', &cat_boundaries_);
+    Common::StringToArray<uint32_t>(get("cat_threshold"), ' ', &cat_threshold_);
+  } else {
+    cat_boundaries_.push_back(0);
+  }
+  threshold_in_bin_.assign(num_leaves_ - 1, 0);
+  leaf_depth_.assign(num_leaves_, 0);
+  leaf_parent_.assign(num_leaves_, -1);
+  leaf_weight_.resize(num_leaves_, 0.0);
+  leaf_count_.resize(num_leaves_, 0);
+  internal_value_.resize(num_leaves_ - 1, 0.0);
+  internal_weight_.resize(num_leaves_ - 1, 0.0);
+  internal_count_.resize(num_leaves_ - 1, 0);
+  split_gain_.resize(num_leaves_ - 1, 0.0f);
+}
+
+std::string Tree::ToJSON() const {
+  std::stringstream ss;
+  ss.precision(17);
+  ss << "{";
+  ss << "\"num_leaves\":" << num_leaves_ << ",";
+  ss << "\"num_cat\":" << num_cat_ << ",";
+  ss << "\"shrinkage\":" << shrinkage_ << ",";
+  ss << "\"tree_structure\":";
+  // recursive node dump
+  std::function<void(int)> dump = [&](int node) {
+    if (node >= 0) {
+      ss << "{\"split_index\":" << node
+         << ",\"split_feature\":" << split_feature_[node]
+         << ",\"split_gain\":" << split_gain_[node]
+         << ",\"threshold\":" << Common::DoubleToStr(threshold_[node])
+         << ",\"decision_type\":\"" << (IsCategoricalSplit(node) ? "==" : "<=") << "\""
+         << ",\"default_left\":" << ((decision_type_[node] & kDefaultLeftMask) ? "true" : "false")
+         << ",\"missing_type\":\"";
+      int mt = (decision_type_[node] >> 2) & 3;
+      ss << (mt == 0 ? "None" : (mt == 1 ? "Zero" : "NaN")) << "\""
+         << ",\"internal_value\":" << internal_value_[node]
+         << ",\"internal_weight\":" << internal_weight_[node]
+         << ",\"internal_count\":" << internal_count_[node]
+         << ",\"left_child\":";
+      dump(left_child_[node]);
+      ss << ",\"right_child\":";
+      dump(right_child_[node]);
+      ss << "}";
+    } else {
+      int leaf = ~node;
+      ss << "{\"leaf_index\":" << leaf
+         << ",\"leaf_value\":" << Common::DoubleToStr(leaf_value_[leaf])
+         << ",\"leaf_weight\":" << leaf_weight_[leaf]
+         << ",\"leaf_count\":" << leaf_count_[leaf] << "}";
+    }
+  };
+  if (num_leaves_ > 1) dump(0);
+  else dump(~0);
+  ss << "}";
+  return ss.str();
+}
+
+}  // namespace migbm

@@ -1,0 +1,669 @@
+"""ctypes bindings: Dataset / Booster over the LGBM_* C ABI of lib_migbm.so.
+
+Capability parity target: reference python-package/lightgbm/basic.py (Dataset lazy
+construction from numpy/pandas/scipy/file, Booster train/predict/save). Fresh
+implementation sized to the implemented C API surface.
+"""
+import ctypes
+import json
+import os
+from pathlib import Path
+
+import numpy as np
+
+from .compat import PANDAS_INSTALLED, SCIPY_INSTALLED, pd_DataFrame, pd_Series, scipy_sparse
+from .libpath import find_lib_path
+
+__all__ = ["Dataset", "Booster", "LightGBMError", "register_logger"]
+
+_DTYPE_F32, _DTYPE_F64, _DTYPE_I32, _DTYPE_I64 = 0, 1, 2, 3
+_PREDICT_NORMAL, _PREDICT_RAW, _PREDICT_LEAF, _PREDICT_CONTRIB = 0, 1, 2, 3
+
+
+class LightGBMError(Exception):
+    """Error from the native library."""
+
+
+def _load_lib():
+    lib = ctypes.cdll.LoadLibrary(find_lib_path()[0])
+    lib.LGBM_GetLastError.restype = ctypes.c_char_p
+    return lib
+
+
+_LIB = _load_lib()
+
+_LOG_CALLBACK = None
+
+
+def register_logger(logger, info_method_name="info", warning_method_name="warning"):
+    """Redirect native logging into a Python logger."""
+    global _LOG_CALLBACK
+    cb_type = ctypes.CFUNCTYPE(None, ctypes.c_char_p)
+
+    def _cb(msg):
+        text = msg.decode("utf-8", errors="replace").rstrip("\n")
+        if "[Warning]" in text:
+            getattr(logger, warning_method_name)(text)
+        else:
+            getattr(logger, info_method_name)(text)
+
+    _LOG_CALLBACK = cb_type(_cb)  # keep alive
+    _safe_call(_LIB.LGBM_RegisterLogCallback(_LOG_CALLBACK))
+
+
+def _safe_call(ret):
+    if ret != 0:
+        raise LightGBMError(_LIB.LGBM_GetLastError().decode("utf-8"))
+
+
+def _c_str(string):
+    return ctypes.c_char_p(str(string).encode("utf-8"))
+
+
+def _param_dict_to_str(params):
+    if not params:
+        return ""
+    pairs = []
+    for k, v in params.items():
+        if isinstance(v, (list, tuple, set)):
+            pairs.append(f"{k}={','.join(map(str, v))}")
+        elif isinstance(v, bool):
+            pairs.append(f"{k}={'true' if v else 'false'}")
+        elif v is None:
+            continue
+        else:
+            pairs.append(f"{k}={v}")
+    return " ".join(pairs)
+
+
+def _to_2d_float64(data):
+    if PANDAS_INSTALLED and isinstance(data, pd_DataFrame):
+        data = data.values
+    arr = np.asarray(data)
+    if arr.ndim == 1:
+        arr = arr.reshape(1, -1)
+    return np.ascontiguousarray(arr, dtype=np.float64)
+
+
+def _np_float32(data):
+    if PANDAS_INSTALLED and isinstance(data, (pd_Series, pd_DataFrame)):
+        data = data.values
+    return np.ascontiguousarray(np.asarray(data).ravel(), dtype=np.float32)
+
+
+class Dataset:
+    """Binned training dataset (parity: reference lgb.Dataset)."""
+
+    def __init__(self, data, label=None, reference=None, weight=None, group=None,
+                 init_score=None, feature_name="auto", categorical_feature="auto",
+                 params=None, free_raw_data=True, position=None):
+        self.data = data
+        self.label = label
+        self.reference = reference
+        self.weight = weight
+        self.group = group
+        self.init_score = init_score
+        self.position = position
+        self.feature_name = feature_name
+        self.categorical_feature = categorical_feature
+        self.params = dict(params) if params else {}
+        self.free_raw_data = free_raw_data
+        self._handle = None
+        self.used_indices = None
+        self._predictor = None
+
+    # ------------------------------------------------------------ construction
+    @property
+    def handle(self):
+        return self._handle
+
+    def construct(self):
+        if self._handle is not None:
+            return self
+        if self.reference is not None:
+            self.reference.construct()
+        params = dict(self.params)
+        cat = self._resolve_categorical(params)
+        if cat:
+            params["categorical_feature"] = ",".join(str(c) for c in cat)
+        param_str = _param_dict_to_str(params)
+        ref_handle = self.reference._handle if self.reference is not None else None
+
+        if isinstance(self.data, (str, Path)):
+            out = ctypes.c_void_p()
+            _safe_call(_LIB.LGBM_DatasetCreateFromFile(
+                _c_str(str(self.data)), _c_str(param_str), ref_handle, ctypes.byref(out)))
+            self._handle = out
+        elif SCIPY_INSTALLED and scipy_sparse is not None and scipy_sparse.issparse(self.data):
+            csr = self.data.tocsr()
+            out = ctypes.c_void_p()
+            indptr = np.ascontiguousarray(csr.indptr, dtype=np.int32)
+            indices = np.ascontiguousarray(csr.indices, dtype=np.int32)
+            values = np.ascontiguousarray(csr.data, dtype=np.float64)
+            _safe_call(_LIB.LGBM_DatasetCreateFromCSR(
+                indptr.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_I32),
+                indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+                values.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
+                ctypes.c_int64(len(indptr)), ctypes.c_int64(len(values)),
+                ctypes.c_int64(csr.shape[1]), _c_str(param_str), ref_handle,
+                ctypes.byref(out)))
+            self._handle = out
+        else:
+            arr = _to_2d_float64(self.data)
+            out = ctypes.c_void_p()
+            _safe_call(_LIB.LGBM_DatasetCreateFromMat(
+                arr.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
+                ctypes.c_int32(arr.shape[0]), ctypes.c_int32(arr.shape[1]),
+                ctypes.c_int(1), _c_str(param_str), ref_handle, ctypes.byref(out)))
+            self._handle = out
+
+        # fields
+        if self.label is not None:
+            self.set_label(self.label)
+        if self.weight is not None:
+            self.set_weight(self.weight)
+        if self.group is not None:
+            self.set_group(self.group)
+        if self.init_score is not None:
+            self.set_init_score(self.init_score)
+        if self.position is not None:
+            self.set_position(self.position)
+        # feature names
+        names = self._resolve_feature_names()
+        if names is not None:
+            arr_names = (ctypes.c_char_p * len(names))(*[n.encode("utf-8") for n in names])
+            _safe_call(_LIB.LGBM_DatasetSetFeatureNames(
+                self._handle, arr_names, ctypes.c_int(len(names))))
+        if self.free_raw_data:
+            self.data = None
+        return self
+
+    def _resolve_feature_names(self):
+        if isinstance(self.feature_name, (list, tuple)):
+            return list(self.feature_name)
+        if self.feature_name == "auto" and PANDAS_INSTALLED and isinstance(self.data, pd_DataFrame):
+            return [str(c) for c in self.data.columns]
+        return None
+
+    def _resolve_categorical(self, params):
+        cf = self.categorical_feature
+        if cf is None or cf == "auto":
+            if PANDAS_INSTALLED and isinstance(self.data, pd_DataFrame):
+                cats = [i for i, d in enumerate(self.data.dtypes)
+                        if str(d) in ("category", "object")]
+                return cats
+            return []
+        out = []
+        names = self._resolve_feature_names()
+        for c in cf:
+            if isinstance(c, str):
+                if names and c in names:
+                    out.append(names.index(c))
+            else:
+                out.append(int(c))
+        return out
+
+    def __del__(self):
+        try:
+            if self._handle is not None:
+                _safe_call(_LIB.LGBM_DatasetFree(self._handle))
+                self._handle = None
+        except Exception:
+            pass
+
+    # ------------------------------------------------------------ fields
+    def _set_float_field(self, name, data):
+        self.construct()
+        arr = _np_float32(data)
+        _safe_call(_LIB.LGBM_DatasetSetField(
+            self._handle, _c_str(name), arr.ctypes.data_as(ctypes.c_void_p),
+            ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_F32)))
+
+    def set_label(self, label):
+        self.label = label
+        if self._handle is not None:
+            self._set_float_field("label", label)
+        return self
+
+    def set_weight(self, weight):
+        self.weight = weight
+        if self._handle is not None and weight is not None:
+            self._set_float_field("weight", weight)
+        return self
+
+    def set_group(self, group):
+        self.group = group
+        if self._handle is not None and group is not None:
+            arr = np.ascontiguousarray(np.asarray(group).ravel(), dtype=np.int32)
+            _safe_call(_LIB.LGBM_DatasetSetField(
+                self._handle, _c_str("group"), arr.ctypes.data_as(ctypes.c_void_p),
+                ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_I32)))
+        return self
+
+    def set_position(self, position):
+        self.position = position
+        if self._handle is not None and position is not None:
+            arr = np.ascontiguousarray(np.asarray(position).ravel(), dtype=np.int32)
+            _safe_call(_LIB.LGBM_DatasetSetField(
+                self._handle, _c_str("position"), arr.ctypes.data_as(ctypes.c_void_p),
+                ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_I32)))
+        return self
+
+    def set_init_score(self, init_score):
+        self.init_score = init_score
+        if self._handle is not None and init_score is not None:
+            arr = np.ascontiguousarray(np.asarray(init_score).ravel(), dtype=np.float64)
+            _safe_call(_LIB.LGBM_DatasetSetField(
+                self._handle, _c_str("init_score"), arr.ctypes.data_as(ctypes.c_void_p),
+                ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_F64)))
+        return self
+
+    def get_field(self, name):
+        self.construct()
+        out_len = ctypes.c_int(0)
+        out_ptr = ctypes.c_void_p()
+        out_type = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_DatasetGetField(
+            self._handle, _c_str(name), ctypes.byref(out_len), ctypes.byref(out_ptr),
+            ctypes.byref(out_type)))
+        if not out_ptr.value or out_len.value == 0:
+            return None
+        n = out_len.value
+        if out_type.value == _DTYPE_F32:
+            return np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_float)),
+                                         shape=(n,)).copy()
+        if out_type.value == _DTYPE_F64:
+            return np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_double)),
+                                         shape=(n,)).copy()
+        return np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_int32)),
+                                     shape=(n,)).copy()
+
+    def get_label(self):
+        return self.get_field("label")
+
+    def get_weight(self):
+        return self.get_field("weight")
+
+    def get_group(self):
+        b = self.get_field("group")
+        if b is None:
+            return None
+        return np.diff(b)
+
+    def get_init_score(self):
+        return self.get_field("init_score")
+
+    def num_data(self):
+        self.construct()
+        out = ctypes.c_int32(0)
+        _safe_call(_LIB.LGBM_DatasetGetNumData(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def num_feature(self):
+        self.construct()
+        out = ctypes.c_int32(0)
+        _safe_call(_LIB.LGBM_DatasetGetNumFeature(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def feature_num_bin(self, feature):
+        self.construct()
+        out = ctypes.c_int32(0)
+        _safe_call(_LIB.LGBM_DatasetGetFeatureNumBin(
+            self._handle, ctypes.c_int(feature), ctypes.byref(out)))
+        return out.value
+
+    def save_binary(self, filename):
+        self.construct()
+        _safe_call(_LIB.LGBM_DatasetSaveBinary(self._handle, _c_str(str(filename))))
+        return self
+
+    def subset(self, used_indices, params=None):
+        self.construct()
+        used = np.ascontiguousarray(np.asarray(used_indices).ravel(), dtype=np.int32)
+        out = ctypes.c_void_p()
+        _safe_call(_LIB.LGBM_DatasetGetSubset(
+            self._handle, used.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+            ctypes.c_int32(len(used)), _c_str(_param_dict_to_str(params or {})),
+            ctypes.byref(out)))
+        ds = Dataset(None, params=params or dict(self.params))
+        ds._handle = out
+        ds.used_indices = used
+        ds.reference = self
+        return ds
+
+    def create_valid(self, data, label=None, weight=None, group=None, init_score=None,
+                     params=None, position=None):
+        return Dataset(data, label=label, reference=self, weight=weight, group=group,
+                       init_score=init_score, params=params or dict(self.params),
+                       position=position)
+
+    def get_feature_name(self):
+        self.construct()
+        return _get_string_buffer(_LIB.LGBM_DatasetGetFeatureNames, self._handle)
+
+
+def _get_string_buffer(fn, handle):
+    n = ctypes.c_int(0)
+    buf_len = ctypes.c_size_t(0)
+    # probe pass
+    dummy = (ctypes.c_char_p * 1)(ctypes.addressof(ctypes.create_string_buffer(1)))
+    fn(handle, ctypes.c_int(0), ctypes.byref(n), ctypes.c_size_t(0),
+       ctypes.byref(buf_len), dummy)
+    if n.value == 0:
+        return []
+    bufs = [ctypes.create_string_buffer(buf_len.value) for _ in range(n.value)]
+    arr = (ctypes.c_char_p * n.value)(*[ctypes.addressof(b) for b in bufs])
+    out_n = ctypes.c_int(0)
+    out_len = ctypes.c_size_t(0)
+    _safe_call(fn(handle, ctypes.c_int(n.value), ctypes.byref(out_n), buf_len,
+                  ctypes.byref(out_len), arr))
+    return [b.value.decode("utf-8") for b in bufs]
+
+
+class Booster:
+    """Gradient-boosting model handle (parity: reference lgb.Booster)."""
+
+    def __init__(self, params=None, train_set=None, model_file=None, model_str=None):
+        self.params = dict(params) if params else {}
+        self._handle = None
+        self._train_set = None
+        self._valid_sets = []
+        self.best_iteration = -1
+        self.best_score = {}
+        self._name_valid_sets = []
+        self._network_initialized = False
+        if train_set is not None:
+            if not isinstance(train_set, Dataset):
+                raise TypeError("train_set must be a Dataset")
+            train_set.construct()
+            out = ctypes.c_void_p()
+            _safe_call(_LIB.LGBM_BoosterCreate(
+                train_set._handle, _c_str(_param_dict_to_str(self.params)),
+                ctypes.byref(out)))
+            self._handle = out
+            self._train_set = train_set
+        elif model_file is not None:
+            out = ctypes.c_void_p()
+            out_iters = ctypes.c_int(0)
+            _safe_call(_LIB.LGBM_BoosterCreateFromModelfile(
+                _c_str(str(model_file)), ctypes.byref(out_iters), ctypes.byref(out)))
+            self._handle = out
+            self.best_iteration = -1
+        elif model_str is not None:
+            self.model_from_string(model_str)
+        else:
+            raise TypeError("Need train_set, model_file or model_str")
+
+    @property
+    def handle(self):
+        return self._handle
+
+    def __del__(self):
+        try:
+            if self._handle is not None:
+                _safe_call(_LIB.LGBM_BoosterFree(self._handle))
+                self._handle = None
+        except Exception:
+            pass
+
+    # ------------------------------------------------------------ training
+    def add_valid(self, data, name):
+        data.construct()
+        _safe_call(_LIB.LGBM_BoosterAddValidData(self._handle, data._handle))
+        self._valid_sets.append(data)
+        self._name_valid_sets.append(name)
+        return self
+
+    def reset_parameter(self, params):
+        self.params.update(params)
+        _safe_call(_LIB.LGBM_BoosterResetParameter(
+            self._handle, _c_str(_param_dict_to_str(params))))
+        return self
+
+    def update(self, train_set=None, fobj=None):
+        if train_set is not None and train_set is not self._train_set:
+            train_set.construct()
+            _safe_call(_LIB.LGBM_BoosterResetTrainingData(self._handle, train_set._handle))
+            self._train_set = train_set
+        if fobj is None:
+            is_finished = ctypes.c_int(0)
+            _safe_call(_LIB.LGBM_BoosterUpdateOneIter(self._handle, ctypes.byref(is_finished)))
+            return is_finished.value == 1
+        grad, hess = fobj(self.__inner_predict(0), self._train_set)
+        return self.__boost(grad, hess)
+
+    def __boost(self, grad, hess):
+        grad = np.ascontiguousarray(np.asarray(grad).ravel(order="F"), dtype=np.float32)
+        hess = np.ascontiguousarray(np.asarray(hess).ravel(order="F"), dtype=np.float32)
+        is_finished = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_BoosterUpdateOneIterCustom(
+            self._handle, grad.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            hess.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            ctypes.byref(is_finished)))
+        return is_finished.value == 1
+
+    def rollback_one_iter(self):
+        _safe_call(_LIB.LGBM_BoosterRollbackOneIter(self._handle))
+        return self
+
+    def current_iteration(self):
+        out = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_BoosterGetCurrentIteration(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def num_model_per_iteration(self):
+        out = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_BoosterNumModelPerIteration(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def num_trees(self):
+        out = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_BoosterNumberOfTotalModel(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def num_feature(self):
+        out = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_BoosterGetNumFeature(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def feature_name(self):
+        return _get_string_buffer(_LIB.LGBM_BoosterGetFeatureNames, self._handle)
+
+    def upper_bound(self):
+        out = ctypes.c_double(0)
+        _safe_call(_LIB.LGBM_BoosterGetUpperBoundValue(self._handle, ctypes.byref(out)))
+        return out.value
+
+    def lower_bound(self):
+        out = ctypes.c_double(0)
+        _safe_call(_LIB.LGBM_BoosterGetLowerBoundValue(self._handle, ctypes.byref(out)))
+        return out.value
+
+    # ------------------------------------------------------------ evaluation
+    def _eval_names(self):
+        return _get_string_buffer(_LIB.LGBM_BoosterGetEvalNames, self._handle)
+
+    def eval_train(self, feval=None):
+        return self.__inner_eval("training", 0, feval)
+
+    def eval_valid(self, feval=None):
+        out = []
+        for i in range(len(self._valid_sets)):
+            out.extend(self.__inner_eval(self._name_valid_sets[i], i + 1, feval))
+        return out
+
+    def eval(self, data, name, feval=None):
+        idx = None
+        for i, v in enumerate(self._valid_sets):
+            if v is data:
+                idx = i + 1
+        if idx is None and data is self._train_set:
+            idx = 0
+        if idx is None:
+            raise ValueError("Data must be added with add_valid first")
+        return self.__inner_eval(name, idx, feval)
+
+    def __inner_eval(self, name, data_idx, feval=None):
+        names = self._eval_names()
+        out = []
+        if names:
+            res = np.zeros(len(names), dtype=np.float64)
+            out_len = ctypes.c_int(0)
+            _safe_call(_LIB.LGBM_BoosterGetEval(
+                self._handle, ctypes.c_int(data_idx), ctypes.byref(out_len),
+                res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+            higher_better = [n in ("auc", "ndcg", "map", "average_precision") or
+                             n.startswith(("auc", "ndcg@", "map@")) for n in names]
+            for i in range(out_len.value):
+                out.append((name, names[i], res[i], higher_better[i]))
+        if feval is not None:
+            ds = self._train_set if data_idx == 0 else self._valid_sets[data_idx - 1]
+            fe = feval(self.__inner_predict(data_idx), ds)
+            if isinstance(fe, list):
+                for (fn, fv, fb) in fe:
+                    out.append((name, fn, fv, fb))
+            else:
+                fn, fv, fb = fe
+                out.append((name, fn, fv, fb))
+        return out
+
+    def __inner_predict(self, data_idx):
+        n = ctypes.c_int64(0)
+        _safe_call(_LIB.LGBM_BoosterGetNumPredict(self._handle, ctypes.c_int(data_idx),
+                                                  ctypes.byref(n)))
+        res = np.zeros(n.value, dtype=np.float64)
+        out_len = ctypes.c_int64(0)
+        _safe_call(_LIB.LGBM_BoosterGetPredict(
+            self._handle, ctypes.c_int(data_idx), ctypes.byref(out_len),
+            res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+        return res
+
+    # ------------------------------------------------------------ prediction
+    def predict(self, data, start_iteration=0, num_iteration=None, raw_score=False,
+                pred_leaf=False, pred_contrib=False, validate_features=False, **kwargs):
+        if num_iteration is None:
+            num_iteration = self.best_iteration if self.best_iteration > 0 else -1
+        ptype = _PREDICT_NORMAL
+        if raw_score:
+            ptype = _PREDICT_RAW
+        if pred_leaf:
+            ptype = _PREDICT_LEAF
+        if pred_contrib:
+            ptype = _PREDICT_CONTRIB
+        arr = _to_2d_float64(data)
+        nrow, ncol = arr.shape
+        n = ctypes.c_int64(0)
+        _safe_call(_LIB.LGBM_BoosterCalcNumPredict(
+            self._handle, ctypes.c_int(nrow), ctypes.c_int(ptype),
+            ctypes.c_int(start_iteration), ctypes.c_int(num_iteration), ctypes.byref(n)))
+        res = np.zeros(n.value, dtype=np.float64)
+        out_len = ctypes.c_int64(0)
+        _safe_call(_LIB.LGBM_BoosterPredictForMat(
+            self._handle, arr.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
+            ctypes.c_int32(nrow), ctypes.c_int32(ncol), ctypes.c_int(1),
+            ctypes.c_int(ptype), ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+            _c_str(""), ctypes.byref(out_len),
+            res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+        per_row = out_len.value // nrow if nrow else 0
+        if per_row > 1:
+            res = res.reshape(nrow, per_row)
+        if pred_leaf:
+            res = res.astype(np.int32)
+        return res
+
+    def refit(self, data, label, decay_rate=0.9, **kwargs):
+        leaf_preds = self.predict(data, pred_leaf=True)
+        if leaf_preds.ndim == 1:
+            leaf_preds = leaf_preds.reshape(-1, 1)
+        nrow, ncol = leaf_preds.shape
+        new_params = dict(self.params)
+        new_params["refit_decay_rate"] = decay_rate
+        train_set = Dataset(data, label=label, params=new_params)
+        new_booster = Booster(new_params, train_set)
+        # copy model
+        _safe_call(_LIB.LGBM_BoosterMerge(new_booster._handle, self._handle))
+        arr = np.ascontiguousarray(leaf_preds.astype(np.int32))
+        _safe_call(_LIB.LGBM_BoosterRefit(
+            new_booster._handle, arr.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+            ctypes.c_int32(nrow), ctypes.c_int32(ncol)))
+        return new_booster
+
+    # ------------------------------------------------------------ serialization
+    def save_model(self, filename, num_iteration=None, start_iteration=0,
+                   importance_type="split"):
+        imp = 0 if importance_type == "split" else 1
+        if num_iteration is None:
+            num_iteration = self.best_iteration if self.best_iteration > 0 else -1
+        _safe_call(_LIB.LGBM_BoosterSaveModel(
+            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+            ctypes.c_int(imp), _c_str(str(filename))))
+        return self
+
+    def model_to_string(self, num_iteration=None, start_iteration=0,
+                        importance_type="split"):
+        imp = 0 if importance_type == "split" else 1
+        if num_iteration is None:
+            num_iteration = self.best_iteration if self.best_iteration > 0 else -1
+        out_len = ctypes.c_int64(0)
+        _safe_call(_LIB.LGBM_BoosterSaveModelToString(
+            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+            ctypes.c_int(imp), ctypes.c_int64(0), ctypes.byref(out_len), None))
+        buf = ctypes.create_string_buffer(out_len.value)
+        _safe_call(_LIB.LGBM_BoosterSaveModelToString(
+            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+            ctypes.c_int(imp), out_len, ctypes.byref(out_len), buf))
+        return buf.value.decode("utf-8")
+
+    def model_from_string(self, model_str):
+        if self._handle is not None:
+            _safe_call(_LIB.LGBM_BoosterFree(self._handle))
+        out = ctypes.c_void_p()
+        out_iters = ctypes.c_int(0)
+        _safe_call(_LIB.LGBM_BoosterLoadModelFromString(
+            _c_str(model_str), ctypes.byref(out_iters), ctypes.byref(out)))
+        self._handle = out
+        return self
+
+    def dump_model(self, num_iteration=None, start_iteration=0, importance_type="split"):
+        imp = 0 if importance_type == "split" else 1
+        if num_iteration is None:
+            num_iteration = self.best_iteration if self.best_iteration > 0 else -1
+        out_len = ctypes.c_int64(0)
+        _safe_call(_LIB.LGBM_BoosterDumpModel(
+            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+            ctypes.c_int(imp), ctypes.c_int64(0), ctypes.byref(out_len), None))
+        buf = ctypes.create_string_buffer(out_len.value)
+        _safe_call(_LIB.LGBM_BoosterDumpModel(
+            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+            ctypes.c_int(imp), out_len, ctypes.byref(out_len), buf))
+        return json.loads(buf.value.decode("utf-8"))
+
+    def feature_importance(self, importance_type="split", iteration=None):
+        imp = 0 if importance_type == "split" else 1
+        if iteration is None:
+            iteration = -1
+        nf = self.num_feature()
+        res = np.zeros(nf, dtype=np.float64)
+        _safe_call(_LIB.LGBM_BoosterFeatureImportance(
+            self._handle, ctypes.c_int(iteration), ctypes.c_int(imp),
+            res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+        if importance_type == "split":
+            return res.astype(np.int64)
+        return res
+
+    def free_dataset(self):
+        self._train_set = None
+        self._valid_sets = []
+        return self
+
+    def free_network(self):
+        _safe_call(_LIB.LGBM_NetworkFree())
+        self._network_initialized = False
+        return self
+
+    def set_network(self, machines=None, local_listen_port=12400, listen_time_out=120,
+                    num_machines=1):
+        from .parallel import init_network_from_torch_distributed
+        init_network_from_torch_distributed()
+        self._network_initialized = True
+        return self
