@@ -36,6 +36,22 @@ class TreeLearner {
                                data_size_t total_num_data, const data_size_t* bag_indices,
                                data_size_t bag_cnt, const double* train_score) = 0;
 
+  // ---- device-resident boosting hooks (HIP learner; no-ops on CPU learners).
+  // The GBDT engine keeps one code path: when IsHIPLearner() is true the learner owns
+  // device score/grad/hess buffers and these hooks keep them in sync.
+  virtual bool IsHIPLearner() const { return false; }
+  /*! true if grad/hess for this objective are computed by a device kernel */
+  virtual bool DeviceObjectiveSupported(const std::string& objective_name) const {
+    (void)objective_name;
+    return false;
+  }
+  /*! compute grad/hess on device from device scores (device objectives only) */
+  virtual void DeviceBoosting(const class ObjectiveFunction* obj) { (void)obj; }
+  /*! add a constant to the device score vector (boost_from_average) */
+  virtual void DeviceAddInitScore(double v) { (void)v; }
+  /*! download the device train scores into a host buffer (for metrics / custom obj) */
+  virtual void DownloadTrainScore(double* dst) { (void)dst; }
+
   static TreeLearner* Create(const std::string& learner_type, const std::string& device_type,
                              const Config* config);
 };
@@ -141,6 +157,9 @@ class SerialTreeLearner : public TreeLearner {
   Random extra_rng_{0};
   int iter_counter_ = 0;
 };
+
+/*! set by the HIP learner's static registrar when the device module is linked in */
+extern TreeLearner* (*g_create_hip_learner)(const Config*);
 
 }  // namespace migbm
 

@@ -451,7 +451,8 @@ void Tree::AddPredictionToScore(const Dataset* data, data_size_t num_data, doubl
         const int cat_idx = static_cast<int>(threshold_in_bin_[node]);
         const uint32_t* bits = cat_threshold_.data() + cat_boundaries_[cat_idx];
         const int n_words = cat_boundaries_[cat_idx + 1] - cat_boundaries_[cat_idx];
-        node = ((bin >> 5) < static_cast<uint32_t>(n_words) && ((bits[bin >> 5] >> (bin & 31)) & 1))
+        const int cat = static_cast<int>(data->FeatureBinMapper(f)->BinToValue(bin));
+        node = (cat >= 0 && (cat >> 5) < n_words && ((bits[cat >> 5] >> (cat & 31)) & 1))
                    ? left_child_[node] : right_child_[node];
       } else {
         // NaN bin (if any) is the last bin; honor default direction
@@ -488,7 +489,8 @@ void Tree::AddPredictionToScore(const Dataset* data, const data_size_t* used_ind
         const int cat_idx = static_cast<int>(threshold_in_bin_[node]);
         const uint32_t* bits = cat_threshold_.data() + cat_boundaries_[cat_idx];
         const int n_words = cat_boundaries_[cat_idx + 1] - cat_boundaries_[cat_idx];
-        node = ((bin >> 5) < static_cast<uint32_t>(n_words) && ((bits[bin >> 5] >> (bin & 31)) & 1))
+        const int cat = static_cast<int>(data->FeatureBinMapper(f)->BinToValue(bin));
+        node = (cat >= 0 && (cat >> 5) < n_words && ((bits[cat >> 5] >> (cat & 31)) & 1))
                    ? left_child_[node] : right_child_[node];
       } else {
         const BinMapper* m = data->FeatureBinMapper(f);

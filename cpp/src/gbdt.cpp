@@ -123,6 +123,8 @@ void GBDT::Init(const Config* config, const Dataset* train_data,
     num_data_ = train_data_->num_data();
     max_feature_idx_ = train_data_->num_total_features() - 1;
     feature_names_ = train_data_->feature_names();
+    // snapshot bin infos so model saving never needs the (possibly freed) dataset
+    feature_infos_ = Common::SplitAny(train_data_->FeatureInfoString().c_str(), " ");
     label_idx_ = 0;
     tree_learner_.reset(TreeLearner::Create(config_->tree_learner, config_->device_type,
                                             config_));
@@ -200,6 +202,7 @@ double GBDT::BoostFromAverage(int class_id, bool update_scores) {
     double init_score = objective_->BoostFromScore(class_id);
     init_score = Network::GlobalSyncUpByMean(init_score);
     if (std::fabs(init_score) > kEpsilon && update_scores) {
+      if (tree_learner_->IsHIPLearner()) tree_learner_->DeviceAddInitScore(init_score);
       double* sc = train_score_.data() + static_cast<size_t>(class_id) * num_data_;
 #pragma omp parallel for schedule(static)
       for (data_size_t i = 0; i < num_data_; ++i) sc[i] += init_score;
@@ -217,12 +220,22 @@ double GBDT::BoostFromAverage(int class_id, bool update_scores) {
 
 bool GBDT::TrainOneIter(const score_t* gradients, const score_t* hessians) {
   std::vector<double> init_scores(num_tree_per_iteration_, 0.0);
+  const bool dev = tree_learner_->IsHIPLearner();
+  const bool dev_obj = dev && objective_ != nullptr &&
+                       tree_learner_->DeviceObjectiveSupported(objective_->GetName());
   if (gradients == nullptr || hessians == nullptr) {
     for (int c = 0; c < num_tree_per_iteration_; ++c)
       init_scores[c] = BoostFromAverage(c, true);
-    objective_->GetGradients(train_score_.data(), gradients_.data(), hessians_.data());
-    gradients = gradients_.data();
-    hessians = hessians_.data();
+    if (dev_obj) {
+      tree_learner_->DeviceBoosting(objective_);
+      gradients = gradients_.data();   // sentinel (ignored by the HIP learner)
+      hessians = hessians_.data();
+    } else {
+      if (dev) tree_learner_->DownloadTrainScore(train_score_.data());
+      objective_->GetGradients(train_score_.data(), gradients_.data(), hessians_.data());
+      gradients = gradients_.data();
+      hessians = hessians_.data();
+    }
   } else {
     // custom objective: copy so bagging/GOSS can modify
     std::copy(gradients, gradients + gradients_.size(), gradients_.begin());
@@ -327,6 +340,9 @@ void GBDT::UpdateScore(const Tree* tree, int cur_tree_id) {
 
 std::vector<double> GBDT::GetEvalAt(int data_idx) const {
   std::vector<double> out;
+  if (data_idx == 0 && tree_learner_ && tree_learner_->IsHIPLearner()) {
+    tree_learner_->DownloadTrainScore(const_cast<double*>(train_score_.data()));
+  }
   if (data_idx == 0) {
     for (auto* m : training_metrics_) {
       auto r = m->Eval(train_score_.data(), objective_);
@@ -464,6 +480,8 @@ void GBDT::PredictContrib(const double* features, double* output, int start_iter
 }
 
 const double* GBDT::GetTrainingScore(int64_t* out_len) const {
+  if (tree_learner_ && tree_learner_->IsHIPLearner())
+    tree_learner_->DownloadTrainScore(const_cast<double*>(train_score_.data()));
   *out_len = static_cast<int64_t>(train_score_.size());
   return train_score_.data();
 }
@@ -474,6 +492,8 @@ int64_t GBDT::GetNumPredictAt(int data_idx) const {
 }
 
 void GBDT::GetPredictAt(int data_idx, double* result, int64_t* out_len) const {
+  if (data_idx == 0 && tree_learner_ && tree_learner_->IsHIPLearner())
+    tree_learner_->DownloadTrainScore(const_cast<double*>(train_score_.data()));
   const std::vector<double>* sc =
       data_idx == 0 ? &train_score_ : &valid_score_[data_idx - 1];
   data_size_t n = data_idx == 0 ? num_data_ : valid_data_[data_idx - 1]->num_data();

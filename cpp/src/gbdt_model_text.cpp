@@ -43,8 +43,7 @@ std::string GBDT::SaveModelToString(int start_iter, int num_iter,
   else if (!objective_name_.empty()) ss << "objective=" << objective_name_ << '\n';
   if (average_output_) ss << "average_output" << '\n';
   ss << "feature_names=" << Common::Join(feature_names_, " ") << '\n';
-  if (train_data_ != nullptr) ss << "feature_infos=" << train_data_->FeatureInfoString() << '\n';
-  else if (!feature_infos_.empty()) ss << "feature_infos=" << Common::Join(feature_infos_, " ") << '\n';
+  if (!feature_infos_.empty()) ss << "feature_infos=" << Common::Join(feature_infos_, " ") << '\n';
   else {
     std::vector<std::string> none(max_feature_idx_ + 1, "none");
     ss << "feature_infos=" << Common::Join(none, " ") << '\n';

@@ -72,8 +72,8 @@ double UnwoundPathSum(const PathElem* path, int depth, int path_index) {
 void TreeSHAPRec(const Tree* tree, const double* x, double* phi, int node, int depth,
                  PathElem* parent_path, double parent_zero_fraction,
                  double parent_one_fraction, int parent_feature_index) {
-  // copy parent path
-  PathElem* path = parent_path + depth;
+  // copy parent path (fresh storage segment past the parent's depth+1 entries)
+  PathElem* path = parent_path + depth + 1;
   for (int i = 0; i < depth; ++i) path[i] = parent_path[i];
   ExtendPath(path, depth, parent_zero_fraction, parent_one_fraction, parent_feature_index);
 

@@ -239,6 +239,14 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     const BinMapper* m = train_data_->FeatureBinMapper(f);
     const int right_leaf = num_leaves;
 
+    // partition first so the tree records exact child counts
+    partition_.Split(best_leaf, right_leaf, MakeGoLeft(s));
+    data_size_t left_cnt_actual = partition_.leaf_count(best_leaf);
+    data_size_t right_cnt_actual = partition_.leaf_count(right_leaf);
+    GlobalChildCounts(&left_cnt_actual, &right_cnt_actual);
+    s.left_count = left_cnt_actual;
+    s.right_count = right_cnt_actual;
+
     // tree structure update
     if (!s.cat_bitset_inner.empty()) {
       // map bin-level bitset to category-value bitset for prediction on raw values
@@ -268,12 +276,6 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                   s.left_count, s.right_count, s.left_sum_hessian, s.right_sum_hessian,
                   static_cast<float>(s.gain), m->missing_type(), s.default_left);
     }
-
-    // partition rows
-    partition_.Split(best_leaf, right_leaf, MakeGoLeft(s));
-    data_size_t left_cnt_actual = partition_.leaf_count(best_leaf);
-    data_size_t right_cnt_actual = partition_.leaf_count(right_leaf);
-    GlobalChildCounts(&left_cnt_actual, &right_cnt_actual);
 
     // child contexts
     const double parent_out = (s.left_output * s.left_sum_hessian +

@@ -652,8 +652,13 @@ class Booster:
         return res
 
     def free_dataset(self):
-        self._train_set = None
-        self._valid_sets = []
+        # The native GBDT keeps raw pointers into the Dataset; keep the Python objects
+        # alive for the Booster's lifetime (the reference uses shared_ptr ownership in
+        # the C++ Booster for the same reason).
+        self._kept_refs = getattr(self, "_kept_refs", [])
+        if self._train_set is not None:
+            self._kept_refs.append(self._train_set)
+        self._kept_refs.extend(self._valid_sets)
         return self
 
     def free_network(self):

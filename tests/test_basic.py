@@ -1,0 +1,80 @@
+"""Dataset construction / binning / fields (parity target: reference test_basic.py)."""
+import numpy as np
+import pytest
+
+import lightgbm_amd as lgb
+
+
+def test_dataset_from_numpy():
+    X = np.random.RandomState(0).randn(500, 5)
+    y = np.random.RandomState(1).rand(500)
+    d = lgb.Dataset(X, label=y).construct()
+    assert d.num_data() == 500
+    assert d.num_feature() == 5
+    np.testing.assert_allclose(d.get_label(), y.astype(np.float32), rtol=1e-6)
+
+
+def test_dataset_feature_num_bin():
+    rng = np.random.RandomState(0)
+    X = rng.randn(1000, 3)
+    X[:, 1] = rng.randint(0, 5, size=1000)  # few distinct values
+    d = lgb.Dataset(X, params={"max_bin": 63}).construct()
+    assert d.feature_num_bin(0) <= 63
+    assert d.feature_num_bin(1) <= 6
+
+
+def test_dataset_weights_group():
+    X = np.random.RandomState(0).randn(100, 3)
+    y = np.random.RandomState(1).rand(100)
+    w = np.arange(100, dtype=np.float32) + 1
+    d = lgb.Dataset(X, label=y, weight=w).construct()
+    np.testing.assert_allclose(d.get_weight(), w, rtol=1e-6)
+    g = np.array([30, 30, 40], dtype=np.int32)
+    d2 = lgb.Dataset(X, label=y, group=g).construct()
+    np.testing.assert_array_equal(d2.get_group(), g)
+
+
+def test_dataset_nan_handling():
+    rng = np.random.RandomState(0)
+    X = rng.randn(1000, 2)
+    X[::7, 0] = np.nan
+    y = (np.nan_to_num(X[:, 0]) > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 10)
+    pred = bst.predict(X)
+    assert np.all(np.isfinite(pred))
+
+
+def test_dataset_from_csr():
+    import scipy.sparse as sp
+    rng = np.random.RandomState(0)
+    X = sp.random(300, 10, density=0.3, random_state=0, format="csr")
+    y = rng.rand(300)
+    d = lgb.Dataset(X, label=y).construct()
+    assert d.num_data() == 300
+    assert d.num_feature() == 10
+
+
+def test_dataset_subset():
+    X = np.random.RandomState(0).randn(200, 4)
+    y = np.random.RandomState(1).rand(200)
+    d = lgb.Dataset(X, label=y).construct()
+    sub = d.subset(np.arange(50))
+    assert sub.num_data() == 50
+    np.testing.assert_allclose(sub.get_label(), y[:50].astype(np.float32), rtol=1e-6)
+
+
+def test_dataset_save_binary(tmp_path):
+    X = np.random.RandomState(0).randn(200, 4)
+    y = np.random.RandomState(1).rand(200)
+    d = lgb.Dataset(X, label=y).construct()
+    f = tmp_path / "data.bin"
+    d.save_binary(str(f))
+    d2 = lgb.Dataset(str(f)).construct()
+    assert d2.num_data() == 200
+    np.testing.assert_allclose(d2.get_label(), y.astype(np.float32), rtol=1e-6)
+
+
+def test_feature_names():
+    X = np.random.RandomState(0).randn(100, 3)
+    d = lgb.Dataset(X, label=np.zeros(100), feature_name=["a", "b", "c"]).construct()
+    assert d.get_feature_name() == ["a", "b", "c"]

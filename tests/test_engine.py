@@ -1,0 +1,277 @@
+"""Training engine tests (parity target: reference test_engine.py, core subset)."""
+import numpy as np
+import pytest
+
+import lightgbm_amd as lgb
+
+
+def _binary_data(n=5000, d=10, seed=42):
+    rng = np.random.RandomState(seed)
+    X = rng.randn(n, d)
+    logit = 2 * X[:, 0] - 1.5 * X[:, 1] + X[:, 2] * X[:, 3] + 0.5 * rng.randn(n)
+    y = (logit > 0).astype(np.float32)
+    return X, y
+
+
+def _regression_data(n=5000, d=10, seed=7):
+    rng = np.random.RandomState(seed)
+    X = rng.randn(n, d)
+    y = 3 * X[:, 0] + np.sin(X[:, 1]) + 0.1 * rng.randn(n)
+    return X, y.astype(np.float32)
+
+
+def test_binary_auc():
+    X, y = _binary_data()
+    ev = {}
+    bst = lgb.train({"objective": "binary", "metric": "auc", "verbosity": -1},
+                    lgb.Dataset(X[:4000], label=y[:4000]), 50,
+                    valid_sets=[lgb.Dataset(X[:4000], label=y[:4000]).create_valid(
+                        X[4000:], label=y[4000:])],
+                    callbacks=[lgb.record_evaluation(ev)])
+    assert ev["valid_0"]["auc"][-1] > 0.93
+    assert ev["valid_0"]["auc"][-1] > ev["valid_0"]["auc"][0]
+
+
+def test_regression_l2():
+    X, y = _regression_data()
+    bst = lgb.train({"objective": "regression", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 60)
+    pred = bst.predict(X)
+    mse = float(np.mean((pred - y) ** 2))
+    assert mse < 0.1 * float(np.var(y))
+
+
+@pytest.mark.parametrize("objective", ["regression_l1", "huber", "fair", "quantile", "mape"])
+def test_regression_objectives_run(objective):
+    X, y = _regression_data(n=2000)
+    y = np.abs(y) + 0.1
+    bst = lgb.train({"objective": objective, "verbosity": -1}, lgb.Dataset(X, label=y), 15)
+    assert np.all(np.isfinite(bst.predict(X[:50])))
+
+
+@pytest.mark.parametrize("objective", ["poisson", "gamma", "tweedie"])
+def test_positive_objectives_run(objective):
+    rng = np.random.RandomState(0)
+    X = rng.randn(2000, 5)
+    y = np.exp(0.5 * X[:, 0]) + 0.1
+    bst = lgb.train({"objective": objective, "verbosity": -1}, lgb.Dataset(X, label=y), 15)
+    pred = bst.predict(X[:50])
+    assert np.all(pred > 0)
+
+
+def test_multiclass():
+    rng = np.random.RandomState(0)
+    X = rng.randn(3000, 6)
+    y = (X[:, 0] + 0.3 * rng.randn(3000) > 0.5).astype(int) + \
+        (X[:, 1] + 0.3 * rng.randn(3000) > 0).astype(int)
+    bst = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+                    lgb.Dataset(X, label=y.astype(np.float32)), 30)
+    pred = bst.predict(X)
+    assert pred.shape == (3000, 3)
+    np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-6)
+    acc = (pred.argmax(axis=1) == y).mean()
+    assert acc > 0.7
+
+
+def test_lambdarank():
+    rng = np.random.RandomState(0)
+    n_queries = 100
+    rows = []
+    labels = []
+    groups = []
+    for q in range(n_queries):
+        nq = rng.randint(5, 30)
+        Xq = rng.randn(nq, 5)
+        rel = (Xq[:, 0] + 0.5 * rng.randn(nq) > 0.5).astype(int) * 2
+        rows.append(Xq)
+        labels.append(rel)
+        groups.append(nq)
+    X = np.vstack(rows)
+    y = np.concatenate(labels).astype(np.float32)
+    ev = {}
+    train = lgb.Dataset(X, label=y, group=np.array(groups, dtype=np.int32))
+    bst = lgb.train({"objective": "lambdarank", "metric": "ndcg", "eval_at": [5],
+                     "verbosity": -1}, train, 30,
+                    valid_sets=[train], valid_names=["train"],
+                    callbacks=[lgb.record_evaluation(ev)])
+    assert ev["train"]["ndcg@5"][-1] > 0.80
+
+
+def test_early_stopping():
+    X, y = _binary_data()
+    train = lgb.Dataset(X[:4000], label=y[:4000])
+    valid = train.create_valid(X[4000:], label=y[4000:])
+    bst = lgb.train({"objective": "binary", "metric": "binary_logloss", "verbosity": -1},
+                    train, 500, valid_sets=[valid],
+                    callbacks=[lgb.early_stopping(5, verbose=False)])
+    assert bst.best_iteration > 0
+    assert bst.best_iteration < 500
+
+
+def test_model_save_load_roundtrip(tmp_path):
+    X, y = _binary_data(n=2000)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    pred1 = bst.predict(X[:100])
+    f = tmp_path / "model.txt"
+    bst.save_model(str(f))
+    content = f.read_text()
+    # model text v4 markers (format parity with the reference)
+    assert content.startswith("tree\nversion=v4\n")
+    assert "end of trees" in content
+    assert "feature_importances:" in content
+    assert "parameters:" in content
+    bst2 = lgb.Booster(model_file=str(f))
+    pred2 = bst2.predict(X[:100])
+    np.testing.assert_allclose(pred1, pred2, rtol=1e-12)
+
+
+def test_continue_training():
+    X, y = _binary_data(n=2000)
+    train = lgb.Dataset(X, label=y)
+    bst1 = lgb.train({"objective": "binary", "verbosity": -1}, train, 10)
+    model = bst1.model_to_string()
+    train2 = lgb.Dataset(X, label=y)
+    bst2 = lgb.train({"objective": "binary", "verbosity": -1}, train2, 10,
+                     init_model=bst1)
+    assert bst2.num_trees() == 20
+
+
+def test_custom_objective():
+    X, y = _binary_data(n=2000)
+
+    def logloss_obj(preds, train_data):
+        labels = train_data.get_label()
+        p = 1.0 / (1.0 + np.exp(-preds))
+        return (p - labels).astype(np.float32), (p * (1 - p)).astype(np.float32)
+
+    bst = lgb.train({"objective": "none", "verbosity": -1}, lgb.Dataset(X, label=y), 30,
+                    fobj=logloss_obj)
+    pred_raw = bst.predict(X, raw_score=True)
+    p = 1 / (1 + np.exp(-pred_raw))
+    acc = ((p > 0.5) == y).mean()
+    assert acc > 0.85
+
+
+def test_custom_metric():
+    X, y = _binary_data(n=2000)
+
+    def accuracy(preds, ds):
+        labels = ds.get_label()
+        return "accuracy", float(((preds > 0.5) == labels).mean()), True
+
+    train = lgb.Dataset(X[:1500], label=y[:1500])
+    valid = train.create_valid(X[1500:], label=y[1500:])
+    ev = {}
+    lgb.train({"objective": "binary", "metric": "none", "verbosity": -1}, train, 10,
+              valid_sets=[valid], feval=accuracy, callbacks=[lgb.record_evaluation(ev)])
+    assert "accuracy" in ev["valid_0"]
+    assert ev["valid_0"]["accuracy"][-1] > 0.8
+
+
+@pytest.mark.parametrize("boosting", ["dart", "rf"])
+def test_other_boosting_modes(boosting):
+    X, y = _binary_data(n=3000)
+    params = {"objective": "binary", "boosting": boosting, "verbosity": -1}
+    if boosting == "rf":
+        params.update({"bagging_freq": 1, "bagging_fraction": 0.7})
+    bst = lgb.train(params, lgb.Dataset(X, label=y), 20)
+    pred = bst.predict(X)
+    acc = ((pred > 0.5) == y).mean()
+    assert acc > 0.75
+
+
+def test_bagging_and_feature_fraction():
+    X, y = _binary_data()
+    bst = lgb.train({"objective": "binary", "bagging_freq": 1, "bagging_fraction": 0.6,
+                     "feature_fraction": 0.7, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 30)
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.85
+
+
+def test_goss():
+    X, y = _binary_data()
+    bst = lgb.train({"objective": "binary", "data_sample_strategy": "goss",
+                     "verbosity": -1}, lgb.Dataset(X, label=y), 30)
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.85
+
+
+def test_categorical_feature():
+    rng = np.random.RandomState(0)
+    n = 4000
+    cat = rng.randint(0, 8, size=n)
+    X = np.column_stack([cat.astype(float), rng.randn(n)])
+    effect = np.array([2.0, -1.0, 0.5, -2.0, 1.5, 0.0, -0.5, 1.0])
+    y = (effect[cat] + 0.3 * rng.randn(n) > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5},
+                    lgb.Dataset(X, label=y, categorical_feature=[0]), 30)
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.85
+
+
+def test_feature_importance():
+    X, y = _binary_data()
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    imp_split = bst.feature_importance("split")
+    imp_gain = bst.feature_importance("gain")
+    assert imp_split.shape == (10,)
+    # features 0,1 drive the label; they should dominate gain
+    assert np.argsort(imp_gain)[-2:].tolist() in ([0, 1], [1, 0]) or \
+        imp_gain[0] + imp_gain[1] > 0.5 * imp_gain.sum()
+
+
+def test_predict_types():
+    X, y = _binary_data(n=1000)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "num_leaves": 7},
+                    lgb.Dataset(X, label=y), 5)
+    raw = bst.predict(X[:10], raw_score=True)
+    prob = bst.predict(X[:10])
+    np.testing.assert_allclose(prob, 1 / (1 + np.exp(-raw)), rtol=1e-9)
+    leaves = bst.predict(X[:10], pred_leaf=True)
+    assert leaves.shape == (10, 5)
+    assert leaves.max() < 7
+    contrib = bst.predict(X[:10], pred_contrib=True)
+    assert contrib.shape == (10, 11)
+    np.testing.assert_allclose(contrib.sum(axis=1), raw, rtol=1e-5, atol=1e-5)
+
+
+def test_cv():
+    X, y = _binary_data(n=2000)
+    res = lgb.cv({"objective": "binary", "metric": "auc", "verbosity": -1},
+                 lgb.Dataset(X, label=y), num_boost_round=10, nfold=3)
+    assert "valid auc-mean" in res
+    assert len(res["valid auc-mean"]) == 10
+    assert res["valid auc-mean"][-1] > 0.85
+
+
+def test_dump_model_json():
+    X, y = _binary_data(n=1000)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 3)
+    d = bst.dump_model()
+    assert d["num_class"] == 1
+    assert len(d["tree_info"]) == 3
+    assert "tree_structure" in d["tree_info"][0]
+
+
+def test_monotone_constraints():
+    rng = np.random.RandomState(0)
+    X = rng.rand(3000, 2)
+    y = (2 * X[:, 0] + 0.1 * rng.randn(3000)).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "monotone_constraints": [1, 0],
+                     "verbosity": -1}, lgb.Dataset(X, label=y), 30)
+    # predictions must be monotone non-decreasing in feature 0
+    xs = np.linspace(0.05, 0.95, 20)
+    grid = np.column_stack([xs, np.full(20, 0.5)])
+    pred = bst.predict(grid)
+    assert np.all(np.diff(pred) >= -1e-9)
+
+
+def test_weights_affect_training():
+    X, y = _binary_data(n=2000)
+    w = np.where(y > 0, 10.0, 1.0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1},
+                    lgb.Dataset(X, label=y, weight=w), 20)
+    pred = bst.predict(X)
+    # heavy positive weights push average prediction up
+    assert pred.mean() > y.mean()
