@@ -147,8 +147,11 @@ TreeLearner* TreeLearner::Create(const std::string& learner_type,
                                  const std::string& device_type, const Config* config) {
   const bool gpu = device_type == "gpu" || device_type == "cuda";
   if (gpu) {
-    if (g_create_hip_learner != nullptr) return g_create_hip_learner(config);
-    Log::Warning("HIP tree learner not available in this build; falling back to CPU");
+    // fail loudly rather than silently training on CPU when a GPU was requested
+    if (g_create_hip_learner == nullptr)
+      Log::Fatal("device_type=%s requested but the HIP learner module is not linked",
+                 device_type.c_str());
+    return g_create_hip_learner(config);
   }
   if (learner_type == "serial") return new SerialTreeLearner(config);
   if (learner_type == "feature") return new FeatureParallelTreeLearner(config);
