@@ -98,6 +98,10 @@ class Tree {
   void AddPredictionToScore(const class Dataset* data, const data_size_t* used_indices,
                             data_size_t num_data, double* score) const;
 
+  /*! replace leaf counts with exact values and recompute internal counts bottom-up
+   *  (the HIP learner builds with hessian-approx counts and fixes them here). */
+  void OverrideLeafCounts(const std::vector<int>& counts);
+
   std::string ToString() const;   // model-text v4 tree block
   std::string ToJSON() const;
 

@@ -94,6 +94,19 @@ int Tree::SplitCategorical(int leaf, int feature, int real_feature,
   return num_leaves_ - 1;
 }
 
+void Tree::OverrideLeafCounts(const std::vector<int>& counts) {
+  for (int l = 0; l < num_leaves_ && l < static_cast<int>(counts.size()); ++l)
+    leaf_count_[l] = counts[l];
+  if (num_leaves_ <= 1) return;
+  std::function<int(int)> rec = [&](int node) -> int {
+    if (node < 0) return leaf_count_[~node];
+    int c = rec(left_child_[node]) + rec(right_child_[node]);
+    internal_count_[node] = c;
+    return c;
+  };
+  rec(0);
+}
+
 double Tree::GetUpperBoundValue() const {
   double mx = leaf_value_[0];
   for (int i = 1; i < num_leaves_; ++i) mx = std::max(mx, leaf_value_[i]);
