@@ -141,16 +141,19 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
   float* ghist = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
                  static_cast<size_t>(part_bin_base) * 2;
 
-  extern __shared__ float lh[];  // NCOPIES * part_bins * 2
+  // LDS layout [bin][NCOPIES][g,h]: the NCOPIES copies of one bin sit in ADJACENT
+  // words, i.e. different banks — the degenerate all-rows-same-bin leaf serializes
+  // through NCOPIES banks in parallel instead of one.
+  extern __shared__ float lh[];  // part_bins * NCOPIES * 2
   __shared__ int loff[256];
   const int nfeat = feat_end - feat_begin;
   for (int i = threadIdx.x; i < nfeat; i += blockDim.x)
-    loff[i] = (fm[feat_begin + i].bin_off - part_bin_base) * 2;
-  const int nelem = part_bins * 2;
-  for (int i = threadIdx.x; i < nelem * NCOPIES; i += blockDim.x) lh[i] = 0.0f;
+    loff[i] = fm[feat_begin + i].bin_off - part_bin_base;
+  const int nelem = part_bins * 2 * NCOPIES;
+  for (int i = threadIdx.x; i < nelem; i += blockDim.x) lh[i] = 0.0f;
   __syncthreads();
 
-  float* my = lh + (threadIdx.x % NCOPIES) * nelem;
+  const int my_copy = (threadIdx.x % NCOPIES) * 2;
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
   const int nthreads = blockDim.x * gridDim.x;
   const int c0 = feat_begin & ~15;
@@ -167,17 +170,19 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
         const int f = c + j;
         if (f < feat_begin || f >= feat_end) continue;
         const int b = (w[j >> 2] >> ((j & 3) * 8)) & 0xFF;
-        float* dst = my + loff[f - feat_begin] + b * 2;
+        float* dst = lh + (loff[f - feat_begin] + b) * (2 * NCOPIES) + my_copy;
         atomicAdd(dst, gi);
         atomicAdd(dst + 1, hi);
       }
     }
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < nelem; i += blockDim.x) {
-    float v = lh[i];
+  for (int i = threadIdx.x; i < part_bins * 2; i += blockDim.x) {
+    const int bin = i >> 1;
+    const int gh = i & 1;
+    float v = 0.0f;
 #pragma unroll
-    for (int cpy = 1; cpy < NCOPIES; ++cpy) v += lh[cpy * nelem + i];
+    for (int cpy = 0; cpy < NCOPIES; ++cpy) v += lh[bin * 2 * NCOPIES + cpy * 2 + gh];
     if (v != 0.0f) atomicAdd(&ghist[i], v);
   }
 }
@@ -458,91 +463,137 @@ __global__ void k_best_overall(const SplitRec* __restrict__ leaf_best, int num_l
 }
 
 // ------------------------------------------------------------------ partition
-/*! single-pass wave-aggregated atomic partition of leaf L: lefts packed ascending from
- *  segment start, rights packed descending from segment end (non-stable; order within a
- *  leaf carries no semantics here — histogram accumulation is atomic anyway). */
-__global__ void k_partition(const uint8_t* __restrict__ colbins,
-                            const uint32_t* __restrict__ idx_base, uint32_t* __restrict__ tmp_base,
-                            const int* __restrict__ leaf_begin, const int* __restrict__ leaf_cnt,
-                            int L, int thr_bin, int nan_bin, int default_left, int cat_onehot,
-                            int* __restrict__ ctr) {
+/*! Deterministic block-scan partition of leaf L (grid-stride, chunked by 256-row
+ *  blocks round-robined over the grid). Three phases: mark+count per block, block
+ *  offset scan, ranked scatter. No same-address global atomics. */
+__device__ __forceinline__ int part_decide(int b, int thr_bin, int nan_bin, int default_left,
+                                           int cat_onehot) {
+  if (cat_onehot) return b == thr_bin ? 1 : 0;
+  if (nan_bin >= 0 && b == nan_bin) return default_left;
+  return b <= thr_bin ? 1 : 0;
+}
+
+__global__ void k_part_mark(const uint8_t* __restrict__ colbins,
+                            const uint32_t* __restrict__ idx_base,
+                            const int* __restrict__ leaf_begin,
+                            const int* __restrict__ leaf_cnt, int L, int thr_bin, int nan_bin,
+                            int default_left, int cat_onehot,
+                            const uint32_t* __restrict__ bits, int n_words,
+                            uint8_t* __restrict__ marks, int* __restrict__ block_cnt) {
+  __shared__ int s_cnt[4];
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const uint32_t* idx = idx_base + begin;
-  uint32_t* tmp = tmp_base + begin;
-  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
-  const int lane = threadIdx.x & 63;
-  const int nthreads = blockDim.x * gridDim.x;
-  const uint64_t lt_mask = (1ull << lane) - 1;
-  for (int i = tid; (i - lane) < cnt; i += nthreads) {
-    const bool active = i < cnt;
-    int go = 0;
-    uint32_t rv = 0;
-    if (active) {
-      rv = idx[i];
-      const int b = colbins[rv];
-      if (cat_onehot) go = b == thr_bin ? 1 : 0;
-      else if (nan_bin >= 0 && b == nan_bin) go = default_left;
-      else go = b <= thr_bin ? 1 : 0;
-    }
-    const uint64_t bl = __ballot(active && go);
-    const uint64_t br = __ballot(active && !go);
-    const int lrank = __popcll(bl & lt_mask);
-    const int rrank = __popcll(br & lt_mask);
-    int lbase = 0, rbase = 0;
-    if (lane == 0) {
-      lbase = atomicAdd(&ctr[0], __popcll(bl));
-      rbase = atomicAdd(&ctr[1], __popcll(br));
-    }
-    lbase = __shfl(lbase, 0);
-    rbase = __shfl(rbase, 0);
-    if (active) {
-      if (go) tmp[lbase + lrank] = rv;
-      else tmp[cnt - 1 - (rbase + rrank)] = rv;
-    }
+  const int chunk_stride = gridDim.x * blockDim.x;
+  int local = 0;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt; i += chunk_stride) {
+    const int b = colbins[idx[i]];
+    int go;
+    if (bits != nullptr) go = ((b >> 5) < n_words && ((bits[b >> 5] >> (b & 31)) & 1)) ? 1 : 0;
+    else go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot);
+    marks[i] = static_cast<uint8_t>(go);
+    local += go;
+  }
+  for (int d = 32; d > 0; d >>= 1) local += __shfl_down(local, d);
+  const int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) s_cnt[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int c = 0;
+    for (int w = 0; w < static_cast<int>(blockDim.x / 64); ++w) c += s_cnt[w];
+    block_cnt[blockIdx.x] = c;
   }
 }
 
-/*! categorical-subset partition variant (bitset over bins). */
-__global__ void k_partition_cat(const uint8_t* __restrict__ colbins,
-                                const uint32_t* __restrict__ idx_base,
-                                uint32_t* __restrict__ tmp_base,
-                                const int* __restrict__ leaf_begin,
-                                const int* __restrict__ leaf_cnt, int L,
-                                const uint32_t* __restrict__ bits, int n_words,
-                                int* __restrict__ ctr) {
+/*! exclusive scan over per-block left counts + per-block own-row counts (closed form
+ *  for the grid-stride decomposition); emits absolute left/right write bases. */
+__global__ void k_part_scan(const int* __restrict__ block_cnt, int nblocks,
+                            const int* __restrict__ leaf_cnt, int L,
+                            int* __restrict__ l_off, int* __restrict__ r_off,
+                            int* __restrict__ ctr) {
+  // single block; serial but tiny (nblocks <= 4096)
+  if (threadIdx.x != 0) return;
+  const int cnt = leaf_cnt[L];
+  const int bs = 256;
+  const int nchunks = (cnt + bs - 1) / bs;
+  int total_left = 0;
+  for (int b = 0; b < nblocks; ++b) total_left += block_cnt[b];
+  int lacc = 0, oacc = 0;
+  const int q = nblocks > 0 ? nchunks / nblocks : 0;
+  const int r = nblocks > 0 ? nchunks % nblocks : 0;
+  const int last_chunk_block = nchunks > 0 ? (nchunks - 1) % nblocks : 0;
+  const int last_chunk_size = nchunks > 0 ? cnt - (nchunks - 1) * bs : 0;
+  for (int b = 0; b < nblocks; ++b) {
+    l_off[b] = lacc;
+    r_off[b] = total_left + (oacc - lacc);
+    lacc += block_cnt[b];
+    int own = (q + (b < r ? 1 : 0)) * bs;
+    if (nchunks > 0 && b == last_chunk_block) own -= bs - last_chunk_size;
+    oacc += own;
+  }
+  ctr[0] = total_left;
+}
+
+__global__ void k_part_scatter(const uint32_t* __restrict__ idx_base,
+                               uint32_t* __restrict__ tmp_base,
+                               const int* __restrict__ leaf_begin,
+                               const int* __restrict__ leaf_cnt, int L,
+                               const uint8_t* __restrict__ marks,
+                               const int* __restrict__ l_off, const int* __restrict__ r_off) {
+  __shared__ int s_l[4], s_n[4];
+  __shared__ int s_lbase, s_rbase;
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const uint32_t* idx = idx_base + begin;
   uint32_t* tmp = tmp_base + begin;
-  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
   const int lane = threadIdx.x & 63;
-  const int nthreads = blockDim.x * gridDim.x;
+  const int wave = threadIdx.x / 64;
   const uint64_t lt_mask = (1ull << lane) - 1;
-  for (int i = tid; (i - lane) < cnt; i += nthreads) {
-    const bool active = i < cnt;
+  if (threadIdx.x == 0) {
+    s_lbase = l_off[blockIdx.x];
+    s_rbase = r_off[blockIdx.x];
+  }
+  __syncthreads();
+  const int chunk_stride = gridDim.x * blockDim.x;
+  // block-uniform loop bound: __syncthreads below requires all waves to iterate together
+  for (int base = blockIdx.x * blockDim.x; base < cnt; base += chunk_stride) {
+    const int i0 = base + static_cast<int>(threadIdx.x);
+    const bool active = i0 < cnt;
     int go = 0;
     uint32_t rv = 0;
     if (active) {
-      rv = idx[i];
-      const int b = colbins[rv];
-      go = (b >> 5) < n_words && ((bits[b >> 5] >> (b & 31)) & 1) ? 1 : 0;
+      rv = idx[i0];
+      go = marks[i0];
     }
     const uint64_t bl = __ballot(active && go);
-    const uint64_t br = __ballot(active && !go);
+    const uint64_t bn = __ballot(active);
     const int lrank = __popcll(bl & lt_mask);
-    const int rrank = __popcll(br & lt_mask);
-    int lbase = 0, rbase = 0;
+    const int nrank = __popcll(bn & lt_mask);
     if (lane == 0) {
-      lbase = atomicAdd(&ctr[0], __popcll(bl));
-      rbase = atomicAdd(&ctr[1], __popcll(br));
+      s_l[wave] = __popcll(bl);
+      s_n[wave] = __popcll(bn);
     }
-    lbase = __shfl(lbase, 0);
-    rbase = __shfl(rbase, 0);
+    __syncthreads();
+    int wl = 0, wn = 0;
+    for (int w = 0; w < wave; ++w) {
+      wl += s_l[w];
+      wn += s_n[w];
+    }
+    int tot_l = wl, tot_n = wn;
+    for (int w = wave; w < static_cast<int>(blockDim.x / 64); ++w) {
+      tot_l += s_l[w];
+      tot_n += s_n[w];
+    }
     if (active) {
-      if (go) tmp[lbase + lrank] = rv;
-      else tmp[cnt - 1 - (rbase + rrank)] = rv;
+      if (go) tmp[s_lbase + wl + lrank] = rv;
+      else tmp[s_rbase + (wn + nrank) - (wl + lrank)] = rv;
     }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      s_lbase += tot_l;
+      s_rbase += tot_n - tot_l;
+    }
+    __syncthreads();
   }
 }
 
@@ -733,8 +784,8 @@ class HIPTreeLearner : public TreeLearner {
   void LaunchHist(int leafA, int leafB, int spare_slot, int approx_cnt);
   void LaunchBestSplit(int leafA, int leafB);
   int HistBlocksFor(int approx_cnt) const {
-    int b = (std::max(approx_cnt, 1) + kHistBlock * 4 - 1) / (kHistBlock * 4);
-    return std::min(4096, std::max(1, b));
+    int b = (std::max(approx_cnt, 1) + kHistBlock * 16 - 1) / (kHistBlock * 16);
+    return std::min(2048, std::max(1, b));
   }
 
   const Config* config_;
@@ -758,6 +809,8 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<double> d_score_;
   DevBuf<float> d_label_, d_weight_;
   DevBuf<uint32_t> d_idx_, d_idx_tmp_;
+  DevBuf<uint8_t> d_marks_;
+  DevBuf<int> d_block_cnt_, d_block_loff_, d_block_roff_;
   DevBuf<int> d_ctr_;
   DevBuf<int64_t> d_gbuf_;
   DevBuf<float> d_hist_;
@@ -877,13 +930,17 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   }
   d_idx_.Alloc(num_data_);
   d_idx_tmp_.Alloc(num_data_);
+  d_marks_.Alloc(num_data_);
+  d_block_cnt_.Alloc(4097);
+  d_block_loff_.Alloc(4097);
+  d_block_roff_.Alloc(4097);
   d_ctr_.Alloc(2);
   d_gbuf_.Alloc(2);
   const int nl = config_->num_leaves;
   d_hist_.Alloc(static_cast<size_t>(nl) * total_bins_ * 2);
   d_feat_best_.Alloc(static_cast<size_t>(2) * nf_);
   d_leaf_best_.Alloc(nl);
-  d_winner_.Alloc(1);
+  d_winner_.Alloc(2);  // rec + trailing winner-leaf int
   d_winner_leaf_.Alloc(1);
   d_leaf_stats_.Alloc(nl);
   d_feat_mask_.Alloc(nf_);
@@ -1059,10 +1116,9 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   int num_leaves = 1;
   for (int split_i = 0; split_i < nl - 1; ++split_i) {
     hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
-                       d_leaf_best_.ptr, num_leaves, d_winner_.ptr, d_winner_leaf_.ptr);
-    HIP_OK(hipMemcpyAsync(h_winner_, d_winner_.ptr, sizeof(hipk::SplitRec),
-                          hipMemcpyDeviceToHost, stream_));
-    HIP_OK(hipMemcpyAsync(h_winner_leaf_, d_winner_leaf_.ptr, sizeof(int),
+                       d_leaf_best_.ptr, num_leaves, d_winner_.ptr,
+                       reinterpret_cast<int*>(d_winner_.ptr + 1));
+    HIP_OK(hipMemcpyAsync(h_winner_, d_winner_.ptr, sizeof(hipk::SplitRec) + sizeof(int),
                           hipMemcpyDeviceToHost, stream_));
     HIP_OK(hipStreamSynchronize(stream_));
     const int L = *h_winner_leaf_;
@@ -1094,24 +1150,27 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
     }
 
     // partition + finalize (all device side; no host sync)
-    HIP_OK(hipMemsetAsync(d_ctr_.ptr, 0, 2 * sizeof(int), stream_));
     const int part_blocks =
-        std::min(40000, std::max(1, (approx_cnt_[L] * 5 / 4 + 4096) / kHistBlock));
+        std::min(4096, std::max(1, (approx_cnt_[L] * 5 / 4 + 4096) / kHistBlock));
     const uint8_t* col = d_cols_.ptr + static_cast<size_t>(f) * num_data_;
+    const uint32_t* cat_bits_arg = nullptr;
     if (mapper->bin_type() == BinType::kCategorical) {
       uint32_t one_bin_bits[8] = {0};
       one_bin_bits[winner.bin >> 5] = 1u << (winner.bin & 31);
       HIP_OK(hipMemcpyAsync(d_cat_bits_.ptr, one_bin_bits, sizeof(one_bin_bits),
                             hipMemcpyHostToDevice, stream_));
-      hipLaunchKernelGGL(hipk::k_partition_cat, dim3(part_blocks), dim3(kHistBlock), 0,
-                         stream_, col, d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr,
-                         d_leaf_cnt_.ptr, L, d_cat_bits_.ptr, 8, d_ctr_.ptr);
-    } else {
-      hipLaunchKernelGGL(hipk::k_partition, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
-                         col, d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
-                         L, winner.bin, m.is_cat ? -1 : m.nan_bin, winner.default_left, 0,
-                         d_ctr_.ptr);
+      cat_bits_arg = d_cat_bits_.ptr;
     }
+    hipLaunchKernelGGL(hipk::k_part_mark, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
+                       col, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L, winner.bin,
+                       m.is_cat ? -1 : m.nan_bin, winner.default_left, 0, cat_bits_arg, 8,
+                       d_marks_.ptr, d_block_cnt_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(64), 0, stream_, d_block_cnt_.ptr,
+                       part_blocks, d_leaf_cnt_.ptr, L, d_block_loff_.ptr, d_block_roff_.ptr,
+                       d_ctr_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_scatter, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
+                       d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L,
+                       d_marks_.ptr, d_block_loff_.ptr, d_block_roff_.ptr);
     hipLaunchKernelGGL(hipk::k_copy_back, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
                        d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L);
     hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
