@@ -505,33 +505,70 @@ __global__ void k_part_mark(const uint8_t* __restrict__ colbins,
   }
 }
 
-/*! exclusive scan over per-block left counts + per-block own-row counts (closed form
- *  for the grid-stride decomposition); emits absolute left/right write bases. */
+/*! parallel exclusive scan over per-block left counts and own-row counts (closed form
+ *  for the grid-stride chunk decomposition). l_off[b] = left write base; r_off[b] holds
+ *  own_prefix - left_prefix (scatter adds total_left from ctr); ctr[0] = total_left. */
 __global__ void k_part_scan(const int* __restrict__ block_cnt, int nblocks,
                             const int* __restrict__ leaf_cnt, int L,
                             int* __restrict__ l_off, int* __restrict__ r_off,
                             int* __restrict__ ctr) {
-  // single block; serial but tiny (nblocks <= 4096)
-  if (threadIdx.x != 0) return;
+  __shared__ int s_wl[4], s_wo[4];
+  __shared__ int carry_l, carry_o;
+  if (threadIdx.x == 0) {
+    carry_l = 0;
+    carry_o = 0;
+  }
+  __syncthreads();
   const int cnt = leaf_cnt[L];
   const int bs = 256;
   const int nchunks = (cnt + bs - 1) / bs;
-  int total_left = 0;
-  for (int b = 0; b < nblocks; ++b) total_left += block_cnt[b];
-  int lacc = 0, oacc = 0;
   const int q = nblocks > 0 ? nchunks / nblocks : 0;
   const int r = nblocks > 0 ? nchunks % nblocks : 0;
-  const int last_chunk_block = nchunks > 0 ? (nchunks - 1) % nblocks : 0;
-  const int last_chunk_size = nchunks > 0 ? cnt - (nchunks - 1) * bs : 0;
-  for (int b = 0; b < nblocks; ++b) {
-    l_off[b] = lacc;
-    r_off[b] = total_left + (oacc - lacc);
-    lacc += block_cnt[b];
-    int own = (q + (b < r ? 1 : 0)) * bs;
-    if (nchunks > 0 && b == last_chunk_block) own -= bs - last_chunk_size;
-    oacc += own;
+  const int last_b = nchunks > 0 ? (nchunks - 1) % nblocks : 0;
+  const int last_sz = nchunks > 0 ? cnt - (nchunks - 1) * bs : 0;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x / 64;
+  for (int base = 0; base < nblocks; base += blockDim.x) {
+    const int b = base + static_cast<int>(threadIdx.x);
+    int vl = b < nblocks ? block_cnt[b] : 0;
+    int vo = 0;
+    if (b < nblocks) {
+      vo = (q + (b < r ? 1 : 0)) * bs;
+      if (nchunks > 0 && b == last_b) vo -= bs - last_sz;
+    }
+    int il = vl, io = vo;
+    for (int d = 1; d < 64; d <<= 1) {
+      const int tl = __shfl_up(il, d);
+      const int to = __shfl_up(io, d);
+      if (lane >= d) {
+        il += tl;
+        io += to;
+      }
+    }
+    if (lane == 63) {
+      s_wl[wave] = il;
+      s_wo[wave] = io;
+    }
+    __syncthreads();
+    int wbl = 0, wbo = 0;
+    for (int w = 0; w < wave; ++w) {
+      wbl += s_wl[w];
+      wbo += s_wo[w];
+    }
+    const int excl_l = carry_l + wbl + il - vl;
+    const int excl_o = carry_o + wbo + io - vo;
+    if (b < nblocks) {
+      l_off[b] = excl_l;
+      r_off[b] = excl_o - excl_l;
+    }
+    __syncthreads();
+    if (threadIdx.x == blockDim.x - 1) {
+      carry_l += wbl + il;
+      carry_o += wbo + io;
+    }
+    __syncthreads();
   }
-  ctr[0] = total_left;
+  if (threadIdx.x == 0) ctr[0] = carry_l;
 }
 
 __global__ void k_part_scatter(const uint32_t* __restrict__ idx_base,
@@ -539,7 +576,8 @@ __global__ void k_part_scatter(const uint32_t* __restrict__ idx_base,
                                const int* __restrict__ leaf_begin,
                                const int* __restrict__ leaf_cnt, int L,
                                const uint8_t* __restrict__ marks,
-                               const int* __restrict__ l_off, const int* __restrict__ r_off) {
+                               const int* __restrict__ l_off, const int* __restrict__ r_off,
+                               const int* __restrict__ ctr) {
   __shared__ int s_l[4], s_n[4];
   __shared__ int s_lbase, s_rbase;
   const int begin = leaf_begin[L];
@@ -551,7 +589,7 @@ __global__ void k_part_scatter(const uint32_t* __restrict__ idx_base,
   const uint64_t lt_mask = (1ull << lane) - 1;
   if (threadIdx.x == 0) {
     s_lbase = l_off[blockIdx.x];
-    s_rbase = r_off[blockIdx.x];
+    s_rbase = ctr[0] + r_off[blockIdx.x];
   }
   __syncthreads();
   const int chunk_stride = gridDim.x * blockDim.x;
@@ -1165,12 +1203,12 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        col, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L, winner.bin,
                        m.is_cat ? -1 : m.nan_bin, winner.default_left, 0, cat_bits_arg, 8,
                        d_marks_.ptr, d_block_cnt_.ptr);
-    hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(64), 0, stream_, d_block_cnt_.ptr,
+    hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(256), 0, stream_, d_block_cnt_.ptr,
                        part_blocks, d_leaf_cnt_.ptr, L, d_block_loff_.ptr, d_block_roff_.ptr,
                        d_ctr_.ptr);
     hipLaunchKernelGGL(hipk::k_part_scatter, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
                        d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L,
-                       d_marks_.ptr, d_block_loff_.ptr, d_block_roff_.ptr);
+                       d_marks_.ptr, d_block_loff_.ptr, d_block_roff_.ptr, d_ctr_.ptr);
     hipLaunchKernelGGL(hipk::k_copy_back, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
                        d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L);
     hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
