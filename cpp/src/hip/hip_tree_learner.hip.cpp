@@ -141,15 +141,16 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
   float* ghist = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
                  static_cast<size_t>(part_bin_base) * 2;
 
-  // LDS layout [bin][NCOPIES][g,h]: the NCOPIES copies of one bin sit in ADJACENT
-  // words, i.e. different banks — the degenerate all-rows-same-bin leaf serializes
-  // through NCOPIES banks in parallel instead of one.
-  extern __shared__ float lh[];  // part_bins * NCOPIES * 2
+  // LDS layout [bin][NCOPIES][g,h] with a non-power-of-2 bin stride (2*NCOPIES+2):
+  // copies of one bin sit in adjacent banks (same-bin leaves drain through NCOPIES
+  // banks in parallel) and consecutive bins rotate across all 32 banks.
+  constexpr int kBinStride = 2 * NCOPIES + 2;
+  extern __shared__ float lh[];  // part_bins * kBinStride
   __shared__ int loff[256];
   const int nfeat = feat_end - feat_begin;
   for (int i = threadIdx.x; i < nfeat; i += blockDim.x)
     loff[i] = fm[feat_begin + i].bin_off - part_bin_base;
-  const int nelem = part_bins * 2 * NCOPIES;
+  const int nelem = part_bins * kBinStride;
   for (int i = threadIdx.x; i < nelem; i += blockDim.x) lh[i] = 0.0f;
   __syncthreads();
 
@@ -170,7 +171,7 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
         const int f = c + j;
         if (f < feat_begin || f >= feat_end) continue;
         const int b = (w[j >> 2] >> ((j & 3) * 8)) & 0xFF;
-        float* dst = lh + (loff[f - feat_begin] + b) * (2 * NCOPIES) + my_copy;
+        float* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
         atomicAdd(dst, gi);
         atomicAdd(dst + 1, hi);
       }
@@ -182,7 +183,7 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
     const int gh = i & 1;
     float v = 0.0f;
 #pragma unroll
-    for (int cpy = 0; cpy < NCOPIES; ++cpy) v += lh[bin * 2 * NCOPIES + cpy * 2 + gh];
+    for (int cpy = 0; cpy < NCOPIES; ++cpy) v += lh[bin * kBinStride + cpy * 2 + gh];
     if (v != 0.0f) atomicAdd(&ghist[i], v);
   }
 }
@@ -822,7 +823,9 @@ class HIPTreeLearner : public TreeLearner {
   void LaunchHist(int leafA, int leafB, int spare_slot, int approx_cnt);
   void LaunchBestSplit(int leafA, int leafB);
   int HistBlocksFor(int approx_cnt) const {
-    int b = (std::max(approx_cnt, 1) + kHistBlock * 16 - 1) / (kHistBlock * 16);
+    // LDS atomic throughput is per-CU: spread even small leaves over many blocks
+    // (~256 rows each); cap so the per-block flush stays amortized at the root.
+    int b = (std::max(approx_cnt, 1) + 255) / 256;
     return std::min(2048, std::max(1, b));
   }
 
@@ -918,7 +921,7 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   auto build_partitions = [&](int copies) -> bool {
     feat_partitions_.clear();
     part_bin_range_.clear();
-    const int max_bins = kLdsBudget / (2 * sizeof(float) * copies);
+    const int max_bins = kLdsBudget / ((2 * copies + 2) * sizeof(float));
     int begin = 0;
     while (begin < nf_) {
       int end = begin;
@@ -1050,7 +1053,7 @@ void HIPTreeLearner::LaunchHist(int leafA, int leafB, int spare_slot, int approx
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
     const auto [fb, fe] = feat_partitions_[pr];
     const auto [bin_base, bins] = part_bin_range_[pr];
-    const size_t lds = static_cast<size_t>(bins) * 2 * sizeof(float) * n_copies_;
+    const size_t lds = static_cast<size_t>(bins) * (2 * n_copies_ + 2) * sizeof(float);
     switch (n_copies_) {
       case 4:
         hipLaunchKernelGGL(hipk::k_hist<4>, dim3(blocks), dim3(kHistBlock), lds, stream_,
