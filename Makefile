@@ -30,7 +30,14 @@ else
   EXTRA_LIBS = -lamdhip64 -lrccl
 endif
 
-all: $(TARGET)
+CLI := lightgbm_amd/bin/migbm
+
+all: $(TARGET) $(CLI)
+
+$(CLI): cpp/cli/main.cpp $(TARGET)
+	mkdir -p lightgbm_amd/bin
+	$(CXX) $(CXXFLAGS) cpp/cli/main.cpp -Llightgbm_amd/lib -l_migbm \
+	  -Wl,-rpath,'$$ORIGIN/../lib' -o $(CLI)
 
 $(BUILD):
 	mkdir -p $(BUILD) $(BUILD)/hip $(LIBDIR)
@@ -45,7 +52,7 @@ $(TARGET): $(HOST_OBJS) $(HIP_OBJS) | $(BUILD)
 	$(LINKER) $(LDFLAGS) $(HOST_OBJS) $(HIP_OBJS) $(EXTRA_LIBS) -o $@
 
 clean:
-	rm -rf $(BUILD) $(TARGET)
+	rm -rf $(BUILD) $(TARGET) $(CLI)
 
 -include $(BUILD)/*.d $(BUILD)/hip/*.d
 

@@ -1,0 +1,199 @@
+/*! migbm CLI — `migbm config=train.conf [key=value ...]`.
+ *  Capability parity target: reference src/main.cpp + src/application/application.cpp
+ *  (tasks train / predict / refit; config-file + argv key=value parsing). */
+#include "migbm/boosting.h"
+#include "migbm/common.h"
+#include "migbm/config.h"
+#include "migbm/dataset.h"
+#include "migbm/metric.h"
+#include "migbm/objective.h"
+
+#include <cstdio>
+#include <fstream>
+#include <memory>
+#include <string>
+
+namespace migbm {
+
+class Application {
+ public:
+  explicit Application(int argc, char** argv) {
+    std::unordered_map<std::string, std::string> params;
+    for (int i = 1; i < argc; ++i) {
+      auto kv = Common::Split(argv[i], '=');
+      if (kv.size() == 2) params[Common::ToLower(Common::Trim(kv[0]))] = Common::Trim(kv[1]);
+    }
+    // config file first, argv overrides
+    auto it = params.find("config");
+    if (it != params.end()) {
+      std::ifstream f(it->second);
+      if (!f.good()) Log::Fatal("Cannot open config file %s", it->second.c_str());
+      std::string line;
+      std::unordered_map<std::string, std::string> file_params;
+      while (std::getline(f, line)) {
+        auto hash = line.find('#');
+        if (hash != std::string::npos) line = line.substr(0, hash);
+        line = Common::Trim(line);
+        if (line.empty()) continue;
+        auto eq = line.find('=');
+        if (eq == std::string::npos) continue;
+        file_params[Common::ToLower(Common::Trim(line.substr(0, eq)))] =
+            Common::Trim(line.substr(eq + 1));
+      }
+      for (auto& kv : params) file_params[kv.first] = kv.second;  // argv wins
+      params = std::move(file_params);
+    }
+    config_.Set(params);
+  }
+
+  void Run() {
+    if (config_.task == "train" || config_.task == "refit") Train();
+    else if (config_.task == "predict" || config_.task == "prediction" ||
+             config_.task == "test")
+      Predict();
+    else if (config_.task == "convert_model")
+      ConvertModel();
+    else
+      Log::Fatal("Unknown task %s", config_.task.c_str());
+  }
+
+ private:
+  void LoadData() {
+    if (config_.data.empty()) Log::Fatal("No training data (data=...) specified");
+    DatasetLoader loader(config_);
+    train_data_ = loader.LoadFromFile(config_.data.c_str());
+    Log::Info("Loaded train data: %d rows, %d features", train_data_->num_data(),
+              train_data_->num_total_features());
+    for (auto& vf : config_.valid) {
+      if (vf.empty()) continue;
+      valid_data_.push_back(loader.LoadFromFileAlignWithOtherDataset(vf.c_str(),
+                                                                     train_data_.get()));
+      Log::Info("Loaded valid data %s: %d rows", vf.c_str(),
+                valid_data_.back()->num_data());
+    }
+    if (config_.save_binary) {
+      std::string out = config_.data + ".bin";
+      train_data_->SaveBinaryFile(out.c_str());
+      Log::Info("Saved binary dataset to %s", out.c_str());
+    }
+  }
+
+  void Train() {
+    LoadData();
+    objective_.reset(ObjectiveFunction::Create(config_.objective, config_));
+    if (objective_) objective_->Init(train_data_->metadata(), train_data_->num_data());
+    auto metric_names = config_.metric;
+    if (metric_names.empty())
+      metric_names.push_back(config_.objective == "regression" ? "l2" : config_.objective);
+    for (auto& name : metric_names) {
+      std::unique_ptr<Metric> m(Metric::Create(name, config_));
+      if (m) {
+        m->Init(train_data_->metadata(), train_data_->num_data());
+        train_metrics_.push_back(std::move(m));
+      }
+    }
+    boosting_.reset(GBDT::CreateBoosting(
+        config_.boosting, config_.input_model.empty() ? nullptr
+                                                      : config_.input_model.c_str()));
+    std::vector<const Metric*> tm;
+    for (auto& m : train_metrics_) tm.push_back(m.get());
+    boosting_->Init(&config_, train_data_.get(), objective_.get(), tm);
+    for (auto& vd : valid_data_) {
+      std::vector<std::unique_ptr<Metric>> vms;
+      for (auto& name : metric_names) {
+        std::unique_ptr<Metric> m(Metric::Create(name, config_));
+        if (m) {
+          m->Init(vd->metadata(), vd->num_data());
+          vms.push_back(std::move(m));
+        }
+      }
+      std::vector<const Metric*> vmp;
+      for (auto& m : vms) vmp.push_back(m.get());
+      valid_metrics_.push_back(std::move(vms));
+      boosting_->AddValidDataset(vd.get(), vmp);
+    }
+    Log::Info("Started training for %d iterations...", config_.num_iterations);
+    double t0 = Timer::Now();
+    boosting_->Train(config_.snapshot_freq, config_.output_model);
+    Log::Info("Finished training in %.3f seconds", Timer::Now() - t0);
+    boosting_->SaveModelToFile(0, -1, 0, config_.output_model.c_str());
+    Log::Info("Model saved to %s", config_.output_model.c_str());
+  }
+
+  void Predict() {
+    if (config_.input_model.empty()) Log::Fatal("task=predict requires input_model=");
+    boosting_.reset(GBDT::CreateBoosting("gbdt", config_.input_model.c_str()));
+    auto rows = LoadRawRowsForPredict(config_.data.c_str(), config_,
+                                      boosting_->MaxFeatureIdx() + 1);
+    int ptype = 0;
+    if (config_.predict_raw_score) ptype = 1;
+    if (config_.predict_leaf_index) ptype = 2;
+    if (config_.predict_contrib) ptype = 3;
+    const int ncol = boosting_->MaxFeatureIdx() + 1;
+    const int per_row = boosting_->NumPredictOneRow(config_.start_iteration_predict,
+                                                    config_.num_iteration_predict,
+                                                    ptype == 2, ptype == 3);
+    FILE* fp = fopen(config_.output_result.c_str(), "w");
+    if (!fp) Log::Fatal("Cannot open %s", config_.output_result.c_str());
+    std::vector<double> out(per_row);
+    std::vector<double> feats(ncol);
+    for (auto& r : rows) {
+      for (int c = 0; c < ncol; ++c)
+        feats[c] = c < static_cast<int>(r.size()) ? r[c] : 0.0;
+      switch (ptype) {
+        case 0: boosting_->Predict(feats.data(), out.data(),
+                                   config_.start_iteration_predict,
+                                   config_.num_iteration_predict); break;
+        case 1: boosting_->PredictRaw(feats.data(), out.data(),
+                                      config_.start_iteration_predict,
+                                      config_.num_iteration_predict); break;
+        case 2: boosting_->PredictLeafIndex(feats.data(), out.data(),
+                                            config_.start_iteration_predict,
+                                            config_.num_iteration_predict); break;
+        case 3: boosting_->PredictContrib(feats.data(), out.data(),
+                                          config_.start_iteration_predict,
+                                          config_.num_iteration_predict); break;
+      }
+      for (int k = 0; k < per_row; ++k) {
+        if (k) fputc('\t', fp);
+        fprintf(fp, "%.17g", out[k]);
+      }
+      fputc('\n', fp);
+    }
+    fclose(fp);
+    Log::Info("Predictions written to %s", config_.output_result.c_str());
+  }
+
+  void ConvertModel() {
+    if (config_.input_model.empty()) Log::Fatal("convert_model requires input_model=");
+    boosting_.reset(GBDT::CreateBoosting("gbdt", config_.input_model.c_str()));
+    std::string json = boosting_->DumpModel(0, -1, 0);
+    std::string out = config_.output_model.empty() ? "model.json" : config_.output_model;
+    FILE* fp = fopen(out.c_str(), "w");
+    if (!fp) Log::Fatal("Cannot open %s", out.c_str());
+    fwrite(json.data(), 1, json.size(), fp);
+    fclose(fp);
+    Log::Info("Model JSON written to %s", out.c_str());
+  }
+
+  Config config_;
+  std::unique_ptr<Dataset> train_data_;
+  std::vector<std::unique_ptr<Dataset>> valid_data_;
+  std::unique_ptr<ObjectiveFunction> objective_;
+  std::vector<std::unique_ptr<Metric>> train_metrics_;
+  std::vector<std::vector<std::unique_ptr<Metric>>> valid_metrics_;
+  std::unique_ptr<GBDT> boosting_;
+};
+
+}  // namespace migbm
+
+int main(int argc, char** argv) {
+  try {
+    migbm::Application app(argc, argv);
+    app.Run();
+  } catch (const std::exception& ex) {
+    fprintf(stderr, "%s\n", ex.what());
+    return 1;
+  }
+  return 0;
+}

@@ -1,0 +1,73 @@
+"""CLI application tests (parity target: reference test_consistency.py — CLI-config
+examples vs Python API on the same data)."""
+import subprocess
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+import lightgbm_amd as lgb
+
+REPO = Path(__file__).resolve().parent.parent
+CLI = REPO / "lightgbm_amd" / "bin" / "migbm"
+EXAMPLE = REPO / "examples" / "binary_classification"
+
+
+@pytest.fixture(scope="module")
+def cli_model(tmp_path_factory):
+    out = tmp_path_factory.mktemp("cli")
+    model = out / "model.txt"
+    r = subprocess.run(
+        [str(CLI), f"config={EXAMPLE/'train.conf'}", f"data={EXAMPLE/'binary.train'}",
+         f"valid_data={EXAMPLE/'binary.test'}", f"output_model={model}",
+         "num_trees=30", "verbosity=-1"],
+        capture_output=True, text=True, timeout=300, cwd=str(out))
+    assert r.returncode == 0, r.stderr
+    return model
+
+
+def _load_tsv(path):
+    data = np.loadtxt(path, delimiter="\t")
+    return data[:, 1:], data[:, 0]
+
+
+def test_cli_trains_and_predicts(cli_model, tmp_path):
+    assert cli_model.exists()
+    result = tmp_path / "pred.txt"
+    r = subprocess.run(
+        [str(CLI), "task=predict", f"data={EXAMPLE/'binary.test'}",
+         f"input_model={cli_model}", f"output_result={result}"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    preds = np.loadtxt(result)
+    X, y = _load_tsv(EXAMPLE / "binary.test")
+    assert len(preds) == len(y)
+    acc = ((preds > 0.5) == y).mean()
+    assert acc > 0.75
+
+
+def test_cli_python_consistency(cli_model):
+    """The Python API must produce the same predictions from the CLI-trained model."""
+    X, y = _load_tsv(EXAMPLE / "binary.test")
+    bst = lgb.Booster(model_file=str(cli_model))
+    pred_py = bst.predict(X)
+    # CLI predict on the same file
+    import subprocess, tempfile
+    with tempfile.TemporaryDirectory() as td:
+        result = Path(td) / "pred.txt"
+        subprocess.run([str(CLI), "task=predict", f"data={EXAMPLE/'binary.test'}",
+                        f"input_model={cli_model}", f"output_result={result}"],
+                       capture_output=True, timeout=300, check=True)
+        pred_cli = np.loadtxt(result)
+    np.testing.assert_allclose(pred_py, pred_cli, rtol=1e-10)
+
+
+def test_cli_python_same_training():
+    """Training via Python on the example files reaches similar quality."""
+    X, y = _load_tsv(EXAMPLE / "binary.train")
+    Xt, yt = _load_tsv(EXAMPLE / "binary.test")
+    bst = lgb.train({"objective": "binary", "num_leaves": 63, "min_data_in_leaf": 50,
+                     "min_sum_hessian_in_leaf": 5.0, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 30)
+    acc = ((bst.predict(Xt) > 0.5) == yt).mean()
+    assert acc > 0.75

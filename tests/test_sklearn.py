@@ -1,0 +1,84 @@
+"""sklearn wrapper tests (parity target: reference test_sklearn.py, core subset)."""
+import numpy as np
+import pytest
+
+import lightgbm_amd as lgb
+
+
+def test_regressor():
+    rng = np.random.RandomState(0)
+    X = rng.randn(3000, 8)
+    y = 2 * X[:, 0] + np.sin(X[:, 1]) + 0.1 * rng.randn(3000)
+    m = lgb.LGBMRegressor(n_estimators=50, random_state=0)
+    m.fit(X, y)
+    assert m.score(X, y) > 0.9
+    assert m.n_features_in_ == 8
+    assert len(m.feature_importances_) == 8
+
+
+def test_classifier_binary():
+    rng = np.random.RandomState(1)
+    X = rng.randn(3000, 6)
+    y = np.where(X[:, 0] + 0.5 * rng.randn(3000) > 0, "pos", "neg")
+    m = lgb.LGBMClassifier(n_estimators=30)
+    m.fit(X, y)
+    assert set(m.classes_) == {"neg", "pos"}
+    proba = m.predict_proba(X[:10])
+    assert proba.shape == (10, 2)
+    np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-9)
+    assert m.score(X, y) > 0.9
+
+
+def test_classifier_multiclass():
+    rng = np.random.RandomState(2)
+    X = rng.randn(3000, 6)
+    y = (X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)
+    m = lgb.LGBMClassifier(n_estimators=30)
+    m.fit(X, y)
+    assert m.n_classes_ == 3
+    assert m.predict_proba(X[:5]).shape == (5, 3)
+    assert m.score(X, y) > 0.8
+
+
+def test_classifier_class_weight():
+    rng = np.random.RandomState(3)
+    X = rng.randn(3000, 4)
+    y = (X[:, 0] + 0.8 * rng.randn(3000) > 1.0).astype(int)  # imbalanced
+    m = lgb.LGBMClassifier(n_estimators=20, class_weight="balanced")
+    m.fit(X, y)
+    # balanced weighting should raise recall on the minority class
+    pred = m.predict(X)
+    recall = (pred[y == 1] == 1).mean()
+    assert recall > 0.5
+
+
+def test_ranker():
+    rng = np.random.RandomState(4)
+    groups = [20] * 50
+    X = rng.randn(sum(groups), 5)
+    y = (X[:, 0] + 0.3 * rng.randn(len(X)) > 0.5).astype(int)
+    m = lgb.LGBMRanker(n_estimators=20)
+    m.fit(X, y, group=groups)
+    pred = m.predict(X[:20])
+    assert pred.shape == (20,)
+
+
+def test_early_stopping_fit():
+    rng = np.random.RandomState(5)
+    X = rng.randn(4000, 6)
+    y = X[:, 0] + 0.2 * rng.randn(4000)
+    m = lgb.LGBMRegressor(n_estimators=500)
+    m.fit(X[:3000], y[:3000], eval_set=[(X[3000:], y[3000:])],
+          eval_metric="l2", early_stopping_rounds=5)
+    assert m.best_iteration_ > 0
+    assert m.best_iteration_ < 500
+    assert "valid_0" in m.evals_result_
+
+
+def test_get_set_params():
+    m = lgb.LGBMRegressor(num_leaves=15, custom_thing=7)
+    p = m.get_params()
+    assert p["num_leaves"] == 15
+    assert p["custom_thing"] == 7
+    m.set_params(num_leaves=31)
+    assert m.get_params()["num_leaves"] == 31
