@@ -70,6 +70,12 @@ struct SplitRec {
   int valid;
 };
 
+struct LogEntry {
+  SplitRec rec;
+  int leaf;
+  int pad;
+};
+
 struct LeafStat {
   double sum_g, sum_h;
   int cnt;  // exact GLOBAL row count (allreduced in multi-GPU)
@@ -129,10 +135,14 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
                        const uint32_t* __restrict__ idx_base,
                        const int* __restrict__ leaf_begin, const int* __restrict__ leaf_cnt,
                        const LeafStat* __restrict__ stats, const int* __restrict__ leaf_slot,
-                       int leafA, int leafB, const float* __restrict__ g,
+                       const int* __restrict__ leafA_ptr, const int* __restrict__ counters,
+                       int leafB_from_counters, const float* __restrict__ g,
                        const float* __restrict__ h, const FeatMeta* __restrict__ fm,
                        int feat_begin, int feat_end, int part_bin_base, int part_bins,
                        float* __restrict__ hist_base, size_t slot_stride) {
+  const int leafA = *leafA_ptr;
+  if (leafA < 0) return;
+  const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
   int leaf = leafA;
   if (leafB >= 0 && stats[leafB].cnt < stats[leafA].cnt) leaf = leafB;
   const int begin = leaf_begin[leaf];
@@ -188,15 +198,25 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
   }
 }
 
-__global__ void k_hist_zero(float* hist, int n) {
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) hist[i] = 0.0f;
+__global__ void k_hist_zero(float* hist_base, size_t slot_stride,
+                            const int* __restrict__ counters, int slot_from_counters,
+                            int literal_slot, const int* __restrict__ leafA_ptr, int n) {
+  if (*leafA_ptr < 0) return;
+  const int slot = slot_from_counters ? counters[0] - 1 : literal_slot;
+  float* hist = hist_base + static_cast<size_t>(slot) * slot_stride;
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (int i = tid; i < n; i += blockDim.x * gridDim.x) hist[i] = 0.0f;
 }
 
 /*! larger-child histogram = parent (in place at the larger leaf's slot) - smaller. */
 __global__ void k_hist_subtract(float* __restrict__ hist_base, size_t slot_stride,
                                 const int* __restrict__ leaf_slot,
-                                const LeafStat* __restrict__ stats, int L, int R, int n) {
+                                const LeafStat* __restrict__ stats,
+                                const int* __restrict__ Lptr,
+                                const int* __restrict__ counters, int n) {
+  const int L = *Lptr;
+  if (L < 0) return;
+  const int R = counters[0] - 1;
   const int smaller = stats[L].cnt <= stats[R].cnt ? L : R;
   const int larger = smaller == L ? R : L;
   float* big = hist_base + static_cast<size_t>(leaf_slot[larger]) * slot_stride;
@@ -207,7 +227,8 @@ __global__ void k_hist_subtract(float* __restrict__ hist_base, size_t slot_strid
 
 // ------------------------------------------------------------------ root setup
 __global__ void k_init_root(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
-                            int used_cnt, int64_t* gbuf) {
+                            int used_cnt, int64_t* gbuf, int* counters, int* root_leaf,
+                            int* minus1) {
   leaf_begin[0] = 0;
   leaf_cnt[0] = used_cnt;
   leaf_slot[0] = 0;
@@ -215,6 +236,10 @@ __global__ void k_init_root(int* leaf_begin, int* leaf_cnt, int* leaf_slot, Leaf
   stats[0].sum_h = 0.0;
   stats[0].cnt = used_cnt;
   gbuf[0] = used_cnt;
+  counters[0] = 1;  // num_leaves
+  counters[1] = 0;  // split log index
+  *root_leaf = 0;
+  *minus1 = -1;
 }
 
 __global__ void k_root_sums(const uint32_t* __restrict__ idx, int cnt,
@@ -257,11 +282,14 @@ __global__ void k_set_root_global_cnt(LeafStat* stats, const int64_t* gbuf) {
 /*! one wave per (feature, child); blockIdx.y selects leafA/leafB. */
 __global__ void __launch_bounds__(64) k_best_feat(
     const float* __restrict__ hist_base, size_t slot_stride, const int* __restrict__ leaf_slot,
-    const FeatMeta* __restrict__ fm, int nf, const LeafStat* __restrict__ stats, int leafA,
-    int leafB, GainParams p, const int8_t* __restrict__ feat_mask,
+    const FeatMeta* __restrict__ fm, int nf, const LeafStat* __restrict__ stats,
+    const int* __restrict__ leafA_ptr, const int* __restrict__ counters,
+    int leafB_from_counters, GainParams p, const int8_t* __restrict__ feat_mask,
     SplitRec* __restrict__ out) {
   const int f = blockIdx.x;
   const int which = blockIdx.y;
+  const int leafA = *leafA_ptr;
+  const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
   const int leaf = which == 0 ? leafA : leafB;
   if (f >= nf || leaf < 0) return;
   const int lane = threadIdx.x;
@@ -390,8 +418,12 @@ __global__ void __launch_bounds__(64) k_best_feat(
 
 /*! reduce per-feature records to one per leaf; blockIdx.x = which child. */
 __global__ void k_best_leaf(const SplitRec* __restrict__ feat_best, int nf,
-                            SplitRec* __restrict__ leaf_best, int leafA, int leafB) {
+                            SplitRec* __restrict__ leaf_best,
+                            const int* __restrict__ leafA_ptr,
+                            const int* __restrict__ counters, int leafB_from_counters) {
   const int which = blockIdx.x;
+  const int leafA = *leafA_ptr;
+  const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
   const int leaf = which == 0 ? leafA : leafB;
   if (leaf < 0) return;
   const SplitRec* cand = feat_best + which * nf;
@@ -430,8 +462,10 @@ __global__ void k_best_leaf(const SplitRec* __restrict__ feat_best, int nf,
   }
 }
 
-__global__ void k_best_overall(const SplitRec* __restrict__ leaf_best, int num_leaves,
+__global__ void k_best_overall(const SplitRec* __restrict__ leaf_best,
+                               const int* __restrict__ counters,
                                SplitRec* __restrict__ winner, int* __restrict__ winner_leaf) {
+  const int num_leaves = counters[0];
   __shared__ int s_idx[256];
   __shared__ double s_gain[256];
   const int tid = threadIdx.x;
@@ -474,24 +508,31 @@ __device__ __forceinline__ int part_decide(int b, int thr_bin, int nan_bin, int 
   return b <= thr_bin ? 1 : 0;
 }
 
-__global__ void k_part_mark(const uint8_t* __restrict__ colbins,
-                            const uint32_t* __restrict__ idx_base,
+__global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
                             const int* __restrict__ leaf_begin,
-                            const int* __restrict__ leaf_cnt, int L, int thr_bin, int nan_bin,
-                            int default_left, int cat_onehot,
-                            const uint32_t* __restrict__ bits, int n_words,
-                            uint8_t* __restrict__ marks, int* __restrict__ block_cnt) {
+                            const int* __restrict__ leaf_cnt,
+                            const int* __restrict__ Lptr, const SplitRec* __restrict__ win,
+                            const FeatMeta* __restrict__ fm, const uint8_t* __restrict__ cols,
+                            int num_data, uint8_t* __restrict__ marks,
+                            int* __restrict__ block_cnt) {
   __shared__ int s_cnt[4];
+  const int L = *Lptr;
+  if (L < 0) return;
+  const int f = win->feature;
+  const FeatMeta m = fm[f];
+  const uint8_t* colbins2 = cols + static_cast<size_t>(f) * num_data;
+  const int thr_bin = win->bin;
+  const int nan_bin = m.is_cat ? -1 : m.nan_bin;
+  const int default_left = win->default_left;
+  const int cat_onehot = m.is_cat;
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const uint32_t* idx = idx_base + begin;
   const int chunk_stride = gridDim.x * blockDim.x;
   int local = 0;
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt; i += chunk_stride) {
-    const int b = colbins[idx[i]];
-    int go;
-    if (bits != nullptr) go = ((b >> 5) < n_words && ((bits[b >> 5] >> (b & 31)) & 1)) ? 1 : 0;
-    else go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot);
+    const int b = colbins2[idx[i]];
+    const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot);
     marks[i] = static_cast<uint8_t>(go);
     local += go;
   }
@@ -510,9 +551,12 @@ __global__ void k_part_mark(const uint8_t* __restrict__ colbins,
  *  for the grid-stride chunk decomposition). l_off[b] = left write base; r_off[b] holds
  *  own_prefix - left_prefix (scatter adds total_left from ctr); ctr[0] = total_left. */
 __global__ void k_part_scan(const int* __restrict__ block_cnt, int nblocks,
-                            const int* __restrict__ leaf_cnt, int L,
+                            const int* __restrict__ leaf_cnt,
+                            const int* __restrict__ Lptr,
                             int* __restrict__ l_off, int* __restrict__ r_off,
                             int* __restrict__ ctr) {
+  const int L = *Lptr;
+  if (L < 0) return;
   __shared__ int s_wl[4], s_wo[4];
   __shared__ int carry_l, carry_o;
   if (threadIdx.x == 0) {
@@ -575,10 +619,13 @@ __global__ void k_part_scan(const int* __restrict__ block_cnt, int nblocks,
 __global__ void k_part_scatter(const uint32_t* __restrict__ idx_base,
                                uint32_t* __restrict__ tmp_base,
                                const int* __restrict__ leaf_begin,
-                               const int* __restrict__ leaf_cnt, int L,
+                               const int* __restrict__ leaf_cnt,
+                               const int* __restrict__ Lptr,
                                const uint8_t* __restrict__ marks,
                                const int* __restrict__ l_off, const int* __restrict__ r_off,
                                const int* __restrict__ ctr) {
+  const int L = *Lptr;
+  if (L < 0) return;
   __shared__ int s_l[4], s_n[4];
   __shared__ int s_lbase, s_rbase;
   const int begin = leaf_begin[L];
@@ -639,7 +686,10 @@ __global__ void k_part_scatter(const uint32_t* __restrict__ idx_base,
 __global__ void k_copy_back(const uint32_t* __restrict__ tmp_base,
                             uint32_t* __restrict__ idx_base,
                             const int* __restrict__ leaf_begin,
-                            const int* __restrict__ leaf_cnt, int L) {
+                            const int* __restrict__ leaf_cnt,
+                            const int* __restrict__ Lptr) {
+  const int L = *Lptr;
+  if (L < 0) return;
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
@@ -647,14 +697,29 @@ __global__ void k_copy_back(const uint32_t* __restrict__ tmp_base,
     idx_base[begin + i] = tmp_base[begin + i];
 }
 
-__global__ void k_store_left(const int* __restrict__ ctr, int64_t* gbuf) {
+__global__ void k_store_left(const int* __restrict__ ctr, const int* __restrict__ Lptr,
+                             int64_t* gbuf) {
+  if (*Lptr < 0) return;
   gbuf[0] = ctr[0];
 }
 
-/*! device-side split bookkeeping: segments, stats, slot map. One thread. */
+/*! device-side split bookkeeping: segments, stats, slot map, split log. One thread. */
 __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
-                           const SplitRec* __restrict__ winner, int L, int R, int spare_slot,
-                           const int* __restrict__ ctr, const int64_t* __restrict__ gbuf) {
+                           const SplitRec* __restrict__ winner,
+                           const int* __restrict__ Lptr, int* counters,
+                           LogEntry* __restrict__ log, const int* __restrict__ ctr,
+                           const int64_t* __restrict__ gbuf) {
+  const int L = *Lptr;
+  if (L < 0) {
+    log[counters[1]].leaf = -1;  // terminator for the host replay
+    return;
+  }
+  const int R = counters[0];
+  const int spare_slot = R;
+  log[counters[1]].rec = *winner;
+  log[counters[1]].leaf = L;
+  counters[1] += 1;
+  counters[0] += 1;
   const int left_local = ctr[0];
   const int parent_local = leaf_cnt[L];
   leaf_begin[R] = leaf_begin[L] + left_local;
@@ -820,8 +885,9 @@ class HIPTreeLearner : public TreeLearner {
 
  private:
   void UploadGradients(const score_t* g, const score_t* h);
-  void LaunchHist(int leafA, int leafB, int spare_slot, int approx_cnt);
-  void LaunchBestSplit(int leafA, int leafB);
+  void LaunchHist(const int* leafA_ptr, int leafB_from_counters, int blocks);
+  void ReduceSpareHist(int spare_slot);
+  void LaunchBestSplit(const int* leafA_ptr, int leafB_from_counters);
   int HistBlocksFor(int approx_cnt) const {
     // LDS atomic throughput is per-CU: spread even small leaves over many blocks
     // (~256 rows each); cap so the per-block flush stays amortized at the root.
@@ -859,6 +925,9 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<hipk::SplitRec> d_leaf_best_;
   DevBuf<hipk::SplitRec> d_winner_;
   DevBuf<int> d_winner_leaf_;
+  DevBuf<int> d_counters_;      // [num_leaves, split_log_index]
+  DevBuf<int> d_root_leaf_, d_minus1_;
+  DevBuf<hipk::LogEntry> d_split_log_;
   DevBuf<hipk::LeafStat> d_leaf_stats_;
   DevBuf<int8_t> d_feat_mask_;
   DevBuf<int> d_leaf_begin_, d_leaf_cnt_, d_leaf_slot_;
@@ -870,8 +939,9 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<double> d_tw_out_;
   DevBuf<uint32_t> d_oob_;
 
-  hipk::SplitRec* h_winner_ = nullptr;  // pinned: winner rec + leaf id
+  hipk::SplitRec* h_winner_ = nullptr;  // pinned staging (legacy; log path reads below)
   int* h_winner_leaf_ = nullptr;
+  std::vector<hipk::LogEntry> host_log_;
 
   bool grads_on_device_ = false;
   bool weights_present_ = false;
@@ -983,6 +1053,10 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_leaf_best_.Alloc(nl);
   d_winner_.Alloc(2);  // rec + trailing winner-leaf int
   d_winner_leaf_.Alloc(1);
+  d_counters_.Alloc(2);
+  d_root_leaf_.Alloc(1);
+  d_minus1_.Alloc(1);
+  d_split_log_.Alloc(nl);
   d_leaf_stats_.Alloc(nl);
   d_feat_mask_.Alloc(nf_);
   d_leaf_begin_.Alloc(nl);
@@ -1042,14 +1116,14 @@ void HIPTreeLearner::DeviceBoosting(const ObjectiveFunction* obj) {
   grads_on_device_ = true;
 }
 
-void HIPTreeLearner::LaunchHist(int leafA, int leafB, int spare_slot, int approx_cnt) {
-  // zero the spare slot, then accumulate the smaller child's histogram into it
-  float* spare = d_hist_.ptr + static_cast<size_t>(spare_slot) * total_bins_ * 2;
+void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
+                                int blocks) {
   const int n_elem = total_bins_ * 2;
-  hipLaunchKernelGGL(hipk::k_hist_zero, dim3((n_elem + 255) / 256), dim3(256), 0, stream_,
-                     spare, n_elem);
-  const int blocks = HistBlocksFor(approx_cnt);
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
+  // zero the spare slot (R = counters[0]-1 after finalize; literal slot 0 for the root)
+  hipLaunchKernelGGL(hipk::k_hist_zero, dim3(256), dim3(256), 0, stream_, d_hist_.ptr,
+                     slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
+                     n_elem);
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
     const auto [fb, fe] = feat_partitions_[pr];
     const auto [bin_base, bins] = part_bin_range_[pr];
@@ -1058,34 +1132,39 @@ void HIPTreeLearner::LaunchHist(int leafA, int leafB, int spare_slot, int approx
       case 4:
         hipLaunchKernelGGL(hipk::k_hist<4>, dim3(blocks), dim3(kHistBlock), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA,
-                           leafB, d_grad_.ptr, d_hess_.ptr, d_feat_meta_.ptr, fb, fe,
-                           bin_base, bins, d_hist_.ptr, slot_stride);
+                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
+                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                           slot_stride);
         break;
       case 2:
         hipLaunchKernelGGL(hipk::k_hist<2>, dim3(blocks), dim3(kHistBlock), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA,
-                           leafB, d_grad_.ptr, d_hess_.ptr, d_feat_meta_.ptr, fb, fe,
-                           bin_base, bins, d_hist_.ptr, slot_stride);
+                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
+                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                           slot_stride);
         break;
       default:
         hipLaunchKernelGGL(hipk::k_hist<1>, dim3(blocks), dim3(kHistBlock), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA,
-                           leafB, d_grad_.ptr, d_hess_.ptr, d_feat_meta_.ptr, fb, fe,
-                           bin_base, bins, d_hist_.ptr, slot_stride);
+                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
+                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                           slot_stride);
     }
-  }
-  // multi-GPU: sum the smaller child's histogram across ranks over xGMI
-  auto& comm = GpuComm::Get();
-  if (comm.active()) {
-    NCCL_OK(ncclAllReduce(spare, spare, static_cast<size_t>(n_elem), ncclFloat32, ncclSum,
-                          comm.comm, stream_));
   }
 }
 
-void HIPTreeLearner::LaunchBestSplit(int leafA, int leafB) {
+void HIPTreeLearner::ReduceSpareHist(int spare_slot) {
+  auto& comm = GpuComm::Get();
+  if (!comm.active()) return;
+  float* spare = d_hist_.ptr + static_cast<size_t>(spare_slot) * total_bins_ * 2;
+  NCCL_OK(ncclAllReduce(spare, spare, static_cast<size_t>(total_bins_) * 2, ncclFloat32,
+                        ncclSum, comm.comm, stream_));
+}
+
+void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counters) {
   hipk::GainParams p;
   p.l1 = config_->lambda_l1;
   p.l2 = config_->lambda_l2;
@@ -1094,14 +1173,15 @@ void HIPTreeLearner::LaunchBestSplit(int leafA, int leafB) {
   p.min_gain_to_split = config_->min_gain_to_split;
   p.min_data = config_->min_data_in_leaf;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
-  const int ny = leafB >= 0 ? 2 : 1;
+  const int ny = leafB_from_counters ? 2 : 1;
   hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_, d_hist_.ptr,
                      slot_stride, d_leaf_slot_.ptr, d_feat_meta_.ptr, nf_,
-                     d_leaf_stats_.ptr, leafA, leafB, p,
+                     d_leaf_stats_.ptr, leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
                      feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
                      d_feat_best_.ptr);
   hipLaunchKernelGGL(hipk::k_best_leaf, dim3(ny), dim3(256), 0, stream_, d_feat_best_.ptr,
-                     nf_, d_leaf_best_.ptr, leafA, leafB);
+                     nf_, d_leaf_best_.ptr, leafA_ptr, d_counters_.ptr,
+                     leafB_from_counters);
 }
 
 Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, bool) {
@@ -1129,12 +1209,11 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
     hipLaunchKernelGGL(hipk::k_iota, dim3((num_data_ + 255) / 256), dim3(256), 0, stream_,
                        d_idx_.ptr, num_data_);
   }
-  approx_cnt_[0] = static_cast<int>(used_cnt_);
-
   // root setup + stats (global count via RCCL when distributed)
   hipLaunchKernelGGL(hipk::k_init_root, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
                      d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr,
-                     static_cast<int>(used_cnt_), d_gbuf_.ptr);
+                     static_cast<int>(used_cnt_), d_gbuf_.ptr, d_counters_.ptr,
+                     d_root_leaf_.ptr, d_minus1_.ptr);
   {
     const int blocks = std::min(2048, (static_cast<int>(used_cnt_) + 255) / 256);
     hipLaunchKernelGGL(hipk::k_root_sums, dim3(blocks), dim3(256), 0, stream_, d_idx_.ptr,
@@ -1151,100 +1230,86 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        d_leaf_stats_.ptr, d_gbuf_.ptr);
   }
 
-  LaunchHist(0, -1, /*spare_slot=*/0, approx_cnt_[0]);
-  LaunchBestSplit(0, -1);
+  // root histogram + best split (leafB disabled via literal -1 pointer semantics)
+  LaunchHist(d_root_leaf_.ptr, 0, HistBlocksFor(static_cast<int>(used_cnt_)));
+  ReduceSpareHist(0);
+  LaunchBestSplit(d_root_leaf_.ptr, 0);
 
-  int num_leaves = 1;
+  // ---- device-driven split loop: ZERO host syncs; winner decisions accumulate in
+  // d_split_log_ and the tree is replayed on the host after one end-of-tree download.
+  const int kPartBlocks = 1024;
+  const int kLoopHistBlocks = 1024;
   for (int split_i = 0; split_i < nl - 1; ++split_i) {
     hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
-                       d_leaf_best_.ptr, num_leaves, d_winner_.ptr,
-                       reinterpret_cast<int*>(d_winner_.ptr + 1));
-    HIP_OK(hipMemcpyAsync(h_winner_, d_winner_.ptr, sizeof(hipk::SplitRec) + sizeof(int),
-                          hipMemcpyDeviceToHost, stream_));
-    HIP_OK(hipStreamSynchronize(stream_));
-    const int L = *h_winner_leaf_;
-    if (L < 0) break;
-    const hipk::SplitRec winner = *h_winner_;
-    const int R = num_leaves;
-    const int f = winner.feature;
-    const BinMapper* mapper = train_data_->FeatureBinMapper(f);
-    const int orig_f = train_data_->RealFeatureIndex(f);
-    const hipk::FeatMeta& m = feat_meta_host_[f];
-
-    // tree structure (approx counts; fixed from exact device counts after the tree)
-    const double right_g = 0.0;  // right sums derivable from stats; not needed here
-    (void)right_g;
-    if (mapper->bin_type() == BinType::kCategorical) {
-      const int cat = static_cast<int>(mapper->BinToValue(winner.bin));
-      std::vector<uint32_t> bits(std::max(cat, 0) / 32 + 1, 0);
-      if (cat >= 0) bits[cat >> 5] |= 1u << (cat & 31);
-      tree->SplitCategorical(L, f, orig_f, bits.data(), static_cast<int>(bits.size()),
-                             winner.left_out, winner.right_out, winner.left_cnt,
-                             winner.right_cnt, winner.left_h,
-                             std::max(0.0, 0.0), static_cast<float>(winner.gain),
-                             mapper->missing_type());
-    } else {
-      tree->Split(L, f, orig_f, winner.bin, mapper->BinToValue(winner.bin), winner.left_out,
-                  winner.right_out, winner.left_cnt, winner.right_cnt, winner.left_h, 0.0,
-                  static_cast<float>(winner.gain), mapper->missing_type(),
-                  winner.default_left != 0);
-    }
-
-    // partition + finalize (all device side; no host sync)
-    const int part_blocks =
-        std::min(4096, std::max(1, (approx_cnt_[L] * 5 / 4 + 4096) / kHistBlock));
-    const uint8_t* col = d_cols_.ptr + static_cast<size_t>(f) * num_data_;
-    const uint32_t* cat_bits_arg = nullptr;
-    if (mapper->bin_type() == BinType::kCategorical) {
-      uint32_t one_bin_bits[8] = {0};
-      one_bin_bits[winner.bin >> 5] = 1u << (winner.bin & 31);
-      HIP_OK(hipMemcpyAsync(d_cat_bits_.ptr, one_bin_bits, sizeof(one_bin_bits),
-                            hipMemcpyHostToDevice, stream_));
-      cat_bits_arg = d_cat_bits_.ptr;
-    }
-    hipLaunchKernelGGL(hipk::k_part_mark, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
-                       col, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L, winner.bin,
-                       m.is_cat ? -1 : m.nan_bin, winner.default_left, 0, cat_bits_arg, 8,
+                       d_leaf_best_.ptr, d_counters_.ptr, d_winner_.ptr,
+                       d_winner_leaf_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_mark, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
+                       d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_winner_leaf_.ptr,
+                       d_winner_.ptr, d_feat_meta_.ptr, d_cols_.ptr, num_data_,
                        d_marks_.ptr, d_block_cnt_.ptr);
     hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(256), 0, stream_, d_block_cnt_.ptr,
-                       part_blocks, d_leaf_cnt_.ptr, L, d_block_loff_.ptr, d_block_roff_.ptr,
-                       d_ctr_.ptr);
-    hipLaunchKernelGGL(hipk::k_part_scatter, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
-                       d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L,
-                       d_marks_.ptr, d_block_loff_.ptr, d_block_roff_.ptr, d_ctr_.ptr);
-    hipLaunchKernelGGL(hipk::k_copy_back, dim3(part_blocks), dim3(kHistBlock), 0, stream_,
-                       d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, L);
+                       kPartBlocks, d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_block_loff_.ptr,
+                       d_block_roff_.ptr, d_ctr_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_scatter, dim3(kPartBlocks), dim3(kHistBlock), 0,
+                       stream_, d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr,
+                       d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_marks_.ptr,
+                       d_block_loff_.ptr, d_block_roff_.ptr, d_ctr_.ptr);
+    hipLaunchKernelGGL(hipk::k_copy_back, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
+                       d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
+                       d_winner_leaf_.ptr);
     hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
-                       d_gbuf_.ptr);
+                       d_winner_leaf_.ptr, d_gbuf_.ptr);
     if (comm.active()) {
       NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
                             stream_));
     }
     hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
                        d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
-                       L, R, /*spare_slot=*/R, d_ctr_.ptr, d_gbuf_.ptr);
-
-    approx_cnt_[L] = std::max(1, winner.left_cnt);
-    approx_cnt_[R] = std::max(1, winner.right_cnt);
-    ++num_leaves;
-
-    const int small_approx = std::min(approx_cnt_[L], approx_cnt_[R]);
-    LaunchHist(L, R, /*spare_slot=*/R, small_approx);
+                       d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
+                       d_gbuf_.ptr);
+    LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks);
+    ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
     {
       const int n_elem = total_bins_ * 2;
       hipLaunchKernelGGL(hipk::k_hist_subtract, dim3((n_elem + 1023) / 1024), dim3(256), 0,
                          stream_, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
-                         d_leaf_slot_.ptr, d_leaf_stats_.ptr, L, R, n_elem);
+                         d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_leaf_.ptr,
+                         d_counters_.ptr, n_elem);
     }
-    LaunchBestSplit(L, R);
+    LaunchBestSplit(d_winner_leaf_.ptr, 1);
   }
 
-  // end of tree: one download of the exact leaf layout
+  // ---- one download: split log + exact leaf layout; replay the tree on the host
+  host_log_.resize(nl);
+  HIP_OK(hipMemcpyAsync(host_log_.data(), d_split_log_.ptr, sizeof(hipk::LogEntry) * nl,
+                        hipMemcpyDeviceToHost, stream_));
   HIP_OK(hipMemcpyAsync(leaf_begin_.data(), d_leaf_begin_.ptr, sizeof(int) * nl,
                         hipMemcpyDeviceToHost, stream_));
   HIP_OK(hipMemcpyAsync(leaf_cnt_.data(), d_leaf_cnt_.ptr, sizeof(int) * nl,
                         hipMemcpyDeviceToHost, stream_));
   HIP_OK(hipStreamSynchronize(stream_));
+
+  for (int i = 0; i < nl - 1; ++i) {
+    const hipk::LogEntry& e = host_log_[i];
+    if (e.leaf < 0) break;
+    const hipk::SplitRec& w = e.rec;
+    const int L = e.leaf;
+    const int f = w.feature;
+    const BinMapper* mapper = train_data_->FeatureBinMapper(f);
+    const int orig_f = train_data_->RealFeatureIndex(f);
+    if (mapper->bin_type() == BinType::kCategorical) {
+      const int cat = static_cast<int>(mapper->BinToValue(w.bin));
+      std::vector<uint32_t> bits(std::max(cat, 0) / 32 + 1, 0);
+      if (cat >= 0) bits[cat >> 5] |= 1u << (cat & 31);
+      tree->SplitCategorical(L, f, orig_f, bits.data(), static_cast<int>(bits.size()),
+                             w.left_out, w.right_out, w.left_cnt, w.right_cnt, w.left_h,
+                             0.0, static_cast<float>(w.gain), mapper->missing_type());
+    } else {
+      tree->Split(L, f, orig_f, w.bin, mapper->BinToValue(w.bin), w.left_out, w.right_out,
+                  w.left_cnt, w.right_cnt, w.left_h, 0.0, static_cast<float>(w.gain),
+                  mapper->missing_type(), w.default_left != 0);
+    }
+  }
   // fix the tree's leaf counts with exact values (internal counts recomputed inside)
   {
     std::vector<int> counts(tree->num_leaves());
