@@ -1121,7 +1121,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
   const int n_elem = total_bins_ * 2;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
   // zero the spare slot (R = counters[0]-1 after finalize; literal slot 0 for the root)
-  hipLaunchKernelGGL(hipk::k_hist_zero, dim3(256), dim3(256), 0, stream_, d_hist_.ptr,
+  hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
                      slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
                      n_elem);
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
@@ -1237,8 +1237,8 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
 
   // ---- device-driven split loop: ZERO host syncs; winner decisions accumulate in
   // d_split_log_ and the tree is replayed on the host after one end-of-tree download.
-  const int kPartBlocks = 1024;
-  const int kLoopHistBlocks = 1024;
+  const int kPartBlocks = 256;
+  const int kLoopHistBlocks = 768;
   for (int split_i = 0; split_i < nl - 1; ++split_i) {
     hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
                        d_leaf_best_.ptr, d_counters_.ptr, d_winner_.ptr,
