@@ -1012,8 +1012,22 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     }
     return true;
   };
-  while (n_copies_ >= 1 && !build_partitions(n_copies_)) n_copies_ /= 2;
-  if (n_copies_ < 1) Log::Fatal("Feature bin footprint exceeds LDS budget");
+  // prefer a SINGLE feature partition (one pass over the row bytes) over more
+  // privatized copies: fewer passes saves idx/grad/hess re-reads at every leaf
+  {
+    int best_copies = -1;
+    size_t best_parts = SIZE_MAX;
+    for (int c : {4, 2, 1}) {
+      if (!build_partitions(c)) continue;
+      if (feat_partitions_.size() < best_parts) {
+        best_parts = feat_partitions_.size();
+        best_copies = c;
+      }
+    }
+    if (best_copies < 0) Log::Fatal("Feature bin footprint exceeds LDS budget");
+    n_copies_ = best_copies;
+    build_partitions(n_copies_);
+  }
 
   const auto& view = train_data->GetRowMajorView();
   if (view.is16) Log::Fatal("HIP learner currently supports max_bin<=255 (uint8 bins)");
