@@ -65,6 +65,10 @@ class GBDT {
 
   /*! raw-feature single-row prediction */
   void PredictRaw(const double* features, double* output, int start_iter, int num_iter) const;
+  /*! margin-based early stop across trees (parity: prediction_early_stop.cpp) */
+  void PredictRawEarlyStop(const double* features, double* output, int start_iter,
+                           int num_iter, int round_period, double margin_threshold,
+                           bool multiclass) const;
   void Predict(const double* features, double* output, int start_iter, int num_iter) const;
   void PredictLeafIndex(const double* features, double* output, int start_iter,
                         int num_iter) const;

@@ -12,6 +12,7 @@
 #include "tree.h"
 
 #include <memory>
+#include <set>
 #include <vector>
 
 namespace migbm {
@@ -156,6 +157,10 @@ class SerialTreeLearner : public TreeLearner {
   Random feature_rng_{0};
   Random extra_rng_{0};
   int iter_counter_ = 0;
+  // CEGB / interaction-constraint state
+  std::vector<int8_t> cegb_feature_used_;
+  std::vector<std::set<int>> leaf_branch_features_;
+  std::vector<std::set<int>> interaction_groups_;
 };
 
 /*! set by the HIP learner's static registrar when the device module is linked in */

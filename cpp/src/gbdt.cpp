@@ -445,6 +445,38 @@ void GBDT::PredictRaw(const double* features, double* output, int start_iter,
   }
 }
 
+void GBDT::PredictRawEarlyStop(const double* features, double* output, int start_iter,
+                               int num_iter, int round_period, double margin_threshold,
+                               bool multiclass) const {
+  const int total_iters = static_cast<int>(models_.size()) / num_tree_per_iteration_;
+  int end_iter = num_iter <= 0 ? total_iters : std::min(total_iters, start_iter + num_iter);
+  for (int c = 0; c < num_tree_per_iteration_; ++c) output[c] = 0.0;
+  int since_check = 0;
+  for (int it = start_iter; it < end_iter; ++it) {
+    for (int c = 0; c < num_tree_per_iteration_; ++c)
+      output[c] += models_[static_cast<size_t>(it) * num_tree_per_iteration_ + c]
+                       ->Predict(features);
+    if (++since_check >= round_period) {
+      since_check = 0;
+      double margin;
+      if (multiclass) {
+        double best = output[0], second = -1e308;
+        for (int c = 1; c < num_tree_per_iteration_; ++c) {
+          if (output[c] > best) { second = best; best = output[c]; }
+          else if (output[c] > second) second = output[c];
+        }
+        margin = best - second;
+      } else {
+        margin = 2.0 * std::fabs(output[0]);
+      }
+      if (margin >= margin_threshold) break;
+    }
+  }
+  if (average_output_ && end_iter > start_iter) {
+    for (int c = 0; c < num_tree_per_iteration_; ++c) output[c] /= (end_iter - start_iter);
+  }
+}
+
 void GBDT::Predict(const double* features, double* output, int start_iter, int num_iter) const {
   PredictRaw(features, output, start_iter, num_iter);
   if (objective_ != nullptr) objective_->ConvertOutput(output, output);

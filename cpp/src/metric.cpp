@@ -435,10 +435,33 @@ Metric* Metric::Create(const std::string& name, const Config& cfg) {
       return -y * std::pow(p, 1 - rho) / (1 - rho) + std::pow(p, 2 - rho) / (2 - rho);
     });
   }
-  if (name == "r2")
-    // r2 needs label variance; approximate via 1 - mse/var at eval time is stateful;
-    // provide as pointwise mse then converted in python layer. Keep simple mse-based proxy.
-    return new PointwiseMetric("l2", true, [](double y, double p) { return (y - p) * (y - p); });
+  if (name == "r2") {
+    class R2Metric : public PointwiseMetric {
+     public:
+      R2Metric() : PointwiseMetric("r2", true,
+                                   [](double y, double p) { return (y - p) * (y - p); }) {}
+      double factor_to_bigger_better() const override { return -1.0; }
+      std::vector<double> Eval(const double* score,
+                               const ObjectiveFunction* obj) const override {
+        auto mse = PointwiseMetric::Eval(score, obj);
+        double mean = 0, sw = 0;
+        for (data_size_t i = 0; i < num_data_; ++i) {
+          double w = weights_ ? weights_[i] : 1.0;
+          mean += w * label_[i];
+          sw += w;
+        }
+        mean /= std::max(sw, 1.0);
+        double var = 0;
+        for (data_size_t i = 0; i < num_data_; ++i) {
+          double w = weights_ ? weights_[i] : 1.0;
+          var += w * (label_[i] - mean) * (label_[i] - mean);
+        }
+        var /= std::max(sw, 1.0);
+        return {var > 0 ? 1.0 - mse[0] / var : 0.0};
+      }
+    };
+    return new R2Metric();
+  }
   if (name == "binary_logloss" || name == "logloss") return new BinaryLoglossMetric();
   if (name == "binary_error")
     return new PointwiseMetric("binary_error", true, [](double y, double p) {

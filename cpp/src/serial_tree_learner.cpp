@@ -6,6 +6,7 @@
 
 #include <algorithm>
 #include <numeric>
+#include <set>
 
 namespace migbm {
 
@@ -67,6 +68,25 @@ void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian
   leaf_ctx_.resize(config_->num_leaves);
   feature_rng_ = Random(config_->feature_fraction_seed);
   extra_rng_ = Random(config_->extra_seed);
+  cegb_feature_used_.assign(train_data_->num_total_features(), 0);
+  leaf_branch_features_.assign(config_->num_leaves, {});
+  interaction_groups_.clear();
+  if (!config_->interaction_constraints.empty()) {
+    // format: "[0,1,2],[2,3]"
+    std::string sdef = config_->interaction_constraints;
+    size_t pos = 0;
+    while ((pos = sdef.find('[', pos)) != std::string::npos) {
+      size_t end = sdef.find(']', pos);
+      if (end == std::string::npos) break;
+      std::set<int> grp;
+      for (auto& tok : Common::Split(sdef.substr(pos + 1, end - pos - 1).c_str(), ',')) {
+        auto t = Common::Trim(tok);
+        if (!t.empty()) grp.insert(atoi(t.c_str()));
+      }
+      if (!grp.empty()) interaction_groups_.push_back(std::move(grp));
+      pos = end + 1;
+    }
+  }
 }
 
 void SerialTreeLearner::ResetTrainingData(const Dataset* train_data) {
@@ -128,9 +148,29 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
   const int nf = train_data_->num_features();
   std::vector<int8_t> node_mask;
   const std::vector<int8_t>* mask = &is_feature_used_;
-  if (config_->feature_fraction_bynode < 1.0) {
-    node_mask = SampleFeatures(true);
+  if (config_->feature_fraction_bynode < 1.0 || !interaction_groups_.empty()) {
+    node_mask = config_->feature_fraction_bynode < 1.0 ? SampleFeatures(true)
+                                                       : std::vector<int8_t>(nf, 1);
     for (int f = 0; f < nf; ++f) node_mask[f] &= is_feature_used_[f];
+    if (!interaction_groups_.empty()) {
+      // interaction constraints: feature f allowed iff some group contains the
+      // leaf's branch features plus f (parity: reference ColSampler constraints)
+      const auto& used = leaf_branch_features_[leaf];
+      for (int f = 0; f < nf; ++f) {
+        if (!node_mask[f]) continue;
+        const int orig = train_data_->RealFeatureIndex(f);
+        bool ok = false;
+        for (const auto& grp : interaction_groups_) {
+          if (!grp.count(orig)) continue;
+          bool covers = true;
+          for (int u : used) {
+            if (!grp.count(u)) { covers = false; break; }
+          }
+          if (covers) { ok = true; break; }
+        }
+        node_mask[f] = ok ? 1 : 0;
+      }
+    }
     mask = &node_mask;
   }
   std::vector<SplitInfo> cand(nf);
@@ -160,7 +200,28 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
     cand[f].feature = f;
   }
   for (int f = 0; f < nf; ++f) {
-    if (cand[f].IsValid() && cand[f] > best) best = cand[f];
+    if (!cand[f].IsValid()) continue;
+    const int orig = train_data_->RealFeatureIndex(f);
+    // feature_contri: multiplicative per-feature gain reweighting
+    if (!config_->feature_contri.empty() &&
+        orig < static_cast<int>(config_->feature_contri.size())) {
+      cand[f].gain *= config_->feature_contri[orig];
+    }
+    // CEGB: cost-efficient gradient boosting gain penalty
+    if (config_->cegb_tradeoff > 0.0 &&
+        (config_->cegb_penalty_split > 0.0 ||
+         !config_->cegb_penalty_feature_coupled.empty() ||
+         !config_->cegb_penalty_feature_lazy.empty())) {
+      double penalty = config_->cegb_penalty_split;
+      if (orig < static_cast<int>(config_->cegb_penalty_feature_coupled.size()) &&
+          !cegb_feature_used_[orig])
+        penalty += config_->cegb_penalty_feature_coupled[orig];
+      if (orig < static_cast<int>(config_->cegb_penalty_feature_lazy.size()))
+        penalty += config_->cegb_penalty_feature_lazy[orig] * ctx.num_data;
+      cand[f].gain -= config_->cegb_tradeoff * penalty;
+      if (cand[f].gain <= 0) continue;
+    }
+    if (cand[f] > best) best = cand[f];
   }
 }
 
@@ -275,6 +336,14 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                   train_data_->RealThreshold(f, s.threshold), s.left_output, s.right_output,
                   s.left_count, s.right_count, s.left_sum_hessian, s.right_sum_hessian,
                   static_cast<float>(s.gain), m->missing_type(), s.default_left);
+    }
+
+    // bookkeeping for CEGB / interaction constraints
+    cegb_feature_used_[orig_f] = 1;
+    if (!interaction_groups_.empty()) {
+      leaf_branch_features_[right_leaf] = leaf_branch_features_[best_leaf];
+      leaf_branch_features_[best_leaf].insert(orig_f);
+      leaf_branch_features_[right_leaf].insert(orig_f);
     }
 
     // child contexts
