@@ -160,6 +160,8 @@ class Dataset {
   void PushRawRow(data_size_t row, const double* values, int ncol);
 
   std::vector<int8_t> categorical_flags_;   // per original column
+  bool has_raw() const { return !raw_values_.empty(); }
+  float raw_value(int inner, data_size_t row) const { return raw_values_[inner][row]; }
 
  private:
   friend class DatasetLoader;
@@ -174,6 +176,7 @@ class Dataset {
   std::vector<uint32_t> hist_offsets_;
   int num_total_bin_ = 0;
   Metadata metadata_;
+  std::vector<std::vector<float>> raw_values_;  // per inner feature (linear_tree only)
   std::vector<std::string> feature_names_;
   mutable RowMajorView row_view_;
   mutable bool row_view_built_ = false;

@@ -43,7 +43,7 @@ class Tree {
     if (num_leaves_ > 1) {
       int node = 0;
       while (node >= 0) node = Decision(feature_values, node);
-      return LeafOutput(~node);
+      return is_linear_ ? LeafOutputLinear(~node, feature_values) : LeafOutput(~node);
     }
     return num_leaves_ == 1 ? leaf_value_[0] : 0.0;
   }
@@ -81,10 +81,19 @@ class Tree {
     shrinkage_ *= rate;
     for (int i = 0; i < num_leaves_; ++i) leaf_value_[i] = MaybeRound(leaf_value_[i] * rate);
     for (int i = 0; i < num_leaves_ - 1; ++i) internal_value_[i] *= rate;
+    if (is_linear_) {
+      for (int i = 0; i < num_leaves_; ++i) {
+        leaf_const_[i] *= rate;
+        for (auto& c : leaf_coeff_[i]) c *= rate;
+      }
+    }
   }
   void AddBias(double val) {
     for (int i = 0; i < num_leaves_; ++i) leaf_value_[i] = MaybeRound(val + leaf_value_[i]);
     for (int i = 0; i < num_leaves_ - 1; ++i) internal_value_[i] += val;
+    if (is_linear_) {
+      for (int i = 0; i < num_leaves_; ++i) leaf_const_[i] += val;
+    }
   }
   void AsConstantTree(double val, int count = 0) {
     num_leaves_ = 1;
@@ -110,6 +119,33 @@ class Tree {
   const std::vector<uint32_t>& cat_threshold() const { return cat_threshold_; }
   int num_cat() const { return num_cat_; }
   bool is_linear() const { return is_linear_; }
+  void SetLinear(bool v) {
+    is_linear_ = v;
+    if (v && leaf_const_.empty()) {
+      leaf_const_.assign(max_leaves_, 0.0);
+      leaf_features_.resize(max_leaves_);
+      leaf_features_inner_.resize(max_leaves_);
+      leaf_coeff_.resize(max_leaves_);
+    }
+  }
+  /*! set a leaf's linear model: output = const + sum(coeff[i] * x[feat[i]]) */
+  void SetLeafLinear(int leaf, double constant, const std::vector<int>& feats_real,
+                     const std::vector<int>& feats_inner, const std::vector<double>& coeff) {
+    leaf_const_[leaf] = constant;
+    leaf_features_[leaf] = feats_real;
+    leaf_features_inner_[leaf] = feats_inner;
+    leaf_coeff_[leaf] = coeff;
+  }
+  double LeafOutputLinear(int leaf, const double* feature_values) const {
+    if (!is_linear_ || leaf_coeff_[leaf].empty()) return leaf_value_[leaf];
+    double out = leaf_const_[leaf];
+    for (size_t i = 0; i < leaf_coeff_[leaf].size(); ++i) {
+      const double v = feature_values[leaf_features_[leaf][i]];
+      if (std::isnan(v)) return leaf_value_[leaf];  // missing -> piecewise-constant value
+      out += leaf_coeff_[leaf][i] * v;
+    }
+    return out;
+  }
 
   /*! expected maximum value |leaf output| for bound calc */
   double GetUpperBoundValue() const;
@@ -166,6 +202,10 @@ class Tree {
   // categorical bitsets
   std::vector<int> cat_boundaries_;
   std::vector<uint32_t> cat_threshold_;
+  // linear leaves (linear_tree=true)
+  std::vector<double> leaf_const_;
+  std::vector<std::vector<int>> leaf_features_, leaf_features_inner_;
+  std::vector<std::vector<double>> leaf_coeff_;
 };
 
 }  // namespace migbm

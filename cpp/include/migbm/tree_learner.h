@@ -136,6 +136,8 @@ class SerialTreeLearner : public TreeLearner {
   hist_t* HistSlot(int slot) { return hist_store_.data() + static_cast<size_t>(slot) * 2 * train_data_->num_total_bin(); }
   /*! per-node feature sampling mask (feature_fraction / bynode / interaction constraints) */
   std::vector<int8_t> SampleFeatures(bool per_node);
+  /*! linear-tree post-pass: ridge-weighted LS fit per leaf */
+  void CalculateLinear(Tree* tree);
   /*! predicate for partition: does row go left under split s of inner feature f? */
   std::function<bool(data_size_t)> MakeGoLeft(const SplitInfo& s) const;
 

@@ -125,6 +125,16 @@ void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& ge
     const BinMapper* m = bin_mappers_[f].get();
     for (data_size_t i = 0; i < nrow; ++i) columns_[f].Set(i, m->ValueToBin(get(i, c)));
   }
+  if (cfg.linear_tree) {
+    raw_values_.resize(nf);
+#pragma omp parallel for schedule(dynamic, 1)
+    for (int f = 0; f < nf; ++f) {
+      raw_values_[f].resize(nrow);
+      const int c = real_feature_index_[f];
+      for (data_size_t i = 0; i < nrow; ++i)
+        raw_values_[f][i] = static_cast<float>(get(i, c));
+    }
+  }
   metadata_.Init(nrow, false, false);
   if (feature_names_.empty()) {
     for (int c = 0; c < ncol; ++c) feature_names_.push_back("Column_" + std::to_string(c));
@@ -465,7 +475,23 @@ void Tree::AddPredictionToScore(const Dataset* data, data_size_t num_data, doubl
         }
       }
     }
-    score[i] += leaf_value_[~node];
+    if (is_linear_) {
+      const int leaf = ~node;
+      double out = leaf_const_.empty() || leaf_coeff_[leaf].empty()
+                       ? leaf_value_[leaf] : leaf_const_[leaf];
+      if (!leaf_const_.empty() && !leaf_coeff_[leaf].empty()) {
+        bool ok = true;
+        for (size_t k = 0; k < leaf_coeff_[leaf].size(); ++k) {
+          const float v = data->raw_value(leaf_features_inner_[leaf][k], i);
+          if (std::isnan(v)) { ok = false; break; }
+          out += leaf_coeff_[leaf][k] * v;
+        }
+        if (!ok) out = leaf_value_[leaf];
+      }
+      score[i] += out;
+    } else {
+      score[i] += leaf_value_[~node];
+    }
   }
 }
 
@@ -502,7 +528,23 @@ void Tree::AddPredictionToScore(const Dataset* data, const data_size_t* used_ind
         }
       }
     }
-    score[r] += leaf_value_[~node];
+    if (is_linear_) {
+      const int leaf = ~node;
+      double out = leaf_const_.empty() || leaf_coeff_[leaf].empty()
+                       ? leaf_value_[leaf] : leaf_const_[leaf];
+      if (!leaf_const_.empty() && !leaf_coeff_[leaf].empty()) {
+        bool ok = true;
+        for (size_t k = 0; k < leaf_coeff_[leaf].size(); ++k) {
+          const float v = data->raw_value(leaf_features_inner_[leaf][k], r);
+          if (std::isnan(v)) { ok = false; break; }
+          out += leaf_coeff_[leaf][k] * v;
+        }
+        if (!ok) out = leaf_value_[leaf];
+      }
+      score[r] += out;
+    } else {
+      score[r] += leaf_value_[~node];
+    }
   }
 }
 

@@ -376,7 +376,105 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
     FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);
   }
+  if (config_->linear_tree) CalculateLinear(tree.get());
   return tree.release();
+}
+
+/*! Fit a ridge-regularized weighted linear model in every leaf over the numerical
+ *  features on the leaf's branch path (parity: reference LinearTreeLearner, Eigen-free:
+ *  small Cholesky solve; targets -g/h weighted by h). */
+void SerialTreeLearner::CalculateLinear(Tree* tree) {
+  if (!train_data_->has_raw()) {
+    Log::Warning("linear_tree requires raw values; dataset was built without them");
+    return;
+  }
+  tree->SetLinear(true);
+  const int nl = tree->num_leaves();
+  if (nl <= 1) return;
+  // collect per-leaf branch features (numerical, unique, inner indices)
+  std::vector<std::vector<int>> leaf_feats(nl);
+  std::function<void(int, std::vector<int>&)> walk = [&](int node, std::vector<int>& path) {
+    if (node < 0) {
+      leaf_feats[~node] = path;
+      return;
+    }
+    const int fi = tree->split_feature_inner(node);
+    const bool is_num = !tree->IsCategoricalSplit(node);
+    bool added = false;
+    if (is_num && std::find(path.begin(), path.end(), fi) == path.end()) {
+      path.push_back(fi);
+      added = true;
+    }
+    walk(tree->left_child(node), path);
+    walk(tree->right_child(node), path);
+    if (added) path.pop_back();
+  };
+  std::vector<int> path;
+  walk(0, path);
+
+#pragma omp parallel for schedule(dynamic)
+  for (int l = 0; l < nl; ++l) {
+    const auto& feats = leaf_feats[l];
+    const int k = static_cast<int>(feats.size());
+    data_size_t cnt;
+    const data_size_t* idx = partition_.GetIndexOnLeaf(l, &cnt);
+    if (k == 0 || cnt < static_cast<data_size_t>(k + 2)) continue;
+    const int dim = k + 1;  // coefficients + intercept
+    std::vector<double> A(dim * dim, 0.0), b(dim, 0.0), z(dim);
+    bool has_nan = false;
+    for (data_size_t i = 0; i < cnt; ++i) {
+      const data_size_t r = idx[i];
+      for (int j = 0; j < k; ++j) {
+        z[j] = train_data_->raw_value(feats[j], r);
+        if (std::isnan(z[j])) { has_nan = true; break; }
+      }
+      if (has_nan) break;
+      z[k] = 1.0;
+      const double h = hessians_[r];
+      const double g = gradients_[r];
+      for (int a = 0; a < dim; ++a) {
+        for (int c2 = a; c2 < dim; ++c2) A[a * dim + c2] += h * z[a] * z[c2];
+        b[a] += -g * z[a];
+      }
+    }
+    if (has_nan) continue;  // leaves with missing values stay piecewise-constant
+    for (int a = 0; a < k; ++a) A[a * dim + a] += config_->linear_lambda + config_->lambda_l2;
+    A[k * dim + k] += config_->lambda_l2;
+    // Cholesky solve (upper triangle filled)
+    std::vector<double> Lm(dim * dim, 0.0);
+    bool ok = true;
+    for (int a = 0; a < dim && ok; ++a) {
+      for (int c2 = 0; c2 <= a; ++c2) {
+        double sum = A[std::min(a, c2) * dim + std::max(a, c2)];
+        for (int t = 0; t < c2; ++t) sum -= Lm[a * dim + t] * Lm[c2 * dim + t];
+        if (a == c2) {
+          if (sum <= 1e-12) { ok = false; break; }
+          Lm[a * dim + a] = std::sqrt(sum);
+        } else {
+          Lm[a * dim + c2] = sum / Lm[c2 * dim + c2];
+        }
+      }
+    }
+    if (!ok) continue;
+    std::vector<double> y(dim), beta(dim);
+    for (int a = 0; a < dim; ++a) {
+      double sum = b[a];
+      for (int t = 0; t < a; ++t) sum -= Lm[a * dim + t] * y[t];
+      y[a] = sum / Lm[a * dim + a];
+    }
+    for (int a = dim - 1; a >= 0; --a) {
+      double sum = y[a];
+      for (int t = a + 1; t < dim; ++t) sum -= Lm[t * dim + a] * beta[t];
+      beta[a] = sum / Lm[a * dim + a];
+    }
+    bool finite = true;
+    for (double v : beta) finite &= std::isfinite(v);
+    if (!finite) continue;
+    std::vector<int> feats_real(k);
+    for (int j = 0; j < k; ++j) feats_real[j] = train_data_->RealFeatureIndex(feats[j]);
+    std::vector<double> coeffs(beta.begin(), beta.begin() + k);
+    tree->SetLeafLinear(l, beta[k], feats_real, feats, coeffs);
+  }
 }
 
 void SerialTreeLearner::AddPredictionToScore(const Tree* tree, double* out_score) {

@@ -146,6 +146,25 @@ std::string Tree::ToString() const {
     ss << "cat_threshold=" << Common::ArrayToString(cat_threshold_.data(), cat_threshold_.size()) << '\n';
   }
   ss << "is_linear=" << (is_linear_ ? 1 : 0) << '\n';
+  if (is_linear_) {
+    std::vector<double> consts(num_leaves_);
+    std::vector<int> nfeat(num_leaves_);
+    std::vector<int> flat_feats;
+    std::vector<double> flat_coefs;
+    for (int l = 0; l < num_leaves_; ++l) {
+      consts[l] = l < static_cast<int>(leaf_const_.size()) ? leaf_const_[l] : 0.0;
+      const auto& ff = l < static_cast<int>(leaf_features_.size()) ? leaf_features_[l]
+                                                                   : std::vector<int>();
+      nfeat[l] = static_cast<int>(ff.size());
+      for (int f : ff) flat_feats.push_back(f);
+      if (l < static_cast<int>(leaf_coeff_.size()))
+        for (double c : leaf_coeff_[l]) flat_coefs.push_back(c);
+    }
+    ss << "leaf_const=" << Common::ArrayToString(consts.data(), consts.size()) << '\n';
+    ss << "num_features=" << Common::ArrayToString(nfeat.data(), nfeat.size()) << '\n';
+    ss << "leaf_features=" << Common::ArrayToString(flat_feats.data(), flat_feats.size()) << '\n';
+    ss << "leaf_coeff=" << Common::ArrayToString(flat_coefs.data(), flat_coefs.size()) << '\n';
+  }
   ss << "shrinkage=" << Common::DoubleToStr(shrinkage_) << '\n';
   return ss.str();
 }
@@ -222,6 +241,27 @@ Tree::Tree(const char* str, size_t* used_len) {
     Common::StringToArray<uint32_t>(get("cat_threshold"), ' ', &cat_threshold_);
   } else {
     cat_boundaries_.push_back(0);
+  }
+  if (is_linear_) {
+    SetLinear(true);
+    std::vector<double> consts;
+    std::vector<int> nfeat, flat_feats;
+    std::vector<double> flat_coefs;
+    Common::StringToArray<double>(get("leaf_const"), ' ', &consts);
+    Common::StringToArray<int>(get("num_features"), ' ', &nfeat);
+    Common::StringToArray<int>(get("leaf_features"), ' ', &flat_feats);
+    Common::StringToArray<double>(get("leaf_coeff"), ' ', &flat_coefs);
+    size_t off = 0;
+    for (int l = 0; l < num_leaves_ && l < static_cast<int>(nfeat.size()); ++l) {
+      leaf_const_[l] = l < static_cast<int>(consts.size()) ? consts[l] : 0.0;
+      const int k = nfeat[l];
+      std::vector<int> ff(flat_feats.begin() + off, flat_feats.begin() + off + k);
+      std::vector<double> cc(flat_coefs.begin() + off, flat_coefs.begin() + off + k);
+      leaf_features_[l] = ff;
+      leaf_features_inner_[l] = ff;
+      leaf_coeff_[l] = cc;
+      off += k;
+    }
   }
   threshold_in_bin_.assign(num_leaves_ - 1, 0);
   leaf_depth_.assign(num_leaves_, 0);

@@ -275,3 +275,69 @@ def test_weights_affect_training():
     pred = bst.predict(X)
     # heavy positive weights push average prediction up
     assert pred.mean() > y.mean()
+
+
+def test_linear_tree():
+    rng = np.random.RandomState(11)
+    X = rng.rand(4000, 4) * 4
+    y = (3.0 * X[:, 0] + 2.0 * X[:, 1] + 0.05 * rng.randn(4000)).astype(np.float32)
+    const_bst = lgb.train({"objective": "regression", "verbosity": -1, "num_leaves": 4},
+                          lgb.Dataset(X, label=y), 20)
+    lin_bst = lgb.train({"objective": "regression", "verbosity": -1, "num_leaves": 4,
+                         "linear_tree": True}, lgb.Dataset(X, label=y), 20)
+    mse_const = float(np.mean((const_bst.predict(X) - y) ** 2))
+    mse_lin = float(np.mean((lin_bst.predict(X) - y) ** 2))
+    assert mse_lin < 0.5 * mse_const
+    s = lin_bst.model_to_string()
+    assert "is_linear=1" in s and "leaf_coeff=" in s
+    b2 = lgb.Booster(model_str=s)
+    np.testing.assert_allclose(b2.predict(X[:50]), lin_bst.predict(X[:50]), rtol=1e-9)
+
+
+def test_cegb_penalty_reduces_features():
+    rng = np.random.RandomState(12)
+    X = rng.randn(3000, 10)
+    y = (X[:, 0] + 0.3 * X[:, 1] + 0.2 * rng.randn(3000)).astype(np.float32)
+    free = lgb.train({"objective": "regression", "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    pen = lgb.train({"objective": "regression", "verbosity": -1,
+                     "cegb_penalty_feature_coupled": [0.0] + [1e5] * 9},
+                    lgb.Dataset(X, label=y), 20)
+    nf_free = int((free.feature_importance() > 0).sum())
+    nf_pen = int((pen.feature_importance() > 0).sum())
+    assert nf_pen <= nf_free
+    assert pen.feature_importance()[0] > 0
+
+
+def test_interaction_constraints():
+    rng = np.random.RandomState(13)
+    X = rng.randn(3000, 4)
+    y = (X[:, 0] * X[:, 1] + X[:, 2] + 0.1 * rng.randn(3000)).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "verbosity": -1,
+                     "interaction_constraints": "[0,1],[2,3]"},
+                    lgb.Dataset(X, label=y), 20)
+    # every branch must stay within one group
+    model = bst.dump_model()
+
+    def check(node, path):
+        if "leaf_index" in node:
+            if path:
+                assert set(path) <= {0, 1} or set(path) <= {2, 3}, path
+            return
+        check(node["left_child"], path + [node["split_feature"]])
+        check(node["right_child"], path + [node["split_feature"]])
+
+    for t in model["tree_info"]:
+        if "split_index" in t["tree_structure"]:
+            check(t["tree_structure"], [])
+
+
+def test_r2_metric():
+    rng = np.random.RandomState(14)
+    X = rng.randn(2000, 5)
+    y = (X[:, 0] + 0.1 * rng.randn(2000)).astype(np.float32)
+    ev = {}
+    train = lgb.Dataset(X, label=y)
+    lgb.train({"objective": "regression", "metric": "r2", "verbosity": -1}, train, 20,
+              valid_sets=[train], valid_names=["train"],
+              callbacks=[lgb.record_evaluation(ev)])
+    assert ev["train"]["r2"][-1] > 0.9
