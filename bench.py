@@ -64,6 +64,9 @@ def main():
     ap.add_argument("--num-leaves", type=int, default=255)
     ap.add_argument("--device", default="gpu", choices=["gpu", "cpu"])
     ap.add_argument("--valid-rows", type=int, default=500_000)
+    ap.add_argument("--quantized", action="store_true",
+                    help="use_quantized_grad: packed int histograms (reference's "
+                         "quantized training mode; halves LDS atomics + wire bytes)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -132,6 +135,8 @@ def main():
         "verbosity": 0,
         "num_threads": 0,
     }
+    if args.quantized:
+        params["use_quantized_grad"] = True
     booster = lgb.Booster(params=params, train_set=train)
 
     def barrier_sync():
@@ -176,7 +181,7 @@ def main():
             "higher_is_better": False,
             "scaling": "weak",
             "vs_baseline": sec_per_iter / BASELINE_SEC_PER_ITER,
-            "dtype": "fp32hist+fp64gain",
+            "dtype": "int16hist+fp64gain" if args.quantized else "fp32hist+fp64gain",
             "data": "synthetic",
             "auc": auc,
             "config": {

@@ -120,6 +120,130 @@ __device__ __forceinline__ double d_leaf_gain(double g, double h, const GainPara
   return d_gain_out(g, h, out, p);
 }
 
+/*! absmax of grad/hess over the used rows -> scales (double[2]). */
+__global__ void k_grad_absmax(const uint32_t* __restrict__ idx, int cnt,
+                              const float* __restrict__ g, const float* __restrict__ h,
+                              float* __restrict__ out_max /* [2], pre-zeroed */) {
+  float mg = 0, mh = 0;
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (int i = tid; i < cnt; i += blockDim.x * gridDim.x) {
+    const uint32_t r = idx[i];
+    mg = fmaxf(mg, fabsf(g[r]));
+    mh = fmaxf(mh, fabsf(h[r]));
+  }
+  for (int d = 32; d > 0; d >>= 1) {
+    mg = fmaxf(mg, __shfl_down(mg, d));
+    mh = fmaxf(mh, __shfl_down(mh, d));
+  }
+  if ((threadIdx.x & 63) == 0) {
+    atomicMax(reinterpret_cast<int*>(&out_max[0]), __float_as_int(mg));
+    atomicMax(reinterpret_cast<int*>(&out_max[1]), __float_as_int(mh));
+  }
+}
+
+/*! stochastic-rounded quantization of (g,h) into a packed int32 (g<<16 | h).
+ *  Range: g in [-levels, levels], h in [0, 2*levels]; scales derived from absmax. */
+__global__ void k_grad_quantize(const float* __restrict__ g, const float* __restrict__ h,
+                                int n, const float* __restrict__ absmax, int levels,
+                                int use_stochastic, uint32_t seed,
+                                int32_t* __restrict__ packed,
+                                float* __restrict__ scales /* [2] out */) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i == 0) {
+    scales[0] = absmax[0] > 0 ? absmax[0] / levels : 1.0f;
+    scales[1] = absmax[1] > 0 ? absmax[1] / (2.0f * levels) : 1.0f;
+  }
+  if (i >= n) return;
+  const float gs = absmax[0] > 0 ? absmax[0] / levels : 1.0f;
+  const float hs = absmax[1] > 0 ? absmax[1] / (2.0f * levels) : 1.0f;
+  float rg = 0.5f, rh = 0.5f;
+  if (use_stochastic) {
+    uint32_t x = (static_cast<uint32_t>(i) * 2654435761u) ^ seed;
+    x ^= x >> 16;
+    x *= 2246822519u;
+    x ^= x >> 13;
+    rg = (x & 0xFFFF) * (1.0f / 65536.0f);
+    rh = ((x >> 16) & 0xFFFF) * (1.0f / 65536.0f);
+  }
+  int gq = static_cast<int>(floorf(g[i] / gs + rg));
+  int hq = static_cast<int>(floorf(h[i] / hs + rh));
+  gq = max(-levels, min(levels, gq));
+  hq = max(0, min(2 * levels, hq));
+  packed[i] = (gq << 16) | (hq & 0xFFFF);
+}
+
+/*! packed-int histogram: ONE ds_add per (row, feature); per-block partial sums are
+ *  dequantized into the same fp32 global histogram at flush, so the best-split scan
+ *  is unchanged. hi16 = sum of gq (signed), lo16 = sum of hq (non-negative; bounded
+ *  by rows_per_block * 2*levels < 2^15 via grid sizing). */
+template <int NCOPIES>
+__global__ void k_hist_q(const uint8_t* __restrict__ rows, int stride,
+                         const uint32_t* __restrict__ idx_base,
+                         const int* __restrict__ leaf_begin, const int* __restrict__ leaf_cnt,
+                         const LeafStat* __restrict__ stats, const int* __restrict__ leaf_slot,
+                         const int* __restrict__ leafA_ptr, const int* __restrict__ counters,
+                         int leafB_from_counters, const int32_t* __restrict__ packed,
+                         const float* __restrict__ scales, const FeatMeta* __restrict__ fm,
+                         int feat_begin, int feat_end, int part_bin_base, int part_bins,
+                         float* __restrict__ hist_base, size_t slot_stride) {
+  const int leafA = *leafA_ptr;
+  if (leafA < 0) return;
+  const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
+  int leaf = leafA;
+  if (leafB >= 0 && stats[leafB].cnt < stats[leafA].cnt) leaf = leafB;
+  const int begin = leaf_begin[leaf];
+  const int cnt = leaf_cnt[leaf];
+  const uint32_t* idx = idx_base + begin;
+  float* ghist = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
+                 static_cast<size_t>(part_bin_base) * 2;
+
+  constexpr int kBinStride = NCOPIES + 1;  // ints per bin (non-pow2 bank rotation)
+  extern __shared__ int lhq[];
+  __shared__ int loff[256];
+  const int nfeat = feat_end - feat_begin;
+  for (int i = threadIdx.x; i < nfeat; i += blockDim.x)
+    loff[i] = fm[feat_begin + i].bin_off - part_bin_base;
+  const int nelem = part_bins * kBinStride;
+  for (int i = threadIdx.x; i < nelem; i += blockDim.x) lhq[i] = 0;
+  __syncthreads();
+
+  const int my_copy = threadIdx.x % NCOPIES;
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const int nthreads = blockDim.x * gridDim.x;
+  const int c0 = feat_begin & ~15;
+  for (int i = tid; i < cnt; i += nthreads) {
+    const uint32_t r = idx[i];
+    const int32_t gh = packed[r];
+    const uint8_t* rp = rows + static_cast<size_t>(r) * stride;
+    for (int c = c0; c < feat_end; c += 16) {
+      const uint4 v = *reinterpret_cast<const uint4*>(rp + c);
+      const uint32_t w[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int f = c + j;
+        if (f < feat_begin || f >= feat_end) continue;
+        const int b = (w[j >> 2] >> ((j & 3) * 8)) & 0xFF;
+        atomicAdd(&lhq[(loff[f - feat_begin] + b) * kBinStride + my_copy], gh);
+      }
+    }
+  }
+  __syncthreads();
+  const float gs = scales[0], hs = scales[1];
+  for (int i = threadIdx.x; i < part_bins; i += blockDim.x) {
+    int G = 0, H = 0;
+#pragma unroll
+    for (int cpy = 0; cpy < NCOPIES; ++cpy) {
+      const int v = lhq[i * kBinStride + cpy];
+      // v = G_c * 65536 + H_c with 0 <= H_c < 2^16 (bounds guaranteed by grid sizing)
+      const int gq = v >> 16;
+      G += gq;
+      H += v - (gq << 16);
+    }
+    if (G != 0) atomicAdd(&ghist[i * 2], static_cast<float>(G) * gs);
+    if (H != 0) atomicAdd(&ghist[i * 2 + 1], static_cast<float>(H) * hs);
+  }
+}
+
 __global__ void k_iota(uint32_t* p, int n) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) p[i] = static_cast<uint32_t>(i);
@@ -913,6 +1037,8 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<uint8_t> d_cols_;
   DevBuf<hipk::FeatMeta> d_feat_meta_;
   DevBuf<float> d_grad_, d_hess_;
+  DevBuf<int32_t> d_grad_packed_;
+  DevBuf<float> d_grad_absmax_, d_grad_scales_;
   DevBuf<double> d_score_;
   DevBuf<float> d_label_, d_weight_;
   DevBuf<uint32_t> d_idx_, d_idx_tmp_;
@@ -944,6 +1070,9 @@ class HIPTreeLearner : public TreeLearner {
   std::vector<hipk::LogEntry> host_log_;
 
   bool grads_on_device_ = false;
+  bool quantized_ = false;
+  int quant_levels_ = 2;
+  uint32_t quant_seed_ = 0x9E3779B9u;
   bool weights_present_ = false;
   const data_size_t* bag_indices_ = nullptr;
   data_size_t bag_cnt_ = 0;
@@ -1042,6 +1171,13 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
 
   d_grad_.Alloc(num_data_);
   d_hess_.Alloc(num_data_);
+  quantized_ = config_->use_quantized_grad;
+  quant_levels_ = std::max(1, config_->num_grad_quant_bins / 2);
+  if (quantized_) {
+    d_grad_packed_.Alloc(num_data_);
+    d_grad_absmax_.Alloc(2);
+    d_grad_scales_.Alloc(2);
+  }
   d_score_.Alloc(num_data_);
   HIP_OK(hipMemset(d_score_.ptr, 0, sizeof(double) * num_data_));
   d_label_.Alloc(num_data_);
@@ -1141,6 +1277,38 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
     const auto [fb, fe] = feat_partitions_[pr];
     const auto [bin_base, bins] = part_bin_range_[pr];
+    if (quantized_) {
+      const size_t ldsq = static_cast<size_t>(bins) * (n_copies_ + 1) * sizeof(int);
+      switch (n_copies_) {
+        case 4:
+          hipLaunchKernelGGL(hipk::k_hist_q<4>, dim3(blocks), dim3(kHistBlock), ldsq,
+                             stream_, d_rows_.ptr, row_stride_, d_idx_.ptr,
+                             d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_leaf_stats_.ptr,
+                             d_leaf_slot_.ptr, leafA_ptr, d_counters_.ptr,
+                             leafB_from_counters, d_grad_packed_.ptr, d_grad_scales_.ptr,
+                             d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                             slot_stride);
+          break;
+        case 2:
+          hipLaunchKernelGGL(hipk::k_hist_q<2>, dim3(blocks), dim3(kHistBlock), ldsq,
+                             stream_, d_rows_.ptr, row_stride_, d_idx_.ptr,
+                             d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_leaf_stats_.ptr,
+                             d_leaf_slot_.ptr, leafA_ptr, d_counters_.ptr,
+                             leafB_from_counters, d_grad_packed_.ptr, d_grad_scales_.ptr,
+                             d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                             slot_stride);
+          break;
+        default:
+          hipLaunchKernelGGL(hipk::k_hist_q<1>, dim3(blocks), dim3(kHistBlock), ldsq,
+                             stream_, d_rows_.ptr, row_stride_, d_idx_.ptr,
+                             d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_leaf_stats_.ptr,
+                             d_leaf_slot_.ptr, leafA_ptr, d_counters_.ptr,
+                             leafB_from_counters, d_grad_packed_.ptr, d_grad_scales_.ptr,
+                             d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                             slot_stride);
+      }
+      continue;
+    }
     const size_t lds = static_cast<size_t>(bins) * (2 * n_copies_ + 2) * sizeof(float);
     switch (n_copies_) {
       case 4:
@@ -1222,6 +1390,17 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
     used_cnt_ = num_data_;
     hipLaunchKernelGGL(hipk::k_iota, dim3((num_data_ + 255) / 256), dim3(256), 0, stream_,
                        d_idx_.ptr, num_data_);
+  }
+  if (quantized_) {
+    HIP_OK(hipMemsetAsync(d_grad_absmax_.ptr, 0, 2 * sizeof(float), stream_));
+    hipLaunchKernelGGL(hipk::k_grad_absmax, dim3(512), dim3(256), 0, stream_, d_idx_.ptr,
+                       static_cast<int>(used_cnt_), d_grad_.ptr, d_hess_.ptr,
+                       d_grad_absmax_.ptr);
+    quant_seed_ = quant_seed_ * 1664525u + 1013904223u;
+    hipLaunchKernelGGL(hipk::k_grad_quantize, dim3((num_data_ + 255) / 256), dim3(256), 0,
+                       stream_, d_grad_.ptr, d_hess_.ptr, num_data_, d_grad_absmax_.ptr,
+                       quant_levels_, config_->stochastic_rounding ? 1 : 0, quant_seed_,
+                       d_grad_packed_.ptr, d_grad_scales_.ptr);
   }
   // root setup + stats (global count via RCCL when distributed)
   hipLaunchKernelGGL(hipk::k_init_root, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,

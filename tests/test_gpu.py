@@ -95,3 +95,18 @@ def test_gpu_weights():
               "num_leaves": 31}
     bst = lgb.train(params, lgb.Dataset(X, label=y, weight=w), 10)
     assert bst.predict(X[:5000]).mean() > y.mean()
+
+
+def test_gpu_quantized_grad():
+    """use_quantized_grad: packed-int histograms must stay at quality parity."""
+    X, y = _binary_data(n=150_000)
+    Xv, yv = _binary_data(n=50_000, seed=55)
+    aucs = {}
+    for q in (False, True):
+        params = {"objective": "binary", "device_type": "gpu", "max_bin": 63,
+                  "num_leaves": 63, "min_data_in_leaf": 1,
+                  "min_sum_hessian_in_leaf": 100, "verbosity": 0,
+                  "use_quantized_grad": q}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 30)
+        aucs[q] = _auc(yv, bst.predict(Xv))
+    assert abs(aucs[True] - aucs[False]) < 5e-3, aucs
