@@ -159,6 +159,7 @@ class SerialTreeLearner : public TreeLearner {
   Random feature_rng_{0};
   Random extra_rng_{0};
   int iter_counter_ = 0;
+  bool build_both_children_ = false;  // voting-parallel: no histogram subtraction
   // CEGB / interaction-constraint state
   std::vector<int8_t> cegb_feature_used_;
   std::vector<std::set<int>> leaf_branch_features_;

@@ -141,6 +141,118 @@ class FeatureParallelTreeLearner : public SerialTreeLearner {
   }
 };
 
+/*! Voting-parallel learner (PV-Tree): each rank proposes its local top-k split
+ *  features, a global vote selects 2k features, and ONLY those features' histogram
+ *  ranges are globally reduced — O(k·bins) wire bytes instead of O(num_total_bin).
+ *  Both children build their own histograms (subtraction is invalid when only the
+ *  voted ranges are global). Parity: reference voting_parallel_tree_learner.cpp. */
+class VotingParallelTreeLearner : public SerialTreeLearner {
+ public:
+  explicit VotingParallelTreeLearner(const Config* config) : SerialTreeLearner(config) {}
+
+  void Init(const Dataset* train_data, bool is_constant_hessian) override {
+    SerialTreeLearner::Init(train_data, is_constant_hessian);
+    build_both_children_ = Network::is_distributed();
+    voted_mask_.assign(config_->num_leaves,
+                       std::vector<int8_t>(train_data->num_features(), 1));
+  }
+
+ protected:
+  void ReduceRootStats(double* sum_g, double* sum_h, data_size_t* cnt) override {
+    if (!Network::is_distributed()) return;
+    double v[2] = {*sum_g, *sum_h};
+    Network::AllreduceSum(v, 2);
+    *sum_g = v[0];
+    *sum_h = v[1];
+    *cnt = static_cast<data_size_t>(Network::GlobalSyncUpBySum(static_cast<int64_t>(*cnt)));
+  }
+  void GlobalChildCounts(data_size_t* left_cnt, data_size_t* right_cnt) override {
+    if (!Network::is_distributed()) return;
+    int64_t v[2] = {*left_cnt, *right_cnt};
+    Network::AllreduceSum(v, 2);
+    *left_cnt = static_cast<data_size_t>(v[0]);
+    *right_cnt = static_cast<data_size_t>(v[1]);
+  }
+
+  void OnHistogramReady(int leaf) override {
+    if (!Network::is_distributed()) return;
+    const int nf = train_data_->num_features();
+    const int k = std::min(config_->top_k, nf);
+    hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
+    const LeafContext& ctx = leaf_ctx_[leaf];
+    // 1. local candidate gain per feature
+    std::vector<std::pair<double, int>> gains(nf);
+#pragma omp parallel for schedule(static)
+    for (int f = 0; f < nf; ++f) {
+      SplitInfo si;
+      const BinMapper* m = train_data_->FeatureBinMapper(f);
+      const hist_t* fh = hist + 2 * train_data_->hist_offset(f);
+      if (m->bin_type() == BinType::kCategorical) {
+        FindBestThresholdCategorical(fh, m->num_bin(), ctx, *config_, &si);
+      } else {
+        FindBestThresholdNumerical(fh, m->num_bin(), m->num_numeric_bin(), m->nan_bin(),
+                                   ctx, *config_, 0, -1, &si);
+      }
+      gains[f] = {si.IsValid() ? si.gain : -1e308, f};
+    }
+    // 2. local top-k proposal
+    std::partial_sort(gains.begin(), gains.begin() + k, gains.end(),
+                      [](auto& a, auto& b) { return a.first > b.first; });
+    std::vector<int32_t> proposal(k);
+    for (int i = 0; i < k; ++i) proposal[i] = gains[i].second;
+    // 3. global vote: allgather proposals, count votes, take global top 2k
+    const int world = Network::num_machines();
+    std::vector<int32_t> all(static_cast<size_t>(world) * k);
+    Network::Allgather(reinterpret_cast<const char*>(proposal.data()),
+                       static_cast<int>(k * sizeof(int32_t)),
+                       reinterpret_cast<char*>(all.data()));
+    std::vector<int> votes(nf, 0);
+    for (int32_t f : all)
+      if (f >= 0 && f < nf) votes[f]++;
+    std::vector<int> order(nf);
+    for (int f = 0; f < nf; ++f) order[f] = f;
+    std::sort(order.begin(), order.end(), [&](int a, int b) {
+      return votes[a] > votes[b] || (votes[a] == votes[b] && a < b);
+    });
+    const int n_sel = std::min(2 * k, nf);
+    auto& mask = voted_mask_[leaf];
+    mask.assign(nf, 0);
+    for (int i = 0; i < n_sel; ++i)
+      if (votes[order[i]] > 0) mask[order[i]] = 1;
+    // 4. reduce ONLY the voted features' histogram ranges (compact buffer)
+    std::vector<hist_t> compact;
+    for (int f = 0; f < nf; ++f) {
+      if (!mask[f]) continue;
+      const hist_t* fh = hist + 2 * train_data_->hist_offset(f);
+      compact.insert(compact.end(), fh, fh + 2 * train_data_->FeatureNumBin(f));
+    }
+    Network::AllreduceSum(compact.data(), compact.size());
+    size_t off = 0;
+    for (int f = 0; f < nf; ++f) {
+      if (!mask[f]) continue;
+      hist_t* fh = hist + 2 * train_data_->hist_offset(f);
+      std::copy(compact.begin() + off,
+                compact.begin() + off + 2 * train_data_->FeatureNumBin(f), fh);
+      off += 2 * train_data_->FeatureNumBin(f);
+    }
+  }
+
+  void FindBestSplitForLeaf(int leaf, const LeafContext& ctx) override {
+    if (!Network::is_distributed()) {
+      SerialTreeLearner::FindBestSplitForLeaf(leaf, ctx);
+      return;
+    }
+    std::vector<int8_t> saved = is_feature_used_;
+    const int nf = train_data_->num_features();
+    for (int f = 0; f < nf; ++f) is_feature_used_[f] &= voted_mask_[leaf][f];
+    SerialTreeLearner::FindBestSplitForLeaf(leaf, ctx);
+    is_feature_used_ = saved;
+  }
+
+ private:
+  std::vector<std::vector<int8_t>> voted_mask_;
+};
+
 TreeLearner* (*g_create_hip_learner)(const Config*) = nullptr;
 
 TreeLearner* TreeLearner::Create(const std::string& learner_type,
@@ -155,11 +267,8 @@ TreeLearner* TreeLearner::Create(const std::string& learner_type,
   }
   if (learner_type == "serial") return new SerialTreeLearner(config);
   if (learner_type == "feature") return new FeatureParallelTreeLearner(config);
-  if (learner_type == "data" || learner_type == "voting") {
-    // voting currently maps to the data-parallel learner (full histogram reduction);
-    // the vote-filtered reduction is a wire-size optimization tracked for the GPU path.
-    return new DataParallelTreeLearner(config);
-  }
+  if (learner_type == "data") return new DataParallelTreeLearner(config);
+  if (learner_type == "voting") return new VotingParallelTreeLearner(config);
   Log::Fatal("Unknown tree learner %s", learner_type.c_str());
   return nullptr;
 }

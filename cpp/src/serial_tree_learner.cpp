@@ -370,8 +370,17 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     const data_size_t* small_idx = partition_.GetIndexOnLeaf(small_leaf, &small_cnt);
     ComputeHistogram(small_leaf, small_cnt, small_idx);
     OnHistogramReady(small_leaf);
-    // in-place: parent_slot -= small_slot -> becomes large hist
-    SubtractHistogram(large_leaf, parent_slot, spare_slot);
+    if (build_both_children_) {
+      // voting-parallel: each child's histogram is built + selectively reduced on its
+      // own voted features (subtraction is invalid when only voted ranges are global)
+      data_size_t large_cnt;
+      const data_size_t* large_idx = partition_.GetIndexOnLeaf(large_leaf, &large_cnt);
+      ComputeHistogram(large_leaf, large_cnt, large_idx);
+      OnHistogramReady(large_leaf);
+    } else {
+      // in-place: parent_slot -= small_slot -> becomes large hist
+      SubtractHistogram(large_leaf, parent_slot, spare_slot);
+    }
 
     FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
     FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);

@@ -91,3 +91,17 @@ def test_feature_parallel_runs(tmp_path):
          "--master-port", "29542", str(script), str(REPO)],
         capture_output=True, text=True, timeout=300, env=env)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+
+
+def test_voting_parallel_identical_models(tmp_path):
+    script = tmp_path / "worker_vp.py"
+    script.write_text(WORKER.replace('"tree_learner": "data"', '"tree_learner": "voting", "top_k": 3'))
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29543", str(script), str(REPO)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "DIST_OK" in r.stdout
