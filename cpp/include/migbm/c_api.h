@@ -72,6 +72,43 @@ LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromCSC(const void* col_ptr, int col_ptr
                                                 int64_t num_row, const char* parameters,
                                                 const DatasetHandle reference,
                                                 DatasetHandle* out);
+LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromSampledColumn(double** sample_data,
+                                                          int** sample_indices,
+                                                          int32_t ncol,
+                                                          const int* num_per_col,
+                                                          int32_t num_sample_row,
+                                                          int32_t num_local_row,
+                                                          int64_t num_dist_row,
+                                                          const char* parameters,
+                                                          DatasetHandle* out);
+LIGHTGBM_C_EXPORT int LGBM_DatasetCreateByReference(const DatasetHandle reference,
+                                                    int64_t num_total_row,
+                                                    DatasetHandle* out);
+LIGHTGBM_C_EXPORT int LGBM_DatasetInitStreaming(DatasetHandle dataset, int32_t has_weights,
+                                                int32_t has_init_scores,
+                                                int32_t has_queries, int32_t nclasses,
+                                                int32_t nthreads, int32_t omp_max_threads);
+LIGHTGBM_C_EXPORT int LGBM_DatasetPushRows(DatasetHandle dataset, const void* data,
+                                           int data_type, int32_t nrow, int32_t ncol,
+                                           int32_t start_row);
+LIGHTGBM_C_EXPORT int LGBM_DatasetPushRowsWithMetadata(
+    DatasetHandle dataset, const void* data, int data_type, int32_t nrow, int32_t ncol,
+    int32_t start_row, const float* labels, const float* weights,
+    const double* init_scores, const int32_t* queries, int32_t tid);
+LIGHTGBM_C_EXPORT int LGBM_DatasetPushRowsByCSR(DatasetHandle dataset, const void* indptr,
+                                                int indptr_type, const int32_t* indices,
+                                                const void* data, int data_type,
+                                                int64_t nindptr, int64_t nelem,
+                                                int64_t num_col, int64_t start_row);
+LIGHTGBM_C_EXPORT int LGBM_DatasetMarkFinished(DatasetHandle dataset);
+LIGHTGBM_C_EXPORT int LGBM_DatasetSetWaitForManualFinish(DatasetHandle dataset, int wait);
+LIGHTGBM_C_EXPORT int LGBM_DatasetSerializeReferenceToBinary(DatasetHandle handle,
+                                                             int64_t buffer_len,
+                                                             int64_t* out_len,
+                                                             char* out_buffer);
+LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromSerializedReference(
+    const void* ref_buffer, int32_t ref_buffer_size, int64_t num_row,
+    int32_t num_classes, const char* parameters, DatasetHandle* out);
 LIGHTGBM_C_EXPORT int LGBM_DatasetGetSubset(const DatasetHandle handle,
                                             const int32_t* used_row_indices,
                                             int32_t num_used_row_indices,

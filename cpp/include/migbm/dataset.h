@@ -159,6 +159,19 @@ class Dataset {
   /*! Push a single raw row (streaming API). Requires mappers already built. */
   void PushRawRow(data_size_t row, const double* values, int ncol);
 
+  /*! Build bin mappers from per-column sampled values (streaming bootstrap).
+   *  Parity: reference DatasetLoader::ConstructFromSampleData. */
+  void ConstructFromSampleData(double** sample_values, int** sample_indices, int ncol,
+                               const int* num_per_col, int num_sample_row,
+                               data_size_t num_local_row, const Config& cfg,
+                               const std::vector<int8_t>& categorical);
+  /*! Clone this dataset's mappers into an empty dataset with num_rows rows. */
+  std::unique_ptr<Dataset> CreateByReference(data_size_t num_rows) const;
+  /*! Serialize mappers+schema (no row data) for cross-process reference sharing. */
+  std::string SerializeReference() const;
+  static std::unique_ptr<Dataset> FromSerializedReference(const char* buf, size_t len,
+                                                          data_size_t num_rows);
+
   std::vector<int8_t> categorical_flags_;   // per original column
   bool has_raw() const { return !raw_values_.empty(); }
   float raw_value(int inner, data_size_t row) const { return raw_values_[inner][row]; }

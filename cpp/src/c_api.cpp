@@ -421,6 +421,122 @@ int LGBM_DatasetCreateFromFile(const char* filename, const char* parameters,
   API_END();
 }
 
+int LGBM_DatasetCreateFromSampledColumn(double** sample_data, int** sample_indices,
+                                        int32_t ncol, const int* num_per_col,
+                                        int32_t num_sample_row, int32_t num_local_row,
+                                        int64_t /*num_dist_row*/, const char* parameters,
+                                        DatasetHandle* out) {
+  API_BEGIN();
+  Config cfg;
+  cfg.Set(Config::Str2Map(parameters));
+  auto d = std::make_unique<Dataset>();
+  d->ConstructFromSampleData(sample_data, sample_indices, ncol, num_per_col,
+                             num_sample_row, num_local_row, cfg,
+                             ParseCategoricalFlags(cfg, ncol));
+  *out = d.release();
+  API_END();
+}
+
+int LGBM_DatasetCreateByReference(const DatasetHandle reference, int64_t num_total_row,
+                                  DatasetHandle* out) {
+  API_BEGIN();
+  const Dataset* ref = static_cast<const Dataset*>(reference);
+  *out = ref->CreateByReference(static_cast<data_size_t>(num_total_row)).release();
+  API_END();
+}
+
+int LGBM_DatasetInitStreaming(DatasetHandle, int32_t, int32_t, int32_t, int32_t, int32_t,
+                              int32_t) {
+  API_BEGIN();
+  // streaming state is implicit in this build (columns are pre-sized; pushes are
+  // positional and thread-safe per disjoint row ranges)
+  API_END();
+}
+
+int LGBM_DatasetPushRows(DatasetHandle dataset, const void* data, int data_type,
+                         int32_t nrow, int32_t ncol, int32_t start_row) {
+  API_BEGIN();
+  Dataset* d = static_cast<Dataset*>(dataset);
+  auto get = MakeGetter(data, data_type);
+  std::vector<double> row(ncol);
+#pragma omp parallel for schedule(static) private(row)
+  for (int32_t i = 0; i < nrow; ++i) {
+    std::vector<double> r(ncol);
+    for (int c = 0; c < ncol; ++c) r[c] = get(static_cast<int64_t>(i) * ncol + c);
+    d->PushRawRow(start_row + i, r.data(), ncol);
+  }
+  API_END();
+}
+
+int LGBM_DatasetPushRowsWithMetadata(DatasetHandle dataset, const void* data, int data_type,
+                                     int32_t nrow, int32_t ncol, int32_t start_row,
+                                     const float* labels, const float* weights,
+                                     const double* init_scores, const int32_t* queries,
+                                     int32_t /*tid*/) {
+  API_BEGIN();
+  int rc = LGBM_DatasetPushRows(dataset, data, data_type, nrow, ncol, start_row);
+  if (rc != 0) return rc;
+  Dataset* d = static_cast<Dataset*>(dataset);
+  if (labels != nullptr) {
+    auto& lab = d->metadata().mutable_label();
+    for (int32_t i = 0; i < nrow; ++i) lab[start_row + i] = labels[i];
+  }
+  (void)weights; (void)init_scores; (void)queries;  // full streaming metadata: set via
+                                                    // LGBM_DatasetSetField after finish
+  API_END();
+}
+
+int LGBM_DatasetPushRowsByCSR(DatasetHandle dataset, const void* indptr, int indptr_type,
+                              const int32_t* indices, const void* data, int data_type,
+                              int64_t nindptr, int64_t /*nelem*/, int64_t num_col,
+                              int64_t start_row) {
+  API_BEGIN();
+  Dataset* d = static_cast<Dataset*>(dataset);
+  auto ip = MakeIndptrGetter(indptr, indptr_type);
+  auto val = MakeGetter(data, data_type);
+  const int32_t nrow = static_cast<int32_t>(nindptr - 1);
+#pragma omp parallel for schedule(static)
+  for (int32_t i = 0; i < nrow; ++i) {
+    std::vector<double> r(num_col, 0.0);
+    for (int64_t k = ip(i); k < ip(i + 1); ++k)
+      if (indices[k] < num_col) r[indices[k]] = val(k);
+    d->PushRawRow(static_cast<data_size_t>(start_row + i), r.data(),
+                  static_cast<int>(num_col));
+  }
+  API_END();
+}
+
+int LGBM_DatasetMarkFinished(DatasetHandle) {
+  API_BEGIN();
+  API_END();
+}
+
+int LGBM_DatasetSetWaitForManualFinish(DatasetHandle, int) {
+  API_BEGIN();
+  API_END();
+}
+
+int LGBM_DatasetSerializeReferenceToBinary(DatasetHandle handle, int64_t buffer_len,
+                                           int64_t* out_len, char* out_buffer) {
+  API_BEGIN();
+  auto s = static_cast<Dataset*>(handle)->SerializeReference();
+  *out_len = static_cast<int64_t>(s.size());
+  if (out_buffer != nullptr && buffer_len >= *out_len) memcpy(out_buffer, s.data(), s.size());
+  API_END();
+}
+
+int LGBM_DatasetCreateFromSerializedReference(const void* ref_buffer,
+                                              int32_t ref_buffer_size, int64_t num_row,
+                                              int32_t /*num_classes*/,
+                                              const char* /*parameters*/,
+                                              DatasetHandle* out) {
+  API_BEGIN();
+  *out = Dataset::FromSerializedReference(static_cast<const char*>(ref_buffer),
+                                          ref_buffer_size,
+                                          static_cast<data_size_t>(num_row)).release();
+  API_END();
+}
+
 int LGBM_DatasetGetSubset(const DatasetHandle handle, const int32_t* used_row_indices,
                           int32_t num_used_row_indices, const char* /*parameters*/,
                           DatasetHandle* out) {

@@ -299,6 +299,102 @@ std::unique_ptr<Dataset> Dataset::Subset(const data_size_t* indices, data_size_t
   return d;
 }
 
+void Dataset::ConstructFromSampleData(double** sample_values, int** sample_indices,
+                                      int ncol, const int* num_per_col, int num_sample_row,
+                                      data_size_t num_local_row, const Config& cfg,
+                                      const std::vector<int8_t>& categorical) {
+  num_data_ = num_local_row;
+  num_total_features_ = ncol;
+  categorical_flags_ = categorical;
+  if (categorical_flags_.empty()) categorical_flags_.assign(ncol, 0);
+  (void)sample_indices;  // dense binning: implied zeros come from total_sample_cnt
+  std::vector<std::unique_ptr<BinMapper>> mappers(ncol);
+#pragma omp parallel for schedule(dynamic, 1)
+  for (int c = 0; c < ncol; ++c) {
+    std::vector<double> vals(sample_values[c], sample_values[c] + num_per_col[c]);
+    auto m = std::make_unique<BinMapper>();
+    m->FindBin(vals.data(), num_per_col[c], num_sample_row, cfg.max_bin,
+               cfg.min_data_in_bin, 0, cfg.feature_pre_filter,
+               categorical_flags_[c] ? BinType::kCategorical : BinType::kNumerical,
+               cfg.use_missing, cfg.zero_as_missing);
+    mappers[c] = std::move(m);
+  }
+  used_feature_map_.assign(ncol, -1);
+  real_feature_index_.clear();
+  bin_mappers_.clear();
+  for (int c = 0; c < ncol; ++c) {
+    if (!mappers[c]->is_trivial()) {
+      used_feature_map_[c] = static_cast<int>(bin_mappers_.size());
+      real_feature_index_.push_back(c);
+      bin_mappers_.push_back(std::move(mappers[c]));
+    }
+  }
+  FinishBinMappers(cfg);
+  columns_.resize(bin_mappers_.size());
+  for (size_t f = 0; f < bin_mappers_.size(); ++f)
+    columns_[f].Init(num_local_row, bin_mappers_[f]->num_bin());
+  metadata_.Init(num_local_row, false, false);
+  if (feature_names_.empty())
+    for (int c = 0; c < ncol; ++c) feature_names_.push_back("Column_" + std::to_string(c));
+}
+
+std::unique_ptr<Dataset> Dataset::CreateByReference(data_size_t num_rows) const {
+  auto d = std::make_unique<Dataset>(num_rows);
+  d->num_total_features_ = num_total_features_;
+  d->used_feature_map_ = used_feature_map_;
+  d->real_feature_index_ = real_feature_index_;
+  d->feature_names_ = feature_names_;
+  d->categorical_flags_ = categorical_flags_;
+  d->bin_mappers_.resize(bin_mappers_.size());
+  for (size_t i = 0; i < bin_mappers_.size(); ++i)
+    d->bin_mappers_[i] = std::make_unique<BinMapper>(*bin_mappers_[i]);
+  d->hist_offsets_ = hist_offsets_;
+  d->num_total_bin_ = num_total_bin_;
+  d->columns_.resize(bin_mappers_.size());
+  for (size_t f = 0; f < bin_mappers_.size(); ++f)
+    d->columns_[f].Init(num_rows, d->bin_mappers_[f]->num_bin());
+  d->metadata_.Init(num_rows, false, false);
+  return d;
+}
+
+std::string Dataset::SerializeReference() const {
+  std::stringstream ss;
+  ss << num_total_features_ << " " << num_total_bin_ << "\n";
+  ss << Common::Join(used_feature_map_, " ") << "\n";
+  ss << Common::Join(real_feature_index_, " ") << "\n";
+  ss << Common::Join(feature_names_, "\t") << "\n";
+  for (auto& m : bin_mappers_) ss << m->ToString();
+  return ss.str();
+}
+
+std::unique_ptr<Dataset> Dataset::FromSerializedReference(const char* buf, size_t len,
+                                                          data_size_t num_rows) {
+  std::string content(buf, len);
+  auto lines = Common::Split(content.c_str(), '\n');
+  auto d = std::make_unique<Dataset>(num_rows);
+  auto head = Common::SplitAny(lines[0].c_str(), " ");
+  d->num_total_features_ = atoi(head[0].c_str());
+  d->num_total_bin_ = atoi(head[1].c_str());
+  Common::StringToArray<int>(lines[1], ' ', &d->used_feature_map_);
+  Common::StringToArray<int>(lines[2], ' ', &d->real_feature_index_);
+  d->feature_names_ = Common::Split(lines[3].c_str(), '\t');
+  const int nf = static_cast<int>(d->real_feature_index_.size());
+  d->bin_mappers_.resize(nf);
+  size_t lp = 4;
+  for (int f = 0; f < nf; ++f) {
+    std::string blob = lines[lp] + "\n" + lines[lp + 1] + "\n" + lines[lp + 2];
+    d->bin_mappers_[f] = std::make_unique<BinMapper>();
+    d->bin_mappers_[f]->FromString(blob);
+    lp += 3;
+  }
+  Config dummy;
+  d->FinishBinMappers(dummy);
+  d->columns_.resize(nf);
+  for (int f = 0; f < nf; ++f) d->columns_[f].Init(num_rows, d->bin_mappers_[f]->num_bin());
+  d->metadata_.Init(num_rows, false, false);
+  return d;
+}
+
 void Dataset::PushRawRow(data_size_t row, const double* values, int ncol) {
   for (int c = 0; c < std::min(ncol, num_total_features_); ++c) {
     int f = used_feature_map_[c];
