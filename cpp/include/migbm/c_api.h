@@ -109,6 +109,19 @@ LIGHTGBM_C_EXPORT int LGBM_DatasetSerializeReferenceToBinary(DatasetHandle handl
 LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromSerializedReference(
     const void* ref_buffer, int32_t ref_buffer_size, int64_t num_row,
     int32_t num_classes, const char* parameters, DatasetHandle* out);
+struct ArrowArray;
+struct ArrowSchema;
+LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromArrow(int64_t n_chunks,
+                                                  const struct ArrowArray* chunks,
+                                                  const struct ArrowSchema* schema,
+                                                  const char* parameters,
+                                                  const DatasetHandle reference,
+                                                  DatasetHandle* out);
+LIGHTGBM_C_EXPORT int LGBM_DatasetSetFieldFromArrow(DatasetHandle handle,
+                                                    const char* field_name,
+                                                    int64_t n_chunks,
+                                                    const struct ArrowArray* chunks,
+                                                    const struct ArrowSchema* schema);
 LIGHTGBM_C_EXPORT int LGBM_DatasetGetSubset(const DatasetHandle handle,
                                             const int32_t* used_row_indices,
                                             int32_t num_used_row_indices,

@@ -2,6 +2,8 @@
  *  wrapper, dataset ingestion paths (mat/CSR/CSC/file) and prediction entry points. */
 #include "migbm/c_api.h"
 
+#include "migbm/arrow.h"
+
 #include "migbm/boosting.h"
 #include "migbm/common.h"
 #include "migbm/config.h"
@@ -534,6 +536,163 @@ int LGBM_DatasetCreateFromSerializedReference(const void* ref_buffer,
   *out = Dataset::FromSerializedReference(static_cast<const char*>(ref_buffer),
                                           ref_buffer_size,
                                           static_cast<data_size_t>(num_row)).release();
+  API_END();
+}
+
+// ------------------------------------------------------------------ Arrow ingestion
+namespace {
+
+std::function<double(int64_t)> ArrowColGetter(const ArrowArray* a, const char* fmt) {
+  const uint8_t* validity = a->n_buffers > 0 ? static_cast<const uint8_t*>(a->buffers[0])
+                                             : nullptr;
+  const void* data = a->n_buffers > 1 ? a->buffers[1] : nullptr;
+  const int64_t off = a->offset;
+  auto is_null = [validity, off](int64_t i) {
+    if (validity == nullptr) return false;
+    const int64_t j = i + off;
+    return ((validity[j >> 3] >> (j & 7)) & 1) == 0;
+  };
+  const double kNaN = std::numeric_limits<double>::quiet_NaN();
+  switch (fmt[0]) {
+    case 'g': {
+      const double* p = static_cast<const double*>(data);
+      return [p, off, is_null, kNaN](int64_t i) { return is_null(i) ? kNaN : p[i + off]; };
+    }
+    case 'f': {
+      const float* p = static_cast<const float*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'l': {
+      const int64_t* p = static_cast<const int64_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'L': {
+      const uint64_t* p = static_cast<const uint64_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'i': {
+      const int32_t* p = static_cast<const int32_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'I': {
+      const uint32_t* p = static_cast<const uint32_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 's': {
+      const int16_t* p = static_cast<const int16_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'c': {
+      const int8_t* p = static_cast<const int8_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'C': {
+      const uint8_t* p = static_cast<const uint8_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        return is_null(i) ? kNaN : static_cast<double>(p[i + off]);
+      };
+    }
+    case 'b': {
+      const uint8_t* p = static_cast<const uint8_t*>(data);
+      return [p, off, is_null, kNaN](int64_t i) {
+        if (is_null(i)) return kNaN;
+        const int64_t j = i + off;
+        return static_cast<double>((p[j >> 3] >> (j & 7)) & 1);
+      };
+    }
+  }
+  Log::Fatal("Unsupported Arrow column format '%s'", fmt);
+  return nullptr;
+}
+
+}  // namespace
+
+int LGBM_DatasetCreateFromArrow(int64_t n_chunks, const struct ArrowArray* chunks,
+                                const struct ArrowSchema* schema, const char* parameters,
+                                const DatasetHandle reference, DatasetHandle* out) {
+  API_BEGIN();
+  MIGBM_CHECK(schema->n_children > 0);
+  const int ncol = static_cast<int>(schema->n_children);
+  // per-chunk per-column getters + chunk row offsets
+  std::vector<int64_t> chunk_start(n_chunks + 1, 0);
+  std::vector<std::vector<std::function<double(int64_t)>>> getters(n_chunks);
+  for (int64_t c = 0; c < n_chunks; ++c) {
+    chunk_start[c + 1] = chunk_start[c] + chunks[c].length;
+    MIGBM_CHECK_EQ(chunks[c].n_children, schema->n_children);
+    getters[c].resize(ncol);
+    for (int f = 0; f < ncol; ++f)
+      getters[c][f] = ArrowColGetter(chunks[c].children[f], schema->children[f]->format);
+  }
+  const data_size_t nrow = static_cast<data_size_t>(chunk_start[n_chunks]);
+  auto at = [&](data_size_t r, int col) -> double {
+    // chunk lookup (chunks are few; linear scan with memo would do, binary search is fine)
+    int64_t lo = 0, hi = n_chunks - 1;
+    while (lo < hi) {
+      const int64_t mid = (lo + hi + 1) >> 1;
+      if (r >= chunk_start[mid]) lo = mid;
+      else hi = mid - 1;
+    }
+    return getters[lo][col](r - chunk_start[lo]);
+  };
+  if (reference != nullptr) {
+    *out = static_cast<const Dataset*>(reference)->CreateValid(at, nrow).release();
+  } else {
+    Config cfg;
+    cfg.Set(Config::Str2Map(parameters));
+    auto d = std::make_unique<Dataset>();
+    // column names from the schema
+    std::vector<std::string> names;
+    for (int f = 0; f < ncol; ++f)
+      names.push_back(schema->children[f]->name ? schema->children[f]->name
+                                                : "Column_" + std::to_string(f));
+    d->ConstructFromMat(at, nrow, ncol, cfg, ParseCategoricalFlags(cfg, ncol));
+    d->set_feature_names(names);
+    *out = d.release();
+  }
+  API_END();
+}
+
+int LGBM_DatasetSetFieldFromArrow(DatasetHandle handle, const char* field_name,
+                                  int64_t n_chunks, const struct ArrowArray* chunks,
+                                  const struct ArrowSchema* schema) {
+  API_BEGIN();
+  // single flat numeric column expected
+  int64_t total = 0;
+  for (int64_t c = 0; c < n_chunks; ++c) total += chunks[c].length;
+  std::vector<double> vals;
+  vals.reserve(total);
+  for (int64_t c = 0; c < n_chunks; ++c) {
+    auto get = ArrowColGetter(&chunks[c], schema->format);
+    for (int64_t i = 0; i < chunks[c].length; ++i) vals.push_back(get(i));
+  }
+  std::string name(field_name);
+  Dataset* d = static_cast<Dataset*>(handle);
+  if (name == "label" || name == "weight") {
+    std::vector<float> f(vals.begin(), vals.end());
+    if (name == "label") d->metadata().SetLabel(f.data(), static_cast<data_size_t>(f.size()));
+    else d->metadata().SetWeights(f.data(), static_cast<data_size_t>(f.size()));
+  } else if (name == "group" || name == "query") {
+    std::vector<int32_t> g(vals.begin(), vals.end());
+    d->metadata().SetQuery(g.data(), static_cast<data_size_t>(g.size()));
+  } else if (name == "init_score") {
+    d->metadata().SetInitScore(vals.data(), static_cast<int64_t>(vals.size()));
+  } else {
+    Log::Fatal("Unknown field %s", field_name);
+  }
   API_END();
 }
 

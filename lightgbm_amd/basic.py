@@ -91,6 +91,52 @@ def _np_float32(data):
     return np.ascontiguousarray(np.asarray(data).ravel(), dtype=np.float32)
 
 
+def _is_pyarrow_table(data):
+    try:
+        import pyarrow as pa
+        return isinstance(data, pa.Table)
+    except ImportError:
+        return False
+
+
+def _create_dataset_from_arrow(table, param_str, ref_handle):
+    """Export record batches via the Arrow C data interface and build the dataset."""
+    import pyarrow as pa  # noqa: F401
+
+    class _ArrowArray(ctypes.Structure):
+        _fields_ = [("length", ctypes.c_int64), ("null_count", ctypes.c_int64),
+                    ("offset", ctypes.c_int64), ("n_buffers", ctypes.c_int64),
+                    ("n_children", ctypes.c_int64), ("buffers", ctypes.c_void_p),
+                    ("children", ctypes.c_void_p), ("dictionary", ctypes.c_void_p),
+                    ("release", ctypes.c_void_p), ("private_data", ctypes.c_void_p)]
+
+    class _ArrowSchema(ctypes.Structure):
+        _fields_ = [("format", ctypes.c_char_p), ("name", ctypes.c_char_p),
+                    ("metadata", ctypes.c_char_p), ("flags", ctypes.c_int64),
+                    ("n_children", ctypes.c_int64), ("children", ctypes.c_void_p),
+                    ("dictionary", ctypes.c_void_p), ("release", ctypes.c_void_p),
+                    ("private_data", ctypes.c_void_p)]
+
+    batches = table.to_batches()
+    if not batches:
+        raise ValueError("empty pyarrow Table")
+    chunks = (_ArrowArray * len(batches))()
+    schema = _ArrowSchema()
+    for i, batch in enumerate(batches):
+        batch._export_to_c(ctypes.addressof(chunks[i]),
+                           ctypes.addressof(schema) if i == 0 else None)             if False else None
+        # export array always; schema once
+        if i == 0:
+            batch._export_to_c(ctypes.addressof(chunks[i]), ctypes.addressof(schema))
+        else:
+            batch._export_to_c(ctypes.addressof(chunks[i]))
+    out = ctypes.c_void_p()
+    _safe_call(_LIB.LGBM_DatasetCreateFromArrow(
+        ctypes.c_int64(len(batches)), chunks, ctypes.byref(schema), _c_str(param_str),
+        ref_handle, ctypes.byref(out)))
+    return out
+
+
 class Dataset:
     """Binned training dataset (parity: reference lgb.Dataset)."""
 
@@ -129,7 +175,9 @@ class Dataset:
         param_str = _param_dict_to_str(params)
         ref_handle = self.reference._handle if self.reference is not None else None
 
-        if isinstance(self.data, (str, Path)):
+        if _is_pyarrow_table(self.data):
+            self._handle = _create_dataset_from_arrow(self.data, param_str, ref_handle)
+        elif isinstance(self.data, (str, Path)):
             out = ctypes.c_void_p()
             _safe_call(_LIB.LGBM_DatasetCreateFromFile(
                 _c_str(str(self.data)), _c_str(param_str), ref_handle, ctypes.byref(out)))

@@ -1,0 +1,47 @@
+"""Arrow ingestion tests (parity target: reference tests/python_package_test/test_arrow.py)."""
+import numpy as np
+import pytest
+
+pa = pytest.importorskip("pyarrow")
+
+import lightgbm_amd as lgb
+
+
+def test_dataset_from_arrow_table():
+    rng = np.random.RandomState(0)
+    n = 3000
+    cols = {f"f{i}": rng.randn(n) for i in range(5)}
+    cols["f2"] = rng.randint(0, 100, size=n).astype(np.int64)  # integer column
+    table = pa.table(cols)
+    y = (np.asarray(cols["f0"]) > 0).astype(np.float32)
+    ds = lgb.Dataset(table, label=y).construct()
+    assert ds.num_data() == n
+    assert ds.num_feature() == 5
+    assert ds.get_feature_name() == ["f0", "f1", "f2", "f3", "f4"]
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, ds, 10)
+    X = np.column_stack([np.asarray(cols[f"f{i}"], dtype=np.float64) for i in range(5)])
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.9
+
+
+def test_arrow_with_nulls():
+    rng = np.random.RandomState(1)
+    n = 1000
+    vals = rng.randn(n)
+    mask = rng.rand(n) < 0.1
+    arr = pa.array(np.where(mask, np.nan, vals), from_pandas=True)  # nulls from NaN
+    table = pa.table({"a": arr, "b": pa.array(rng.randn(n))})
+    y = (vals > 0).astype(np.float32)
+    ds = lgb.Dataset(table, label=y).construct()
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, ds, 5)
+    assert np.all(np.isfinite(bst.predict(np.column_stack([vals, rng.randn(n)]))))
+
+
+def test_arrow_multi_chunk():
+    rng = np.random.RandomState(2)
+    t1 = pa.table({"x": rng.randn(500), "z": rng.randn(500)})
+    t2 = pa.table({"x": rng.randn(700), "z": rng.randn(700)})
+    table = pa.concat_tables([t1, t2])
+    y = np.zeros(1200, dtype=np.float32)
+    ds = lgb.Dataset(table, label=y).construct()
+    assert ds.num_data() == 1200
