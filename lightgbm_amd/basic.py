@@ -123,9 +123,7 @@ def _create_dataset_from_arrow(table, param_str, ref_handle):
     chunks = (_ArrowArray * len(batches))()
     schema = _ArrowSchema()
     for i, batch in enumerate(batches):
-        batch._export_to_c(ctypes.addressof(chunks[i]),
-                           ctypes.addressof(schema) if i == 0 else None)             if False else None
-        # export array always; schema once
+        # export every array; the schema only once
         if i == 0:
             batch._export_to_c(ctypes.addressof(chunks[i]), ctypes.addressof(schema))
         else:
