@@ -40,6 +40,42 @@ def make_higgs_like(n, d, seed, dtype=np.float32):
     return X, y
 
 
+def make_mslr_like(n_rows_target, d, seed):
+    """Synthetic MSLR-WEB30K-shaped ranking data: ~120-doc queries, 136 features,
+    graded relevance 0-4 driven by a few features."""
+    rng = np.random.RandomState(seed)
+    qsizes = []
+    total = 0
+    while total < n_rows_target:
+        sz = int(rng.randint(40, 200))
+        qsizes.append(sz)
+        total += sz
+    n = int(np.sum(qsizes))
+    X = rng.randn(n, d).astype(np.float32)
+    rel = np.clip(1.3 * X[:, 0] + 0.8 * X[:, 1] * (X[:, 2] > 0) +
+                  0.9 * rng.randn(n) + 1.0, 0, 4).astype(np.float32)
+    return X, np.floor(rel).astype(np.float32), np.asarray(qsizes, dtype=np.int32)
+
+
+def ndcg_at_k(y, p, qsizes, k=10):
+    start = 0
+    total, nq = 0.0, 0
+    for sz in qsizes:
+        yy = y[start:start + sz]
+        pp = p[start:start + sz]
+        start += sz
+        order = np.argsort(-pp, kind="stable")
+        gains = (2.0 ** yy) - 1
+        kk = min(k, sz)
+        disc = 1.0 / np.log2(2 + np.arange(kk))
+        dcg = float((gains[order[:kk]] * disc).sum())
+        idcg = float((np.sort(gains)[::-1][:kk] * disc).sum())
+        if idcg > 0:
+            total += dcg / idcg
+            nq += 1
+    return total / max(1, nq)
+
+
 def auc_score(y, p):
     order = np.argsort(-p, kind="stable")
     ys = y[order]
@@ -64,6 +100,8 @@ def main():
     ap.add_argument("--num-leaves", type=int, default=255)
     ap.add_argument("--device", default="gpu", choices=["gpu", "cpu"])
     ap.add_argument("--valid-rows", type=int, default=500_000)
+    ap.add_argument("--task", default="binary", choices=["binary", "ranking"],
+                    help="ranking = MSLR-WEB30K-shaped lambdarank (3Mx136 default)")
     ap.add_argument("--quantized", action="store_true",
                     help="use_quantized_grad: packed int histograms (reference's "
                          "quantized training mode; halves LDS atomics + wire bytes)")
@@ -112,19 +150,30 @@ def main():
     # same reference dataset from a common-seed sample, then bin their own shard
     # against it (deterministic; no broadcast needed).
     t0 = time.time()
+    ranking = args.task == "ranking"
+    if ranking and args.rows == 10_000_000:
+        args.rows = 3_000_000
+        args.features = 136
     d = args.features
-    ref_X, ref_y = make_higgs_like(min(200_000, args.rows), d, seed=1234)
-    params_ds = {"max_bin": args.max_bin, "bin_construct_sample_cnt": 200_000}
-    ref = lgb.Dataset(ref_X, label=ref_y, params=params_ds).construct()
-    X, y = make_higgs_like(args.rows, d, seed=100 + rank)
-    train = ref.create_valid(X, label=y)
+    if ranking:
+        ref_X, ref_y, ref_q = make_mslr_like(min(200_000, args.rows), d, seed=1234)
+        params_ds = {"max_bin": args.max_bin, "bin_construct_sample_cnt": 200_000}
+        ref = lgb.Dataset(ref_X, label=ref_y, group=ref_q, params=params_ds).construct()
+        X, y, qsizes = make_mslr_like(args.rows, d, seed=100 + rank)
+        train = ref.create_valid(X, label=y, group=qsizes)
+    else:
+        ref_X, ref_y = make_higgs_like(min(200_000, args.rows), d, seed=1234)
+        params_ds = {"max_bin": args.max_bin, "bin_construct_sample_cnt": 200_000}
+        ref = lgb.Dataset(ref_X, label=ref_y, params=params_ds).construct()
+        X, y = make_higgs_like(args.rows, d, seed=100 + rank)
+        train = ref.create_valid(X, label=y)
     train.construct()
     del X
     data_s = time.time() - t0
 
     params = {
-        "objective": "binary",
-        "metric": "auc",
+        "objective": "lambdarank" if ranking else "binary",
+        "metric": "ndcg" if ranking else "auc",
         "device_type": "gpu" if use_gpu else "cpu",
         "tree_learner": "data" if world > 1 else "serial",
         "max_bin": args.max_bin,
@@ -165,9 +214,14 @@ def main():
     # AUC on a held-out synthetic valid set (outside the timed region, rank 0 only)
     auc = None
     if rank == 0 and args.valid_rows > 0:
-        Xv, yv = make_higgs_like(args.valid_rows, d, seed=99991)
-        pred = booster.predict(Xv)
-        auc = auc_score(yv, pred)
+        if ranking:
+            Xv, yv, qv = make_mslr_like(args.valid_rows, d, seed=99991)
+            pred = booster.predict(Xv)
+            auc = ndcg_at_k(yv, pred, qv, k=10)
+        else:
+            Xv, yv = make_higgs_like(args.valid_rows, d, seed=99991)
+            pred = booster.predict(Xv)
+            auc = auc_score(yv, pred)
 
     if rank == 0:
         result = {
@@ -184,8 +238,9 @@ def main():
             "dtype": "int16hist+fp64gain" if args.quantized else "fp32hist+fp64gain",
             "data": "synthetic",
             "auc": auc,
+            "quality_metric": "ndcg@10" if ranking else "auc",
             "config": {
-                "model": "higgs-gbdt-binary",
+                "model": "mslr-lambdarank" if ranking else "higgs-gbdt-binary",
                 "rows_per_gpu": args.rows,
                 "features": args.features,
                 "max_bin": args.max_bin,

@@ -894,6 +894,115 @@ __global__ void k_grad_l2(const double* __restrict__ score, const float* __restr
   h[i] = static_cast<float>(w);
 }
 
+/*! lambdarank NDCG gradients: one block per query. Scores are argsorted in LDS
+ *  (bitonic, padded to a power of two <= 1024); pairwise lambdas accumulate into
+ *  LDS grad/hess; optional per-query normalization. Capability parity: the
+ *  reference's GetGradientsKernel_LambdarankNDCG (fresh wave64 implementation). */
+__global__ void __launch_bounds__(256) k_grad_lambdarank(
+    const double* __restrict__ score, const float* __restrict__ label,
+    const int* __restrict__ qb, int num_queries, const double* __restrict__ inv_max_dcg,
+    const double* __restrict__ label_gain, double sigmoid, int trunc, int norm,
+    float* __restrict__ out_g, float* __restrict__ out_h) {
+  __shared__ float s_score[1024];
+  __shared__ short s_idx[1024];
+  __shared__ float s_g[1024], s_h[1024];
+  __shared__ double s_suml[4];
+  const int q = blockIdx.x;
+  if (q >= num_queries) return;
+  const int start = qb[q];
+  const int cnt = qb[q + 1] - start;
+  if (cnt > 1024) return;  // oversized queries are handled by the host fallback
+  int pow2 = 1;
+  while (pow2 < cnt) pow2 <<= 1;
+  for (int i = threadIdx.x; i < pow2; i += blockDim.x) {
+    s_score[i] = i < cnt ? static_cast<float>(score[start + i]) : -3.0e38f;
+    s_idx[i] = static_cast<short>(i);
+    if (i < cnt) {
+      s_g[i] = 0.0f;
+      s_h[i] = 0.0f;
+    }
+  }
+  __syncthreads();
+  // bitonic sort descending by score
+  for (int ksz = 2; ksz <= pow2; ksz <<= 1) {
+    for (int j = ksz >> 1; j > 0; j >>= 1) {
+      for (int i = threadIdx.x; i < pow2; i += blockDim.x) {
+        const int ixj = i ^ j;
+        if (ixj > i) {
+          const bool up = (i & ksz) == 0;  // descending overall
+          const bool swap = up ? (s_score[i] < s_score[ixj]) : (s_score[i] > s_score[ixj]);
+          if (swap) {
+            const float ts = s_score[i];
+            s_score[i] = s_score[ixj];
+            s_score[ixj] = ts;
+            const short ti = s_idx[i];
+            s_idx[i] = s_idx[ixj];
+            s_idx[ixj] = ti;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  const double imd = inv_max_dcg[q];
+  if (imd <= 0) {
+    for (int i = threadIdx.x; i < cnt; i += blockDim.x) {
+      out_g[start + i] = 0.0f;
+      out_h[start + i] = 0.0f;
+    }
+    return;
+  }
+  const int t = min(trunc, cnt);
+  const float best = s_score[0];
+  const float worst = s_score[cnt - 1];
+  double local_suml = 0.0;
+  // pair (i, j) with i < t, i < j < cnt, strided over threads
+  const int64_t n_pairs = static_cast<int64_t>(t) * cnt;
+  for (int64_t p = threadIdx.x; p < n_pairs; p += blockDim.x) {
+    const int i = static_cast<int>(p / cnt);
+    const int j = static_cast<int>(p % cnt);
+    if (j <= i) continue;
+    const int di = s_idx[i], dj = s_idx[j];
+    const float li = label[start + di], lj = label[start + dj];
+    if (li == lj) continue;
+    int hi_rank, lo_rank, hi_doc, lo_doc;
+    if (li > lj) {
+      hi_rank = i; lo_rank = j; hi_doc = di; lo_doc = dj;
+    } else {
+      hi_rank = j; lo_rank = i; hi_doc = dj; lo_doc = di;
+    }
+    const double high_gain = label_gain[static_cast<int>(label[start + hi_doc])];
+    const double low_gain = label_gain[static_cast<int>(label[start + lo_doc])];
+    const double ds = score[start + hi_doc] - score[start + lo_doc];
+    const double high_disc = 1.0 / log2(2.0 + hi_rank);
+    const double low_disc = 1.0 / log2(2.0 + lo_rank);
+    double delta = fabs((high_gain - low_gain) * (high_disc - low_disc) * imd);
+    if (best != worst) delta /= (0.01 + fabs(ds));
+    double p_lambda = 1.0 / (1.0 + exp(sigmoid * ds));
+    double p_hess = p_lambda * (1.0 - p_lambda);
+    p_lambda *= -sigmoid * delta;
+    p_hess *= sigmoid * sigmoid * delta;
+    atomicAdd(&s_g[hi_doc], static_cast<float>(p_lambda));
+    atomicAdd(&s_h[hi_doc], static_cast<float>(p_hess));
+    atomicAdd(&s_g[lo_doc], static_cast<float>(-p_lambda));
+    atomicAdd(&s_h[lo_doc], static_cast<float>(p_hess));
+    local_suml -= 2.0 * p_lambda;
+  }
+  // block reduce sum_lambdas
+  for (int d = 32; d > 0; d >>= 1) local_suml += __shfl_down(local_suml, d);
+  const int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) s_suml[wave] = local_suml;
+  __syncthreads();
+  double suml = 0;
+  for (int w = 0; w < static_cast<int>(blockDim.x / 64); ++w) suml += s_suml[w];
+  double nf = 1.0;
+  if (norm && suml > 0) nf = log2(1.0 + suml) / suml;
+  for (int i = threadIdx.x; i < cnt; i += blockDim.x) {
+    out_g[start + i] = static_cast<float>(s_g[i] * nf);
+    out_h[start + i] = static_cast<float>(s_h[i] * nf);
+  }
+}
+
 __global__ void k_score_add_const(double* score, int n, double v) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) score[i] += v;
@@ -977,7 +1086,9 @@ class HIPTreeLearner : public TreeLearner {
 
   bool IsHIPLearner() const override { return true; }
   bool DeviceObjectiveSupported(const std::string& name) const override {
-    return name == "binary" || name == "regression";
+    if (name == "binary" || name == "regression") return true;
+    if (name == "lambdarank") return rank_ok_;
+    return false;
   }
 
   void Init(const Dataset* train_data, bool is_constant_hessian) override;
@@ -1039,6 +1150,11 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<float> d_grad_, d_hess_;
   DevBuf<int32_t> d_grad_packed_;
   DevBuf<float> d_grad_absmax_, d_grad_scales_;
+  // lambdarank (device ranking objective)
+  bool rank_ok_ = false;
+  DevBuf<int> d_qb_;
+  DevBuf<double> d_inv_max_dcg_, d_label_gain_;
+  int num_queries_ = 0;
   DevBuf<double> d_score_;
   DevBuf<float> d_label_, d_weight_;
   DevBuf<uint32_t> d_idx_, d_idx_tmp_;
@@ -1227,6 +1343,49 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   leaf_begin_.resize(nl);
   leaf_cnt_.resize(nl);
   approx_cnt_.resize(nl);
+
+  // ranking metadata (device lambdarank gradients)
+  rank_ok_ = false;
+  const data_size_t* qb = train_data->metadata().query_boundaries();
+  if (qb != nullptr) {
+    num_queries_ = train_data->metadata().num_queries();
+    int max_docs = 0;
+    for (int q = 0; q < num_queries_; ++q)
+      max_docs = std::max<int>(max_docs, qb[q + 1] - qb[q]);
+    if (max_docs <= 1024) {
+      rank_ok_ = true;
+      d_qb_.Alloc(num_queries_ + 1);
+      HIP_OK(hipMemcpy(d_qb_.ptr, qb, sizeof(int) * (num_queries_ + 1),
+                       hipMemcpyHostToDevice));
+      // label-gain table + per-query inverse max DCG (labels are static)
+      std::vector<double> lg = config_->label_gain;
+      if (lg.empty())
+        for (int i = 0; i < 31; ++i) lg.push_back((1u << i) - 1.0);
+      d_label_gain_.Alloc(lg.size());
+      HIP_OK(hipMemcpy(d_label_gain_.ptr, lg.data(), sizeof(double) * lg.size(),
+                       hipMemcpyHostToDevice));
+      const label_t* lab = train_data->metadata().label();
+      std::vector<double> imd(num_queries_);
+      const int trunc = config_->lambdarank_truncation_level;
+#pragma omp parallel for schedule(static)
+      for (int q = 0; q < num_queries_; ++q) {
+        std::vector<double> gains;
+        for (data_size_t i = qb[q]; i < qb[q + 1]; ++i)
+          gains.push_back(lg[static_cast<int>(lab[i])]);
+        std::sort(gains.begin(), gains.end(), std::greater<double>());
+        double dcg = 0;
+        const int k = std::min<int>(trunc, static_cast<int>(gains.size()));
+        for (int i = 0; i < k; ++i) dcg += gains[i] / std::log2(2.0 + i);
+        imd[q] = dcg > 0 ? 1.0 / dcg : 0.0;
+      }
+      d_inv_max_dcg_.Alloc(num_queries_);
+      HIP_OK(hipMemcpy(d_inv_max_dcg_.ptr, imd.data(), sizeof(double) * num_queries_,
+                       hipMemcpyHostToDevice));
+    } else {
+      Log::Warning("lambdarank on GPU supports queries up to 1024 docs; the largest "
+                   "query has %d -> gradients fall back to the host", max_docs);
+    }
+  }
   Log::Info("HIP tree learner: %d rows, %d features, %d bins, %zu LDS partitions, "
             "%d hist copies",
             num_data_, nf_, total_bins_, feat_partitions_.size(), n_copies_);
@@ -1260,6 +1419,12 @@ void HIPTreeLearner::DeviceBoosting(const ObjectiveFunction* obj) {
     hipLaunchKernelGGL(hipk::k_grad_l2, g, b, 0, stream_, d_score_.ptr, d_label_.ptr,
                        weights_present_ ? d_weight_.ptr : nullptr, n, d_grad_.ptr,
                        d_hess_.ptr);
+  } else if (name == "lambdarank") {
+    hipLaunchKernelGGL(hipk::k_grad_lambdarank, dim3(num_queries_), dim3(256), 0, stream_,
+                       d_score_.ptr, d_label_.ptr, d_qb_.ptr, num_queries_,
+                       d_inv_max_dcg_.ptr, d_label_gain_.ptr, config_->sigmoid,
+                       config_->lambdarank_truncation_level,
+                       config_->lambdarank_norm ? 1 : 0, d_grad_.ptr, d_hess_.ptr);
   } else {
     Log::Fatal("DeviceBoosting called for unsupported objective %s", name.c_str());
   }

@@ -110,3 +110,24 @@ def test_gpu_quantized_grad():
         bst = lgb.train(params, lgb.Dataset(X, label=y), 30)
         aucs[q] = _auc(yv, bst.predict(Xv))
     assert abs(aucs[True] - aucs[False]) < 5e-3, aucs
+
+
+def test_gpu_lambdarank():
+    """Device lambdarank gradients: quality parity with the CPU objective."""
+    rng = np.random.RandomState(9)
+    groups = rng.randint(20, 120, size=400)
+    n = int(groups.sum())
+    X = rng.randn(n, 20).astype(np.float32)
+    rel = np.clip((X[:, 0] * 1.5 + 0.5 * rng.randn(n) + 1.2), 0, 4).astype(int)
+    y = rel.astype(np.float32)
+    ndcgs = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "lambdarank", "metric": "ndcg", "eval_at": [10],
+                  "device_type": dev, "num_leaves": 31, "verbosity": 0}
+        ev = {}
+        train = lgb.Dataset(X, label=y, group=groups.astype(np.int32))
+        lgb.train(params, train, 30, valid_sets=[train], valid_names=["t"],
+                  callbacks=[lgb.record_evaluation(ev)])
+        ndcgs[dev] = ev["t"]["ndcg@10"][-1]
+    assert ndcgs["gpu"] > 0.85, ndcgs
+    assert abs(ndcgs["gpu"] - ndcgs["cpu"]) < 0.02, ndcgs
