@@ -59,7 +59,7 @@ std::string GBDT::SaveModelToString(int start_iter, int num_iter,
   }
   ss << "tree_sizes=" << Common::Join(tree_sizes, " ") << '\n';
   ss << '\n';
-  for (auto& s : tree_strs) ss << s << '\n';
+  for (auto& blk : tree_strs) ss << blk;  // sizes index into this byte stream exactly
   ss << "end of trees" << '\n';
   ss << '\n';
 

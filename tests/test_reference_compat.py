@@ -1,0 +1,113 @@
+"""Cross-ecosystem model-format compatibility: models trained by migbm must load and
+predict identically in the REFERENCE LightGBM, and vice versa (the SURVEY phase-1 exit
+criterion). The reference oracle (built by tools/build_reference_oracle.sh) runs in a
+SUBPROCESS — both libraries export the LGBM_* symbols, so they cannot share a process.
+Skipped when /root/reference/lib_lightgbm.so is absent."""
+import json
+import subprocess
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+import lightgbm_amd as lgb
+
+REF_LIB = Path("/root/reference/lib_lightgbm.so")
+pytestmark = pytest.mark.skipif(not REF_LIB.exists(),
+                                reason="reference oracle not built "
+                                       "(tools/build_reference_oracle.sh)")
+
+_REF_WORKER = r"""
+import ctypes, json, sys
+import numpy as np
+lib = ctypes.cdll.LoadLibrary("/root/reference/lib_lightgbm.so")
+lib.LGBM_GetLastError.restype = ctypes.c_char_p
+def ok(ret):
+    assert ret == 0, lib.LGBM_GetLastError().decode()
+cmd = json.loads(sys.argv[1])
+X = np.load(cmd["x"])
+if cmd["op"] == "predict":
+    h = ctypes.c_void_p(); it = ctypes.c_int(0)
+    ok(lib.LGBM_BoosterCreateFromModelfile(cmd["model"].encode(), ctypes.byref(it), ctypes.byref(h)))
+    arr = np.ascontiguousarray(X, dtype=np.float64)
+    out = np.zeros(len(X), dtype=np.float64); n = ctypes.c_int64(0)
+    ok(lib.LGBM_BoosterPredictForMat(h, arr.ctypes.data_as(ctypes.c_void_p), 1,
+        ctypes.c_int32(arr.shape[0]), ctypes.c_int32(arr.shape[1]), 1, 0, 0, -1, b"",
+        ctypes.byref(n), out.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+    np.save(cmd["out"], out)
+elif cmd["op"] == "train":
+    y = np.load(cmd["y"])
+    arr = np.ascontiguousarray(X, dtype=np.float64)
+    ds = ctypes.c_void_p()
+    ok(lib.LGBM_DatasetCreateFromMat(arr.ctypes.data_as(ctypes.c_void_p), 1,
+        ctypes.c_int32(arr.shape[0]), ctypes.c_int32(arr.shape[1]), 1,
+        cmd["ds_params"].encode(), None, ctypes.byref(ds)))
+    lab = np.ascontiguousarray(y, dtype=np.float32)
+    ok(lib.LGBM_DatasetSetField(ds, b"label", lab.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_int(len(lab)), 0))
+    bst = ctypes.c_void_p()
+    ok(lib.LGBM_BoosterCreate(ds, cmd["params"].encode(), ctypes.byref(bst)))
+    fin = ctypes.c_int(0)
+    for _ in range(cmd["iters"]):
+        ok(lib.LGBM_BoosterUpdateOneIter(bst, ctypes.byref(fin)))
+    ok(lib.LGBM_BoosterSaveModel(bst, 0, -1, 0, cmd["model"].encode()))
+print("REF_OK")
+"""
+
+
+def _ref(cmd, tmp_path):
+    worker = tmp_path / "ref_worker.py"
+    worker.write_text(_REF_WORKER)
+    r = subprocess.run([sys.executable, str(worker), json.dumps(cmd)],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0 and "REF_OK" in r.stdout, r.stdout + r.stderr
+
+
+def _ref_predict(model_file, X, tmp_path):
+    np.save(tmp_path / "x.npy", X)
+    _ref({"op": "predict", "model": str(model_file), "x": str(tmp_path / "x.npy"),
+          "out": str(tmp_path / "out.npy")}, tmp_path)
+    return np.load(tmp_path / "out.npy")
+
+
+def _data(n=4000, d=8, seed=0):
+    rng = np.random.RandomState(seed)
+    X = rng.randn(n, d)
+    y = ((X[:, 0] - 0.5 * X[:, 1] + X[:, 2] * X[:, 3] + 0.4 * rng.randn(n)) > 0)
+    return X, y.astype(np.float32)
+
+
+def test_our_model_loads_in_reference(tmp_path):
+    X, y = _data()
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "num_leaves": 31},
+                    lgb.Dataset(X, label=y), 20)
+    ours = bst.predict(X[:500])
+    f = tmp_path / "migbm_model.txt"
+    bst.save_model(str(f))
+    theirs = _ref_predict(f, X[:500], tmp_path)
+    np.testing.assert_allclose(ours, theirs, rtol=1e-9, atol=1e-12)
+
+
+def test_reference_model_loads_in_ours(tmp_path):
+    X, y = _data(seed=3)
+    np.save(tmp_path / "x.npy", X)
+    np.save(tmp_path / "y.npy", y)
+    f = tmp_path / "reference_model.txt"
+    _ref({"op": "train", "x": str(tmp_path / "x.npy"), "y": str(tmp_path / "y.npy"),
+          "model": str(f), "ds_params": "max_bin=255",
+          "params": "objective=binary verbosity=-1 num_leaves=31", "iters": 20}, tmp_path)
+    theirs = _ref_predict(f, X[:500], tmp_path)
+    mine = lgb.Booster(model_file=str(f)).predict(X[:500])
+    np.testing.assert_allclose(mine, theirs, rtol=1e-9, atol=1e-12)
+
+
+def test_regression_model_cross(tmp_path):
+    rng = np.random.RandomState(5)
+    X = rng.randn(3000, 6)
+    y = (2 * X[:, 0] + np.sin(X[:, 1])).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "verbosity": -1}, lgb.Dataset(X, label=y), 15)
+    f = tmp_path / "reg.txt"
+    bst.save_model(str(f))
+    np.testing.assert_allclose(bst.predict(X[:300]), _ref_predict(f, X[:300], tmp_path),
+                               rtol=1e-9, atol=1e-12)
