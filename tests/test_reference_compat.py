@@ -111,3 +111,72 @@ def test_regression_model_cross(tmp_path):
     bst.save_model(str(f))
     np.testing.assert_allclose(bst.predict(X[:300]), _ref_predict(f, X[:300], tmp_path),
                                rtol=1e-9, atol=1e-12)
+
+
+def test_categorical_model_cross(tmp_path):
+    rng = np.random.RandomState(7)
+    n = 4000
+    cat = rng.randint(0, 10, size=n)
+    X = np.column_stack([cat.astype(float), rng.randn(n), rng.randn(n)])
+    effect = rng.randn(10) * 2
+    y = (effect[cat] + 0.3 * rng.randn(n) > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5},
+                    lgb.Dataset(X, label=y, categorical_feature=[0]), 15)
+    f = tmp_path / "cat.txt"
+    bst.save_model(str(f))
+    theirs = _ref_predict(f, X[:500], tmp_path)
+    np.testing.assert_allclose(bst.predict(X[:500]), theirs, rtol=1e-9, atol=1e-12)
+
+
+def test_multiclass_model_cross(tmp_path):
+    rng = np.random.RandomState(8)
+    X = rng.randn(3000, 5)
+    y = ((X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)).astype(np.float32)
+    bst = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 10)
+    f = tmp_path / "mc.txt"
+    bst.save_model(str(f))
+    # reference PredictForMat on multiclass returns nrow x nclass row-major
+    import ctypes  # noqa: F401  (worker handles shape via out size)
+    np.save(tmp_path / "x.npy", X[:200])
+    worker = tmp_path / "ref_worker_mc.py"
+    worker.write_text(_REF_WORKER.replace(
+        "out = np.zeros(len(X), dtype=np.float64)",
+        "out = np.zeros(len(X) * 3, dtype=np.float64)"))
+    import subprocess as sp
+    r = sp.run([sys.executable, str(worker), json.dumps(
+        {"op": "predict", "model": str(f), "x": str(tmp_path / "x.npy"),
+         "out": str(tmp_path / "out.npy")})], capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    theirs = np.load(tmp_path / "out.npy").reshape(200, 3)
+    np.testing.assert_allclose(bst.predict(X[:200]), theirs, rtol=1e-9, atol=1e-12)
+
+
+def test_training_quality_parity_with_reference(tmp_path):
+    """Same data, same params: our CPU learner and the reference must reach the same
+    quality (not identical trees — binning differs — but equal AUC to ~1e-2)."""
+    X, y = _data(n=20000, seed=11)
+    params = dict(objective="binary", num_leaves=31, learning_rate=0.1, verbosity=-1)
+    bst = lgb.train(dict(params), lgb.Dataset(X[:16000], label=y[:16000]), 60)
+    ours_pred = bst.predict(X[16000:])
+    np.save(tmp_path / "x.npy", X[:16000])
+    np.save(tmp_path / "y.npy", y[:16000])
+    f = tmp_path / "ref_model.txt"
+    _ref({"op": "train", "x": str(tmp_path / "x.npy"), "y": str(tmp_path / "y.npy"),
+          "model": str(f), "ds_params": "max_bin=255",
+          "params": "objective=binary verbosity=-1 num_leaves=31 learning_rate=0.1",
+          "iters": 60}, tmp_path)
+    ref_pred = _ref_predict(f, X[16000:], tmp_path)
+
+    def auc(yy, pp):
+        order = np.argsort(-pp, kind="stable")
+        ys = yy[order]
+        npos = ys.sum()
+        nneg = len(ys) - npos
+        ranks = np.arange(1, len(ys) + 1)
+        return 1.0 - (ranks[ys > 0].sum() - npos * (npos + 1) / 2) / (npos * nneg)
+
+    a_ours = auc(y[16000:], ours_pred)
+    a_ref = auc(y[16000:], ref_pred)
+    assert abs(a_ours - a_ref) < 0.01, (a_ours, a_ref)
+    assert a_ours > 0.9
