@@ -117,8 +117,21 @@ class Dataset {
   int FeatureNumBin(int inner) const { return bin_mappers_[inner]->num_bin(); }
   uint32_t hist_offset(int inner) const { return hist_offsets_[inner]; }
   int num_total_bin() const { return num_total_bin_; }
-  uint32_t GetBin(data_size_t row, int inner) const { return columns_[inner].Get(row); }
-  const BinColumn& column(int inner) const { return columns_[inner]; }
+  /*! feature-level bin for a row, decoding EFB bundle columns when present */
+  inline uint32_t GetBin(data_size_t row, int inner) const {
+    const uint32_t v = columns_[col_of_feature_[inner]].Get(row);
+    if (!feature_bundled_[inner]) return v;
+    const uint32_t off = off_in_col_[inner];
+    const uint32_t span = static_cast<uint32_t>(bin_mappers_[inner]->num_bin()) - 1;
+    return (v >= off && v < off + span) ? v - off + 1 : 0;
+  }
+  int num_columns() const { return static_cast<int>(columns_.size()); }
+  int feature_column(int inner) const { return col_of_feature_[inner]; }
+  bool feature_bundled(int inner) const { return feature_bundled_[inner] != 0; }
+  bool has_bundles() const { return has_bundles_; }
+  /*! features stored in column c (one entry for unbundled columns) */
+  const std::vector<int>& column_features(int c) const { return column_features_[c]; }
+  const BinColumn& column(int c) const { return columns_[c]; }
 
   Metadata& metadata() { return metadata_; }
   const Metadata& metadata() const { return metadata_; }
@@ -179,13 +192,31 @@ class Dataset {
  private:
   friend class DatasetLoader;
   void FinishBinMappers(const Config& cfg);
+  /*! EFB: greedily bundle sparse mutually-exclusive features into shared columns.
+   *  Fills col_of_feature_/off_in_col_/column_features_ (identity when disabled).
+   *  Parity: reference Dataset::FindGroups/FastFeatureBundling (re-derived). */
+  void PlanBundles(const Config& cfg, const std::function<double(data_size_t, int)>& get,
+                   const std::vector<data_size_t>& sample_idx);
+  void CopyBundlingFrom(const Dataset& other) {
+    col_of_feature_ = other.col_of_feature_;
+    off_in_col_ = other.off_in_col_;
+    feature_bundled_ = other.feature_bundled_;
+    column_features_ = other.column_features_;
+    has_bundles_ = other.has_bundles_;
+  }
 
   data_size_t num_data_ = 0;
   int num_total_features_ = 0;
   std::vector<int> used_feature_map_;        // orig -> inner (-1 trivial)
   std::vector<int> real_feature_index_;      // inner -> orig
   std::vector<std::unique_ptr<BinMapper>> bin_mappers_;  // per inner feature
+  // storage columns: one per unbundled feature + one per EFB bundle
   std::vector<BinColumn> columns_;
+  std::vector<int> col_of_feature_;          // inner feature -> column
+  std::vector<uint32_t> off_in_col_;         // bundle offset (bundled features)
+  std::vector<int8_t> feature_bundled_;
+  std::vector<std::vector<int>> column_features_;
+  bool has_bundles_ = false;
   std::vector<uint32_t> hist_offsets_;
   int num_total_bin_ = 0;
   Metadata metadata_;

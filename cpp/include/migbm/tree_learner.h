@@ -120,8 +120,28 @@ class SerialTreeLearner : public TreeLearner {
  protected:
   /*! Histogram for `leaf` into its slot; optionally by subtraction (parent - sibling). */
   void ComputeHistogram(int leaf, data_size_t cnt, const data_size_t* indices);
-  /*! hook after a leaf histogram is built (data-parallel: global allreduce). */
-  virtual void OnHistogramReady(int leaf) { (void)leaf; }
+  /*! hook after a leaf histogram is built. Base: reconstruct the shared default bin
+   *  of EFB-bundled features from the leaf totals (the EFB form of FixHistogram).
+   *  Distributed learners reduce first, then call this base. */
+  virtual void OnHistogramReady(int leaf) { MaterializeBundledBin0(leaf); }
+  void MaterializeBundledBin0(int leaf) {
+    if (!train_data_->has_bundles()) return;
+    hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
+    const LeafContext& ctx = leaf_ctx_[leaf];
+    const int nf = train_data_->num_features();
+    for (int f = 0; f < nf; ++f) {
+      if (!train_data_->feature_bundled(f)) continue;
+      hist_t* fh = hist + 2 * train_data_->hist_offset(f);
+      double g = 0, h = 0;
+      const int nb = train_data_->FeatureNumBin(f);
+      for (int b = 1; b < nb; ++b) {
+        g += fh[2 * b];
+        h += fh[2 * b + 1];
+      }
+      fh[0] = ctx.sum_gradient - g;
+      fh[1] = ctx.sum_hessian - h;
+    }
+  }
   /*! hook to globalize root stats (data-parallel: allreduce). */
   virtual void ReduceRootStats(double* sum_g, double* sum_h, data_size_t* cnt) {
     (void)sum_g; (void)sum_h; (void)cnt;

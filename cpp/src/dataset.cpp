@@ -115,15 +115,36 @@ void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& ge
   if (bin_mappers_.empty())
     Log::Warning("All features are trivial (constant); no informative splits possible");
   FinishBinMappers(cfg);
-  // 4. bin all values
-  columns_.resize(bin_mappers_.size());
+  // 4. plan EFB bundles, then bin all values into columns
+  PlanBundles(cfg, get, sample_idx);
   const int nf = static_cast<int>(bin_mappers_.size());
+  const int ncols = static_cast<int>(column_features_.size());
+  columns_.resize(ncols);
 #pragma omp parallel for schedule(dynamic, 1)
-  for (int f = 0; f < nf; ++f) {
-    columns_[f].Init(nrow, bin_mappers_[f]->num_bin());
-    const int c = real_feature_index_[f];
-    const BinMapper* m = bin_mappers_[f].get();
-    for (data_size_t i = 0; i < nrow; ++i) columns_[f].Set(i, m->ValueToBin(get(i, c)));
+  for (int col = 0; col < ncols; ++col) {
+    const auto& members = column_features_[col];
+    if (members.size() == 1 && !feature_bundled_[members[0]]) {
+      const int f = members[0];
+      columns_[col].Init(nrow, bin_mappers_[f]->num_bin());
+      const int c = real_feature_index_[f];
+      const BinMapper* m = bin_mappers_[f].get();
+      for (data_size_t i = 0; i < nrow; ++i) columns_[col].Set(i, m->ValueToBin(get(i, c)));
+    } else {
+      // bundle column: value 0 = every member at its default(zero) bin;
+      // member j's non-default bin b (>=1) stored as off_j + b - 1
+      int col_bins = 1;
+      for (int f : members) col_bins += bin_mappers_[f]->num_bin() - 1;
+      columns_[col].Init(nrow, col_bins);
+      for (int f : members) {
+        const int c = real_feature_index_[f];
+        const BinMapper* m = bin_mappers_[f].get();
+        const uint32_t off = off_in_col_[f];
+        for (data_size_t i = 0; i < nrow; ++i) {
+          const uint32_t b = m->ValueToBin(get(i, c));
+          if (b > 0) columns_[col].Set(i, off + b - 1);  // conflicts: last writer wins
+        }
+      }
+    }
   }
   if (cfg.linear_tree) {
     raw_values_.resize(nf);
@@ -154,17 +175,143 @@ std::unique_ptr<Dataset> Dataset::CreateValid(
     d->bin_mappers_[i] = std::make_unique<BinMapper>(*bin_mappers_[i]);
   d->hist_offsets_ = hist_offsets_;
   d->num_total_bin_ = num_total_bin_;
-  const int nf = static_cast<int>(bin_mappers_.size());
-  d->columns_.resize(nf);
+  d->CopyBundlingFrom(*this);
+  const int ncols = num_columns();
+  d->columns_.resize(ncols);
 #pragma omp parallel for schedule(dynamic, 1)
-  for (int f = 0; f < nf; ++f) {
-    d->columns_[f].Init(nrow, d->bin_mappers_[f]->num_bin());
-    const int c = real_feature_index_[f];
-    const BinMapper* m = d->bin_mappers_[f].get();
-    for (data_size_t i = 0; i < nrow; ++i) d->columns_[f].Set(i, m->ValueToBin(get(i, c)));
+  for (int col = 0; col < ncols; ++col) {
+    const auto& members = column_features_[col];
+    if (members.size() == 1 && !feature_bundled_[members[0]]) {
+      const int f = members[0];
+      d->columns_[col].Init(nrow, d->bin_mappers_[f]->num_bin());
+      const int c = real_feature_index_[f];
+      const BinMapper* m = d->bin_mappers_[f].get();
+      for (data_size_t i = 0; i < nrow; ++i)
+        d->columns_[col].Set(i, m->ValueToBin(get(i, c)));
+    } else {
+      int col_bins = 1;
+      for (int f : members) col_bins += bin_mappers_[f]->num_bin() - 1;
+      d->columns_[col].Init(nrow, col_bins);
+      for (int f : members) {
+        const int c = real_feature_index_[f];
+        const BinMapper* m = d->bin_mappers_[f].get();
+        const uint32_t off = off_in_col_[f];
+        for (data_size_t i = 0; i < nrow; ++i) {
+          const uint32_t b = m->ValueToBin(get(i, c));
+          if (b > 0) d->columns_[col].Set(i, off + b - 1);
+        }
+      }
+    }
   }
   d->metadata_.Init(nrow, false, false);
   return d;
+}
+
+void Dataset::PlanBundles(const Config& cfg,
+                          const std::function<double(data_size_t, int)>& get,
+                          const std::vector<data_size_t>& sample_idx) {
+  const int nf = static_cast<int>(bin_mappers_.size());
+  col_of_feature_.assign(nf, -1);
+  off_in_col_.assign(nf, 0);
+  feature_bundled_.assign(nf, 0);
+  column_features_.clear();
+  has_bundles_ = false;
+  // candidates: sparse (>=80% zeros), non-negative (zero maps to bin 0), no NaN bin
+  std::vector<int> cand;
+  if (cfg.enable_bundle) {
+    for (int f = 0; f < nf; ++f) {
+      const BinMapper* m = bin_mappers_[f].get();
+      if (m->bin_type() == BinType::kNumerical && m->missing_type() == MissingType::kNone &&
+          m->ValueToBin(0.0) == 0 && m->sparse_rate() >= 0.8 && m->num_bin() >= 2) {
+        cand.push_back(f);
+      }
+    }
+  }
+  if (static_cast<int>(cand.size()) >= 2) {
+    // nonzero bitmap per candidate over the binning sample
+    const int ns = static_cast<int>(sample_idx.size());
+    const int words = (ns + 63) / 64;
+    std::vector<std::vector<uint64_t>> nz(cand.size(), std::vector<uint64_t>(words, 0));
+    std::vector<int> nz_cnt(cand.size(), 0);
+#pragma omp parallel for schedule(dynamic, 1)
+    for (size_t k = 0; k < cand.size(); ++k) {
+      const int c = real_feature_index_[cand[k]];
+      for (int i = 0; i < ns; ++i) {
+        const double v = get(sample_idx[i], c);
+        if (v != 0.0 && !std::isnan(v)) {
+          nz[k][i >> 6] |= 1ull << (i & 63);
+          nz_cnt[k]++;
+        }
+      }
+    }
+    // greedy conflict-bounded bundling (densest candidates first)
+    std::vector<size_t> order(cand.size());
+    for (size_t k = 0; k < order.size(); ++k) order[k] = k;
+    std::sort(order.begin(), order.end(),
+              [&](size_t a, size_t b) { return nz_cnt[a] > nz_cnt[b]; });
+    const int max_conflict = static_cast<int>(cfg.max_conflict_rate * ns);
+    struct Bundle {
+      std::vector<size_t> members;
+      std::vector<uint64_t> occupied;
+      int total_bins = 1;
+      int conflicts = 0;
+    };
+    std::vector<Bundle> bundles;
+    for (size_t k : order) {
+      const int fbins = bin_mappers_[cand[k]]->num_bin() - 1;
+      bool placed = false;
+      for (auto& b : bundles) {
+        if (b.total_bins + fbins > 255) continue;
+        int conf = 0;
+        for (int w = 0; w < words; ++w)
+          conf += __builtin_popcountll(b.occupied[w] & nz[k][w]);
+        if (b.conflicts + conf <= max_conflict) {
+          for (int w = 0; w < words; ++w) b.occupied[w] |= nz[k][w];
+          b.members.push_back(k);
+          b.total_bins += fbins;
+          b.conflicts += conf;
+          placed = true;
+          break;
+        }
+      }
+      if (!placed) {
+        Bundle b;
+        b.members = {k};
+        b.occupied = nz[k];
+        b.total_bins = 1 + fbins;
+        bundles.push_back(std::move(b));
+      }
+    }
+    // realize bundles with >=2 members
+    for (auto& b : bundles) {
+      if (b.members.size() < 2) continue;
+      const int col = static_cast<int>(column_features_.size());
+      std::vector<int> feats;
+      uint32_t off = 1;
+      for (size_t k : b.members) {
+        const int f = cand[k];
+        feats.push_back(f);
+        col_of_feature_[f] = col;
+        off_in_col_[f] = off;
+        feature_bundled_[f] = 1;
+        off += bin_mappers_[f]->num_bin() - 1;
+      }
+      column_features_.push_back(std::move(feats));
+      has_bundles_ = true;
+    }
+    if (has_bundles_)
+      Log::Info("EFB: bundled %d sparse features into %zu columns",
+                static_cast<int>(std::count(feature_bundled_.begin(),
+                                            feature_bundled_.end(), 1)),
+                column_features_.size());
+  }
+  // unbundled features get their own columns
+  for (int f = 0; f < nf; ++f) {
+    if (col_of_feature_[f] < 0) {
+      col_of_feature_[f] = static_cast<int>(column_features_.size());
+      column_features_.push_back({f});
+    }
+  }
 }
 
 namespace {
@@ -202,7 +349,16 @@ void HistInner(const BIN_T* bins, const data_size_t* idx, data_size_t n,
 void Dataset::ConstructHistogramForFeature(int f, const data_size_t* data_indices,
                                            data_size_t num_data, const score_t* og,
                                            const score_t* oh, hist_t* out) const {
-  const BinColumn& col = columns_[f];
+  if (feature_bundled_[f]) {
+    for (data_size_t i = 0; i < num_data; ++i) {
+      const data_size_t r = data_indices ? data_indices[i] : i;
+      const uint32_t b = GetBin(r, f);
+      out[2 * b] += og[i];
+      out[2 * b + 1] += oh[i];
+    }
+    return;
+  }
+  const BinColumn& col = columns_[col_of_feature_[f]];
   const bool use_idx = data_indices != nullptr;
   if (col.is16()) {
     if (use_idx) HistInner<uint16_t, true>(col.data16(), data_indices, num_data, og, oh, out);
@@ -217,13 +373,43 @@ void Dataset::ConstructHistograms(const std::vector<int8_t>& is_feature_used,
                                   const data_size_t* data_indices, data_size_t num_data,
                                   const score_t* og, const score_t* oh, hist_t* hist) const {
   if (num_data <= 0) return;
-  const int nf = num_features();
+  const int ncols = num_columns();
 #pragma omp parallel for schedule(dynamic)
-  for (int f = 0; f < nf; ++f) {
-    if (!is_feature_used[f]) continue;
-    hist_t* out = hist + 2 * hist_offsets_[f];
-    std::fill(out, out + 2 * bin_mappers_[f]->num_bin(), 0.0);
-    ConstructHistogramForFeature(f, data_indices, num_data, og, oh, out);
+  for (int col = 0; col < ncols; ++col) {
+    const auto& members = column_features_[col];
+    bool any_used = false;
+    for (int f : members) any_used |= is_feature_used[f] != 0;
+    if (!any_used) continue;
+    for (int f : members)
+      std::fill(hist + 2 * hist_offsets_[f],
+                hist + 2 * (hist_offsets_[f] + bin_mappers_[f]->num_bin()), 0.0);
+    if (members.size() == 1 && !feature_bundled_[members[0]]) {
+      const int f = members[0];
+      ConstructHistogramForFeature(f, data_indices, num_data, og, oh,
+                                   hist + 2 * hist_offsets_[f]);
+    } else {
+      // bundle column: one pass fills every member's non-default bins; the shared
+      // default (value 0) is reconstructed from leaf totals in the learner
+      // (SerialTreeLearner::OnHistogramReady), the EFB analogue of FixHistogram.
+      int col_bins = 1;
+      for (int f : members) col_bins += bin_mappers_[f]->num_bin() - 1;
+      std::vector<uint32_t> map(col_bins, UINT32_MAX);
+      for (int f : members) {
+        const uint32_t off = off_in_col_[f];
+        for (int b = 1; b < bin_mappers_[f]->num_bin(); ++b)
+          map[off + b - 1] = hist_offsets_[f] + b;
+      }
+      const BinColumn& bc = columns_[col];
+      for (data_size_t i = 0; i < num_data; ++i) {
+        const data_size_t r = data_indices ? data_indices[i] : i;
+        const uint32_t v = bc.Get(r);
+        if (v == 0) continue;
+        const uint32_t hb = map[v];
+        if (hb == UINT32_MAX) continue;
+        hist[2 * hb] += og[i];
+        hist[2 * hb + 1] += oh[i];
+      }
+    }
   }
 }
 
@@ -242,7 +428,7 @@ const Dataset::RowMajorView& Dataset::GetRowMajorView() const {
   if (row_view_built_) return row_view_;
   const int nf = num_features();
   bool any16 = false;
-  for (int f = 0; f < nf; ++f) any16 |= columns_[f].is16();
+  for (int f = 0; f < nf; ++f) any16 |= bin_mappers_[f]->num_bin() > 256;
   // pad row stride to 16 elements for aligned vector loads on device
   int stride = (nf + 15) & ~15;
   row_view_.row_stride = stride;
@@ -252,14 +438,14 @@ const Dataset::RowMajorView& Dataset::GetRowMajorView() const {
 #pragma omp parallel for schedule(static)
     for (data_size_t i = 0; i < num_data_; ++i) {
       uint8_t* row = row_view_.data.data() + static_cast<size_t>(i) * stride;
-      for (int f = 0; f < nf; ++f) row[f] = static_cast<uint8_t>(columns_[f].Get(i));
+      for (int f = 0; f < nf; ++f) row[f] = static_cast<uint8_t>(GetBin(i, f));
     }
   } else {
     row_view_.data16.assign(static_cast<size_t>(num_data_) * stride, 0);
 #pragma omp parallel for schedule(static)
     for (data_size_t i = 0; i < num_data_; ++i) {
       uint16_t* row = row_view_.data16.data() + static_cast<size_t>(i) * stride;
-      for (int f = 0; f < nf; ++f) row[f] = static_cast<uint16_t>(columns_[f].Get(i));
+      for (int f = 0; f < nf; ++f) row[f] = static_cast<uint16_t>(GetBin(i, f));
     }
   }
   row_view_built_ = true;
@@ -278,12 +464,13 @@ std::unique_ptr<Dataset> Dataset::Subset(const data_size_t* indices, data_size_t
     d->bin_mappers_[i] = std::make_unique<BinMapper>(*bin_mappers_[i]);
   d->hist_offsets_ = hist_offsets_;
   d->num_total_bin_ = num_total_bin_;
-  const int nf = num_features();
-  d->columns_.resize(nf);
+  d->CopyBundlingFrom(*this);
+  const int ncols = num_columns();
+  d->columns_.resize(ncols);
 #pragma omp parallel for schedule(dynamic, 1)
-  for (int f = 0; f < nf; ++f) {
-    d->columns_[f].Init(n, d->bin_mappers_[f]->num_bin());
-    for (data_size_t i = 0; i < n; ++i) d->columns_[f].Set(i, columns_[f].Get(indices[i]));
+  for (int col = 0; col < ncols; ++col) {
+    d->columns_[col].Init(n, columns_[col].is16() ? 65536 : 256);
+    for (data_size_t i = 0; i < n; ++i) d->columns_[col].Set(i, columns_[col].Get(indices[i]));
   }
   // metadata subset
   d->metadata_.Init(n, false, false);
@@ -330,6 +517,17 @@ void Dataset::ConstructFromSampleData(double** sample_values, int** sample_indic
     }
   }
   FinishBinMappers(cfg);
+  {
+    const int nfX = static_cast<int>(bin_mappers_.size());
+    col_of_feature_.resize(nfX);
+    off_in_col_.assign(nfX, 0);
+    feature_bundled_.assign(nfX, 0);
+    column_features_.clear();
+    for (int f = 0; f < nfX; ++f) {
+      col_of_feature_[f] = f;
+      column_features_.push_back({f});
+    }
+  }
   columns_.resize(bin_mappers_.size());
   for (size_t f = 0; f < bin_mappers_.size(); ++f)
     columns_[f].Init(num_local_row, bin_mappers_[f]->num_bin());
@@ -350,9 +548,10 @@ std::unique_ptr<Dataset> Dataset::CreateByReference(data_size_t num_rows) const 
     d->bin_mappers_[i] = std::make_unique<BinMapper>(*bin_mappers_[i]);
   d->hist_offsets_ = hist_offsets_;
   d->num_total_bin_ = num_total_bin_;
-  d->columns_.resize(bin_mappers_.size());
-  for (size_t f = 0; f < bin_mappers_.size(); ++f)
-    d->columns_[f].Init(num_rows, d->bin_mappers_[f]->num_bin());
+  d->CopyBundlingFrom(*this);
+  d->columns_.resize(columns_.size());
+  for (size_t c = 0; c < columns_.size(); ++c)
+    d->columns_[c].Init(num_rows, columns_[c].is16() ? 65536 : 256);
   d->metadata_.Init(num_rows, false, false);
   return d;
 }
@@ -363,6 +562,13 @@ std::string Dataset::SerializeReference() const {
   ss << Common::Join(used_feature_map_, " ") << "\n";
   ss << Common::Join(real_feature_index_, " ") << "\n";
   ss << Common::Join(feature_names_, "\t") << "\n";
+  ss << Common::Join(col_of_feature_, " ") << "\n";
+  ss << Common::Join(off_in_col_, " ") << "\n";
+  {
+    std::vector<int> fb(feature_bundled_.begin(), feature_bundled_.end());
+    ss << Common::Join(fb, " ") << "\n";
+  }
+  ss << columns_.size() << "\n";
   for (auto& m : bin_mappers_) ss << m->ToString();
   return ss.str();
 }
@@ -378,9 +584,17 @@ std::unique_ptr<Dataset> Dataset::FromSerializedReference(const char* buf, size_
   Common::StringToArray<int>(lines[1], ' ', &d->used_feature_map_);
   Common::StringToArray<int>(lines[2], ' ', &d->real_feature_index_);
   d->feature_names_ = Common::Split(lines[3].c_str(), '\t');
+  Common::StringToArray<int>(lines[4], ' ', &d->col_of_feature_);
+  Common::StringToArray<uint32_t>(lines[5], ' ', &d->off_in_col_);
+  {
+    std::vector<int> fb;
+    Common::StringToArray<int>(lines[6], ' ', &fb);
+    d->feature_bundled_.assign(fb.begin(), fb.end());
+  }
+  const int ref_ncols = atoi(lines[7].c_str());
   const int nf = static_cast<int>(d->real_feature_index_.size());
   d->bin_mappers_.resize(nf);
-  size_t lp = 4;
+  size_t lp = 8;
   for (int f = 0; f < nf; ++f) {
     std::string blob = lines[lp] + "\n" + lines[lp + 1] + "\n" + lines[lp + 2];
     d->bin_mappers_[f] = std::make_unique<BinMapper>();
@@ -389,16 +603,34 @@ std::unique_ptr<Dataset> Dataset::FromSerializedReference(const char* buf, size_
   }
   Config dummy;
   d->FinishBinMappers(dummy);
-  d->columns_.resize(nf);
-  for (int f = 0; f < nf; ++f) d->columns_[f].Init(num_rows, d->bin_mappers_[f]->num_bin());
+  d->column_features_.assign(ref_ncols, {});
+  for (int f = 0; f < nf; ++f) d->column_features_[d->col_of_feature_[f]].push_back(f);
+  for (int8_t fb : d->feature_bundled_) d->has_bundles_ |= fb != 0;
+  d->columns_.resize(ref_ncols);
+  for (int c = 0; c < ref_ncols; ++c) {
+    int col_bins = 0;
+    for (int f : d->column_features_[c]) {
+      if (d->feature_bundled_[f]) col_bins += d->bin_mappers_[f]->num_bin() - 1;
+      else col_bins = std::max(col_bins, d->bin_mappers_[f]->num_bin());
+    }
+    if (!d->column_features_[c].empty() && d->feature_bundled_[d->column_features_[c][0]])
+      col_bins += 1;  // shared default bin
+    d->columns_[c].Init(num_rows, col_bins);
+  }
   d->metadata_.Init(num_rows, false, false);
   return d;
 }
 
 void Dataset::PushRawRow(data_size_t row, const double* values, int ncol) {
   for (int c = 0; c < std::min(ncol, num_total_features_); ++c) {
-    int f = used_feature_map_[c];
-    if (f >= 0) columns_[f].Set(row, bin_mappers_[f]->ValueToBin(values[c]));
+    const int f = used_feature_map_[c];
+    if (f < 0) continue;
+    const uint32_t b = bin_mappers_[f]->ValueToBin(values[c]);
+    if (!feature_bundled_[f]) {
+      columns_[col_of_feature_[f]].Set(row, b);
+    } else if (b > 0) {
+      columns_[col_of_feature_[f]].Set(row, off_in_col_[f] + b - 1);
+    }
   }
 }
 
@@ -414,6 +646,13 @@ void Dataset::SaveBinaryFile(const char* filename) const {
   ss << Common::Join(used_feature_map_, " ") << "\n";
   ss << Common::Join(real_feature_index_, " ") << "\n";
   ss << Common::Join(feature_names_, "\t") << "\n";
+  ss << Common::Join(col_of_feature_, " ") << "\n";
+  ss << Common::Join(off_in_col_, " ") << "\n";
+  {
+    std::vector<int> fb(feature_bundled_.begin(), feature_bundled_.end());
+    ss << Common::Join(fb, " ") << "\n";
+  }
+  ss << columns_.size() << "\n";
   for (auto& m : bin_mappers_) ss << m->ToString();
   std::string header = ss.str();
   uint64_t hlen = header.size();
@@ -468,9 +707,17 @@ std::unique_ptr<Dataset> Dataset::LoadFromBinFile(const char* filename) {
   Common::StringToArray<int>(lines[1], ' ', &d->used_feature_map_);
   Common::StringToArray<int>(lines[2], ' ', &d->real_feature_index_);
   d->feature_names_ = Common::Split(lines[3].c_str(), '\t');
+  Common::StringToArray<int>(lines[4], ' ', &d->col_of_feature_);
+  Common::StringToArray<uint32_t>(lines[5], ' ', &d->off_in_col_);
+  {
+    std::vector<int> fb;
+    Common::StringToArray<int>(lines[6], ' ', &fb);
+    d->feature_bundled_.assign(fb.begin(), fb.end());
+  }
+  const int n_cols_stored = atoi(lines[7].c_str());
   int nf = static_cast<int>(d->real_feature_index_.size());
   d->bin_mappers_.resize(nf);
-  size_t line_pos = 4;
+  size_t line_pos = 8;
   for (int f = 0; f < nf; ++f) {
     std::string blob = lines[line_pos] + "\n" + lines[line_pos + 1] + "\n" + lines[line_pos + 2];
     d->bin_mappers_[f] = std::make_unique<BinMapper>();
@@ -479,8 +726,11 @@ std::unique_ptr<Dataset> Dataset::LoadFromBinFile(const char* filename) {
   }
   Config dummy;
   d->FinishBinMappers(dummy);
-  d->columns_.resize(nf);
-  for (int f = 0; f < nf; ++f) {
+  d->column_features_.assign(n_cols_stored, {});
+  for (int f = 0; f < nf; ++f) d->column_features_[d->col_of_feature_[f]].push_back(f);
+  for (int8_t fb : d->feature_bundled_) d->has_bundles_ |= fb != 0;
+  d->columns_.resize(n_cols_stored);
+  for (int f = 0; f < n_cols_stored; ++f) {
     uint8_t is16;
     MIGBM_CHECK_EQ(fread(&is16, 1, 1, fp), 1u);
     d->columns_[f].Init(d->num_data_, is16 ? 65536 : 256);

@@ -1280,9 +1280,23 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_rows_.Alloc(view.data.size());
   HIP_OK(hipMemcpy(d_rows_.ptr, view.data.data(), view.data.size(), hipMemcpyHostToDevice));
   d_cols_.Alloc(static_cast<size_t>(nf_) * num_data_);
-  for (int f = 0; f < nf_; ++f) {
-    HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_,
-                     train_data->column(f).data8(), num_data_, hipMemcpyHostToDevice));
+  {
+    // per-feature dense bins on device (EFB bundles are decoded here: the GPU keeps
+    // feature-major layouts; 288GB HBM makes the unbundled copy cheap)
+    std::vector<uint8_t> colbuf(num_data_);
+    for (int f = 0; f < nf_; ++f) {
+      if (train_data->feature_bundled(f)) {
+#pragma omp parallel for schedule(static)
+        for (int i = 0; i < num_data_; ++i)
+          colbuf[i] = static_cast<uint8_t>(train_data->GetBin(i, f));
+        HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_, colbuf.data(),
+                         num_data_, hipMemcpyHostToDevice));
+      } else {
+        HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_,
+                         train_data->column(train_data->feature_column(f)).data8(),
+                         num_data_, hipMemcpyHostToDevice));
+      }
+    }
   }
 
   d_grad_.Alloc(num_data_);

@@ -462,7 +462,10 @@ Metric* Metric::Create(const std::string& name, const Config& cfg) {
     };
     return new R2Metric();
   }
-  if (name == "binary_logloss" || name == "logloss") return new BinaryLoglossMetric();
+  if (name == "binary_logloss" || name == "logloss" || name == "binary")
+    return new BinaryLoglossMetric();
+  if (name == "regression_l1")
+    return new PointwiseMetric("l1", true, [](double y, double p) { return std::fabs(y - p); });
   if (name == "binary_error")
     return new PointwiseMetric("binary_error", true, [](double y, double p) {
       return (p > 0.5 ? 1.0 : 0.0) != (y > 0 ? 1.0 : 0.0) ? 1.0 : 0.0;

@@ -100,9 +100,11 @@ class DataParallelTreeLearner : public SerialTreeLearner {
 
  protected:
   void OnHistogramReady(int leaf) override {
-    if (!Network::is_distributed()) return;
-    hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
-    Network::AllreduceSum(hist, 2 * static_cast<size_t>(train_data_->num_total_bin()));
+    if (Network::is_distributed()) {
+      hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
+      Network::AllreduceSum(hist, 2 * static_cast<size_t>(train_data_->num_total_bin()));
+    }
+    SerialTreeLearner::OnHistogramReady(leaf);  // EFB default-bin reconstruction
   }
   void ReduceRootStats(double* sum_g, double* sum_h, data_size_t* cnt) override {
     if (!Network::is_distributed()) return;
@@ -235,6 +237,7 @@ class VotingParallelTreeLearner : public SerialTreeLearner {
                 compact.begin() + off + 2 * train_data_->FeatureNumBin(f), fh);
       off += 2 * train_data_->FeatureNumBin(f);
     }
+    SerialTreeLearner::OnHistogramReady(leaf);  // EFB default-bin reconstruction
   }
 
   void FindBestSplitForLeaf(int leaf, const LeafContext& ctx) override {

@@ -227,14 +227,13 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
 
 std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& s) const {
   const int f = s.feature;
-  const BinColumn& col = train_data_->column(f);
+  const Dataset* data = train_data_;
   const BinMapper* m = train_data_->FeatureBinMapper(f);
   if (!s.cat_bitset_inner.empty()) {
-    // categorical
     std::vector<uint32_t> bits = s.cat_bitset_inner;
     const int nwords = static_cast<int>(bits.size());
-    return [&col, bits, nwords](data_size_t row) {
-      uint32_t b = col.Get(row);
+    return [data, f, bits, nwords](data_size_t row) {
+      uint32_t b = data->GetBin(row, f);
       return (b >> 5) < static_cast<uint32_t>(nwords) && ((bits[b >> 5] >> (b & 31)) & 1);
     };
   }
@@ -242,11 +241,11 @@ std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& 
   const int nanb = m->nan_bin();
   const bool default_left = s.default_left;
   if (nanb < 0) {
-    return [&col, thr](data_size_t row) { return col.Get(row) <= thr; };
+    return [data, f, thr](data_size_t row) { return data->GetBin(row, f) <= thr; };
   }
   const uint32_t nb = static_cast<uint32_t>(nanb);
-  return [&col, thr, nb, default_left](data_size_t row) {
-    uint32_t b = col.Get(row);
+  return [data, f, thr, nb, default_left](data_size_t row) {
+    uint32_t b = data->GetBin(row, f);
     if (b == nb) return default_left;
     return b <= thr;
   };

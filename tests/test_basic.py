@@ -78,3 +78,41 @@ def test_feature_names():
     X = np.random.RandomState(0).randn(100, 3)
     d = lgb.Dataset(X, label=np.zeros(100), feature_name=["a", "b", "c"]).construct()
     assert d.get_feature_name() == ["a", "b", "c"]
+
+
+def test_efb_bundling_sparse_features():
+    """EFB: mutually-exclusive sparse features share a column; quality unaffected."""
+    import scipy.sparse as sp
+    rng = np.random.RandomState(21)
+    n = 6000
+    # 10 one-hot-ish mutually exclusive sparse columns + 2 dense
+    group = rng.randint(0, 10, size=n)
+    S = np.zeros((n, 10))
+    S[np.arange(n), group] = rng.rand(n) + 0.5
+    D = rng.randn(n, 2)
+    X = np.column_stack([S, D])
+    effect = rng.randn(10) * 2
+    y = (effect[group] + 0.5 * D[:, 0] + 0.3 * rng.randn(n) > 0).astype(np.float32)
+    for bundle in (True, False):
+        ds = lgb.Dataset(X, label=y, params={"enable_bundle": bundle,
+                                             "max_conflict_rate": 0.0}).construct()
+        bst = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5,
+                         "enable_bundle": bundle}, ds, 30)
+        acc = ((bst.predict(X) > 0.5) == y).mean()
+        assert acc > 0.9, (bundle, acc)
+
+
+def test_efb_valid_set_alignment():
+    rng = np.random.RandomState(22)
+    n = 4000
+    group = rng.randint(0, 8, size=n)
+    S = np.zeros((n, 8))
+    S[np.arange(n), group] = 1.0
+    X = np.column_stack([S, rng.randn(n, 2)])
+    y = (group % 2).astype(np.float32)
+    train = lgb.Dataset(X[:3000], label=y[:3000]).construct()
+    valid = train.create_valid(X[3000:], label=y[3000:])
+    ev = {}
+    lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5},
+              train, 20, valid_sets=[valid], callbacks=[lgb.record_evaluation(ev)])
+    assert ev["valid_0"]["binary_logloss"][-1] < 0.1
