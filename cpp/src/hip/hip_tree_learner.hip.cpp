@@ -1080,7 +1080,6 @@ class HIPTreeLearner : public TreeLearner {
  public:
   explicit HIPTreeLearner(const Config* config) : config_(config) {}
   ~HIPTreeLearner() override {
-    InvalidateGraph();
     if (h_winner_) hipHostFree(h_winner_);
     if (stream_) hipStreamDestroy(stream_);
   }
@@ -1096,10 +1095,7 @@ class HIPTreeLearner : public TreeLearner {
   void ResetTrainingData(const Dataset* train_data) override {
     Init(train_data, is_constant_hessian_);
   }
-  void ResetConfig(const Config* config) override {
-    config_ = config;
-    InvalidateGraph();
-  }
+  void ResetConfig(const Config* config) override { config_ = config; }
   Tree* Train(const score_t* gradients, const score_t* hessians, bool is_first) override;
   void SetBaggingData(const Dataset* subset, const data_size_t* used_indices,
                       data_size_t num_data) override {
@@ -1127,13 +1123,6 @@ class HIPTreeLearner : public TreeLearner {
   void LaunchHist(const int* leafA_ptr, int leafB_from_counters, int blocks);
   void ReduceSpareHist(int spare_slot);
   void LaunchBestSplit(const int* leafA_ptr, int leafB_from_counters);
-  void LaunchSplitLoop(int nl);
-  void InvalidateGraph() {
-    if (graph_exec_) {
-      hipGraphExecDestroy(graph_exec_);
-      graph_exec_ = nullptr;
-    }
-  }
   int HistBlocksFor(int approx_cnt) const {
     // LDS atomic throughput is per-CU: spread even small leaves over many blocks
     // (~256 rows each); cap so the per-block flush stays amortized at the root.
@@ -1145,7 +1134,6 @@ class HIPTreeLearner : public TreeLearner {
   const Dataset* train_data_ = nullptr;
   bool is_constant_hessian_ = false;
   hipStream_t stream_ = nullptr;
-  hipGraphExec_t graph_exec_ = nullptr;
 
   int num_data_ = 0;
   int nf_ = 0;
@@ -1223,7 +1211,6 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   total_bins_ = train_data->num_total_bin();
   feature_rng_ = Random(config_->feature_fraction_seed);
 
-  InvalidateGraph();
   if (!stream_) HIP_OK(hipStreamCreate(&stream_));
   if (!h_winner_) {
     HIP_OK(hipHostMalloc(reinterpret_cast<void**>(&h_winner_),
@@ -1558,51 +1545,6 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                      leafB_from_counters);
 }
 
-void HIPTreeLearner::LaunchSplitLoop(int nl) {
-  auto& comm = GpuComm::Get();
-  const int kPartBlocks = 256;
-  const int kLoopHistBlocks = 768;
-  for (int split_i = 0; split_i < nl - 1; ++split_i) {
-    hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
-                       d_leaf_best_.ptr, d_counters_.ptr, d_winner_.ptr,
-                       d_winner_leaf_.ptr);
-    hipLaunchKernelGGL(hipk::k_part_mark, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
-                       d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_winner_leaf_.ptr,
-                       d_winner_.ptr, d_feat_meta_.ptr, d_cols_.ptr, num_data_,
-                       d_marks_.ptr, d_block_cnt_.ptr);
-    hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(256), 0, stream_, d_block_cnt_.ptr,
-                       kPartBlocks, d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_block_loff_.ptr,
-                       d_block_roff_.ptr, d_ctr_.ptr);
-    hipLaunchKernelGGL(hipk::k_part_scatter, dim3(kPartBlocks), dim3(kHistBlock), 0,
-                       stream_, d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr,
-                       d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_marks_.ptr,
-                       d_block_loff_.ptr, d_block_roff_.ptr, d_ctr_.ptr);
-    hipLaunchKernelGGL(hipk::k_copy_back, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
-                       d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
-                       d_winner_leaf_.ptr);
-    hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
-                       d_winner_leaf_.ptr, d_gbuf_.ptr);
-    if (comm.active()) {
-      NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
-                            stream_));
-    }
-    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
-                       d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
-                       d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
-                       d_gbuf_.ptr);
-    LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks);
-    ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
-    {
-      const int n_elem = total_bins_ * 2;
-      hipLaunchKernelGGL(hipk::k_hist_subtract, dim3((n_elem + 1023) / 1024), dim3(256), 0,
-                         stream_, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
-                         d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_leaf_.ptr,
-                         d_counters_.ptr, n_elem);
-    }
-    LaunchBestSplit(d_winner_leaf_.ptr, 1);
-  }
-}
-
 Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, bool) {
   const int nl = config_->num_leaves;
   auto tree = std::make_unique<Tree>(nl);
@@ -1667,22 +1609,47 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
 
   // ---- device-driven split loop: ZERO host syncs; winner decisions accumulate in
   // d_split_log_ and the tree is replayed on the host after one end-of-tree download.
-  // Single-GPU: the whole loop (11 kernels x num_leaves-1 splits, all device-side
-  // arguments) is captured into ONE hipGraph on the first tree and replayed per tree.
-  if (!comm.active()) {
-    if (graph_exec_ == nullptr) {
-      HIP_OK(hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal));
-      LaunchSplitLoop(nl);
-      hipGraph_t g = nullptr;
-      HIP_OK(hipStreamEndCapture(stream_, &g));
-      HIP_OK(hipGraphInstantiate(&graph_exec_, g, nullptr, nullptr, 0));
-      HIP_OK(hipGraphDestroy(g));
+  const int kPartBlocks = 256;
+  const int kLoopHistBlocks = 768;
+  for (int split_i = 0; split_i < nl - 1; ++split_i) {
+    hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
+                       d_leaf_best_.ptr, d_counters_.ptr, d_winner_.ptr,
+                       d_winner_leaf_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_mark, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
+                       d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_winner_leaf_.ptr,
+                       d_winner_.ptr, d_feat_meta_.ptr, d_cols_.ptr, num_data_,
+                       d_marks_.ptr, d_block_cnt_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(256), 0, stream_, d_block_cnt_.ptr,
+                       kPartBlocks, d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_block_loff_.ptr,
+                       d_block_roff_.ptr, d_ctr_.ptr);
+    hipLaunchKernelGGL(hipk::k_part_scatter, dim3(kPartBlocks), dim3(kHistBlock), 0,
+                       stream_, d_idx_.ptr, d_idx_tmp_.ptr, d_leaf_begin_.ptr,
+                       d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_marks_.ptr,
+                       d_block_loff_.ptr, d_block_roff_.ptr, d_ctr_.ptr);
+    hipLaunchKernelGGL(hipk::k_copy_back, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
+                       d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
+                       d_winner_leaf_.ptr);
+    hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
+                       d_winner_leaf_.ptr, d_gbuf_.ptr);
+    if (comm.active()) {
+      NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
+                            stream_));
     }
-    HIP_OK(hipGraphLaunch(graph_exec_, stream_));
-  } else {
-    LaunchSplitLoop(nl);
+    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
+                       d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
+                       d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
+                       d_gbuf_.ptr);
+    LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks);
+    ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
+    {
+      const int n_elem = total_bins_ * 2;
+      hipLaunchKernelGGL(hipk::k_hist_subtract, dim3((n_elem + 1023) / 1024), dim3(256), 0,
+                         stream_, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
+                         d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_leaf_.ptr,
+                         d_counters_.ptr, n_elem);
+    }
+    LaunchBestSplit(d_winner_leaf_.ptr, 1);
   }
-
 
   // ---- one download: split log + exact leaf layout; replay the tree on the host
   host_log_.resize(nl);
