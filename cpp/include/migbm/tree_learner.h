@@ -180,6 +180,16 @@ class SerialTreeLearner : public TreeLearner {
   Random extra_rng_{0};
   int iter_counter_ = 0;
   bool build_both_children_ = false;  // voting-parallel: no histogram subtraction
+  // forced splits (forcedsplits_filename JSON)
+  struct ForcedNode {
+    int feature = -1;
+    double threshold = 0.0;
+    std::unique_ptr<ForcedNode> left, right;
+  };
+  std::unique_ptr<ForcedNode> forced_root_;
+  std::vector<const ForcedNode*> forced_of_leaf_;
+  bool MakeForcedSplit(int leaf, const LeafContext& ctx, const ForcedNode* node,
+                       SplitInfo* out);
   // CEGB / interaction-constraint state
   std::vector<int8_t> cegb_feature_used_;
   std::vector<std::set<int>> leaf_branch_features_;

@@ -372,6 +372,72 @@ class MapMetric : public Metric {
   const data_size_t* query_boundaries_ = nullptr;
 };
 
+/*! auc_mu: multiclass AUC (average of one-vs-one partition AUCs; parity target:
+ *  reference multiclass_metric.hpp AucMuMetric). */
+class AucMuMetric : public Metric {
+ public:
+  explicit AucMuMetric(const Config& cfg) : names_({"auc_mu"}), nc_(cfg.num_class) {}
+  void Init(const Metadata& meta, data_size_t num_data) override {
+    num_data_ = num_data;
+    label_ = meta.label();
+    weights_ = meta.weights();
+  }
+  const std::vector<std::string>& GetName() const override { return names_; }
+  double factor_to_bigger_better() const override { return -1.0; }
+  std::vector<double> Eval(const double* score, const ObjectiveFunction*) const override {
+    double total = 0;
+    int pairs = 0;
+    for (int a = 0; a < nc_; ++a) {
+      for (int b = a + 1; b < nc_; ++b) {
+        // rows of classes a,b ranked by s_a - s_b; AUC of class a vs b
+        std::vector<std::pair<double, int>> v;
+        std::vector<double> w;
+        for (data_size_t i = 0; i < num_data_; ++i) {
+          const int c = static_cast<int>(label_[i]);
+          if (c != a && c != b) continue;
+          const double sa = score[static_cast<size_t>(a) * num_data_ + i];
+          const double sb = score[static_cast<size_t>(b) * num_data_ + i];
+          v.push_back({sa - sb, c == a ? 1 : 0});
+          w.push_back(weights_ ? weights_[i] : 1.0);
+        }
+        std::vector<size_t> order(v.size());
+        std::iota(order.begin(), order.end(), size_t{0});
+        std::sort(order.begin(), order.end(),
+                  [&](size_t x, size_t y) { return v[x].first > v[y].first; });
+        double acc_pos = 0, acc_neg = 0, cp = 0, cn = 0, auc = 0;
+        double prev = std::numeric_limits<double>::infinity();
+        for (size_t k = 0; k < order.size(); ++k) {
+          const auto& e = v[order[k]];
+          if (e.first != prev) {
+            auc += cn * (acc_pos + cp * 0.5);
+            acc_pos += cp;
+            acc_neg += cn;
+            cp = cn = 0;
+            prev = e.first;
+          }
+          if (e.second) cp += w[order[k]];
+          else cn += w[order[k]];
+        }
+        auc += cn * (acc_pos + cp * 0.5);
+        acc_pos += cp;
+        acc_neg += cn;
+        if (acc_pos > 0 && acc_neg > 0) {
+          total += auc / (acc_pos * acc_neg);
+          ++pairs;
+        }
+      }
+    }
+    return {pairs > 0 ? total / pairs : 1.0};
+  }
+
+ private:
+  std::vector<std::string> names_;
+  int nc_;
+  data_size_t num_data_ = 0;
+  const label_t* label_ = nullptr;
+  const label_t* weights_ = nullptr;
+};
+
 }  // namespace
 
 Metric* Metric::Create(const std::string& name, const Config& cfg) {
@@ -476,6 +542,7 @@ Metric* Metric::Create(const std::string& name, const Config& cfg) {
       name == "multiclassova")
     return new MultiLoglossMetric(cfg.num_class);
   if (name == "multi_error") return new MultiErrorMetric(cfg.num_class, cfg.multi_error_top_k);
+  if (name == "auc_mu") return new AucMuMetric(cfg);
   if (name == "ndcg" || name == "lambdarank" || name == "rank_xendcg") return new NDCGMetric(cfg);
   if (name == "map" || name == "mean_average_precision") return new MapMetric(cfg);
   if (name == "cross_entropy" || name == "xentropy")

@@ -581,6 +581,16 @@ class LambdarankNDCG : public ObjectiveFunction {
     weights_ = meta.weights();
     query_boundaries_ = meta.query_boundaries();
     num_queries_ = meta.num_queries();
+    positions_ = meta.positions();
+    if (positions_ != nullptr) {
+      int max_pos = 0;
+      for (data_size_t i = 0; i < num_data_; ++i)
+        max_pos = std::max(max_pos, positions_[i]);
+      position_bias_.assign(max_pos + 1, 1.0);
+      bias_num_.assign(max_pos + 1, 0.0);
+      bias_den_.assign(max_pos + 1, 0.0);
+      Log::Info("Lambdarank position debiasing enabled (%d positions)", max_pos + 1);
+    }
     if (query_boundaries_ == nullptr)
       Log::Fatal("Lambdarank requires query information (group)");
     // inverse max DCG per query
@@ -599,12 +609,30 @@ class LambdarankNDCG : public ObjectiveFunction {
     }
   }
   void GetGradients(const double* score, score_t* grad, score_t* hess) const override {
+    if (positions_ != nullptr) {
+      std::fill(bias_num_.begin(), bias_num_.end(), 0.0);
+      std::fill(bias_den_.begin(), bias_den_.end(), 0.0);
+    }
 #pragma omp parallel for schedule(guided)
     for (data_size_t q = 0; q < num_queries_; ++q) {
       GetGradientsForOneQuery(q, score, grad, hess);
     }
-    if (norm_) {
-      // per-query lambda normalization happens inside; nothing global
+    if (positions_ != nullptr) UpdatePositionBiasFactors();
+  }
+
+  /*! unbiased lambdarank: re-estimate per-position click propensity from the
+   *  accumulated lambda mass (parity: reference rank_objective.hpp position
+   *  debiasing — re-derived update rule). */
+  void UpdatePositionBiasFactors() const {
+    double base = bias_den_[0] > 0 ? bias_num_[0] / bias_den_[0] : 1.0;
+    if (base <= 0) return;
+    for (size_t k = 0; k < position_bias_.size(); ++k) {
+      if (bias_den_[k] > 0) {
+        const double est = (bias_num_[k] / bias_den_[k]) / base;
+        // smoothed multiplicative update, clamped for stability
+        position_bias_[k] = std::min(10.0, std::max(0.1, 0.9 * position_bias_[k] +
+                                                              0.1 * est));
+      }
     }
   }
   void GetGradientsForOneQuery(data_size_t q, const double* score_all, score_t* grad_all,
@@ -660,6 +688,22 @@ class LambdarankNDCG : public ObjectiveFunction {
         double p_hessian = p_lambda * (1.0 - p_lambda);
         p_lambda *= -sigmoid_ * delta_pair_ndcg;
         p_hessian *= sigmoid_ * sigmoid_ * delta_pair_ndcg;
+        if (positions_ != nullptr) {
+          // inverse-propensity weighting by the displayed positions
+          const int ph = positions_[s + high];
+          const int pl = positions_[s + low];
+          const double w_ipw = 1.0 / (position_bias_[ph] * position_bias_[pl]);
+          p_lambda *= w_ipw;
+          p_hessian *= w_ipw;
+#pragma omp atomic
+          bias_num_[ph] += std::fabs(p_lambda);
+#pragma omp atomic
+          bias_den_[ph] += 1.0;
+#pragma omp atomic
+          bias_num_[pl] += std::fabs(p_lambda);
+#pragma omp atomic
+          bias_den_[pl] += 1.0;
+        }
         grad[high] += static_cast<score_t>(p_lambda);
         hess[high] += static_cast<score_t>(p_hessian);
         grad[low] -= static_cast<score_t>(p_lambda);
@@ -694,6 +738,8 @@ class LambdarankNDCG : public ObjectiveFunction {
   int truncation_level_;
   std::vector<double> label_gain_;
   std::vector<double> inverse_max_dcg_;
+  const int32_t* positions_ = nullptr;
+  mutable std::vector<double> position_bias_, bias_num_, bias_den_;
 };
 
 class RankXENDCG : public ObjectiveFunction {

@@ -5,6 +5,7 @@
 #include "migbm/objective.h"
 
 #include <algorithm>
+#include <fstream>
 #include <numeric>
 #include <set>
 
@@ -69,6 +70,48 @@ void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian
   feature_rng_ = Random(config_->feature_fraction_seed);
   extra_rng_ = Random(config_->extra_seed);
   cegb_feature_used_.assign(train_data_->num_total_features(), 0);
+  forced_of_leaf_.assign(config_->num_leaves, nullptr);
+  forced_root_.reset();
+  if (!config_->forcedsplits_filename.empty()) {
+    std::ifstream jf(config_->forcedsplits_filename);
+    if (!jf.good()) {
+      Log::Warning("Cannot open forced splits file %s", config_->forcedsplits_filename.c_str());
+    } else {
+      std::string content((std::istreambuf_iterator<char>(jf)),
+                          std::istreambuf_iterator<char>());
+      size_t pos = 0;
+      std::function<std::unique_ptr<ForcedNode>()> parse = [&]() -> std::unique_ptr<ForcedNode> {
+        auto skip = [&]() { while (pos < content.size() && isspace(content[pos])) ++pos; };
+        skip();
+        if (pos >= content.size() || content[pos] != '{') return nullptr;
+        ++pos;
+        auto node = std::make_unique<ForcedNode>();
+        while (pos < content.size() && content[pos] != '}') {
+          skip();
+          if (content[pos] == ',') { ++pos; continue; }
+          if (content[pos] != '"') break;
+          size_t kend = content.find('"', pos + 1);
+          std::string key = content.substr(pos + 1, kend - pos - 1);
+          pos = content.find(':', kend) + 1;
+          skip();
+          if (key == "feature") node->feature = atoi(content.c_str() + pos);
+          else if (key == "threshold") node->threshold = atof(content.c_str() + pos);
+          if (key == "left") node->left = parse();
+          else if (key == "right") node->right = parse();
+          else {  // skip number token
+            while (pos < content.size() && content[pos] != ',' && content[pos] != '}') ++pos;
+            continue;
+          }
+          skip();
+        }
+        if (pos < content.size() && content[pos] == '}') ++pos;
+        return node;
+      };
+      forced_root_ = parse();
+      if (forced_root_) Log::Info("Loaded forced splits from %s",
+                                  config_->forcedsplits_filename.c_str());
+    }
+  }
   leaf_branch_features_.assign(config_->num_leaves, {});
   interaction_groups_.clear();
   if (!config_->interaction_constraints.empty()) {
@@ -225,6 +268,42 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
   }
 }
 
+bool SerialTreeLearner::MakeForcedSplit(int leaf, const LeafContext& ctx,
+                                        const ForcedNode* node, SplitInfo* out) {
+  const int inner = train_data_->InnerFeatureIndex(node->feature);
+  if (inner < 0) return false;
+  const BinMapper* m = train_data_->FeatureBinMapper(inner);
+  if (m->bin_type() != BinType::kNumerical) return false;
+  int bin = static_cast<int>(m->ValueToBin(node->threshold));
+  bin = std::min(bin, m->num_numeric_bin() - 2);
+  if (bin < 0) return false;
+  const hist_t* fh = HistSlot(leaf_to_slot_[leaf]) + 2 * train_data_->hist_offset(inner);
+  double gl = 0, hl = 0;
+  for (int b = 0; b <= bin; ++b) {
+    gl += fh[2 * b];
+    hl += fh[2 * b + 1];
+  }
+  out->Reset();
+  out->feature = inner;
+  out->threshold = static_cast<uint32_t>(bin);
+  out->default_left = false;
+  out->left_sum_gradient = gl;
+  out->left_sum_hessian = hl;
+  out->right_sum_gradient = ctx.sum_gradient - gl;
+  out->right_sum_hessian = ctx.sum_hessian - hl;
+  const double cf = ctx.num_data > 0 && ctx.sum_hessian > 0
+                        ? ctx.num_data / ctx.sum_hessian : 1.0;
+  out->left_count = static_cast<data_size_t>(Common::RoundInt(hl * cf));
+  out->right_count = ctx.num_data - out->left_count;
+  out->left_output = GainMath::CalculateSplittedLeafOutput(
+      gl, hl, config_->lambda_l1, config_->lambda_l2, config_->max_delta_step);
+  out->right_output = GainMath::CalculateSplittedLeafOutput(
+      out->right_sum_gradient, out->right_sum_hessian, config_->lambda_l1,
+      config_->lambda_l2, config_->max_delta_step);
+  out->gain = 1e30;  // forced splits take precedence over gain selection
+  return true;
+}
+
 std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& s) const {
   const int f = s.feature;
   const Dataset* data = train_data_;
@@ -282,14 +361,28 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
   FindBestSplitForLeaf(0, leaf_ctx_[0]);
 
   int num_leaves = 1;
+  std::fill(forced_of_leaf_.begin(), forced_of_leaf_.end(), nullptr);
+  forced_of_leaf_[0] = forced_root_.get();
   for (int split_i = 0; split_i < config_->num_leaves - 1; ++split_i) {
-    // pick best leaf
+    // forced splits first (in discovery order), then best-gain leaf
     int best_leaf = -1;
-    double best_gain = 0.0;
+    SplitInfo forced_split;
     for (int l = 0; l < num_leaves; ++l) {
-      if (best_split_per_leaf_[l].IsValid() && best_split_per_leaf_[l].gain > best_gain) {
-        best_gain = best_split_per_leaf_[l].gain;
+      if (forced_of_leaf_[l] != nullptr &&
+          MakeForcedSplit(l, leaf_ctx_[l], forced_of_leaf_[l], &forced_split)) {
         best_leaf = l;
+        best_split_per_leaf_[l] = forced_split;
+        break;
+      }
+      if (forced_of_leaf_[l] != nullptr) forced_of_leaf_[l] = nullptr;  // unusable
+    }
+    if (best_leaf < 0) {
+      double best_gain = 0.0;
+      for (int l = 0; l < num_leaves; ++l) {
+        if (best_split_per_leaf_[l].IsValid() && best_split_per_leaf_[l].gain > best_gain) {
+          best_gain = best_split_per_leaf_[l].gain;
+          best_leaf = l;
+        }
       }
     }
     if (best_leaf < 0) break;
@@ -335,6 +428,13 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                   train_data_->RealThreshold(f, s.threshold), s.left_output, s.right_output,
                   s.left_count, s.right_count, s.left_sum_hessian, s.right_sum_hessian,
                   static_cast<float>(s.gain), m->missing_type(), s.default_left);
+    }
+
+    // propagate forced-split children
+    {
+      const ForcedNode* fn = forced_of_leaf_[best_leaf];
+      forced_of_leaf_[best_leaf] = fn != nullptr ? fn->left.get() : nullptr;
+      forced_of_leaf_[right_leaf] = fn != nullptr ? fn->right.get() : nullptr;
     }
 
     // bookkeeping for CEGB / interaction constraints

@@ -341,3 +341,48 @@ def test_r2_metric():
               valid_sets=[train], valid_names=["train"],
               callbacks=[lgb.record_evaluation(ev)])
     assert ev["train"]["r2"][-1] > 0.9
+
+
+def test_forced_splits(tmp_path):
+    rng = np.random.RandomState(31)
+    X = rng.randn(3000, 4)
+    y = (X[:, 0] + 0.3 * X[:, 1] + 0.2 * rng.randn(3000)).astype(np.float32)
+    fs = tmp_path / "forced.json"
+    fs.write_text('{"feature": 3, "threshold": 0.0, '
+                  '"left": {"feature": 2, "threshold": 0.5}}')
+    bst = lgb.train({"objective": "regression", "verbosity": -1,
+                     "forcedsplits_filename": str(fs)}, lgb.Dataset(X, label=y), 5)
+    model = bst.dump_model()
+    for t in model["tree_info"]:
+        root = t["tree_structure"]
+        if "split_index" not in root:
+            continue
+        assert root["split_feature"] == 3
+        assert abs(root["threshold"] - 0.0) < 0.2
+        left = root["left_child"]
+        if "split_index" in left:
+            assert left["split_feature"] == 2
+
+
+def test_position_debias_runs():
+    rng = np.random.RandomState(32)
+    groups = [25] * 80
+    n = sum(groups)
+    X = rng.randn(n, 5)
+    y = np.clip((X[:, 0] + 0.5 * rng.randn(n) + 1), 0, 3).astype(int).astype(np.float32)
+    pos = np.concatenate([np.arange(g) for g in groups]).astype(np.int32)
+    ds = lgb.Dataset(X, label=y, group=np.array(groups, dtype=np.int32), position=pos)
+    bst = lgb.train({"objective": "lambdarank", "verbosity": -1}, ds, 10)
+    assert np.all(np.isfinite(bst.predict(X[:50])))
+
+
+def test_auc_mu():
+    rng = np.random.RandomState(33)
+    X = rng.randn(3000, 5)
+    y = ((X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)).astype(np.float32)
+    ev = {}
+    train = lgb.Dataset(X, label=y)
+    lgb.train({"objective": "multiclass", "num_class": 3, "metric": "auc_mu",
+               "verbosity": -1}, train, 20, valid_sets=[train], valid_names=["t"],
+              callbacks=[lgb.record_evaluation(ev)])
+    assert ev["t"]["auc_mu"][-1] > 0.9
