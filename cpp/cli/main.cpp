@@ -47,7 +47,8 @@ class Application {
   }
 
   void Run() {
-    if (config_.task == "train" || config_.task == "refit") Train();
+    if (config_.task == "train") Train();
+    else if (config_.task == "refit" || config_.task == "refit_tree") Refit();
     else if (config_.task == "predict" || config_.task == "prediction" ||
              config_.task == "test")
       Predict();
@@ -162,6 +163,35 @@ class Application {
     }
     fclose(fp);
     Log::Info("Predictions written to %s", config_.output_result.c_str());
+  }
+
+  void Refit() {
+    if (config_.input_model.empty()) Log::Fatal("task=refit requires input_model=");
+    LoadData();
+    boosting_.reset(GBDT::CreateBoosting("gbdt", config_.input_model.c_str()));
+    objective_.reset(ObjectiveFunction::Create(boosting_->ObjectiveName(), config_));
+    if (objective_) objective_->Init(train_data_->metadata(), train_data_->num_data());
+    std::vector<const Metric*> none;
+    boosting_->ResetTrainingData(train_data_.get(), objective_.get(), none);
+    // leaf assignments of the current model on the new data
+    const int ntrees = boosting_->NumberOfTotalModel();
+    const data_size_t n = train_data_->num_data();
+    std::vector<int32_t> leaf_preds(static_cast<size_t>(n) * ntrees);
+    const int ncol = boosting_->MaxFeatureIdx() + 1;
+    auto rows = LoadRawRowsForPredict(config_.data.c_str(), config_, ncol);
+#pragma omp parallel for schedule(static)
+    for (data_size_t i = 0; i < n; ++i) {
+      std::vector<double> feats(ncol, 0.0);
+      for (int c = 0; c < ncol && c < static_cast<int>(rows[i].size()); ++c)
+        feats[c] = rows[i][c];
+      std::vector<double> out(ntrees);
+      boosting_->PredictLeafIndex(feats.data(), out.data(), 0, -1);
+      for (int t = 0; t < ntrees; ++t)
+        leaf_preds[static_cast<size_t>(i) * ntrees + t] = static_cast<int32_t>(out[t]);
+    }
+    boosting_->RefitTree(leaf_preds.data(), n, ntrees);
+    boosting_->SaveModelToFile(0, -1, 0, config_.output_model.c_str());
+    Log::Info("Refitted model saved to %s", config_.output_model.c_str());
   }
 
   void ConvertModel() {
