@@ -121,6 +121,8 @@ def main():
 
     use_gpu = args.device == "gpu" and torch.cuda.is_available()
     if use_gpu:
+        # modulo clamp lets N ranks share fewer GPUs (testing on a 1-GPU box)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
 
     import lightgbm_amd as lgb
