@@ -146,7 +146,9 @@ def main():
                 payload = [None]
             dist.broadcast_object_list(payload, src=0)
             idb = payload[0]
-            assert _LIB.LGBM_GPUNetworkInit(ctypes.c_int(world), ctypes.c_int(rank), idb) == 0
+            rc = _LIB.LGBM_GPUNetworkInit(ctypes.c_int(world), ctypes.c_int(rank), idb)
+            if rc != 0:
+                raise RuntimeError(f"rank {rank}: RCCL communicator init failed (see stderr)")
 
     # ---- data: bin mappers must be identical on every rank -> all ranks build the
     # same reference dataset from a common-seed sample, then bin their own shard

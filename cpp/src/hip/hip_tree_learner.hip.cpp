@@ -1818,7 +1818,11 @@ extern "C" __attribute__((visibility("default"))) int LGBM_GPUNetworkInit(int wo
     c.rank = rank;
     migbm::Log::Info("RCCL communicator initialized: rank %d / %d", rank, world);
     return 0;
+  } catch (const std::exception& ex) {
+    fprintf(stderr, "LGBM_GPUNetworkInit failed: %s\n", ex.what());
+    return -1;
   } catch (...) {
+    fprintf(stderr, "LGBM_GPUNetworkInit failed: unknown error\n");
     return -1;
   }
 }
