@@ -875,18 +875,14 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
     if (threadIdx.x == 0) log[counters[1]].leaf = -1;  // terminator for the host replay
     return;
   }
-  // zero the spare histogram slot cooperatively (spare = current num_leaves).
-  // Every thread snapshots counters[0] BEFORE the barrier; thread 0 only mutates
-  // the counters after it, so the pointer arithmetic cannot race the increment.
-  const int spare_snapshot = counters[0];
+  // zero the spare histogram slot cooperatively (spare = current num_leaves)
   {
-    float* spare = hist_base + static_cast<size_t>(spare_snapshot) * slot_stride;
+    float* spare = hist_base + static_cast<size_t>(counters[0]) * slot_stride;
     for (int i = threadIdx.x; i < n_elem; i += blockDim.x) spare[i] = 0.0f;
   }
-  __syncthreads();
   if (threadIdx.x != 0) return;
   if (!use_gbuf) gbuf[0] = ctr[0];  // single-GPU: local count IS the global count
-  const int R = spare_snapshot;
+  const int R = counters[0];
   const int spare_slot = R;
   log[counters[1]].rec = *winner;
   log[counters[1]].leaf = L;
