@@ -540,63 +540,27 @@ __global__ void __launch_bounds__(64) k_best_feat(
   }
 }
 
-/*! single block: reduce per-feature records into leaf_best for both children, then
- *  argmax over all leaves -> the winner record for the NEXT split (fused former
- *  k_best_leaf + k_best_overall). */
+/*! reduce per-feature records to one per leaf; blockIdx.x = which child. */
 __global__ void k_best_leaf(const SplitRec* __restrict__ feat_best, int nf,
                             SplitRec* __restrict__ leaf_best,
                             const int* __restrict__ leafA_ptr,
-                            const int* __restrict__ counters, int leafB_from_counters,
-                            SplitRec* __restrict__ winner, int* __restrict__ winner_leaf) {
+                            const int* __restrict__ counters, int leafB_from_counters) {
+  const int which = blockIdx.x;
+  const int leafA = *leafA_ptr;
+  const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
+  const int leaf = which == 0 ? leafA : leafB;
+  if (leaf < 0) return;
+  const SplitRec* cand = feat_best + which * nf;
   __shared__ int s_idx[256];
   __shared__ double s_gain[256];
   const int tid = threadIdx.x;
-  const int leafA = *leafA_ptr;
-  const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
-  for (int which = 0; which < 2; ++which) {
-    const int leaf = which == 0 ? leafA : leafB;
-    if (leaf < 0) continue;
-    const SplitRec* cand = feat_best + which * nf;
-    int bi = -1;
-    double bg = -1e308;
-    for (int f = tid; f < nf; f += blockDim.x) {
-      if (cand[f].valid &&
-          (cand[f].gain > bg || (cand[f].gain == bg && (bi < 0 || f < bi)))) {
-        bg = cand[f].gain;
-        bi = f;
-      }
-    }
-    s_idx[tid] = bi;
-    s_gain[tid] = bg;
-    __syncthreads();
-    for (int s = blockDim.x / 2; s > 0; s >>= 1) {
-      if (tid < s) {
-        if (s_idx[tid + s] >= 0 &&
-            (s_idx[tid] < 0 || s_gain[tid + s] > s_gain[tid] ||
-             (s_gain[tid + s] == s_gain[tid] && s_idx[tid + s] < s_idx[tid]))) {
-          s_idx[tid] = s_idx[tid + s];
-          s_gain[tid] = s_gain[tid + s];
-        }
-      }
-      __syncthreads();
-    }
-    if (tid == 0) {
-      if (s_idx[0] >= 0) leaf_best[leaf] = cand[s_idx[0]];
-      else {
-        leaf_best[leaf].valid = 0;
-        leaf_best[leaf].gain = -1e308;
-      }
-    }
-    __syncthreads();
-  }
-  // overall argmax -> next split's winner
-  const int num_leaves = counters[0];
   int bi = -1;
-  double bg = 0.0;
-  for (int l = tid; l < num_leaves; l += blockDim.x) {
-    if (leaf_best[l].valid && leaf_best[l].gain > bg) {
-      bg = leaf_best[l].gain;
-      bi = l;
+  double bg = -1e308;
+  for (int f = tid; f < nf; f += blockDim.x) {
+    if (cand[f].valid &&
+        (cand[f].gain > bg || (cand[f].gain == bg && (bi < 0 || f < bi)))) {
+      bg = cand[f].gain;
+      bi = f;
     }
   }
   s_idx[tid] = bi;
@@ -614,8 +578,11 @@ __global__ void k_best_leaf(const SplitRec* __restrict__ feat_best, int nf,
     __syncthreads();
   }
   if (tid == 0) {
-    *winner_leaf = s_idx[0];
-    if (s_idx[0] >= 0) *winner = leaf_best[s_idx[0]];
+    if (s_idx[0] >= 0) leaf_best[leaf] = cand[s_idx[0]];
+    else {
+      leaf_best[leaf].valid = 0;
+      leaf_best[leaf].gain = -1e308;
+    }
   }
 }
 
@@ -860,28 +827,17 @@ __global__ void k_store_left(const int* __restrict__ ctr, const int* __restrict_
   gbuf[0] = ctr[0];
 }
 
-/*! device-side split bookkeeping: segments, stats, slot map, split log; the whole
- *  block also zeroes the spare histogram slot (fused former k_hist_zero). Thread 0
- *  additionally publishes the local left count for the RCCL path (fused k_store_left;
- *  pass use_gbuf=0 single-GPU and the count is read from ctr directly). */
+/*! device-side split bookkeeping: segments, stats, slot map, split log. One thread. */
 __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
                            const SplitRec* __restrict__ winner,
                            const int* __restrict__ Lptr, int* counters,
                            LogEntry* __restrict__ log, const int* __restrict__ ctr,
-                           int64_t* __restrict__ gbuf, int use_gbuf,
-                           float* __restrict__ hist_base, size_t slot_stride, int n_elem) {
+                           const int64_t* __restrict__ gbuf) {
   const int L = *Lptr;
   if (L < 0) {
-    if (threadIdx.x == 0) log[counters[1]].leaf = -1;  // terminator for the host replay
+    log[counters[1]].leaf = -1;  // terminator for the host replay
     return;
   }
-  // zero the spare histogram slot cooperatively (spare = current num_leaves)
-  {
-    float* spare = hist_base + static_cast<size_t>(counters[0]) * slot_stride;
-    for (int i = threadIdx.x; i < n_elem; i += blockDim.x) spare[i] = 0.0f;
-  }
-  if (threadIdx.x != 0) return;
-  if (!use_gbuf) gbuf[0] = ctr[0];  // single-GPU: local count IS the global count
   const int R = counters[0];
   const int spare_slot = R;
   log[counters[1]].rec = *winner;
@@ -1491,14 +1447,12 @@ void HIPTreeLearner::DeviceBoosting(const ObjectiveFunction* obj) {
 
 void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
                                 int blocks) {
+  const int n_elem = total_bins_ * 2;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
-  if (!leafB_from_counters) {
-    // root call: zero slot 0 here (loop iterations get their spare slot zeroed
-    // inside the fused finalize kernel)
-    const int n_elem = total_bins_ * 2;
-    hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
-                       slot_stride, d_counters_.ptr, 0, 0, leafA_ptr, n_elem);
-  }
+  // zero the spare slot (R = counters[0]-1 after finalize; literal slot 0 for the root)
+  hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
+                     slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
+                     n_elem);
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
     const auto [fb, fe] = feat_partitions_[pr];
     const auto [bin_base, bins] = part_bin_range_[pr];
@@ -1586,11 +1540,9 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                      d_leaf_stats_.ptr, leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
                      feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
                      d_feat_best_.ptr);
-  (void)ny;
-  hipLaunchKernelGGL(hipk::k_best_leaf, dim3(1), dim3(256), 0, stream_, d_feat_best_.ptr,
+  hipLaunchKernelGGL(hipk::k_best_leaf, dim3(ny), dim3(256), 0, stream_, d_feat_best_.ptr,
                      nf_, d_leaf_best_.ptr, leafA_ptr, d_counters_.ptr,
-                     leafB_from_counters, d_winner_.ptr,
-                     reinterpret_cast<int*>(d_winner_.ptr + 1));
+                     leafB_from_counters);
 }
 
 Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, bool) {
@@ -1659,8 +1611,10 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   // d_split_log_ and the tree is replayed on the host after one end-of-tree download.
   const int kPartBlocks = 256;
   const int kLoopHistBlocks = 768;
-  // the fused k_best_leaf of the previous split already produced d_winner_
   for (int split_i = 0; split_i < nl - 1; ++split_i) {
+    hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
+                       d_leaf_best_.ptr, d_counters_.ptr, d_winner_.ptr,
+                       d_winner_leaf_.ptr);
     hipLaunchKernelGGL(hipk::k_part_mark, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
                        d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_winner_leaf_.ptr,
                        d_winner_.ptr, d_feat_meta_.ptr, d_cols_.ptr, num_data_,
@@ -1675,17 +1629,16 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
     hipLaunchKernelGGL(hipk::k_copy_back, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
                        d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
                        d_winner_leaf_.ptr);
+    hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
+                       d_winner_leaf_.ptr, d_gbuf_.ptr);
     if (comm.active()) {
-      hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
-                         d_winner_leaf_.ptr, d_gbuf_.ptr);
       NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
                             stream_));
     }
-    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(256), 0, stream_, d_leaf_begin_.ptr,
+    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
                        d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
                        d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
-                       d_gbuf_.ptr, comm.active() ? 1 : 0, d_hist_.ptr,
-                       static_cast<size_t>(total_bins_) * 2, total_bins_ * 2);
+                       d_gbuf_.ptr);
     LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks);
     ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
     {
