@@ -14,7 +14,7 @@ import numpy as np
 from .compat import PANDAS_INSTALLED, SCIPY_INSTALLED, pd_DataFrame, pd_Series, scipy_sparse
 from .libpath import find_lib_path
 
-__all__ = ["Dataset", "Booster", "LightGBMError", "register_logger"]
+__all__ = ["Dataset", "Booster", "LightGBMError", "register_logger", "Sequence"]
 
 _DTYPE_F32, _DTYPE_F64, _DTYPE_I32, _DTYPE_I64 = 0, 1, 2, 3
 _PREDICT_NORMAL, _PREDICT_RAW, _PREDICT_LEAF, _PREDICT_CONTRIB = 0, 1, 2, 3
@@ -89,6 +89,80 @@ def _np_float32(data):
     if PANDAS_INSTALLED and isinstance(data, (pd_Series, pd_DataFrame)):
         data = data.values
     return np.ascontiguousarray(np.asarray(data).ravel(), dtype=np.float32)
+
+
+class Sequence:
+    """Generic random-access data source (parity: reference basic.py Sequence ABC).
+
+    Subclass and implement __getitem__ (row -> 1D numpy array) and __len__; pass
+    instances (or a list of them) as Dataset data. batch_size controls the chunk
+    size used while streaming rows into the dataset.
+    """
+
+    batch_size = 4096
+
+    def __getitem__(self, idx):
+        raise NotImplementedError("Sequence subclasses must implement __getitem__")
+
+    def __len__(self):
+        raise NotImplementedError("Sequence subclasses must implement __len__")
+
+
+def _is_sequence_input(data):
+    if isinstance(data, Sequence):
+        return True
+    return isinstance(data, list) and len(data) > 0 and all(
+        isinstance(s, Sequence) for s in data)
+
+
+def _create_dataset_from_seqs(seqs, param_str, ref_handle):
+    if isinstance(seqs, Sequence):
+        seqs = [seqs]
+    total = sum(len(s) for s in seqs)
+    first = np.asarray(seqs[0][0], dtype=np.float64).ravel()
+    ncol = len(first)
+    # sample rows for binning (bounded), then create + push in batches
+    sample_rows = []
+    stride = max(1, total // 200_000)
+    for s in seqs:
+        for i in range(0, len(s), stride):
+            sample_rows.append(np.asarray(s[i], dtype=np.float64).ravel())
+    sample = np.vstack(sample_rows)
+    out = ctypes.c_void_p()
+    if ref_handle is not None:
+        raise LightGBMError("Sequence input with a reference dataset is not supported yet")
+    # build mappers from the sample via the sampled-column API
+    col_ptrs = (ctypes.POINTER(ctypes.c_double) * ncol)()
+    idx_ptrs = (ctypes.POINTER(ctypes.c_int) * ncol)()
+    keep = []
+    nper = (ctypes.c_int * ncol)()
+    n_s = sample.shape[0]
+    for c in range(ncol):
+        vals = np.ascontiguousarray(sample[:, c])
+        idxs = np.arange(n_s, dtype=np.int32)
+        keep.append((vals, idxs))
+        col_ptrs[c] = vals.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+        idx_ptrs[c] = idxs.ctypes.data_as(ctypes.POINTER(ctypes.c_int))
+        nper[c] = n_s
+    _safe_call(_LIB.LGBM_DatasetCreateFromSampledColumn(
+        col_ptrs, idx_ptrs, ctypes.c_int32(ncol), nper, ctypes.c_int32(n_s),
+        ctypes.c_int32(total), ctypes.c_int64(total), _c_str(param_str),
+        ctypes.byref(out)))
+    # push rows in batches
+    start = 0
+    for s in seqs:
+        bs = getattr(s, "batch_size", 4096) or 4096
+        for b0 in range(0, len(s), bs):
+            rows = [np.asarray(s[i], dtype=np.float64).ravel()
+                    for i in range(b0, min(b0 + bs, len(s)))]
+            chunk = np.ascontiguousarray(np.vstack(rows))
+            _safe_call(_LIB.LGBM_DatasetPushRows(
+                out, chunk.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
+                ctypes.c_int32(chunk.shape[0]), ctypes.c_int32(ncol),
+                ctypes.c_int32(start)))
+            start += chunk.shape[0]
+    _safe_call(_LIB.LGBM_DatasetMarkFinished(out))
+    return out
 
 
 def _is_pyarrow_table(data):
@@ -175,6 +249,8 @@ class Dataset:
 
         if _is_pyarrow_table(self.data):
             self._handle = _create_dataset_from_arrow(self.data, param_str, ref_handle)
+        elif _is_sequence_input(self.data):
+            self._handle = _create_dataset_from_seqs(self.data, param_str, ref_handle)
         elif isinstance(self.data, (str, Path)):
             out = ctypes.c_void_p()
             _safe_call(_LIB.LGBM_DatasetCreateFromFile(
@@ -696,6 +772,39 @@ class Booster:
         if importance_type == "split":
             return res.astype(np.int64)
         return res
+
+    def trees_to_dataframe(self):
+        """Model structure as a pandas DataFrame (parity: reference Booster method)."""
+        from .compat import PANDAS_INSTALLED
+        if not PANDAS_INSTALLED:
+            raise ImportError("pandas is required for trees_to_dataframe")
+        import pandas as pd
+        model = self.dump_model()
+        rows = []
+
+        def walk(tree_index, node, parent=None):
+            if "leaf_index" in node:
+                rows.append(dict(tree_index=tree_index,
+                                 node_index=f"{tree_index}-L{node['leaf_index']}",
+                                 parent_index=parent, split_feature=None,
+                                 threshold=None, decision_type=None,
+                                 value=node["leaf_value"], count=node.get("leaf_count"),
+                                 weight=node.get("leaf_weight")))
+                return
+            ni = f"{tree_index}-S{node['split_index']}"
+            rows.append(dict(tree_index=tree_index, node_index=ni, parent_index=parent,
+                             split_feature=node["split_feature"],
+                             threshold=node["threshold"],
+                             decision_type=node["decision_type"],
+                             value=node.get("internal_value"),
+                             count=node.get("internal_count"),
+                             weight=node.get("internal_weight")))
+            walk(tree_index, node["left_child"], ni)
+            walk(tree_index, node["right_child"], ni)
+
+        for t in model["tree_info"]:
+            walk(t["tree_index"], t["tree_structure"])
+        return pd.DataFrame(rows)
 
     def free_dataset(self):
         # The native GBDT keeps raw pointers into the Dataset; keep the Python objects

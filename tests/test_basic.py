@@ -116,3 +116,33 @@ def test_efb_valid_set_alignment():
     lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5},
               train, 20, valid_sets=[valid], callbacks=[lgb.record_evaluation(ev)])
     assert ev["valid_0"]["binary_logloss"][-1] < 0.1
+
+
+def test_sequence_input():
+    class NpSeq(lgb.basic.Sequence):
+        def __init__(self, arr):
+            self.arr = arr
+
+        def __getitem__(self, i):
+            return self.arr[i]
+
+        def __len__(self):
+            return len(self.arr)
+
+    rng = np.random.RandomState(41)
+    X = rng.randn(3000, 5)
+    y = (X[:, 0] > 0).astype(np.float32)
+    ds = lgb.Dataset([NpSeq(X[:1500]), NpSeq(X[1500:])], label=y)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, ds, 10)
+    assert ((bst.predict(X) > 0.5) == y).mean() > 0.9
+
+
+def test_trees_to_dataframe():
+    rng = np.random.RandomState(42)
+    X = rng.randn(500, 3)
+    y = (X[:, 0] > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "num_leaves": 7},
+                    lgb.Dataset(X, label=y), 3)
+    df = bst.trees_to_dataframe()
+    assert set(df["tree_index"]) == {0, 1, 2}
+    assert "threshold" in df.columns
