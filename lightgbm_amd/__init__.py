@@ -4,7 +4,7 @@ Drop-in `import lightgbm_amd as lgb` replacement for the reference Python packag
 (python-package/lightgbm/__init__.py parity), backed by lib_migbm.so: a from-scratch
 C++/HIP (gfx950) implementation with RCCL-over-xGMI multi-GPU training.
 """
-from .basic import Booster, Dataset, LightGBMError, register_logger
+from .basic import Booster, Dataset, LightGBMError, Sequence, register_logger
 from .callback import EarlyStopException, early_stopping, log_evaluation, \
     record_evaluation, reset_parameter
 from .engine import CVBooster, cv, train
@@ -24,7 +24,7 @@ except ImportError:
 __version__ = "0.1.0"
 
 __all__ = [
-    "Dataset", "Booster", "LightGBMError", "register_logger",
+    "Dataset", "Booster", "LightGBMError", "register_logger", "Sequence",
     "train", "cv", "CVBooster",
     "early_stopping", "log_evaluation", "record_evaluation", "reset_parameter",
     "EarlyStopException",
