@@ -97,6 +97,9 @@ struct LeafContext {
   data_size_t num_data = 0;
   double parent_output = 0.0;   // for path smoothing
   int depth = 0;
+  // monotone-constraint output bounds (BasicLeafConstraints propagation)
+  double out_lo = -std::numeric_limits<double>::infinity();
+  double out_hi = std::numeric_limits<double>::infinity();
 };
 
 /*! Best numerical threshold for one feature.

@@ -55,6 +55,9 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
                                                         leaf.parent_output);
       double ro = GainMath::CalculateSplittedLeafOutput(sgr, shr, l1, l2, mds, smooth, rc,
                                                         leaf.parent_output);
+      // BasicLeafConstraints: clamp candidate outputs into the leaf's inherited bounds
+      lo = std::min(std::max(lo, leaf.out_lo), leaf.out_hi);
+      ro = std::min(std::max(ro, leaf.out_lo), leaf.out_hi);
       if (monotone_constraint != 0) {
         if (monotone_constraint > 0 && lo > ro) continue;
         if (monotone_constraint < 0 && lo < ro) continue;
@@ -85,6 +88,8 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
   out->right_output = GainMath::CalculateSplittedLeafOutput(
       out->right_sum_gradient, out->right_sum_hessian, l1, l2, mds, smooth, out->right_count,
       leaf.parent_output);
+  out->left_output = std::min(std::max(out->left_output, leaf.out_lo), leaf.out_hi);
+  out->right_output = std::min(std::max(out->right_output, leaf.out_lo), leaf.out_hi);
   out->monotone_type = monotone_constraint;
 }
 

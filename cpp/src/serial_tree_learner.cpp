@@ -450,10 +450,24 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                                s.right_output * s.right_sum_hessian) /
                               std::max(s.left_sum_hessian + s.right_sum_hessian, kEpsilon);
     int parent_depth = leaf_ctx_[best_leaf].depth;
+    const double b_lo = leaf_ctx_[best_leaf].out_lo;
+    const double b_hi = leaf_ctx_[best_leaf].out_hi;
     leaf_ctx_[best_leaf] = {s.left_sum_gradient, s.left_sum_hessian, left_cnt_actual,
-                            parent_out, parent_depth + 1};
+                            parent_out, parent_depth + 1, b_lo, b_hi};
     leaf_ctx_[right_leaf] = {s.right_sum_gradient, s.right_sum_hessian, right_cnt_actual,
-                             parent_out, parent_depth + 1};
+                             parent_out, parent_depth + 1, b_lo, b_hi};
+    if (s.monotone_type != 0) {
+      // BasicLeafConstraints: descendants of the low side may not exceed the split
+      // midpoint, and vice versa — monotonicity holds for the whole subtree
+      const double mid = (s.left_output + s.right_output) / 2.0;
+      if (s.monotone_type > 0) {
+        leaf_ctx_[best_leaf].out_hi = std::min(b_hi, mid);
+        leaf_ctx_[right_leaf].out_lo = std::max(b_lo, mid);
+      } else {
+        leaf_ctx_[best_leaf].out_lo = std::max(b_lo, mid);
+        leaf_ctx_[right_leaf].out_hi = std::min(b_hi, mid);
+      }
+    }
     ++num_leaves;
 
     // histograms: build smaller child, subtract for larger.

@@ -267,6 +267,25 @@ def test_monotone_constraints():
     assert np.all(np.diff(pred) >= -1e-9)
 
 
+def test_monotone_constraints_deep_propagation():
+    # deep trees + adversarial interaction: monotonicity must hold at EVERY slice
+    # (requires leaf-bound propagation, not just sibling-level rejection)
+    rng = np.random.RandomState(7)
+    X = rng.rand(6000, 3)
+    y = (1.5 * X[:, 0] - 1.2 * X[:, 1] + 2.0 * np.sin(6 * X[:, 2]) +
+         0.05 * rng.randn(6000)).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "monotone_constraints": [1, -1, 0],
+                     "num_leaves": 63, "learning_rate": 0.1, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 60)
+    xs = np.linspace(0.02, 0.98, 25)
+    for other in (0.1, 0.5, 0.9):
+        for z in (0.2, 0.8):
+            g0 = np.column_stack([xs, np.full(25, other), np.full(25, z)])
+            assert np.all(np.diff(bst.predict(g0)) >= -1e-9)     # +1 on f0
+            g1 = np.column_stack([np.full(25, other), xs, np.full(25, z)])
+            assert np.all(np.diff(bst.predict(g1)) <= 1e-9)      # -1 on f1
+
+
 def test_weights_affect_training():
     X, y = _binary_data(n=2000)
     w = np.where(y > 0, 10.0, 1.0).astype(np.float32)
