@@ -197,13 +197,20 @@ class Application {
   void ConvertModel() {
     if (config_.input_model.empty()) Log::Fatal("convert_model requires input_model=");
     boosting_.reset(GBDT::CreateBoosting("gbdt", config_.input_model.c_str()));
-    std::string json = boosting_->DumpModel(0, -1, 0);
-    std::string out = config_.output_model.empty() ? "model.json" : config_.output_model;
+    std::string text;
+    std::string out;
+    if (config_.convert_model_language == "cpp") {
+      text = boosting_->ModelToIfElse(-1);
+      out = config_.convert_model.empty() ? "gbdt_prediction.cpp" : config_.convert_model;
+    } else {
+      text = boosting_->DumpModel(0, -1, 0);
+      out = config_.output_model.empty() ? "model.json" : config_.output_model;
+    }
     FILE* fp = fopen(out.c_str(), "w");
     if (!fp) Log::Fatal("Cannot open %s", out.c_str());
-    fwrite(json.data(), 1, json.size(), fp);
+    fwrite(text.data(), 1, text.size(), fp);
     fclose(fp);
-    Log::Info("Model JSON written to %s", out.c_str());
+    Log::Info("Converted model written to %s", out.c_str());
   }
 
   Config config_;

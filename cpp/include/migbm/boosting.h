@@ -86,6 +86,8 @@ class GBDT {
   bool SaveModelToFile(int start_iter, int num_iter, int feature_importance_type,
                        const char* filename) const;
   std::string DumpModel(int start_iter, int num_iter, int feature_importance_type) const;
+  /*! whole model as a self-contained C++ source file (if-else codegen). */
+  std::string ModelToIfElse(int num_iteration) const;
   bool LoadModelFromString(const char* str, size_t len);
 
   std::vector<double> FeatureImportance(int num_iter, int importance_type) const;

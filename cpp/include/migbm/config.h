@@ -89,6 +89,8 @@ struct Config {
   std::string interaction_constraints = "";
   int verbosity = 1;
   std::string input_model = "";
+  std::string convert_model_language = "";        // "" (json) or "cpp"
+  std::string convert_model = "gbdt_prediction.cpp";
   std::string output_model = "LightGBM_model.txt";
   int snapshot_freq = -1;
   bool use_quantized_grad = false;

@@ -112,6 +112,10 @@ class Tree {
   void OverrideLeafCounts(const std::vector<int>& counts);
 
   std::string ToString() const;   // model-text v4 tree block
+  /*! standalone C++ if-else code for this tree (reference parity: convert_model
+   *  with convert_model_language=cpp). Emits PredictTree<idx> and
+   *  PredictTree<idx>LeafIndex functions; categorical bitsets become static arrays. */
+  std::string ToIfElse(int index) const;
   std::string ToJSON() const;
 
   /*! leaf ids in pre-order; maps categorical bitset words. */

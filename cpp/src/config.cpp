@@ -215,6 +215,8 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
     else if (k == "interaction_constraints") interaction_constraints = v;
     else if (k == "verbosity") verbosity = ParseInt(v);
     else if (k == "input_model") input_model = v;
+    else if (k == "convert_model_language") convert_model_language = v;
+    else if (k == "convert_model") convert_model = v;
     else if (k == "output_model") output_model = v;
     else if (k == "snapshot_freq") snapshot_freq = ParseInt(v);
     else if (k == "use_quantized_grad") use_quantized_grad = ParseBool(v);

@@ -3,7 +3,9 @@
  *  names and layout match so models interchange with the reference ecosystem. */
 #include "migbm/boosting.h"
 
+#include <iomanip>
 #include <map>
+#include <sstream>
 
 namespace migbm {
 
@@ -226,6 +228,81 @@ GBDT* GBDT::CreateBoosting(const std::string& type, const char* model_filename) 
     ret->LoadModelFromString(buf.data(), buf.size());
   }
   return ret;
+}
+
+
+std::string GBDT::ModelToIfElse(int num_iteration) const {
+  // Self-contained C++ source for the whole model (reference parity:
+  // convert_model task with convert_model_language=cpp). Compiles with any
+  // C++11 compiler; only <cmath> is required.
+  int total_iters = num_tree_per_iteration_ > 0
+                        ? static_cast<int>(models_.size()) / num_tree_per_iteration_ : 0;
+  int end_iter = num_iteration > 0 ? std::min(num_iteration, total_iters) : total_iters;
+  const int n_models = end_iter * num_tree_per_iteration_;
+  std::stringstream ss;
+  ss << std::setprecision(17);
+  ss << "// Generated by migbm convert_model (if-else codegen).\n"
+     << "// objective: " << objective_tostring_ << "\n"
+     << "#include <cmath>\n\n"
+     << "#define MIGBM_NUM_FEATURES " << (max_feature_idx_ + 1) << "\n"
+     << "#define MIGBM_NUM_CLASSES " << num_class_ << "\n"
+     << "#define MIGBM_NUM_TREES " << n_models << "\n\n"
+     << "static inline bool NumericalDecision(double v, int missing_type, bool default_left,\n"
+     << "                                     double threshold) {\n"
+     << "  if (std::isnan(v) && missing_type != 2) v = 0.0;\n"
+     << "  if ((missing_type == 1 && v == 0.0) || (missing_type == 2 && std::isnan(v)))\n"
+     << "    return default_left;\n"
+     << "  return v <= threshold;\n"
+     << "}\n\n"
+     << "static inline bool CategoricalDecision(double v, const unsigned int* bits, int n_words) {\n"
+     << "  if (std::isnan(v)) return false;\n"
+     << "  int c = static_cast<int>(v);\n"
+     << "  if (c < 0) return false;\n"
+     << "  return (c >> 5) < n_words && ((bits[c >> 5] >> (c & 31)) & 1U);\n"
+     << "}\n\n";
+  for (int i = 0; i < n_models; ++i) ss << models_[i]->ToIfElse(i);
+  // raw score per class
+  ss << "void PredictRaw(const double* features, double* output) {\n";
+  for (int k = 0; k < num_tree_per_iteration_; ++k) {
+    ss << "  output[" << k << "] = 0.0";
+    for (int i = k; i < n_models; i += num_tree_per_iteration_)
+      ss << "\n      + PredictTree" << i << "(features)";
+    ss << ";\n";
+  }
+  if (average_output_ && end_iter > 0) {
+    for (int k = 0; k < num_tree_per_iteration_; ++k)
+      ss << "  output[" << k << "] /= " << end_iter << ".0;\n";
+  }
+  ss << "}\n\n";
+  // transformed prediction matching the training objective
+  const std::string& obj = objective_name_;
+  double sigmoid = 1.0;
+  {
+    size_t p = objective_tostring_.find("sigmoid:");
+    if (p != std::string::npos) sigmoid = atof(objective_tostring_.c_str() + p + 8);
+  }
+  ss << "void Predict(const double* features, double* output) {\n"
+     << "  PredictRaw(features, output);\n";
+  if (obj == "binary" || obj == "cross_entropy" || obj == "xentropy" ||
+      obj == "multiclassova" || obj == "ova") {
+    ss << "  for (int k = 0; k < MIGBM_NUM_CLASSES; ++k)\n"
+       << "    output[k] = 1.0 / (1.0 + std::exp(-" << sigmoid << " * output[k]));\n";
+  } else if (obj == "multiclass" || obj == "softmax") {
+    ss << "  double wmax = output[0];\n"
+       << "  for (int k = 1; k < MIGBM_NUM_CLASSES; ++k) if (output[k] > wmax) wmax = output[k];\n"
+       << "  double wsum = 0.0;\n"
+       << "  for (int k = 0; k < MIGBM_NUM_CLASSES; ++k) { output[k] = std::exp(output[k] - wmax); wsum += output[k]; }\n"
+       << "  for (int k = 0; k < MIGBM_NUM_CLASSES; ++k) output[k] /= wsum;\n";
+  } else if (obj == "poisson" || obj == "gamma" || obj == "tweedie") {
+    ss << "  for (int k = 0; k < MIGBM_NUM_CLASSES; ++k) output[k] = std::exp(output[k]);\n";
+  }
+  ss << "}\n\n";
+  // leaf indices (refit / feature engineering)
+  ss << "void PredictLeafIndex(const double* features, int* output) {\n";
+  for (int i = 0; i < n_models; ++i)
+    ss << "  output[" << i << "] = PredictTree" << i << "LeafIndex(features);\n";
+  ss << "}\n";
+  return ss.str();
 }
 
 }  // namespace migbm

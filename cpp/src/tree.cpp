@@ -3,7 +3,10 @@
 #include "migbm/tree.h"
 #include "migbm/dataset.h"
 
+#include <functional>
+#include <iomanip>
 #include <map>
+#include <sstream>
 
 namespace migbm {
 
@@ -313,6 +316,71 @@ std::string Tree::ToJSON() const {
   if (num_leaves_ > 1) dump(0);
   else dump(~0);
   ss << "}";
+  return ss.str();
+}
+
+
+
+std::string Tree::ToIfElse(int index) const {
+  // Standalone nested-if codegen. Decision semantics mirror NumericalDecision /
+  // CategoricalDecision exactly (missing handling included); the generated file
+  // depends only on <cmath>.
+  std::stringstream ss;
+  ss << std::setprecision(17);
+  // categorical bitset words for this tree, one flat static array
+  if (!cat_threshold_.empty()) {
+    ss << "static const unsigned int cat_bits_" << index << "[] = {";
+    for (size_t i = 0; i < cat_threshold_.size(); ++i) {
+      if (i) ss << ",";
+      ss << cat_threshold_[i] << "u";
+    }
+    ss << "};\n";
+  }
+  // recursive emitters for value and leaf-index variants
+  std::function<void(int, int, bool)> emit = [&](int node, int depth, bool leaf_index) {
+    std::string ind(static_cast<size_t>(depth) * 2 + 2, ' ');
+    if (node < 0) {  // leaf
+      int leaf = ~node;
+      if (leaf_index) {
+        ss << ind << "return " << leaf << ";\n";
+      } else {
+        ss << ind << "return " << leaf_value_[leaf] << ";\n";
+      }
+      return;
+    }
+    const int fid = split_feature_[node];
+    if (decision_type_[node] & kCategoricalMask) {
+      const int cat_idx = static_cast<int>(threshold_[node]);
+      const int off = cat_boundaries_[cat_idx];
+      const int n_words = cat_boundaries_[cat_idx + 1] - off;
+      ss << ind << "if (CategoricalDecision(arr[" << fid << "], cat_bits_" << index
+         << " + " << off << ", " << n_words << ")) {\n";
+    } else {
+      const int8_t dt = decision_type_[node];
+      const int missing_type = (dt >> 2) & 3;
+      const bool default_left = (dt & kDefaultLeftMask) != 0;
+      ss << ind << "if (NumericalDecision(arr[" << fid << "], " << missing_type << ", "
+         << (default_left ? "true" : "false") << ", " << threshold_[node] << ")) {\n";
+    }
+    emit(left_child_[node], depth + 1, leaf_index);
+    ss << ind << "} else {\n";
+    emit(right_child_[node], depth + 1, leaf_index);
+    ss << ind << "}\n";
+  };
+  ss << "double PredictTree" << index << "(const double* arr) {\n";
+  if (num_leaves_ <= 1) {
+    ss << "  return " << (leaf_value_.empty() ? 0.0 : leaf_value_[0]) << ";\n";
+  } else {
+    emit(0, 0, false);
+  }
+  ss << "}\n\n";
+  ss << "int PredictTree" << index << "LeafIndex(const double* arr) {\n";
+  if (num_leaves_ <= 1) {
+    ss << "  return 0;\n";
+  } else {
+    emit(0, 0, true);
+  }
+  ss << "}\n\n";
   return ss.str();
 }
 
