@@ -223,6 +223,29 @@ LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMatSingleRow(
     BoosterHandle handle, const void* data, int data_type, int ncol, int is_row_major,
     int predict_type, int start_iteration, int num_iteration, const char* parameter,
     int64_t* out_len, double* out_result);
+/*! predict over an array of row pointers (reference c_api.h LGBM_BoosterPredictForMats) */
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMats(BoosterHandle handle, const void** data,
+                                                 int data_type, int32_t nrow, int32_t ncol,
+                                                 int predict_type, int start_iteration,
+                                                 int num_iteration, const char* parameter,
+                                                 int64_t* out_len, double* out_result);
+/*! single-row fast path: bind predict config once, then per-row calls with no re-parsing
+ *  (reference c_api.h FastConfigHandle family) */
+typedef void* FastConfigHandle;
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMatSingleRowFastInit(
+    BoosterHandle handle, int predict_type, int start_iteration, int num_iteration,
+    int data_type, int32_t ncol, const char* parameter, FastConfigHandle* out_fastConfig);
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMatSingleRowFast(FastConfigHandle fastConfig_handle,
+                                                             const void* data, int64_t* out_len,
+                                                             double* out_result);
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForCSRSingleRowFastInit(
+    BoosterHandle handle, int predict_type, int start_iteration, int num_iteration,
+    int data_type, int64_t num_col, const char* parameter, FastConfigHandle* out_fastConfig);
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForCSRSingleRowFast(
+    FastConfigHandle fastConfig_handle, const void* indptr, int indptr_type,
+    const int32_t* indices, const void* data, int64_t nindptr, int64_t nelem,
+    int64_t* out_len, double* out_result);
+LIGHTGBM_C_EXPORT int LGBM_FastConfigFree(FastConfigHandle fastConfig);
 LIGHTGBM_C_EXPORT int LGBM_BoosterSaveModel(BoosterHandle handle, int start_iteration,
                                             int num_iteration, int feature_importance_type,
                                             const char* filename);

@@ -1096,6 +1096,81 @@ int LGBM_BoosterPredictForCSRSingleRow(BoosterHandle handle, const void* indptr,
                                    num_iteration, parameter, out_len, out_result);
 }
 
+int LGBM_BoosterPredictForMats(BoosterHandle handle, const void** data, int data_type,
+                               int32_t nrow, int32_t ncol, int predict_type,
+                               int start_iteration, int num_iteration, const char*,
+                               int64_t* out_len, double* out_result) {
+  API_BEGIN();
+  auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
+  auto row_getter = [data, data_type, ncol](int64_t r, double* o) {
+    auto get = MakeGetter(data[r], data_type);
+    for (int c = 0; c < ncol; ++c) o[c] = get(c);
+  };
+  PredictRows(b, row_getter, nrow, ncol, predict_type, start_iteration, num_iteration,
+              out_result);
+  *out_len = static_cast<int64_t>(nrow) *
+             b->NumPredictOneRow(start_iteration, num_iteration,
+                                 predict_type == C_API_PREDICT_LEAF_INDEX,
+                                 predict_type == C_API_PREDICT_CONTRIB);
+  API_END();
+}
+
+/*! bound single-row predict config (reference FastConfig semantics: parameters and
+ *  shape parsed once at Init; per-row calls do no setup work) */
+struct FastConfig {
+  BoosterHandle booster;
+  int predict_type, start_iteration, num_iteration, data_type;
+  int64_t ncol;
+};
+
+int LGBM_BoosterPredictForMatSingleRowFastInit(BoosterHandle handle, int predict_type,
+                                               int start_iteration, int num_iteration,
+                                               int data_type, int32_t ncol, const char*,
+                                               FastConfigHandle* out_fastConfig) {
+  API_BEGIN();
+  *out_fastConfig = new FastConfig{handle, predict_type, start_iteration, num_iteration,
+                                   data_type, ncol};
+  API_END();
+}
+
+int LGBM_BoosterPredictForMatSingleRowFast(FastConfigHandle fastConfig_handle,
+                                           const void* data, int64_t* out_len,
+                                           double* out_result) {
+  auto* fc = static_cast<FastConfig*>(fastConfig_handle);
+  return LGBM_BoosterPredictForMat(fc->booster, data, fc->data_type, 1,
+                                   static_cast<int32_t>(fc->ncol), 1, fc->predict_type,
+                                   fc->start_iteration, fc->num_iteration, "", out_len,
+                                   out_result);
+}
+
+int LGBM_BoosterPredictForCSRSingleRowFastInit(BoosterHandle handle, int predict_type,
+                                               int start_iteration, int num_iteration,
+                                               int data_type, int64_t num_col, const char*,
+                                               FastConfigHandle* out_fastConfig) {
+  API_BEGIN();
+  *out_fastConfig = new FastConfig{handle, predict_type, start_iteration, num_iteration,
+                                   data_type, num_col};
+  API_END();
+}
+
+int LGBM_BoosterPredictForCSRSingleRowFast(FastConfigHandle fastConfig_handle,
+                                           const void* indptr, int indptr_type,
+                                           const int32_t* indices, const void* data,
+                                           int64_t nindptr, int64_t nelem, int64_t* out_len,
+                                           double* out_result) {
+  auto* fc = static_cast<FastConfig*>(fastConfig_handle);
+  return LGBM_BoosterPredictForCSR(fc->booster, indptr, indptr_type, indices, data,
+                                   fc->data_type, nindptr, nelem, fc->ncol, fc->predict_type,
+                                   fc->start_iteration, fc->num_iteration, "", out_len,
+                                   out_result);
+}
+
+int LGBM_FastConfigFree(FastConfigHandle fastConfig) {
+  API_BEGIN();
+  delete static_cast<FastConfig*>(fastConfig);
+  API_END();
+}
+
 int LGBM_BoosterPredictForFile(BoosterHandle handle, const char* data_filename,
                                int data_has_header, int predict_type, int start_iteration,
                                int num_iteration, const char* parameter,

@@ -146,3 +146,44 @@ def test_trees_to_dataframe():
     df = bst.trees_to_dataframe()
     assert set(df["tree_index"]) == {0, 1, 2}
     assert "threshold" in df.columns
+
+
+def test_c_api_predict_for_mats_and_fast_path():
+    """LGBM_BoosterPredictForMats (row-pointer array) and the FastConfig single-row
+    path must match Booster.predict (reference c_api.h parity)."""
+    import ctypes
+    from lightgbm_amd.basic import _LIB
+    rng = np.random.RandomState(5)
+    X = rng.rand(400, 6)
+    y = (X[:, 0] + X[:, 1] > 1.0).astype(np.float64)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 8)
+    expected = bst.predict(X[:16])
+    h = bst._handle
+
+    # --- PredictForMats: array of row pointers
+    rows = np.ascontiguousarray(X[:16], dtype=np.float64)
+    ptrs = (ctypes.c_void_p * 16)(*[rows[i].ctypes.data for i in range(16)])
+    out = np.zeros(16, dtype=np.float64)
+    out_len = ctypes.c_int64(0)
+    rc = _LIB.LGBM_BoosterPredictForMats(
+        h, ptrs, ctypes.c_int(1), ctypes.c_int32(16), ctypes.c_int32(6), ctypes.c_int(0),
+        ctypes.c_int(0), ctypes.c_int(-1), b"", ctypes.byref(out_len),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+    assert rc == 0 and out_len.value == 16
+    np.testing.assert_allclose(out, expected, rtol=1e-12)
+
+    # --- FastConfig single-row path
+    fc = ctypes.c_void_p()
+    rc = _LIB.LGBM_BoosterPredictForMatSingleRowFastInit(
+        h, ctypes.c_int(0), ctypes.c_int(0), ctypes.c_int(-1), ctypes.c_int(1),
+        ctypes.c_int32(6), b"", ctypes.byref(fc))
+    assert rc == 0
+    one = np.zeros(1, dtype=np.float64)
+    for i in range(4):
+        row = np.ascontiguousarray(X[i], dtype=np.float64)
+        rc = _LIB.LGBM_BoosterPredictForMatSingleRowFast(
+            fc, row.ctypes.data_as(ctypes.c_void_p), ctypes.byref(out_len),
+            one.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        assert rc == 0
+        assert abs(one[0] - expected[i]) < 1e-12
+    assert _LIB.LGBM_FastConfigFree(fc) == 0
