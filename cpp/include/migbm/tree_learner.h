@@ -167,6 +167,8 @@ class SerialTreeLearner : public TreeLearner {
   const score_t* gradients_ = nullptr;
   const score_t* hessians_ = nullptr;
   std::vector<score_t> ordered_grad_, ordered_hess_;
+  std::vector<score_t> quant_grad_, quant_hess_;  // CPU quantized-training grids
+  uint32_t quant_seed_ = 0x9E3779B9u;
   DataPartition partition_;
   std::vector<hist_t> hist_store_;          // num_leaves slots x 2*num_total_bin
   std::vector<int> leaf_to_slot_;

@@ -334,6 +334,41 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                                bool /*is_first_tree*/) {
   gradients_ = gradients;
   hessians_ = hessians;
+  if (config_->use_quantized_grad) {
+    // CPU gradient discretization (reference GradientDiscretizer semantics):
+    // stochastic rounding onto a num_grad_quant_bins grid scaled by absmax.
+    // Values stay score_t so the whole histogram pipeline is unchanged; sums of
+    // grid multiples are exact in fp64, so subtraction keeps its guarantee.
+    const data_size_t n = train_data_->num_data();
+    quant_grad_.resize(n);
+    quant_hess_.resize(n);
+    double gmax = 0.0, hmax = 0.0;
+#pragma omp parallel for schedule(static) reduction(max : gmax, hmax)
+    for (data_size_t i = 0; i < n; ++i) {
+      gmax = std::max(gmax, std::fabs(static_cast<double>(gradients[i])));
+      hmax = std::max(hmax, static_cast<double>(hessians[i]));
+    }
+    const int levels = std::max(1, config_->num_grad_quant_bins / 2);
+    const double gs = gmax > 0 ? gmax / levels : 1.0;
+    const double hs = hmax > 0 ? hmax / (2.0 * levels) : 1.0;
+    quant_seed_ = quant_seed_ * 1664525u + 1013904223u;
+    const uint32_t seed = quant_seed_;
+#pragma omp parallel for schedule(static)
+    for (data_size_t i = 0; i < n; ++i) {
+      uint32_t x = (static_cast<uint32_t>(i) * 2654435761u) ^ seed;
+      x ^= x >> 16; x *= 2246822519u; x ^= x >> 13;
+      const double rg = (x & 0xFFFF) * (1.0 / 65536.0);
+      const double rh = ((x >> 16) & 0xFFFF) * (1.0 / 65536.0);
+      int gq = static_cast<int>(std::floor(gradients[i] / gs + rg));
+      int hq = static_cast<int>(std::floor(hessians[i] / hs + rh));
+      gq = std::max(-levels, std::min(levels, gq));
+      hq = std::max(0, std::min(2 * levels, hq));
+      quant_grad_[i] = static_cast<score_t>(gq * gs);
+      quant_hess_[i] = static_cast<score_t>(hq * hs);
+    }
+    gradients_ = quant_grad_.data();
+    hessians_ = quant_hess_.data();
+  }
   ++iter_counter_;
   is_feature_used_ = SampleFeatures(false);
 
@@ -497,6 +532,23 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
 
     FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
     FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);
+  }
+  if (config_->use_quantized_grad && config_->quant_train_renew_leaf) {
+    // renew leaf outputs from the UNquantized gradients (reference
+    // quant_train_renew_leaf): removes discretization bias from the values
+    // while splits stay those chosen on the quantized histograms.
+    for (int l = 0; l < tree->num_leaves(); ++l) {
+      data_size_t cnt;
+      const data_size_t* idx = partition_.GetIndexOnLeaf(l, &cnt);
+      double sg = 0.0, sh = 0.0;
+#pragma omp parallel for schedule(static) reduction(+ : sg, sh)
+      for (data_size_t i = 0; i < cnt; ++i) {
+        sg += gradients[idx[i]];
+        sh += hessians[idx[i]];
+      }
+      tree->SetLeafOutput(l, GainMath::CalculateSplittedLeafOutput(
+          sg, sh, config_->lambda_l1, config_->lambda_l2, config_->max_delta_step));
+    }
   }
   if (config_->linear_tree) CalculateLinear(tree.get());
   return tree.release();

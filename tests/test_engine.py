@@ -405,3 +405,23 @@ def test_auc_mu():
                "verbosity": -1}, train, 20, valid_sets=[train], valid_names=["t"],
               callbacks=[lgb.record_evaluation(ev)])
     assert ev["t"]["auc_mu"][-1] > 0.9
+
+
+def test_quantized_training_cpu():
+    """use_quantized_grad on CPU: discretized gradients still reach near-baseline
+    quality, and quant_train_renew_leaf changes (de-biases) leaf values."""
+    X, y = _binary_data(n=4000)
+    base = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 30)
+    q = lgb.train({"objective": "binary", "verbosity": -1, "use_quantized_grad": True,
+                   "num_grad_quant_bins": 8}, lgb.Dataset(X, label=y), 30)
+    from sklearn.metrics import roc_auc_score
+    auc_b = roc_auc_score(y, base.predict(X))
+    auc_q = roc_auc_score(y, q.predict(X))
+    assert auc_q > auc_b - 0.03
+    # predictions must actually differ (quantization is active)
+    assert not np.allclose(base.predict(X), q.predict(X))
+    r = lgb.train({"objective": "binary", "verbosity": -1, "use_quantized_grad": True,
+                   "num_grad_quant_bins": 8, "quant_train_renew_leaf": True},
+                  lgb.Dataset(X, label=y), 30)
+    assert not np.allclose(q.predict(X), r.predict(X))
+    assert roc_auc_score(y, r.predict(X)) > auc_b - 0.03
