@@ -207,6 +207,14 @@ class Dataset {
 
   data_size_t num_data_ = 0;
   int num_total_features_ = 0;
+
+ public:
+  /*! Append every feature (and its storage columns) of `other` to this dataset.
+   *  Same row count required. Reference parity: LGBM_DatasetAddFeaturesFrom /
+   *  Dataset::AddFeaturesFrom (src/io/dataset.cpp). */
+  void AddFeaturesFrom(const Dataset* other);
+
+ private:
   std::vector<int> used_feature_map_;        // orig -> inner (-1 trivial)
   std::vector<int> real_feature_index_;      // inner -> orig
   std::vector<std::unique_ptr<BinMapper>> bin_mappers_;  // per inner feature

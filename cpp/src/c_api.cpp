@@ -742,6 +742,12 @@ int LGBM_DatasetDumpText(DatasetHandle handle, const char* filename) {
   API_END();
 }
 
+int LGBM_DatasetAddFeaturesFrom(DatasetHandle target, DatasetHandle source) {
+  API_BEGIN();
+  static_cast<Dataset*>(target)->AddFeaturesFrom(static_cast<Dataset*>(source));
+  API_END();
+}
+
 int LGBM_DatasetSetField(DatasetHandle handle, const char* field_name, const void* field_data,
                          int num_element, int type) {
   API_BEGIN();
@@ -828,12 +834,6 @@ int LGBM_DatasetGetFeatureNumBin(DatasetHandle handle, int feature, int32_t* out
   Dataset* d = static_cast<Dataset*>(handle);
   int inner = d->InnerFeatureIndex(feature);
   *out = inner >= 0 ? d->FeatureNumBin(inner) : 0;
-  API_END();
-}
-
-int LGBM_DatasetAddFeaturesFrom(DatasetHandle, DatasetHandle) {
-  API_BEGIN();
-  Log::Fatal("DatasetAddFeaturesFrom is not yet supported by migbm");
   API_END();
 }
 

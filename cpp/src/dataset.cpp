@@ -413,6 +413,52 @@ void Dataset::ConstructHistograms(const std::vector<int8_t>& is_feature_used,
   }
 }
 
+void Dataset::AddFeaturesFrom(const Dataset* other) {
+  if (other->num_data_ != num_data_)
+    Log::Fatal("Cannot add features from other Dataset with a different number of rows");
+  const int col_base = static_cast<int>(columns_.size());
+  const int feat_base = num_features();
+  const int orig_base = num_total_features_;
+  // columns + per-feature storage mapping
+  for (const auto& c : other->columns_) columns_.push_back(c);
+  for (size_t f = 0; f < other->bin_mappers_.size(); ++f) {
+    bin_mappers_.push_back(std::make_unique<BinMapper>(*other->bin_mappers_[f]));
+    col_of_feature_.push_back(other->col_of_feature_[f] + col_base);
+    off_in_col_.push_back(other->off_in_col_[f]);
+    feature_bundled_.push_back(other->feature_bundled_[f]);
+    real_feature_index_.push_back(other->real_feature_index_[f] + orig_base);
+  }
+  for (const auto& members : other->column_features_) {
+    std::vector<int> shifted;
+    for (int f : members) shifted.push_back(f + feat_base);
+    column_features_.push_back(std::move(shifted));
+  }
+  has_bundles_ = has_bundles_ || other->has_bundles_;
+  // original-index bookkeeping
+  for (int o = 0; o < other->num_total_features_; ++o) {
+    const int inner = other->used_feature_map_[o];
+    used_feature_map_.push_back(inner >= 0 ? inner + feat_base : -1);
+  }
+  num_total_features_ += other->num_total_features_;
+  for (int o = 0; o < other->num_total_features_; ++o) {
+    feature_names_.push_back(o < static_cast<int>(other->feature_names_.size())
+                                 ? other->feature_names_[o]
+                                 : "Column_" + std::to_string(orig_base + o));
+  }
+  // rebuild hist offsets (append other's bins after ours)
+  for (size_t f = 0; f < other->bin_mappers_.size(); ++f) {
+    hist_offsets_.push_back(static_cast<uint32_t>(num_total_bin_) + other->hist_offsets_[f] -
+                            (other->hist_offsets_.empty() ? 0 : other->hist_offsets_[0]));
+  }
+  num_total_bin_ += other->num_total_bin_;
+  if (!raw_values_.empty() || !other->raw_values_.empty()) {
+    raw_values_.resize(feat_base);
+    for (const auto& rv : other->raw_values_) raw_values_.push_back(rv);
+  }
+  row_view_built_ = false;
+  row_view_ = RowMajorView();
+}
+
 void Dataset::set_feature_names(const std::vector<std::string>& names) {
   feature_names_ = names;
 }

@@ -31,10 +31,20 @@ class BaggingStrategy : public SampleStrategy {
     bag_indices_.reserve(n);
     const bool posneg = cfg_->pos_bagging_fraction < 1.0 || cfg_->neg_bagging_fraction < 1.0;
     const label_t* label = data_->metadata().label();
-    for (data_size_t i = 0; i < n; ++i) {
-      double frac = cfg_->bagging_fraction;
-      if (posneg) frac = label[i] > 0 ? cfg_->pos_bagging_fraction : cfg_->neg_bagging_fraction;
-      if (rng_.NextFloat() < frac) bag_indices_.push_back(i);
+    const data_size_t* qb = data_->metadata().query_boundaries();
+    if (cfg_->bagging_by_query && qb != nullptr) {
+      // sample whole queries: keeps ranking groups intact in the bag
+      const data_size_t nq = data_->metadata().num_queries();
+      for (data_size_t q = 0; q < nq; ++q) {
+        if (rng_.NextFloat() < cfg_->bagging_fraction)
+          for (data_size_t i = qb[q]; i < qb[q + 1]; ++i) bag_indices_.push_back(i);
+      }
+    } else {
+      for (data_size_t i = 0; i < n; ++i) {
+        double frac = cfg_->bagging_fraction;
+        if (posneg) frac = label[i] > 0 ? cfg_->pos_bagging_fraction : cfg_->neg_bagging_fraction;
+        if (rng_.NextFloat() < frac) bag_indices_.push_back(i);
+      }
     }
     bag_cnt_ = static_cast<data_size_t>(bag_indices_.size());
     learner->SetBaggingData(nullptr, bag_indices_.data(), bag_cnt_);

@@ -427,6 +427,16 @@ class Dataset:
         _safe_call(_LIB.LGBM_DatasetGetNumFeature(self._handle, ctypes.byref(out)))
         return out.value
 
+    def add_features_from(self, other):
+        """Add features of `other` Dataset to this one (both must be constructed
+        and have the same number of rows). Reference parity: Dataset.add_features_from."""
+        self.construct()
+        other.construct()
+        _safe_call(_LIB.LGBM_DatasetAddFeaturesFrom(self._handle, other._handle))
+        if isinstance(self.feature_name, list) and isinstance(other.feature_name, list):
+            self.feature_name = self.feature_name + other.feature_name
+        return self
+
     def feature_num_bin(self, feature):
         self.construct()
         out = ctypes.c_int32(0)

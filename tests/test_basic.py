@@ -187,3 +187,26 @@ def test_c_api_predict_for_mats_and_fast_path():
         assert rc == 0
         assert abs(one[0] - expected[i]) < 1e-12
     assert _LIB.LGBM_FastConfigFree(fc) == 0
+
+
+def test_add_features_from():
+    """Dataset.add_features_from appends another dataset's features in place."""
+    rng = np.random.RandomState(11)
+    Xa, Xb = rng.rand(1200, 3), rng.rand(1200, 4)
+    y = (Xa[:, 0] + Xb[:, 0] > 1.0).astype(np.float64)
+    da = lgb.Dataset(Xa, label=y, free_raw_data=False)
+    db = lgb.Dataset(Xb, free_raw_data=False)
+    da.construct()
+    db.construct()
+    da.add_features_from(db)
+    assert da.num_feature() == 7
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, da, 20)
+    # features from BOTH sources must be usable: combined model beats Xa-only
+    bst_a = lgb.train({"objective": "binary", "verbosity": -1},
+                      lgb.Dataset(Xa, label=y), 20)
+    X_full = np.hstack([Xa, Xb])
+    acc_joint = ((bst.predict(X_full) > 0.5) == y).mean()
+    acc_a = ((bst_a.predict(Xa) > 0.5) == y).mean()
+    assert acc_joint > acc_a + 0.02
+    imp = bst.feature_importance()
+    assert imp[:3].sum() > 0 and imp[3:].sum() > 0

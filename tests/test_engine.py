@@ -425,3 +425,24 @@ def test_quantized_training_cpu():
                   lgb.Dataset(X, label=y), 30)
     assert not np.allclose(q.predict(X), r.predict(X))
     assert roc_auc_score(y, r.predict(X)) > auc_b - 0.03
+
+
+def test_bagging_by_query():
+    """bagging_by_query samples whole query groups for ranking bags."""
+    rng = np.random.RandomState(2)
+    rows, labels, groups = [], [], []
+    for q in range(120):
+        nq = rng.randint(5, 20)
+        Xq = rng.randn(nq, 5)
+        rel = (Xq[:, 0] > 0.3).astype(int)
+        rows.append(Xq); labels.append(rel); groups.append(nq)
+    X = np.vstack(rows)
+    y = np.concatenate(labels).astype(np.float32)
+    ev = {}
+    train = lgb.Dataset(X, label=y, group=np.array(groups, dtype=np.int32))
+    lgb.train({"objective": "lambdarank", "metric": "ndcg", "eval_at": [5],
+               "bagging_by_query": True, "bagging_fraction": 0.5, "bagging_freq": 1,
+               "verbosity": -1}, train, 25,
+              valid_sets=[train], valid_names=["train"],
+              callbacks=[lgb.record_evaluation(ev)])
+    assert ev["train"]["ndcg@5"][-1] > 0.75
