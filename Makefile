@@ -51,9 +51,22 @@ $(BUILD)/hip/%.o: cpp/src/hip/%.hip.cpp | $(BUILD)
 $(TARGET): $(HOST_OBJS) $(HIP_OBJS) | $(BUILD)
 	$(LINKER) $(LDFLAGS) $(HOST_OBJS) $(HIP_OBJS) $(EXTRA_LIBS) -o $@
 
+# ---- ASAN lane: host-only (no HIP) CLI under AddressSanitizer.
+# Usage: make asan && ./build_asan/migbm_asan config=examples/binary_classification/train.conf
+ASAN_BUILD := build_asan
+ASAN_FLAGS := -O1 -g -std=c++17 -fopenmp -fsanitize=address -fno-omit-frame-pointer \
+              -Icpp/include -DMIGBM_NO_HIP
+ASAN_CLI   := $(ASAN_BUILD)/migbm_asan
+
+asan: $(ASAN_CLI)
+
+$(ASAN_CLI): $(HOST_SRCS) cpp/cli/main.cpp
+	mkdir -p $(ASAN_BUILD)
+	$(CXX) $(ASAN_FLAGS) $(HOST_SRCS) cpp/cli/main.cpp -o $(ASAN_CLI)
+
 clean:
-	rm -rf $(BUILD) $(TARGET) $(CLI)
+	rm -rf $(BUILD) $(TARGET) $(CLI) $(ASAN_BUILD)
 
 -include $(BUILD)/*.d $(BUILD)/hip/*.d
 
-.PHONY: all clean
+.PHONY: all clean asan

@@ -21,7 +21,9 @@ class Application {
     std::unordered_map<std::string, std::string> params;
     for (int i = 1; i < argc; ++i) {
       auto kv = Common::Split(argv[i], '=');
-      if (kv.size() == 2) params[Common::ToLower(Common::Trim(kv[0]))] = Common::Trim(kv[1]);
+      if (kv.size() == 2)
+        params[Config::ResolveAlias(Common::ToLower(Common::Trim(kv[0])))] =
+            Common::Trim(kv[1]);
     }
     // config file first, argv overrides
     auto it = params.find("config");
@@ -37,7 +39,7 @@ class Application {
         if (line.empty()) continue;
         auto eq = line.find('=');
         if (eq == std::string::npos) continue;
-        file_params[Common::ToLower(Common::Trim(line.substr(0, eq)))] =
+        file_params[Config::ResolveAlias(Common::ToLower(Common::Trim(line.substr(0, eq))))] =
             Common::Trim(line.substr(eq + 1));
       }
       for (auto& kv : params) file_params[kv.first] = kv.second;  // argv wins

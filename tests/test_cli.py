@@ -112,3 +112,15 @@ int main() {
                          timeout=300, check=True)
     pred_gen = np.array([float(t) for t in res.stdout.split()])
     np.testing.assert_allclose(pred_gen, bst.predict(X[:200]), rtol=1e-12)
+
+
+def test_cli_arg_overrides_config_alias():
+    """An argv param must override its ALIAS form in the config file
+    (num_iterations on argv vs num_trees in train.conf)."""
+    import subprocess, tempfile
+    with tempfile.TemporaryDirectory() as td:
+        out = Path(td) / "m.txt"
+        subprocess.run([str(CLI), "config=train.conf", "num_iterations=3",
+                        f"output_model={out}"],
+                       cwd=EXAMPLE, capture_output=True, timeout=300, check=True)
+        assert out.read_text().count("Tree=") == 3
