@@ -541,60 +541,61 @@ __global__ void __launch_bounds__(64) k_best_feat(
 }
 
 /*! reduce per-feature records to one per leaf; blockIdx.x = which child. */
-__global__ void k_best_leaf(const SplitRec* __restrict__ feat_best, int nf,
-                            SplitRec* __restrict__ leaf_best,
-                            const int* __restrict__ leafA_ptr,
-                            const int* __restrict__ counters, int leafB_from_counters) {
-  const int which = blockIdx.x;
+__global__ void k_best_leaf_overall(const SplitRec* __restrict__ feat_best, int nf,
+                                    SplitRec* __restrict__ leaf_best,
+                                    const int* __restrict__ leafA_ptr,
+                                    const int* __restrict__ counters,
+                                    int leafB_from_counters,
+                                    SplitRec* __restrict__ winner,
+                                    int* __restrict__ winner_leaf) {
+  // Fused per-leaf argmax (both fresh children) + overall winner selection.
+  // One block: the per-leaf reduction is tiny (nf recs), and fusing removes two
+  // kernel launches per split from the device-driven loop.
   const int leafA = *leafA_ptr;
   const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
-  const int leaf = which == 0 ? leafA : leafB;
-  if (leaf < 0) return;
-  const SplitRec* cand = feat_best + which * nf;
-  __shared__ int s_idx[256];
-  __shared__ double s_gain[256];
-  const int tid = threadIdx.x;
-  int bi = -1;
-  double bg = -1e308;
-  for (int f = tid; f < nf; f += blockDim.x) {
-    if (cand[f].valid &&
-        (cand[f].gain > bg || (cand[f].gain == bg && (bi < 0 || f < bi)))) {
-      bg = cand[f].gain;
-      bi = f;
-    }
-  }
-  s_idx[tid] = bi;
-  s_gain[tid] = bg;
-  __syncthreads();
-  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
-    if (tid < s) {
-      if (s_idx[tid + s] >= 0 &&
-          (s_idx[tid] < 0 || s_gain[tid + s] > s_gain[tid] ||
-           (s_gain[tid + s] == s_gain[tid] && s_idx[tid + s] < s_idx[tid]))) {
-        s_idx[tid] = s_idx[tid + s];
-        s_gain[tid] = s_gain[tid + s];
-      }
-    }
-    __syncthreads();
-  }
-  if (tid == 0) {
-    if (s_idx[0] >= 0) leaf_best[leaf] = cand[s_idx[0]];
-    else {
-      leaf_best[leaf].valid = 0;
-      leaf_best[leaf].gain = -1e308;
-    }
-  }
-}
-
-__global__ void k_best_overall(const SplitRec* __restrict__ leaf_best,
-                               const int* __restrict__ counters,
-                               SplitRec* __restrict__ winner, int* __restrict__ winner_leaf) {
   const int num_leaves = counters[0];
   __shared__ int s_idx[256];
   __shared__ double s_gain[256];
   const int tid = threadIdx.x;
+  for (int which = 0; which < 2; ++which) {
+    const int leaf = which == 0 ? leafA : leafB;
+    if (leaf < 0) continue;
+    const SplitRec* cand = feat_best + which * nf;
+    int bi = -1;
+    double bg = -1e308;
+    for (int f = tid; f < nf; f += blockDim.x) {
+      if (cand[f].valid &&
+          (cand[f].gain > bg || (cand[f].gain == bg && (bi < 0 || f < bi)))) {
+        bg = cand[f].gain;
+        bi = f;
+      }
+    }
+    s_idx[tid] = bi;
+    s_gain[tid] = bg;
+    __syncthreads();
+    for (int r = blockDim.x / 2; r > 0; r >>= 1) {
+      if (tid < r) {
+        if (s_idx[tid + r] >= 0 &&
+            (s_idx[tid] < 0 || s_gain[tid + r] > s_gain[tid] ||
+             (s_gain[tid + r] == s_gain[tid] && s_idx[tid + r] < s_idx[tid]))) {
+          s_idx[tid] = s_idx[tid + r];
+          s_gain[tid] = s_gain[tid + r];
+        }
+      }
+      __syncthreads();
+    }
+    if (tid == 0) {
+      if (s_idx[0] >= 0) leaf_best[leaf] = cand[s_idx[0]];
+      else {
+        leaf_best[leaf].valid = 0;
+        leaf_best[leaf].gain = -1e308;
+      }
+    }
+    __syncthreads();
+  }
+  // overall winner over every live leaf (requires strictly positive improvement)
   int bi = -1;
-  double bg = 0.0;  // require strictly positive improvement
+  double bg = 0.0;
   for (int l = tid; l < num_leaves; l += blockDim.x) {
     if (leaf_best[l].valid && leaf_best[l].gain > bg) {
       bg = leaf_best[l].gain;
@@ -604,13 +605,13 @@ __global__ void k_best_overall(const SplitRec* __restrict__ leaf_best,
   s_idx[tid] = bi;
   s_gain[tid] = bg;
   __syncthreads();
-  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
-    if (tid < s) {
-      if (s_idx[tid + s] >= 0 &&
-          (s_idx[tid] < 0 || s_gain[tid + s] > s_gain[tid] ||
-           (s_gain[tid + s] == s_gain[tid] && s_idx[tid + s] < s_idx[tid]))) {
-        s_idx[tid] = s_idx[tid + s];
-        s_gain[tid] = s_gain[tid + s];
+  for (int r = blockDim.x / 2; r > 0; r >>= 1) {
+    if (tid < r) {
+      if (s_idx[tid + r] >= 0 &&
+          (s_idx[tid] < 0 || s_gain[tid + r] > s_gain[tid] ||
+           (s_gain[tid + r] == s_gain[tid] && s_idx[tid + r] < s_idx[tid]))) {
+        s_idx[tid] = s_idx[tid + r];
+        s_gain[tid] = s_gain[tid + r];
       }
     }
     __syncthreads();
@@ -811,9 +812,12 @@ __global__ void k_copy_back(const uint32_t* __restrict__ tmp_base,
                             uint32_t* __restrict__ idx_base,
                             const int* __restrict__ leaf_begin,
                             const int* __restrict__ leaf_cnt,
-                            const int* __restrict__ Lptr) {
+                            const int* __restrict__ Lptr, const int* __restrict__ ctr,
+                            int64_t* gbuf) {
   const int L = *Lptr;
   if (L < 0) return;
+  // fused store-left: publish the local left count for the (optional) RCCL sum
+  if (blockIdx.x == 0 && threadIdx.x == 0) gbuf[0] = ctr[0];
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
@@ -821,23 +825,44 @@ __global__ void k_copy_back(const uint32_t* __restrict__ tmp_base,
     idx_base[begin + i] = tmp_base[begin + i];
 }
 
-__global__ void k_store_left(const int* __restrict__ ctr, const int* __restrict__ Lptr,
-                             int64_t* gbuf) {
-  if (*Lptr < 0) return;
-  gbuf[0] = ctr[0];
-}
+__device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_slot,
+                                    LeafStat* stats, const SplitRec* winner, int L,
+                                    int* counters, LogEntry* log, const int* ctr,
+                                    const int64_t* gbuf);
 
-/*! device-side split bookkeeping: segments, stats, slot map, split log. One thread. */
+/*! device-side split bookkeeping: segments, stats, slot map, split log. Thread 0
+ *  does the bookkeeping; the whole block then zeroes the spare histogram slot the
+ *  upcoming smaller-child build will accumulate into (fused k_hist_zero). */
 __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
                            const SplitRec* __restrict__ winner,
                            const int* __restrict__ Lptr, int* counters,
                            LogEntry* __restrict__ log, const int* __restrict__ ctr,
-                           const int64_t* __restrict__ gbuf) {
-  const int L = *Lptr;
-  if (L < 0) {
-    log[counters[1]].leaf = -1;  // terminator for the host replay
-    return;
+                           const int64_t* __restrict__ gbuf, float* hist_base,
+                           size_t slot_stride, int n_elem) {
+  __shared__ int s_spare;
+  if (threadIdx.x == 0) {
+    s_spare = -1;
+    const int L = *Lptr;
+    if (L < 0) {
+      log[counters[1]].leaf = -1;  // terminator for the host replay
+    } else {
+      s_spare = counters[0];
+      FinalizeBookkeeping(leaf_begin, leaf_cnt, leaf_slot, stats, winner, L, counters, log,
+                          ctr, gbuf);
+    }
   }
+  __syncthreads();
+  if (s_spare < 0) return;
+  float* hist = hist_base + static_cast<size_t>(s_spare) * slot_stride;
+  for (int i = threadIdx.x; i < n_elem; i += blockDim.x) hist[i] = 0.0f;
+}
+
+/*! single-thread bookkeeping body (called from k_finalize thread 0). */
+__device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_slot,
+                                    LeafStat* stats, const SplitRec* __restrict__ winner,
+                                    int L, int* counters, LogEntry* __restrict__ log,
+                                    const int* __restrict__ ctr,
+                                    const int64_t* __restrict__ gbuf) {
   const int R = counters[0];
   const int spare_slot = R;
   log[counters[1]].rec = *winner;
@@ -867,6 +892,7 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
     leaf_slot[R] = spare_slot;  // L keeps old slot
   }
 }
+
 
 // ------------------------------------------------------------------ boosting kernels
 __global__ void k_grad_binary(const double* __restrict__ score,
@@ -1120,7 +1146,7 @@ class HIPTreeLearner : public TreeLearner {
 
  private:
   void UploadGradients(const score_t* g, const score_t* h);
-  void LaunchHist(const int* leafA_ptr, int leafB_from_counters, int blocks);
+  void LaunchHist(const int* leafA_ptr, int leafB_from_counters, int blocks, bool zero_spare = true);
   void ReduceSpareHist(int spare_slot);
   void LaunchBestSplit(const int* leafA_ptr, int leafB_from_counters);
   int HistBlocksFor(int approx_cnt) const {
@@ -1446,13 +1472,16 @@ void HIPTreeLearner::DeviceBoosting(const ObjectiveFunction* obj) {
 }
 
 void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
-                                int blocks) {
+                                int blocks, bool zero_spare) {
   const int n_elem = total_bins_ * 2;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
-  // zero the spare slot (R = counters[0]-1 after finalize; literal slot 0 for the root)
-  hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
-                     slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
-                     n_elem);
+  // zero the spare slot (literal slot 0 for the root). In the split loop the spare
+  // was already zeroed by the fused k_finalize, so the launch is skipped.
+  if (zero_spare) {
+    hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
+                       slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
+                       n_elem);
+  }
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
     const auto [fb, fe] = feat_partitions_[pr];
     const auto [bin_base, bins] = part_bin_range_[pr];
@@ -1540,9 +1569,10 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                      d_leaf_stats_.ptr, leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
                      feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
                      d_feat_best_.ptr);
-  hipLaunchKernelGGL(hipk::k_best_leaf, dim3(ny), dim3(256), 0, stream_, d_feat_best_.ptr,
-                     nf_, d_leaf_best_.ptr, leafA_ptr, d_counters_.ptr,
-                     leafB_from_counters);
+  (void)ny;
+  hipLaunchKernelGGL(hipk::k_best_leaf_overall, dim3(1), dim3(256), 0, stream_,
+                     d_feat_best_.ptr, nf_, d_leaf_best_.ptr, leafA_ptr, d_counters_.ptr,
+                     leafB_from_counters, d_winner_.ptr, d_winner_leaf_.ptr);
 }
 
 Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, bool) {
@@ -1612,9 +1642,8 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   const int kPartBlocks = 256;
   const int kLoopHistBlocks = 768;
   for (int split_i = 0; split_i < nl - 1; ++split_i) {
-    hipLaunchKernelGGL(hipk::k_best_overall, dim3(1), dim3(256), 0, stream_,
-                       d_leaf_best_.ptr, d_counters_.ptr, d_winner_.ptr,
-                       d_winner_leaf_.ptr);
+    // winner for this split was already selected by the fused k_best_leaf_overall
+    // at the end of the previous split (or of the root best-split pass)
     hipLaunchKernelGGL(hipk::k_part_mark, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
                        d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_winner_leaf_.ptr,
                        d_winner_.ptr, d_feat_meta_.ptr, d_cols_.ptr, num_data_,
@@ -1628,18 +1657,17 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        d_block_loff_.ptr, d_block_roff_.ptr, d_ctr_.ptr);
     hipLaunchKernelGGL(hipk::k_copy_back, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
                        d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
-                       d_winner_leaf_.ptr);
-    hipLaunchKernelGGL(hipk::k_store_left, dim3(1), dim3(1), 0, stream_, d_ctr_.ptr,
-                       d_winner_leaf_.ptr, d_gbuf_.ptr);
+                       d_winner_leaf_.ptr, d_ctr_.ptr, d_gbuf_.ptr);
     if (comm.active()) {
       NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
                             stream_));
     }
-    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
+    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(256), 0, stream_, d_leaf_begin_.ptr,
                        d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
                        d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
-                       d_gbuf_.ptr);
-    LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks);
+                       d_gbuf_.ptr, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
+                       total_bins_ * 2);
+    LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks, /*zero_spare=*/false);
     ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
     {
       const int n_elem = total_bins_ * 2;
