@@ -1406,6 +1406,16 @@ class HIPTreeLearner : public TreeLearner {
   std::vector<int8_t> feat_mask_host_;
 
   static constexpr int kHistBlock = 256;
+  /*! hist-kernel workgroup size (k_hist is blockDim-agnostic; partition kernels
+   *  stay at kHistBlock). MIGBM_HIST_THREADS overrides for experiments. */
+  static int HistThreads() {
+    static int v = [] {
+      const char* e = getenv("MIGBM_HIST_THREADS");
+      int t = e ? atoi(e) : 256;
+      return t < 64 ? 64 : (t > 1024 ? 1024 : t);
+    }();
+    return v;
+  }
   // 64 KB is the no-opt-in workgroup LDS limit; CDNA4 physically has 160 KB per
   // CU and >64 KB dynamic LDS is enabled per-kernel via hipFuncSetAttribute.
   // MIGBM_LDS_BUDGET (bytes) overrides for experiments.
@@ -1481,7 +1491,8 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   {
     int best_copies = -1;
     size_t best_parts = SIZE_MAX;
-    for (int c : {4, 2, 1}) {
+    for (int c : {8, 4, 2, 1}) {
+      if (c == 8 && config_->use_quantized_grad) continue;  // k_hist_q has no <8>
       if (!build_partitions(c)) continue;
       if (feat_partitions_.size() < best_parts) {
         best_parts = feat_partitions_.size();
@@ -1493,7 +1504,8 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     build_partitions(n_copies_);
     if (LdsBudget() > 64 * 1024) {
       // opt the hist kernels into >64 KB dynamic LDS (CDNA4: up to 160 KB per WG)
-      for (const void* k : {reinterpret_cast<const void*>(&hipk::k_hist<4>),
+      for (const void* k : {reinterpret_cast<const void*>(&hipk::k_hist<8>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<4>),
                             reinterpret_cast<const void*>(&hipk::k_hist<2>),
                             reinterpret_cast<const void*>(&hipk::k_hist<1>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<4>),
@@ -1733,8 +1745,16 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
     }
     const size_t lds = static_cast<size_t>(bins) * (2 * n_copies_ + 2) * sizeof(float);
     switch (n_copies_) {
+      case 8:
+        hipLaunchKernelGGL(hipk::k_hist<8>, dim3(blocks), dim3(HistThreads()), lds, stream_,
+                           d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
+                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
+                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
+                           slot_stride);
+        break;
       case 4:
-        hipLaunchKernelGGL(hipk::k_hist<4>, dim3(blocks), dim3(kHistBlock), lds, stream_,
+        hipLaunchKernelGGL(hipk::k_hist<4>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
                            d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
@@ -1742,7 +1762,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
                            slot_stride);
         break;
       case 2:
-        hipLaunchKernelGGL(hipk::k_hist<2>, dim3(blocks), dim3(kHistBlock), lds, stream_,
+        hipLaunchKernelGGL(hipk::k_hist<2>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
                            d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
@@ -1750,7 +1770,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
                            slot_stride);
         break;
       default:
-        hipLaunchKernelGGL(hipk::k_hist<1>, dim3(blocks), dim3(kHistBlock), lds, stream_,
+        hipLaunchKernelGGL(hipk::k_hist<1>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
                            d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
