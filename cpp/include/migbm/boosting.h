@@ -99,6 +99,7 @@ class GBDT {
   int LabelIdx() const { return label_idx_; }
   const std::vector<std::string>& FeatureNames() const { return feature_names_; }
   const std::string& ObjectiveName() const { return objective_name_; }
+  const std::string& LoadedParameter() const { return loaded_parameter_; }
   const Tree* GetTree(int i) const { return models_[i].get(); }
   Tree* GetMutableTree(int i) { return models_[i].get(); }
   double GetLeafValue(int tree_idx, int leaf_idx) const {

@@ -39,6 +39,8 @@ typedef void* BoosterHandle;
 LIGHTGBM_C_EXPORT const char* LGBM_GetLastError();
 LIGHTGBM_C_EXPORT int LGBM_RegisterLogCallback(void (*callback)(const char*));
 LIGHTGBM_C_EXPORT int LGBM_SetMaxThreads(int num_threads);
+LIGHTGBM_C_EXPORT int LGBM_GetMaxThreads(int* out);
+LIGHTGBM_C_EXPORT int LGBM_SetLastError(const char* msg);
 LIGHTGBM_C_EXPORT int LGBM_DumpParamAliases(int64_t buffer_len, int64_t* out_len,
                                             char* out_str);
 LIGHTGBM_C_EXPORT int LGBM_GetSampleCount(int32_t num_total_row, const char* parameters,
@@ -95,6 +97,15 @@ LIGHTGBM_C_EXPORT int LGBM_DatasetPushRowsWithMetadata(
     DatasetHandle dataset, const void* data, int data_type, int32_t nrow, int32_t ncol,
     int32_t start_row, const float* labels, const float* weights,
     const double* init_scores, const int32_t* queries, int32_t tid);
+LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromCSRFunc(void* get_row_funptr, int num_rows,
+                                                    int64_t num_col, const char* parameters,
+                                                    const DatasetHandle reference,
+                                                    DatasetHandle* out);
+LIGHTGBM_C_EXPORT int LGBM_DatasetPushRowsByCSRWithMetadata(
+    DatasetHandle dataset, const void* indptr, int indptr_type, const int32_t* indices,
+    const void* data, int data_type, int64_t nindptr, int64_t nelem, int64_t start_row,
+    const float* labels, const float* weights, const double* init_scores,
+    const int32_t* queries, int32_t tid);
 LIGHTGBM_C_EXPORT int LGBM_DatasetPushRowsByCSR(DatasetHandle dataset, const void* indptr,
                                                 int indptr_type, const int32_t* indices,
                                                 const void* data, int data_type,
@@ -224,6 +235,29 @@ LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMatSingleRow(
     int predict_type, int start_iteration, int num_iteration, const char* parameter,
     int64_t* out_len, double* out_result);
 /*! predict over an array of row pointers (reference c_api.h LGBM_BoosterPredictForMats) */
+LIGHTGBM_C_EXPORT int LGBM_BoosterValidateFeatureNames(BoosterHandle handle,
+                                                       const char** data_names,
+                                                       int data_num_features);
+LIGHTGBM_C_EXPORT int LGBM_BoosterGetLoadedParam(BoosterHandle handle, int64_t buffer_len,
+                                                 int64_t* out_len, char* out_str);
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForCSC(BoosterHandle handle, const void* col_ptr,
+                                                int col_ptr_type, const int32_t* indices,
+                                                const void* data, int data_type,
+                                                int64_t ncol_ptr, int64_t nelem,
+                                                int64_t num_row, int predict_type,
+                                                int start_iteration, int num_iteration,
+                                                const char* parameter, int64_t* out_len,
+                                                double* out_result);
+/*! sparse (CSR) output of SHAP contributions; free with LGBM_BoosterFreePredictSparse */
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictSparseOutput(
+    BoosterHandle handle, const void* indptr, int indptr_type, const int32_t* indices,
+    const void* data, int data_type, int64_t nindptr, int64_t nelem, int64_t num_col,
+    int predict_type, int start_iteration, int num_iteration, const char* parameter,
+    int matrix_type, int64_t* out_len, void** out_indptr, int32_t** out_indices,
+    void** out_data);
+LIGHTGBM_C_EXPORT int LGBM_BoosterFreePredictSparse(void* indptr, int32_t* indices,
+                                                    void* data, int indptr_type,
+                                                    int data_type);
 LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMats(BoosterHandle handle, const void** data,
                                                  int data_type, int32_t nrow, int32_t ncol,
                                                  int predict_type, int start_iteration,

@@ -47,6 +47,33 @@ class Metadata {
 
   std::vector<label_t>& mutable_label() { return label_; }
 
+  /*! streaming-push metadata (reference PushRows*WithMetadata semantics): values
+   *  land positionally; query ids are run-length encoded into boundaries by
+   *  FinalizeStreamedQueries (called from LGBM_DatasetMarkFinished). */
+  void SetStreamedWeights(int64_t start, const float* w, int32_t n) {
+    if (weights_.size() != static_cast<size_t>(num_data_)) weights_.assign(num_data_, 1.0f);
+    for (int32_t i = 0; i < n; ++i) weights_[start + i] = w[i];
+  }
+  void SetStreamedInitScores(int64_t start, const double* s, int32_t n) {
+    if (init_score_.size() < static_cast<size_t>(num_data_))
+      init_score_.assign(num_data_, 0.0);
+    for (int32_t i = 0; i < n; ++i) init_score_[start + i] = s[i];
+  }
+  void SetStreamedQueryIds(int64_t start, const int32_t* q, int32_t n) {
+    if (streamed_query_ids_.size() != static_cast<size_t>(num_data_))
+      streamed_query_ids_.assign(num_data_, 0);
+    for (int32_t i = 0; i < n; ++i) streamed_query_ids_[start + i] = q[i];
+  }
+  void FinalizeStreamedQueries() {
+    if (streamed_query_ids_.empty()) return;
+    query_boundaries_.clear();
+    query_boundaries_.push_back(0);
+    for (data_size_t i = 1; i < num_data_; ++i)
+      if (streamed_query_ids_[i] != streamed_query_ids_[i - 1]) query_boundaries_.push_back(i);
+    query_boundaries_.push_back(num_data_);
+    streamed_query_ids_.clear();
+  }
+
  private:
   data_size_t num_data_ = 0;
   std::vector<label_t> label_;
@@ -55,6 +82,7 @@ class Metadata {
   std::vector<label_t> query_weights_;
   std::vector<double> init_score_;
   std::vector<int32_t> positions_;
+  std::vector<int32_t> streamed_query_ids_;
 };
 
 /*! Dense bin column; uint8 when num_bin<=256 else uint16. */

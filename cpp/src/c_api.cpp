@@ -233,6 +233,17 @@ int LGBM_SetMaxThreads(int num_threads) {
   API_END();
 }
 
+int LGBM_GetMaxThreads(int* out) {
+  API_BEGIN();
+  *out = omp_get_max_threads();
+  API_END();
+}
+
+int LGBM_SetLastError(const char* msg) {
+  migbm::g_last_error = msg != nullptr ? msg : "";
+  return 0;
+}
+
 int LGBM_DumpParamAliases(int64_t buffer_len, int64_t* out_len, char* out_str) {
   API_BEGIN();
   std::stringstream ss;
@@ -483,8 +494,62 @@ int LGBM_DatasetPushRowsWithMetadata(DatasetHandle dataset, const void* data, in
     auto& lab = d->metadata().mutable_label();
     for (int32_t i = 0; i < nrow; ++i) lab[start_row + i] = labels[i];
   }
-  (void)weights; (void)init_scores; (void)queries;  // full streaming metadata: set via
-                                                    // LGBM_DatasetSetField after finish
+  if (weights != nullptr) d->metadata().SetStreamedWeights(start_row, weights, nrow);
+  if (init_scores != nullptr) d->metadata().SetStreamedInitScores(start_row, init_scores, nrow);
+  if (queries != nullptr) d->metadata().SetStreamedQueryIds(start_row, queries, nrow);
+  API_END();
+}
+
+int LGBM_DatasetCreateFromCSRFunc(void* get_row_funptr, int num_rows, int64_t num_col,
+                                  const char* parameters, const DatasetHandle reference,
+                                  DatasetHandle* out) {
+  API_BEGIN();
+  // row provider is a std::function supplying (index, value) pairs per row
+  // (reference c_api.cpp semantics for the SWIG/R ingestion path)
+  using RowFunc = std::function<void(int idx, std::vector<std::pair<int, double>>&)>;
+  RowFunc* get_row = reinterpret_cast<RowFunc*>(get_row_funptr);
+  std::vector<std::vector<std::pair<int, double>>> rows(num_rows);
+  for (int i = 0; i < num_rows; ++i) (*get_row)(i, rows[i]);
+  auto getter = [&rows](data_size_t r, int c) -> double {
+    for (auto& kv : rows[r])
+      if (kv.first == c) return kv.second;
+    return 0.0;
+  };
+  Config cfg;
+  cfg.Set(Config::Str2Map(parameters));
+  if (reference != nullptr) {
+    *out = static_cast<const Dataset*>(reference)
+               ->CreateValid(getter, num_rows).release();
+  } else {
+    auto d = std::make_unique<Dataset>(num_rows);
+    d->ConstructFromMat(getter, num_rows, static_cast<int>(num_col), cfg,
+                        ParseCategoricalFlags(cfg, static_cast<int>(num_col)));
+    *out = d.release();
+  }
+  API_END();
+}
+
+int LGBM_DatasetPushRowsByCSRWithMetadata(DatasetHandle dataset, const void* indptr,
+                                          int indptr_type, const int32_t* indices,
+                                          const void* data, int data_type, int64_t nindptr,
+                                          int64_t nelem, int64_t start_row,
+                                          const float* labels, const float* weights,
+                                          const double* init_scores,
+                                          const int32_t* queries, int32_t /*tid*/) {
+  API_BEGIN();
+  Dataset* d = static_cast<Dataset*>(dataset);
+  const int64_t num_col = d->num_total_features();
+  int rc = LGBM_DatasetPushRowsByCSR(dataset, indptr, indptr_type, indices, data, data_type,
+                                     nindptr, nelem, num_col, start_row);
+  if (rc != 0) return rc;
+  const int32_t nrow = static_cast<int32_t>(nindptr - 1);
+  if (labels != nullptr) {
+    auto& lab = d->metadata().mutable_label();
+    for (int32_t i = 0; i < nrow; ++i) lab[start_row + i] = labels[i];
+  }
+  if (weights != nullptr) d->metadata().SetStreamedWeights(start_row, weights, nrow);
+  if (init_scores != nullptr) d->metadata().SetStreamedInitScores(start_row, init_scores, nrow);
+  if (queries != nullptr) d->metadata().SetStreamedQueryIds(start_row, queries, nrow);
   API_END();
 }
 
@@ -508,8 +573,9 @@ int LGBM_DatasetPushRowsByCSR(DatasetHandle dataset, const void* indptr, int ind
   API_END();
 }
 
-int LGBM_DatasetMarkFinished(DatasetHandle) {
+int LGBM_DatasetMarkFinished(DatasetHandle dataset) {
   API_BEGIN();
+  static_cast<Dataset*>(dataset)->metadata().FinalizeStreamedQueries();
   API_END();
 }
 
@@ -985,6 +1051,54 @@ int LGBM_BoosterGetFeatureNames(BoosterHandle handle, const int len, int* out_le
   API_END();
 }
 
+int LGBM_BoosterValidateFeatureNames(BoosterHandle handle, const char** data_names,
+                                     int data_num_features) {
+  API_BEGIN();
+  const auto& names = static_cast<BoosterWrapper*>(handle)->boosting()->FeatureNames();
+  if (static_cast<int>(names.size()) != data_num_features)
+    Log::Fatal("Expected %d features but data has %d features",
+               static_cast<int>(names.size()), data_num_features);
+  for (int i = 0; i < data_num_features; ++i) {
+    if (names[i] != data_names[i])
+      Log::Fatal("Expected '%s' at position %d but found '%s'", names[i].c_str(), i,
+                 data_names[i]);
+  }
+  API_END();
+}
+
+int LGBM_BoosterGetLoadedParam(BoosterHandle handle, int64_t buffer_len, int64_t* out_len,
+                               char* out_str) {
+  API_BEGIN();
+  // "[key: value]" lines of the model's parameters block rendered as JSON;
+  // fresh (in-memory trained) boosters fall back to their live config echo
+  auto* w = static_cast<BoosterWrapper*>(handle);
+  std::string raw = w->boosting()->LoadedParameter();
+  if (raw.empty()) raw = w->config().SaveHyperParameters();
+  std::stringstream js;
+  js << "{";
+  bool first = true;
+  for (auto& line : Common::Split(raw.c_str(), '\n')) {
+    auto t = Common::Trim(line);
+    if (t.size() < 3 || t.front() != '[' || t.back() != ']') continue;
+    t = t.substr(1, t.size() - 2);
+    auto colon = t.find(':');
+    if (colon == std::string::npos) continue;
+    std::string k = Common::Trim(t.substr(0, colon));
+    std::string v = Common::Trim(t.substr(colon + 1));
+    if (!first) js << ",";
+    first = false;
+    char* endp = nullptr;
+    strtod(v.c_str(), &endp);
+    const bool numeric = !v.empty() && endp != nullptr && *endp == '\0';
+    js << "\"" << k << "\":";
+    if (numeric) js << v;
+    else js << "\"" << v << "\"";
+  }
+  js << "}";
+  return CopyToBuffer(js.str(), buffer_len, out_len, out_str);
+  API_END();
+}
+
 int LGBM_BoosterGetNumFeature(BoosterHandle handle, int* out_len) {
   API_BEGIN();
   *out_len = static_cast<BoosterWrapper*>(handle)->boosting()->MaxFeatureIdx() + 1;
@@ -1094,6 +1208,98 @@ int LGBM_BoosterPredictForCSRSingleRow(BoosterHandle handle, const void* indptr,
   return LGBM_BoosterPredictForCSR(handle, indptr, indptr_type, indices, data, data_type,
                                    nindptr, nelem, num_col, predict_type, start_iteration,
                                    num_iteration, parameter, out_len, out_result);
+}
+
+int LGBM_BoosterPredictForCSC(BoosterHandle handle, const void* col_ptr, int col_ptr_type,
+                              const int32_t* indices, const void* data, int data_type,
+                              int64_t ncol_ptr, int64_t /*nelem*/, int64_t num_row,
+                              int predict_type, int start_iteration, int num_iteration,
+                              const char*, int64_t* out_len, double* out_result) {
+  API_BEGIN();
+  auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
+  auto cp = MakeIndptrGetter(col_ptr, col_ptr_type);
+  auto val = MakeGetter(data, data_type);
+  const int64_t num_col = ncol_ptr - 1;
+  // transpose CSC -> dense row-major once (predict inputs are caller-bounded)
+  std::vector<double> dense(static_cast<size_t>(num_row) * num_col, 0.0);
+  for (int64_t c = 0; c < num_col; ++c)
+    for (int64_t k = cp(c); k < cp(c + 1); ++k)
+      if (indices[k] < num_row)
+        dense[static_cast<size_t>(indices[k]) * num_col + c] = val(k);
+  auto row_getter = [&dense, num_col](int64_t r, double* o) {
+    memcpy(o, dense.data() + static_cast<size_t>(r) * num_col, sizeof(double) * num_col);
+  };
+  PredictRows(b, row_getter, num_row, static_cast<int>(num_col), predict_type,
+              start_iteration, num_iteration, out_result);
+  *out_len = num_row * b->NumPredictOneRow(start_iteration, num_iteration,
+                                           predict_type == C_API_PREDICT_LEAF_INDEX,
+                                           predict_type == C_API_PREDICT_CONTRIB);
+  API_END();
+}
+
+int LGBM_BoosterPredictSparseOutput(BoosterHandle handle, const void* indptr,
+                                    int indptr_type, const int32_t* indices,
+                                    const void* data, int data_type, int64_t nindptr,
+                                    int64_t /*nelem*/, int64_t num_col, int predict_type,
+                                    int start_iteration, int num_iteration, const char*,
+                                    int matrix_type, int64_t* out_len, void** out_indptr,
+                                    int32_t** out_indices, void** out_data) {
+  API_BEGIN();
+  if (predict_type != C_API_PREDICT_CONTRIB)
+    Log::Fatal("PredictSparseOutput only supports contribution (SHAP) prediction");
+  if (matrix_type != 0)
+    Log::Fatal("PredictSparseOutput currently emits CSR only (matrix_type=0)");
+  auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
+  auto ip = MakeIndptrGetter(indptr, indptr_type);
+  auto val = MakeGetter(data, data_type);
+  const int64_t nrow = nindptr - 1;
+  const int per_row = b->NumPredictOneRow(start_iteration, num_iteration, false, true);
+  std::vector<double> dense(static_cast<size_t>(nrow) * per_row);
+  auto row_getter = [&, indices, num_col](int64_t r, double* o) {
+    std::fill(o, o + num_col, 0.0);
+    for (int64_t k = ip(r); k < ip(r + 1); ++k)
+      if (indices[k] < num_col) o[indices[k]] = val(k);
+  };
+  PredictRows(b, row_getter, nrow, static_cast<int>(num_col), predict_type,
+              start_iteration, num_iteration, dense.data());
+  // compact nonzeros into a fresh CSR triple owned by the caller
+  std::vector<int64_t> rp(nrow + 1, 0);
+  for (int64_t r = 0; r < nrow; ++r) {
+    int64_t nz = 0;
+    for (int j = 0; j < per_row; ++j)
+      if (dense[static_cast<size_t>(r) * per_row + j] != 0.0) ++nz;
+    rp[r + 1] = rp[r] + nz;
+  }
+  const int64_t total = rp[nrow];
+  int64_t* o_indptr = new int64_t[nrow + 1];
+  int32_t* o_indices = new int32_t[std::max<int64_t>(total, 1)];
+  double* o_data = new double[std::max<int64_t>(total, 1)];
+  memcpy(o_indptr, rp.data(), sizeof(int64_t) * (nrow + 1));
+  for (int64_t r = 0; r < nrow; ++r) {
+    int64_t w = rp[r];
+    for (int j = 0; j < per_row; ++j) {
+      const double v = dense[static_cast<size_t>(r) * per_row + j];
+      if (v != 0.0) {
+        o_indices[w] = j;
+        o_data[w] = v;
+        ++w;
+      }
+    }
+  }
+  *out_len = total;
+  *out_indptr = o_indptr;
+  *out_indices = o_indices;
+  *out_data = o_data;
+  API_END();
+}
+
+int LGBM_BoosterFreePredictSparse(void* indptr, int32_t* indices, void* data,
+                                  int /*indptr_type*/, int /*data_type*/) {
+  API_BEGIN();
+  delete[] static_cast<int64_t*>(indptr);
+  delete[] indices;
+  delete[] static_cast<double*>(data);
+  API_END();
 }
 
 int LGBM_BoosterPredictForMats(BoosterHandle handle, const void** data, int data_type,
