@@ -258,6 +258,12 @@ LIGHTGBM_C_EXPORT int LGBM_BoosterPredictSparseOutput(
 LIGHTGBM_C_EXPORT int LGBM_BoosterFreePredictSparse(void* indptr, int32_t* indices,
                                                     void* data, int indptr_type,
                                                     int data_type);
+LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForArrow(BoosterHandle handle, int64_t n_chunks,
+                                                  const struct ArrowArray* chunks,
+                                                  const struct ArrowSchema* schema,
+                                                  int predict_type, int start_iteration,
+                                                  int num_iteration, const char* parameter,
+                                                  int64_t* out_len, double* out_result);
 LIGHTGBM_C_EXPORT int LGBM_BoosterPredictForMats(BoosterHandle handle, const void** data,
                                                  int data_type, int32_t nrow, int32_t ncol,
                                                  int predict_type, int start_iteration,

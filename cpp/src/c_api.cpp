@@ -1302,6 +1302,41 @@ int LGBM_BoosterFreePredictSparse(void* indptr, int32_t* indices, void* data,
   API_END();
 }
 
+int LGBM_BoosterPredictForArrow(BoosterHandle handle, int64_t n_chunks,
+                                const struct ArrowArray* chunks,
+                                const struct ArrowSchema* schema, int predict_type,
+                                int start_iteration, int num_iteration, const char*,
+                                int64_t* out_len, double* out_result) {
+  API_BEGIN();
+  auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
+  MIGBM_CHECK(schema->n_children > 0);
+  const int ncol = static_cast<int>(schema->n_children);
+  std::vector<int64_t> chunk_start(n_chunks + 1, 0);
+  std::vector<std::vector<std::function<double(int64_t)>>> getters(n_chunks);
+  for (int64_t c = 0; c < n_chunks; ++c) {
+    chunk_start[c + 1] = chunk_start[c] + chunks[c].length;
+    getters[c].resize(ncol);
+    for (int f = 0; f < ncol; ++f)
+      getters[c][f] = ArrowColGetter(chunks[c].children[f], schema->children[f]->format);
+  }
+  const int64_t nrow = chunk_start[n_chunks];
+  auto row_getter = [&](int64_t r, double* o) {
+    int64_t lo = 0, hi = n_chunks - 1;
+    while (lo < hi) {
+      const int64_t mid = (lo + hi + 1) >> 1;
+      if (r >= chunk_start[mid]) lo = mid;
+      else hi = mid - 1;
+    }
+    for (int f = 0; f < ncol; ++f) o[f] = getters[lo][f](r - chunk_start[lo]);
+  };
+  PredictRows(b, row_getter, nrow, ncol, predict_type, start_iteration, num_iteration,
+              out_result);
+  *out_len = nrow * b->NumPredictOneRow(start_iteration, num_iteration,
+                                        predict_type == C_API_PREDICT_LEAF_INDEX,
+                                        predict_type == C_API_PREDICT_CONTRIB);
+  API_END();
+}
+
 int LGBM_BoosterPredictForMats(BoosterHandle handle, const void** data, int data_type,
                                int32_t nrow, int32_t ncol, int predict_type,
                                int start_iteration, int num_iteration, const char*,

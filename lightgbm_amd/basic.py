@@ -173,8 +173,9 @@ def _is_pyarrow_table(data):
         return False
 
 
-def _create_dataset_from_arrow(table, param_str, ref_handle):
-    """Export record batches via the Arrow C data interface and build the dataset."""
+def _export_arrow_table(table):
+    """Export a pyarrow Table over the Arrow C data interface.
+    Returns (n_chunks, chunks_array, schema) ctypes objects."""
     import pyarrow as pa  # noqa: F401
 
     class _ArrowArray(ctypes.Structure):
@@ -197,14 +198,35 @@ def _create_dataset_from_arrow(table, param_str, ref_handle):
     chunks = (_ArrowArray * len(batches))()
     schema = _ArrowSchema()
     for i, batch in enumerate(batches):
-        # export every array; the schema only once
         if i == 0:
             batch._export_to_c(ctypes.addressof(chunks[i]), ctypes.addressof(schema))
         else:
             batch._export_to_c(ctypes.addressof(chunks[i]))
+    return len(batches), chunks, schema
+
+
+def _create_dataset_from_arrow(table, param_str, ref_handle):
+    """Export record batches via the Arrow C data interface and build the dataset."""
+    import pyarrow as pa  # noqa: F401
+
+    class _ArrowArray(ctypes.Structure):
+        _fields_ = [("length", ctypes.c_int64), ("null_count", ctypes.c_int64),
+                    ("offset", ctypes.c_int64), ("n_buffers", ctypes.c_int64),
+                    ("n_children", ctypes.c_int64), ("buffers", ctypes.c_void_p),
+                    ("children", ctypes.c_void_p), ("dictionary", ctypes.c_void_p),
+                    ("release", ctypes.c_void_p), ("private_data", ctypes.c_void_p)]
+
+    class _ArrowSchema(ctypes.Structure):
+        _fields_ = [("format", ctypes.c_char_p), ("name", ctypes.c_char_p),
+                    ("metadata", ctypes.c_char_p), ("flags", ctypes.c_int64),
+                    ("n_children", ctypes.c_int64), ("children", ctypes.c_void_p),
+                    ("dictionary", ctypes.c_void_p), ("release", ctypes.c_void_p),
+                    ("private_data", ctypes.c_void_p)]
+
+    n_chunks, chunks, schema = _export_arrow_table(table)
     out = ctypes.c_void_p()
     _safe_call(_LIB.LGBM_DatasetCreateFromArrow(
-        ctypes.c_int64(len(batches)), chunks, ctypes.byref(schema), _c_str(param_str),
+        ctypes.c_int64(n_chunks), chunks, ctypes.byref(schema), _c_str(param_str),
         ref_handle, ctypes.byref(out)))
     return out
 
@@ -682,6 +704,27 @@ class Booster:
             ptype = _PREDICT_LEAF
         if pred_contrib:
             ptype = _PREDICT_CONTRIB
+        if _is_pyarrow_table(data):
+            nrow = data.num_rows
+            n = ctypes.c_int64(0)
+            _safe_call(_LIB.LGBM_BoosterCalcNumPredict(
+                self._handle, ctypes.c_int(nrow), ctypes.c_int(ptype),
+                ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+                ctypes.byref(n)))
+            res = np.zeros(n.value, dtype=np.float64)
+            out_len = ctypes.c_int64(0)
+            n_chunks, chunks, schema = _export_arrow_table(data)
+            _safe_call(_LIB.LGBM_BoosterPredictForArrow(
+                self._handle, ctypes.c_int64(n_chunks), chunks, ctypes.byref(schema),
+                ctypes.c_int(ptype), ctypes.c_int(start_iteration),
+                ctypes.c_int(num_iteration), _c_str(""), ctypes.byref(out_len),
+                res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+            per_row = out_len.value // nrow if nrow else 0
+            if per_row > 1:
+                res = res.reshape(nrow, per_row)
+            if pred_leaf:
+                res = res.astype(np.int32)
+            return res
         arr = _to_2d_float64(data)
         nrow, ncol = arr.shape
         n = ctypes.c_int64(0)

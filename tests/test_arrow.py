@@ -45,3 +45,17 @@ def test_arrow_multi_chunk():
     y = np.zeros(1200, dtype=np.float32)
     ds = lgb.Dataset(table, label=y).construct()
     assert ds.num_data() == 1200
+
+
+def test_predict_for_arrow():
+    """Booster.predict accepts a pyarrow Table (LGBM_BoosterPredictForArrow)."""
+    pa = pytest.importorskip("pyarrow")
+    rng = np.random.RandomState(3)
+    X = rng.rand(500, 4)
+    y = (X[:, 0] + X[:, 1] > 1.0).astype(np.float64)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 10)
+    table = pa.table({f"f{i}": X[:, i] for i in range(4)})
+    pred_arrow = bst.predict(table)
+    np.testing.assert_allclose(pred_arrow, bst.predict(X), rtol=1e-12)
+    contrib = bst.predict(table, pred_contrib=True)
+    assert contrib.shape == (500, 5)
