@@ -14,6 +14,7 @@
 
 #include <memory>
 #include <string>
+#include <algorithm>
 #include <vector>
 
 namespace migbm {
@@ -85,27 +86,66 @@ class Metadata {
   std::vector<int32_t> streamed_query_ids_;
 };
 
-/*! Dense bin column; uint8 when num_bin<=256 else uint16. */
+/*! Bin column: dense (uint8 when num_bin<=256 else uint16) or sparse.
+ *  Sparse mode stores (sorted row, bin) pairs for rows whose bin differs from the
+ *  column's default bin — chosen at construction for high-default-fraction
+ *  features (capability parity: reference SparseBin / is_enable_sparse; fresh
+ *  row-list design instead of delta encoding). */
 class BinColumn {
  public:
   void Init(data_size_t n, int num_bin) {
+    is_sparse_ = false;
     is16_ = num_bin > 256;
     if (is16_) d16_.assign(n, 0);
     else d8_.assign(n, 0);
   }
+  void InitSparse(data_size_t n, int num_bin, uint32_t default_bin) {
+    is_sparse_ = true;
+    is16_ = num_bin > 256;
+    num_data_ = n;
+    default_bin_ = default_bin;
+    s_rows_.clear();
+    s_bins_.clear();
+  }
+  /*! dense write, or sparse append (sparse: rows must arrive in ascending order) */
   inline void Set(data_size_t i, uint32_t b) {
+    if (is_sparse_) {
+      if (b == default_bin_) return;
+      s_rows_.push_back(i);
+      s_bins_.push_back(b);
+      return;
+    }
     if (is16_) d16_[i] = static_cast<uint16_t>(b);
     else d8_[i] = static_cast<uint8_t>(b);
   }
-  inline uint32_t Get(data_size_t i) const { return is16_ ? d16_[i] : d8_[i]; }
+  inline uint32_t Get(data_size_t i) const {
+    if (is_sparse_) {
+      auto it = std::lower_bound(s_rows_.begin(), s_rows_.end(), i);
+      if (it != s_rows_.end() && *it == i) return s_bins_[it - s_rows_.begin()];
+      return default_bin_;
+    }
+    return is16_ ? d16_[i] : d8_[i];
+  }
   bool is16() const { return is16_; }
+  bool is_sparse() const { return is_sparse_; }
+  uint32_t default_bin() const { return default_bin_; }
+  data_size_t nnz() const { return static_cast<data_size_t>(s_rows_.size()); }
+  const data_size_t* sparse_rows() const { return s_rows_.data(); }
+  const uint16_t* sparse_bins() const { return s_bins_.data(); }
   const uint8_t* data8() const { return d8_.data(); }
   const uint16_t* data16() const { return d16_.data(); }
+  std::vector<data_size_t>& mutable_sparse_rows() { return s_rows_; }
+  std::vector<uint16_t>& mutable_sparse_bins() { return s_bins_; }
 
  private:
   bool is16_ = false;
+  bool is_sparse_ = false;
+  uint32_t default_bin_ = 0;
+  data_size_t num_data_ = 0;
   std::vector<uint8_t> d8_;
   std::vector<uint16_t> d16_;
+  std::vector<data_size_t> s_rows_;   // ascending
+  std::vector<uint16_t> s_bins_;
 };
 
 class Dataset {
@@ -129,7 +169,26 @@ class Dataset {
   void ConstructHistograms(const std::vector<int8_t>& is_feature_used,
                            const data_size_t* data_indices, data_size_t num_data,
                            const score_t* ordered_grad, const score_t* ordered_hess,
-                           hist_t* hist) const;
+                           hist_t* hist) const {
+    ConstructHistograms(is_feature_used, data_indices, num_data, ordered_grad, ordered_hess,
+                        nullptr, nullptr, nullptr, hist);
+  }
+  /*! Full form: row_grad/row_hess are the FULL per-row gradient arrays and
+   *  in_leaf is a per-row membership byte mask — both required only when the
+   *  dataset has sparse columns and the leaf is a strict subset (the sparse path
+   *  iterates column nonzeros, which are row- not position-indexed). */
+  void ConstructHistograms(const std::vector<int8_t>& is_feature_used,
+                           const data_size_t* data_indices, data_size_t num_data,
+                           const score_t* ordered_grad, const score_t* ordered_hess,
+                           const score_t* row_grad, const score_t* row_hess,
+                           const uint8_t* in_leaf, hist_t* hist) const;
+  bool has_sparse() const { return has_sparse_; }
+  /*! default bin of a sparse feature (the bin MaterializeDefaultBins must
+   *  reconstruct from leaf totals), or -1 for dense features. */
+  int feature_sparse_default_bin(int f) const {
+    const BinColumn& c = columns_[col_of_feature_[f]];
+    return c.is_sparse() ? static_cast<int>(c.default_bin()) : -1;
+  }
 
   /*! Histogram for one inner feature into out[2*bin]. */
   void ConstructHistogramForFeature(int fidx, const data_size_t* data_indices,
@@ -231,6 +290,7 @@ class Dataset {
     feature_bundled_ = other.feature_bundled_;
     column_features_ = other.column_features_;
     has_bundles_ = other.has_bundles_;
+    has_sparse_ = other.has_sparse_;
   }
 
   data_size_t num_data_ = 0;
@@ -253,6 +313,7 @@ class Dataset {
   std::vector<int8_t> feature_bundled_;
   std::vector<std::vector<int>> column_features_;
   bool has_bundles_ = false;
+  bool has_sparse_ = false;
   std::vector<uint32_t> hist_offsets_;
   int num_total_bin_ = 0;
   Metadata metadata_;

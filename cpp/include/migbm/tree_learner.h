@@ -123,25 +123,33 @@ class SerialTreeLearner : public TreeLearner {
   /*! hook after a leaf histogram is built. Base: reconstruct the shared default bin
    *  of EFB-bundled features from the leaf totals (the EFB form of FixHistogram).
    *  Distributed learners reduce first, then call this base. */
-  virtual void OnHistogramReady(int leaf) { MaterializeBundledBin0(leaf); }
-  void MaterializeBundledBin0(int leaf) {
-    if (!train_data_->has_bundles()) return;
+  virtual void OnHistogramReady(int leaf) { MaterializeDefaultBins(leaf); }
+  /*! Reconstruct implicit bins from leaf totals: the shared default bin of
+   *  EFB-bundled features AND the default bin of sparse columns (the migbm
+   *  analogue of the reference's FixHistogram). */
+  void MaterializeDefaultBins(int leaf) {
+    if (!train_data_->has_bundles() && !train_data_->has_sparse()) return;
     hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
     const LeafContext& ctx = leaf_ctx_[leaf];
     const int nf = train_data_->num_features();
     for (int f = 0; f < nf; ++f) {
-      if (!train_data_->feature_bundled(f)) continue;
+      int def_bin = -1;
+      if (train_data_->feature_bundled(f)) def_bin = 0;
+      else def_bin = train_data_->feature_sparse_default_bin(f);
+      if (def_bin < 0) continue;
       hist_t* fh = hist + 2 * train_data_->hist_offset(f);
       double g = 0, h = 0;
       const int nb = train_data_->FeatureNumBin(f);
-      for (int b = 1; b < nb; ++b) {
+      for (int b = 0; b < nb; ++b) {
+        if (b == def_bin) continue;
         g += fh[2 * b];
         h += fh[2 * b + 1];
       }
-      fh[0] = ctx.sum_gradient - g;
-      fh[1] = ctx.sum_hessian - h;
+      fh[2 * def_bin] = ctx.sum_gradient - g;
+      fh[2 * def_bin + 1] = ctx.sum_hessian - h;
     }
   }
+  void MaterializeBundledBin0(int leaf) { MaterializeDefaultBins(leaf); }  // legacy name
   /*! hook to globalize root stats (data-parallel: allreduce). */
   virtual void ReduceRootStats(double* sum_g, double* sum_h, data_size_t* cnt) {
     (void)sum_g; (void)sum_h; (void)cnt;
@@ -167,6 +175,7 @@ class SerialTreeLearner : public TreeLearner {
   const score_t* gradients_ = nullptr;
   const score_t* hessians_ = nullptr;
   std::vector<score_t> ordered_grad_, ordered_hess_;
+  std::vector<uint8_t> in_leaf_mask_;   // sparse-column hist membership scratch
   std::vector<score_t> quant_grad_, quant_hess_;  // CPU quantized-training grids
   uint32_t quant_seed_ = 0x9E3779B9u;
   DataPartition partition_;

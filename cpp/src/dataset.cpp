@@ -125,9 +125,20 @@ void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& ge
     const auto& members = column_features_[col];
     if (members.size() == 1 && !feature_bundled_[members[0]]) {
       const int f = members[0];
-      columns_[col].Init(nrow, bin_mappers_[f]->num_bin());
       const int c = real_feature_index_[f];
       const BinMapper* m = bin_mappers_[f].get();
+      // sparse decision on the binning sample: store only non-default rows when
+      // >=80% of values land in the zero/default bin (reference is_enable_sparse)
+      bool sparse = false;
+      uint32_t def_bin = m->ValueToBin(0.0);
+      if (cfg.is_enable_sparse && nrow >= 1024) {
+        int def_cnt = 0;
+        for (int i = 0; i < ns; ++i)
+          if (m->ValueToBin(get(sample_idx[i], c)) == def_bin) ++def_cnt;
+        sparse = def_cnt >= static_cast<int>(0.8 * ns);
+      }
+      if (sparse) columns_[col].InitSparse(nrow, m->num_bin(), def_bin);
+      else columns_[col].Init(nrow, m->num_bin());
       for (data_size_t i = 0; i < nrow; ++i) columns_[col].Set(i, m->ValueToBin(get(i, c)));
     } else {
       // bundle column: value 0 = every member at its default(zero) bin;
@@ -146,6 +157,7 @@ void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& ge
       }
     }
   }
+  for (const auto& col : columns_) has_sparse_ = has_sparse_ || col.is_sparse();
   if (cfg.linear_tree) {
     raw_values_.resize(nf);
 #pragma omp parallel for schedule(dynamic, 1)
@@ -183,9 +195,12 @@ std::unique_ptr<Dataset> Dataset::CreateValid(
     const auto& members = column_features_[col];
     if (members.size() == 1 && !feature_bundled_[members[0]]) {
       const int f = members[0];
-      d->columns_[col].Init(nrow, d->bin_mappers_[f]->num_bin());
       const int c = real_feature_index_[f];
       const BinMapper* m = d->bin_mappers_[f].get();
+      if (columns_[col].is_sparse())
+        d->columns_[col].InitSparse(nrow, m->num_bin(), columns_[col].default_bin());
+      else
+        d->columns_[col].Init(nrow, m->num_bin());
       for (data_size_t i = 0; i < nrow; ++i)
         d->columns_[col].Set(i, m->ValueToBin(get(i, c)));
     } else {
@@ -371,7 +386,9 @@ void Dataset::ConstructHistogramForFeature(int f, const data_size_t* data_indice
 
 void Dataset::ConstructHistograms(const std::vector<int8_t>& is_feature_used,
                                   const data_size_t* data_indices, data_size_t num_data,
-                                  const score_t* og, const score_t* oh, hist_t* hist) const {
+                                  const score_t* og, const score_t* oh,
+                                  const score_t* row_grad, const score_t* row_hess,
+                                  const uint8_t* in_leaf, hist_t* hist) const {
   if (num_data <= 0) return;
   const int ncols = num_columns();
 #pragma omp parallel for schedule(dynamic)
@@ -383,7 +400,40 @@ void Dataset::ConstructHistograms(const std::vector<int8_t>& is_feature_used,
     for (int f : members)
       std::fill(hist + 2 * hist_offsets_[f],
                 hist + 2 * (hist_offsets_[f] + bin_mappers_[f]->num_bin()), 0.0);
-    if (members.size() == 1 && !feature_bundled_[members[0]]) {
+    if (members.size() == 1 && !feature_bundled_[members[0]] &&
+        columns_[col].is_sparse()) {
+      // sparse column: accumulate the non-default bins only; the default bin is
+      // reconstructed from leaf totals by MaterializeDefaultBins (same mechanism
+      // as the EFB shared-default fix). Two regimes, both deterministic:
+      //   - nonzero scan (needs the in_leaf mask) when nnz is small vs the leaf
+      //   - per-row binary search otherwise
+      const int f = members[0];
+      const BinColumn& bc = columns_[col];
+      hist_t* fh = hist + 2 * hist_offsets_[f];
+      const data_size_t nnz = bc.nnz();
+      const bool whole_data = data_indices == nullptr;  // position == row
+      const bool masked_scan = in_leaf != nullptr && row_grad != nullptr &&
+                               static_cast<int64_t>(nnz) <= static_cast<int64_t>(num_data) * 8;
+      if (whole_data || masked_scan) {
+        const data_size_t* rows = bc.sparse_rows();
+        const uint16_t* bins = bc.sparse_bins();
+        for (data_size_t k = 0; k < nnz; ++k) {
+          const data_size_t r = rows[k];
+          if (!whole_data && !in_leaf[r]) continue;
+          fh[2 * bins[k]] += whole_data ? og[r] : row_grad[r];
+          fh[2 * bins[k] + 1] += whole_data ? oh[r] : row_hess[r];
+        }
+      } else {
+        // subset without a membership mask: per-row lookup (O(cnt log nnz))
+        for (data_size_t i = 0; i < num_data; ++i) {
+          const data_size_t r = data_indices[i];
+          const uint32_t b = bc.Get(r);
+          if (b == bc.default_bin()) continue;
+          fh[2 * b] += og[i];
+          fh[2 * b + 1] += oh[i];
+        }
+      }
+    } else if (members.size() == 1 && !feature_bundled_[members[0]]) {
       const int f = members[0];
       ConstructHistogramForFeature(f, data_indices, num_data, og, oh,
                                    hist + 2 * hist_offsets_[f]);
@@ -518,6 +568,7 @@ std::unique_ptr<Dataset> Dataset::Subset(const data_size_t* indices, data_size_t
     d->columns_[col].Init(n, columns_[col].is16() ? 65536 : 256);
     for (data_size_t i = 0; i < n; ++i) d->columns_[col].Set(i, columns_[col].Get(indices[i]));
   }
+  d->has_sparse_ = false;  // subset columns are materialized dense
   // metadata subset
   d->metadata_.Init(n, false, false);
   std::vector<float> lab(n);
@@ -707,8 +758,23 @@ void Dataset::SaveBinaryFile(const char* filename) const {
   for (auto& col : columns_) {
     uint8_t is16 = col.is16();
     fwrite(&is16, 1, 1, fp);
-    if (is16) fwrite(col.data16(), sizeof(uint16_t), num_data_, fp);
-    else fwrite(col.data8(), sizeof(uint8_t), num_data_, fp);
+    if (col.is_sparse()) {
+      // serialize densified (format unchanged; sparsity is re-decided at load
+      // only for in-memory construction paths)
+      if (is16) {
+        std::vector<uint16_t> buf(num_data_);
+        for (data_size_t i = 0; i < num_data_; ++i) buf[i] = static_cast<uint16_t>(col.Get(i));
+        fwrite(buf.data(), sizeof(uint16_t), num_data_, fp);
+      } else {
+        std::vector<uint8_t> buf(num_data_);
+        for (data_size_t i = 0; i < num_data_; ++i) buf[i] = static_cast<uint8_t>(col.Get(i));
+        fwrite(buf.data(), sizeof(uint8_t), num_data_, fp);
+      }
+    } else if (is16) {
+      fwrite(col.data16(), sizeof(uint16_t), num_data_, fp);
+    } else {
+      fwrite(col.data8(), sizeof(uint8_t), num_data_, fp);
+    }
   }
   // metadata
   uint8_t has_w = metadata_.weights() != nullptr;

@@ -1528,15 +1528,17 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     // feature-major layouts; 288GB HBM makes the unbundled copy cheap)
     std::vector<uint8_t> colbuf(num_data_);
     for (int f = 0; f < nf_; ++f) {
-      if (train_data->feature_bundled(f)) {
+      const auto& col = train_data->column(train_data->feature_column(f));
+      if (train_data->feature_bundled(f) || col.is_sparse()) {
+        // bundled features decode; sparse columns densify (the device layout is
+        // dense row-major + col-major — sparsity is a host-memory concern)
 #pragma omp parallel for schedule(static)
         for (int i = 0; i < num_data_; ++i)
           colbuf[i] = static_cast<uint8_t>(train_data->GetBin(i, f));
         HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_, colbuf.data(),
                          num_data_, hipMemcpyHostToDevice));
       } else {
-        HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_,
-                         train_data->column(train_data->feature_column(f)).data8(),
+        HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_, col.data8(),
                          num_data_, hipMemcpyHostToDevice));
       }
     }

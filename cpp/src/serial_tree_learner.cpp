@@ -170,8 +170,22 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
   }
   hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
   std::fill(hist, hist + 2 * train_data_->num_total_bin(), 0.0);
-  train_data_->ConstructHistograms(is_feature_used_, indices, cnt, ordered_grad_.data(),
-                                   ordered_hess_.data(), hist);
+  if (train_data_->has_sparse()) {
+    // membership mask for the sparse-column nonzero scan; cleared after use so
+    // the buffer never needs a full memset
+    if (in_leaf_mask_.size() != static_cast<size_t>(train_data_->num_data()))
+      in_leaf_mask_.assign(train_data_->num_data(), 0);
+#pragma omp parallel for schedule(static)
+    for (data_size_t i = 0; i < cnt; ++i) in_leaf_mask_[indices[i]] = 1;
+    train_data_->ConstructHistograms(is_feature_used_, indices, cnt, ordered_grad_.data(),
+                                     ordered_hess_.data(), gradients_, hessians_,
+                                     in_leaf_mask_.data(), hist);
+#pragma omp parallel for schedule(static)
+    for (data_size_t i = 0; i < cnt; ++i) in_leaf_mask_[indices[i]] = 0;
+  } else {
+    train_data_->ConstructHistograms(is_feature_used_, indices, cnt, ordered_grad_.data(),
+                                     ordered_hess_.data(), hist);
+  }
 }
 
 void SerialTreeLearner::SubtractHistogram(int /*dst_leaf*/, int parent_slot, int sibling_slot) {

@@ -704,6 +704,35 @@ class Booster:
             ptype = _PREDICT_LEAF
         if pred_contrib:
             ptype = _PREDICT_CONTRIB
+        if SCIPY_INSTALLED and scipy_sparse is not None and scipy_sparse.issparse(data):
+            csr = data.tocsr()
+            nrow = csr.shape[0]
+            n = ctypes.c_int64(0)
+            _safe_call(_LIB.LGBM_BoosterCalcNumPredict(
+                self._handle, ctypes.c_int(nrow), ctypes.c_int(ptype),
+                ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
+                ctypes.byref(n)))
+            res = np.zeros(n.value, dtype=np.float64)
+            out_len = ctypes.c_int64(0)
+            indptr = np.ascontiguousarray(csr.indptr, dtype=np.int32)
+            indices = np.ascontiguousarray(csr.indices, dtype=np.int32)
+            vals = np.ascontiguousarray(csr.data, dtype=np.float64)
+            _safe_call(_LIB.LGBM_BoosterPredictForCSR(
+                self._handle, indptr.ctypes.data_as(ctypes.c_void_p),
+                ctypes.c_int(_DTYPE_I32),
+                indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+                vals.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
+                ctypes.c_int64(len(indptr)), ctypes.c_int64(csr.nnz),
+                ctypes.c_int64(csr.shape[1]), ctypes.c_int(ptype),
+                ctypes.c_int(start_iteration), ctypes.c_int(num_iteration), _c_str(""),
+                ctypes.byref(out_len),
+                res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
+            per_row = out_len.value // nrow if nrow else 0
+            if per_row > 1:
+                res = res.reshape(nrow, per_row)
+            if pred_leaf:
+                res = res.astype(np.int32)
+            return res
         if _is_pyarrow_table(data):
             nrow = data.num_rows
             n = ctypes.c_int64(0)
