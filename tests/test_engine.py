@@ -446,3 +446,111 @@ def test_bagging_by_query():
               valid_sets=[train], valid_names=["train"],
               callbacks=[lgb.record_evaluation(ev)])
     assert ev["train"]["ndcg@5"][-1] > 0.75
+
+
+def test_max_depth_enforced():
+    X, y = _binary_data()
+    bst = lgb.train({"objective": "binary", "max_depth": 3, "num_leaves": 255,
+                     "verbosity": -1}, lgb.Dataset(X, label=y), 10)
+    d = bst.dump_model()
+
+    def depth(node, cur=0):
+        if "leaf_index" in node:
+            return cur
+        return max(depth(node["left_child"], cur + 1),
+                   depth(node["right_child"], cur + 1))
+    for t in d["tree_info"]:
+        assert depth(t["tree_structure"]) <= 3
+
+
+def test_min_gain_to_split_prunes():
+    X, y = _binary_data()
+    free = lgb.train({"objective": "binary", "verbosity": -1},
+                     lgb.Dataset(X, label=y), 10)
+    strict = lgb.train({"objective": "binary", "min_gain_to_split": 50.0,
+                        "verbosity": -1}, lgb.Dataset(X, label=y), 10)
+
+    def count_leaves(d):
+        total = 0
+        for t in d["tree_info"]:
+            total += t["num_leaves"]
+        return total
+    assert count_leaves(strict.dump_model()) < count_leaves(free.dump_model())
+
+
+def test_init_score_shifts_training():
+    X, y = _binary_data(n=3000)
+    base = lgb.train({"objective": "binary", "verbosity": -1},
+                     lgb.Dataset(X, label=y), 5)
+    init = np.full(3000, 4.0)  # strong positive prior
+    shifted = lgb.train({"objective": "binary", "verbosity": -1},
+                        lgb.Dataset(X, label=y, init_score=init), 5)
+    # raw predictions exclude the init score; the boosted part must differ
+    assert not np.allclose(base.predict(X, raw_score=True),
+                           shifted.predict(X, raw_score=True))
+
+
+def test_quantile_objective_hits_quantile():
+    rng = np.random.RandomState(0)
+    X = rng.rand(8000, 3)
+    y = (X[:, 0] * 2 + rng.exponential(1.0, 8000)).astype(np.float32)
+    for alpha in (0.2, 0.8):
+        bst = lgb.train({"objective": "quantile", "alpha": alpha, "verbosity": -1,
+                         "num_leaves": 15}, lgb.Dataset(X, label=y), 80)
+        frac_below = (y <= bst.predict(X)).mean()
+        assert abs(frac_below - alpha) < 0.08
+
+
+def test_categorical_with_missing():
+    rng = np.random.RandomState(1)
+    n = 4000
+    cat = rng.randint(0, 6, n).astype(float)
+    cat[rng.rand(n) < 0.15] = np.nan
+    X = np.column_stack([cat, rng.randn(n)])
+    eff = np.array([2.0, -2.0, 1.0, -1.0, 0.5, -0.5])
+    y = np.where(np.isnan(cat), 0.3, eff[np.nan_to_num(cat).astype(int)])
+    y = (y + 0.2 * rng.randn(n) > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1,
+                     "categorical_feature": [0]}, lgb.Dataset(X, label=y), 30)
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.8
+
+
+def test_refit_changes_values_not_structure():
+    X, y = _binary_data(n=3000, seed=3)
+    bst = lgb.train({"objective": "binary", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 10)
+    X2, y2 = _binary_data(n=3000, seed=99)
+    refitted = bst.refit(X2, y2, decay_rate=0.5)
+    d1, d2 = bst.dump_model(), refitted.dump_model()
+    for t1, t2 in zip(d1["tree_info"], d2["tree_info"]):
+
+        def structure(node):
+            if "leaf_index" in node:
+                return ("leaf",)
+            return (node["split_feature"], round(node["threshold"], 9),
+                    structure(node["left_child"]), structure(node["right_child"]))
+        assert structure(t1["tree_structure"]) == structure(t2["tree_structure"])
+    assert not np.allclose(bst.predict(X), refitted.predict(X))
+
+
+def test_feature_importance_types_differ():
+    X, y = _binary_data()
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    split_imp = bst.feature_importance(importance_type="split")
+    gain_imp = bst.feature_importance(importance_type="gain")
+    assert split_imp.dtype.kind in "iu" or np.allclose(split_imp, split_imp.astype(int))
+    assert gain_imp.sum() > 0
+    # the strongest feature by gain should be one of the true signal features 0..3
+    assert int(np.argmax(gain_imp)) in (0, 1, 2, 3)
+
+
+def test_early_stopping_first_metric_only():
+    X, y = _binary_data()
+    train = lgb.Dataset(X[:4000], label=y[:4000])
+    valid = train.create_valid(X[4000:], label=y[4000:])
+    bst = lgb.train({"objective": "binary", "metric": ["binary_logloss", "auc"],
+                     "first_metric_only": True, "verbosity": -1}, train, 300,
+                    valid_sets=[valid], callbacks=[lgb.early_stopping(5, verbose=False,
+                                                                      first_metric_only=True)])
+    assert 0 < bst.best_iteration < 300
