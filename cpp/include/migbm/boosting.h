@@ -69,6 +69,9 @@ class GBDT {
   void PredictRawEarlyStop(const double* features, double* output, int start_iter,
                            int num_iter, int round_period, double margin_threshold,
                            bool multiclass) const;
+  void ConvertRawToOutput(double* output) const {
+    if (objective_ != nullptr) objective_->ConvertOutput(output, output);
+  }
   void Predict(const double* features, double* output, int start_iter, int num_iter) const;
   void PredictLeafIndex(const double* features, double* output, int start_iter,
                         int num_iter) const;

@@ -168,10 +168,24 @@ class BoosterWrapper {
 
 static void PredictRows(GBDT* b, const std::function<void(int64_t, double*)>& row_getter,
                         int64_t nrow, int ncol, int predict_type, int start_iter, int num_iter,
-                        double* out) {
+                        double* out, const char* parameter = nullptr) {
   const int per_row = b->NumPredictOneRow(start_iter, num_iter,
                                           predict_type == C_API_PREDICT_LEAF_INDEX,
                                           predict_type == C_API_PREDICT_CONTRIB);
+  // prediction early stopping (reference pred_early_stop params): skip remaining
+  // iterations once the margin clears the threshold at a check round
+  bool early_stop = false;
+  int es_freq = 10;
+  double es_margin = 10.0;
+  if (parameter != nullptr && *parameter != '\0') {
+    for (auto& kv : Config::Str2Map(parameter)) {
+      if (kv.first == "pred_early_stop") early_stop = kv.second == "true" || kv.second == "1";
+      else if (kv.first == "pred_early_stop_freq") es_freq = atoi(kv.second.c_str());
+      else if (kv.first == "pred_early_stop_margin") es_margin = atof(kv.second.c_str());
+    }
+  }
+  const bool es_usable = early_stop && (predict_type == C_API_PREDICT_NORMAL ||
+                                        predict_type == C_API_PREDICT_RAW_SCORE);
 #pragma omp parallel
   {
     std::vector<double> features(ncol);
@@ -179,6 +193,12 @@ static void PredictRows(GBDT* b, const std::function<void(int64_t, double*)>& ro
     for (int64_t i = 0; i < nrow; ++i) {
       row_getter(i, features.data());
       double* o = out + i * per_row;
+      if (es_usable) {
+        b->PredictRawEarlyStop(features.data(), o, start_iter, num_iter, es_freq,
+                               es_margin, per_row > 1);
+        if (predict_type == C_API_PREDICT_NORMAL) b->ConvertRawToOutput(o);
+        continue;
+      }
       switch (predict_type) {
         case C_API_PREDICT_NORMAL: b->Predict(features.data(), o, start_iter, num_iter); break;
         case C_API_PREDICT_RAW_SCORE: b->PredictRaw(features.data(), o, start_iter, num_iter); break;
@@ -1140,7 +1160,7 @@ int LGBM_BoosterCalcNumPredict(BoosterHandle handle, int num_row, int predict_ty
 
 int LGBM_BoosterPredictForMat(BoosterHandle handle, const void* data, int data_type,
                               int32_t nrow, int32_t ncol, int is_row_major, int predict_type,
-                              int start_iteration, int num_iteration, const char*,
+                              int start_iteration, int num_iteration, const char* parameter,
                               int64_t* out_len, double* out_result) {
   API_BEGIN();
   auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
@@ -1156,7 +1176,7 @@ int LGBM_BoosterPredictForMat(BoosterHandle handle, const void* data, int data_t
     };
   }
   PredictRows(b, row_getter, nrow, ncol, predict_type, start_iteration, num_iteration,
-              out_result);
+              out_result, parameter);
   *out_len = static_cast<int64_t>(nrow) *
              b->NumPredictOneRow(start_iteration, num_iteration,
                                  predict_type == C_API_PREDICT_LEAF_INDEX,

@@ -697,6 +697,10 @@ class Booster:
                 pred_leaf=False, pred_contrib=False, validate_features=False, **kwargs):
         if num_iteration is None:
             num_iteration = self.best_iteration if self.best_iteration > 0 else -1
+        pred_param = _param_dict_to_str(
+            {k: v for k, v in kwargs.items()
+             if k in ("pred_early_stop", "pred_early_stop_freq",
+                      "pred_early_stop_margin")})
         ptype = _PREDICT_NORMAL
         if raw_score:
             ptype = _PREDICT_RAW
@@ -766,7 +770,7 @@ class Booster:
             self._handle, arr.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
             ctypes.c_int32(nrow), ctypes.c_int32(ncol), ctypes.c_int(1),
             ctypes.c_int(ptype), ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
-            _c_str(""), ctypes.byref(out_len),
+            _c_str(pred_param), ctypes.byref(out_len),
             res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
         per_row = out_len.value // nrow if nrow else 0
         if per_row > 1:

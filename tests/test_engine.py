@@ -554,3 +554,45 @@ def test_early_stopping_first_metric_only():
                     valid_sets=[valid], callbacks=[lgb.early_stopping(5, verbose=False,
                                                                       first_metric_only=True)])
     assert 0 < bst.best_iteration < 300
+
+
+def test_ndcg_matches_manual():
+    """ndcg@k reported by the metric matches a direct computation."""
+    rng = np.random.RandomState(0)
+    rows, labels, groups = [], [], []
+    for q in range(50):
+        nq = rng.randint(6, 15)
+        Xq = rng.randn(nq, 4)
+        rel = rng.randint(0, 3, nq)
+        rows.append(Xq); labels.append(rel); groups.append(nq)
+    X = np.vstack(rows); y = np.concatenate(labels).astype(np.float32)
+    g = np.array(groups, dtype=np.int32)
+    train = lgb.Dataset(X, label=y, group=g)
+    ev = {}
+    bst = lgb.train({"objective": "lambdarank", "metric": "ndcg", "eval_at": [3],
+                     "verbosity": -1}, train, 5, valid_sets=[train],
+                    valid_names=["t"], callbacks=[lgb.record_evaluation(ev)])
+    reported = ev["t"]["ndcg@3"][-1]
+    # manual NDCG@3
+    scores = bst.predict(X)
+    pos = 0
+    total = 0.0
+    for nq in groups:
+        s, rel = scores[pos:pos+nq], y[pos:pos+nq]
+        pos += nq
+        order = np.argsort(-s, kind="stable")
+        dcg = sum((2**rel[order[i]] - 1) / np.log2(i + 2) for i in range(min(3, nq)))
+        ideal = np.sort(rel)[::-1]
+        idcg = sum((2**ideal[i] - 1) / np.log2(i + 2) for i in range(min(3, nq)))
+        total += dcg / idcg if idcg > 0 else 1.0
+    np.testing.assert_allclose(reported, total / len(groups), rtol=1e-6)
+
+
+def test_prediction_early_stopping():
+    X, y = _binary_data(n=2000)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 60)
+    exact = bst.predict(X[:200])
+    fast = bst.predict(X[:200], pred_early_stop=True, pred_early_stop_freq=10,
+                       pred_early_stop_margin=1.5)
+    # labels agree even where probabilities differ slightly
+    assert (((exact > 0.5) == (fast > 0.5)).mean()) > 0.98
