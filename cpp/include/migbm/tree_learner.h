@@ -176,6 +176,9 @@ class SerialTreeLearner : public TreeLearner {
   const score_t* hessians_ = nullptr;
   std::vector<score_t> ordered_grad_, ordered_hess_;
   std::vector<uint8_t> in_leaf_mask_;   // sparse-column hist membership scratch
+  // LOCAL (pre-reduce) leaf gradient totals of the last ComputeHistogram call;
+  // distributed learners use them to materialize default bins before reducing
+  double local_leaf_sum_g_ = 0.0, local_leaf_sum_h_ = 0.0;
   std::vector<score_t> quant_grad_, quant_hess_;  // CPU quantized-training grids
   uint32_t quant_seed_ = 0x9E3779B9u;
   DataPartition partition_;

@@ -182,6 +182,26 @@ class VotingParallelTreeLearner : public SerialTreeLearner {
     const int k = std::min(config_->top_k, nf);
     hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
     const LeafContext& ctx = leaf_ctx_[leaf];
+    // 0. pre-vote: materialize bundled/sparse default bins from LOCAL leaf totals
+    // (the base fix at the end uses GLOBAL totals, valid only after the reduce;
+    // voting on unmaterialized histograms would skew candidates for EFB/sparse
+    // features). Local totals = full-bin sum of any dense unbundled feature.
+    if (train_data_->has_bundles() || train_data_->has_sparse()) {
+      for (int f = 0; f < nf; ++f) {
+        int def_bin = train_data_->feature_bundled(f)
+                          ? 0 : train_data_->feature_sparse_default_bin(f);
+        if (def_bin < 0) continue;
+        hist_t* fh = hist + 2 * train_data_->hist_offset(f);
+        double g = 0, h = 0;
+        for (int b = 0; b < train_data_->FeatureNumBin(f); ++b) {
+          if (b == def_bin) continue;
+          g += fh[2 * b];
+          h += fh[2 * b + 1];
+        }
+        fh[2 * def_bin] = local_leaf_sum_g_ - g;
+        fh[2 * def_bin + 1] = local_leaf_sum_h_ - h;
+      }
+    }
     // 1. local candidate gain per feature
     std::vector<std::pair<double, int>> gains(nf);
 #pragma omp parallel for schedule(static)
@@ -237,7 +257,10 @@ class VotingParallelTreeLearner : public SerialTreeLearner {
                 compact.begin() + off + 2 * train_data_->FeatureNumBin(f), fh);
       off += 2 * train_data_->FeatureNumBin(f);
     }
-    SerialTreeLearner::OnHistogramReady(leaf);  // EFB default-bin reconstruction
+    // voted features' default bins are already globally correct (the reduce summed
+    // per-rank locally-materialized defaults); non-voted features stay local-only
+    // and are never scanned (FindBestSplitForLeaf restricts to voted_mask_), so the
+    // base global fix — which would mix global totals with local bins — is skipped.
   }
 
   void FindBestSplitForLeaf(int leaf, const LeafContext& ctx) override {

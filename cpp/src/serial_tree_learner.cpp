@@ -2,6 +2,7 @@
  *  subtraction. Parity target: reference src/treelearner/serial_tree_learner.cpp (Train,
  *  FindBestSplits, Split) — algorithm re-implemented fresh. */
 #include "migbm/tree_learner.h"
+#include "migbm/network.h"
 #include "migbm/objective.h"
 
 #include <algorithm>
@@ -170,6 +171,17 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
   }
   hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
   std::fill(hist, hist + 2 * train_data_->num_total_bin(), 0.0);
+  if ((train_data_->has_bundles() || train_data_->has_sparse()) &&
+      Network::is_distributed()) {
+    double sg = 0.0, sh = 0.0;
+#pragma omp parallel for schedule(static) reduction(+ : sg, sh)
+    for (data_size_t i = 0; i < cnt; ++i) {
+      sg += ordered_grad_[i];
+      sh += ordered_hess_[i];
+    }
+    local_leaf_sum_g_ = sg;
+    local_leaf_sum_h_ = sh;
+  }
   if (train_data_->has_sparse()) {
     // membership mask for the sparse-column nonzero scan; cleared after use so
     // the buffer never needs a full memset

@@ -105,3 +105,38 @@ def test_voting_parallel_identical_models(tmp_path):
         capture_output=True, text=True, timeout=300, env=env)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "DIST_OK" in r.stdout
+
+
+def test_data_and_voting_parallel_with_sparse_efb(tmp_path):
+    """Distributed learners over mostly-zero data: sparse bin columns (and any EFB
+    bundles) must reconstruct default bins from LOCAL totals before the reduce —
+    per-rank models identical, quality holds."""
+    w = WORKER.replace(
+        "Xref = rng.randn(5000, 6)",
+        "Xref = rng.randn(5000, 6); Xref[np.random.RandomState(1).rand(5000, 6) < 0.9] = 0.0")
+    w = w.replace(
+        "X = rng.randn(8000, 6)",
+        "X = rng.randn(8000, 6); X[np.random.RandomState(2 + rank).rand(8000, 6) < 0.9] = 0.0")
+    w = w.replace(
+        "Xv = rng.randn(4000, 6)",
+        "Xv = rng.randn(4000, 6); Xv[np.random.RandomState(3).rand(4000, 6) < 0.9] = 0.0")
+    w = w.replace('y = (X[:, 0] + 0.5 * X[:, 1] + 0.3 * rng.randn(8000) > 0)',
+                  'y = (X[:, 0] + 0.5 * X[:, 1] + 0.1 * rng.randn(8000) > 0.2)')
+    w = w.replace('yref = (Xref[:, 0] + 0.5 * Xref[:, 1] > 0)',
+                  'yref = (Xref[:, 0] + 0.5 * Xref[:, 1] > 0.2)')
+    w = w.replace('yv = (Xv[:, 0] + 0.5 * Xv[:, 1] > 0)',
+                  'yv = (Xv[:, 0] + 0.5 * Xv[:, 1] > 0.2)')
+    w = w.replace("assert acc > 0.9", "assert acc > 0.8")
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    for port, learner in ((29544, '"tree_learner": "data"'),
+                          (29545, '"tree_learner": "voting", "top_k": 3')):
+        script = tmp_path / f"worker_sp_{port}.py"
+        script.write_text(w.replace('"tree_learner": "data"', learner))
+        r = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+             "--master-port", str(port), str(script), str(REPO)],
+            capture_output=True, text=True, timeout=300, env=env)
+        assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+        assert "DIST_OK" in r.stdout
