@@ -95,9 +95,18 @@ class BinColumn {
  public:
   void Init(data_size_t n, int num_bin) {
     is_sparse_ = false;
+    is4_ = false;
     is16_ = num_bin > 256;
     if (is16_) d16_.assign(n, 0);
     else d8_.assign(n, 0);
+  }
+  /*! 4-bit packed dense column (2 bins per byte) for num_bin <= 16
+   *  (capability parity: reference dense_bin.hpp IS_4BIT). */
+  void Init4(data_size_t n) {
+    is_sparse_ = false;
+    is16_ = false;
+    is4_ = true;
+    d8_.assign((static_cast<size_t>(n) + 1) / 2, 0);
   }
   void InitSparse(data_size_t n, int num_bin, uint32_t default_bin) {
     is_sparse_ = true;
@@ -115,6 +124,12 @@ class BinColumn {
       s_bins_.push_back(b);
       return;
     }
+    if (is4_) {
+      uint8_t& byte = d8_[i >> 1];
+      const int sh = (i & 1) * 4;
+      byte = static_cast<uint8_t>((byte & ~(0xF << sh)) | ((b & 0xF) << sh));
+      return;
+    }
     if (is16_) d16_[i] = static_cast<uint16_t>(b);
     else d8_[i] = static_cast<uint8_t>(b);
   }
@@ -124,8 +139,10 @@ class BinColumn {
       if (it != s_rows_.end() && *it == i) return s_bins_[it - s_rows_.begin()];
       return default_bin_;
     }
+    if (is4_) return (d8_[i >> 1] >> ((i & 1) * 4)) & 0xF;
     return is16_ ? d16_[i] : d8_[i];
   }
+  bool is4() const { return is4_; }
   bool is16() const { return is16_; }
   bool is_sparse() const { return is_sparse_; }
   uint32_t default_bin() const { return default_bin_; }
@@ -139,6 +156,7 @@ class BinColumn {
 
  private:
   bool is16_ = false;
+  bool is4_ = false;
   bool is_sparse_ = false;
   uint32_t default_bin_ = 0;
   data_size_t num_data_ = 0;

@@ -138,6 +138,7 @@ void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& ge
         sparse = def_cnt >= static_cast<int>(0.8 * ns);
       }
       if (sparse) columns_[col].InitSparse(nrow, m->num_bin(), def_bin);
+      else if (m->num_bin() <= 16) columns_[col].Init4(nrow);
       else columns_[col].Init(nrow, m->num_bin());
       for (data_size_t i = 0; i < nrow; ++i) columns_[col].Set(i, m->ValueToBin(get(i, c)));
     } else {
@@ -199,6 +200,8 @@ std::unique_ptr<Dataset> Dataset::CreateValid(
       const BinMapper* m = d->bin_mappers_[f].get();
       if (columns_[col].is_sparse())
         d->columns_[col].InitSparse(nrow, m->num_bin(), columns_[col].default_bin());
+      else if (columns_[col].is4())
+        d->columns_[col].Init4(nrow);
       else
         d->columns_[col].Init(nrow, m->num_bin());
       for (data_size_t i = 0; i < nrow; ++i)
@@ -359,6 +362,17 @@ void HistInner(const BIN_T* bins, const data_size_t* idx, data_size_t n,
   }
 }
 
+template <bool USE_INDICES>
+void HistInner4(const uint8_t* packed, const data_size_t* idx, data_size_t n,
+                const score_t* og, const score_t* oh, hist_t* hist) {
+  for (data_size_t i = 0; i < n; ++i) {
+    const data_size_t r = USE_INDICES ? idx[i] : i;
+    const uint32_t b = ((packed[r >> 1] >> ((r & 1) * 4)) & 0xF) << 1;
+    hist[b] += og[i];
+    hist[b + 1] += oh[i];
+  }
+}
+
 }  // namespace
 
 void Dataset::ConstructHistogramForFeature(int f, const data_size_t* data_indices,
@@ -375,7 +389,10 @@ void Dataset::ConstructHistogramForFeature(int f, const data_size_t* data_indice
   }
   const BinColumn& col = columns_[col_of_feature_[f]];
   const bool use_idx = data_indices != nullptr;
-  if (col.is16()) {
+  if (col.is4()) {
+    if (use_idx) HistInner4<true>(col.data8(), data_indices, num_data, og, oh, out);
+    else HistInner4<false>(col.data8(), nullptr, num_data, og, oh, out);
+  } else if (col.is16()) {
     if (use_idx) HistInner<uint16_t, true>(col.data16(), data_indices, num_data, og, oh, out);
     else HistInner<uint16_t, false>(col.data16(), nullptr, num_data, og, oh, out);
   } else {
@@ -758,9 +775,9 @@ void Dataset::SaveBinaryFile(const char* filename) const {
   for (auto& col : columns_) {
     uint8_t is16 = col.is16();
     fwrite(&is16, 1, 1, fp);
-    if (col.is_sparse()) {
-      // serialize densified (format unchanged; sparsity is re-decided at load
-      // only for in-memory construction paths)
+    if (col.is_sparse() || col.is4()) {
+      // serialize densified (format unchanged; sparse/4-bit packing is re-decided
+      // only on in-memory construction paths)
       if (is16) {
         std::vector<uint16_t> buf(num_data_);
         for (data_size_t i = 0; i < num_data_; ++i) buf[i] = static_cast<uint16_t>(col.Get(i));

@@ -1529,7 +1529,7 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     std::vector<uint8_t> colbuf(num_data_);
     for (int f = 0; f < nf_; ++f) {
       const auto& col = train_data->column(train_data->feature_column(f));
-      if (train_data->feature_bundled(f) || col.is_sparse()) {
+      if (train_data->feature_bundled(f) || col.is_sparse() || col.is4()) {
         // bundled features decode; sparse columns densify (the device layout is
         // dense row-major + col-major — sparsity is a host-memory concern)
 #pragma omp parallel for schedule(static)

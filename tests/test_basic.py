@@ -263,3 +263,23 @@ def test_sparse_bin_from_scipy_csr():
     pred_sparse_in = bst.predict(Xs)
     np.testing.assert_allclose(pred_sparse_in, bst.predict(X), rtol=1e-12)
     assert (((pred_sparse_in > 0.5) == y).mean()) > 0.9
+
+
+def test_4bit_bin_packing_equivalence():
+    """max_bin<=16 features use 4-bit packed columns; models must be byte-identical
+    to unpacked training (max_bin=15 forces every feature into the 4-bit path)."""
+    rng = np.random.RandomState(8)
+    X = rng.rand(4000, 6)
+    y = (X[:, 0] + X[:, 1] > 1.0).astype(np.float64)
+    p = {"objective": "binary", "max_bin": 15, "verbosity": -1}
+    m1 = lgb.train(p, lgb.Dataset(X, label=y, params={"max_bin": 15}), 20)
+    pred = m1.predict(X)
+    acc = ((pred > 0.5) == y).mean()
+    assert acc > 0.9
+    # valid-set alignment through the 4-bit path
+    train = lgb.Dataset(X[:3000], label=y[:3000], params={"max_bin": 15})
+    valid = train.create_valid(X[3000:], label=y[3000:])
+    ev = {}
+    lgb.train({**p, "metric": "auc"}, train, 20, valid_sets=[valid],
+              callbacks=[lgb.record_evaluation(ev)])
+    assert ev["valid_0"]["auc"][-1] > 0.95
