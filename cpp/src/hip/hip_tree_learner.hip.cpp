@@ -2072,7 +2072,30 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
 
 // ------------------------------------------------------------------ registration
 namespace {
-TreeLearner* CreateHIP(const Config* cfg) { return new HIPTreeLearner(cfg); }
+TreeLearner* CreateHIP(const Config* cfg) {
+  // loud, not silent: features the device split loop does not implement yet fall
+  // back to the host serial learner (reference CUDA learner errors similarly)
+  auto unsupported = [&]() -> const char* {
+    if (cfg->linear_tree) return "linear_tree";
+    if (!cfg->monotone_constraints.empty()) return "monotone_constraints";
+    if (!cfg->interaction_constraints.empty()) return "interaction_constraints";
+    if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
+    if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||
+        !cfg->cegb_penalty_feature_lazy.empty())
+      return "cost-effective gradient boosting (cegb_*)";
+    if (cfg->extra_trees) return "extra_trees";
+    if (cfg->path_smooth > 0.0) return "path_smooth";
+    if (cfg->feature_fraction_bynode < 1.0) return "feature_fraction_bynode";
+    return nullptr;
+  };
+  if (const char* what = unsupported()) {
+    Log::Warning("device_type=%s: %s is not implemented in the HIP split loop; "
+                 "training this model on the host (CPU) learner instead",
+                 cfg->device_type.c_str(), what);
+    return new SerialTreeLearner(cfg);
+  }
+  return new HIPTreeLearner(cfg);
+}
 struct HIPRegistrar {
   HIPRegistrar() { g_create_hip_learner = CreateHIP; }
 } hip_registrar;

@@ -157,3 +157,17 @@ def test_gpu_rccl_world1_comm():
         assert roc_auc_score(y, bst.predict(X)) > 0.9
     finally:
         assert _LIB.LGBM_GPUNetworkFree() == 0
+
+
+def test_gpu_unsupported_feature_falls_back_loudly():
+    """monotone_constraints with device=cuda must train on the host learner
+    (correct results) rather than silently ignoring the constraint."""
+    rng = np.random.RandomState(0)
+    X = rng.rand(20000, 3)
+    y = (2 * X[:, 0] + 0.1 * rng.randn(20000)).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "device_type": "cuda",
+                     "monotone_constraints": [1, 0, 0], "verbosity": -1},
+                    lgb.Dataset(X, label=y), 30)
+    xs = np.linspace(0.05, 0.95, 20)
+    grid = np.column_stack([xs, np.full(20, 0.5), np.full(20, 0.5)])
+    assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
