@@ -353,7 +353,9 @@ __global__ void k_hist_subtract(float* __restrict__ hist_base, size_t slot_strid
 // ------------------------------------------------------------------ root setup
 __global__ void k_init_root(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
                             int used_cnt, int64_t* gbuf, int* counters, int* root_leaf,
-                            int* minus1) {
+                            int* minus1, double* leaf_bounds) {
+  leaf_bounds[0] = -1e308;  // monotone output bounds of the root leaf
+  leaf_bounds[1] = 1e308;
   leaf_begin[0] = 0;
   leaf_cnt[0] = used_cnt;
   leaf_slot[0] = 0;
@@ -410,6 +412,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
     const FeatMeta* __restrict__ fm, int nf, const LeafStat* __restrict__ stats,
     const int* __restrict__ leafA_ptr, const int* __restrict__ counters,
     int leafB_from_counters, GainParams p, const int8_t* __restrict__ feat_mask,
+    const int8_t* __restrict__ mono, const double* __restrict__ leaf_bounds,
     SplitRec* __restrict__ out) {
   const int f = blockIdx.x;
   const int which = blockIdx.y;
@@ -446,6 +449,11 @@ __global__ void __launch_bounds__(64) k_best_feat(
   double best_gain = -1e308;
   int best_bin = -1, best_dl = 0;
   double best_lg = 0, best_lh = 0;
+  // monotone constraints: per-leaf output bounds (BasicLeafConstraints, same
+  // clamp-then-score semantics as the CPU oracle) + per-feature direction
+  const int8_t mc = mono != nullptr ? mono[f] : static_cast<int8_t>(0);
+  const double blo = mono != nullptr ? leaf_bounds[2 * leaf] : -1e308;
+  const double bhi = mono != nullptr ? leaf_bounds[2 * leaf + 1] : 1e308;
 
   if (m.is_cat) {
     for (int b = lane; b < m.num_bin; b += 64) {
@@ -496,7 +504,11 @@ __global__ void __launch_bounds__(64) k_best_feat(
         const int rc = num_data - lc;
         if (hl < p.min_hess || lc < p.min_data) continue;
         if (hr < p.min_hess || rc < p.min_data) continue;
-        const double lo = d_leaf_out(gl, hl, p), ro = d_leaf_out(gr, hr, p);
+        double lo = d_leaf_out(gl, hl, p), ro = d_leaf_out(gr, hr, p);
+        lo = fmin(fmax(lo, blo), bhi);
+        ro = fmin(fmax(ro, blo), bhi);
+        if (mc > 0 && lo > ro) continue;
+        if (mc < 0 && lo < ro) continue;
         const double gain = d_gain_out(gl, hl, lo, p) + d_gain_out(gr, hr, ro, p);
         if (gain <= min_gain_shift) continue;
         if (gain > best_gain || (gain == best_gain && b < best_bin)) {
@@ -534,8 +546,13 @@ __global__ void __launch_bounds__(64) k_best_feat(
     rec.left_g = best_lg;
     rec.left_h = best_lh;
     const double rg = sum_g - best_lg, rh = sum_h - best_lh;
-    rec.left_out = d_leaf_out(best_lg, best_lh, p);
-    rec.right_out = d_leaf_out(rg, rh, p);
+    if (m.is_cat) {
+      rec.left_out = d_leaf_out(best_lg, best_lh, p);
+      rec.right_out = d_leaf_out(rg, rh, p);
+    } else {
+      rec.left_out = fmin(fmax(d_leaf_out(best_lg, best_lh, p), blo), bhi);
+      rec.right_out = fmin(fmax(d_leaf_out(rg, rh, p), blo), bhi);
+    }
     rec.left_cnt = static_cast<int>(best_lh * cnt_factor + 0.5);
     rec.right_cnt = num_data - rec.left_cnt;
   }
@@ -1007,7 +1024,8 @@ __global__ void k_copy_back(const uint32_t* __restrict__ tmp_base,
 __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_slot,
                                     LeafStat* stats, const SplitRec* winner, int L,
                                     int* counters, LogEntry* log, const int* ctr,
-                                    const int64_t* gbuf);
+                                    const int64_t* gbuf, const int8_t* mono,
+                                    double* leaf_bounds);
 
 /*! device-side split bookkeeping: segments, stats, slot map, split log. Thread 0
  *  does the bookkeeping; the whole block then zeroes the spare histogram slot the
@@ -1017,7 +1035,8 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
                            const int* __restrict__ Lptr, int* counters,
                            LogEntry* __restrict__ log, const int* __restrict__ ctr,
                            const int64_t* __restrict__ gbuf, float* hist_base,
-                           size_t slot_stride, int n_elem) {
+                           size_t slot_stride, int n_elem,
+                           const int8_t* __restrict__ mono, double* leaf_bounds) {
   __shared__ int s_spare;
   if (threadIdx.x == 0) {
     s_spare = -1;
@@ -1027,7 +1046,7 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
     } else {
       s_spare = counters[0];
       FinalizeBookkeeping(leaf_begin, leaf_cnt, leaf_slot, stats, winner, L, counters, log,
-                          ctr, gbuf);
+                          ctr, gbuf, mono, leaf_bounds);
     }
   }
   __syncthreads();
@@ -1041,7 +1060,8 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
                                     LeafStat* stats, const SplitRec* __restrict__ winner,
                                     int L, int* counters, LogEntry* __restrict__ log,
                                     const int* __restrict__ ctr,
-                                    const int64_t* __restrict__ gbuf) {
+                                    const int64_t* __restrict__ gbuf, const int8_t* mono,
+                                    double* leaf_bounds) {
   const int R = counters[0];
   const int spare_slot = R;
   log[counters[1]].rec = *winner;
@@ -1069,6 +1089,24 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
     leaf_slot[R] = old_slot;
   } else {
     leaf_slot[R] = spare_slot;  // L keeps old slot
+  }
+  if (mono != nullptr) {
+    // BasicLeafConstraints bound propagation (mirrors the host learner): children
+    // inherit the parent bounds; a monotone split pins each side at the midpoint
+    double lo = leaf_bounds[2 * L], hi = leaf_bounds[2 * L + 1];
+    leaf_bounds[2 * R] = lo;
+    leaf_bounds[2 * R + 1] = hi;
+    const int8_t mc = mono[w.feature];
+    if (mc != 0) {
+      const double mid = (w.left_out + w.right_out) / 2.0;
+      if (mc > 0) {
+        leaf_bounds[2 * L + 1] = fmin(hi, mid);
+        leaf_bounds[2 * R] = fmax(lo, mid);
+      } else {
+        leaf_bounds[2 * L] = fmax(lo, mid);
+        leaf_bounds[2 * R + 1] = fmin(hi, mid);
+      }
+    }
   }
 }
 
@@ -1376,6 +1414,8 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<int> d_root_leaf_, d_minus1_;
   DevBuf<hipk::LogEntry> d_split_log_;
   DevBuf<hipk::LeafStat> d_leaf_stats_;
+  DevBuf<double> d_leaf_bounds_;   // [2*num_leaves] monotone output bounds
+  DevBuf<int8_t> d_mono_;          // per inner feature, only when constraints set
   DevBuf<int8_t> d_feat_mask_;
   DevBuf<int> d_leaf_begin_, d_leaf_cnt_, d_leaf_slot_;
   DevBuf<int> d_sorted_begin_;
@@ -1393,6 +1433,7 @@ class HIPTreeLearner : public TreeLearner {
   bool grads_on_device_ = false;
   bool quantized_ = false;
   bool coop_launch_ = false;   // fused cooperative partition kernel available
+  bool use_mono_ = false;      // monotone constraints active (bounds tracked on device)
   int quant_levels_ = 2;
   uint32_t quant_seed_ = 0x9E3779B9u;
   bool weights_present_ = false;
@@ -1592,6 +1633,24 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_minus1_.Alloc(1);
   d_split_log_.Alloc(nl);
   d_leaf_stats_.Alloc(nl);
+  d_leaf_bounds_.Alloc(2 * static_cast<size_t>(nl));
+  use_mono_ = false;
+  {
+    const auto& mcs = config_->monotone_constraints;
+    if (!mcs.empty()) {
+      std::vector<int8_t> mono_host(nf_, 0);
+      for (int f = 0; f < nf_; ++f) {
+        const int orig = train_data->RealFeatureIndex(f);
+        if (orig < static_cast<int>(mcs.size()))
+          mono_host[f] = static_cast<int8_t>(mcs[orig]);
+        use_mono_ = use_mono_ || mono_host[f] != 0;
+      }
+      if (use_mono_) {
+        d_mono_.Alloc(nf_);
+        HIP_OK(hipMemcpy(d_mono_.ptr, mono_host.data(), nf_, hipMemcpyHostToDevice));
+      }
+    }
+  }
   d_feat_mask_.Alloc(nf_);
   d_leaf_begin_.Alloc(nl);
   d_leaf_cnt_.Alloc(nl);
@@ -1804,6 +1863,7 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                      slot_stride, d_leaf_slot_.ptr, d_feat_meta_.ptr, nf_,
                      d_leaf_stats_.ptr, leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
                      feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
+                     use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
                      d_feat_best_.ptr);
   (void)ny;
   hipLaunchKernelGGL(hipk::k_best_leaf_overall, dim3(1), dim3(256), 0, stream_,
@@ -1851,7 +1911,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   hipLaunchKernelGGL(hipk::k_init_root, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
                      d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr,
                      static_cast<int>(used_cnt_), d_gbuf_.ptr, d_counters_.ptr,
-                     d_root_leaf_.ptr, d_minus1_.ptr);
+                     d_root_leaf_.ptr, d_minus1_.ptr, d_leaf_bounds_.ptr);
   {
     const int blocks = std::min(2048, (static_cast<int>(used_cnt_) + 255) / 256);
     hipLaunchKernelGGL(hipk::k_root_sums, dim3(blocks), dim3(256), 0, stream_, d_idx_.ptr,
@@ -1926,7 +1986,8 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
                        d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
                        d_gbuf_.ptr, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
-                       total_bins_ * 2);
+                       total_bins_ * 2, use_mono_ ? d_mono_.ptr : nullptr,
+                       d_leaf_bounds_.ptr);
     LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks, /*zero_spare=*/false);
     ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
     {
@@ -2077,7 +2138,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
     if (cfg->linear_tree) return "linear_tree";
-    if (!cfg->monotone_constraints.empty()) return "monotone_constraints";
     if (!cfg->interaction_constraints.empty()) return "interaction_constraints";
     if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
     if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||

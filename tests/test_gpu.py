@@ -159,15 +159,21 @@ def test_gpu_rccl_world1_comm():
         assert _LIB.LGBM_GPUNetworkFree() == 0
 
 
-def test_gpu_unsupported_feature_falls_back_loudly():
-    """monotone_constraints with device=cuda must train on the host learner
-    (correct results) rather than silently ignoring the constraint."""
+def test_gpu_monotone_constraints():
+    """monotone constraints run IN the device split loop (per-leaf bound array +
+    midpoint propagation in k_finalize); predictions must be globally monotone."""
     rng = np.random.RandomState(0)
     X = rng.rand(20000, 3)
     y = (2 * X[:, 0] + 0.1 * rng.randn(20000)).astype(np.float32)
     bst = lgb.train({"objective": "regression", "device_type": "cuda",
-                     "monotone_constraints": [1, 0, 0], "verbosity": -1},
-                    lgb.Dataset(X, label=y), 30)
-    xs = np.linspace(0.05, 0.95, 20)
-    grid = np.column_stack([xs, np.full(20, 0.5), np.full(20, 0.5)])
-    assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
+                     "monotone_constraints": [1, 0, 0], "num_leaves": 63,
+                     "verbosity": -1}, lgb.Dataset(X, label=y), 40)
+    xs = np.linspace(0.02, 0.98, 30)
+    for other in (0.1, 0.5, 0.9):
+        grid = np.column_stack([xs, np.full(30, other), np.full(30, other)])
+        assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
+    # and the fallback path still exists for a genuinely unimplemented feature
+    bst2 = lgb.train({"objective": "regression", "device_type": "cuda",
+                      "linear_tree": True, "verbosity": -1},
+                     lgb.Dataset(X, label=y), 5)
+    assert np.isfinite(bst2.predict(X[:10])).all()
