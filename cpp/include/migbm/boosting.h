@@ -25,6 +25,9 @@ class SampleStrategy {
   virtual ~SampleStrategy() = default;
   virtual void Bagging(int iter, TreeLearner* learner, score_t* gradients,
                        score_t* hessians) = 0;
+  /*! true when Bagging() reads/modifies the gradient arrays (GOSS): device
+   *  boosting must fall back to host gradients for such strategies. */
+  virtual bool NeedsGradients() const { return false; }
   const std::vector<data_size_t>& bag_indices() const { return bag_indices_; }
   data_size_t bag_cnt() const { return bag_cnt_; }
   bool is_use_subset() const { return false; }

@@ -61,6 +61,7 @@ class GOSSStrategy : public SampleStrategy {
  public:
   GOSSStrategy(const Config* cfg, const Dataset* data, int num_tree_per_iter)
       : cfg_(cfg), data_(data), ntpi_(num_tree_per_iter), rng_(cfg->bagging_seed) {}
+  bool NeedsGradients() const override { return true; }
   void Bagging(int iter, TreeLearner* learner, score_t* gradients,
                score_t* hessians) override {
     (void)iter;
@@ -231,7 +232,11 @@ double GBDT::BoostFromAverage(int class_id, bool update_scores) {
 bool GBDT::TrainOneIter(const score_t* gradients, const score_t* hessians) {
   std::vector<double> init_scores(num_tree_per_iteration_, 0.0);
   const bool dev = tree_learner_->IsHIPLearner();
+  // GOSS reads and rescales gradients on the host, so device-resident boosting
+  // is disabled for gradient-dependent sample strategies (grads are computed on
+  // the host and uploaded by the HIP learner instead)
   const bool dev_obj = dev && objective_ != nullptr &&
+                       !sample_strategy_->NeedsGradients() &&
                        tree_learner_->DeviceObjectiveSupported(objective_->GetName());
   if (gradients == nullptr || hessians == nullptr) {
     for (int c = 0; c < num_tree_per_iteration_; ++c)

@@ -177,3 +177,16 @@ def test_gpu_monotone_constraints():
                       "linear_tree": True, "verbosity": -1},
                      lgb.Dataset(X, label=y), 5)
     assert np.isfinite(bst2.predict(X[:10])).all()
+
+
+def test_gpu_goss():
+    """GOSS on GPU: gradients computed on host (device boosting disabled for
+    gradient-dependent sampling), amplified correctly, trained on device."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(60000, 10)
+    y = (2 * X[:, 0] - X[:, 1] > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "device_type": "cuda",
+                     "data_sample_strategy": "goss", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 25)
+    from sklearn.metrics import roc_auc_score
+    assert roc_auc_score(y, bst.predict(X)) > 0.95
