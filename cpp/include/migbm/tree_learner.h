@@ -52,6 +52,8 @@ class TreeLearner {
   virtual void DeviceAddInitScore(double v) { (void)v; }
   /*! download the device train scores into a host buffer (for metrics / custom obj) */
   virtual void DownloadTrainScore(double* dst) { (void)dst; }
+  /*! push host-modified train scores back to the device (DART drop/renormalize). */
+  virtual void UploadTrainScore(const double* src) { (void)src; }
 
   static TreeLearner* Create(const std::string& learner_type, const std::string& device_type,
                              const Config* config);

@@ -632,6 +632,10 @@ std::vector<int> DART::DroppingTrees() {
 
 bool DART::TrainOneIter(const score_t* gradients, const score_t* hessians) {
   auto dropped = DroppingTrees();
+  // device-resident scores: DART's drop/renormalize surgery happens on the host
+  // copy, synced down before and up after each host-side mutation
+  const bool dev = tree_learner_->IsHIPLearner();
+  if (dev) tree_learner_->DownloadTrainScore(train_score_.data());
   auto apply_tree_all = [&](const Tree& t, int c) {
     // add t's prediction to train and all valid score buffers for class c
     const_cast<Tree&>(t).AddPredictionToScore(
@@ -650,7 +654,9 @@ bool DART::TrainOneIter(const score_t* gradients, const score_t* hessians) {
       apply_tree_all(neg, c);
     }
   }
+  if (dev) tree_learner_->UploadTrainScore(train_score_.data());
   bool stop = GBDT::TrainOneIter(gradients, hessians);
+  if (!stop && dev) tree_learner_->DownloadTrainScore(train_score_.data());
   if (!stop) {
     // normalize: new tree trained against the residual without dropped trees
     const double k = static_cast<double>(dropped.size());
@@ -669,6 +675,7 @@ bool DART::TrainOneIter(const score_t* gradients, const score_t* hessians) {
         apply_tree_all(*old, c);
       }
     }
+    if (dev) tree_learner_->UploadTrainScore(train_score_.data());
   }
   return stop;
 }

@@ -1360,6 +1360,10 @@ class HIPTreeLearner : public TreeLearner {
     HIP_OK(hipStreamSynchronize(stream_));
     HIP_OK(hipMemcpy(dst, d_score_.ptr, sizeof(double) * num_data_, hipMemcpyDeviceToHost));
   }
+  void UploadTrainScore(const double* src) override {
+    HIP_OK(hipStreamSynchronize(stream_));
+    HIP_OK(hipMemcpy(d_score_.ptr, src, sizeof(double) * num_data_, hipMemcpyHostToDevice));
+  }
 
  private:
   void UploadGradients(const score_t* g, const score_t* h);
@@ -2137,6 +2141,8 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // loud, not silent: features the device split loop does not implement yet fall
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
+    if (cfg->num_class > 1) return "multiclass (per-class device score buffers)";
+    if (cfg->boosting == "rf") return "random forest score bookkeeping";
     if (cfg->linear_tree) return "linear_tree";
     if (!cfg->interaction_constraints.empty()) return "interaction_constraints";
     if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";

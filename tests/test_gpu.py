@@ -190,3 +190,19 @@ def test_gpu_goss():
                     lgb.Dataset(X, label=y), 25)
     from sklearn.metrics import roc_auc_score
     assert roc_auc_score(y, bst.predict(X)) > 0.95
+
+
+def test_gpu_dart():
+    """DART on GPU: device scores synced around the host-side drop/renormalize."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(50000, 8)
+    y = (X[:, 0] + 0.5 * X[:, 1] > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "device_type": "cuda", "boosting": "dart",
+                     "drop_rate": 0.3, "verbosity": -1}, lgb.Dataset(X, label=y), 25)
+    from sklearn.metrics import roc_auc_score
+    auc = roc_auc_score(y, bst.predict(X))
+    assert auc > 0.95
+    # cross-check against CPU DART quality (same config)
+    cpu = lgb.train({"objective": "binary", "boosting": "dart", "drop_rate": 0.3,
+                     "verbosity": -1}, lgb.Dataset(X, label=y), 25)
+    assert abs(auc - roc_auc_score(y, cpu.predict(X))) < 0.02
