@@ -87,7 +87,21 @@ struct GainParams {
   double l1, l2, mds;
   double min_hess, min_gain_to_split;
   int min_data;
+  // sync-free per-split sampling (device hash RNG keyed on split counter):
+  float bynode_frac;   // feature_fraction_bynode (1.0 = off)
+  int extra_trees;     // 1 = evaluate one hashed random threshold per feature
+  uint32_t rng_seed;   // per-tree seed component
 };
+
+__device__ __forceinline__ uint32_t d_hash3(uint32_t a, uint32_t b, uint32_t c) {
+  uint32_t x = a * 0x9E3779B1u ^ b * 0x85EBCA77u ^ c * 0xC2B2AE3Du;
+  x ^= x >> 16;
+  x *= 0x7FEB352Du;
+  x ^= x >> 15;
+  x *= 0x846CA68Bu;
+  x ^= x >> 16;
+  return x;
+}
 
 struct FeatMeta {
   int bin_off;
@@ -428,6 +442,17 @@ __global__ void __launch_bounds__(64) k_best_feat(
     rec.feature = f;
   }
   if (feat_mask != nullptr && !feat_mask[f]) return;
+  const int split_idx = counters[1];
+  if (p.bynode_frac < 1.0f) {
+    // deterministic per-(split, child, feature) Bernoulli mask; one hashed anchor
+    // feature per node is always kept so a node can never lose every feature
+    const int anchor = static_cast<int>(
+        d_hash3(p.rng_seed, static_cast<uint32_t>(split_idx * 2 + which), 0xFFFFu) %
+        static_cast<uint32_t>(nf));
+    const uint32_t hv = d_hash3(p.rng_seed, static_cast<uint32_t>(split_idx * 2 + which),
+                                static_cast<uint32_t>(f));
+    if (f != anchor && (hv & 0xFFFFFF) * (1.0f / 16777216.0f) >= p.bynode_frac) return;
+  }
   const LeafStat st = stats[leaf];
   const double sum_g = st.sum_g;
   const double sum_h = st.sum_h;
@@ -494,6 +519,13 @@ __global__ void __launch_bounds__(64) k_best_feat(
       carry_g += __shfl(gb, 63);
       carry_h += __shfl(hb, 63);
       if (b > t_max) continue;
+      if (p.extra_trees) {
+        // extra_trees: only the hashed random threshold of this feature is eligible
+        const int rand_t = static_cast<int>(
+            d_hash3(p.rng_seed ^ 0xA5A5A5A5u, static_cast<uint32_t>(split_idx * 2 + which),
+                    static_cast<uint32_t>(f)) % static_cast<uint32_t>(t_max + 1));
+        if (b != rand_t) continue;
+      }
       const int n_var = has_nan ? 2 : 1;
       for (int v = 0; v < n_var; ++v) {
         const bool ml = v == 1;
@@ -1438,6 +1470,7 @@ class HIPTreeLearner : public TreeLearner {
   bool quantized_ = false;
   bool coop_launch_ = false;   // fused cooperative partition kernel available
   bool use_mono_ = false;      // monotone constraints active (bounds tracked on device)
+  uint32_t bynode_seed_ = 0x1234ABCDu;  // per-tree component of the device sampling hash
   int quant_levels_ = 2;
   uint32_t quant_seed_ = 0x9E3779B9u;
   bool weights_present_ = false;
@@ -1861,6 +1894,9 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.min_hess = config_->min_sum_hessian_in_leaf;
   p.min_gain_to_split = config_->min_gain_to_split;
   p.min_data = config_->min_data_in_leaf;
+  p.bynode_frac = static_cast<float>(config_->feature_fraction_bynode);
+  p.extra_trees = config_->extra_trees ? 1 : 0;
+  p.rng_seed = bynode_seed_;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
   const int ny = leafB_from_counters ? 2 : 1;
   hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_, d_hist_.ptr,
@@ -1877,6 +1913,9 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
 
 Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, bool) {
   const int nl = config_->num_leaves;
+  bynode_seed_ = bynode_seed_ * 1664525u + 1013904223u +
+                 static_cast<uint32_t>(config_->extra_seed +
+                                       config_->feature_fraction_seed * 2654435761u);
   auto tree = std::make_unique<Tree>(nl);
 
   if (!grads_on_device_) UploadGradients(gradients, hessians);
@@ -2149,9 +2188,7 @@ TreeLearner* CreateHIP(const Config* cfg) {
     if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||
         !cfg->cegb_penalty_feature_lazy.empty())
       return "cost-effective gradient boosting (cegb_*)";
-    if (cfg->extra_trees) return "extra_trees";
     if (cfg->path_smooth > 0.0) return "path_smooth";
-    if (cfg->feature_fraction_bynode < 1.0) return "feature_fraction_bynode";
     return nullptr;
   };
   if (const char* what = unsupported()) {

@@ -206,3 +206,23 @@ def test_gpu_dart():
     cpu = lgb.train({"objective": "binary", "boosting": "dart", "drop_rate": 0.3,
                      "verbosity": -1}, lgb.Dataset(X, label=y), 25)
     assert abs(auc - roc_auc_score(y, cpu.predict(X))) < 0.02
+
+
+def test_gpu_extra_trees_and_bynode():
+    """extra_trees + feature_fraction_bynode run in the device loop via the
+    sync-free hash sampler; models train to reasonable quality and differ from
+    the unsampled model."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(60000, 10)
+    y = (2 * X[:, 0] - X[:, 1] + 0.3 * rng.randn(60000) > 0).astype(np.float32)
+    from sklearn.metrics import roc_auc_score
+    base = lgb.train({"objective": "binary", "device_type": "cuda", "verbosity": -1},
+                     lgb.Dataset(X, label=y), 25)
+    et = lgb.train({"objective": "binary", "device_type": "cuda", "extra_trees": True,
+                    "verbosity": -1}, lgb.Dataset(X, label=y), 25)
+    bn = lgb.train({"objective": "binary", "device_type": "cuda",
+                    "feature_fraction_bynode": 0.5, "verbosity": -1},
+                   lgb.Dataset(X, label=y), 25)
+    for m in (et, bn):
+        assert roc_auc_score(y, m.predict(X)) > 0.9
+        assert not np.allclose(m.predict(X[:100]), base.predict(X[:100]))
