@@ -1,6 +1,7 @@
 /*! migbm metrics: regression family, binary (logloss/error/auc/ap), multiclass, ranking
  *  (ndcg@/map@), cross-entropy. Parity target: reference src/metric/*. */
 #include "migbm/metric.h"
+#include "migbm/network.h"
 
 #include <algorithm>
 #include <numeric>
@@ -8,6 +9,18 @@
 namespace migbm {
 
 namespace {
+
+/*! distributed eval: reduce (sum, weight) over every rank's shard so all ranks
+ *  report the metric of the UNION (parity: reference Network::GlobalSync). */
+inline double GlobalAvg(double sum, double w) {
+  if (Network::is_distributed()) {
+    double buf[2] = {sum, w};
+    Network::AllreduceSum(buf, 2);
+    sum = buf[0];
+    w = buf[1];
+  }
+  return sum / std::max(1.0, w);
+}
 
 /*! generic pointwise metric: avg of per-row loss (weighted). */
 class PointwiseMetric : public Metric {
@@ -36,7 +49,16 @@ class PointwiseMetric : public Metric {
       double w = weights_ ? weights_[i] : 1.0;
       sum += w * loss_(label_[i], pred);
     }
-    double r = final_ ? final_(sum, sum_w_) : sum / std::max(1.0, sum_w_);
+    double gsum = sum, gw = sum_w_;
+    if (Network::is_distributed()) {
+      // distributed eval: the metric is over the UNION of rank shards
+      // (parity: reference metrics' Network::GlobalSync of loss sums)
+      double buf[2] = {gsum, gw};
+      Network::AllreduceSum(buf, 2);
+      gsum = buf[0];
+      gw = buf[1];
+    }
+    double r = final_ ? final_(gsum, gw) : gsum / std::max(1.0, gw);
     return {r};
   }
 
@@ -164,7 +186,7 @@ class BinaryLoglossMetric : public Metric {
       double w = weights_ ? weights_[i] : 1.0;
       sum += w * (label_[i] > 0 ? -std::log(p) : -std::log(1.0 - p));
     }
-    return {sum / std::max(1.0, sum_w_)};
+    return {GlobalAvg(sum, sum_w_)};
   }
 
  private:
@@ -206,7 +228,7 @@ class MultiLoglossMetric : public Metric {
       double w = weights_ ? weights_[i] : 1.0;
       sum += -w * std::log(p);
     }
-    return {sum / std::max(1.0, sum_w_)};
+    return {GlobalAvg(sum, sum_w_)};
   }
 
  private:
@@ -246,7 +268,7 @@ class MultiErrorMetric : public Metric {
       double w = weights_ ? weights_[i] : 1.0;
       if (num_better >= top_k_) sum += w;
     }
-    return {sum / std::max(1.0, sum_w_)};
+    return {GlobalAvg(sum, sum_w_)};
   }
 
  private:
@@ -308,7 +330,15 @@ class NDCGMetric : public Metric {
         result[ki] += qw * tmp[ki];
       }
     }
-    for (auto& r : result) r /= std::max(1.0, sum_qw_);
+    double gqw = sum_qw_;
+    if (Network::is_distributed()) {
+      std::vector<double> buf(result);
+      buf.push_back(gqw);
+      Network::AllreduceSum(buf.data(), buf.size());
+      for (size_t i = 0; i < result.size(); ++i) result[i] = buf[i];
+      gqw = buf.back();
+    }
+    for (auto& r : result) r /= std::max(1.0, gqw);
     return result;
   }
 
@@ -360,7 +390,15 @@ class MapMetric : public Metric {
         result[ki] += hits > 0 ? ap / hits : 0.0;
       }
     }
-    for (auto& r : result) r /= std::max<data_size_t>(1, num_queries_);
+    double gq = static_cast<double>(num_queries_);
+    if (Network::is_distributed()) {
+      std::vector<double> buf(result);
+      buf.push_back(gq);
+      Network::AllreduceSum(buf.data(), buf.size());
+      for (size_t i = 0; i < result.size(); ++i) result[i] = buf[i];
+      gq = buf.back();
+    }
+    for (auto& r : result) r /= std::max(1.0, gq);
     return result;
   }
 

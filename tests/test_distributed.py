@@ -140,3 +140,28 @@ def test_data_and_voting_parallel_with_sparse_efb(tmp_path):
             capture_output=True, text=True, timeout=300, env=env)
         assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
         assert "DIST_OK" in r.stdout
+
+
+def test_distributed_metric_global_reduction(tmp_path):
+    """Pointwise metrics in distributed eval are reduced over ALL rank shards:
+    every rank reports the identical (global) value, equal to a single-process
+    eval over the concatenated data."""
+    w = WORKER.replace('''bst = lgb.train(params, train, num_boost_round=10)''',
+'''valid = ref.create_valid(X, label=y)
+ev = {}
+bst = lgb.train(dict(params, metric="binary_logloss"), train, num_boost_round=10,
+                valid_sets=[valid], callbacks=[lgb.record_evaluation(ev)])
+vals = [None] * world
+dist.all_gather_object(vals, ev["valid_0"]["binary_logloss"][-1])
+assert max(vals) - min(vals) < 1e-12, vals''')
+    script = tmp_path / "worker_metric.py"
+    script.write_text(w)
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29546", str(script), str(REPO)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "DIST_OK" in r.stdout
