@@ -552,6 +552,25 @@ class Booster:
     def handle(self):
         return self._handle
 
+    def __getstate__(self):
+        """Pickle as the model text (reference Booster pickling semantics):
+        native handle and dataset references are dropped, the model is
+        reconstructed from its string form on unpickle."""
+        state = self.__dict__.copy()
+        if state.get("_handle") is not None:
+            state["_model_str"] = self.model_to_string(num_iteration=-1)
+        state["_handle"] = None
+        state["_train_set"] = None
+        state["_valid_sets"] = []
+        state["_kept_refs"] = []
+        return state
+
+    def __setstate__(self, state):
+        model_str = state.pop("_model_str", None)
+        self.__dict__.update(state)
+        if model_str is not None:
+            self.model_from_string(model_str)
+
     def __del__(self):
         try:
             if self._handle is not None:

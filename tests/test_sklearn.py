@@ -82,3 +82,18 @@ def test_get_set_params():
     assert p["custom_thing"] == 7
     m.set_params(num_leaves=31)
     assert m.get_params()["num_leaves"] == 31
+
+
+def test_pickle_roundtrip():
+    """Booster and sklearn estimators pickle via model-text reconstruction."""
+    import pickle
+    rng = np.random.RandomState(0)
+    X = rng.randn(2000, 6)
+    y = (X[:, 0] > 0).astype(int)
+    m = lgb.LGBMClassifier(n_estimators=10, verbosity=-1).fit(X, y)
+    m2 = pickle.loads(pickle.dumps(m))
+    np.testing.assert_array_equal(m2.predict(X[:100]), m.predict(X[:100]))
+    np.testing.assert_allclose(m2.predict_proba(X[:100]), m.predict_proba(X[:100]),
+                               rtol=1e-12)
+    b2 = pickle.loads(pickle.dumps(m.booster_))
+    np.testing.assert_allclose(b2.predict(X[:50]), m.booster_.predict(X[:50]), rtol=1e-12)
