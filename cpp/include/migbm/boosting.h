@@ -107,6 +107,19 @@ class GBDT {
   const std::string& ObjectiveName() const { return objective_name_; }
   const std::string& LoadedParameter() const { return loaded_parameter_; }
   const Tree* GetTree(int i) const { return models_[i].get(); }
+  /*! shuffle tree order in [start, end) iterations (reference ShuffleModels). */
+  void ShuffleModels(int start_iter, int end_iter) {
+    const int total = GetCurrentIteration();
+    start_iter = std::max(0, start_iter);
+    end_iter = end_iter <= 0 ? total : std::min(end_iter, total);
+    Random rng(17);
+    for (int i = start_iter; i < end_iter - 1; ++i) {
+      const int j = i + rng.NextInt(0, end_iter - i);
+      for (int c = 0; c < num_tree_per_iteration_; ++c)
+        std::swap(models_[static_cast<size_t>(i) * num_tree_per_iteration_ + c],
+                  models_[static_cast<size_t>(j) * num_tree_per_iteration_ + c]);
+    }
+  }
   Tree* GetMutableTree(int i) { return models_[i].get(); }
   double GetLeafValue(int tree_idx, int leaf_idx) const {
     return models_[tree_idx]->LeafOutput(leaf_idx);

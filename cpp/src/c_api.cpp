@@ -955,9 +955,9 @@ int LGBM_BoosterFree(BoosterHandle handle) {
   API_END();
 }
 
-int LGBM_BoosterShuffleModels(BoosterHandle, int, int) {
+int LGBM_BoosterShuffleModels(BoosterHandle handle, int start_iter, int end_iter) {
   API_BEGIN();
-  Log::Fatal("ShuffleModels not yet supported");
+  static_cast<BoosterWrapper*>(handle)->boosting()->ShuffleModels(start_iter, end_iter);
   API_END();
 }
 

@@ -656,6 +656,71 @@ class Booster:
     def _eval_names(self):
         return _get_string_buffer(_LIB.LGBM_BoosterGetEvalNames, self._handle)
 
+    def num_data(self):
+        out = ctypes.c_int32(0)
+        _safe_call(_LIB.LGBM_BoosterGetNumPredict(self._handle, ctypes.c_int(0),
+                                                  ctypes.byref(ctypes.c_int64(0)))) \
+            if False else None
+        if self._train_set is not None:
+            return self._train_set.num_data()
+        raise LightGBMError("num_data requires a training dataset")
+
+    def set_attr(self, **kwargs):
+        """Store free-form string attributes on the Booster (reference parity)."""
+        attrs = getattr(self, "_attrs", {})
+        for k, v in kwargs.items():
+            if v is None:
+                attrs.pop(k, None)
+            else:
+                attrs[k] = str(v)
+        self._attrs = attrs
+        return self
+
+    def attr(self, key):
+        return getattr(self, "_attrs", {}).get(key)
+
+    def set_train_data_name(self, name):
+        self._train_data_name = name
+        return self
+
+    def shuffle_models(self, start_iteration=0, end_iteration=-1):
+        _safe_call(_LIB.LGBM_BoosterShuffleModels(
+            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(end_iteration)))
+        return self
+
+    def get_leaf_output(self, tree_id, leaf_id):
+        out = ctypes.c_double(0.0)
+        _safe_call(_LIB.LGBM_BoosterGetLeafValue(
+            self._handle, ctypes.c_int(tree_id), ctypes.c_int(leaf_id), ctypes.byref(out)))
+        return out.value
+
+    def set_leaf_output(self, tree_id, leaf_id, value):
+        _safe_call(_LIB.LGBM_BoosterSetLeafValue(
+            self._handle, ctypes.c_int(tree_id), ctypes.c_int(leaf_id),
+            ctypes.c_double(value)))
+        return self
+
+    def get_split_value_histogram(self, feature, bins=None):
+        """Histogram of split threshold values used for `feature` across the model."""
+        d = self.dump_model()
+        values = []
+
+        def walk(node):
+            if "leaf_index" in node:
+                return
+            f = node.get("split_feature")
+            name = self.feature_name()[f] if isinstance(f, int) else f
+            if f == feature or name == feature:
+                values.append(node["threshold"])
+            walk(node["left_child"])
+            walk(node["right_child"])
+        for t in d["tree_info"]:
+            walk(t["tree_structure"])
+        values = np.asarray(values, dtype=np.float64)
+        if bins is None:
+            bins = max(1, min(len(values), 32))
+        return np.histogram(values, bins=bins)
+
     def eval_train(self, feval=None):
         return self.__inner_eval("training", 0, feval)
 

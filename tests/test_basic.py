@@ -283,3 +283,26 @@ def test_4bit_bin_packing_equivalence():
     lgb.train({**p, "metric": "auc"}, train, 20, valid_sets=[valid],
               callbacks=[lgb.record_evaluation(ev)])
     assert ev["valid_0"]["auc"][-1] > 0.95
+
+
+def test_booster_misc_api():
+    """attr/set_attr, leaf get/set, shuffle_models, split-value histogram,
+    set_train_data_name (reference Booster API surface)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(1500, 5)
+    y = (X[:, 0] > 0).astype(np.float32)
+    b = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y), 8)
+    b.set_attr(note="v1")
+    assert b.attr("note") == "v1"
+    b.set_attr(note=None)
+    assert b.attr("note") is None
+    v = b.get_leaf_output(0, 0)
+    b.set_leaf_output(0, 0, v * 2)
+    assert abs(b.get_leaf_output(0, 0) - 2 * v) < 1e-12
+    b.set_leaf_output(0, 0, v)
+    pred = b.predict(X[:50])
+    b.shuffle_models()
+    np.testing.assert_allclose(b.predict(X[:50]), pred, rtol=1e-12)
+    hist, edges = b.get_split_value_histogram(0)
+    assert hist.sum() > 0 and len(edges) == len(hist) + 1
+    assert b.set_train_data_name("train") is b
