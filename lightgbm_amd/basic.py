@@ -363,6 +363,58 @@ class Dataset:
             self._handle, _c_str(name), arr.ctypes.data_as(ctypes.c_void_p),
             ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_F32)))
 
+    def set_field(self, field_name, data):
+        """Generic field setter (reference parity)."""
+        if field_name in ("label", "weight"):
+            self._set_float_field(field_name, data)
+        elif field_name == "group":
+            self.set_group(data)
+        elif field_name == "init_score":
+            self.set_init_score(data)
+        elif field_name == "position":
+            self.set_position(data)
+        else:
+            raise LightGBMError(f"Unknown field {field_name}")
+        return self
+
+    def get_position(self):
+        return self.get_field("position")
+
+    def get_data(self):
+        """Return the raw data this Dataset was built from (if still referenced)."""
+        if getattr(self, "free_raw_data", True) and self._handle is not None \
+                and self.data is None:
+            raise LightGBMError("Raw data was freed (construct with free_raw_data=False)")
+        return self.data
+
+    def get_params(self):
+        return dict(self.params)
+
+    def set_feature_name(self, feature_name):
+        self.feature_name = list(feature_name)
+        return self
+
+    def set_categorical_feature(self, categorical_feature):
+        if self._handle is not None:
+            raise LightGBMError(
+                "set_categorical_feature must be called before construct()")
+        self.categorical_feature = categorical_feature
+        return self
+
+    def set_reference(self, reference):
+        if self._handle is not None:
+            raise LightGBMError("set_reference must be called before construct()")
+        self.reference = reference
+        return self
+
+    def get_ref_chain(self, ref_limit=100):
+        chain = []
+        node = self
+        while node is not None and len(chain) < ref_limit:
+            chain.append(node)
+            node = getattr(node, "reference", None)
+        return set(chain)
+
     def set_label(self, label):
         self.label = label
         if self._handle is not None:

@@ -306,3 +306,24 @@ def test_booster_misc_api():
     hist, edges = b.get_split_value_histogram(0)
     assert hist.sum() > 0 and len(edges) == len(hist) + 1
     assert b.set_train_data_name("train") is b
+
+
+def test_dataset_misc_api():
+    """set_field/get_data/get_params/set_reference/set_categorical_feature/
+    get_ref_chain (reference Dataset API surface)."""
+    rng = np.random.RandomState(0)
+    X = rng.rand(500, 4)
+    y = (X[:, 0] > 0.5).astype(np.float32)
+    ds = lgb.Dataset(X, label=y, free_raw_data=False, params={"max_bin": 31})
+    ds.set_categorical_feature([3])
+    ds.construct()
+    ds.set_field("weight", np.ones(500, dtype=np.float32) * 2)
+    np.testing.assert_allclose(ds.get_field("weight"), 2.0)
+    assert ds.get_data() is X
+    assert ds.get_params()["max_bin"] == 31
+    valid = lgb.Dataset(X[:100], label=y[:100]).set_reference(ds)
+    valid.construct()
+    assert ds in valid.get_ref_chain()
+    import pytest as _pt
+    with _pt.raises(lgb.LightGBMError):
+        ds.set_reference(valid)  # already constructed
