@@ -201,6 +201,13 @@ class Dataset {
                            const score_t* row_grad, const score_t* row_hess,
                            const uint8_t* in_leaf, hist_t* hist) const;
   bool has_sparse() const { return has_sparse_; }
+  /*! Row-wise histogram over the packed row-major view: one pass over the rows,
+   *  per-thread private histograms merged at the end (the migbm analogue of the
+   *  reference's row-wise MultiValBin mode; selection is empirical, see
+   *  SerialTreeLearner::ComputeHistogram). Covers every feature. */
+  void ConstructHistogramsRowWise(const data_size_t* data_indices, data_size_t num_data,
+                                  const score_t* ordered_grad,
+                                  const score_t* ordered_hess, hist_t* hist) const;
   /*! default bin of a sparse feature (the bin MaterializeDefaultBins must
    *  reconstruct from leaf totals), or -1 for dense features. */
   int feature_sparse_default_bin(int f) const {

@@ -178,6 +178,12 @@ class SerialTreeLearner : public TreeLearner {
   const score_t* hessians_ = nullptr;
   std::vector<score_t> ordered_grad_, ordered_hess_;
   std::vector<uint8_t> in_leaf_mask_;   // sparse-column hist membership scratch
+  // empirical col-wise vs row-wise histogram selection (reference
+  // TrainingShareStates behavior): the first two root histograms are timed, one
+  // per mode, then the faster mode is locked in. -1 undecided, 0 col, 1 row.
+  int hist_mode_ = -1;
+  double hist_trial_time_[2] = {0.0, 0.0};
+  int hist_trials_done_ = 0;
   // LOCAL (pre-reduce) leaf gradient totals of the last ComputeHistogram call;
   // distributed learners use them to materialize default bins before reducing
   double local_leaf_sum_g_ = 0.0, local_leaf_sum_h_ = 0.0;

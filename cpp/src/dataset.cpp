@@ -526,6 +526,63 @@ void Dataset::AddFeaturesFrom(const Dataset* other) {
   row_view_ = RowMajorView();
 }
 
+void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
+                                         data_size_t num_data, const score_t* og,
+                                         const score_t* oh, hist_t* hist) const {
+  const RowMajorView& view = GetRowMajorView();
+  const int nf = num_features();
+  const size_t hist_elems = 2 * static_cast<size_t>(num_total_bin_);
+  const int nthreads = omp_get_max_threads();
+  // per-thread private histograms (tiny: num_total_bin*16B each), merged below
+  static thread_local std::vector<hist_t> priv;  // reused across calls
+  std::vector<hist_t*> priv_ptrs(nthreads, nullptr);
+#pragma omp parallel num_threads(nthreads)
+  {
+    const int tid = omp_get_thread_num();
+    priv.assign(hist_elems, 0.0);
+    priv_ptrs[tid] = priv.data();
+    hist_t* h = priv.data();
+    if (!view.is16) {
+      const uint8_t* base = view.data.data();
+      const int stride = view.row_stride;
+#pragma omp for schedule(static)
+      for (data_size_t i = 0; i < num_data; ++i) {
+        const data_size_t r = data_indices ? data_indices[i] : i;
+        const uint8_t* row = base + static_cast<size_t>(r) * stride;
+        const double g = og[i], hv = oh[i];
+        for (int f = 0; f < nf; ++f) {
+          const uint32_t b = (hist_offsets_[f] + row[f]) << 1;
+          h[b] += g;
+          h[b + 1] += hv;
+        }
+      }
+    } else {
+      const uint16_t* base = view.data16.data();
+      const int stride = view.row_stride;
+#pragma omp for schedule(static)
+      for (data_size_t i = 0; i < num_data; ++i) {
+        const data_size_t r = data_indices ? data_indices[i] : i;
+        const uint16_t* row = base + static_cast<size_t>(r) * stride;
+        const double g = og[i], hv = oh[i];
+        for (int f = 0; f < nf; ++f) {
+          const uint32_t b = (hist_offsets_[f] + row[f]) << 1;
+          h[b] += g;
+          h[b + 1] += hv;
+        }
+      }
+    }
+    // parallel merge: each thread owns a slice of the output histogram
+#pragma omp barrier
+#pragma omp for schedule(static)
+    for (int64_t e = 0; e < static_cast<int64_t>(hist_elems); ++e) {
+      double acc = 0.0;
+      for (int t = 0; t < nthreads; ++t)
+        if (priv_ptrs[t] != nullptr) acc += priv_ptrs[t][e];
+      hist[e] = acc;
+    }
+  }
+}
+
 void Dataset::set_feature_names(const std::vector<std::string>& names) {
   feature_names_ = names;
 }

@@ -3,6 +3,8 @@
  *  FindBestSplits, Split) — algorithm re-implemented fresh. */
 #include "migbm/tree_learner.h"
 #include "migbm/network.h"
+
+#include <chrono>
 #include "migbm/objective.h"
 
 #include <algorithm>
@@ -171,6 +173,41 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
   }
   hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
   std::fill(hist, hist + 2 * train_data_->num_total_bin(), 0.0);
+  // ---- histogram mode: empirical col-wise vs row-wise choice, reference
+  // TrainingShareStates-style. Row-wise is only a candidate when every feature
+  // is in play and the dataset is dense (the row view materializes all bins).
+  const bool row_possible = !train_data_->has_sparse() &&
+                            config_->feature_fraction >= 1.0 &&
+                            !config_->force_col_wise &&
+                            train_data_->num_features() >= 2;
+  if (hist_mode_ < 0) {
+    if (config_->force_row_wise && row_possible) hist_mode_ = 1;
+    else if (!row_possible) hist_mode_ = 0;
+  }
+  const bool big_leaf = cnt * 2 >= train_data_->num_data();
+  if (hist_mode_ < 0 && big_leaf) {
+    // trial: time this (root) histogram in the mode not yet measured
+    const int mode = hist_trials_done_;  // 0 = col first, 1 = row second
+    const auto t0 = std::chrono::steady_clock::now();
+    if (mode == 0) {
+      train_data_->ConstructHistograms(is_feature_used_, indices, cnt,
+                                       ordered_grad_.data(), ordered_hess_.data(), hist);
+    } else {
+      train_data_->ConstructHistogramsRowWise(indices, cnt, ordered_grad_.data(),
+                                              ordered_hess_.data(), hist);
+    }
+    hist_trial_time_[mode] =
+        std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+    if (++hist_trials_done_ >= 2)
+      hist_mode_ = hist_trial_time_[1] < hist_trial_time_[0] ? 1 : 0;
+    return;
+  }
+  if (hist_mode_ == 1 || (hist_mode_ < 0 && hist_trials_done_ == 1)) {
+    // locked row-wise, or mid-trial small leaves while row was measured last
+    train_data_->ConstructHistogramsRowWise(indices, cnt, ordered_grad_.data(),
+                                            ordered_hess_.data(), hist);
+    return;
+  }
   if ((train_data_->has_bundles() || train_data_->has_sparse()) &&
       Network::is_distributed()) {
     double sg = 0.0, sh = 0.0;

@@ -596,3 +596,18 @@ def test_prediction_early_stopping():
                        pred_early_stop_margin=1.5)
     # labels agree even where probabilities differ slightly
     assert (((exact > 0.5) == (fast > 0.5)).mean()) > 0.98
+
+
+def test_hist_mode_forced_equivalence():
+    """force_col_wise and force_row_wise must train models of equal quality
+    (different summation order, same splits on well-separated data)."""
+    X, y = _binary_data(n=6000)
+    col = lgb.train({"objective": "binary", "force_col_wise": True, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 20)
+    row = lgb.train({"objective": "binary", "force_row_wise": True, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 20)
+    acc_c = ((col.predict(X) > 0.5) == y).mean()
+    acc_r = ((row.predict(X) > 0.5) == y).mean()
+    assert acc_c > 0.9 and acc_r > 0.9
+    np.testing.assert_allclose(col.predict(X[:200]), row.predict(X[:200]),
+                               rtol=1e-6, atol=1e-6)
