@@ -709,10 +709,6 @@ class Booster:
         return _get_string_buffer(_LIB.LGBM_BoosterGetEvalNames, self._handle)
 
     def num_data(self):
-        out = ctypes.c_int32(0)
-        _safe_call(_LIB.LGBM_BoosterGetNumPredict(self._handle, ctypes.c_int(0),
-                                                  ctypes.byref(ctypes.c_int64(0)))) \
-            if False else None
         if self._train_set is not None:
             return self._train_set.num_data()
         raise LightGBMError("num_data requires a training dataset")
