@@ -1529,6 +1529,14 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   for (int f = 0; f < nf_; ++f) {
     const BinMapper* m = train_data->FeatureBinMapper(f);
     if (m->num_bin() > 256) Log::Fatal("HIP learner currently supports max_bin<=255");
+    if (m->bin_type() == BinType::kCategorical &&
+        m->num_bin() > config_->max_cat_to_onehot) {
+      Log::Warning("HIP learner evaluates categorical feature %d with one-hot splits "
+                   "only (%d categories > max_cat_to_onehot=%d; the CPU learner's "
+                   "sorted-subset scan is not on the device yet)",
+                   train_data->RealFeatureIndex(f), m->num_bin(),
+                   config_->max_cat_to_onehot);
+    }
     feat_meta_host_[f] = {static_cast<int>(train_data->hist_offset(f)), m->num_bin(),
                           m->num_numeric_bin(), m->nan_bin(),
                           m->bin_type() == BinType::kCategorical ? 1 : 0};
