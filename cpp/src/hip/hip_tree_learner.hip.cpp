@@ -64,6 +64,7 @@ struct SplitRec {
   double gain;
   double left_g, left_h;
   double left_out, right_out;
+  unsigned long long cat_mask;  // bin-level subset (categorical sorted scan); 0 = one-hot/numeric
   int left_cnt, right_cnt;  // hessian-derived approx (host launch-sizing hints)
   int feature;
   int bin;
@@ -91,6 +92,9 @@ struct GainParams {
   float bynode_frac;   // feature_fraction_bynode (1.0 = off)
   int extra_trees;     // 1 = evaluate one hashed random threshold per feature
   uint32_t rng_seed;   // per-tree seed component
+  // categorical scan (parity: FindBestThresholdCategorical CPU oracle)
+  double cat_l2, cat_smooth;
+  int max_cat_to_onehot, max_cat_threshold;
 };
 
 __device__ __forceinline__ uint32_t d_hash3(uint32_t a, uint32_t b, uint32_t c) {
@@ -125,6 +129,26 @@ __device__ __forceinline__ double d_gain_out(double g, double h, double out,
                                              const GainParams& p) {
   double s = d_thl1(g, p.l1);
   return -(2.0 * s * out + (h + p.l2) * out * out);
+}
+__device__ __forceinline__ double d_leaf_out_l2(double g, double h, const GainParams& p,
+                                                double l2v) {
+  double r = -d_thl1(g, p.l1) / (h + l2v);
+  if (p.mds > 0.0 && fabs(r) > p.mds) r = r > 0 ? p.mds : -p.mds;
+  return r;
+}
+__device__ __forceinline__ double d_gain_out_l2(double g, double h, double out,
+                                                const GainParams& p, double l2v) {
+  double s = d_thl1(g, p.l1);
+  return -(2.0 * s * out + (h + l2v) * out * out);
+}
+__device__ __forceinline__ double d_split_gain_l2(double gl, double hl, double gr, double hr,
+                                                  const GainParams& p, double l2v) {
+  if (p.mds <= 0.0) {
+    const double sl = d_thl1(gl, p.l1), sr = d_thl1(gr, p.l1);
+    return sl * sl / (hl + l2v) + sr * sr / (hr + l2v);
+  }
+  const double lo = d_leaf_out_l2(gl, hl, p, l2v), ro = d_leaf_out_l2(gr, hr, p, l2v);
+  return d_gain_out_l2(gl, hl, lo, p, l2v) + d_gain_out_l2(gr, hr, ro, p, l2v);
 }
 __device__ __forceinline__ double d_leaf_gain(double g, double h, const GainParams& p) {
   if (p.mds <= 0.0) {
@@ -440,6 +464,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
   if (lane == 0) {
     rec.valid = 0;
     rec.feature = f;
+    rec.cat_mask = 0;
   }
   if (feat_mask != nullptr && !feat_mask[f]) return;
   const int split_idx = counters[1];
@@ -481,14 +506,138 @@ __global__ void __launch_bounds__(64) k_best_feat(
   const double bhi = mono != nullptr ? leaf_bounds[2 * leaf + 1] : 1e308;
 
   if (m.is_cat) {
+    const double l2c = p.l2 + p.cat_l2;
+    if (m.num_bin > p.max_cat_to_onehot && m.num_bin <= 64) {
+      // sorted-subset scan (CPU-oracle parity: FindBestThresholdCategorical).
+      // One wave: lane = bin; bitonic sort by grad/(hess+cat_smooth); prefix
+      // scans give the k-subset sums from both ends; the winner's bin subset is
+      // reconstructed as a 64-bit mask carried in SplitRec.cat_mask.
+      int orig = lane;
+      double gb = 0.0, hb = 0.0, ratio = 1e300;
+      bool elig = false;
+      if (lane < m.num_bin) {
+        gb = fh[2 * lane];
+        hb = fh[2 * lane + 1];
+        elig = hb * cnt_factor >= p.cat_smooth;
+        ratio = elig ? gb / (hb + p.cat_smooth) : 1e300;
+      }
+      // wave64 bitonic sort ascending by (ratio, orig)
+      for (int k2 = 2; k2 <= 64; k2 <<= 1) {
+        for (int j = k2 >> 1; j > 0; j >>= 1) {
+          const double r2 = __shfl_xor(ratio, j);
+          const double g2 = __shfl_xor(gb, j);
+          const double h2 = __shfl_xor(hb, j);
+          const int b2 = __shfl_xor(orig, j);
+          const bool lower = (lane & j) == 0;
+          const bool asc = (lane & k2) == 0;
+          const bool other_lt = r2 < ratio || (r2 == ratio && b2 < orig);
+          const bool take = lower ? (other_lt == asc) : (other_lt != asc);
+          if (take) {
+            ratio = r2;
+            gb = g2;
+            hb = h2;
+            orig = b2;
+          }
+        }
+      }
+      const uint64_t elig_ballot = __ballot(ratio < 1e300);
+      const int n_elig = __popcll(elig_ballot);
+      if (n_elig >= 2) {
+        // inclusive prefix sums over the sorted order
+        double pg = gb, ph = hb;
+        for (int d = 1; d < 64; d <<= 1) {
+          const double tg = __shfl_up(pg, d);
+          const double th = __shfl_up(ph, d);
+          if (lane >= d) {
+            pg += tg;
+            ph += th;
+          }
+        }
+        const int limit = min(p.max_cat_threshold, n_elig - 1);
+        const double tot_g = __shfl(pg, n_elig - 1);
+        const double tot_h = __shfl(ph, n_elig - 1);
+        double bg_cat = -1e308;
+        int b_dir = 0;
+        double b_lg = 0, b_lh = 0;
+        // dir-1 ("from the high end") prefix lookups, hoisted so every lane
+        // participates in the shuffles (no divergent-source reads)
+        const int lo_idx_raw = n_elig - 2 - lane;
+        const double pre_g = __shfl(pg, lo_idx_raw < 0 ? 0 : lo_idx_raw);
+        const double pre_h = __shfl(ph, lo_idx_raw < 0 ? 0 : lo_idx_raw);
+        // lane i evaluates subset size k = i+1 for both directions
+        if (lane < limit) {
+          for (int dir = 0; dir < 2; ++dir) {
+            double sgl, shl;
+            if (dir == 0) {
+              sgl = pg;
+              shl = ph;
+            } else {
+              if (lo_idx_raw < 0) continue;
+              sgl = tot_g - pre_g;
+              shl = tot_h - pre_h;
+            }
+            const double sgr = sum_g - sgl, shr = sum_h - shl;
+            const int lc = static_cast<int>(shl * cnt_factor + 0.5);
+            const int rc = num_data - lc;
+            if (shl < p.min_hess || lc < p.min_data) continue;
+            if (shr < p.min_hess || rc < p.min_data) continue;
+            const double gain = d_split_gain_l2(sgl, shl, sgr, shr, p, l2c);
+            if (gain <= min_gain_shift) continue;
+            if (gain > bg_cat) {
+              bg_cat = gain;
+              b_dir = dir;
+              b_lg = sgl;
+              b_lh = shl;
+            }
+          }
+        }
+        // wave argmax over lanes (tie: smaller subset wins)
+        int b_k = lane;
+        for (int d = 32; d > 0; d >>= 1) {
+          const double og = __shfl_xor(bg_cat, d);
+          const int ok = __shfl_xor(b_k, d);
+          const int od = __shfl_xor(b_dir, d);
+          const double olg = __shfl_xor(b_lg, d);
+          const double olh = __shfl_xor(b_lh, d);
+          if (og > bg_cat || (og == bg_cat && ok < b_k)) {
+            bg_cat = og;
+            b_k = ok;
+            b_dir = od;
+            b_lg = olg;
+            b_lh = olh;
+          }
+        }
+        if (bg_cat > -1e307) {
+          // membership mask over ORIGINAL bins for the winning subset
+          const bool member = b_dir == 0 ? (lane <= b_k)
+                                         : (lane >= n_elig - 1 - b_k && lane < n_elig);
+          uint64_t mask = member && lane < n_elig ? (1ull << orig) : 0ull;
+          for (int d = 32; d > 0; d >>= 1) mask |= __shfl_xor(mask, d);
+          if (lane == 0) {
+            rec.valid = 1;
+            rec.gain = bg_cat - min_gain_shift + p.min_gain_to_split;
+            rec.feature = f;
+            rec.bin = b_k + 1;  // #cats on the left (CPU threshold semantics)
+            rec.default_left = 0;
+            rec.cat_mask = mask;
+            rec.left_g = b_lg;
+            rec.left_h = b_lh;
+            rec.left_out = d_leaf_out_l2(b_lg, b_lh, p, l2c);
+            rec.right_out = d_leaf_out_l2(sum_g - b_lg, sum_h - b_lh, p, l2c);
+            rec.left_cnt = static_cast<int>(b_lh * cnt_factor + 0.5);
+            rec.right_cnt = num_data - rec.left_cnt;
+          }
+        }
+      }
+      return;
+    }
     for (int b = lane; b < m.num_bin; b += 64) {
       const double gl = fh[2 * b], hl = fh[2 * b + 1];
       const double gr = sum_g - gl, hr = sum_h - hl;
       const int lc = static_cast<int>(hl * cnt_factor + 0.5);
       const int rc = num_data - lc;
       if (hl < p.min_hess || lc < p.min_data || hr < p.min_hess || rc < p.min_data) continue;
-      const double lo = d_leaf_out(gl, hl, p), ro = d_leaf_out(gr, hr, p);
-      const double gain = d_gain_out(gl, hl, lo, p) + d_gain_out(gr, hr, ro, p);
+      const double gain = d_split_gain_l2(gl, hl, gr, hr, p, l2c);
       if (gain <= min_gain_shift) continue;
       if (gain > best_gain || (gain == best_gain && b < best_bin)) {
         best_gain = gain;
@@ -677,8 +826,11 @@ __global__ void k_best_leaf_overall(const SplitRec* __restrict__ feat_best, int 
  *  blocks round-robined over the grid). Three phases: mark+count per block, block
  *  offset scan, ranked scatter. No same-address global atomics. */
 __device__ __forceinline__ int part_decide(int b, int thr_bin, int nan_bin, int default_left,
-                                           int cat_onehot) {
-  if (cat_onehot) return b == thr_bin ? 1 : 0;
+                                           int is_cat, unsigned long long cat_mask) {
+  if (is_cat) {
+    if (cat_mask != 0) return (cat_mask >> b) & 1ull;  // sorted-subset split
+    return b == thr_bin ? 1 : 0;                       // one-hot split
+  }
   if (nan_bin >= 0 && b == nan_bin) return default_left;
   return b <= thr_bin ? 1 : 0;
 }
@@ -700,6 +852,7 @@ __global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
   const int nan_bin = m.is_cat ? -1 : m.nan_bin;
   const int default_left = win->default_left;
   const int cat_onehot = m.is_cat;
+  const unsigned long long cmask = win->cat_mask;
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const uint32_t* idx = idx_base + begin;
@@ -707,7 +860,7 @@ __global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
   int local = 0;
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt; i += chunk_stride) {
     const int b = colbins2[idx[i]];
-    const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot);
+    const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot, cmask);
     marks[i] = static_cast<uint8_t>(go);
     local += go;
   }
@@ -899,11 +1052,12 @@ __global__ void k_part_fused(const uint32_t* __restrict__ idx_base,
     const int nan_bin = m.is_cat ? -1 : m.nan_bin;
     const int default_left = win->default_left;
     const int cat_onehot = m.is_cat;
+    const unsigned long long cmask = win->cat_mask;
     const uint32_t* idx = idx_base + begin;
     int local = 0;
     for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt; i += chunk_stride) {
       const int b = colbins2[idx[i]];
-      const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot);
+      const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot, cmask);
       marks[i] = static_cast<uint8_t>(go);
       local += go;
     }
@@ -1529,13 +1683,10 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   for (int f = 0; f < nf_; ++f) {
     const BinMapper* m = train_data->FeatureBinMapper(f);
     if (m->num_bin() > 256) Log::Fatal("HIP learner currently supports max_bin<=255");
-    if (m->bin_type() == BinType::kCategorical &&
-        m->num_bin() > config_->max_cat_to_onehot) {
+    if (m->bin_type() == BinType::kCategorical && m->num_bin() > 64) {
       Log::Warning("HIP learner evaluates categorical feature %d with one-hot splits "
-                   "only (%d categories > max_cat_to_onehot=%d; the CPU learner's "
-                   "sorted-subset scan is not on the device yet)",
-                   train_data->RealFeatureIndex(f), m->num_bin(),
-                   config_->max_cat_to_onehot);
+                   "only (%d categories exceed the 64-bin device sorted-subset scan)",
+                   train_data->RealFeatureIndex(f), m->num_bin());
     }
     feat_meta_host_[f] = {static_cast<int>(train_data->hist_offset(f)), m->num_bin(),
                           m->num_numeric_bin(), m->nan_bin(),
@@ -1903,6 +2054,10 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.min_gain_to_split = config_->min_gain_to_split;
   p.min_data = config_->min_data_in_leaf;
   p.bynode_frac = static_cast<float>(config_->feature_fraction_bynode);
+  p.cat_l2 = config_->cat_l2;
+  p.cat_smooth = config_->cat_smooth;
+  p.max_cat_to_onehot = config_->max_cat_to_onehot;
+  p.max_cat_threshold = config_->max_cat_threshold;
   p.extra_trees = config_->extra_trees ? 1 : 0;
   p.rng_seed = bynode_seed_;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
@@ -2070,9 +2225,21 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
     const BinMapper* mapper = train_data_->FeatureBinMapper(f);
     const int orig_f = train_data_->RealFeatureIndex(f);
     if (mapper->bin_type() == BinType::kCategorical) {
-      const int cat = static_cast<int>(mapper->BinToValue(w.bin));
-      std::vector<uint32_t> bits(std::max(cat, 0) / 32 + 1, 0);
-      if (cat >= 0) bits[cat >> 5] |= 1u << (cat & 31);
+      // value-level bitset from the winner: 64-bit bin subset (sorted scan) or a
+      // single one-hot bin
+      std::vector<int> cats;
+      if (w.cat_mask != 0) {
+        for (int b = 0; b < 64; ++b)
+          if ((w.cat_mask >> b) & 1ull)
+            cats.push_back(static_cast<int>(mapper->BinToValue(b)));
+      } else {
+        cats.push_back(static_cast<int>(mapper->BinToValue(w.bin)));
+      }
+      int max_cat = 0;
+      for (int c : cats) max_cat = std::max(max_cat, c);
+      std::vector<uint32_t> bits(max_cat / 32 + 1, 0);
+      for (int c : cats)
+        if (c >= 0) bits[c >> 5] |= 1u << (c & 31);
       tree->SplitCategorical(L, f, orig_f, bits.data(), static_cast<int>(bits.size()),
                              w.left_out, w.right_out, w.left_cnt, w.right_cnt, w.left_h,
                              0.0, static_cast<float>(w.gain), mapper->missing_type());

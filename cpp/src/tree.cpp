@@ -290,9 +290,26 @@ std::string Tree::ToJSON() const {
     if (node >= 0) {
       ss << "{\"split_index\":" << node
          << ",\"split_feature\":" << split_feature_[node]
-         << ",\"split_gain\":" << split_gain_[node]
-         << ",\"threshold\":" << Common::DoubleToStr(threshold_[node])
-         << ",\"decision_type\":\"" << (IsCategoricalSplit(node) ? "==" : "<=") << "\""
+         << ",\"split_gain\":" << split_gain_[node];
+      if (IsCategoricalSplit(node)) {
+        // reference JSON: categorical threshold = "c1||c2||..." category values
+        const int cat_idx = static_cast<int>(threshold_[node]);
+        const uint32_t* bits = cat_threshold_.data() + cat_boundaries_[cat_idx];
+        const int n_words = cat_boundaries_[cat_idx + 1] - cat_boundaries_[cat_idx];
+        std::string cats;
+        for (int w = 0; w < n_words; ++w) {
+          for (int b = 0; b < 32; ++b) {
+            if ((bits[w] >> b) & 1u) {
+              if (!cats.empty()) cats += "||";
+              cats += std::to_string(w * 32 + b);
+            }
+          }
+        }
+        ss << ",\"threshold\":\"" << cats << "\"";
+      } else {
+        ss << ",\"threshold\":" << Common::DoubleToStr(threshold_[node]);
+      }
+      ss << ",\"decision_type\":\"" << (IsCategoricalSplit(node) ? "==" : "<=") << "\""
          << ",\"default_left\":" << ((decision_type_[node] & kDefaultLeftMask) ? "true" : "false")
          << ",\"missing_type\":\"";
       int mt = (decision_type_[node] >> 2) & 3;

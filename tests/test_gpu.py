@@ -226,3 +226,34 @@ def test_gpu_extra_trees_and_bynode():
     for m in (et, bn):
         assert roc_auc_score(y, m.predict(X)) > 0.9
         assert not np.allclose(m.predict(X[:100]), base.predict(X[:100]))
+
+
+def test_gpu_categorical_sorted_subset():
+    """Categorical features with 4 < cats <= 64 use the device wave64
+    sorted-subset scan; quality must match the CPU learner's subset splits."""
+    rng = np.random.RandomState(0)
+    n = 60000
+    cat = rng.randint(0, 24, n).astype(float)
+    X = np.column_stack([cat, rng.randn(n)])
+    eff = rng.randn(24) * 1.5
+    y = (eff[cat.astype(int)] + 0.4 * rng.randn(n) > 0).astype(np.float32)
+    from sklearn.metrics import roc_auc_score
+    p = {"objective": "binary", "categorical_feature": [0], "verbosity": -1,
+         "num_leaves": 31}
+    gpu = lgb.train({**p, "device_type": "cuda"}, lgb.Dataset(X, label=y), 25)
+    cpu = lgb.train(p, lgb.Dataset(X, label=y), 25)
+    auc_gpu = roc_auc_score(y, gpu.predict(X))
+    auc_cpu = roc_auc_score(y, cpu.predict(X))
+    assert auc_gpu > 0.85
+    assert abs(auc_gpu - auc_cpu) < 0.02
+    # the model must actually contain multi-category subset splits
+    d = gpu.dump_model()
+
+    def has_subset(node):
+        if "leaf_index" in node:
+            return False
+        if node.get("decision_type") == "==" and \
+                len(str(node.get("threshold", "")).split("||")) > 1:
+            return True
+        return has_subset(node["left_child"]) or has_subset(node["right_child"])
+    assert any(has_subset(t["tree_structure"]) for t in d["tree_info"])
