@@ -50,6 +50,9 @@ class TreeLearner {
   virtual void DeviceBoosting(const class ObjectiveFunction* obj) { (void)obj; }
   /*! add a constant to the device score vector (boost_from_average) */
   virtual void DeviceAddInitScore(double v) { (void)v; }
+  /*! multiclass: select which class's device score buffer subsequent
+   *  Train/UpdateScore/DeviceAddInitScore calls operate on. */
+  virtual void SetClassOffset(int class_id) { (void)class_id; }
   /*! download the device train scores into a host buffer (for metrics / custom obj) */
   virtual void DownloadTrainScore(double* dst) { (void)dst; }
   /*! push host-modified train scores back to the device (DART drop/renormalize). */

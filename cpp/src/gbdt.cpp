@@ -213,7 +213,10 @@ double GBDT::BoostFromAverage(int class_id, bool update_scores) {
     double init_score = objective_->BoostFromScore(class_id);
     init_score = Network::GlobalSyncUpByMean(init_score);
     if (std::fabs(init_score) > kEpsilon && update_scores) {
-      if (tree_learner_->IsHIPLearner()) tree_learner_->DeviceAddInitScore(init_score);
+      if (tree_learner_->IsHIPLearner()) {
+        tree_learner_->SetClassOffset(class_id);
+        tree_learner_->DeviceAddInitScore(init_score);
+      }
       double* sc = train_score_.data() + static_cast<size_t>(class_id) * num_data_;
 #pragma omp parallel for schedule(static)
       for (data_size_t i = 0; i < num_data_; ++i) sc[i] += init_score;
@@ -265,6 +268,7 @@ bool GBDT::TrainOneIter(const score_t* gradients, const score_t* hessians) {
   for (int c = 0; c < num_tree_per_iteration_; ++c) {
     const size_t off = static_cast<size_t>(c) * num_data_;
     std::unique_ptr<Tree> new_tree(new Tree(2));
+    tree_learner_->SetClassOffset(c);
     if (objective_ == nullptr || objective_->ClassNeedTrain(c)) {
       new_tree.reset(tree_learner_->Train(gradients + off, hessians + off, models_.empty()));
     }

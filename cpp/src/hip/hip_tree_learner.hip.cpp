@@ -1540,15 +1540,20 @@ class HIPTreeLearner : public TreeLearner {
   void DeviceBoosting(const ObjectiveFunction* obj) override;
   void DeviceAddInitScore(double v) override {
     hipLaunchKernelGGL(hipk::k_score_add_const, dim3((num_data_ + 255) / 256), dim3(256), 0,
-                       stream_, d_score_.ptr, num_data_, v);
+                       stream_, ScorePtr(), num_data_, v);
+  }
+  void SetClassOffset(int class_id) override {
+    cur_class_ = class_id < num_class_score_ ? class_id : 0;
   }
   void DownloadTrainScore(double* dst) override {
     HIP_OK(hipStreamSynchronize(stream_));
-    HIP_OK(hipMemcpy(dst, d_score_.ptr, sizeof(double) * num_data_, hipMemcpyDeviceToHost));
+    HIP_OK(hipMemcpy(dst, d_score_.ptr,
+                     sizeof(double) * num_data_ * num_class_score_, hipMemcpyDeviceToHost));
   }
   void UploadTrainScore(const double* src) override {
     HIP_OK(hipStreamSynchronize(stream_));
-    HIP_OK(hipMemcpy(d_score_.ptr, src, sizeof(double) * num_data_, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(d_score_.ptr, src,
+                     sizeof(double) * num_data_ * num_class_score_, hipMemcpyHostToDevice));
   }
 
  private:
@@ -1625,6 +1630,9 @@ class HIPTreeLearner : public TreeLearner {
   bool coop_launch_ = false;   // fused cooperative partition kernel available
   bool use_mono_ = false;      // monotone constraints active (bounds tracked on device)
   uint32_t bynode_seed_ = 0x1234ABCDu;  // per-tree component of the device sampling hash
+  int num_class_score_ = 1;   // classes in the device score buffer
+  int cur_class_ = 0;         // class selected by SetClassOffset
+  double* ScorePtr() { return d_score_.ptr + static_cast<size_t>(cur_class_) * num_data_; }
   int quant_levels_ = 2;
   uint32_t quant_seed_ = 0x9E3779B9u;
   bool weights_present_ = false;
@@ -1799,8 +1807,10 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     d_grad_absmax_.Alloc(2);
     d_grad_scales_.Alloc(2);
   }
-  d_score_.Alloc(num_data_);
-  HIP_OK(hipMemset(d_score_.ptr, 0, sizeof(double) * num_data_));
+  num_class_score_ = std::max(1, config_->num_class);
+  d_score_.Alloc(static_cast<size_t>(num_data_) * num_class_score_);
+  HIP_OK(hipMemset(d_score_.ptr, 0,
+                   sizeof(double) * num_data_ * num_class_score_));
   d_label_.Alloc(num_data_);
   HIP_OK(hipMemcpy(d_label_.ptr, train_data->metadata().label(), sizeof(float) * num_data_,
                    hipMemcpyHostToDevice));
@@ -2276,7 +2286,7 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
                         hipMemcpyHostToDevice, stream_));
   const int n = static_cast<int>(used_cnt_);
   hipLaunchKernelGGL(hipk::k_score_update, dim3((n + 255) / 256), dim3(256), 0, stream_,
-                     d_idx_.ptr, d_sorted_begin_.ptr, nl, n, d_leaf_out_.ptr, d_score_.ptr);
+                     d_idx_.ptr, d_sorted_begin_.ptr, nl, n, d_leaf_out_.ptr, ScorePtr());
   if (bag_indices_ != nullptr && bag_cnt_ > 0 &&
       bag_cnt_ < static_cast<data_size_t>(num_data_)) {
     std::vector<uint32_t> oob;
@@ -2320,7 +2330,7 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
                          stream_, d_cols_.ptr, num_data_, d_tw_feat_.ptr, d_tw_thr_.ptr,
                          d_tw_left_.ptr, d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr,
                          d_tw_out_.ptr, d_oob_.ptr, static_cast<int>(oob.size()),
-                         d_score_.ptr);
+                         ScorePtr());
     }
   }
 }
@@ -2336,7 +2346,7 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
   HIP_OK(hipMemcpy(idx.data(), d_idx_.ptr, sizeof(uint32_t) * used_cnt_,
                    hipMemcpyDeviceToHost));
   std::vector<double> score(num_data_);
-  HIP_OK(hipMemcpy(score.data(), d_score_.ptr, sizeof(double) * num_data_,
+  HIP_OK(hipMemcpy(score.data(), ScorePtr(), sizeof(double) * num_data_,
                    hipMemcpyDeviceToHost));
   const int nl = tree->num_leaves();
   for (int l = 0; l < nl; ++l) {
@@ -2355,7 +2365,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // loud, not silent: features the device split loop does not implement yet fall
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
-    if (cfg->num_class > 1) return "multiclass (per-class device score buffers)";
     if (cfg->boosting == "rf") return "random forest score bookkeeping";
     if (cfg->linear_tree) return "linear_tree";
     if (!cfg->interaction_constraints.empty()) return "interaction_constraints";

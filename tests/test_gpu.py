@@ -257,3 +257,18 @@ def test_gpu_categorical_sorted_subset():
             return True
         return has_subset(node["left_child"]) or has_subset(node["right_child"])
     assert any(has_subset(t["tree_structure"]) for t in d["tree_info"])
+
+
+def test_gpu_multiclass():
+    """Multiclass on GPU: per-class device score buffers; host softmax gradients,
+    device histograms/partition per class tree."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(60000, 8)
+    y = ((X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)).astype(np.float32)
+    bst = lgb.train({"objective": "multiclass", "num_class": 3, "device_type": "cuda",
+                     "verbosity": -1}, lgb.Dataset(X, label=y), 25)
+    pred = bst.predict(X)
+    assert pred.shape == (60000, 3)
+    np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-6)
+    acc = (pred.argmax(axis=1) == y).mean()
+    assert acc > 0.85
