@@ -2,7 +2,7 @@
 predict identically in the REFERENCE LightGBM, and vice versa (the SURVEY phase-1 exit
 criterion). The reference oracle (built by tools/build_reference_oracle.sh) runs in a
 SUBPROCESS — both libraries export the LGBM_* symbols, so they cannot share a process.
-Skipped when /root/reference/lib_lightgbm.so is absent."""
+Skipped when tools/oracle/lib_lightgbm.so is absent."""
 import json
 import subprocess
 import sys
@@ -13,7 +13,7 @@ import pytest
 
 import lightgbm_amd as lgb
 
-REF_LIB = Path("/root/reference/lib_lightgbm.so")
+REF_LIB = Path(__file__).resolve().parent.parent / "tools" / "oracle" / "lib_lightgbm.so"
 pytestmark = pytest.mark.skipif(not REF_LIB.exists(),
                                 reason="reference oracle not built "
                                        "(tools/build_reference_oracle.sh)")
@@ -21,7 +21,7 @@ pytestmark = pytest.mark.skipif(not REF_LIB.exists(),
 _REF_WORKER = r"""
 import ctypes, json, sys
 import numpy as np
-lib = ctypes.cdll.LoadLibrary("/root/reference/lib_lightgbm.so")
+lib = ctypes.cdll.LoadLibrary(sys.argv[2])
 lib.LGBM_GetLastError.restype = ctypes.c_char_p
 def ok(ret):
     assert ret == 0, lib.LGBM_GetLastError().decode()
@@ -59,7 +59,7 @@ print("REF_OK")
 def _ref(cmd, tmp_path):
     worker = tmp_path / "ref_worker.py"
     worker.write_text(_REF_WORKER)
-    r = subprocess.run([sys.executable, str(worker), json.dumps(cmd)],
+    r = subprocess.run([sys.executable, str(worker), json.dumps(cmd), str(REF_LIB)],
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0 and "REF_OK" in r.stdout, r.stdout + r.stderr
 
@@ -146,7 +146,8 @@ def test_multiclass_model_cross(tmp_path):
     import subprocess as sp
     r = sp.run([sys.executable, str(worker), json.dumps(
         {"op": "predict", "model": str(f), "x": str(tmp_path / "x.npy"),
-         "out": str(tmp_path / "out.npy")})], capture_output=True, text=True, timeout=300)
+         "out": str(tmp_path / "out.npy")}), str(REF_LIB)],
+               capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     theirs = np.load(tmp_path / "out.npy").reshape(200, 3)
     np.testing.assert_allclose(bst.predict(X[:200]), theirs, rtol=1e-9, atol=1e-12)
@@ -180,3 +181,41 @@ def test_training_quality_parity_with_reference(tmp_path):
     a_ref = auc(y[16000:], ref_pred)
     assert abs(a_ours - a_ref) < 0.01, (a_ours, a_ref)
     assert a_ours > 0.9
+
+
+def test_dart_and_linear_model_cross(tmp_path):
+    """DART and linear-tree model files (is_linear leaf coefficients) must load and
+    predict identically in the reference implementation."""
+    rng = np.random.RandomState(21)
+    X = rng.rand(4000, 5)
+    y = (3 * X[:, 0] + np.sin(5 * X[:, 1])).astype(np.float32)
+    dart = lgb.train({"objective": "regression", "boosting": "dart", "drop_rate": 0.2,
+                      "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    f = tmp_path / "dart.txt"
+    dart.save_model(str(f))
+    np.testing.assert_allclose(dart.predict(X[:300]), _ref_predict(f, X[:300], tmp_path),
+                               rtol=1e-9, atol=1e-12)
+    lin = lgb.train({"objective": "regression", "linear_tree": True, "verbosity": -1,
+                     "num_leaves": 15}, lgb.Dataset(X, label=y), 15)
+    f2 = tmp_path / "linear.txt"
+    lin.save_model(str(f2))
+    np.testing.assert_allclose(lin.predict(X[:300]), _ref_predict(f2, X[:300], tmp_path),
+                               rtol=1e-7, atol=1e-10)
+
+
+def test_subset_categorical_model_cross(tmp_path):
+    """Multi-category subset splits (cat_threshold bitsets spanning many cats) must
+    route identically in the reference."""
+    rng = np.random.RandomState(31)
+    n = 6000
+    cat = rng.randint(0, 40, size=n)
+    X = np.column_stack([cat.astype(float), rng.randn(n)])
+    effect = rng.randn(40) * 2
+    y = (effect[cat] + 0.3 * rng.randn(n) > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5,
+                     "max_cat_to_onehot": 4}, lgb.Dataset(X, label=y,
+                                                          categorical_feature=[0]), 20)
+    f = tmp_path / "catsub.txt"
+    bst.save_model(str(f))
+    np.testing.assert_allclose(bst.predict(X[:500]), _ref_predict(f, X[:500], tmp_path),
+                               rtol=1e-9, atol=1e-12)

@@ -7,7 +7,13 @@
 set -e
 SHIM=/tmp/lgbshim
 BUILD=/tmp/lgbref
-mkdir -p $SHIM/fmt $SHIM/Eigen $BUILD
+SRC=/tmp/lgbrefsrc
+OUT="$(cd "$(dirname "$0")" && pwd)/oracle"
+# /root/reference must stay pristine: LightGBM's CMake drops lib_lightgbm.so into
+# the SOURCE root, so build from a disposable copy of the tree.
+mkdir -p $SHIM/fmt $SHIM/Eigen $BUILD "$OUT"
+rm -rf $SRC
+cp -r /root/reference $SRC
 
 cat > $SHIM/fast_double_parser.h <<'EOF'
 #pragma once
@@ -53,6 +59,7 @@ EOF
 cp "$(dirname "$0")/eigen_shim.h" $SHIM/Eigen/Dense
 
 cd $BUILD
-cmake /root/reference -DCMAKE_BUILD_TYPE=Release -DCMAKE_CXX_FLAGS="-I$SHIM" > /dev/null
+cmake $SRC -DCMAKE_BUILD_TYPE=Release -DCMAKE_CXX_FLAGS="-I$SHIM" > /dev/null
 make -j16 2>/dev/null || make -j16
-echo "oracle ready: /root/reference/lib_lightgbm.so"
+cp $SRC/lib_lightgbm.so "$OUT/lib_lightgbm.so"
+echo "oracle ready: $OUT/lib_lightgbm.so"
