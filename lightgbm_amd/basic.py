@@ -76,9 +76,36 @@ def _param_dict_to_str(params):
     return " ".join(pairs)
 
 
-def _to_2d_float64(data):
+def _pandas_to_float64(df, pandas_categorical=None):
+    """DataFrame -> float64 matrix, mapping category dtypes to their codes
+    (reference _data_from_pandas semantics). When `pandas_categorical` (a list of
+    category-value lists, in categorical-column order) is given, codes are aligned
+    to it — the training-time mapping — otherwise the mapping is recorded and
+    returned. Returns (array, pandas_categorical)."""
+    import pandas as pd
+    cat_cols = [c for c in df.columns if str(df[c].dtype) == "category"]
+    recorded = pandas_categorical is not None
+    if pandas_categorical is None:
+        pandas_categorical = [list(df[c].cat.categories) for c in cat_cols]
+    if not cat_cols:
+        return np.ascontiguousarray(df.to_numpy(), dtype=np.float64), pandas_categorical
+    df = df.copy()
+    for i, c in enumerate(cat_cols):
+        cats = pandas_categorical[i] if i < len(pandas_categorical) else \
+            list(df[c].cat.categories)
+        codes = pd.Categorical(df[c], categories=cats).codes.astype(np.float64)
+        codes[codes < 0] = np.nan  # unseen category / NaN -> missing
+        df[c] = codes
+    _ = recorded
+    return np.ascontiguousarray(df.to_numpy(dtype=np.float64)), pandas_categorical
+
+
+def _to_2d_float64(data, pandas_categorical=None):
     if PANDAS_INSTALLED and isinstance(data, pd_DataFrame):
-        data = data.values
+        arr, _ = _pandas_to_float64(data, pandas_categorical)
+        if arr.ndim == 1:
+            arr = arr.reshape(1, -1)
+        return arr
     arr = np.asarray(data)
     if arr.ndim == 1:
         arr = arr.reshape(1, -1)
@@ -248,6 +275,7 @@ class Dataset:
         self.categorical_feature = categorical_feature
         self.params = dict(params) if params else {}
         self.free_raw_data = free_raw_data
+        self.pandas_categorical = None
         self._handle = None
         self.used_indices = None
         self._predictor = None
@@ -293,7 +321,10 @@ class Dataset:
                 ctypes.byref(out)))
             self._handle = out
         else:
-            arr = _to_2d_float64(self.data)
+            if PANDAS_INSTALLED and isinstance(self.data, pd_DataFrame):
+                arr, self.pandas_categorical = _pandas_to_float64(self.data)
+            else:
+                arr = _to_2d_float64(self.data)
             out = ctypes.c_void_p()
             _safe_call(_LIB.LGBM_DatasetCreateFromMat(
                 arr.ctypes.data_as(ctypes.c_void_p), ctypes.c_int(_DTYPE_F64),
@@ -578,6 +609,7 @@ class Booster:
         self.best_score = {}
         self._name_valid_sets = []
         self._network_initialized = False
+        self.pandas_categorical = None
         if train_set is not None:
             if not isinstance(train_set, Dataset):
                 raise TypeError("train_set must be a Dataset")
@@ -588,6 +620,7 @@ class Booster:
                 ctypes.byref(out)))
             self._handle = out
             self._train_set = train_set
+            self.pandas_categorical = getattr(train_set, "pandas_categorical", None)
         elif model_file is not None:
             out = ctypes.c_void_p()
             out_iters = ctypes.c_int(0)
@@ -595,6 +628,16 @@ class Booster:
                 _c_str(str(model_file)), ctypes.byref(out_iters), ctypes.byref(out)))
             self._handle = out
             self.best_iteration = -1
+            # trailing pandas_categorical line (reference format)
+            try:
+                with open(model_file, "rb") as fh:
+                    tail = fh.read()[-65536:].decode("utf-8", errors="replace")
+                for line in reversed(tail.splitlines()):
+                    if line.startswith("pandas_categorical:"):
+                        self.pandas_categorical = json.loads(line[len("pandas_categorical:"):])
+                        break
+            except (OSError, ValueError):
+                pass
         elif model_str is not None:
             self.model_from_string(model_str)
         else:
@@ -890,7 +933,7 @@ class Booster:
             if pred_leaf:
                 res = res.astype(np.int32)
             return res
-        arr = _to_2d_float64(data)
+        arr = _to_2d_float64(data, getattr(self, "pandas_categorical", None))
         nrow, ncol = arr.shape
         n = ctypes.c_int64(0)
         _safe_call(_LIB.LGBM_BoosterCalcNumPredict(
@@ -952,9 +995,20 @@ class Booster:
         _safe_call(_LIB.LGBM_BoosterSaveModelToString(
             self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
             ctypes.c_int(imp), out_len, ctypes.byref(out_len), buf))
-        return buf.value.decode("utf-8")
+        text = buf.value.decode("utf-8")
+        if getattr(self, "pandas_categorical", None) is not None:
+            text += "\npandas_categorical:" + json.dumps(self.pandas_categorical) + "\n"
+        return text
 
     def model_from_string(self, model_str):
+        idx = model_str.rfind("pandas_categorical:")
+        if idx >= 0:
+            try:
+                self.pandas_categorical = json.loads(
+                    model_str[idx + len("pandas_categorical:"):].splitlines()[0])
+            except ValueError:
+                pass
+            model_str = model_str[:idx]
         if self._handle is not None:
             _safe_call(_LIB.LGBM_BoosterFree(self._handle))
         out = ctypes.c_void_p()

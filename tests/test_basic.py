@@ -327,3 +327,30 @@ def test_dataset_misc_api():
     import pytest as _pt
     with _pt.raises(lgb.LightGBMError):
         ds.set_reference(valid)  # already constructed
+
+
+def test_pandas_categorical_dtype():
+    """pandas category-dtype columns: auto-detected as categorical, codes aligned
+    through save/load via the pandas_categorical model line (reference parity)."""
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(0)
+    n = 4000
+    df = pd.DataFrame({"a": rng.randn(n),
+                       "b": pd.Categorical(rng.choice(["x", "y", "z"], n)),
+                       "c": rng.rand(n)})
+    y = ((df["b"] == "x").values & (df["a"] > 0)).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1},
+                    lgb.Dataset(df, label=y), 20)
+    pred = bst.predict(df)
+    assert ((pred > 0.5) == y).mean() > 0.98
+    m = bst.model_to_string()
+    assert "pandas_categorical:" in m
+    b2 = lgb.Booster(model_str=m)
+    # a frame with re-ordered categories must map through the stored category list
+    df2 = df.copy()
+    df2["b"] = pd.Categorical(df["b"], categories=["z", "x", "y"])
+    np.testing.assert_allclose(b2.predict(df2), pred, rtol=1e-12)
+    # unseen category becomes missing, not a crash
+    df3 = df.head(10).copy()
+    df3["b"] = pd.Categorical(["w"] * 10, categories=["w"])
+    assert np.isfinite(b2.predict(df3)).all()
