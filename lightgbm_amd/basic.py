@@ -83,6 +83,11 @@ def _pandas_to_float64(df, pandas_categorical=None):
     to it — the training-time mapping — otherwise the mapping is recorded and
     returned. Returns (array, pandas_categorical)."""
     import pandas as pd
+    bad = [str(c) for c in df.columns if str(df[c].dtype) == "object"]
+    if bad:
+        raise LightGBMError(
+            "DataFrame.dtypes must be int, float, bool or category; found object "
+            f"columns {bad} — cast them with .astype('category') first")
     cat_cols = [c for c in df.columns if str(df[c].dtype) == "category"]
     recorded = pandas_categorical is not None
     if pandas_categorical is None:
