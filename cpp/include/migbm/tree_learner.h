@@ -187,6 +187,7 @@ class SerialTreeLearner : public TreeLearner {
   int hist_mode_ = -1;
   double hist_trial_time_[2] = {0.0, 0.0};
   int hist_trials_done_ = 0;
+  bool warned_mc_method_ = false;
   // LOCAL (pre-reduce) leaf gradient totals of the last ComputeHistogram call;
   // distributed learners use them to materialize default bins before reducing
   double local_leaf_sum_g_ = 0.0, local_leaf_sum_h_ = 0.0;

@@ -2367,6 +2367,8 @@ TreeLearner* CreateHIP(const Config* cfg) {
   auto unsupported = [&]() -> const char* {
     if (cfg->boosting == "rf") return "random forest score bookkeeping";
     if (cfg->linear_tree) return "linear_tree";
+    if (cfg->monotone_penalty > 0.0 && !cfg->monotone_constraints.empty())
+      return "monotone_penalty (depth-dependent gain scaling)";
     if (!cfg->interaction_constraints.empty()) return "interaction_constraints";
     if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
     if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||

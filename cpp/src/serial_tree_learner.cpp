@@ -308,6 +308,17 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
   for (int f = 0; f < nf; ++f) {
     if (!cand[f].IsValid()) continue;
     const int orig = train_data_->RealFeatureIndex(f);
+    // monotone_penalty: depth-decaying multiplicative penalty on monotone splits
+    // (parity: reference ComputeMonotoneSplitGainPenalty)
+    if (cand[f].monotone_type != 0 && config_->monotone_penalty > 0.0) {
+      const double pen = config_->monotone_penalty;
+      const int depth = ctx.depth;
+      double factor;
+      if (pen >= depth + 1.0) factor = kEpsilon;
+      else if (pen <= 1.0) factor = 1.0 - pen / std::pow(2.0, depth) + kEpsilon;
+      else factor = 1.0 - std::pow(2.0, pen - 1.0 - depth) + kEpsilon;
+      cand[f].gain *= factor;
+    }
     // feature_contri: multiplicative per-feature gain reweighting
     if (!config_->feature_contri.empty() &&
         orig < static_cast<int>(config_->feature_contri.size())) {
@@ -395,6 +406,13 @@ std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& 
 
 Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians,
                                bool /*is_first_tree*/) {
+  if (!config_->monotone_constraints.empty() &&
+      config_->monotone_constraints_method != "basic" && !warned_mc_method_) {
+    Log::Warning("monotone_constraints_method=%s is not implemented; using the basic "
+                 "method with leaf-bound propagation",
+                 config_->monotone_constraints_method.c_str());
+    warned_mc_method_ = true;
+  }
   gradients_ = gradients;
   hessians_ = hessians;
   if (config_->use_quantized_grad) {

@@ -611,3 +611,18 @@ def test_hist_mode_forced_equivalence():
     assert acc_c > 0.9 and acc_r > 0.9
     np.testing.assert_allclose(col.predict(X[:200]), row.predict(X[:200]),
                                rtol=1e-6, atol=1e-6)
+
+
+def test_monotone_penalty_discourages_shallow_monotone_splits():
+    rng = np.random.RandomState(0)
+    X = rng.rand(4000, 2)
+    y = (2 * X[:, 0] + 2 * X[:, 1] + 0.1 * rng.randn(4000)).astype(np.float32)
+    p = {"objective": "regression", "monotone_constraints": [1, 0], "verbosity": -1}
+    free = lgb.train(p, lgb.Dataset(X, label=y), 10)
+    pen = lgb.train({**p, "monotone_penalty": 2.0}, lgb.Dataset(X, label=y), 10)
+
+    def root_feature(b):
+        return b.dump_model()["tree_info"][0]["tree_structure"]["split_feature"]
+    # with a strong penalty the constrained feature 0 should lose the root split
+    assert root_feature(pen) == 1
+    assert not np.allclose(free.predict(X[:100]), pen.predict(X[:100]))
