@@ -194,6 +194,8 @@ class DART : public GBDT {
   void Normalize(const std::vector<int>& dropped, Tree* new_tree, int cur_tree_id);
   Random drop_rng_{4};
   bool drop_rng_init_ = false;
+  std::vector<double> tree_weight_;  // per iteration; drives weighted (non-uniform) drop
+  double sum_weight_ = 0.0;
 };
 
 class RF : public GBDT {

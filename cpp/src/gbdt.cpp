@@ -617,19 +617,34 @@ void GBDT::RefitTree(const int32_t* leaf_preds, int nrow, int ncol) {
 
 // ------------------------------------------------------------------ DART
 std::vector<int> DART::DroppingTrees() {
+  // parity: reference dart.hpp DroppingTrees — skip_drop gate, then per-tree
+  // Bernoulli with weight-proportional rates (uniform_drop=false default) and a
+  // max_drop-derived rate clamp; an empty selection is a plain boosting step.
   if (!drop_rng_init_) { drop_rng_ = Random(config_->drop_seed); drop_rng_init_ = true; }
   std::vector<int> dropped;
   const int n_iters = iter_;
   if (n_iters == 0) return dropped;
   if (drop_rng_.NextFloat() < config_->skip_drop) return dropped;
-  for (int i = 0; i < n_iters; ++i) {
-    if (drop_rng_.NextFloat() < config_->drop_rate) {
-      dropped.push_back(i);
-      if (static_cast<int>(dropped.size()) >= config_->max_drop) break;
+  double drop_rate = config_->drop_rate;
+  if (!config_->uniform_drop && sum_weight_ > 0) {
+    const double inv_avg = static_cast<double>(tree_weight_.size()) / sum_weight_;
+    if (config_->max_drop > 0)
+      drop_rate = std::min(drop_rate, config_->max_drop * inv_avg / sum_weight_);
+    for (int i = 0; i < n_iters; ++i) {
+      if (drop_rng_.NextFloat() < drop_rate * tree_weight_[i] * inv_avg) {
+        dropped.push_back(i);
+        if (static_cast<int>(dropped.size()) >= config_->max_drop) break;
+      }
     }
-  }
-  if (dropped.empty() && config_->drop_rate > 0) {
-    dropped.push_back(drop_rng_.NextInt(0, n_iters));
+  } else {
+    if (config_->max_drop > 0)
+      drop_rate = std::min(drop_rate, config_->max_drop / static_cast<double>(n_iters));
+    for (int i = 0; i < n_iters; ++i) {
+      if (drop_rng_.NextFloat() < drop_rate) {
+        dropped.push_back(i);
+        if (static_cast<int>(dropped.size()) >= config_->max_drop) break;
+      }
+    }
   }
   return dropped;
 }
@@ -659,25 +674,34 @@ bool DART::TrainOneIter(const score_t* gradients, const score_t* hessians) {
     }
   }
   if (dev) tree_learner_->UploadTrainScore(train_score_.data());
+  // the new tree's weight: lr/(1+k), or the xgboost-mode lr/(lr+k) (reference
+  // dart.hpp shrinkage_rate_); applied directly through the base train path
+  const double k = static_cast<double>(dropped.size());
+  const double lr = config_->learning_rate;
+  const double saved_rate = shrinkage_rate_;
+  if (config_->xgboost_dart_mode)
+    shrinkage_rate_ = dropped.empty() ? lr : lr / (lr + k);
+  else
+    shrinkage_rate_ = lr / (1.0 + k);
+  const double new_weight = shrinkage_rate_;
   bool stop = GBDT::TrainOneIter(gradients, hessians);
+  shrinkage_rate_ = saved_rate;
   if (!stop && dev) tree_learner_->DownloadTrainScore(train_score_.data());
   if (!stop) {
-    // normalize: new tree trained against the residual without dropped trees
-    const double k = static_cast<double>(dropped.size());
-    const double norm_new = 1.0 / (k + 1.0);
+    tree_weight_.push_back(new_weight);
+    sum_weight_ += new_weight;
+    // dropped trees re-enter at factor k/(k+1) (or k/(k+lr) in xgboost mode)
+    const double factor = config_->xgboost_dart_mode ? k / (k + lr) : k / (k + 1.0);
     for (int c = 0; c < num_tree_per_iteration_; ++c) {
-      Tree* new_tree = models_[models_.size() - num_tree_per_iteration_ + c].get();
-      // undo full-weight score update of new tree, reapply scaled
-      Tree adj(*new_tree);
-      adj.Shrinkage(norm_new - 1.0);  // score currently has 1.0x; want norm_new x
-      apply_tree_all(adj, c);
-      new_tree->Shrinkage(norm_new);
-      // shrink dropped trees to k/(k+1) of their weight and re-add them to the scores
       for (int it : dropped) {
         Tree* old = models_[static_cast<size_t>(it) * num_tree_per_iteration_ + c].get();
-        old->Shrinkage(k / (k + 1.0));
+        old->Shrinkage(factor);
         apply_tree_all(*old, c);
       }
+    }
+    for (int it : dropped) {
+      sum_weight_ -= tree_weight_[it] * (1.0 - factor);
+      tree_weight_[it] *= factor;
     }
     if (dev) tree_learner_->UploadTrainScore(train_score_.data());
   }

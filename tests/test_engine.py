@@ -626,3 +626,13 @@ def test_monotone_penalty_discourages_shallow_monotone_splits():
     # with a strong penalty the constrained feature 0 should lose the root split
     assert root_feature(pen) == 1
     assert not np.allclose(free.predict(X[:100]), pen.predict(X[:100]))
+
+
+def test_dart_modes():
+    """uniform_drop / weighted (default) / xgboost_dart_mode all train sane models."""
+    X, y = _binary_data(n=3000)
+    for extra in ({}, {"uniform_drop": True}, {"xgboost_dart_mode": True}):
+        bst = lgb.train({"objective": "binary", "boosting": "dart", "drop_rate": 0.3,
+                         "verbosity": -1, **extra}, lgb.Dataset(X, label=y), 25)
+        acc = ((bst.predict(X) > 0.5) == y).mean()
+        assert acc > 0.8, extra
