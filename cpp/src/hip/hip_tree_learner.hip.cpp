@@ -95,6 +95,7 @@ struct GainParams {
   // categorical scan (parity: FindBestThresholdCategorical CPU oracle)
   double cat_l2, cat_smooth;
   int max_cat_to_onehot, max_cat_threshold;
+  int n_interaction_groups;  // 0 = unconstrained
 };
 
 __device__ __forceinline__ uint32_t d_hash3(uint32_t a, uint32_t b, uint32_t c) {
@@ -391,9 +392,11 @@ __global__ void k_hist_subtract(float* __restrict__ hist_base, size_t slot_strid
 // ------------------------------------------------------------------ root setup
 __global__ void k_init_root(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
                             int used_cnt, int64_t* gbuf, int* counters, int* root_leaf,
-                            int* minus1, double* leaf_bounds) {
+                            int* minus1, double* leaf_bounds,
+                            unsigned long long* leaf_branch) {
   leaf_bounds[0] = -1e308;  // monotone output bounds of the root leaf
   leaf_bounds[1] = 1e308;
+  leaf_branch[0] = 0ull;
   leaf_begin[0] = 0;
   leaf_cnt[0] = used_cnt;
   leaf_slot[0] = 0;
@@ -451,6 +454,8 @@ __global__ void __launch_bounds__(64) k_best_feat(
     const int* __restrict__ leafA_ptr, const int* __restrict__ counters,
     int leafB_from_counters, GainParams p, const int8_t* __restrict__ feat_mask,
     const int8_t* __restrict__ mono, const double* __restrict__ leaf_bounds,
+    const unsigned long long* __restrict__ group_masks,
+    const unsigned long long* __restrict__ leaf_branch,
     SplitRec* __restrict__ out) {
   const int f = blockIdx.x;
   const int which = blockIdx.y;
@@ -467,6 +472,15 @@ __global__ void __launch_bounds__(64) k_best_feat(
     rec.cat_mask = 0;
   }
   if (feat_mask != nullptr && !feat_mask[f]) return;
+  if (p.n_interaction_groups > 0) {
+    // interaction constraints (inner-feature bitmask form, nf <= 64): feature f
+    // is allowed iff some group covers the leaf's branch features plus f
+    const unsigned long long need = leaf_branch[leaf] | (1ull << f);
+    bool ok = false;
+    for (int g = 0; g < p.n_interaction_groups; ++g)
+      ok = ok || (need & ~group_masks[g]) == 0ull;
+    if (!ok) return;
+  }
   const int split_idx = counters[1];
   if (p.bynode_frac < 1.0f) {
     // deterministic per-(split, child, feature) Bernoulli mask; one hashed anchor
@@ -1211,7 +1225,7 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
                                     LeafStat* stats, const SplitRec* winner, int L,
                                     int* counters, LogEntry* log, const int* ctr,
                                     const int64_t* gbuf, const int8_t* mono,
-                                    double* leaf_bounds);
+                                    double* leaf_bounds, unsigned long long* leaf_branch);
 
 /*! device-side split bookkeeping: segments, stats, slot map, split log. Thread 0
  *  does the bookkeeping; the whole block then zeroes the spare histogram slot the
@@ -1222,7 +1236,8 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
                            LogEntry* __restrict__ log, const int* __restrict__ ctr,
                            const int64_t* __restrict__ gbuf, float* hist_base,
                            size_t slot_stride, int n_elem,
-                           const int8_t* __restrict__ mono, double* leaf_bounds) {
+                           const int8_t* __restrict__ mono, double* leaf_bounds,
+                           unsigned long long* leaf_branch) {
   __shared__ int s_spare;
   if (threadIdx.x == 0) {
     s_spare = -1;
@@ -1232,7 +1247,7 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
     } else {
       s_spare = counters[0];
       FinalizeBookkeeping(leaf_begin, leaf_cnt, leaf_slot, stats, winner, L, counters, log,
-                          ctr, gbuf, mono, leaf_bounds);
+                          ctr, gbuf, mono, leaf_bounds, leaf_branch);
     }
   }
   __syncthreads();
@@ -1247,7 +1262,7 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
                                     int L, int* counters, LogEntry* __restrict__ log,
                                     const int* __restrict__ ctr,
                                     const int64_t* __restrict__ gbuf, const int8_t* mono,
-                                    double* leaf_bounds) {
+                                    double* leaf_bounds, unsigned long long* leaf_branch) {
   const int R = counters[0];
   const int spare_slot = R;
   log[counters[1]].rec = *winner;
@@ -1275,6 +1290,11 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
     leaf_slot[R] = old_slot;
   } else {
     leaf_slot[R] = spare_slot;  // L keeps old slot
+  }
+  if (leaf_branch != nullptr) {
+    const unsigned long long b = leaf_branch[L] | (1ull << w.feature);
+    leaf_branch[L] = b;
+    leaf_branch[R] = b;
   }
   if (mono != nullptr) {
     // BasicLeafConstraints bound propagation (mirrors the host learner): children
@@ -1611,6 +1631,9 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<hipk::LeafStat> d_leaf_stats_;
   DevBuf<double> d_leaf_bounds_;   // [2*num_leaves] monotone output bounds
   DevBuf<int8_t> d_mono_;          // per inner feature, only when constraints set
+  DevBuf<unsigned long long> d_leaf_branch_;  // [num_leaves] branch-feature bitmasks
+  DevBuf<unsigned long long> d_group_masks_;  // interaction groups (inner-feature bits)
+  int n_interaction_groups_ = 0;
   DevBuf<int8_t> d_feat_mask_;
   DevBuf<int> d_leaf_begin_, d_leaf_cnt_, d_leaf_slot_;
   DevBuf<int> d_sorted_begin_;
@@ -1840,6 +1863,40 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_split_log_.Alloc(nl);
   d_leaf_stats_.Alloc(nl);
   d_leaf_bounds_.Alloc(2 * static_cast<size_t>(nl));
+  d_leaf_branch_.Alloc(nl);
+  n_interaction_groups_ = 0;
+  if (!config_->interaction_constraints.empty() && nf_ > 64)
+    Log::Fatal("device interaction constraints support up to 64 features (%d present); "
+               "use device_type=cpu for this dataset", nf_);
+  if (!config_->interaction_constraints.empty()) {
+    // parse "[0,1],[2,3]" original-feature groups into inner-feature bitmasks
+    std::vector<unsigned long long> masks;
+    std::vector<int> orig_to_inner(train_data->num_total_features(), -1);
+    for (int f = 0; f < nf_; ++f) orig_to_inner[train_data->RealFeatureIndex(f)] = f;
+    std::string sgrp = config_->interaction_constraints;
+    size_t pos = 0;
+    while ((pos = sgrp.find('[', pos)) != std::string::npos) {
+      size_t end = sgrp.find(']', pos);
+      if (end == std::string::npos) break;
+      unsigned long long msk = 0;
+      for (auto& tok : Common::Split(sgrp.substr(pos + 1, end - pos - 1).c_str(), ',')) {
+        auto t = Common::Trim(tok);
+        if (t.empty()) continue;
+        const int orig = atoi(t.c_str());
+        if (orig >= 0 && orig < static_cast<int>(orig_to_inner.size()) &&
+            orig_to_inner[orig] >= 0)
+          msk |= 1ull << orig_to_inner[orig];
+      }
+      if (msk) masks.push_back(msk);
+      pos = end + 1;
+    }
+    if (!masks.empty()) {
+      n_interaction_groups_ = static_cast<int>(masks.size());
+      d_group_masks_.Alloc(masks.size());
+      HIP_OK(hipMemcpy(d_group_masks_.ptr, masks.data(),
+                       sizeof(unsigned long long) * masks.size(), hipMemcpyHostToDevice));
+    }
+  }
   use_mono_ = false;
   {
     const auto& mcs = config_->monotone_constraints;
@@ -2068,6 +2125,7 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.cat_smooth = config_->cat_smooth;
   p.max_cat_to_onehot = config_->max_cat_to_onehot;
   p.max_cat_threshold = config_->max_cat_threshold;
+  p.n_interaction_groups = n_interaction_groups_;
   p.extra_trees = config_->extra_trees ? 1 : 0;
   p.rng_seed = bynode_seed_;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
@@ -2077,7 +2135,8 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                      d_leaf_stats_.ptr, leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
                      feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
                      use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
-                     d_feat_best_.ptr);
+                     n_interaction_groups_ > 0 ? d_group_masks_.ptr : nullptr,
+                     d_leaf_branch_.ptr, d_feat_best_.ptr);
   (void)ny;
   hipLaunchKernelGGL(hipk::k_best_leaf_overall, dim3(1), dim3(256), 0, stream_,
                      d_feat_best_.ptr, nf_, d_leaf_best_.ptr, leafA_ptr, d_counters_.ptr,
@@ -2127,7 +2186,8 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   hipLaunchKernelGGL(hipk::k_init_root, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
                      d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr,
                      static_cast<int>(used_cnt_), d_gbuf_.ptr, d_counters_.ptr,
-                     d_root_leaf_.ptr, d_minus1_.ptr, d_leaf_bounds_.ptr);
+                     d_root_leaf_.ptr, d_minus1_.ptr, d_leaf_bounds_.ptr,
+                     d_leaf_branch_.ptr);
   {
     const int blocks = std::min(2048, (static_cast<int>(used_cnt_) + 255) / 256);
     hipLaunchKernelGGL(hipk::k_root_sums, dim3(blocks), dim3(256), 0, stream_, d_idx_.ptr,
@@ -2203,7 +2263,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
                        d_gbuf_.ptr, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
                        total_bins_ * 2, use_mono_ ? d_mono_.ptr : nullptr,
-                       d_leaf_bounds_.ptr);
+                       d_leaf_bounds_.ptr, d_leaf_branch_.ptr);
     LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks, /*zero_spare=*/false);
     ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
     {
@@ -2369,7 +2429,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
     if (cfg->linear_tree) return "linear_tree";
     if (cfg->monotone_penalty > 0.0 && !cfg->monotone_constraints.empty())
       return "monotone_penalty (depth-dependent gain scaling)";
-    if (!cfg->interaction_constraints.empty()) return "interaction_constraints";
     if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
     if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||
         !cfg->cegb_penalty_feature_lazy.empty())

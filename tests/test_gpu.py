@@ -272,3 +272,26 @@ def test_gpu_multiclass():
     np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-6)
     acc = (pred.argmax(axis=1) == y).mean()
     assert acc > 0.85
+
+
+def test_gpu_interaction_constraints():
+    """Interaction constraints enforced inside the device loop via per-leaf
+    branch-feature bitmasks: no tree may mix features across groups."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(40000, 4)
+    y = (X[:, 0] * X[:, 1] + X[:, 2] * X[:, 3] > 0).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "device_type": "cuda", "num_leaves": 31,
+                     "interaction_constraints": "[0,1],[2,3]", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 15)
+    d = bst.dump_model()
+
+    def feats(node, acc):
+        if "leaf_index" in node:
+            return
+        acc.add(node["split_feature"])
+        feats(node["left_child"], acc)
+        feats(node["right_child"], acc)
+    for t in d["tree_info"]:
+        used = set()
+        feats(t["tree_structure"], used)
+        assert used <= {0, 1} or used <= {2, 3}, used
