@@ -218,16 +218,22 @@ def cv(params, train_set, num_boost_round=100, folds=None, nfold=5, stratified=T
     results = collections.defaultdict(list)
     for i in range(num_boost_round):
         agg = collections.defaultdict(list)
+        train_agg = collections.defaultdict(list)
         for bst in cvbooster.boosters:
             bst.update(fobj=fobj)
             for (name, metric, value, hb) in bst.eval_valid(feval):
                 agg[metric].append((value, hb))
+            if eval_train_metric:
+                for (name, metric, value, hb) in bst.eval_train(feval):
+                    train_agg[metric].append((value, hb))
         line = []
-        for metric, vals in agg.items():
-            vs = [v for v, _ in vals]
-            results[f"valid {metric}-mean"].append(float(np.mean(vs)))
-            results[f"valid {metric}-stdv"].append(float(np.std(vs)))
-            line.append((("cv_agg"), metric, float(np.mean(vs)), vals[0][1]))
+        for prefix, bucket in (("train", train_agg), ("valid", agg)):
+            for metric, vals in bucket.items():
+                vs = [v for v, _ in vals]
+                results[f"{prefix} {metric}-mean"].append(float(np.mean(vs)))
+                results[f"{prefix} {metric}-stdv"].append(float(np.std(vs)))
+                line.append((("cv_agg"), f"{prefix} {metric}", float(np.mean(vs)),
+                             vals[0][1]))
         if es_cb is not None:
             env = callback_mod.CallbackEnv(model=cvbooster, params=params, iteration=i,
                                            begin_iteration=0, end_iteration=num_boost_round,

@@ -636,3 +636,32 @@ def test_dart_modes():
                          "verbosity": -1, **extra}, lgb.Dataset(X, label=y), 25)
         acc = ((bst.predict(X) > 0.5) == y).mean()
         assert acc > 0.8, extra
+
+
+def test_cv_eval_train_metric_and_plotting_smoke():
+    X, y = _binary_data(n=2000)
+    r = lgb.cv({"objective": "binary", "metric": "auc", "verbosity": -1},
+               lgb.Dataset(X, label=y), 10, nfold=3, eval_train_metric=True,
+               return_cvbooster=True)
+    assert "train auc-mean" in r and "valid auc-mean" in r
+    assert len(r["cvbooster"].boosters) == 3
+    # plotting smoke (matplotlib Agg)
+    try:
+        import matplotlib
+        matplotlib.use("Agg")
+    except ImportError:
+        return
+    bst = lgb.train({"objective": "binary", "metric": "auc", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 5)
+    ev = {}
+    lgb.train({"objective": "binary", "metric": "auc", "verbosity": -1},
+              lgb.Dataset(X, label=y), 5,
+              valid_sets=[lgb.Dataset(X, label=y).create_valid(X, label=y)],
+              callbacks=[lgb.record_evaluation(ev)])
+    ax = lgb.plot_importance(bst)
+    assert ax is not None
+    ax2 = lgb.plot_metric(ev)
+    assert ax2 is not None
+    import pytest as _pt
+    with _pt.raises(ImportError):
+        lgb.plot_tree(bst, tree_index=0)  # graphviz absent here (reference parity)
