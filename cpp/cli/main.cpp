@@ -171,6 +171,7 @@ class Application {
     if (config_.input_model.empty()) Log::Fatal("task=refit requires input_model=");
     LoadData();
     boosting_.reset(GBDT::CreateBoosting("gbdt", config_.input_model.c_str()));
+    boosting_->ResetConfig(&config_);  // file-loaded boosters carry no config
     objective_.reset(ObjectiveFunction::Create(boosting_->ObjectiveName(), config_));
     if (objective_) objective_->Init(train_data_->metadata(), train_data_->num_data());
     std::vector<const Metric*> none;

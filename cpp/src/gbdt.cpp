@@ -163,7 +163,8 @@ void GBDT::ResetTrainingData(const Dataset* train_data, const ObjectiveFunction*
   objective_ = objective;
   training_metrics_ = training_metrics;
   num_data_ = train_data_->num_data();
-  tree_learner_->ResetTrainingData(train_data_);
+  // a file-loaded booster has no learner/config until (re)initialized for training
+  if (tree_learner_ != nullptr) tree_learner_->ResetTrainingData(train_data_);
   train_score_.assign(static_cast<size_t>(num_data_) * num_tree_per_iteration_, 0.0);
   // re-apply existing model to scores
   for (size_t i = 0; i < models_.size(); ++i) {
@@ -173,8 +174,10 @@ void GBDT::ResetTrainingData(const Dataset* train_data, const ObjectiveFunction*
   }
   gradients_.assign(train_score_.size(), 0);
   hessians_.assign(train_score_.size(), 0);
-  sample_strategy_.reset(SampleStrategy::Create(config_, train_data_, objective_,
-                                                num_tree_per_iteration_));
+  if (config_ != nullptr) {
+    sample_strategy_.reset(SampleStrategy::Create(config_, train_data_, objective_,
+                                                  num_tree_per_iteration_));
+  }
 }
 
 void GBDT::ResetConfig(const Config* config) {

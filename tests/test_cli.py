@@ -124,3 +124,16 @@ def test_cli_arg_overrides_config_alias():
                         f"output_model={out}"],
                        cwd=EXAMPLE, capture_output=True, timeout=300, check=True)
         assert out.read_text().count("Tree=") == 3
+
+
+def test_cli_refit(cli_model, tmp_path):
+    """task=refit re-derives leaf values on new data and saves a working model."""
+    import subprocess
+    out = tmp_path / "refit.txt"
+    subprocess.run([str(CLI), "task=refit", f"data={EXAMPLE/'binary.train'}",
+                    f"input_model={cli_model}", f"output_model={out}"],
+                   capture_output=True, timeout=300, check=True)
+    X, y = _load_tsv(EXAMPLE / "binary.test")
+    bst = lgb.Booster(model_file=str(out))
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.7
