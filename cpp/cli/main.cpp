@@ -20,10 +20,12 @@ class Application {
   explicit Application(int argc, char** argv) {
     std::unordered_map<std::string, std::string> params;
     for (int i = 1; i < argc; ++i) {
-      auto kv = Common::Split(argv[i], '=');
-      if (kv.size() == 2)
-        params[Config::ResolveAlias(Common::ToLower(Common::Trim(kv[0])))] =
-            Common::Trim(kv[1]);
+      // split on the FIRST '=' only: values may themselves contain '='
+      std::string a(argv[i]);
+      auto eq = a.find('=');
+      if (eq == std::string::npos || eq == 0) continue;
+      params[Config::ResolveAlias(Common::ToLower(Common::Trim(a.substr(0, eq))))] =
+          Common::Trim(a.substr(eq + 1));
     }
     // config file first, argv overrides
     auto it = params.find("config");

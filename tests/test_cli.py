@@ -137,3 +137,16 @@ def test_cli_refit(cli_model, tmp_path):
     bst = lgb.Booster(model_file=str(out))
     acc = ((bst.predict(X) > 0.5) == y).mean()
     assert acc > 0.7
+
+
+def test_cli_predict_variants(cli_model, tmp_path):
+    """predict_raw_score / predict_leaf_index / predict_contrib output shapes."""
+    import subprocess
+    for flag, ncols in (("predict_raw_score", 1), ("predict_leaf_index", 100),
+                        ("predict_contrib", 29)):
+        out = tmp_path / f"{flag}.txt"
+        subprocess.run([str(CLI), "task=predict", f"data={EXAMPLE/'binary.test'}",
+                        f"input_model={cli_model}", f"output_result={out}",
+                        f"{flag}=true"], capture_output=True, timeout=300, check=True)
+        first = out.read_text().splitlines()[0].split("\t")
+        assert len(first) == ncols, (flag, len(first))
