@@ -142,7 +142,8 @@ def test_cli_refit(cli_model, tmp_path):
 def test_cli_predict_variants(cli_model, tmp_path):
     """predict_raw_score / predict_leaf_index / predict_contrib output shapes."""
     import subprocess
-    for flag, ncols in (("predict_raw_score", 1), ("predict_leaf_index", 100),
+    ntrees = lgb.Booster(model_file=str(cli_model)).num_trees()
+    for flag, ncols in (("predict_raw_score", 1), ("predict_leaf_index", ntrees),
                         ("predict_contrib", 29)):
         out = tmp_path / f"{flag}.txt"
         subprocess.run([str(CLI), "task=predict", f"data={EXAMPLE/'binary.test'}",
