@@ -63,7 +63,7 @@ dist.destroy_process_group()
 """
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 3])
 def test_data_parallel_identical_models(tmp_path, world):
     script = tmp_path / "worker.py"
     script.write_text(WORKER)
