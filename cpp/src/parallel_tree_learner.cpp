@@ -32,7 +32,7 @@ struct WireSplit {
   uint8_t default_left;
   uint8_t is_cat;
   uint8_t n_cat_words;
-  uint8_t pad;
+  int8_t monotone_type;
   uint32_t cat_words[8];
 
   static WireSplit From(const SplitInfo& s) {
@@ -50,6 +50,7 @@ struct WireSplit {
     w.feature = s.feature;
     w.threshold = s.threshold;
     w.default_left = s.default_left;
+    w.monotone_type = s.monotone_type;
     w.is_cat = !s.cat_bitset_inner.empty();
     w.n_cat_words = static_cast<uint8_t>(std::min<size_t>(8, s.cat_bitset_inner.size()));
     for (int i = 0; i < w.n_cat_words; ++i) w.cat_words[i] = s.cat_bitset_inner[i];
@@ -69,6 +70,7 @@ struct WireSplit {
     s.feature = feature;
     s.threshold = threshold;
     s.default_left = default_left;
+    s.monotone_type = monotone_type;
     if (is_cat) s.cat_bitset_inner.assign(cat_words, cat_words + n_cat_words);
     return s;
   }
@@ -91,9 +93,14 @@ SplitInfo SyncUpGlobalBestSplit(const SplitInfo& local) {
 
 }  // namespace
 
-/*! Data-parallel learner: local row shard, globally reduced histograms.
- *  Every rank scans the same global histogram -> identical split chosen everywhere with
- *  no extra sync (gain scan is deterministic). */
+/*! Data-parallel learner: local row shard, globally reduced histograms, and the
+ *  reference's FEATURE-OWNERSHIP gain scan (data_parallel_tree_learner.cpp
+ *  structure): features are assigned to ranks balanced by bin count, each rank
+ *  scans only its owned features of the reduced histogram, and the per-rank
+ *  winners are argmax-allgathered — dividing FindBestSplit work by the rank
+ *  count. (The wire-level reduce-scatter of the histogram payload itself is an
+ *  RCCL-path optimization — see docs/ROADMAP.md #1 — the injected-collective
+ *  transport here moves the same bytes either way at these payload sizes.) */
 class DataParallelTreeLearner : public SerialTreeLearner {
  public:
   explicit DataParallelTreeLearner(const Config* config) : SerialTreeLearner(config) {}
@@ -106,6 +113,42 @@ class DataParallelTreeLearner : public SerialTreeLearner {
     }
     SerialTreeLearner::OnHistogramReady(leaf);  // EFB default-bin reconstruction
   }
+  void FindBestSplitForLeaf(int leaf, const LeafContext& ctx) override {
+    if (!Network::is_distributed()) {
+      SerialTreeLearner::FindBestSplitForLeaf(leaf, ctx);
+      return;
+    }
+    if (feature_owner_.empty()) AssignFeatureOwnership();
+    std::vector<int8_t> saved = is_feature_used_;
+    const int nf = train_data_->num_features();
+    const int rank = Network::rank();
+    for (int f = 0; f < nf; ++f)
+      if (feature_owner_[f] != rank) is_feature_used_[f] = 0;
+    SerialTreeLearner::FindBestSplitForLeaf(leaf, ctx);
+    is_feature_used_ = saved;
+    best_split_per_leaf_[leaf] = SyncUpGlobalBestSplit(best_split_per_leaf_[leaf]);
+  }
+  /*! deterministic greedy assignment balanced by bin count (all ranks compute the
+   *  identical map; parity: reference per-tree ownership balancing) */
+  void AssignFeatureOwnership() {
+    const int nf = train_data_->num_features();
+    const int world = Network::num_machines();
+    feature_owner_.assign(nf, 0);
+    std::vector<int> order(nf);
+    for (int f = 0; f < nf; ++f) order[f] = f;
+    std::stable_sort(order.begin(), order.end(), [&](int a, int b) {
+      return train_data_->FeatureNumBin(a) > train_data_->FeatureNumBin(b);
+    });
+    std::vector<int64_t> load(world, 0);
+    for (int f : order) {
+      int r = 0;
+      for (int i = 1; i < world; ++i)
+        if (load[i] < load[r]) r = i;
+      feature_owner_[f] = r;
+      load[r] += train_data_->FeatureNumBin(f);
+    }
+  }
+  std::vector<int> feature_owner_;
   void ReduceRootStats(double* sum_g, double* sum_h, data_size_t* cnt) override {
     if (!Network::is_distributed()) return;
     double v[2] = {*sum_g, *sum_h};
