@@ -839,6 +839,11 @@ int LGBM_DatasetSetField(DatasetHandle handle, const char* field_name, const voi
   API_BEGIN();
   Dataset* d = static_cast<Dataset*>(handle);
   std::string name(field_name);
+  if ((name == "label" || name == "weight" || name == "position") &&
+      num_element != d->num_data()) {
+    Log::Fatal("Length of %s (%d) differs from the number of rows (%d)", field_name,
+               num_element, d->num_data());
+  }
   if (name == "label" || name == "weight") {
     std::vector<float> buf(num_element);
     if (type == C_API_DTYPE_FLOAT32) {
@@ -1164,6 +1169,9 @@ int LGBM_BoosterPredictForMat(BoosterHandle handle, const void* data, int data_t
                               int64_t* out_len, double* out_result) {
   API_BEGIN();
   auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
+  if (ncol <= b->MaxFeatureIdx())
+    Log::Fatal("The number of features in data (%d) is fewer than it was in training data "
+               "(%d)", ncol, b->MaxFeatureIdx() + 1);
   auto get = MakeGetter(data, data_type);
   std::function<void(int64_t, double*)> row_getter;
   if (is_row_major) {

@@ -354,3 +354,19 @@ def test_pandas_categorical_dtype():
     df3 = df.head(10).copy()
     df3["b"] = pd.Categorical(["w"] * 10, categories=["w"])
     assert np.isfinite(b2.predict(df3)).all()
+
+
+def test_validation_errors():
+    """Length-mismatched labels and too-few predict columns must raise, not
+    silently mis-train/mis-predict."""
+    rng = np.random.RandomState(0)
+    X = rng.rand(100, 4)
+    with pytest.raises(lgb.LightGBMError):
+        lgb.Dataset(X, label=np.zeros(50, dtype=np.float32)).construct()
+    y = (X[:, 0] > 0.5).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5},
+                    lgb.Dataset(X, label=y), 3)
+    with pytest.raises(lgb.LightGBMError):
+        bst.predict(rng.rand(5, 2))
+    # more columns than training is allowed (extras ignored), like the reference
+    assert len(bst.predict(np.column_stack([X[:5], np.zeros(5)]))) == 5
