@@ -120,7 +120,32 @@ class LGBMModel(BaseEstimator):
              eval_group=None, eval_metric=None, early_stopping_rounds=None,
              callbacks=None, categorical_feature="auto", feature_name="auto"):
         params = self._make_params(default_objective)
-        if eval_metric is not None:
+        # sklearn-style CALLABLE objective: (y_true, y_pred[, weight[, group]]) ->
+        # (grad, hess); adapted to the engine's (preds, dataset) form
+        fobj = None
+        if callable(params.get("objective")):
+            user_obj = params["objective"]
+            params["objective"] = "none"
+
+            def fobj(preds, dataset):
+                import inspect
+                args = [dataset.get_label(), preds]
+                n_par = len(inspect.signature(user_obj).parameters)
+                if n_par >= 3:
+                    args.append(dataset.get_weight())
+                if n_par >= 4:
+                    args.append(dataset.get_group())
+                g, h = user_obj(*args)
+                return np.asarray(g, dtype=np.float32), np.asarray(h, dtype=np.float32)
+        feval = None
+        if callable(eval_metric):
+            user_metric = eval_metric
+
+            def feval(preds, dataset):
+                res = user_metric(dataset.get_label(), preds)
+                return res if isinstance(res, tuple) else ("metric", float(res), False)
+            params["metric"] = "none"
+        elif eval_metric is not None:
             params["metric"] = eval_metric
         X = np.asarray(X, dtype=np.float64)
         self._n_features = X.shape[1]
@@ -155,7 +180,8 @@ class LGBMModel(BaseEstimator):
         cbs.append(callback_mod.record_evaluation(self._evals_result))
         self._Booster = train_fn(params, train_set, num_boost_round=self.n_estimators,
                                  valid_sets=valid_sets or None,
-                                 valid_names=names or None, callbacks=cbs)
+                                 valid_names=names or None, callbacks=cbs,
+                                 fobj=fobj, feval=feval)
         self._best_iteration = self._Booster.best_iteration
         return self
 

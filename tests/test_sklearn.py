@@ -97,3 +97,27 @@ def test_pickle_roundtrip():
                                rtol=1e-12)
     b2 = pickle.loads(pickle.dumps(m.booster_))
     np.testing.assert_allclose(b2.predict(X[:50]), m.booster_.predict(X[:50]), rtol=1e-12)
+
+
+def test_callable_objective_and_metric():
+    """sklearn-style callable objective (y_true, y_pred) -> (grad, hess) and
+    callable eval_metric (reference sklearn.py adapters)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(2000, 5)
+    y = (X[:, 0] > 0).astype(np.float32)
+
+    def logistic_obj(y_true, y_pred):
+        p = 1 / (1 + np.exp(-y_pred))
+        return (p - y_true).astype(np.float32), (p * (1 - p)).astype(np.float32)
+
+    def acc_metric(y_true, y_pred):
+        return "my_acc", float(np.mean((y_pred > 0.5) == y_true)), True
+
+    m = lgb.LGBMClassifier(n_estimators=15, objective=logistic_obj, verbosity=-1)
+    m.fit(X, y)
+    raw = m.predict(X, raw_score=True).ravel()
+    assert (((1 / (1 + np.exp(-raw))) > 0.5) == y).mean() > 0.9
+    m2 = lgb.LGBMClassifier(n_estimators=5, verbosity=-1)
+    m2.fit(X, y, eval_set=[(X, y)], eval_metric=acc_metric)
+    assert "my_acc" in m2.evals_result_["valid_0"]
+    assert m2.evals_result_["valid_0"]["my_acc"][-1] > 0.9
