@@ -116,6 +116,16 @@ void DatasetLoader::ParseFile(const char* filename, std::vector<std::vector<doub
   }
 }
 
+/*! .position sidecar (position-debiased ranking): one integer per row. */
+static std::vector<int32_t> ReadPositionSidecar(const char* filename) {
+  std::vector<int32_t> positions;
+  std::vector<std::string> pl;
+  if (ReadLines((std::string(filename) + ".position").c_str(), &pl)) {
+    for (auto& l : pl) positions.push_back(atoi(l.c_str()));
+  }
+  return positions;
+}
+
 std::unique_ptr<Dataset> DatasetLoader::LoadFromFile(const char* filename, int rank,
                                                      int num_machines) {
   std::vector<std::vector<double>> rows;
@@ -144,6 +154,11 @@ std::unique_ptr<Dataset> DatasetLoader::LoadFromFile(const char* filename, int r
     d->metadata().SetWeights(weights.data(), static_cast<data_size_t>(weights.size()));
   if (!groups.empty())
     d->metadata().SetQuery(groups.data(), static_cast<data_size_t>(groups.size()));
+  {
+    auto positions = ReadPositionSidecar(filename);
+    if (!positions.empty())
+      d->metadata().SetPosition(positions.data(), static_cast<data_size_t>(positions.size()));
+  }
   return d;
 }
 
@@ -163,6 +178,11 @@ std::unique_ptr<Dataset> DatasetLoader::LoadFromFileAlignWithOtherDataset(
     d->metadata().SetWeights(weights.data(), static_cast<data_size_t>(weights.size()));
   if (!groups.empty())
     d->metadata().SetQuery(groups.data(), static_cast<data_size_t>(groups.size()));
+  {
+    auto positions = ReadPositionSidecar(filename);
+    if (!positions.empty())
+      d->metadata().SetPosition(positions.data(), static_cast<data_size_t>(positions.size()));
+  }
   return d;
 }
 

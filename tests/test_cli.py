@@ -151,3 +151,29 @@ def test_cli_predict_variants(cli_model, tmp_path):
                         f"{flag}=true"], capture_output=True, timeout=300, check=True)
         first = out.read_text().splitlines()[0].split("\t")
         assert len(first) == ncols, (flag, len(first))
+
+
+def test_position_sidecar_loading(tmp_path):
+    """A <file>.position sidecar feeds position-debiased lambdarank training."""
+    import subprocess
+    rng = np.random.RandomState(0)
+    rows, qsizes, positions = [], [], []
+    for q in range(80):
+        nq = 10
+        Xq = rng.rand(nq, 4)
+        rel = (Xq[:, 0] > 0.5).astype(int)
+        for i in range(nq):
+            rows.append(f"{rel[i]}\t" + "\t".join(f"{v:.6f}" for v in Xq[i]))
+            positions.append(i)
+        qsizes.append(nq)
+    train = tmp_path / "rank.train"
+    train.write_text("\n".join(rows) + "\n")
+    (tmp_path / "rank.train.query").write_text("\n".join(map(str, qsizes)) + "\n")
+    (tmp_path / "rank.train.position").write_text("\n".join(map(str, positions)) + "\n")
+    out = tmp_path / "m.txt"
+    r = subprocess.run([str(CLI), "task=train", "objective=lambdarank",
+                        f"data={train}", "lambdarank_position_debiasing=true",
+                        "num_trees=10", f"output_model={out}"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert out.exists()
