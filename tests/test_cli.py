@@ -172,8 +172,9 @@ def test_position_sidecar_loading(tmp_path):
     (tmp_path / "rank.train.position").write_text("\n".join(map(str, positions)) + "\n")
     out = tmp_path / "m.txt"
     r = subprocess.run([str(CLI), "task=train", "objective=lambdarank",
-                        f"data={train}", "lambdarank_position_debiasing=true",
-                        "num_trees=10", f"output_model={out}"],
+                        f"data={train}", "num_trees=10", f"output_model={out}"],
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert out.exists()
+    # positions were seen -> debiasing engaged (auto when positions exist)
+    assert "position debiasing enabled" in r.stdout
