@@ -177,4 +177,4 @@ def test_position_sidecar_loading(tmp_path):
     assert r.returncode == 0, r.stdout + r.stderr
     assert out.exists()
     # positions were seen -> debiasing engaged (auto when positions exist)
-    assert "position debiasing enabled" in r.stdout
+    assert "position debiasing enabled" in (r.stdout + r.stderr)
