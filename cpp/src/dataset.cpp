@@ -567,6 +567,8 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
 #pragma omp for schedule(static)
         for (data_size_t i = 0; i < num_data; ++i) {
           const data_size_t r = data_indices ? data_indices[i] : i;
+          if (data_indices && i + 8 < num_data)
+            __builtin_prefetch(base + static_cast<size_t>(data_indices[i + 8]) * stride, 0, 1);
           const uint8_t* row = base + static_cast<size_t>(r) * stride;
           const double g = og[i];
           for (int f = 0; f < nf; ++f) {
@@ -621,6 +623,8 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
 #pragma omp for schedule(static)
       for (data_size_t i = 0; i < num_data; ++i) {
         const data_size_t r = data_indices ? data_indices[i] : i;
+        if (data_indices && i + 8 < num_data)
+          __builtin_prefetch(base + static_cast<size_t>(data_indices[i + 8]) * stride, 0, 1);
         const uint8_t* row = base + static_cast<size_t>(r) * stride;
         const double g = og[i], hv = oh[i];
         for (int f = 0; f < nf; ++f) {
