@@ -146,24 +146,20 @@ class Random {
 class Timer {
  public:
   static Timer& Global() { static Timer t; return t; }
+  /*! runtime-enabled: set MIGBM_TIMETAG=1; zero cost when off beyond one branch */
+  static bool Enabled() {
+    static bool e = getenv("MIGBM_TIMETAG") != nullptr;
+    return e;
+  }
   void Start(const std::string& name) {
-#ifdef MIGBM_TIMETAG
-    starts_[name] = Now();
-#else
-    (void)name;
-#endif
+    if (Enabled()) starts_[name] = Now();
   }
   void Stop(const std::string& name) {
-#ifdef MIGBM_TIMETAG
-    totals_[name] += Now() - starts_[name];
-#else
-    (void)name;
-#endif
+    if (Enabled()) totals_[name] += Now() - starts_[name];
   }
   void Print() const {
-#ifdef MIGBM_TIMETAG
-    for (auto& kv : totals_) fprintf(stderr, "[timer] %s: %.3f s\n", kv.first.c_str(), kv.second);
-#endif
+    for (auto& kv : totals_)
+      fprintf(stderr, "[timer] %-22s %.3f s\n", kv.first.c_str(), kv.second);
   }
   static double Now() {
     struct timespec ts;

@@ -171,6 +171,10 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
     ordered_grad_[i] = gradients_[indices[i]];
     ordered_hess_[i] = hessians_[indices[i]];
   }
+  Timer::Global().Start("hist");
+  struct HistPhaseGuard {
+    ~HistPhaseGuard() { Timer::Global().Stop("hist"); }
+  } hist_phase_guard;
   hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
   std::fill(hist, hist + 2 * train_data_->num_total_bin(), 0.0);
   // ---- histogram mode: empirical col-wise vs row-wise choice, reference
@@ -246,6 +250,10 @@ void SerialTreeLearner::SubtractHistogram(int /*dst_leaf*/, int parent_slot, int
 }
 
 void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
+  Timer::Global().Start("find_best_split");
+  struct SplitPhaseGuard {
+    ~SplitPhaseGuard() { Timer::Global().Stop("find_best_split"); }
+  } split_phase_guard;
   SplitInfo& best = best_split_per_leaf_[leaf];
   best.Reset();
   if (config_->max_depth > 0 && ctx.depth >= config_->max_depth) return;
