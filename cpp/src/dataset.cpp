@@ -18,6 +18,11 @@ void Metadata::Init(data_size_t num_data, bool has_weight, bool has_query) {
 }
 
 void Metadata::SetLabel(const float* label, data_size_t n) {
+  for (data_size_t i = 0; i < n; ++i) {
+    if (!std::isfinite(label[i]))
+      Log::Fatal("Label at index %d is not finite (%f); labels must not contain NaN/Inf",
+                 i, static_cast<double>(label[i]));
+  }
   num_data_ = n;
   label_.assign(label, label + n);
 }
