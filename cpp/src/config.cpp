@@ -316,10 +316,14 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
   if (boosting == "random_forest") boosting = "rf";
 
   if (num_threads > 0) omp_set_num_threads(num_threads);
-  if (verbosity <= -1) Log::Level() = LogLevel::Fatal;
-  else if (verbosity == 0) Log::Level() = LogLevel::Warning;
-  else if (verbosity == 1) Log::Level() = LogLevel::Info;
-  else Log::Level() = LogLevel::Debug;
+  // only an EXPLICIT verbosity touches the global log level: lazily-constructed
+  // datasets with default params must not undo a booster's verbosity=-1
+  if (params.count("verbosity")) {
+    if (verbosity <= -1) Log::Level() = LogLevel::Fatal;
+    else if (verbosity == 0) Log::Level() = LogLevel::Warning;
+    else if (verbosity == 1) Log::Level() = LogLevel::Info;
+    else Log::Level() = LogLevel::Debug;
+  }
 }
 
 std::string Config::SaveHyperParameters() const {
