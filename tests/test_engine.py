@@ -665,3 +665,76 @@ def test_cv_eval_train_metric_and_plotting_smoke():
     import pytest as _pt
     with _pt.raises(ImportError):
         lgb.plot_tree(bst, tree_index=0)  # graphviz absent here (reference parity)
+
+
+def test_multiple_valid_sets_and_names():
+    X, y = _binary_data(n=2000)
+    tr = lgb.Dataset(X, label=y)
+    ev = {}
+    lgb.train({"objective": "binary", "metric": "auc", "verbosity": -1}, tr, 5,
+              valid_sets=[tr.create_valid(X[:500], label=y[:500]),
+                          tr.create_valid(X[500:], label=y[500:])],
+              valid_names=["a", "b"], callbacks=[lgb.record_evaluation(ev)])
+    assert set(ev.keys()) == {"a", "b"}
+
+
+def test_early_stopping_min_delta():
+    X, y = _binary_data()
+    tr = lgb.Dataset(X[:4000], label=y[:4000])
+    bst = lgb.train({"objective": "binary", "metric": "auc", "verbosity": -1}, tr, 100,
+                    valid_sets=[tr.create_valid(X[4000:], label=y[4000:])],
+                    callbacks=[lgb.early_stopping(5, min_delta=0.5, verbose=False)])
+    # a 0.5 AUC min_delta is unreachable -> stops at patience
+    assert bst.best_iteration <= 6
+
+
+def test_reset_parameter_callback_changes_lr():
+    X, y = _binary_data(n=2000)
+    sched = [0.3, 0.2, 0.1, 0.05, 0.02]
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y),
+                    len(sched), callbacks=[lgb.reset_parameter(learning_rate=sched)])
+    # shrinkage shows up in per-tree leaf magnitudes: tree0 built at lr 0.3,
+    # tree4 at lr 0.02 -> much smaller outputs
+    d = bst.dump_model()
+
+    def max_abs_leaf(node):
+        if "leaf_index" in node:
+            return abs(node["leaf_value"])
+        return max(max_abs_leaf(node["left_child"]), max_abs_leaf(node["right_child"]))
+    assert max_abs_leaf(d["tree_info"][4]["tree_structure"]) < \
+        max_abs_leaf(d["tree_info"][0]["tree_structure"])
+
+
+@pytest.mark.parametrize("objective", ["cross_entropy", "cross_entropy_lambda"])
+def test_xentropy_probabilistic_labels(objective):
+    rng = np.random.RandomState(0)
+    X = rng.randn(2000, 5)
+    yq = np.clip((X[:, 0] > 0) + 0.2 * rng.rand(2000), 0, 1).astype(np.float32)
+    bst = lgb.train({"objective": objective, "verbosity": -1}, lgb.Dataset(X, label=yq), 10)
+    p = bst.predict(X)
+    assert np.all(p >= 0)
+    if objective == "cross_entropy":
+        assert np.all(p <= 1)  # probability link; the _lambda link is an intensity
+    assert np.corrcoef(p, yq)[0, 1] > 0.7
+
+
+def test_multiclassova():
+    rng = np.random.RandomState(0)
+    X = rng.randn(3000, 5)
+    y = ((X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)).astype(np.float32)
+    bst = lgb.train({"objective": "multiclassova", "num_class": 3, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 20)
+    pred = bst.predict(X)
+    assert pred.shape == (3000, 3)
+    assert (pred.argmax(axis=1) == y).mean() > 0.7
+
+
+def test_average_precision_metric():
+    X, y = _binary_data(n=3000)
+    tr = lgb.Dataset(X, label=y)
+    ev = {}
+    lgb.train({"objective": "binary", "metric": "average_precision", "verbosity": -1},
+              tr, 10, valid_sets=[tr.create_valid(X, label=y)],
+              callbacks=[lgb.record_evaluation(ev)])
+    ap = ev["valid_0"]["average_precision"]
+    assert ap[-1] > 0.9 and ap[-1] >= ap[0] - 1e-9
