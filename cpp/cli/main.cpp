@@ -66,7 +66,11 @@ class Application {
   void LoadData() {
     if (config_.data.empty()) Log::Fatal("No training data (data=...) specified");
     DatasetLoader loader(config_);
-    train_data_ = loader.LoadFromFile(config_.data.c_str());
+    if (Dataset::IsBinFile(config_.data.c_str())) {
+      train_data_ = Dataset::LoadFromBinFile(config_.data.c_str());
+    } else {
+      train_data_ = loader.LoadFromFile(config_.data.c_str());
+    }
     Log::Info("Loaded train data: %d rows, %d features", train_data_->num_data(),
               train_data_->num_total_features());
     for (auto& vf : config_.valid) {

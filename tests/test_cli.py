@@ -178,3 +178,24 @@ def test_position_sidecar_loading(tmp_path):
     assert out.exists()
     # positions were seen -> debiasing engaged (auto when positions exist)
     assert "position debiasing enabled" in (r.stdout + r.stderr)
+
+
+def test_cli_binary_dataset_roundtrip(tmp_path):
+    """save_binary writes <data>.bin; training from the .bin reproduces the exact
+    same trees as training from text."""
+    import subprocess, shutil
+    work = tmp_path / "ex"
+    shutil.copytree(EXAMPLE, work)
+    m1, m2 = tmp_path / "m1.txt", tmp_path / "m2.txt"
+    subprocess.run([str(CLI), "config=train.conf", "save_binary=true", "num_trees=5",
+                    f"output_model={m1}"], cwd=work, capture_output=True,
+                   timeout=300, check=True)
+    assert (work / "binary.train.bin").exists()
+    subprocess.run([str(CLI), "config=train.conf", "data=binary.train.bin",
+                    "num_trees=5", f"output_model={m2}"], cwd=work,
+                   capture_output=True, timeout=300, check=True)
+
+    def trees(p):
+        t = p.read_text()
+        return t[t.index("Tree=0"):t.index("end of trees")]
+    assert trees(m1) == trees(m2)
