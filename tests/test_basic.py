@@ -370,3 +370,39 @@ def test_validation_errors():
         bst.predict(rng.rand(5, 2))
     # more columns than training is allowed (extras ignored), like the reference
     assert len(bst.predict(np.column_stack([X[:5], np.zeros(5)]))) == 5
+
+
+def test_kitchen_sink_mixed_features():
+    """Integration: dense + sparse + 4-bit + categorical + EFB-bundleable features
+    in ONE dataset, with bagging and a valid set — exercises feature-storage
+    interactions that unit tests miss."""
+    rng = np.random.RandomState(42)
+    n = 6000
+    dense = rng.randn(n, 3)                                   # plain dense
+    sparse = np.zeros((n, 4))                                 # ~5% nonzero
+    mask = rng.rand(n, 4) < 0.05
+    sparse[mask] = rng.rand(mask.sum()) + 1
+    lowcard = rng.randint(0, 9, (n, 2)).astype(float)         # 4-bit packable
+    cat = rng.randint(0, 30, n).astype(float)                 # categorical
+    onehot = np.zeros((n, 6))                                 # EFB-bundleable
+    onehot[np.arange(n), rng.randint(0, 6, n)] = 1.0
+    X = np.column_stack([dense, sparse, lowcard, cat, onehot])
+    eff = rng.randn(30)
+    y = ((dense[:, 0] + 2 * (sparse[:, 0] > 0) + eff[cat.astype(int)] * 0.5 +
+          onehot[:, 0]) > 0.5).astype(np.float32)
+    p = {"objective": "binary", "verbosity": -1, "bagging_freq": 1,
+         "bagging_fraction": 0.7, "categorical_feature": [9], "min_data_in_leaf": 5}
+    tr = lgb.Dataset(X[:5000], label=y[:5000])
+    ev = {}
+    bst = lgb.train({**p, "metric": "auc"}, tr, 40,
+                    valid_sets=[tr.create_valid(X[5000:], label=y[5000:])],
+                    callbacks=[lgb.record_evaluation(ev)])
+    assert ev["valid_0"]["auc"][-1] > 0.9
+    # model roundtrips and predicts identically
+    m = bst.model_to_string()
+    np.testing.assert_allclose(lgb.Booster(model_str=m).predict(X[:200]),
+                               bst.predict(X[:200]), rtol=1e-12)
+    # SHAP consistency on the mixed feature set
+    contrib = bst.predict(X[:50], pred_contrib=True)
+    np.testing.assert_allclose(contrib.sum(axis=1),
+                               bst.predict(X[:50], raw_score=True), rtol=1e-6, atol=1e-6)
