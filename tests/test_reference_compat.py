@@ -219,3 +219,28 @@ def test_subset_categorical_model_cross(tmp_path):
     bst.save_model(str(f))
     np.testing.assert_allclose(bst.predict(X[:500]), _ref_predict(f, X[:500], tmp_path),
                                rtol=1e-9, atol=1e-12)
+
+
+def test_kitchen_sink_model_cross(tmp_path):
+    """The mixed-storage model (sparse + 4-bit + categorical + EFB features) must
+    predict identically in the reference implementation."""
+    rng = np.random.RandomState(42)
+    n = 4000
+    dense = rng.randn(n, 3)
+    sparse = np.zeros((n, 4))
+    mask = rng.rand(n, 4) < 0.05
+    sparse[mask] = rng.rand(mask.sum()) + 1
+    lowcard = rng.randint(0, 9, (n, 2)).astype(float)
+    cat = rng.randint(0, 30, n).astype(float)
+    onehot = np.zeros((n, 6))
+    onehot[np.arange(n), rng.randint(0, 6, n)] = 1.0
+    X = np.column_stack([dense, sparse, lowcard, cat, onehot])
+    eff = rng.randn(30)
+    y = ((dense[:, 0] + 2 * (sparse[:, 0] > 0) + eff[cat.astype(int)] * 0.5) > 0.5)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5,
+                     "categorical_feature": [9]},
+                    lgb.Dataset(X, label=y.astype(np.float32)), 25)
+    f = tmp_path / "ks.txt"
+    bst.save_model(str(f))
+    np.testing.assert_allclose(bst.predict(X[:500]), _ref_predict(f, X[:500], tmp_path),
+                               rtol=1e-9, atol=1e-12)
