@@ -5,7 +5,7 @@ import copy
 import numpy as np
 
 from . import callback as callback_mod
-from .basic import Booster, Dataset, LightGBMError
+from .basic import Booster, Dataset
 
 __all__ = ["train", "cv", "CVBooster"]
 
@@ -64,7 +64,6 @@ def train(params, train_set, num_boost_round=100, valid_sets=None, valid_names=N
             other = Booster(model_str=init_model)
         else:
             other = Booster(model_file=init_model)
-        import ctypes
         from .basic import _LIB, _safe_call
         _safe_call(_LIB.LGBM_BoosterMerge(booster._handle, other._handle))
 

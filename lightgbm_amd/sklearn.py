@@ -1,5 +1,4 @@
 """scikit-learn estimator wrappers (parity target: reference python-package/lightgbm/sklearn.py)."""
-import copy
 
 import numpy as np
 
