@@ -184,8 +184,40 @@ std::string GBDT::DumpModel(int start_iter, int num_iter, int feature_importance
     ss << "\"" << feature_names_[i] << "\"";
   }
   ss << "],";
-  ss << "\"monotone_constraints\":[],";
-  ss << "\"feature_infos\":{},";
+  ss << "\"monotone_constraints\":[";
+  if (config_ != nullptr) {
+    for (size_t i = 0; i < config_->monotone_constraints.size(); ++i) {
+      if (i) ss << ",";
+      ss << config_->monotone_constraints[i];
+    }
+  }
+  ss << "],";
+  // feature_infos: {"name": {"min_value":..,"max_value":..,"values":[..]}}
+  // parsed back from the "[lo:hi]" / "cat1:cat2:..." strings of the text format
+  ss << "\"feature_infos\":{";
+  bool first_fi = true;
+  for (size_t i = 0; i < feature_infos_.size() && i < feature_names_.size(); ++i) {
+    const std::string& fi = feature_infos_[i];
+    if (fi == "none" || fi.empty()) continue;
+    if (!first_fi) ss << ",";
+    first_fi = false;
+    ss << "\"" << feature_names_[i] << "\":{";
+    if (fi.front() == '[') {
+      auto colon = fi.find(':');
+      ss << "\"min_value\":" << fi.substr(1, colon - 1) << ",\"max_value\":"
+         << fi.substr(colon + 1, fi.size() - colon - 2) << ",\"values\":[]";
+    } else {
+      ss << "\"min_value\":0,\"max_value\":0,\"values\":[";
+      auto cats = Common::SplitAny(fi.c_str(), ":");
+      for (size_t c = 0; c < cats.size(); ++c) {
+        if (c) ss << ",";
+        ss << cats[c];
+      }
+      ss << "]";
+    }
+    ss << "}";
+  }
+  ss << "},";
   ss << "\"tree_info\":[";
   for (int t = start_iter * num_tree_per_iteration_;
        t < end_iter * num_tree_per_iteration_ && t < static_cast<int>(models_.size()); ++t) {
