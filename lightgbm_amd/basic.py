@@ -1059,25 +1059,42 @@ class Booster:
         model = self.dump_model()
         rows = []
 
-        def walk(tree_index, node, parent=None):
+        feat_names = self.feature_name()
+
+        def walk(tree_index, node, parent=None, depth=1):
             if "leaf_index" in node:
-                rows.append(dict(tree_index=tree_index,
+                rows.append(dict(tree_index=tree_index, node_depth=depth,
                                  node_index=f"{tree_index}-L{node['leaf_index']}",
+                                 left_child=None, right_child=None,
                                  parent_index=parent, split_feature=None,
-                                 threshold=None, decision_type=None,
+                                 split_gain=None, threshold=None, decision_type=None,
+                                 missing_direction=None, missing_type=None,
                                  value=node["leaf_value"], count=node.get("leaf_count"),
                                  weight=node.get("leaf_weight")))
                 return
             ni = f"{tree_index}-S{node['split_index']}"
-            rows.append(dict(tree_index=tree_index, node_index=ni, parent_index=parent,
-                             split_feature=node["split_feature"],
+
+            def child_index(c):
+                return (f"{tree_index}-L{c['leaf_index']}" if "leaf_index" in c
+                        else f"{tree_index}-S{c['split_index']}")
+            f = node["split_feature"]
+            rows.append(dict(tree_index=tree_index, node_depth=depth, node_index=ni,
+                             left_child=child_index(node["left_child"]),
+                             right_child=child_index(node["right_child"]),
+                             parent_index=parent,
+                             split_feature=feat_names[f] if isinstance(f, int) and
+                             f < len(feat_names) else f,
+                             split_gain=node.get("split_gain"),
                              threshold=node["threshold"],
                              decision_type=node["decision_type"],
+                             missing_direction="left" if node.get("default_left")
+                             else "right",
+                             missing_type=node.get("missing_type"),
                              value=node.get("internal_value"),
                              count=node.get("internal_count"),
                              weight=node.get("internal_weight")))
-            walk(tree_index, node["left_child"], ni)
-            walk(tree_index, node["right_child"], ni)
+            walk(tree_index, node["left_child"], ni, depth + 1)
+            walk(tree_index, node["right_child"], ni, depth + 1)
 
         for t in model["tree_info"]:
             walk(t["tree_index"], t["tree_structure"])
