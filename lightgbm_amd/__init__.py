@@ -16,8 +16,10 @@ except ImportError:
     _SKLEARN_EXPORTS = []
 
 try:
-    from .plotting import create_tree_digraph, plot_importance, plot_metric, plot_tree
-    _PLOT_EXPORTS = ["plot_importance", "plot_metric", "plot_tree", "create_tree_digraph"]
+    from .plotting import create_tree_digraph, plot_importance, plot_metric, \
+        plot_split_value_histogram, plot_tree
+    _PLOT_EXPORTS = ["plot_importance", "plot_metric", "plot_tree", "create_tree_digraph",
+                     "plot_split_value_histogram"]
 except ImportError:
     _PLOT_EXPORTS = []
 

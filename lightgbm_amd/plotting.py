@@ -147,3 +147,35 @@ def plot_tree(booster, ax=None, tree_index=0, figsize=None, dpi=None, precision=
     ax.imshow(img)
     ax.axis("off")
     return ax
+
+
+def plot_split_value_histogram(booster, feature, bins=None, ax=None, width_coef=0.8,
+                               xlim=None, ylim=None, title="Split value histogram for "
+                               "feature with @index/name@ @feature@",
+                               xlabel="Feature split value", ylabel="Count", figsize=None,
+                               dpi=None, grid=True, **kwargs):
+    """Plot the histogram of split threshold values used for `feature`
+    (parity: reference plotting.plot_split_value_histogram)."""
+    if not MATPLOTLIB_INSTALLED:
+        raise ImportError("matplotlib is required for plotting")
+    import matplotlib.pyplot as plt
+    hist, edges = booster.get_split_value_histogram(feature, bins=bins)
+    if hist.sum() == 0:
+        raise ValueError(f"Cannot plot split value histogram: feature {feature} "
+                         "was not used in splitting")
+    if ax is None:
+        _, ax = plt.subplots(1, 1, figsize=figsize, dpi=dpi)
+    centers = (edges[:-1] + edges[1:]) / 2
+    ax.bar(centers, hist, width=width_coef * (edges[1] - edges[0]))
+    if xlim is not None:
+        ax.set_xlim(xlim)
+    if ylim is not None:
+        ax.set_ylim(ylim)
+    if title:
+        ax.set_title(title.replace("@index/name@", "name" if isinstance(feature, str)
+                                   else "index").replace("@feature@", str(feature)))
+    ax.set_xlabel(xlabel)
+    ax.set_ylabel(ylabel)
+    if grid:
+        ax.grid(True)
+    return ax
