@@ -16,6 +16,12 @@ except ImportError:
     _SKLEARN_EXPORTS = []
 
 try:
+    from .dask import DaskLGBMClassifier, DaskLGBMRanker, DaskLGBMRegressor
+    _DASK_EXPORTS = ["DaskLGBMRegressor", "DaskLGBMClassifier", "DaskLGBMRanker"]
+except ImportError:
+    _DASK_EXPORTS = []
+
+try:
     from .plotting import create_tree_digraph, plot_importance, plot_metric, \
         plot_split_value_histogram, plot_tree
     _PLOT_EXPORTS = ["plot_importance", "plot_metric", "plot_tree", "create_tree_digraph",
@@ -30,4 +36,4 @@ __all__ = [
     "train", "cv", "CVBooster",
     "early_stopping", "log_evaluation", "record_evaluation", "reset_parameter",
     "EarlyStopException",
-] + _SKLEARN_EXPORTS + _PLOT_EXPORTS
+] + _SKLEARN_EXPORTS + _DASK_EXPORTS + _PLOT_EXPORTS
