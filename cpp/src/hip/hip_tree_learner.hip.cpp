@@ -80,6 +80,7 @@ struct LogEntry {
 
 struct LeafStat {
   double sum_g, sum_h;
+  double parent_out;  // hessian-weighted parent output (path smoothing)
   int cnt;  // exact GLOBAL row count (allreduced in multi-GPU)
   int pad;
 };
@@ -92,6 +93,7 @@ struct GainParams {
   float bynode_frac;   // feature_fraction_bynode (1.0 = off)
   int extra_trees;     // 1 = evaluate one hashed random threshold per feature
   uint32_t rng_seed;   // per-tree seed component
+  double smooth;       // path_smooth (0 = off; smoothing uses LeafStat.parent_out)
   // categorical scan (parity: FindBestThresholdCategorical CPU oracle)
   double cat_l2, cat_smooth;
   int max_cat_to_onehot, max_cat_threshold;
@@ -151,12 +153,25 @@ __device__ __forceinline__ double d_split_gain_l2(double gl, double hl, double g
   const double lo = d_leaf_out_l2(gl, hl, p, l2v), ro = d_leaf_out_l2(gr, hr, p, l2v);
   return d_gain_out_l2(gl, hl, lo, p, l2v) + d_gain_out_l2(gr, hr, ro, p, l2v);
 }
+/*! path smoothing (CPU-oracle parity): out*(n/a)/(n/a+1) + parent/(n/a+1) */
+__device__ __forceinline__ double d_smooth_out(double out, double n, double parent_out,
+                                               const GainParams& p) {
+  if (p.smooth <= 0.0) return out;
+  const double na = n / p.smooth;
+  return out * na / (na + 1.0) + parent_out / (na + 1.0);
+}
 __device__ __forceinline__ double d_leaf_gain(double g, double h, const GainParams& p) {
   if (p.mds <= 0.0) {
     double s = d_thl1(g, p.l1);
     return s * s / (h + p.l2);
   }
   double out = d_leaf_out(g, h, p);
+  return d_gain_out(g, h, out, p);
+}
+__device__ __forceinline__ double d_leaf_gain_sm(double g, double h, double n,
+                                                 double parent_out, const GainParams& p) {
+  if (p.smooth <= 0.0) return d_leaf_gain(g, h, p);
+  const double out = d_smooth_out(d_leaf_out(g, h, p), n, parent_out, p);
   return d_gain_out(g, h, out, p);
 }
 
@@ -442,8 +457,10 @@ __global__ void k_root_sums(const uint32_t* __restrict__ idx, int cnt,
   }
 }
 
-__global__ void k_set_root_global_cnt(LeafStat* stats, const int64_t* gbuf) {
+__global__ void k_set_root_global_cnt(LeafStat* stats, const int64_t* gbuf,
+                                      GainParams p) {
   stats[0].cnt = static_cast<int>(gbuf[0]);
+  stats[0].parent_out = d_leaf_out(stats[0].sum_g, stats[0].sum_h, p);
 }
 
 // ------------------------------------------------------------------ best split
@@ -498,7 +515,8 @@ __global__ void __launch_bounds__(64) k_best_feat(
   const int num_data = st.cnt;
   if (num_data < 2 * p.min_data) return;
   const double cnt_factor = (num_data > 0 && sum_h > 0) ? num_data / sum_h : 1.0;
-  const double parent_gain = d_leaf_gain(sum_g, sum_h, p);
+  const double leaf_parent_out = st.parent_out;
+  const double parent_gain = d_leaf_gain_sm(sum_g, sum_h, num_data, leaf_parent_out, p);
   const double min_gain_shift = parent_gain + p.min_gain_to_split;
 
   const float* fh = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
@@ -699,7 +717,8 @@ __global__ void __launch_bounds__(64) k_best_feat(
         const int rc = num_data - lc;
         if (hl < p.min_hess || lc < p.min_data) continue;
         if (hr < p.min_hess || rc < p.min_data) continue;
-        double lo = d_leaf_out(gl, hl, p), ro = d_leaf_out(gr, hr, p);
+        double lo = d_smooth_out(d_leaf_out(gl, hl, p), lc, leaf_parent_out, p);
+        double ro = d_smooth_out(d_leaf_out(gr, hr, p), rc, leaf_parent_out, p);
         lo = fmin(fmax(lo, blo), bhi);
         ro = fmin(fmax(ro, blo), bhi);
         if (mc > 0 && lo > ro) continue;
@@ -745,8 +764,11 @@ __global__ void __launch_bounds__(64) k_best_feat(
       rec.left_out = d_leaf_out(best_lg, best_lh, p);
       rec.right_out = d_leaf_out(rg, rh, p);
     } else {
-      rec.left_out = fmin(fmax(d_leaf_out(best_lg, best_lh, p), blo), bhi);
-      rec.right_out = fmin(fmax(d_leaf_out(rg, rh, p), blo), bhi);
+      const int w_lc = static_cast<int>(best_lh * cnt_factor + 0.5);
+      rec.left_out = fmin(fmax(d_smooth_out(d_leaf_out(best_lg, best_lh, p), w_lc,
+                                            leaf_parent_out, p), blo), bhi);
+      rec.right_out = fmin(fmax(d_smooth_out(d_leaf_out(rg, rh, p), num_data - w_lc,
+                                             leaf_parent_out, p), blo), bhi);
     }
     rec.left_cnt = static_cast<int>(best_lh * cnt_factor + 0.5);
     rec.right_cnt = num_data - rec.left_cnt;
@@ -1278,12 +1300,17 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
   const int gl = static_cast<int>(gbuf[0]);
   const int gr = parent.cnt - gl;
   const SplitRec w = *winner;
+  const double right_h = parent.sum_h - w.left_h;
+  const double po = (w.left_out * w.left_h + w.right_out * right_h) /
+                    fmax(w.left_h + right_h, 1e-15);
   stats[L].sum_g = w.left_g;
   stats[L].sum_h = w.left_h;
   stats[L].cnt = gl;
+  stats[L].parent_out = po;
   stats[R].sum_g = parent.sum_g - w.left_g;
-  stats[R].sum_h = parent.sum_h - w.left_h;
+  stats[R].sum_h = right_h;
   stats[R].cnt = gr;
+  stats[R].parent_out = po;
   const int old_slot = leaf_slot[L];
   if (gl <= gr) {
     leaf_slot[L] = spare_slot;
@@ -2120,6 +2147,7 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.min_hess = config_->min_sum_hessian_in_leaf;
   p.min_gain_to_split = config_->min_gain_to_split;
   p.min_data = config_->min_data_in_leaf;
+  p.smooth = config_->path_smooth;
   p.bynode_frac = static_cast<float>(config_->feature_fraction_bynode);
   p.cat_l2 = config_->cat_l2;
   p.cat_smooth = config_->cat_smooth;
@@ -2200,8 +2228,16 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                           ncclFloat64, ncclSum, comm.comm, stream_));
     NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
                           stream_));
+  }
+  {
+    // global root count + the root's own output (the smoothing parent for depth-1
+    // candidates); single-GPU gbuf already holds the local count
+    hipk::GainParams rp = {};
+    rp.l1 = config_->lambda_l1;
+    rp.l2 = config_->lambda_l2;
+    rp.mds = config_->max_delta_step;
     hipLaunchKernelGGL(hipk::k_set_root_global_cnt, dim3(1), dim3(1), 0, stream_,
-                       d_leaf_stats_.ptr, d_gbuf_.ptr);
+                       d_leaf_stats_.ptr, d_gbuf_.ptr, rp);
   }
 
   // root histogram + best split (leafB disabled via literal -1 pointer semantics)
@@ -2433,7 +2469,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
     if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||
         !cfg->cegb_penalty_feature_lazy.empty())
       return "cost-effective gradient boosting (cegb_*)";
-    if (cfg->path_smooth > 0.0) return "path_smooth";
     return nullptr;
   };
   if (const char* what = unsupported()) {

@@ -295,3 +295,18 @@ def test_gpu_interaction_constraints():
         used = set()
         feats(t["tree_structure"], used)
         assert used <= {0, 1} or used <= {2, 3}, used
+
+
+def test_gpu_path_smooth():
+    """path_smooth runs in the device gain scan (LeafStat.parent_out chain);
+    smoothed models differ from unsmoothed and keep quality."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(50000, 8)
+    y = (X[:, 0] + 0.5 * X[:, 1] + 0.3 * rng.randn(50000)).astype(np.float32)
+    base = lgb.train({"objective": "regression", "device_type": "cuda",
+                      "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    sm = lgb.train({"objective": "regression", "device_type": "cuda",
+                    "path_smooth": 10.0, "verbosity": -1}, lgb.Dataset(X, label=y), 20)
+    assert not np.allclose(base.predict(X[:100]), sm.predict(X[:100]))
+    r2 = 1 - np.mean((sm.predict(X) - y) ** 2) / np.var(y)
+    assert r2 > 0.8
