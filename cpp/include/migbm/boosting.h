@@ -42,7 +42,9 @@ class SampleStrategy {
 class GBDT {
  public:
   GBDT() = default;
-  virtual ~GBDT() = default;
+  virtual ~GBDT() {
+    if (Timer::Enabled()) Timer::Global().Print();
+  }
 
   virtual void Init(const Config* config, const Dataset* train_data,
                     const ObjectiveFunction* objective,

@@ -95,6 +95,10 @@ class DataPartition {
 
   /*! Stable-partition rows of `leaf` into (leaf, right_leaf) by predicate go_left(row). */
   void Split(int leaf, int right_leaf, const std::function<bool(data_size_t)>& go_left);
+  /*! fast path for plain dense-uint8 numeric splits: the per-row decision is
+   *  inlined (no std::function indirection) — the dominant case. */
+  void SplitDenseU8(int leaf, int right_leaf, const uint8_t* col, uint32_t thr,
+                    int nan_bin, bool default_left);
 
   data_size_t used_cnt() const { return used_cnt_; }
 

@@ -449,7 +449,6 @@ void GBDT::Train(int snapshot_freq, const std::string& model_output_path) {
                       (model_output_path + ".snapshot_iter_" + std::to_string(i + 1)).c_str());
     }
   }
-  if (Timer::Enabled()) Timer::Global().Print();
 }
 
 // ------------------------------------------------------------------ prediction

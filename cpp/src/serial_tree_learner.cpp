@@ -57,6 +57,66 @@ void DataPartition::Split(int leaf, int right_leaf,
   leaf_count_[right_leaf] = cnt - total_left;
 }
 
+void DataPartition::SplitDenseU8(int leaf, int right_leaf, const uint8_t* col, uint32_t thr,
+                                 int nan_bin, bool default_left) {
+  const data_size_t begin = leaf_begin_[leaf];
+  const data_size_t cnt = leaf_count_[leaf];
+  data_size_t* idx = indices_.data() + begin;
+  data_size_t* tmp = temp_.data() + begin;
+  const int max_threads = omp_get_max_threads();
+  int nblock;
+  data_size_t bsize;
+  Threading::BlockInfo<data_size_t>(max_threads, cnt, 1024, &nblock, &bsize);
+  std::vector<data_size_t> left_cnts(nblock, 0);
+  const uint8_t nb = nan_bin >= 0 ? static_cast<uint8_t>(nan_bin) : 255;
+  const bool has_nan = nan_bin >= 0;
+#pragma omp parallel for schedule(static, 1)
+  for (int b = 0; b < nblock; ++b) {
+    data_size_t s = b * bsize, e = std::min(cnt, s + bsize);
+    data_size_t lc = 0;
+    if (!has_nan) {
+      for (data_size_t i = s; i < e; ++i) lc += col[idx[i]] <= thr ? 1 : 0;
+    } else {
+      for (data_size_t i = s; i < e; ++i) {
+        const uint8_t v = col[idx[i]];
+        lc += (v == nb ? default_left : v <= thr) ? 1 : 0;
+      }
+    }
+    left_cnts[b] = lc;
+  }
+  std::vector<data_size_t> loff(nblock + 1, 0), roff(nblock + 1, 0);
+  for (int b = 0; b < nblock; ++b) loff[b + 1] = loff[b] + left_cnts[b];
+  const data_size_t total_left = loff[nblock];
+  roff[0] = total_left;
+  for (int b = 0; b < nblock; ++b) {
+    data_size_t s = b * bsize, e = std::min(cnt, s + bsize);
+    roff[b + 1] = roff[b] + (e - s) - left_cnts[b];
+  }
+#pragma omp parallel for schedule(static, 1)
+  for (int b = 0; b < nblock; ++b) {
+    data_size_t s = b * bsize, e = std::min(cnt, s + bsize);
+    data_size_t lw = loff[b], rw = roff[b];
+    if (!has_nan) {
+      for (data_size_t i = s; i < e; ++i) {
+        const data_size_t r = idx[i];
+        if (col[r] <= thr) tmp[lw++] = r;
+        else tmp[rw++] = r;
+      }
+    } else {
+      for (data_size_t i = s; i < e; ++i) {
+        const data_size_t r = idx[i];
+        const uint8_t v = col[r];
+        if (v == nb ? default_left : v <= thr) tmp[lw++] = r;
+        else tmp[rw++] = r;
+      }
+    }
+  }
+  std::copy(tmp, tmp + cnt, idx);
+  leaf_count_[leaf] = total_left;
+  leaf_begin_[right_leaf] = begin + total_left;
+  leaf_count_[right_leaf] = cnt - total_left;
+}
+
 // ------------------------------------------------------------------ SerialTreeLearner
 void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   train_data_ = train_data;
@@ -517,7 +577,18 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     const int right_leaf = num_leaves;
 
     // partition first so the tree records exact child counts
-    partition_.Split(best_leaf, right_leaf, MakeGoLeft(s));
+    {
+      // fast path: plain dense-uint8 numeric split (the dominant case)
+      const BinMapper* sm = train_data_->FeatureBinMapper(s.feature);
+      const auto& scol = train_data_->column(train_data_->feature_column(s.feature));
+      if (s.cat_bitset_inner.empty() && !train_data_->feature_bundled(s.feature) &&
+          !scol.is_sparse() && !scol.is4() && !scol.is16()) {
+        partition_.SplitDenseU8(best_leaf, right_leaf, scol.data8(), s.threshold,
+                                sm->nan_bin(), s.default_left);
+      } else {
+        partition_.Split(best_leaf, right_leaf, MakeGoLeft(s));
+      }
+    }
     data_size_t left_cnt_actual = partition_.leaf_count(best_leaf);
     data_size_t right_cnt_actual = partition_.leaf_count(right_leaf);
     GlobalChildCounts(&left_cnt_actual, &right_cnt_actual);
