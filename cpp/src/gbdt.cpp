@@ -253,7 +253,9 @@ bool GBDT::TrainOneIter(const score_t* gradients, const score_t* hessians) {
       hessians = hessians_.data();
     } else {
       if (dev) tree_learner_->DownloadTrainScore(train_score_.data());
+      Timer::Global().Start("boosting");
       objective_->GetGradients(train_score_.data(), gradients_.data(), hessians_.data());
+      Timer::Global().Stop("boosting");
       gradients = gradients_.data();
       hessians = hessians_.data();
     }
@@ -273,7 +275,9 @@ bool GBDT::TrainOneIter(const score_t* gradients, const score_t* hessians) {
     std::unique_ptr<Tree> new_tree(new Tree(2));
     tree_learner_->SetClassOffset(c);
     if (objective_ == nullptr || objective_->ClassNeedTrain(c)) {
+      Timer::Global().Start("train_tree");
       new_tree.reset(tree_learner_->Train(gradients + off, hessians + off, models_.empty()));
+      Timer::Global().Stop("train_tree");
     }
     if (new_tree->num_leaves() > 1) {
       should_continue = true;
@@ -284,7 +288,9 @@ bool GBDT::TrainOneIter(const score_t* gradients, const score_t* hessians) {
                                        train_score_.data() + off);
       }
       new_tree->Shrinkage(shrinkage_rate_);
+      Timer::Global().Start("update_score");
       UpdateScore(new_tree.get(), c);
+      Timer::Global().Stop("update_score");
       if (std::fabs(init_scores[c]) > kEpsilon) new_tree->AddBias(init_scores[c]);
     } else {
       // no splits: constant tree with objective-specific output
