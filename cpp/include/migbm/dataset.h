@@ -207,7 +207,15 @@ class Dataset {
    *  SerialTreeLearner::ComputeHistogram). Covers every feature. */
   void ConstructHistogramsRowWise(const data_size_t* data_indices, data_size_t num_data,
                                   const score_t* ordered_grad,
-                                  const score_t* ordered_hess, hist_t* hist) const;
+                                  const score_t* ordered_hess, hist_t* hist) const {
+    ConstructHistogramsRowWise(data_indices, num_data, ordered_grad, ordered_hess, hist,
+                               false);
+  }
+  /*! row_indexed=true: grad/hess arrays are indexed by ROW id (the full per-row
+   *  arrays) — skips the ordered-gather pass entirely. */
+  void ConstructHistogramsRowWise(const data_size_t* data_indices, data_size_t num_data,
+                                  const score_t* grad, const score_t* hess, hist_t* hist,
+                                  bool row_indexed) const;
   /*! default bin of a sparse feature (the bin MaterializeDefaultBins must
    *  reconstruct from leaf totals), or -1 for dense features. */
   int feature_sparse_default_bin(int f) const {

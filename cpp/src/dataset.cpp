@@ -533,7 +533,8 @@ void Dataset::AddFeaturesFrom(const Dataset* other) {
 
 void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
                                          data_size_t num_data, const score_t* og,
-                                         const score_t* oh, hist_t* hist) const {
+                                         const score_t* oh, hist_t* hist,
+                                         bool row_indexed) const {
   const RowMajorView& view = GetRowMajorView();
   const int nf = num_features();
   const size_t hist_elems = 2 * static_cast<size_t>(num_total_bin_);
@@ -542,16 +543,19 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
   // (plain L2/ranking without weights/GOSS), accumulate counts instead of
   // hessians and expand at merge — halves the private-histogram write traffic
   // (reference dense_bin.hpp USE_HESSIAN=false analogue)
+  auto oh_at = [&](data_size_t i) {
+    return oh[row_indexed ? (data_indices ? data_indices[i] : i) : i];
+  };
   bool const_hess = num_data > 0;
   if (const_hess) {
-    const score_t h0 = oh[0];
+    const score_t h0 = oh_at(0);
     bool ok = true;
 #pragma omp parallel for schedule(static) reduction(&& : ok)
-    for (data_size_t i = 0; i < num_data; ++i) ok = ok && oh[i] == h0;
+    for (data_size_t i = 0; i < num_data; ++i) ok = ok && oh_at(i) == h0;
     const_hess = ok;
   }
   if (const_hess) {
-    const double h0 = oh[0];
+    const double h0 = oh_at(0);
     std::vector<int> counts(num_total_bin_, 0);
     static thread_local std::vector<double> priv_g;
     static thread_local std::vector<int> priv_c;
@@ -575,7 +579,7 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
           if (data_indices && i + 8 < num_data)
             __builtin_prefetch(base + static_cast<size_t>(data_indices[i + 8]) * stride, 0, 1);
           const uint8_t* row = base + static_cast<size_t>(r) * stride;
-          const double g = og[i];
+          const double g = og[row_indexed ? r : i];
           for (int f = 0; f < nf; ++f) {
             const uint32_t b = hist_offsets_[f] + row[f];
             gp[b] += g;
@@ -589,7 +593,7 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
         for (data_size_t i = 0; i < num_data; ++i) {
           const data_size_t r = data_indices ? data_indices[i] : i;
           const uint16_t* row = base + static_cast<size_t>(r) * stride;
-          const double g = og[i];
+          const double g = og[row_indexed ? r : i];
           for (int f = 0; f < nf; ++f) {
             const uint32_t b = hist_offsets_[f] + row[f];
             gp[b] += g;
@@ -631,7 +635,7 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
         if (data_indices && i + 8 < num_data)
           __builtin_prefetch(base + static_cast<size_t>(data_indices[i + 8]) * stride, 0, 1);
         const uint8_t* row = base + static_cast<size_t>(r) * stride;
-        const double g = og[i], hv = oh[i];
+        const double g = og[row_indexed ? r : i], hv = oh[row_indexed ? r : i];
         for (int f = 0; f < nf; ++f) {
           const uint32_t b = (hist_offsets_[f] + row[f]) << 1;
           h[b] += g;
@@ -645,7 +649,7 @@ void Dataset::ConstructHistogramsRowWise(const data_size_t* data_indices,
       for (data_size_t i = 0; i < num_data; ++i) {
         const data_size_t r = data_indices ? data_indices[i] : i;
         const uint16_t* row = base + static_cast<size_t>(r) * stride;
-        const double g = og[i], hv = oh[i];
+        const double g = og[row_indexed ? r : i], hv = oh[row_indexed ? r : i];
         for (int f = 0; f < nf; ++f) {
           const uint32_t b = (hist_offsets_[f] + row[f]) << 1;
           h[b] += g;

@@ -225,12 +225,18 @@ std::vector<int8_t> SerialTreeLearner::SampleFeatures(bool per_node) {
 
 void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
                                          const data_size_t* indices) {
-  // gather ordered gradients
+  // ordered-gradient gather, deferred: the row-wise histogram reads gradients by
+  // row id directly and skips this pass entirely
+  bool gathered = false;
+  auto gather = [&]() {
+    if (gathered) return;
+    gathered = true;
 #pragma omp parallel for schedule(static)
-  for (data_size_t i = 0; i < cnt; ++i) {
-    ordered_grad_[i] = gradients_[indices[i]];
-    ordered_hess_[i] = hessians_[indices[i]];
-  }
+    for (data_size_t i = 0; i < cnt; ++i) {
+      ordered_grad_[i] = gradients_[indices[i]];
+      ordered_hess_[i] = hessians_[indices[i]];
+    }
+  };
   Timer::Global().Start("hist");
   struct HistPhaseGuard {
     ~HistPhaseGuard() { Timer::Global().Stop("hist"); }
@@ -254,11 +260,12 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
     const int mode = hist_trials_done_;  // 0 = col first, 1 = row second
     const auto t0 = std::chrono::steady_clock::now();
     if (mode == 0) {
+      gather();
       train_data_->ConstructHistograms(is_feature_used_, indices, cnt,
                                        ordered_grad_.data(), ordered_hess_.data(), hist);
     } else {
-      train_data_->ConstructHistogramsRowWise(indices, cnt, ordered_grad_.data(),
-                                              ordered_hess_.data(), hist);
+      train_data_->ConstructHistogramsRowWise(indices, cnt, gradients_, hessians_, hist,
+                                              /*row_indexed=*/true);
     }
     hist_trial_time_[mode] =
         std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
@@ -268,8 +275,8 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
   }
   if (hist_mode_ == 1 || (hist_mode_ < 0 && hist_trials_done_ == 1)) {
     // locked row-wise, or mid-trial small leaves while row was measured last
-    train_data_->ConstructHistogramsRowWise(indices, cnt, ordered_grad_.data(),
-                                            ordered_hess_.data(), hist);
+    train_data_->ConstructHistogramsRowWise(indices, cnt, gradients_, hessians_, hist,
+                                            /*row_indexed=*/true);
     return;
   }
   if ((train_data_->has_bundles() || train_data_->has_sparse()) &&
@@ -277,12 +284,13 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
     double sg = 0.0, sh = 0.0;
 #pragma omp parallel for schedule(static) reduction(+ : sg, sh)
     for (data_size_t i = 0; i < cnt; ++i) {
-      sg += ordered_grad_[i];
-      sh += ordered_hess_[i];
+      sg += gradients_[indices[i]];
+      sh += hessians_[indices[i]];
     }
     local_leaf_sum_g_ = sg;
     local_leaf_sum_h_ = sh;
   }
+  gather();
   if (train_data_->has_sparse()) {
     // membership mask for the sparse-column nonzero scan; cleared after use so
     // the buffer never needs a full memset
