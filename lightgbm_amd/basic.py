@@ -327,7 +327,10 @@ class Dataset:
             self._handle = out
         else:
             if PANDAS_INSTALLED and isinstance(self.data, pd_DataFrame):
-                arr, self.pandas_categorical = _pandas_to_float64(self.data)
+                # a valid set aligns its category codes to the reference's mapping
+                ref_cats = getattr(self.reference, "pandas_categorical", None) \
+                    if self.reference is not None else None
+                arr, self.pandas_categorical = _pandas_to_float64(self.data, ref_cats)
             else:
                 arr = _to_2d_float64(self.data)
             out = ctypes.c_void_p()

@@ -354,6 +354,15 @@ def test_pandas_categorical_dtype():
     df3 = df.head(10).copy()
     df3["b"] = pd.Categorical(["w"] * 10, categories=["w"])
     assert np.isfinite(b2.predict(df3)).all()
+    # a VALID set with re-ordered categories aligns its codes to the train mapping
+    tr = lgb.Dataset(df[:3000], label=y[:3000])
+    dv = df[3000:].copy()
+    dv["b"] = pd.Categorical(dv["b"], categories=["z", "y", "x"])
+    ev = {}
+    lgb.train({"objective": "binary", "metric": "auc", "verbosity": -1}, tr, 10,
+              valid_sets=[tr.create_valid(dv, label=y[3000:])],
+              callbacks=[lgb.record_evaluation(ev)])
+    assert ev["valid_0"]["auc"][-1] > 0.98
 
 
 def test_validation_errors():
