@@ -146,8 +146,12 @@ class LGBMModel(BaseEstimator):
             params["metric"] = "none"
         elif eval_metric is not None:
             params["metric"] = eval_metric
-        X = np.asarray(X, dtype=np.float64)
-        self._n_features = X.shape[1]
+        from .compat import PANDAS_INSTALLED, pd_DataFrame
+        if PANDAS_INSTALLED and isinstance(X, pd_DataFrame):
+            self._n_features = X.shape[1]  # Dataset handles category-dtype mapping
+        else:
+            X = np.asarray(X, dtype=np.float64)
+            self._n_features = X.shape[1]
         sw = sample_weight
         if self.class_weight is not None and self._classes is not None:
             cw = self.class_weight
@@ -169,8 +173,10 @@ class LGBMModel(BaseEstimator):
             for i, (vx, vy) in enumerate(eval_set):
                 vw = eval_sample_weight[i] if eval_sample_weight else None
                 vg = eval_group[i] if eval_group else None
-                valid_sets.append(train_set.create_valid(np.asarray(vx, dtype=np.float64),
-                                                         label=vy, weight=vw, group=vg))
+                if not (PANDAS_INSTALLED and isinstance(vx, pd_DataFrame)):
+                    vx = np.asarray(vx, dtype=np.float64)
+                valid_sets.append(train_set.create_valid(vx, label=vy, weight=vw,
+                                                         group=vg))
                 names.append(eval_names[i] if eval_names else f"valid_{i}")
         cbs = list(callbacks) if callbacks else []
         if early_stopping_rounds:
@@ -225,7 +231,10 @@ class LGBMModel(BaseEstimator):
 
     def predict(self, X, raw_score=False, start_iteration=0, num_iteration=None,
                 pred_leaf=False, pred_contrib=False, **kwargs):
-        return self.booster_.predict(np.asarray(X, dtype=np.float64), raw_score=raw_score,
+        from .compat import PANDAS_INSTALLED, pd_DataFrame
+        if not (PANDAS_INSTALLED and isinstance(X, pd_DataFrame)):
+            X = np.asarray(X, dtype=np.float64)
+        return self.booster_.predict(X, raw_score=raw_score,
                                      start_iteration=start_iteration,
                                      num_iteration=num_iteration, pred_leaf=pred_leaf,
                                      pred_contrib=pred_contrib)

@@ -121,3 +121,17 @@ def test_callable_objective_and_metric():
     m2.fit(X, y, eval_set=[(X, y)], eval_metric=acc_metric)
     assert "my_acc" in m2.evals_result_["valid_0"]
     assert m2.evals_result_["valid_0"]["my_acc"][-1] > 0.9
+
+
+def test_sklearn_pandas_categorical():
+    """sklearn estimators accept DataFrames with category dtype end to end."""
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(0)
+    n = 2000
+    df = pd.DataFrame({"a": rng.randn(n),
+                       "b": pd.Categorical(rng.choice(["x", "y", "z"], n))})
+    y = ((df["b"] == "x").values & (df["a"] > 0)).astype(int)
+    m = lgb.LGBMClassifier(n_estimators=10, verbosity=-1)
+    m.fit(df, y, eval_set=[(df, y)])
+    assert (m.predict(df) == y).mean() > 0.98
+    assert m.predict_proba(df).shape == (n, 2)
