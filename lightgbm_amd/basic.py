@@ -855,13 +855,15 @@ class Booster:
                 out.append((name, names[i], res[i], higher_better[i]))
         if feval is not None:
             ds = self._train_set if data_idx == 0 else self._valid_sets[data_idx - 1]
-            fe = feval(self.__inner_predict(data_idx), ds)
-            if isinstance(fe, list):
-                for (fn, fv, fb) in fe:
+            preds = self.__inner_predict(data_idx)
+            for fe_fn in (feval if isinstance(feval, (list, tuple)) else [feval]):
+                fe = fe_fn(preds, ds)
+                if isinstance(fe, list):
+                    for (fn, fv, fb) in fe:
+                        out.append((name, fn, fv, fb))
+                else:
+                    fn, fv, fb = fe
                     out.append((name, fn, fv, fb))
-            else:
-                fn, fv, fb = fe
-                out.append((name, fn, fv, fb))
         return out
 
     def __inner_predict(self, data_idx):
