@@ -48,6 +48,19 @@ int ParseLabelIdx(const Config& cfg) {
   return atoi(cfg.label_column.c_str());
 }
 
+/*! resolve a column spec ("3" or "name:foo" against a header line) to an index;
+ *  -1 = unset */
+int ResolveColumn(const std::string& spec, const std::vector<std::string>& header) {
+  if (spec.empty()) return -1;
+  if (Common::StartsWith(spec, "name:")) {
+    const std::string want = spec.substr(5);
+    for (size_t i = 0; i < header.size(); ++i)
+      if (header[i] == want) return static_cast<int>(i);
+    Log::Fatal("Column '%s' not found in the header", want.c_str());
+  }
+  return atoi(spec.c_str());
+}
+
 }  // namespace
 
 void DatasetLoader::ParseFile(const char* filename, std::vector<std::vector<double>>* rows,
@@ -61,7 +74,27 @@ void DatasetLoader::ParseFile(const char* filename, std::vector<std::vector<doub
   if (cfg_.header) first = 1;
   FileFormat fmt = DetectFormat(lines[first]);
   const char* delims = fmt == FileFormat::kCSV ? "," : (fmt == FileFormat::kTSV ? "\t" : " ");
-  const int label_idx = ParseLabelIdx(cfg_);
+  // column specs (weight_column / group_column / ignore_column; "name:" needs header)
+  std::vector<std::string> header_toks;
+  if (cfg_.header) header_toks = Common::SplitAny(lines[0].c_str(), delims);
+  int label_idx = ParseLabelIdx(cfg_);
+  if (Common::StartsWith(cfg_.label_column, "name:"))
+    label_idx = ResolveColumn(cfg_.label_column, header_toks);
+  const int weight_idx = ResolveColumn(cfg_.weight_column, header_toks);
+  const int group_idx = ResolveColumn(cfg_.group_column, header_toks);
+  std::vector<int8_t> ignored;
+  if (!cfg_.ignore_column.empty()) {
+    for (auto& tok : Common::Split(cfg_.ignore_column.c_str(), ',')) {
+      auto t = Common::Trim(tok);
+      if (t.empty()) continue;
+      const int idx = ResolveColumn(t, header_toks);
+      if (idx >= 0) {
+        if (idx >= static_cast<int>(ignored.size())) ignored.resize(idx + 1, 0);
+        ignored[idx] = 1;
+      }
+    }
+  }
+  std::vector<int32_t> row_group_ids;
   int ncol = -1;
   const size_t n_lines = lines.size();
   for (size_t li = first; li < n_lines; ++li) {
@@ -84,8 +117,15 @@ void DatasetLoader::ParseFile(const char* filename, std::vector<std::vector<doub
     } else {
       row.reserve(toks.size() - 1);
       for (size_t t = 0; t < toks.size(); ++t) {
-        if (static_cast<int>(t) == label_idx) {
+        const int ti = static_cast<int>(t);
+        if (ti == label_idx) {
           label = static_cast<float>(atof(toks[t].c_str()));
+        } else if (ti == weight_idx) {
+          weights->push_back(static_cast<float>(atof(toks[t].c_str())));
+        } else if (ti == group_idx) {
+          row_group_ids.push_back(atoi(toks[t].c_str()));
+        } else if (ti < static_cast<int>(ignored.size()) && ignored[ti]) {
+          // dropped
         } else {
           const std::string& s = toks[t];
           if (s.empty() || s == "na" || s == "NA" || s == "nan" || s == "NaN" || s == "?")
@@ -98,6 +138,20 @@ void DatasetLoader::ParseFile(const char* filename, std::vector<std::vector<doub
     ncol = std::max<int>(ncol, static_cast<int>(row.size()));
     rows->push_back(std::move(row));
     labels->push_back(label);
+  }
+  // group column: run-length encode consecutive equal ids into group sizes
+  if (!row_group_ids.empty() && groups->empty()) {
+    int32_t cur = row_group_ids[0], cnt = 0;
+    for (int32_t g : row_group_ids) {
+      if (g == cur) {
+        ++cnt;
+      } else {
+        groups->push_back(cnt);
+        cur = g;
+        cnt = 1;
+      }
+    }
+    groups->push_back(cnt);
   }
   // pad jagged libsvm rows
   for (auto& r : *rows) r.resize(ncol, 0.0);

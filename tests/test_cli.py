@@ -199,3 +199,29 @@ def test_cli_binary_dataset_roundtrip(tmp_path):
         t = p.read_text()
         return t[t.index("Tree=0"):t.index("end of trees")]
     assert trees(m1) == trees(m2)
+
+
+def test_cli_column_specs(tmp_path):
+    """weight_column / group_column / ignore_column with header + name: resolution."""
+    import subprocess
+    rng = np.random.RandomState(0)
+    n = 1200
+    lines = ["label\tw\tqid\tf0\tf1\tjunk"]
+    qid = np.repeat(np.arange(60), 20)
+    for i in range(n):
+        f0, f1 = rng.rand(), rng.rand()
+        y = int(f0 > 0.5)
+        lines.append(f"{y}\t{1.0 + (i % 3)}\t{qid[i]}\t{f0:.6f}\t{f1:.6f}\t999")
+    data = tmp_path / "t.tsv"
+    data.write_text("\n".join(lines) + "\n")
+    out = tmp_path / "m.txt"
+    r = subprocess.run([str(CLI), "task=train", "objective=lambdarank", f"data={data}",
+                        "header=true", "label_column=name:label",
+                        "weight_column=name:w", "group_column=name:qid",
+                        "ignore_column=name:junk", "num_trees=5",
+                        f"output_model={out}"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    txt = out.read_text()
+    # only f0/f1 remain as features
+    assert "max_feature_idx=1" in txt
