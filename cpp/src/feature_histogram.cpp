@@ -118,7 +118,7 @@ void FindBestThresholdCategorical(const hist_t* hist, int num_bin, const LeafCon
     data_size_t lc = static_cast<data_size_t>(Common::RoundInt(shl * cnt_factor));
     data_size_t rc = leaf.num_data - lc;
     if (shl < min_hess || lc < min_cnt) return;
-    if (shr < min_hess || rc < min_cnt) return;
+    if (shr < min_hess || rc < min_cnt || rc < cfg.min_data_per_group) return;
     double gain = GainMath::GetSplitGains(sgl, shl, sgr, shr, l1, l2, cfg.max_delta_step, 0.0,
                                           lc, rc, leaf.parent_output);
     if (gain <= min_gain_shift) return;
@@ -150,12 +150,20 @@ void FindBestThresholdCategorical(const hist_t* hist, int num_bin, const LeafCon
     const int limit = std::min(cfg.max_cat_threshold, n - 1);
     for (int dir = 0; dir < 2; ++dir) {
       double sgl = 0, shl = 0;
+      data_size_t cnt_cur_group = 0;
       std::vector<int> taken;
       for (int k = 0; k < limit; ++k) {
         int b = dir == 0 ? order[k] : order[n - 1 - k];
         sgl += hist[2 * b];
         shl += hist[2 * b + 1];
+        cnt_cur_group += static_cast<data_size_t>(
+            Common::RoundInt(hist[2 * b + 1] * cnt_factor));
         taken.push_back(b);
+        // group gating (parity: reference min_data_per_group): only evaluate a
+        // boundary once the accumulated group since the last evaluation is big
+        // enough — prevents tiny-category overfitting
+        if (cnt_cur_group < cfg.min_data_per_group) continue;
+        cnt_cur_group = 0;
         eval_subset(taken, sgl, shl);
       }
     }
