@@ -244,3 +244,63 @@ def test_kitchen_sink_model_cross(tmp_path):
     bst.save_model(str(f))
     np.testing.assert_allclose(bst.predict(X[:500]), _ref_predict(f, X[:500], tmp_path),
                                rtol=1e-9, atol=1e-12)
+
+
+def test_ranking_quality_parity_with_reference(tmp_path):
+    """Same ranking data and config: our lambdarank reaches the reference's NDCG
+    within tolerance, and the model files interchange."""
+    rng = np.random.RandomState(3)
+    rows, labels, groups = [], [], []
+    for q in range(300):
+        nq = 20
+        Xq = rng.rand(nq, 10)
+        rel = np.clip((3 * Xq[:, 0] + Xq[:, 1] + 0.3 * rng.randn(nq)).astype(int), 0, 4)
+        rows.append(Xq)
+        labels.append(rel)
+        groups.append(nq)
+    X = np.vstack(rows)
+    y = np.concatenate(labels).astype(np.float32)
+    g = np.array(groups, dtype=np.int32)
+    bst = lgb.train({"objective": "lambdarank", "verbosity": -1, "num_leaves": 31},
+                    lgb.Dataset(X, label=y, group=g), 30)
+    f = tmp_path / "rank.txt"
+    bst.save_model(str(f))
+    # model interchange
+    np.testing.assert_allclose(bst.predict(X[:500]), _ref_predict(f, X[:500], tmp_path),
+                               rtol=1e-9, atol=1e-12)
+    # reference trains the same config; compare NDCG@5 of both models on the data
+    np.save(tmp_path / "x.npy", X)
+    np.save(tmp_path / "y.npy", y)
+    worker = tmp_path / "rank_worker.py"
+    worker.write_text(_REF_WORKER.replace(
+        '''ok(lib.LGBM_DatasetSetField(ds, b"label", lab.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_int(len(lab)), 0))''',
+        '''ok(lib.LGBM_DatasetSetField(ds, b"label", lab.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_int(len(lab)), 0))
+    grp = np.full(len(lab) // 20, 20, dtype=np.int32)
+    ok(lib.LGBM_DatasetSetField(ds, b"group", grp.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_int(len(grp)), 2))'''))
+    import subprocess as sp
+    fr = tmp_path / "ref_rank.txt"
+    r = sp.run([sys.executable, str(worker), json.dumps(
+        {"op": "train", "x": str(tmp_path / "x.npy"), "y": str(tmp_path / "y.npy"),
+         "model": str(fr), "ds_params": "max_bin=255",
+         "params": "objective=lambdarank verbosity=-1 num_leaves=31", "iters": 30}),
+        str(REF_LIB)], capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    ref_scores = _ref_predict(fr, X, tmp_path)
+    our_scores = bst.predict(X)
+
+    def ndcg5(scores):
+        total, pos = 0.0, 0
+        for nq in groups:
+            s, rel = scores[pos:pos + nq], y[pos:pos + nq]
+            pos += nq
+            order = np.argsort(-s, kind="stable")
+            dcg = sum((2 ** rel[order[i]] - 1) / np.log2(i + 2) for i in range(5))
+            ideal = np.sort(rel)[::-1]
+            idcg = sum((2 ** ideal[i] - 1) / np.log2(i + 2) for i in range(5))
+            total += dcg / idcg if idcg > 0 else 1.0
+        return total / len(groups)
+    ours, theirs = ndcg5(our_scores), ndcg5(ref_scores)
+    assert ours > theirs - 0.01, (ours, theirs)
