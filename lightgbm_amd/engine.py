@@ -3,6 +3,7 @@ import collections
 import copy
 
 import numpy as np
+from pathlib import Path
 
 from . import callback as callback_mod
 from .basic import Booster, Dataset
@@ -121,11 +122,41 @@ class CVBooster:
     def __init__(self, model_file=None):
         self.boosters = []
         self.best_iteration = -1
+        if model_file is not None:
+            text = Path(model_file).read_text()
+            for chunk in text.split("!!! cv booster fold separator !!!"):
+                chunk = chunk.strip()
+                if chunk:
+                    self.boosters.append(Booster(model_str=chunk))
 
     def _append(self, booster):
         self.boosters.append(booster)
 
+    def save_model(self, filename, num_iteration=None, start_iteration=0,
+                   importance_type="split"):
+        parts = [b.model_to_string(num_iteration=num_iteration,
+                                   start_iteration=start_iteration,
+                                   importance_type=importance_type)
+                 for b in self.boosters]
+        Path(filename).write_text("\n!!! cv booster fold separator !!!\n".join(parts))
+        return self
+
+    def model_to_string(self, **kwargs):
+        return "\n!!! cv booster fold separator !!!\n".join(
+            b.model_to_string(**kwargs) for b in self.boosters)
+
+    def __getstate__(self):
+        return {"best_iteration": self.best_iteration,
+                "_models": [b.model_to_string() for b in self.boosters]}
+
+    def __setstate__(self, state):
+        self.best_iteration = state.get("best_iteration", -1)
+        self.boosters = [Booster(model_str=m) for m in state.get("_models", [])]
+
     def __getattr__(self, name):
+        if name.startswith("__") and name.endswith("__"):
+            raise AttributeError(name)
+
         def handler_function(*args, **kwargs):
             return [getattr(b, name)(*args, **kwargs) for b in self.boosters]
         return handler_function

@@ -738,3 +738,20 @@ def test_average_precision_metric():
               callbacks=[lgb.record_evaluation(ev)])
     ap = ev["valid_0"]["average_precision"]
     assert ap[-1] > 0.9 and ap[-1] >= ap[0] - 1e-9
+
+
+def test_cvbooster_persistence(tmp_path):
+    """CVBooster pickles and saves/loads as a multi-fold model file."""
+    import pickle
+    X, y = _binary_data(n=1000)
+    r = lgb.cv({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=y),
+               5, nfold=3, return_cvbooster=True)
+    cvb = r["cvbooster"]
+    cvb2 = pickle.loads(pickle.dumps(cvb))
+    assert len(cvb2.boosters) == 3
+    f = tmp_path / "cv.txt"
+    cvb.save_model(str(f))
+    cvb3 = lgb.CVBooster(model_file=str(f))
+    a = np.mean([b.predict(X[:50]) for b in cvb.boosters], axis=0)
+    b = np.mean([m.predict(X[:50]) for m in cvb3.boosters], axis=0)
+    np.testing.assert_allclose(a, b, rtol=1e-12)
