@@ -1,6 +1,7 @@
 """scikit-learn estimator wrappers (parity target: reference python-package/lightgbm/sklearn.py)."""
 
 import numpy as np
+from pathlib import Path
 
 from . import callback as callback_mod
 from .basic import Booster, Dataset, LightGBMError
@@ -117,7 +118,8 @@ class LGBMModel(BaseEstimator):
     def _fit(self, X, y, default_objective, sample_weight=None, init_score=None,
              group=None, eval_set=None, eval_names=None, eval_sample_weight=None,
              eval_group=None, eval_metric=None, early_stopping_rounds=None,
-             callbacks=None, categorical_feature="auto", feature_name="auto"):
+             callbacks=None, categorical_feature="auto", feature_name="auto",
+             init_model=None):
         params = self._make_params(default_objective)
         # sklearn-style CALLABLE objective: (y_true, y_pred[, weight[, group]]) ->
         # (grad, hess); adapted to the engine's (preds, dataset) form
@@ -183,10 +185,12 @@ class LGBMModel(BaseEstimator):
             cbs.append(callback_mod.early_stopping(early_stopping_rounds, verbose=False))
         self._evals_result = {}
         cbs.append(callback_mod.record_evaluation(self._evals_result))
+        if init_model is not None and not isinstance(init_model, (str, Path, Booster)):
+            init_model = init_model.booster_   # an estimator
         self._Booster = train_fn(params, train_set, num_boost_round=self.n_estimators,
                                  valid_sets=valid_sets or None,
                                  valid_names=names or None, callbacks=cbs,
-                                 fobj=fobj, feval=feval)
+                                 fobj=fobj, feval=feval, init_model=init_model)
         self._best_iteration = self._Booster.best_iteration
         return self
 

@@ -135,3 +135,12 @@ def test_sklearn_pandas_categorical():
     m.fit(df, y, eval_set=[(df, y)])
     assert (m.predict(df) == y).mean() > 0.98
     assert m.predict_proba(df).shape == (n, 2)
+
+
+def test_fit_init_model_continuation():
+    rng = np.random.RandomState(0)
+    X = rng.randn(1000, 4)
+    y = (X[:, 0] > 0).astype(int)
+    m1 = lgb.LGBMClassifier(n_estimators=5, verbosity=-1).fit(X, y)
+    m2 = lgb.LGBMClassifier(n_estimators=5, verbosity=-1).fit(X, y, init_model=m1)
+    assert m2.booster_.num_trees() == 10
