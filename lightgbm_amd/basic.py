@@ -882,6 +882,13 @@ class Booster:
                 pred_leaf=False, pred_contrib=False, validate_features=False, **kwargs):
         if num_iteration is None:
             num_iteration = self.best_iteration if self.best_iteration > 0 else -1
+        if validate_features and PANDAS_INSTALLED and isinstance(data, pd_DataFrame):
+            model_names = self.feature_name()
+            data_names = [str(c) for c in data.columns]
+            if data_names != model_names:
+                raise LightGBMError(
+                    f"Feature names mismatch: model expects {model_names}, "
+                    f"data has {data_names}")
         pred_param = _param_dict_to_str(
             {k: v for k, v in kwargs.items()
              if k in ("pred_early_stop", "pred_early_stop_freq",
