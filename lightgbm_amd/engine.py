@@ -29,11 +29,16 @@ def train(params, train_set, num_boost_round=100, valid_sets=None, valid_names=N
 
     cbs = list(callbacks) if callbacks else []
     # early_stopping_round in params spawns the callback
+    es_min_delta = 0.0
+    for d_alias in ("early_stopping_min_delta",):
+        if d_alias in params:
+            es_min_delta = float(params.pop(d_alias))
     for alias in ("early_stopping_round", "early_stopping_rounds", "early_stopping",
                   "n_iter_no_change"):
         if alias in params and params[alias]:
             cbs.append(callback_mod.early_stopping(int(params[alias]),
-                                                   first_metric_only=first_metric_only))
+                                                   first_metric_only=first_metric_only,
+                                                   min_delta=es_min_delta))
             params.pop(alias)
             break
     if params.get("verbosity", params.get("verbose", 1)) >= 1 and not any(
