@@ -450,6 +450,22 @@ bool SerialTreeLearner::MakeForcedSplit(int leaf, const LeafContext& ctx,
   out->right_output = GainMath::CalculateSplittedLeafOutput(
       out->right_sum_gradient, out->right_sum_hessian, config_->lambda_l1,
       config_->lambda_l2, config_->max_delta_step);
+  // monotone constraints apply to forced splits too: clamp into the leaf's
+  // inherited bounds and tag the direction so the Train loop propagates bounds
+  out->left_output = std::min(std::max(out->left_output, ctx.out_lo), ctx.out_hi);
+  out->right_output = std::min(std::max(out->right_output, ctx.out_lo), ctx.out_hi);
+  if (!config_->monotone_constraints.empty()) {
+    const int orig = train_data_->RealFeatureIndex(inner);
+    if (orig < static_cast<int>(config_->monotone_constraints.size()))
+      out->monotone_type = static_cast<int8_t>(config_->monotone_constraints[orig]);
+    if (out->monotone_type > 0 && out->left_output > out->right_output) {
+      out->left_output = out->right_output =
+          (out->left_output + out->right_output) / 2.0;
+    } else if (out->monotone_type < 0 && out->left_output < out->right_output) {
+      out->left_output = out->right_output =
+          (out->left_output + out->right_output) / 2.0;
+    }
+  }
   out->gain = 1e30;  // forced splits take precedence over gain selection
   return true;
 }

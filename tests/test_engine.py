@@ -755,3 +755,24 @@ def test_cvbooster_persistence(tmp_path):
     a = np.mean([b.predict(X[:50]) for b in cvb.boosters], axis=0)
     b = np.mean([m.predict(X[:50]) for m in cvb3.boosters], axis=0)
     np.testing.assert_allclose(a, b, rtol=1e-12)
+
+
+def test_forced_splits_respect_monotone(tmp_path):
+    """Forced splits clamp into monotone bounds and propagate them (interaction
+    of two constraint systems)."""
+    import json
+    rng = np.random.RandomState(0)
+    X = rng.rand(4000, 4)
+    y = (2 * X[:, 0] + X[:, 1] > 1.2).astype(np.float32)
+    fs = tmp_path / "forced.json"
+    fs.write_text(json.dumps({"feature": 0, "threshold": 0.5,
+                              "left": {"feature": 1, "threshold": 0.5}}))
+    bst = lgb.train({"objective": "binary", "verbosity": -1,
+                     "forcedsplits_filename": str(fs),
+                     "monotone_constraints": [1, 0, 0, 0],
+                     "bagging_freq": 1, "bagging_fraction": 0.7},
+                    lgb.Dataset(X, label=y), 20)
+    xs = np.linspace(0.05, 0.95, 20)
+    for other in (0.2, 0.5, 0.8):
+        grid = np.column_stack([xs] + [np.full(20, other)] * 3)
+        assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
