@@ -315,6 +315,16 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
   if (boosting == "gbrt") boosting = "gbdt";
   if (boosting == "random_forest") boosting = "rf";
 
+  if (linear_tree) {
+    // reference restrictions: per-leaf linear fits need the serial learner's raw
+    // values and exact partitions
+    if (tree_learner != "serial") {
+      tree_learner = "serial";
+      Log::Warning("Linear tree learner must be serial; forcing tree_learner=serial");
+    }
+    if (zero_as_missing)
+      Log::Fatal("zero_as_missing must be false when fitting linear trees");
+  }
   if (num_threads > 0) omp_set_num_threads(num_threads);
   // only an EXPLICIT verbosity touches the global log level: lazily-constructed
   // datasets with default params must not undo a booster's verbosity=-1
