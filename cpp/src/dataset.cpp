@@ -226,6 +226,18 @@ std::unique_ptr<Dataset> Dataset::CreateValid(
       }
     }
   }
+  if (has_raw()) {
+    // linear-tree scoring reads raw feature values on valid data too
+    const int nf_ = num_features();
+    d->raw_values_.resize(nf_);
+#pragma omp parallel for schedule(dynamic, 1)
+    for (int f = 0; f < nf_; ++f) {
+      d->raw_values_[f].resize(nrow);
+      const int c = real_feature_index_[f];
+      for (data_size_t i = 0; i < nrow; ++i)
+        d->raw_values_[f][i] = static_cast<float>(get(i, c));
+    }
+  }
   d->metadata_.Init(nrow, false, false);
   return d;
 }
@@ -756,6 +768,13 @@ std::unique_ptr<Dataset> Dataset::Subset(const data_size_t* indices, data_size_t
     for (data_size_t i = 0; i < n; ++i) d->columns_[col].Set(i, columns_[col].Get(indices[i]));
   }
   d->has_sparse_ = false;  // subset columns are materialized dense
+  if (has_raw()) {
+    d->raw_values_.resize(raw_values_.size());
+    for (size_t f = 0; f < raw_values_.size(); ++f) {
+      d->raw_values_[f].resize(n);
+      for (data_size_t i = 0; i < n; ++i) d->raw_values_[f][i] = raw_values_[f][indices[i]];
+    }
+  }
   // metadata subset
   d->metadata_.Init(n, false, false);
   std::vector<float> lab(n);
@@ -1120,7 +1139,7 @@ void Tree::AddPredictionToScore(const Dataset* data, data_size_t num_data, doubl
         }
       }
     }
-    if (is_linear_) {
+    if (is_linear_ && data->has_raw()) {
       const int leaf = ~node;
       double out = leaf_const_.empty() || leaf_coeff_[leaf].empty()
                        ? leaf_value_[leaf] : leaf_const_[leaf];
@@ -1173,7 +1192,7 @@ void Tree::AddPredictionToScore(const Dataset* data, const data_size_t* used_ind
         }
       }
     }
-    if (is_linear_) {
+    if (is_linear_ && data->has_raw()) {
       const int leaf = ~node;
       double out = leaf_const_.empty() || leaf_coeff_[leaf].empty()
                        ? leaf_value_[leaf] : leaf_const_[leaf];

@@ -776,3 +776,21 @@ def test_forced_splits_respect_monotone(tmp_path):
     for other in (0.2, 0.5, 0.8):
         grid = np.column_stack([xs] + [np.full(20, other)] * 3)
         assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
+
+
+def test_linear_tree_with_valid_sets():
+    """Linear trees score valid sets through raw feature values (regression: this
+    used to segfault — valid datasets carried no raw values)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(4000, 6)
+    y = (2 * X[:, 0] + np.sin(X[:, 1])).astype(np.float32)
+    tr = lgb.Dataset(X, label=y)
+    ev = {}
+    bst = lgb.train({"objective": "regression", "linear_tree": True, "metric": "l2",
+                     "verbosity": -1}, tr, 20,
+                    valid_sets=[tr.create_valid(X[:800], label=y[:800])],
+                    callbacks=[lgb.record_evaluation(ev)])
+    assert ev["valid_0"]["l2"][-1] < 0.5 * float(np.var(y))
+    # valid eval equals direct prediction MSE (raw values wired correctly)
+    mse = float(np.mean((bst.predict(X[:800]) - y[:800]) ** 2))
+    assert abs(mse - ev["valid_0"]["l2"][-1]) < 1e-6
