@@ -225,3 +225,21 @@ def test_cli_column_specs(tmp_path):
     txt = out.read_text()
     # only f0/f1 remain as features
     assert "max_feature_idx=1" in txt
+
+
+@pytest.mark.parametrize("example", ["binary_classification", "regression",
+                                     "multiclass_classification", "lambdarank"])
+def test_bundled_examples_run(example, tmp_path):
+    """Every bundled CLI example trains and (where configured) predicts."""
+    import subprocess, shutil
+    src = REPO / "examples" / example
+    work = tmp_path / example
+    shutil.copytree(src, work)
+    r = subprocess.run([str(CLI), "config=train.conf", "num_trees=10"], cwd=work,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert list(work.glob("*model.txt")), "no model file written"
+    if (work / "predict.conf").exists():
+        r2 = subprocess.run([str(CLI), "config=predict.conf"], cwd=work,
+                            capture_output=True, text=True, timeout=300)
+        assert r2.returncode == 0, r2.stdout + r2.stderr
