@@ -274,10 +274,21 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
     return;
   }
   if (hist_mode_ == 1 || (hist_mode_ < 0 && hist_trials_done_ == 1)) {
-    // locked row-wise, or mid-trial small leaves while row was measured last
-    train_data_->ConstructHistogramsRowWise(indices, cnt, gradients_, hessians_, hist,
-                                            /*row_indexed=*/true);
-    return;
+    // locked row-wise — but SMALL leaves pay row-wise's fixed per-call cost
+    // (per-thread private-histogram zero + merge) without amortizing it, so they
+    // drop to the col-wise path (hybrid, like the reference's per-leaf choice)
+    static const int kRowWiseMinRows = [] {
+      const char* e = getenv("MIGBM_ROWWISE_MIN_ROWS");
+      return e ? atoi(e) : -1;
+    }();
+    const data_size_t min_rows = kRowWiseMinRows >= 0
+                                     ? kRowWiseMinRows
+                                     : train_data_->num_total_bin();
+    if (cnt >= min_rows) {
+      train_data_->ConstructHistogramsRowWise(indices, cnt, gradients_, hessians_, hist,
+                                              /*row_indexed=*/true);
+      return;
+    }
   }
   if ((train_data_->has_bundles() || train_data_->has_sparse()) &&
       Network::is_distributed()) {
