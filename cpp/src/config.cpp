@@ -315,6 +315,12 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
   if (boosting == "gbrt") boosting = "gbdt";
   if (boosting == "random_forest") boosting = "rf";
 
+  if (deterministic) {
+    // thread-count-invariant training: col-wise histograms accumulate each
+    // feature sequentially, and the root reduction uses fixed-block summation
+    force_col_wise = true;
+    force_row_wise = false;
+  }
   if (linear_tree) {
     // reference restrictions: per-leaf linear fits need the serial learner's raw
     // values and exact partitions

@@ -564,10 +564,32 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
   data_size_t root_cnt;
   const data_size_t* root_idx = partition_.GetIndexOnLeaf(0, &root_cnt);
   double sum_g = 0.0, sum_h = 0.0;
+  if (config_->deterministic) {
+    // fixed 1024-row blocks summed in order: invariant to the thread count
+    const int nblk = static_cast<int>((root_cnt + 1023) / 1024);
+    std::vector<double> bg(nblk, 0.0), bh(nblk, 0.0);
+#pragma omp parallel for schedule(static)
+    for (int b = 0; b < nblk; ++b) {
+      const data_size_t s0 = static_cast<data_size_t>(b) * 1024;
+      const data_size_t e0 = std::min<data_size_t>(root_cnt, s0 + 1024);
+      double g = 0.0, h = 0.0;
+      for (data_size_t i = s0; i < e0; ++i) {
+        g += gradients_[root_idx[i]];
+        h += hessians_[root_idx[i]];
+      }
+      bg[b] = g;
+      bh[b] = h;
+    }
+    for (int b = 0; b < nblk; ++b) {
+      sum_g += bg[b];
+      sum_h += bh[b];
+    }
+  } else {
 #pragma omp parallel for schedule(static) reduction(+ : sum_g, sum_h)
-  for (data_size_t i = 0; i < root_cnt; ++i) {
-    sum_g += gradients_[root_idx[i]];
-    sum_h += hessians_[root_idx[i]];
+    for (data_size_t i = 0; i < root_cnt; ++i) {
+      sum_g += gradients_[root_idx[i]];
+      sum_h += hessians_[root_idx[i]];
+    }
   }
   data_size_t global_root_cnt = root_cnt;
   ReduceRootStats(&sum_g, &sum_h, &global_root_cnt);

@@ -794,3 +794,16 @@ def test_linear_tree_with_valid_sets():
     # valid eval equals direct prediction MSE (raw values wired correctly)
     mse = float(np.mean((bst.predict(X[:800]) - y[:800]) ** 2))
     assert abs(mse - ev["valid_0"]["l2"][-1]) < 1e-6
+
+
+def test_deterministic_thread_invariance():
+    """deterministic=true: identical trees regardless of num_threads."""
+    X, y = _binary_data(n=3000)
+
+    def trees(m):
+        return m[m.index("Tree=0"):m.index("end of trees")]
+    ms = [trees(lgb.train({"objective": "binary", "verbosity": -1,
+                           "deterministic": True, "num_threads": t},
+                          lgb.Dataset(X, label=y), 10).model_to_string())
+          for t in (1, 4, 8)]
+    assert ms[0] == ms[1] == ms[2]
