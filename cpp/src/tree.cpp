@@ -360,9 +360,27 @@ std::string Tree::ToIfElse(int index) const {
       int leaf = ~node;
       if (leaf_index) {
         ss << ind << "return " << leaf << ";\n";
-      } else {
-        ss << ind << "return " << leaf_value_[leaf] << ";\n";
+        return;
       }
+      if (is_linear_ && leaf < static_cast<int>(leaf_coeff_.size()) &&
+          !leaf_coeff_[leaf].empty()) {
+        // linear leaf: const + Σ coeff·x, falling back to the piecewise-constant
+        // value when any used feature is missing (LeafOutputLinear semantics)
+        ss << ind << "{\n";
+        std::string cond;
+        for (size_t i = 0; i < leaf_coeff_[leaf].size(); ++i) {
+          if (i) cond += " || ";
+          cond += "std::isnan(arr[" + std::to_string(leaf_features_[leaf][i]) + "])";
+        }
+        ss << ind << "  if (" << cond << ") return " << leaf_value_[leaf] << ";\n";
+        ss << ind << "  return " << leaf_const_[leaf];
+        for (size_t i = 0; i < leaf_coeff_[leaf].size(); ++i)
+          ss << " + " << leaf_coeff_[leaf][i] << " * arr["
+             << leaf_features_[leaf][i] << "]";
+        ss << ";\n" << ind << "}\n";
+        return;
+      }
+      ss << ind << "return " << leaf_value_[leaf] << ";\n";
       return;
     }
     const int fid = split_feature_[node];

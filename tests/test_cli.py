@@ -243,3 +243,34 @@ def test_bundled_examples_run(example, tmp_path):
         r2 = subprocess.run([str(CLI), "config=predict.conf"], cwd=work,
                             capture_output=True, text=True, timeout=300)
         assert r2.returncode == 0, r2.stdout + r2.stderr
+
+
+def test_convert_model_cpp_linear_trees(tmp_path):
+    """C++ codegen emits linear-leaf expressions (const + coeffs), matching the
+    Booster bit-for-bit."""
+    import subprocess
+    rng = np.random.RandomState(0)
+    X = rng.rand(2000, 3)
+    y = (3 * X[:, 0] + np.sin(4 * X[:, 1])).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "linear_tree": True, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 5)
+    m = tmp_path / "lin.txt"
+    bst.save_model(str(m))
+    gen = tmp_path / "gen.cpp"
+    subprocess.run([str(CLI), "task=convert_model", f"input_model={m}",
+                    "convert_model_language=cpp", f"convert_model={gen}"],
+                   check=True, capture_output=True, timeout=300)
+    main = tmp_path / "main.cpp"
+    main.write_text(gen.read_text() + r"""
+#include <cstdio>
+int main(){double r[3],o[1];
+ while(std::scanf("%lf %lf %lf",r,r+1,r+2)==3){Predict(r,o);std::printf("%.17g\n",o[0]);}
+ return 0;}
+""")
+    exe = tmp_path / "pred"
+    subprocess.run(["g++", "-O1", "-o", str(exe), str(main)], check=True, timeout=300)
+    rows = "\n".join(" ".join("%.17g" % v for v in r) for r in X[:100])
+    res = subprocess.run([str(exe)], input=rows, capture_output=True, text=True,
+                         check=True, timeout=300)
+    gen_pred = np.array([float(t) for t in res.stdout.split()])
+    np.testing.assert_allclose(gen_pred, bst.predict(X[:100]), rtol=1e-10)
