@@ -144,3 +144,20 @@ def test_fit_init_model_continuation():
     m1 = lgb.LGBMClassifier(n_estimators=5, verbosity=-1).fit(X, y)
     m2 = lgb.LGBMClassifier(n_estimators=5, verbosity=-1).fit(X, y, init_model=m1)
     assert m2.booster_.num_trees() == 10
+
+
+def test_sklearn_ecosystem_protocols():
+    """clone / cross_val_score / GridSearchCV interop."""
+    from sklearn.base import clone
+    from sklearn.model_selection import GridSearchCV, cross_val_score
+    rng = np.random.RandomState(0)
+    X = rng.randn(800, 4)
+    y = (X[:, 0] > 0).astype(int)
+    m = clone(lgb.LGBMClassifier(n_estimators=5, verbosity=-1))
+    m.fit(X, y)
+    scores = cross_val_score(lgb.LGBMClassifier(n_estimators=5, verbosity=-1), X, y, cv=3)
+    assert scores.mean() > 0.9
+    gs = GridSearchCV(lgb.LGBMClassifier(verbosity=-1),
+                      {"n_estimators": [3, 5], "num_leaves": [7, 15]}, cv=2)
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.9
