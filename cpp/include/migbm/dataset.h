@@ -181,6 +181,11 @@ class Dataset {
   std::unique_ptr<Dataset> CreateValid(const std::function<double(data_size_t, int)>& get,
                                        data_size_t nrow) const;
 
+  /*! True when `other`'s per-feature bin mappers match this dataset's — required
+   *  before scoring trees trained on `other`'s bins against this data.
+   *  Reference parity: DatasetLoader::CheckAlignWithOtherDataset semantics. */
+  bool AlignsWith(const Dataset* other) const;
+
   /*! Column-wise histogram build over an ordered index subset.
    *  hist layout: per used feature f at hist_offset(f)*2, (sum_grad, sum_hess) pairs.
    *  ordered_grad/hess must be pre-gathered to match data_indices order. */

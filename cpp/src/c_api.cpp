@@ -118,6 +118,11 @@ class BoosterWrapper {
   }
 
   void AddValidData(const Dataset* valid) {
+    if (train_data_ != nullptr && !valid->AlignsWith(train_data_)) {
+      Log::Fatal("Cannot add validation data: its bin mappers differ from the "
+                 "training data's. Construct the valid set with the training "
+                 "Dataset as reference (Dataset(..., reference=train_set)).");
+    }
     std::vector<std::unique_ptr<Metric>> ms;
     auto metric_names = config_.metric;
     if (metric_names.empty() && !config_.objective.empty() && config_.objective != "none")

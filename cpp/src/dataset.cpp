@@ -242,6 +242,22 @@ std::unique_ptr<Dataset> Dataset::CreateValid(
   return d;
 }
 
+bool Dataset::AlignsWith(const Dataset* other) const {
+  if (other == nullptr) return false;
+  if (num_total_features_ != other->num_total_features_) return false;
+  if (bin_mappers_.size() != other->bin_mappers_.size()) return false;
+  for (size_t f = 0; f < bin_mappers_.size(); ++f) {
+    const BinMapper* a = bin_mappers_[f].get();
+    const BinMapper* b = other->bin_mappers_[f].get();
+    if (a->num_bin() != b->num_bin()) return false;
+    if (a->bin_type() != b->bin_type()) return false;
+    if (a->missing_type() != b->missing_type()) return false;
+    if (a->bin_upper_bound() != b->bin_upper_bound()) return false;
+    if (a->bin_2_categorical() != b->bin_2_categorical()) return false;
+  }
+  return true;
+}
+
 void Dataset::PlanBundles(const Config& cfg,
                           const std::function<double(data_size_t, int)>& get,
                           const std::vector<data_size_t>& sample_idx) {

@@ -1387,7 +1387,11 @@ __global__ void __launch_bounds__(256) k_grad_lambdarank(
   if (q >= num_queries) return;
   const int start = qb[q];
   const int cnt = qb[q + 1] - start;
-  if (cnt > 1024) return;  // oversized queries are handled by the host fallback
+  if (cnt > 1024) {
+    // the host pre-check refuses such datasets; reaching here means stale g/h
+    // would silently corrupt training — trap instead of returning
+    __builtin_trap();
+  }
   int pow2 = 1;
   while (pow2 < cnt) pow2 <<= 1;
   for (int i = threadIdx.x; i < pow2; i += blockDim.x) {
@@ -1453,7 +1457,7 @@ __global__ void __launch_bounds__(256) k_grad_lambdarank(
     const double high_disc = 1.0 / log2(2.0 + hi_rank);
     const double low_disc = 1.0 / log2(2.0 + lo_rank);
     double delta = fabs((high_gain - low_gain) * (high_disc - low_disc) * imd);
-    if (best != worst) delta /= (0.01 + fabs(ds));
+    if (norm && best != worst) delta /= (0.01 + fabs(ds));
     double p_lambda = 1.0 / (1.0 + exp(sigmoid * ds));
     double p_hess = p_lambda * (1.0 - p_lambda);
     p_lambda *= -sigmoid * delta;

@@ -681,7 +681,7 @@ class LambdarankNDCG : public ObjectiveFunction {
         double delta_pair_ndcg = (high_gain - low_gain) * (high_disc - low_disc) *
                                  inverse_max_dcg_[q];
         if (delta_pair_ndcg < 0) delta_pair_ndcg = -delta_pair_ndcg;
-        if (best != worst && high_label != low_label)
+        if (norm_ && best != worst && high_label != low_label)
           delta_pair_ndcg /= (0.01 + std::fabs(delta_score));
         // lambda
         double p_lambda = 1.0 / (1.0 + std::exp(sigmoid_ * delta_score));
@@ -766,23 +766,34 @@ class RankXENDCG : public ObjectiveFunction {
       double sum = 0;
       for (data_size_t i = 0; i < cnt; ++i) { rho[i] = std::exp(sc[i] - mx); sum += rho[i]; }
       for (data_size_t i = 0; i < cnt; ++i) rho[i] /= sum;
-      // gumbel-perturbed relevance probabilities (phi)
-      std::vector<double> phi(cnt);
-      double psum = 0;
+      // stochastic ground-truth distribution: phi_i = 2^l_i - u, u ~ U(0,1)
+      // (XE-NDCG-MART; reference rank_objective.hpp:400-456 math)
+      std::vector<double> t(cnt);
       Random& r = rands_[q];
+      double denom = 0;
       for (data_size_t i = 0; i < cnt; ++i) {
-        double g = label_[s + i];
-        double gumbel = -std::log(-std::log(std::max(1e-12f, r.NextFloat())));
-        phi[i] = std::pow(2.0, g) - 1.0 + 0.0 * gumbel;  // deterministic variant
-        psum += phi[i];
+        t[i] = std::pow(2.0, static_cast<int>(label_[s + i])) -
+               static_cast<double>(r.NextFloat());
+        denom += t[i];
       }
-      if (psum <= 0) {
-        for (data_size_t i = 0; i < cnt; ++i) { grad[s + i] = 0; hess[s + i] = 0; }
-        continue;
-      }
-      for (data_size_t i = 0; i < cnt; ++i) phi[i] /= psum;
+      const double inv_denom = 1.0 / std::max(1e-15, denom);
+      // first-order term, then two softmax-weighted correction orders
+      double sum_l1 = 0.0;
       for (data_size_t i = 0; i < cnt; ++i) {
-        grad[s + i] = static_cast<score_t>(rho[i] - phi[i]);
+        const double term = rho[i] - t[i] * inv_denom;
+        grad[s + i] = static_cast<score_t>(term);
+        t[i] = term / (1.0 - rho[i]);
+        sum_l1 += t[i];
+      }
+      double sum_l2 = 0.0;
+      for (data_size_t i = 0; i < cnt; ++i) {
+        const double term = rho[i] * (sum_l1 - t[i]);
+        grad[s + i] += static_cast<score_t>(term);
+        t[i] = term / (1.0 - rho[i]);
+        sum_l2 += t[i];
+      }
+      for (data_size_t i = 0; i < cnt; ++i) {
+        grad[s + i] += static_cast<score_t>(rho[i] * (sum_l2 - t[i]));
         hess[s + i] = static_cast<score_t>(rho[i] * (1.0 - rho[i]));
       }
     }

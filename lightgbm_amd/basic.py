@@ -17,6 +17,13 @@ from .libpath import find_lib_path
 __all__ = ["Dataset", "Booster", "LightGBMError", "register_logger", "Sequence"]
 
 _DTYPE_F32, _DTYPE_F64, _DTYPE_I32, _DTYPE_I64 = 0, 1, 2, 3
+
+# Metrics whose C++ factor_to_bigger_better is -1 (higher value = better model);
+# must stay in sync with cpp/src/metric.cpp. Used by early stopping.
+_HIGHER_BETTER_METRICS = frozenset({
+    "auc", "auc_mu", "average_precision", "ndcg", "map",
+    "mean_average_precision", "r2",
+})
 _PREDICT_NORMAL, _PREDICT_RAW, _PREDICT_LEAF, _PREDICT_CONTRIB = 0, 1, 2, 3
 
 
@@ -849,7 +856,7 @@ class Booster:
             _safe_call(_LIB.LGBM_BoosterGetEval(
                 self._handle, ctypes.c_int(data_idx), ctypes.byref(out_len),
                 res.ctypes.data_as(ctypes.POINTER(ctypes.c_double))))
-            higher_better = [n in ("auc", "ndcg", "map", "average_precision") or
+            higher_better = [n in _HIGHER_BETTER_METRICS or
                              n.startswith(("auc", "ndcg@", "map@")) for n in names]
             for i in range(out_len.value):
                 out.append((name, names[i], res[i], higher_better[i]))
@@ -991,12 +998,13 @@ class Booster:
     # ------------------------------------------------------------ serialization
     def save_model(self, filename, num_iteration=None, start_iteration=0,
                    importance_type="split"):
-        imp = 0 if importance_type == "split" else 1
-        if num_iteration is None:
-            num_iteration = self.best_iteration if self.best_iteration > 0 else -1
-        _safe_call(_LIB.LGBM_BoosterSaveModel(
-            self._handle, ctypes.c_int(start_iteration), ctypes.c_int(num_iteration),
-            ctypes.c_int(imp), _c_str(str(filename))))
+        # write via model_to_string so the pandas_categorical trailer survives a
+        # file round-trip (reference Booster.save_model does the same)
+        text = self.model_to_string(num_iteration=num_iteration,
+                                    start_iteration=start_iteration,
+                                    importance_type=importance_type)
+        with open(str(filename), "w") as f:
+            f.write(text)
         return self
 
     def model_to_string(self, num_iteration=None, start_iteration=0,

@@ -82,6 +82,11 @@ def train(params, train_set, num_boost_round=100, valid_sets=None, valid_names=N
                 # training data as eval set: use data_idx 0 via eval_train naming
                 booster._train_as_valid_name = name
                 continue
+            # a valid set must share the training data's bin mappers — bind it
+            # before construction (reference engine.py does the same); an
+            # independently-binned valid set would score trees on wrong bins
+            if vs._handle is None and train_set not in vs.get_ref_chain():
+                vs.set_reference(train_set)
             vs.construct()
             booster.add_valid(vs, name)
 

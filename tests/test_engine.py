@@ -807,3 +807,74 @@ def test_deterministic_thread_invariance():
                           lgb.Dataset(X, label=y), 10).model_to_string())
           for t in (1, 4, 8)]
     assert ms[0] == ms[1] == ms[2]
+
+
+def test_unbound_valid_set_auto_references_train():
+    """A valid set passed without reference= is auto-bound to the training
+    data's bin mappers by train() (independently-binned valid
+    sets silently corrupt eval metrics and early stopping)."""
+    X, y = _regression_data()
+    tr = lgb.Dataset(X[:4000], label=y[:4000])
+    # plain Dataset, NO reference= — subset of the training data
+    va = lgb.Dataset(X[:800], label=y[:800])
+    ev = {}
+    lgb.train({"objective": "regression", "metric": "l2", "verbosity": -1},
+              tr, 30, valid_sets=[va], callbacks=[lgb.record_evaluation(ev)])
+    l2 = ev["valid_0"]["l2"]
+    # on a training subset l2 must improve monotonically-ish and end well below var
+    assert l2[-1] < 0.5 * float(np.var(y[:800]))
+    assert l2[-1] < l2[0]
+
+
+def test_misaligned_valid_set_rejected():
+    """Adding an independently-constructed (differently-binned) valid set to a
+    Booster raises instead of producing garbage metrics."""
+    X, y = _regression_data()
+    tr = lgb.Dataset(X[:4000], label=y[:4000]).construct()
+    va = lgb.Dataset(X[4000:] * 3.7 + 1.0, label=y[4000:]).construct()
+    bst = lgb.Booster(params={"objective": "regression", "verbosity": -1},
+                      train_set=tr)
+    with pytest.raises(lgb.LightGBMError):
+        bst.add_valid(va, "bad")
+
+
+def test_early_stopping_respects_r2_direction():
+    """r2 is higher-better; early stopping must not stop while it improves
+    (ADVICE r1 medium)."""
+    X, y = _regression_data()
+    tr = lgb.Dataset(X[:4000], label=y[:4000])
+    va = tr.create_valid(X[4000:], label=y[4000:])
+    ev = {}
+    bst = lgb.train({"objective": "regression", "metric": "r2", "verbosity": -1,
+                     "learning_rate": 0.1},
+                    tr, 60, valid_sets=[va],
+                    callbacks=[lgb.early_stopping(5), lgb.record_evaluation(ev)])
+    r2 = ev["valid_0"]["r2"]
+    # r2 climbs for many rounds on this problem; a direction bug stops at iter 1
+    assert bst.best_iteration > 10
+    assert max(r2) > 0.8
+
+
+def test_save_model_keeps_pandas_categorical(tmp_path):
+    """save_model writes the pandas_categorical trailer so a file round-trip
+    preserves the training category mapping."""
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(3)
+    n = 2000
+    df = pd.DataFrame({
+        "num": rng.randn(n),
+        "cat": pd.Categorical(rng.choice(["a", "b", "c", "d"], size=n)),
+    })
+    y = (df["num"] + (df["cat"].cat.codes % 2) > 0.3).astype(np.float32)
+    bst = lgb.train({"objective": "binary", "verbosity": -1},
+                    lgb.Dataset(df, label=y), 15)
+    p_before = bst.predict(df)
+    path = tmp_path / "m.txt"
+    bst.save_model(path)
+    bst2 = lgb.Booster(model_file=str(path))
+    assert bst2.pandas_categorical is not None
+    # shuffled category declaration order must still map via the saved trailer
+    df2 = df.copy()
+    df2["cat"] = pd.Categorical(df["cat"].astype(str),
+                                categories=["d", "c", "b", "a"])
+    np.testing.assert_allclose(bst2.predict(df2), p_before, rtol=1e-9)
