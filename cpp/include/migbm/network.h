@@ -32,6 +32,7 @@ class Network {
   /*! element-wise sum allreduce of doubles (histograms, stats). */
   static void AllreduceSum(double* data, size_t n);
   static void AllreduceSum(int64_t* data, size_t n);
+  static void AllreduceSum(float* data, size_t n);
   /*! generic byte allgather: every rank contributes `size` bytes; out = world x size. */
   static void Allgather(const char* input, int size, char* output);
   /*! allgather with per-rank sizes */

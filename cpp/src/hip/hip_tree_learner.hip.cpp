@@ -25,6 +25,7 @@
 
 #include "migbm/tree_learner.h"
 #include "migbm/objective.h"
+#include "migbm/network.h"
 
 #include <algorithm>
 
@@ -47,16 +48,100 @@ namespace migbm {
   } while (0)
 
 // ------------------------------------------------------------------ GPU comm singleton
+/*! Transport-agnostic inter-GPU collectives. Two transports:
+ *  - RCCL (ncclCommInitRank via LGBM_GPUNetworkInit): the production path, one
+ *    process per GPU over xGMI, stream-ordered (no host syncs).
+ *  - host seam (migbm::Network, injected allgather): correctness transport used
+ *    when several ranks share one GPU (RCCL refuses duplicate devices) — this is
+ *    how the multi-rank device code path is proven on a single-GPU box. Each call
+ *    syncs the stream and bounces through host memory.
+ *  Both run the IDENTICAL device-side code; only the reduction transport differs. */
 struct GpuComm {
   ncclComm_t comm = nullptr;
   int world = 1;
   int rank = 0;
-  bool active() const { return comm != nullptr && world > 1; }
+  bool active() const { return comm != nullptr || Network::is_distributed(); }
+  bool rccl() const { return comm != nullptr; }
+  int World() const { return rccl() ? world : Network::num_machines(); }
+  int Rank() const { return rccl() ? rank : Network::rank(); }
+
+  void AllReduce(float* d, size_t n, hipStream_t s) {
+    if (rccl()) {
+      NCCL_OK(ncclAllReduce(d, d, n, ncclFloat32, ncclSum, comm, s));
+    } else {
+      HostBounce<float>(d, n, s);
+    }
+  }
+  void AllReduce(double* d, size_t n, hipStream_t s) {
+    if (rccl()) {
+      NCCL_OK(ncclAllReduce(d, d, n, ncclFloat64, ncclSum, comm, s));
+    } else {
+      HostBounce<double>(d, n, s);
+    }
+  }
+  void AllReduce(int64_t* d, size_t n, hipStream_t s) {
+    if (rccl()) {
+      NCCL_OK(ncclAllReduce(d, d, n, ncclInt64, ncclSum, comm, s));
+    } else {
+      HostBounce<int64_t>(d, n, s);
+    }
+  }
+  /*! per-rank-owned block reduce: rank r ends with the globally-summed
+   *  [off[r], off[r]+cnt[r]) slice; other slices are left partial (garbage).
+   *  RCCL: grouped ncclReduce, one per root. Host seam: plain allreduce (a
+   *  correct superset — owned slices are what the kernels read). */
+  void ReduceBlocks(float* d, const std::vector<size_t>& off, const std::vector<size_t>& cnt,
+                    hipStream_t s) {
+    if (rccl()) {
+      NCCL_OK(ncclGroupStart());
+      for (int r = 0; r < world; ++r) {
+        if (cnt[r] == 0) continue;
+        NCCL_OK(ncclReduce(d + off[r], d + off[r], cnt[r], ncclFloat32, ncclSum, r, comm, s));
+      }
+      NCCL_OK(ncclGroupEnd());
+    } else {
+      HostBounce<float>(d, off.back() + cnt.back(), s);
+    }
+  }
+  /*! byte allgather: world * bytes_per_rank into d_out (rank-major). */
+  void AllGather(const void* d_in, void* d_out, size_t bytes_per_rank, hipStream_t s);
+
   static GpuComm& Get() {
     static GpuComm c;
     return c;
   }
+
+ private:
+  template <typename T>
+  void HostBounce(T* d, size_t n, hipStream_t s) {
+    if (hipStreamSynchronize(s) != hipSuccess)
+      Log::Fatal("GpuComm host-bounce: stream sync failed");
+    staging_.resize(n * sizeof(T));
+    T* h = reinterpret_cast<T*>(staging_.data());
+    if (hipMemcpy(h, d, n * sizeof(T), hipMemcpyDeviceToHost) != hipSuccess)
+      Log::Fatal("GpuComm host-bounce: D2H failed");
+    Network::AllreduceSum(h, n);
+    if (hipMemcpy(d, h, n * sizeof(T), hipMemcpyHostToDevice) != hipSuccess)
+      Log::Fatal("GpuComm host-bounce: H2D failed");
+  }
+  std::vector<char> staging_;
 };
+
+void GpuComm::AllGather(const void* d_in, void* d_out, size_t bytes_per_rank, hipStream_t s) {
+  if (rccl()) {
+    NCCL_OK(ncclAllGather(d_in, d_out, bytes_per_rank, ncclInt8, comm, s));
+  } else {
+    if (hipStreamSynchronize(s) != hipSuccess)
+      Log::Fatal("GpuComm host-bounce: stream sync failed");
+    const int w = Network::num_machines();
+    std::vector<char> h_in(bytes_per_rank), h_out(bytes_per_rank * w);
+    if (hipMemcpy(h_in.data(), d_in, bytes_per_rank, hipMemcpyDeviceToHost) != hipSuccess)
+      Log::Fatal("GpuComm host-bounce: D2H failed");
+    Network::Allgather(h_in.data(), static_cast<int>(bytes_per_rank), h_out.data());
+    if (hipMemcpy(d_out, h_out.data(), bytes_per_rank * w, hipMemcpyHostToDevice) != hipSuccess)
+      Log::Fatal("GpuComm host-bounce: H2D failed");
+  }
+}
 
 namespace hipk {
 
@@ -98,6 +183,9 @@ struct GainParams {
   double cat_l2, cat_smooth;
   int max_cat_to_onehot, max_cat_threshold;
   int n_interaction_groups;  // 0 = unconstrained
+  // distributed feature ownership: this rank scans only features [own_fb, own_fe)
+  // (reduce-scatter mode divides both wire bytes and the gain scan by world)
+  int own_fb, own_fe;
 };
 
 __device__ __forceinline__ uint32_t d_hash3(uint32_t a, uint32_t b, uint32_t c) {
@@ -488,6 +576,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
     rec.feature = f;
     rec.cat_mask = 0;
   }
+  if (f < p.own_fb || f >= p.own_fe) return;  // another rank owns this feature's scan
   if (feat_mask != nullptr && !feat_mask[f]) return;
   if (p.n_interaction_groups > 0) {
     // interaction constraints (inner-feature bitmask form, nf <= 64): feature f
@@ -1344,6 +1433,49 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
 }
 
 
+// -------------------------------------------------------- distributed winner sync
+/*! pack this rank's overall winner (best split over its OWNED features, all
+ *  leaves) into a wire record for the cross-rank allgather. */
+__global__ void k_pack_winner(const SplitRec* __restrict__ winner,
+                              const int* __restrict__ winner_leaf,
+                              LogEntry* __restrict__ wire) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    wire->rec = *winner;
+    wire->leaf = *winner_leaf;
+  }
+}
+
+/*! deterministic argmax over the gathered per-rank winners; every rank picks the
+ *  identical global winner (ties broken by feature index, then leaf). Replaces
+ *  the reference's SyncUpGlobalBestSplit custom allreduce
+ *  (parallel_tree_learner.h:209-232) with allgather + device argmax. */
+__global__ void k_pick_global_winner(const LogEntry* __restrict__ wires, int world,
+                                     SplitRec* __restrict__ winner,
+                                     int* __restrict__ winner_leaf) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  int best = -1;
+  for (int r = 0; r < world; ++r) {
+    if (wires[r].leaf < 0 || !wires[r].rec.valid) continue;
+    if (best < 0) {
+      best = r;
+      continue;
+    }
+    const double g1 = wires[r].rec.gain, g0 = wires[best].rec.gain;
+    if (g1 > g0 ||
+        (g1 == g0 && (wires[r].rec.feature < wires[best].rec.feature ||
+                      (wires[r].rec.feature == wires[best].rec.feature &&
+                       wires[r].leaf < wires[best].leaf)))) {
+      best = r;
+    }
+  }
+  if (best < 0) {
+    *winner_leaf = -1;
+  } else {
+    *winner = wires[best].rec;
+    *winner_leaf = wires[best].leaf;
+  }
+}
+
 // ------------------------------------------------------------------ boosting kernels
 __global__ void k_grad_binary(const double* __restrict__ score,
                               const float* __restrict__ label,
@@ -1612,6 +1744,10 @@ class HIPTreeLearner : public TreeLearner {
   void LaunchHist(const int* leafA_ptr, int leafB_from_counters, int blocks, bool zero_spare = true);
   void ReduceSpareHist(int spare_slot);
   void LaunchBestSplit(const int* leafA_ptr, int leafB_from_counters);
+  /*! balanced contiguous feature-block ownership for reduce-scatter mode */
+  bool SetupOwnership(int world, int rank);
+  /*! allgather per-rank winners + deterministic global argmax (ownership mode) */
+  void SyncGlobalWinner();
   int HistBlocksFor(int approx_cnt) const {
     // LDS atomic throughput is per-CU: spread even small leaves over many blocks
     // (~256 rows each); cap so the per-block flush stays amortized at the root.
@@ -1698,6 +1834,14 @@ class HIPTreeLearner : public TreeLearner {
   std::vector<int> approx_cnt_;              // launch-sizing hints during the tree loop
   Random feature_rng_{0};
   std::vector<int8_t> feat_mask_host_;
+
+  // distributed state (multi-GPU data-parallel)
+  bool dist_ = false;       // any cross-rank transport active this tree
+  bool own_scan_ = false;   // reduce-scatter + per-rank feature-ownership scan
+  int own_fb_ = 0, own_fe_ = 0;            // this rank's owned feature range
+  int own_world_ = 0;                      // world the ownership plan was built for
+  std::vector<size_t> own_off_, own_cnt_;  // per-rank hist float offsets/counts
+  DevBuf<hipk::LogEntry> d_wire_my_, d_wire_all_;
 
   static constexpr int kHistBlock = 256;
   /*! hist-kernel workgroup size (k_hist is blockDim-agnostic; partition kernels
@@ -2137,10 +2281,59 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
 
 void HIPTreeLearner::ReduceSpareHist(int spare_slot) {
   auto& comm = GpuComm::Get();
-  if (!comm.active()) return;
+  if (!dist_) return;
   float* spare = d_hist_.ptr + static_cast<size_t>(spare_slot) * total_bins_ * 2;
-  NCCL_OK(ncclAllReduce(spare, spare, static_cast<size_t>(total_bins_) * 2, ncclFloat32,
-                        ncclSum, comm.comm, stream_));
+  if (own_scan_) {
+    // reduce-scatter analogue: rank r receives the global sum of its owned
+    // feature block only (matches the reference data-parallel learner's
+    // ReduceScatter + owned-feature gain scan, data_parallel_tree_learner.cpp:283-450)
+    comm.ReduceBlocks(spare, own_off_, own_cnt_, stream_);
+  } else {
+    comm.AllReduce(spare, static_cast<size_t>(total_bins_) * 2, stream_);
+  }
+}
+
+bool HIPTreeLearner::SetupOwnership(int world, int rank) {
+  if (world <= 1 || nf_ < world) return false;
+  if (own_world_ == world && !own_off_.empty()) return true;  // plan is per-dataset
+  // contiguous feature blocks balanced by bin count (hist offsets are contiguous
+  // ascending, Dataset::FinishBinMappers)
+  std::vector<int> fb(world + 1, nf_);
+  fb[0] = 0;
+  int f = 0;
+  for (int r = 1; r < world; ++r) {
+    const double target = static_cast<double>(total_bins_) * r / world;
+    while (f < nf_ - (world - r) &&
+           feat_meta_host_[f].bin_off + feat_meta_host_[f].num_bin / 2 < target)
+      ++f;
+    f = std::max(f, fb[r - 1] + 1);  // at least one feature per rank
+    fb[r] = f;
+  }
+  own_off_.assign(world, 0);
+  own_cnt_.assign(world, 0);
+  for (int r = 0; r < world; ++r) {
+    const int b0 = feat_meta_host_[fb[r]].bin_off;
+    const int b1 = fb[r + 1] < nf_ ? feat_meta_host_[fb[r + 1]].bin_off : total_bins_;
+    own_off_[r] = static_cast<size_t>(b0) * 2;
+    own_cnt_[r] = static_cast<size_t>(b1 - b0) * 2;
+  }
+  own_fb_ = fb[rank];
+  own_fe_ = fb[rank + 1];
+  own_world_ = world;
+  if (!d_wire_my_.ptr) {
+    d_wire_my_.Alloc(1);
+    d_wire_all_.Alloc(world);
+  }
+  return true;
+}
+
+void HIPTreeLearner::SyncGlobalWinner() {
+  auto& comm = GpuComm::Get();
+  hipLaunchKernelGGL(hipk::k_pack_winner, dim3(1), dim3(1), 0, stream_, d_winner_.ptr,
+                     d_winner_leaf_.ptr, d_wire_my_.ptr);
+  comm.AllGather(d_wire_my_.ptr, d_wire_all_.ptr, sizeof(hipk::LogEntry), stream_);
+  hipLaunchKernelGGL(hipk::k_pick_global_winner, dim3(1), dim3(1), 0, stream_,
+                     d_wire_all_.ptr, comm.World(), d_winner_.ptr, d_winner_leaf_.ptr);
 }
 
 void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counters) {
@@ -2160,6 +2353,8 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.n_interaction_groups = n_interaction_groups_;
   p.extra_trees = config_->extra_trees ? 1 : 0;
   p.rng_seed = bynode_seed_;
+  p.own_fb = own_scan_ ? own_fb_ : 0;
+  p.own_fe = own_scan_ ? own_fe_ : nf_;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
   const int ny = leafB_from_counters ? 2 : 1;
   hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_, d_hist_.ptr,
@@ -2227,11 +2422,23 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        d_leaf_stats_.ptr);
   }
   auto& comm = GpuComm::Get();
-  if (comm.active()) {
-    NCCL_OK(ncclAllReduce(&d_leaf_stats_.ptr[0].sum_g, &d_leaf_stats_.ptr[0].sum_g, 2,
-                          ncclFloat64, ncclSum, comm.comm, stream_));
-    NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
-                          stream_));
+  dist_ = comm.active();
+  own_scan_ = false;
+  if (dist_ && comm.World() > 1) {
+    // comm-path selection: full-histogram allreduce for small
+    // payloads (latency-bound over xGMI), reduce-scatter + per-rank feature
+    // ownership for large ones (divides wire bytes AND the gain scan by world).
+    // MIGBM_DIST_HIST=allreduce|reduce_scatter overrides the size heuristic.
+    const char* e = getenv("MIGBM_DIST_HIST");
+    const std::string mode = e ? e : "auto";
+    const size_t payload = static_cast<size_t>(total_bins_) * 2 * sizeof(float);
+    const bool want_rs =
+        mode == "reduce_scatter" || (mode == "auto" && payload >= (128u << 10));
+    if (want_rs) own_scan_ = SetupOwnership(comm.World(), comm.Rank());
+  }
+  if (dist_) {
+    comm.AllReduce(&d_leaf_stats_.ptr[0].sum_g, 2, stream_);
+    comm.AllReduce(d_gbuf_.ptr, 1, stream_);
   }
   {
     // global root count + the root's own output (the smoothing parent for depth-1
@@ -2248,6 +2455,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   LaunchHist(d_root_leaf_.ptr, 0, HistBlocksFor(static_cast<int>(used_cnt_)));
   ReduceSpareHist(0);
   LaunchBestSplit(d_root_leaf_.ptr, 0);
+  if (own_scan_) SyncGlobalWinner();
 
   // ---- device-driven split loop: ZERO host syncs; winner decisions accumulate in
   // d_split_log_ and the tree is replayed on the host after one end-of-tree download.
@@ -2294,10 +2502,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                          d_idx_tmp_.ptr, d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr,
                          d_winner_leaf_.ptr, d_ctr_.ptr, d_gbuf_.ptr);
     }
-    if (comm.active()) {
-      NCCL_OK(ncclAllReduce(d_gbuf_.ptr, d_gbuf_.ptr, 1, ncclInt64, ncclSum, comm.comm,
-                            stream_));
-    }
+    if (dist_) comm.AllReduce(d_gbuf_.ptr, 1, stream_);
     hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(256), 0, stream_, d_leaf_begin_.ptr,
                        d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
                        d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
@@ -2314,6 +2519,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                          d_counters_.ptr, n_elem);
     }
     LaunchBestSplit(d_winner_leaf_.ptr, 1);
+    if (own_scan_) SyncGlobalWinner();
   }
 
   // ---- one download: split log + exact leaf layout; replay the tree on the host
@@ -2359,10 +2565,20 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                   mapper->missing_type(), w.default_left != 0);
     }
   }
-  // fix the tree's leaf counts with exact values (internal counts recomputed inside)
+  // fix the tree's leaf counts with exact values (internal counts recomputed
+  // inside). Distributed: LeafStat.cnt carries the GLOBAL count so every rank's
+  // model (and its leaf_count fields) is byte-identical.
   {
     std::vector<int> counts(tree->num_leaves());
-    for (int l = 0; l < tree->num_leaves(); ++l) counts[l] = leaf_cnt_[l];
+    if (dist_) {
+      std::vector<hipk::LeafStat> st(tree->num_leaves());
+      HIP_OK(hipMemcpy(st.data(), d_leaf_stats_.ptr,
+                       sizeof(hipk::LeafStat) * tree->num_leaves(),
+                       hipMemcpyDeviceToHost));
+      for (int l = 0; l < tree->num_leaves(); ++l) counts[l] = st[l].cnt;
+    } else {
+      for (int l = 0; l < tree->num_leaves(); ++l) counts[l] = leaf_cnt_[l];
+    }
     tree->OverrideLeafCounts(counts);
   }
   return tree.release();

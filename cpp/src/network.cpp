@@ -62,6 +62,24 @@ void Network::AllreduceSum(double* data, size_t n) {
   }
 }
 
+void Network::AllreduceSum(float* data, size_t n) {
+  if (num_machines_ <= 1) return;
+  const size_t chunk = 1 << 20;
+  std::vector<char> out;
+  for (size_t off = 0; off < n; off += chunk) {
+    size_t m = std::min(chunk, n - off);
+    out.resize(m * sizeof(float) * num_machines_);
+    Allgather(reinterpret_cast<const char*>(data + off), static_cast<int>(m * sizeof(float)),
+              out.data());
+    const float* gathered = reinterpret_cast<const float*>(out.data());
+    for (size_t i = 0; i < m; ++i) {
+      float s = 0;
+      for (int r = 0; r < num_machines_; ++r) s += gathered[static_cast<size_t>(r) * m + i];
+      data[off + i] = s;
+    }
+  }
+}
+
 void Network::AllreduceSum(int64_t* data, size_t n) {
   if (num_machines_ <= 1) return;
   std::vector<char> out(n * sizeof(int64_t) * num_machines_);

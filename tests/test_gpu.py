@@ -310,3 +310,54 @@ def test_gpu_path_smooth():
     assert not np.allclose(base.predict(X[:100]), sm.predict(X[:100]))
     r2 = 1 - np.mean((sm.predict(X) - y) ** 2) / np.var(y)
     assert r2 > 0.8
+
+
+def _run_dist_gpu(tmp_path, mode, world=2, extra=()):
+    """torchrun `world` ranks on GPU 0 through the device data-parallel learner."""
+    import subprocess, sys, os, json, random
+    from pathlib import Path
+    out = tmp_path / mode
+    out.mkdir(parents=True, exist_ok=True)
+    env = dict(os.environ)
+    env["MIGBM_DIST_HIST"] = mode
+    env["HIP_VISIBLE_DEVICES"] = env.get("HIP_VISIBLE_DEVICES", "0").split(",")[0]
+    port = random.randint(20000, 59000)
+    worker = Path(__file__).parent / "helpers" / "gpu_dist_worker.py"
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           f"--nproc-per-node={world}", "--master-addr", "127.0.0.1",
+           "--master-port", str(port), str(worker), "--out-dir", str(out)] + list(extra)
+    r = subprocess.run(cmd, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, f"worker failed:\n{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
+    models = [(out / f"model_rank{i}.txt").read_text() for i in range(world)]
+    res = json.loads((out / "result.json").read_text())
+    return models, res
+
+
+def test_gpu_dist_world2_allreduce(tmp_path):
+    """Two ranks share GPU 0 (host-seam transport, identical device code path to
+    RCCL): full-histogram allreduce mode. Every rank must build the byte-identical
+    model, and quality must match a single-process run (VERDICT r1 next-round #1)."""
+    models, res = _run_dist_gpu(tmp_path, "allreduce")
+    assert models[0] == models[1]
+    assert res["num_trees"] == 10
+    assert res["auc_full"] > 0.80
+
+
+def test_gpu_dist_world2_reduce_scatter(tmp_path):
+    """Reduce-scatter mode: per-rank feature-block ownership, owner-only gain
+    scan, winner allgather + device argmax. Models identical across ranks and
+    quality parity with the allreduce mode."""
+    m_rs, res_rs = _run_dist_gpu(tmp_path, "reduce_scatter")
+    assert m_rs[0] == m_rs[1]
+    assert res_rs["auc_full"] > 0.80
+    # ownership must not change WHAT is learned, only who scans what: compare
+    # against the allreduce mode at identical seeds/config
+    m_ar, res_ar = _run_dist_gpu(tmp_path, "allreduce")
+    assert abs(res_rs["auc_full"] - res_ar["auc_full"]) < 2e-3
+
+
+def test_gpu_dist_world3_reduce_scatter(tmp_path):
+    """Odd world size exercises unbalanced feature-block ownership."""
+    models, res = _run_dist_gpu(tmp_path, "reduce_scatter", world=3)
+    assert models[0] == models[1] == models[2]
+    assert res["auc_full"] > 0.80
