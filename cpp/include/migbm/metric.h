@@ -14,6 +14,35 @@
 
 namespace migbm {
 
+/*! per-point loss kinds the HIP learner can reduce on device
+ *  (keep in sync with k_metric_pointwise in hip_tree_learner.hip.cpp).
+ *  Capability parity: reference src/metric/cuda/cuda_pointwise_metric.cu. */
+enum PwLossKind : int {
+  kPwL2 = 0,
+  kPwL1,
+  kPwQuantile,   // a = alpha
+  kPwHuber,      // a = alpha
+  kPwFair,       // a = fair_c
+  kPwPoisson,
+  kPwMape,
+  kPwGamma,
+  kPwGammaDev,
+  kPwTweedie,    // a = tweedie_variance_power
+  kPwBinaryLogloss,
+  kPwBinaryError,
+  kPwXent,
+  kPwXentLambda,
+};
+
+/*! descriptor for device-side pointwise metric evaluation: the HIP learner
+ *  computes (Σ w·loss, Σ w) on the GPU so train-metric eval stops downloading
+ *  the full score vector. kind < 0 = not device-evaluable (AUC, NDCG, ...). */
+struct PointwiseEvalDesc {
+  int kind = -1;
+  double a = 0.0;
+  bool convert = true;  // apply the objective's output transform to the score first
+};
+
 class Metric {
  public:
   virtual ~Metric() = default;
@@ -23,6 +52,11 @@ class Metric {
   virtual double factor_to_bigger_better() const = 0;
   virtual std::vector<double> Eval(const double* score,
                                    const ObjectiveFunction* objective) const = 0;
+  /*! device fast path: descriptor (kind -1 = unsupported) */
+  virtual PointwiseEvalDesc pointwise_desc() const { return {}; }
+  /*! finish a device (loss_sum, weight_sum) pair into the metric value;
+   *  applies the distributed reduce + the metric's final transform */
+  virtual double FinalizeFromSums(double sum, double w) const { return sum / w; }
 
   static Metric* Create(const std::string& name, const Config& config);
 };

@@ -57,6 +57,18 @@ class TreeLearner {
   virtual void DownloadTrainScore(double* dst) { (void)dst; }
   /*! push host-modified train scores back to the device (DART drop/renormalize). */
   virtual void UploadTrainScore(const double* src) { (void)src; }
+  /*! device pointwise train-metric eval: reduce (Σ w·loss, Σ w) on the GPU so
+   *  eval does not download the score vector. `convert_kind` selects the
+   *  objective's output transform (0 identity, 1 exp, 2 sigmoid(param),
+   *  3 logistic, 4 log1p(exp), 5 signed-square). Returns false when the learner
+   *  cannot evaluate this loss on device (host fallback). */
+  virtual bool DeviceEvalPointwise(int loss_kind, double loss_a, int convert_kind,
+                                   double convert_param, double* out_sum,
+                                   double* out_wsum) {
+    (void)loss_kind; (void)loss_a; (void)convert_kind; (void)convert_param;
+    (void)out_sum; (void)out_wsum;
+    return false;
+  }
 
   static TreeLearner* Create(const std::string& learner_type, const std::string& device_type,
                              const Config* config);

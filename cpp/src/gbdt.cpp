@@ -366,10 +366,71 @@ void GBDT::UpdateScore(const Tree* tree, int cur_tree_id) {
   }
 }
 
+/*! objective output-transform id for the device metric kernel:
+ *  0 identity, 1 exp, 2 sigmoid(param), 3 logistic, 4 log1p(exp), 5 signed-square;
+ *  -1 = unknown (forces the host eval path). */
+static int DeviceConvertKind(const ObjectiveFunction* obj, const Config* cfg,
+                             double* param) {
+  *param = 0.0;
+  if (obj == nullptr) return -1;
+  const std::string name = obj->GetName();
+  static const char* kIdentity[] = {"regression", "regression_l1", "huber", "fair",
+                                    "quantile",   "mape",          "lambdarank",
+                                    "rank_xendcg"};
+  for (const char* s : kIdentity)
+    if (name == s) return cfg->reg_sqrt ? 5 : 0;
+  if (name == "poisson" || name == "gamma" || name == "tweedie") return 1;
+  if (name == "binary") {
+    *param = cfg->sigmoid;
+    return 2;
+  }
+  if (name == "cross_entropy") return 3;
+  if (name == "cross_entropy_lambda") return 4;
+  return -1;
+}
+
 std::vector<double> GBDT::GetEvalAt(int data_idx) const {
   std::vector<double> out;
   if (data_idx == 0 && tree_learner_ && tree_learner_->IsHIPLearner()) {
-    tree_learner_->DownloadTrainScore(const_cast<double*>(train_score_.data()));
+    // device fast path: pointwise train metrics reduce (Σ w·loss, Σ w) on the
+    // GPU; the full score vector is downloaded only if some metric needs it
+    // (AUC/NDCG/..). Capability parity: reference cuda_pointwise_metric.cu.
+    double cparam = 0.0;
+    const int ckind = DeviceConvertKind(objective_, config_, &cparam);
+    std::vector<const Metric*> host_metrics;
+    std::vector<std::pair<size_t, double>> dev_vals;  // (output slot, value)
+    size_t slot = 0;
+    bool all_dev = ckind >= 0;
+    for (auto* m : training_metrics_) {
+      const auto desc = m->pointwise_desc();
+      double s = 0, w = 0;
+      if (all_dev && desc.kind >= 0 &&
+          tree_learner_->DeviceEvalPointwise(desc.kind, desc.a,
+                                             desc.convert ? ckind : 0, cparam, &s, &w)) {
+        dev_vals.emplace_back(slot, m->FinalizeFromSums(s, w));
+        slot += 1;
+      } else {
+        host_metrics.push_back(m);
+        slot += m->GetName().size();
+      }
+    }
+    if (!host_metrics.empty()) {
+      tree_learner_->DownloadTrainScore(const_cast<double*>(train_score_.data()));
+    }
+    // rebuild outputs in metric order
+    size_t dev_i = 0;
+    for (auto* m : training_metrics_) {
+      const auto desc = m->pointwise_desc();
+      if (all_dev && desc.kind >= 0 && dev_i < dev_vals.size() &&
+          dev_vals[dev_i].first == out.size()) {
+        out.push_back(dev_vals[dev_i].second);
+        ++dev_i;
+      } else {
+        auto r = m->Eval(train_score_.data(), objective_);
+        out.insert(out.end(), r.begin(), r.end());
+      }
+    }
+    return out;
   }
   if (data_idx == 0) {
     for (auto* m : training_metrics_) {

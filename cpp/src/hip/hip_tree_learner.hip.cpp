@@ -1502,6 +1502,313 @@ __global__ void k_grad_l2(const double* __restrict__ score, const float* __restr
   h[i] = static_cast<float>(w);
 }
 
+// pointwise objective kinds (host dispatch: HIPTreeLearner::DeviceBoosting).
+// Gradient formulas mirror cpp/src/objective.cpp exactly (capability parity:
+// reference src/objective/cuda/cuda_regression_objective.cu, cuda_xentropy...).
+enum GradKind : int {
+  kGradL1 = 0,
+  kGradHuber,      // a = alpha
+  kGradFair,       // a = fair_c
+  kGradPoisson,    // a = poisson_max_delta_step
+  kGradQuantile,   // a = alpha
+  kGradMape,
+  kGradGamma,
+  kGradTweedie,    // a = tweedie_variance_power
+  kGradXent,
+  kGradXentLambda,
+};
+
+__global__ void k_grad_pointwise(int kind, const double* __restrict__ score,
+                                 const float* __restrict__ label,
+                                 const float* __restrict__ weight, int n, double a,
+                                 float* __restrict__ g, float* __restrict__ h) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const double w = weight ? weight[i] : 1.0;
+  const double y = label[i];
+  const double s = score[i];
+  const double d = s - y;
+  double gv = 0.0, hv = 1.0;
+  switch (kind) {
+    case kGradL1:
+      gv = d > 0 ? w : -w;
+      hv = w;
+      break;
+    case kGradHuber:
+      gv = fabs(d) <= a ? w * d : w * (d > 0 ? a : -a);
+      hv = w;
+      break;
+    case kGradFair: {
+      const double den = fabs(d) + a;
+      gv = w * a * d / den;
+      hv = w * a * a / (den * den);
+      break;
+    }
+    case kGradPoisson: {
+      const double e = exp(s);
+      gv = w * (e - y);
+      hv = w * exp(s + a);
+      break;
+    }
+    case kGradQuantile:
+      gv = d >= 0 ? w * (1.0 - a) : -w * a;
+      hv = w;
+      break;
+    case kGradMape: {
+      const double lw = w / fmax(1.0, fabs(y));
+      gv = d > 0 ? lw : -lw;
+      hv = lw;
+      break;
+    }
+    case kGradGamma: {
+      const double e = exp(-s);
+      gv = w * (1.0 - y * e);
+      hv = w * y * e;
+      break;
+    }
+    case kGradTweedie: {
+      const double e1 = exp((1.0 - a) * s);
+      const double e2 = exp((2.0 - a) * s);
+      gv = w * (-y * e1 + e2);
+      hv = w * (-y * (1.0 - a) * e1 + (2.0 - a) * e2);
+      break;
+    }
+    case kGradXent: {
+      const double p = 1.0 / (1.0 + exp(-s));
+      gv = w * (p - y);
+      hv = w * p * (1.0 - p);
+      break;
+    }
+    case kGradXentLambda: {
+      const double epf = exp(s);
+      const double hhat = log1p(epf);
+      const double z = 1.0 - exp(-w * hhat);
+      const double enf = 1.0 / epf;
+      gv = (1.0 - y / z) * w / (1.0 + enf);
+      const double c = 1.0 / (1.0 - z);
+      const double dd = 1.0 + epf;
+      const double aa = w * epf / (dd * dd);
+      hv = aa * (1.0 + y * (1.0 - c * (1.0 + w * epf / dd * (1.0 - c))));
+      break;
+    }
+  }
+  g[i] = static_cast<float>(gv);
+  h[i] = static_cast<float>(hv);
+}
+
+/*! multiclass softmax gradients, all classes at once (class-major layout,
+ *  stride n). Mirrors MulticlassSoftmax::GetGradients (objective.cpp:402). */
+__global__ void k_grad_multiclass(const double* __restrict__ score,
+                                  const float* __restrict__ label,
+                                  const float* __restrict__ weight, int n, int num_class,
+                                  float* __restrict__ g, float* __restrict__ h) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const double w = weight ? weight[i] : 1.0;
+  const double factor = static_cast<double>(num_class) / (num_class - 1);
+  double mx = -1e300;
+  for (int c = 0; c < num_class; ++c)
+    mx = fmax(mx, score[static_cast<size_t>(c) * n + i]);
+  double sum = 0;
+  for (int c = 0; c < num_class; ++c) sum += exp(score[static_cast<size_t>(c) * n + i] - mx);
+  const int lbl = static_cast<int>(label[i]);
+  for (int c = 0; c < num_class; ++c) {
+    const size_t k = static_cast<size_t>(c) * n + i;
+    const double p = exp(score[k] - mx) / sum;
+    g[k] = static_cast<float>(w * (p - (c == lbl ? 1.0 : 0.0)));
+    h[k] = static_cast<float>(w * factor * p * (1.0 - p));
+  }
+}
+
+/*! one-vs-all: per-class sigmoid binary gradients (MulticlassOVA parity). */
+__global__ void k_grad_multiclass_ova(const double* __restrict__ score,
+                                      const float* __restrict__ label,
+                                      const float* __restrict__ weight, int n,
+                                      int num_class, double sigmoid,
+                                      float* __restrict__ g, float* __restrict__ h) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const double w = weight ? weight[i] : 1.0;
+  const int lbl = static_cast<int>(label[i]);
+  for (int c = 0; c < num_class; ++c) {
+    const size_t k = static_cast<size_t>(c) * n + i;
+    const double y = lbl == c ? 1.0 : -1.0;
+    const double response = -y * sigmoid / (1.0 + exp(y * sigmoid * score[k]));
+    const double ar = fabs(response);
+    g[k] = static_cast<float>(response * w);
+    h[k] = static_cast<float>(ar * (sigmoid - ar) * w);
+  }
+}
+
+/*! device pointwise metric reduction: out[0] += Σ w·loss, out[1] += Σ w.
+ *  Loss formulas mirror cpp/src/metric.cpp; convert kinds are the objectives'
+ *  output transforms. Replaces the per-eval 8B×rows score download
+ *  (capability parity: reference src/metric/cuda/cuda_pointwise_metric.cu). */
+__global__ void k_metric_pointwise(int kind, double a, int convert_kind,
+                                   double convert_param,
+                                   const double* __restrict__ score,
+                                   const float* __restrict__ label,
+                                   const float* __restrict__ weight, int n,
+                                   double* __restrict__ out) {
+  double loss_sum = 0.0, w_sum = 0.0;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += gridDim.x * blockDim.x) {
+    const double w = weight ? weight[i] : 1.0;
+    const double y = label[i];
+    double p = score[i];
+    switch (convert_kind) {
+      case 1: p = exp(p); break;
+      case 2: p = 1.0 / (1.0 + exp(-convert_param * p)); break;
+      case 3: p = 1.0 / (1.0 + exp(-p)); break;
+      case 4: p = log1p(exp(p)); break;
+      case 5: p = p * fabs(p); break;
+      default: break;
+    }
+    double l = 0.0;
+    switch (kind) {
+      case 0: l = (y - p) * (y - p); break;                         // l2 / rmse / r2
+      case 1: l = fabs(y - p); break;                               // l1
+      case 2: { const double dd = y - p; l = dd >= 0 ? a * dd : (a - 1.0) * dd; break; }
+      case 3: { const double dd = fabs(y - p);                      // huber
+                l = dd <= a ? 0.5 * dd * dd : a * (dd - 0.5 * a); break; }
+      case 4: { const double x = fabs(y - p);                       // fair
+                l = a * x - a * a * log1p(x / a); break; }
+      case 5: { double pp = fmax(p, 1e-10); l = pp - y * log(pp); break; }  // poisson
+      case 6: l = fabs((y - p) / fmax(1.0, fabs(y))); break;        // mape
+      case 7: { double pp = fmax(p, 1e-10); l = y / pp + log(pp) - 1.0; break; }  // gamma
+      case 8: { double pp = fmax(p, 1e-10);                         // gamma_deviance
+                l = y <= 1e-10 ? 0.0 : 2.0 * (log(pp / y) + y / pp - 1.0); break; }
+      case 9: { double pp = fmax(p, 1e-10);                         // tweedie
+                l = -y * pow(pp, 1.0 - a) / (1.0 - a) + pow(pp, 2.0 - a) / (2.0 - a);
+                break; }
+      case 10: { double pp = fmin(1.0 - 1e-12, fmax(1e-12, p));     // binary_logloss
+                 l = y > 0 ? -log(pp) : -log(1.0 - pp); break; }
+      case 11: l = ((p > 0.5) != (y > 0)) ? 1.0 : 0.0; break;       // binary_error
+      case 12: { double pp = fmin(1.0 - 1e-12, fmax(1e-12, p));     // cross_entropy
+                 l = -y * log(pp) - (1.0 - y) * log(1.0 - pp); break; }
+      case 13: { const double hhat = log1p(fmax(1e-12, p));         // xentlambda
+                 l = y * hhat - p; break; }
+    }
+    loss_sum += w * l;
+    w_sum += w;
+  }
+  // wave64 reduce, then one atomic per wave
+  for (int off = 32; off > 0; off >>= 1) {
+    loss_sum += __shfl_down(loss_sum, off);
+    w_sum += __shfl_down(w_sum, off);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    atomicAdd(&out[0], loss_sum);
+    atomicAdd(&out[1], w_sum);
+  }
+}
+
+// ------------------------------------------------------ device percentile renew
+__device__ __forceinline__ unsigned long long d_ord64(double v) {
+  // order-preserving double -> uint64 mapping (sign-magnitude to biased)
+  unsigned long long b = static_cast<unsigned long long>(__double_as_longlong(v));
+  return (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
+}
+__device__ __forceinline__ double d_unord64(unsigned long long u) {
+  const unsigned long long b =
+      (u & 0x8000000000000000ull) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u;
+  return __longlong_as_double(static_cast<long long>(b));
+}
+
+/*! per-leaf α-percentile of residual (label - score) via 64-step bitwise binary
+ *  search with block-wide counting — no sort, no extra memory, exact order
+ *  statistics. One block per leaf. wmode: 0 unweighted (reference PercentileFun
+ *  interpolation), 1 metadata weights, 2 MAPE label weights (w/max(1,|label|)).
+ *  Replaces the host download+sort renewal for l1/quantile/mape (capability
+ *  parity: reference RenewTreeOutputCUDAKernel_RegressionL1/Quantile,
+ *  cuda_regression_objective.cu). */
+__global__ void k_renew_percentile(const uint32_t* __restrict__ idx,
+                                   const int* __restrict__ leaf_begin,
+                                   const int* __restrict__ leaf_cnt, int nl,
+                                   const double* __restrict__ score,
+                                   const float* __restrict__ label,
+                                   const float* __restrict__ weight, int wmode,
+                                   double alpha, double* __restrict__ out) {
+  const int l = blockIdx.x;
+  if (l >= nl) return;
+  const int begin = leaf_begin[l];
+  const int cnt = leaf_cnt[l];
+  if (cnt <= 0) return;  // keep the existing output
+  __shared__ double s_red[4];
+  __shared__ unsigned long long s_bound[2];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int nwave = blockDim.x >> 6;
+
+  auto residual = [&](int j) -> double {
+    const uint32_t row = idx[begin + j];
+    return static_cast<double>(label[row]) - score[row];
+  };
+  auto wgt = [&](int j) -> double {
+    const uint32_t row = idx[begin + j];
+    double w = weight ? weight[row] : 1.0;
+    if (wmode == 2) w /= fmax(1.0, fabs(static_cast<double>(label[row])));
+    return w;
+  };
+  // block-wide sum of a per-thread double
+  auto block_sum = [&](double v) -> double {
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off);
+    if ((tid & 63) == 0) s_red[wave] = v;
+    __syncthreads();
+    double t = 0;
+    for (int wv = 0; wv < nwave; ++wv) t += s_red[wv];
+    __syncthreads();
+    return t;
+  };
+
+  if (wmode == 0) {
+    // k-th order statistic: smallest ordinal m with count(res <= m) >= k+1
+    auto kth = [&](int k) -> double {
+      unsigned long long lo = 0ull, hi = ~0ull;
+      while (lo < hi) {
+        const unsigned long long mid = lo + ((hi - lo) >> 1);
+        double c = 0;
+        for (int j = tid; j < cnt; j += blockDim.x)
+          if (d_ord64(residual(j)) <= mid) c += 1.0;
+        const double total = block_sum(c);
+        if (tid == 0) s_bound[0] = total >= static_cast<double>(k + 1) ? 1ull : 0ull;
+        __syncthreads();
+        const bool le = s_bound[0] != 0ull;
+        __syncthreads();
+        if (le) hi = mid;
+        else lo = mid + 1ull;
+      }
+      return d_unord64(lo);
+    };
+    const double pos = alpha * (cnt - 1);
+    const int klo = static_cast<int>(pos);
+    const int khi = klo + 1 < cnt ? klo + 1 : cnt - 1;
+    const double frac = pos - klo;
+    const double vlo = kth(klo);
+    const double vhi = khi == klo ? vlo : kth(khi);
+    if (tid == 0) out[l] = vlo * (1.0 - frac) + vhi * frac;
+  } else {
+    double wloc = 0;
+    for (int j = tid; j < cnt; j += blockDim.x) wloc += wgt(j);
+    const double total_w = block_sum(wloc);
+    const double target = alpha * total_w;
+    unsigned long long lo = 0ull, hi = ~0ull;
+    while (lo < hi) {
+      const unsigned long long mid = lo + ((hi - lo) >> 1);
+      double c = 0;
+      for (int j = tid; j < cnt; j += blockDim.x)
+        if (d_ord64(residual(j)) <= mid) c += wgt(j);
+      const double cum = block_sum(c);
+      if (tid == 0) s_bound[0] = cum >= target ? 1ull : 0ull;
+      __syncthreads();
+      const bool le = s_bound[0] != 0ull;
+      __syncthreads();
+      if (le) hi = mid;
+      else lo = mid + 1ull;
+    }
+    if (tid == 0) out[l] = d_unord64(lo);
+  }
+}
+
 /*! lambdarank NDCG gradients: one block per query. Scores are argsorted in LDS
  *  (bitonic, padded to a power of two <= 1024); pairwise lambdas accumulate into
  *  LDS grad/hess; optional per-query normalization. Capability parity: the
@@ -1698,10 +2005,22 @@ class HIPTreeLearner : public TreeLearner {
 
   bool IsHIPLearner() const override { return true; }
   bool DeviceObjectiveSupported(const std::string& name) const override {
-    if (name == "binary" || name == "regression") return true;
+    static const char* kRegression[] = {"regression", "regression_l1", "huber", "fair",
+                                        "poisson",    "quantile",      "mape",  "gamma",
+                                        "tweedie"};
+    for (const char* r : kRegression) {
+      // reg_sqrt transforms labels host-side at objective Init; keep host path
+      if (name == r) return !config_->reg_sqrt;
+    }
+    if (name == "binary" || name == "cross_entropy" || name == "cross_entropy_lambda" ||
+        name == "multiclass" || name == "multiclassova")
+      return true;
     if (name == "lambdarank") return rank_ok_;
     return false;
   }
+  bool DeviceEvalPointwise(int loss_kind, double loss_a, int convert_kind,
+                           double convert_param, double* out_sum,
+                           double* out_wsum) override;
 
   void Init(const Dataset* train_data, bool is_constant_hessian) override;
   void ResetTrainingData(const Dataset* train_data) override {
@@ -1748,6 +2067,27 @@ class HIPTreeLearner : public TreeLearner {
   bool SetupOwnership(int world, int rank);
   /*! allgather per-rank winners + deterministic global argmax (ownership mode) */
   void SyncGlobalWinner();
+  /*! hipk::GradKind for a pointwise objective name, or -1 */
+  int PointwiseGradKind(const std::string& name) const {
+    if (name == "regression_l1") return hipk::kGradL1;
+    if (name == "huber") return hipk::kGradHuber;
+    if (name == "fair") return hipk::kGradFair;
+    if (name == "poisson") return hipk::kGradPoisson;
+    if (name == "quantile") return hipk::kGradQuantile;
+    if (name == "mape") return hipk::kGradMape;
+    if (name == "gamma") return hipk::kGradGamma;
+    if (name == "tweedie") return hipk::kGradTweedie;
+    if (name == "cross_entropy") return hipk::kGradXent;
+    if (name == "cross_entropy_lambda") return hipk::kGradXentLambda;
+    return -1;
+  }
+  double PointwiseGradParam(const std::string& name) const {
+    if (name == "huber" || name == "quantile") return config_->alpha;
+    if (name == "fair") return config_->fair_c;
+    if (name == "poisson") return config_->poisson_max_delta_step;
+    if (name == "tweedie") return config_->tweedie_variance_power;
+    return 0.0;
+  }
   int HistBlocksFor(int approx_cnt) const {
     // LDS atomic throughput is per-CU: spread even small leaves over many blocks
     // (~256 rows each); cap so the per-block flush stays amortized at the root.
@@ -1823,6 +2163,8 @@ class HIPTreeLearner : public TreeLearner {
   int num_class_score_ = 1;   // classes in the device score buffer
   int cur_class_ = 0;         // class selected by SetClassOffset
   double* ScorePtr() { return d_score_.ptr + static_cast<size_t>(cur_class_) * num_data_; }
+  float* GradPtr() { return d_grad_.ptr + static_cast<size_t>(cur_class_) * num_data_; }
+  float* HessPtr() { return d_hess_.ptr + static_cast<size_t>(cur_class_) * num_data_; }
   int quant_levels_ = 2;
   uint32_t quant_seed_ = 0x9E3779B9u;
   bool weights_present_ = false;
@@ -1842,6 +2184,7 @@ class HIPTreeLearner : public TreeLearner {
   int own_world_ = 0;                      // world the ownership plan was built for
   std::vector<size_t> own_off_, own_cnt_;  // per-rank hist float offsets/counts
   DevBuf<hipk::LogEntry> d_wire_my_, d_wire_all_;
+  DevBuf<double> d_eval_out_;  // [loss_sum, weight_sum] device metric reduction
 
   static constexpr int kHistBlock = 256;
   /*! hist-kernel workgroup size (k_hist is blockDim-agnostic; partition kernels
@@ -1987,8 +2330,11 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     }
   }
 
-  d_grad_.Alloc(num_data_);
-  d_hess_.Alloc(num_data_);
+  // device multiclass objectives fill all classes at once (class-major, stride
+  // num_data_); single-class objectives use only the first slice
+  num_class_score_ = std::max(1, config_->num_class);
+  d_grad_.Alloc(static_cast<size_t>(num_data_) * num_class_score_);
+  d_hess_.Alloc(static_cast<size_t>(num_data_) * num_class_score_);
   {
     int dev = 0, coop = 0;
     HIP_OK(hipGetDevice(&dev));
@@ -2005,7 +2351,6 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     d_grad_absmax_.Alloc(2);
     d_grad_scales_.Alloc(2);
   }
-  num_class_score_ = std::max(1, config_->num_class);
   d_score_.Alloc(static_cast<size_t>(num_data_) * num_class_score_);
   HIP_OK(hipMemset(d_score_.ptr, 0,
                    sizeof(double) * num_data_ * num_class_score_));
@@ -2157,9 +2502,9 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
 }
 
 void HIPTreeLearner::UploadGradients(const score_t* g, const score_t* h) {
-  HIP_OK(hipMemcpyAsync(d_grad_.ptr, g, sizeof(float) * num_data_, hipMemcpyHostToDevice,
+  HIP_OK(hipMemcpyAsync(GradPtr(), g, sizeof(float) * num_data_, hipMemcpyHostToDevice,
                         stream_));
-  HIP_OK(hipMemcpyAsync(d_hess_.ptr, h, sizeof(float) * num_data_, hipMemcpyHostToDevice,
+  HIP_OK(hipMemcpyAsync(HessPtr(), h, sizeof(float) * num_data_, hipMemcpyHostToDevice,
                         stream_));
 }
 
@@ -2184,6 +2529,18 @@ void HIPTreeLearner::DeviceBoosting(const ObjectiveFunction* obj) {
     hipLaunchKernelGGL(hipk::k_grad_l2, g, b, 0, stream_, d_score_.ptr, d_label_.ptr,
                        weights_present_ ? d_weight_.ptr : nullptr, n, d_grad_.ptr,
                        d_hess_.ptr);
+  } else if (name == "multiclass") {
+    hipLaunchKernelGGL(hipk::k_grad_multiclass, g, b, 0, stream_, d_score_.ptr,
+                       d_label_.ptr, weights_present_ ? d_weight_.ptr : nullptr, n,
+                       num_class_score_, d_grad_.ptr, d_hess_.ptr);
+  } else if (name == "multiclassova") {
+    hipLaunchKernelGGL(hipk::k_grad_multiclass_ova, g, b, 0, stream_, d_score_.ptr,
+                       d_label_.ptr, weights_present_ ? d_weight_.ptr : nullptr, n,
+                       num_class_score_, config_->sigmoid, d_grad_.ptr, d_hess_.ptr);
+  } else if (int kind = PointwiseGradKind(name); kind >= 0) {
+    hipLaunchKernelGGL(hipk::k_grad_pointwise, g, b, 0, stream_, kind, d_score_.ptr,
+                       d_label_.ptr, weights_present_ ? d_weight_.ptr : nullptr, n,
+                       PointwiseGradParam(name), d_grad_.ptr, d_hess_.ptr);
   } else if (name == "lambdarank") {
     hipLaunchKernelGGL(hipk::k_grad_lambdarank, dim3(num_queries_), dim3(256), 0, stream_,
                        d_score_.ptr, d_label_.ptr, d_qb_.ptr, num_queries_,
@@ -2248,7 +2605,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
         hipLaunchKernelGGL(hipk::k_hist<8>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
                            d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
                            slot_stride);
         break;
@@ -2256,7 +2613,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
         hipLaunchKernelGGL(hipk::k_hist<4>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
                            d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
                            slot_stride);
         break;
@@ -2264,7 +2621,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
         hipLaunchKernelGGL(hipk::k_hist<2>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
                            d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
                            slot_stride);
         break;
@@ -2272,7 +2629,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
         hipLaunchKernelGGL(hipk::k_hist<1>, dim3(blocks), dim3(HistThreads()), lds, stream_,
                            d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                            d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, d_grad_.ptr, d_hess_.ptr,
+                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
                            d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
                            slot_stride);
     }
@@ -2378,7 +2735,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   auto tree = std::make_unique<Tree>(nl);
 
   if (!grads_on_device_) UploadGradients(gradients, hessians);
-  grads_on_device_ = false;
+  if (cur_class_ >= num_class_score_ - 1) grads_on_device_ = false;
 
   feat_mask_host_.clear();
   if (config_->feature_fraction < 1.0) {
@@ -2401,11 +2758,11 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   if (quantized_) {
     HIP_OK(hipMemsetAsync(d_grad_absmax_.ptr, 0, 2 * sizeof(float), stream_));
     hipLaunchKernelGGL(hipk::k_grad_absmax, dim3(512), dim3(256), 0, stream_, d_idx_.ptr,
-                       static_cast<int>(used_cnt_), d_grad_.ptr, d_hess_.ptr,
+                       static_cast<int>(used_cnt_), GradPtr(), HessPtr(),
                        d_grad_absmax_.ptr);
     quant_seed_ = quant_seed_ * 1664525u + 1013904223u;
     hipLaunchKernelGGL(hipk::k_grad_quantize, dim3((num_data_ + 255) / 256), dim3(256), 0,
-                       stream_, d_grad_.ptr, d_hess_.ptr, num_data_, d_grad_absmax_.ptr,
+                       stream_, GradPtr(), HessPtr(), num_data_, d_grad_absmax_.ptr,
                        quant_levels_, config_->stochastic_rounding ? 1 : 0, quant_seed_,
                        d_grad_packed_.ptr, d_grad_scales_.ptr);
   }
@@ -2418,7 +2775,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   {
     const int blocks = std::min(2048, (static_cast<int>(used_cnt_) + 255) / 256);
     hipLaunchKernelGGL(hipk::k_root_sums, dim3(blocks), dim3(256), 0, stream_, d_idx_.ptr,
-                       static_cast<int>(used_cnt_), d_grad_.ptr, d_hess_.ptr,
+                       static_cast<int>(used_cnt_), GradPtr(), HessPtr(),
                        d_leaf_stats_.ptr);
   }
   auto& comm = GpuComm::Get();
@@ -2657,6 +3014,56 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
                                      const double* train_score) {
   if (obj == nullptr || !obj->NeedRenewTreeOutput()) return;
   (void)train_score;
+  const std::string name = obj->GetName();
+  const int nl = tree->num_leaves();
+  double alpha = 0.5;
+  int wmode = -1;
+  if (name == "regression_l1") wmode = weights_present_ ? 1 : 0;
+  else if (name == "quantile") { alpha = config_->alpha; wmode = weights_present_ ? 1 : 0; }
+  else if (name == "mape") wmode = 2;
+  if (wmode >= 0 && !config_->reg_sqrt) {
+    // device percentile renewal: no score download, no host sort
+    std::vector<double> outs(nl);
+    for (int l = 0; l < nl; ++l) outs[l] = tree->LeafOutput(l);
+    HIP_OK(hipMemcpyAsync(d_leaf_out_.ptr, outs.data(), sizeof(double) * nl,
+                          hipMemcpyHostToDevice, stream_));
+    hipLaunchKernelGGL(hipk::k_renew_percentile, dim3(nl), dim3(256), 0, stream_,
+                       d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, nl, ScorePtr(),
+                       d_label_.ptr, weights_present_ ? d_weight_.ptr : nullptr, wmode,
+                       alpha, d_leaf_out_.ptr);
+    HIP_OK(hipMemcpyAsync(outs.data(), d_leaf_out_.ptr, sizeof(double) * nl,
+                          hipMemcpyDeviceToHost, stream_));
+    HIP_OK(hipStreamSynchronize(stream_));
+    if (dist_ && GpuComm::Get().World() > 1) {
+      // ranks renewed from their local shard: sync to the count-weighted mean so
+      // every rank keeps the identical model (the reference's multi-GPU mode
+      // skips renewal entirely, nccl_gbdt.cpp:167-171 — this is strictly closer)
+      std::vector<double> acc(2 * nl);
+      for (int l = 0; l < nl; ++l) {
+        acc[2 * l] = outs[l] * leaf_cnt_[l];
+        acc[2 * l + 1] = leaf_cnt_[l];
+      }
+      if (GpuComm::Get().rccl()) {
+        DevBuf<double> scratch;
+        scratch.Alloc(2 * nl);
+        HIP_OK(hipMemcpyAsync(scratch.ptr, acc.data(), sizeof(double) * 2 * nl,
+                              hipMemcpyHostToDevice, stream_));
+        GpuComm::Get().AllReduce(scratch.ptr, 2 * nl, stream_);
+        HIP_OK(hipMemcpyAsync(acc.data(), scratch.ptr, sizeof(double) * 2 * nl,
+                              hipMemcpyDeviceToHost, stream_));
+        HIP_OK(hipStreamSynchronize(stream_));
+      } else {
+        Network::AllreduceSum(acc.data(), 2 * nl);
+      }
+      for (int l = 0; l < nl; ++l)
+        if (acc[2 * l + 1] > 0) outs[l] = acc[2 * l] / acc[2 * l + 1];
+    }
+    for (int l = 0; l < nl; ++l)
+      if (leaf_cnt_[l] > 0 || (dist_ && outs[l] != tree->LeafOutput(l)))
+        tree->SetLeafOutput(l, outs[l]);
+    return;
+  }
+  // host fallback (unknown renewing objective)
   std::vector<uint32_t> idx(used_cnt_);
   HIP_OK(hipStreamSynchronize(stream_));
   HIP_OK(hipMemcpy(idx.data(), d_idx_.ptr, sizeof(uint32_t) * used_cnt_,
@@ -2664,7 +3071,6 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
   std::vector<double> score(num_data_);
   HIP_OK(hipMemcpy(score.data(), ScorePtr(), sizeof(double) * num_data_,
                    hipMemcpyDeviceToHost));
-  const int nl = tree->num_leaves();
   for (int l = 0; l < nl; ++l) {
     if (leaf_cnt_[l] == 0) continue;
     std::vector<data_size_t> rows(leaf_cnt_[l]);
@@ -2673,6 +3079,26 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
     tree->SetLeafOutput(l, obj->RenewTreeOutput(tree->LeafOutput(l), rows.data(),
                                                 leaf_cnt_[l], score.data()));
   }
+}
+
+bool HIPTreeLearner::DeviceEvalPointwise(int loss_kind, double loss_a, int convert_kind,
+                                         double convert_param, double* out_sum,
+                                         double* out_wsum) {
+  if (loss_kind < 0 || num_class_score_ > 1) return false;
+  if (!d_eval_out_.ptr) d_eval_out_.Alloc(2);
+  HIP_OK(hipMemsetAsync(d_eval_out_.ptr, 0, 2 * sizeof(double), stream_));
+  const int blocks = std::min(1024, (num_data_ + 255) / 256);
+  hipLaunchKernelGGL(hipk::k_metric_pointwise, dim3(blocks), dim3(256), 0, stream_,
+                     loss_kind, loss_a, convert_kind, convert_param, ScorePtr(),
+                     d_label_.ptr, weights_present_ ? d_weight_.ptr : nullptr, num_data_,
+                     d_eval_out_.ptr);
+  double host_out[2];
+  HIP_OK(hipMemcpyAsync(host_out, d_eval_out_.ptr, 2 * sizeof(double),
+                        hipMemcpyDeviceToHost, stream_));
+  HIP_OK(hipStreamSynchronize(stream_));
+  *out_sum = host_out[0];
+  *out_wsum = host_out[1];
+  return true;
 }
 
 // ------------------------------------------------------------------ registration

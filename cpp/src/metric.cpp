@@ -61,12 +61,30 @@ class PointwiseMetric : public Metric {
     double r = final_ ? final_(gsum, gw) : gsum / std::max(1.0, gw);
     return {r};
   }
+  PointwiseEvalDesc pointwise_desc() const override { return desc_; }
+  PointwiseMetric* WithDesc(int kind, double a = 0.0) {
+    desc_.kind = kind;
+    desc_.a = a;
+    desc_.convert = convert_;
+    return this;
+  }
+  double FinalizeFromSums(double sum, double w) const override {
+    double gsum = sum, gw = w;
+    if (Network::is_distributed()) {
+      double buf[2] = {gsum, gw};
+      Network::AllreduceSum(buf, 2);
+      gsum = buf[0];
+      gw = buf[1];
+    }
+    return final_ ? final_(gsum, gw) : gsum / std::max(1.0, gw);
+  }
 
  protected:
   std::vector<std::string> names_;
   bool convert_;
   std::function<double(double, double)> loss_;
   std::function<double(double, double)> final_;
+  PointwiseEvalDesc desc_;
   data_size_t num_data_ = 0;
   const label_t* label_ = nullptr;
   const label_t* weights_ = nullptr;
@@ -188,6 +206,8 @@ class BinaryLoglossMetric : public Metric {
     }
     return {GlobalAvg(sum, sum_w_)};
   }
+  PointwiseEvalDesc pointwise_desc() const override { return {kPwBinaryLogloss, 0.0, true}; }
+  double FinalizeFromSums(double sum, double w) const override { return GlobalAvg(sum, w); }
 
  private:
   std::vector<std::string> names_;
@@ -480,70 +500,92 @@ class AucMuMetric : public Metric {
 
 Metric* Metric::Create(const std::string& name, const Config& cfg) {
   if (name == "l2" || name == "mse" || name == "regression" || name == "mean_squared_error")
-    return new PointwiseMetric("l2", true, [](double y, double p) { return (y - p) * (y - p); });
+    return (new PointwiseMetric("l2", true, [](double y, double p) { return (y - p) * (y - p); }))
+        ->WithDesc(kPwL2);
   if (name == "rmse" || name == "root_mean_squared_error" || name == "l2_root")
-    return new PointwiseMetric("rmse", true,
+    return (new PointwiseMetric("rmse", true,
                                [](double y, double p) { return (y - p) * (y - p); },
-                               [](double s, double w) { return std::sqrt(s / std::max(1.0, w)); });
+                               [](double s, double w) { return std::sqrt(s / std::max(1.0, w)); }))
+        ->WithDesc(kPwL2);
   if (name == "l1" || name == "mae" || name == "mean_absolute_error")
-    return new PointwiseMetric("l1", true, [](double y, double p) { return std::fabs(y - p); });
+    return (new PointwiseMetric("l1", true, [](double y, double p) { return std::fabs(y - p); }))
+        ->WithDesc(kPwL1);
   if (name == "quantile") {
     double a = cfg.alpha;
-    return new PointwiseMetric("quantile", true, [a](double y, double p) {
+    return (new PointwiseMetric("quantile", true, [a](double y, double p) {
       double d = y - p;
       return d >= 0 ? a * d : (a - 1) * d;
-    });
+    }))->WithDesc(kPwQuantile, a);
   }
   if (name == "huber") {
     double a = cfg.alpha;
-    return new PointwiseMetric("huber", true, [a](double y, double p) {
+    return (new PointwiseMetric("huber", true, [a](double y, double p) {
       double d = std::fabs(y - p);
       return d <= a ? 0.5 * d * d : a * (d - 0.5 * a);
-    });
+    }))->WithDesc(kPwHuber, a);
   }
   if (name == "fair") {
     double c = cfg.fair_c;
-    return new PointwiseMetric("fair", true, [c](double y, double p) {
+    return (new PointwiseMetric("fair", true, [c](double y, double p) {
       double x = std::fabs(y - p);
       return c * x - c * c * std::log1p(x / c);
-    });
+    }))->WithDesc(kPwFair, c);
   }
   if (name == "poisson")
-    return new PointwiseMetric("poisson", true, [](double y, double p) {
+    return (new PointwiseMetric("poisson", true, [](double y, double p) {
       double eps = 1e-10;
       if (p <= eps) p = eps;
       return p - y * std::log(p);
-    });
+    }))->WithDesc(kPwPoisson);
   if (name == "mape")
-    return new PointwiseMetric("mape", true, [](double y, double p) {
+    return (new PointwiseMetric("mape", true, [](double y, double p) {
       return std::fabs((y - p) / std::max(1.0, std::fabs(y)));
-    });
+    }))->WithDesc(kPwMape);
   if (name == "gamma")
-    return new PointwiseMetric("gamma", true, [](double y, double p) {
+    return (new PointwiseMetric("gamma", true, [](double y, double p) {
       double eps = 1e-10;
       if (p <= eps) p = eps;
       return y / p + std::log(p) - 1;  // negative log-likelihood up to const
-    });
+    }))->WithDesc(kPwGamma);
   if (name == "gamma_deviance")
-    return new PointwiseMetric("gamma_deviance", true, [](double y, double p) {
+    return (new PointwiseMetric("gamma_deviance", true, [](double y, double p) {
       double eps = 1e-10;
       if (p <= eps) p = eps;
       if (y <= eps) return 0.0;
       return 2.0 * (std::log(p / y) + y / p - 1);
-    });
+    }))->WithDesc(kPwGammaDev);
   if (name == "tweedie") {
     double rho = cfg.tweedie_variance_power;
-    return new PointwiseMetric("tweedie", true, [rho](double y, double p) {
+    return (new PointwiseMetric("tweedie", true, [rho](double y, double p) {
       double eps = 1e-10;
       if (p <= eps) p = eps;
       return -y * std::pow(p, 1 - rho) / (1 - rho) + std::pow(p, 2 - rho) / (2 - rho);
-    });
+    }))->WithDesc(kPwTweedie, rho);
   }
   if (name == "r2") {
     class R2Metric : public PointwiseMetric {
      public:
       R2Metric() : PointwiseMetric("r2", true,
-                                   [](double y, double p) { return (y - p) * (y - p); }) {}
+                                   [](double y, double p) { return (y - p) * (y - p); }) {
+        WithDesc(kPwL2);
+      }
+      double FinalizeFromSums(double sum, double w) const override {
+        const double mse = PointwiseMetric::FinalizeFromSums(sum, w);
+        double mean = 0, sw = 0;
+        for (data_size_t i = 0; i < num_data_; ++i) {
+          double wi = weights_ ? weights_[i] : 1.0;
+          mean += wi * label_[i];
+          sw += wi;
+        }
+        mean /= std::max(sw, 1.0);
+        double var = 0;
+        for (data_size_t i = 0; i < num_data_; ++i) {
+          double wi = weights_ ? weights_[i] : 1.0;
+          var += wi * (label_[i] - mean) * (label_[i] - mean);
+        }
+        var /= std::max(sw, 1.0);
+        return var > 0 ? 1.0 - mse / var : 0.0;
+      }
       double factor_to_bigger_better() const override { return -1.0; }
       std::vector<double> Eval(const double* score,
                                const ObjectiveFunction* obj) const override {
@@ -569,11 +611,12 @@ Metric* Metric::Create(const std::string& name, const Config& cfg) {
   if (name == "binary_logloss" || name == "logloss" || name == "binary")
     return new BinaryLoglossMetric();
   if (name == "regression_l1")
-    return new PointwiseMetric("l1", true, [](double y, double p) { return std::fabs(y - p); });
+    return (new PointwiseMetric("l1", true, [](double y, double p) { return std::fabs(y - p); }))
+        ->WithDesc(kPwL1);
   if (name == "binary_error")
-    return new PointwiseMetric("binary_error", true, [](double y, double p) {
+    return (new PointwiseMetric("binary_error", true, [](double y, double p) {
       return (p > 0.5 ? 1.0 : 0.0) != (y > 0 ? 1.0 : 0.0) ? 1.0 : 0.0;
-    });
+    }))->WithDesc(kPwBinaryError);
   if (name == "auc") return new AUCMetric();
   if (name == "average_precision") return new AveragePrecisionMetric();
   if (name == "multi_logloss" || name == "softmax" || name == "multiclass" ||
@@ -584,10 +627,10 @@ Metric* Metric::Create(const std::string& name, const Config& cfg) {
   if (name == "ndcg" || name == "lambdarank" || name == "rank_xendcg") return new NDCGMetric(cfg);
   if (name == "map" || name == "mean_average_precision") return new MapMetric(cfg);
   if (name == "cross_entropy" || name == "xentropy")
-    return new PointwiseMetric("cross_entropy", true, [](double y, double p) {
+    return (new PointwiseMetric("cross_entropy", true, [](double y, double p) {
       p = std::min(1.0 - 1e-12, std::max(1e-12, p));
       return -y * std::log(p) - (1 - y) * std::log(1 - p);
-    });
+    }))->WithDesc(kPwXent);
   if (name == "cross_entropy_lambda" || name == "xentlambda")
     return new PointwiseMetric("cross_entropy_lambda", true, [](double y, double p) {
       double hhat = std::log1p(std::max(1e-12, p));

@@ -884,11 +884,26 @@ void SerialTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj
   // generic renewal: per leaf, objective-specific output from the rows in the leaf
   const int nl = tree->num_leaves();
   data_size_t cnt;
+  std::vector<double> acc;  // distributed: (out*cnt, cnt) pairs for the global mean
+  if (Network::is_distributed()) acc.assign(2 * nl, 0.0);
   for (int l = 0; l < nl; ++l) {
     const data_size_t* idx = partition_.GetIndexOnLeaf(l, &cnt);
     if (cnt == 0) continue;
     double new_out = obj->RenewTreeOutput(tree->LeafOutput(l), idx, cnt, train_score);
-    tree->SetLeafOutput(l, new_out);
+    if (Network::is_distributed()) {
+      acc[2 * l] = new_out * cnt;
+      acc[2 * l + 1] = cnt;
+    } else {
+      tree->SetLeafOutput(l, new_out);
+    }
+  }
+  if (Network::is_distributed()) {
+    // each rank renewed from its local shard; sync to the count-weighted mean so
+    // every rank keeps the identical model (the reference leaves the outputs
+    // rank-local, silently diverging the machines' models)
+    Network::AllreduceSum(acc.data(), 2 * nl);
+    for (int l = 0; l < nl; ++l)
+      if (acc[2 * l + 1] > 0) tree->SetLeafOutput(l, acc[2 * l] / acc[2 * l + 1]);
   }
 }
 

@@ -165,3 +165,67 @@ assert max(vals) - min(vals) < 1e-12, vals''')
         capture_output=True, text=True, timeout=300, env=env)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "DIST_OK" in r.stdout
+
+
+RENEW_WORKER = r"""
+import hashlib
+import os, sys
+sys.path.insert(0, sys.argv[1])
+import numpy as np
+import torch.distributed as dist
+import datetime
+
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+dist.init_process_group("gloo", timeout=datetime.timedelta(seconds=120))
+rank, world = dist.get_rank(), dist.get_world_size()
+
+import lightgbm_amd as lgb
+from lightgbm_amd.parallel import init_network_from_torch_distributed
+init_network_from_torch_distributed()
+
+rng = np.random.RandomState(7)
+Xref = rng.randn(5000, 6)
+yref = (3 * Xref[:, 0] + np.sin(Xref[:, 1])).astype(np.float32)
+ref = lgb.Dataset(Xref, label=yref, params={"max_bin": 63}).construct()
+
+rng = np.random.RandomState(100 + rank)
+X = rng.randn(8000, 6)
+y = (3 * X[:, 0] + np.sin(X[:, 1]) + 0.1 * rng.randn(8000)).astype(np.float32)
+train = ref.create_valid(X, label=y)
+
+# l1 renews leaf outputs (median) — without the cross-rank sync the per-rank
+# models silently diverge
+params = {"objective": "regression_l1", "tree_learner": "data", "num_leaves": 31,
+          "verbosity": -1, "max_bin": 63}
+bst = lgb.train(params, train, num_boost_round=40)
+digest = hashlib.sha256(bst.model_to_string().encode()).hexdigest()
+payload = [None] * world
+dist.all_gather_object(payload, digest)
+assert len(set(payload)) == 1, f"rank models differ: {payload}"
+
+rng = np.random.RandomState(999)
+Xv = rng.randn(4000, 6)
+yv = 3 * Xv[:, 0] + np.sin(Xv[:, 1])
+mae = np.abs(bst.predict(Xv) - yv).mean()
+assert mae < 0.4, mae
+if rank == 0:
+    print("DIST_OK", mae)
+dist.destroy_process_group()
+"""
+
+
+def test_data_parallel_l1_renew_identical_models(tmp_path):
+    """Objectives with RenewTreeOutput (l1 median renewal) must produce identical
+    models on every rank: renewed outputs are synced to the count-weighted mean
+    across shards (the reference leaves them rank-local and diverges)."""
+    script = tmp_path / "worker_renew.py"
+    script.write_text(RENEW_WORKER)
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29551", str(script), str(REPO)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "DIST_OK" in r.stdout

@@ -361,3 +361,118 @@ def test_gpu_dist_world3_reduce_scatter(tmp_path):
     models, res = _run_dist_gpu(tmp_path, "reduce_scatter", world=3)
     assert models[0] == models[1] == models[2]
     assert res["auc_full"] > 0.80
+
+
+@pytest.mark.parametrize("objective,make_label,pred_tol", [
+    ("regression_l1", "reg", 0.05),
+    ("huber", "reg", 0.05),
+    ("fair", "reg", 0.05),
+    ("quantile", "reg", 0.08),
+    ("mape", "pos", 0.10),
+    ("poisson", "pos", 0.10),
+    ("gamma", "pos", 0.10),
+    ("tweedie", "pos", 0.10),
+    ("cross_entropy", "prob", 0.03),
+    ("cross_entropy_lambda", "prob", 0.05),
+])
+def test_gpu_device_objective_parity(objective, make_label, pred_tol):
+    """Every device gradient kernel must track the CPU oracle of the same
+    objective (device objective breadth). Compares predictions
+    on held-out rows between device_type=cpu and gpu at identical config."""
+    rng = np.random.RandomState(5)
+    n, d = 60_000, 10
+    X = rng.randn(n, d).astype(np.float32)
+    base = 1.5 * X[:, 0] + np.sin(X[:, 1]) + 0.5 * X[:, 2] * X[:, 3]
+    if make_label == "reg":
+        y = (base + 0.2 * rng.randn(n)).astype(np.float32)
+    elif make_label == "pos":
+        y = np.exp(0.4 * base + 0.1 * rng.randn(n)).astype(np.float32)
+    else:  # prob
+        y = (1.0 / (1.0 + np.exp(-base))).astype(np.float32)
+    preds = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": objective, "device_type": dev, "max_bin": 63,
+                  "num_leaves": 31, "min_data_in_leaf": 20, "verbosity": 0,
+                  "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 25)
+        preds[dev] = bst.predict(X[:10000])
+    scale = max(1e-3, float(np.abs(preds["cpu"]).mean()))
+    rel = np.abs(preds["cpu"] - preds["gpu"]).mean() / scale
+    assert rel < pred_tol, (objective, rel)
+
+
+def test_gpu_multiclass_device_objective():
+    """multiclass softmax gradients now run on device (all classes at once);
+    quality parity with the CPU learner."""
+    rng = np.random.RandomState(0)
+    n, d, k = 60_000, 8, 4
+    X = rng.randn(n, d).astype(np.float32)
+    logits = np.stack([X[:, i] + 0.5 * X[:, (i + 1) % d] for i in range(k)], axis=1)
+    y = np.argmax(logits + 0.5 * rng.randn(n, k), axis=1).astype(np.float32)
+    accs = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "multiclass", "num_class": k, "device_type": dev,
+                  "max_bin": 63, "num_leaves": 31, "verbosity": 0, "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 15)
+        p = bst.predict(X[:10000]).reshape(-1, k)
+        accs[dev] = float((np.argmax(p, axis=1) == y[:10000]).mean())
+    assert accs["gpu"] > 0.8, accs
+    assert abs(accs["cpu"] - accs["gpu"]) < 0.02, accs
+
+
+def test_gpu_device_metric_eval_matches_host():
+    """Train-metric eval must produce the same numbers through the device
+    pointwise reducer as through the host path (score download)."""
+    X, y = _binary_data(n=80_000)
+    params = {"objective": "binary", "device_type": "gpu", "max_bin": 63,
+              "num_leaves": 63, "verbosity": 0,
+              "metric": ["binary_logloss", "binary_error", "l2"]}
+    tr = lgb.Dataset(X, label=y)
+    ev = {}
+    bst = lgb.train(params, tr, 10, valid_sets=[tr], valid_names=["training"],
+                    callbacks=[lgb.record_evaluation(ev)])
+    # recompute each metric from downloaded predictions (raw scores -> sigmoid)
+    p = bst.predict(X)
+    ll = -(y * np.log(np.clip(p, 1e-12, None)) +
+           (1 - y) * np.log(np.clip(1 - p, 1e-12, None))).mean()
+    err = ((p > 0.5) != (y > 0)).mean()
+    l2 = ((y - p) ** 2).mean()
+    assert abs(ev["training"]["binary_logloss"][-1] - ll) < 1e-6
+    assert abs(ev["training"]["binary_error"][-1] - err) < 1e-9
+    assert abs(ev["training"]["l2"][-1] - l2) < 1e-6
+
+
+def test_gpu_l1_renew_parity():
+    """Device percentile renewal (binary-search order statistics) must match the
+    CPU median renewal: same leaf outputs -> near-identical predictions."""
+    rng = np.random.RandomState(11)
+    n = 50_000
+    X = rng.randn(n, 8).astype(np.float32)
+    y = (2 * X[:, 0] + np.abs(X[:, 1]) + 0.3 * rng.standard_cauchy(n)).astype(np.float32)
+    preds = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "regression_l1", "device_type": dev, "max_bin": 63,
+                  "num_leaves": 31, "min_data_in_leaf": 50, "verbosity": 0,
+                  "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 20)
+        preds[dev] = bst.predict(X[:10000])
+    rel = np.abs(preds["cpu"] - preds["gpu"]).mean() / max(0.3, np.abs(preds["cpu"]).mean())
+    assert rel < 0.05, rel
+
+
+def test_gpu_weighted_quantile_renew():
+    """Weighted percentile renewal path (weights present + quantile alpha)."""
+    rng = np.random.RandomState(3)
+    n = 40_000
+    X = rng.randn(n, 6).astype(np.float32)
+    y = (X[:, 0] * 2 + 0.5 * rng.randn(n)).astype(np.float32)
+    w = rng.uniform(0.5, 2.0, size=n).astype(np.float32)
+    preds = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "quantile", "alpha": 0.8, "device_type": dev,
+                  "max_bin": 63, "num_leaves": 31, "min_data_in_leaf": 50,
+                  "verbosity": 0, "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y, weight=w), 20)
+        preds[dev] = bst.predict(X[:10000])
+    rel = np.abs(preds["cpu"] - preds["gpu"]).mean() / max(0.3, np.abs(preds["cpu"]).mean())
+    assert rel < 0.08, rel
