@@ -2021,6 +2021,10 @@ class HIPTreeLearner : public TreeLearner {
   bool DeviceEvalPointwise(int loss_kind, double loss_a, int convert_kind,
                            double convert_param, double* out_sum,
                            double* out_wsum) override;
+  /*! kernel-level unit-test probe: run the root k_hist on device and compare
+   *  bin-for-bin against the fp64 host histogram oracle
+   *  (Dataset::ConstructHistograms). Returns the max relative error. */
+  double DebugRootHistMaxRelErr(const score_t* g, const score_t* h);
 
   void Init(const Dataset* train_data, bool is_constant_hessian) override;
   void ResetTrainingData(const Dataset* train_data) override {
@@ -3101,6 +3105,44 @@ bool HIPTreeLearner::DeviceEvalPointwise(int loss_kind, double loss_a, int conve
   return true;
 }
 
+double HIPTreeLearner::DebugRootHistMaxRelErr(const score_t* g, const score_t* h) {
+  cur_class_ = 0;
+  UploadGradients(g, h);
+  used_cnt_ = num_data_;
+  hipLaunchKernelGGL(hipk::k_iota, dim3((num_data_ + 255) / 256), dim3(256), 0, stream_,
+                     d_idx_.ptr, num_data_);
+  hipLaunchKernelGGL(hipk::k_init_root, dim3(1), dim3(1), 0, stream_, d_leaf_begin_.ptr,
+                     d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr,
+                     static_cast<int>(used_cnt_), d_gbuf_.ptr, d_counters_.ptr,
+                     d_root_leaf_.ptr, d_minus1_.ptr, d_leaf_bounds_.ptr,
+                     d_leaf_branch_.ptr);
+  {
+    const int blocks = std::min(2048, (static_cast<int>(used_cnt_) + 255) / 256);
+    hipLaunchKernelGGL(hipk::k_root_sums, dim3(blocks), dim3(256), 0, stream_, d_idx_.ptr,
+                       static_cast<int>(used_cnt_), GradPtr(), HessPtr(),
+                       d_leaf_stats_.ptr);
+  }
+  LaunchHist(d_root_leaf_.ptr, 0, HistBlocksFor(static_cast<int>(used_cnt_)));
+  HIP_OK(hipStreamSynchronize(stream_));
+  std::vector<float> dev(static_cast<size_t>(total_bins_) * 2);
+  HIP_OK(hipMemcpy(dev.data(), d_hist_.ptr, sizeof(float) * dev.size(),
+                   hipMemcpyDeviceToHost));
+  // fp64 host oracle over the identical rows/gradients
+  std::vector<double> host(dev.size(), 0.0);
+  std::vector<int8_t> used(train_data_->num_features(), 1);
+  std::vector<data_size_t> idx(num_data_);
+  for (int i = 0; i < num_data_; ++i) idx[i] = i;
+  train_data_->ConstructHistograms(used, idx.data(), num_data_, g, h, host.data());
+  double max_rel = 0.0;
+  for (size_t i = 0; i < dev.size(); ++i) {
+    const double hv = host[i];
+    const double dv = dev[i];
+    const double err = std::fabs(dv - hv) / std::max(1.0, std::fabs(hv));
+    max_rel = std::max(max_rel, err);
+  }
+  return max_rel;
+}
+
 // ------------------------------------------------------------------ registration
 namespace {
 TreeLearner* CreateHIP(const Config* cfg) {
@@ -3181,4 +3223,26 @@ extern "C" __attribute__((visibility("default"))) int LGBM_GPUNetworkFree() {
 
 extern "C" __attribute__((visibility("default"))) int LGBM_GPUSetDevice(int device) {
   return hipSetDevice(device) == hipSuccess ? 0 : -1;
+}
+
+/*! kernel-level unit-test hook (tests/test_gpu.py): device root histogram vs the
+ *  fp64 host oracle, bin for bin. Not part of the public LGBM_* surface. */
+extern "C" __attribute__((visibility("default"))) int MIGBM_DebugDeviceRootHist(
+    void* dataset_handle, const char* params, const float* grad, const float* hess,
+    double* out_max_rel_err) {
+  try {
+    auto* ds = static_cast<migbm::Dataset*>(dataset_handle);
+    auto cfgmap = migbm::Config::Str2Map(params);
+    migbm::Config cfg;
+    cfg.Set(cfgmap);
+    migbm::HIPTreeLearner learner(&cfg);
+    learner.Init(ds, false);
+    *out_max_rel_err = learner.DebugRootHistMaxRelErr(grad, hess);
+    return 0;
+  } catch (const std::exception& ex) {
+    fprintf(stderr, "MIGBM_DebugDeviceRootHist: %s\n", ex.what());
+    return -1;
+  } catch (...) {
+    return -1;
+  }
 }
