@@ -145,7 +145,15 @@ class Random {
 /*! Phase timer (capability: reference Common::Timer / FunctionTimer). */
 class Timer {
  public:
-  static Timer& Global() { static Timer t; return t; }
+  static Timer& Global() {
+    static Timer t;
+    static bool reg = [] {
+      if (Enabled()) atexit([] { Timer::Global().Print(); });
+      return true;
+    }();
+    (void)reg;
+    return t;
+  }
   /*! runtime-enabled: set MIGBM_TIMETAG=1; zero cost when off beyond one branch */
   static bool Enabled() {
     static bool e = getenv("MIGBM_TIMETAG") != nullptr;

@@ -203,6 +203,10 @@ class SerialTreeLearner : public TreeLearner {
   int hist_mode_ = -1;
   double hist_trial_time_[2] = {0.0, 0.0};
   int hist_trials_done_ = 0;
+  // per-tree hessian classification (1 = constant -> count mode) and the
+  // interleaved (g,h) pair array the varying-hessian row-wise loop reads
+  int tree_const_hess_ = -1;
+  std::vector<score_t> gh_;
   bool warned_mc_method_ = false;
   // LOCAL (pre-reduce) leaf gradient totals of the last ComputeHistogram call;
   // distributed learners use them to materialize default bins before reducing

@@ -285,8 +285,31 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
                                      ? kRowWiseMinRows
                                      : train_data_->num_total_bin();
     if (cnt >= min_rows) {
-      train_data_->ConstructHistogramsRowWise(indices, cnt, gradients_, hessians_, hist,
-                                              /*row_indexed=*/true);
+      if (tree_const_hess_ < 0) {
+        // first row-wise call of this tree is the root: classify the hessians
+        // once (constant -> count mode) and, if varying, build the interleaved
+        // (g,h) pair array so the hot loop does ONE 8B load per row
+        const score_t h0 = hessians_[indices[0]];
+        bool ok = true;
+#pragma omp parallel for schedule(static) reduction(&& : ok)
+        for (data_size_t i = 0; i < cnt; ++i) ok = ok && hessians_[indices[i]] == h0;
+        tree_const_hess_ = ok ? 1 : 0;
+        if (!ok) {
+          gh_.resize(2 * static_cast<size_t>(train_data_->num_data()));
+#pragma omp parallel for schedule(static)
+          for (data_size_t i = 0; i < cnt; ++i) {
+            const data_size_t r = indices[i];
+            gh_[2 * static_cast<size_t>(r)] = gradients_[r];
+            gh_[2 * static_cast<size_t>(r) + 1] = hessians_[r];
+          }
+        }
+      }
+      if (tree_const_hess_ == 1) {
+        train_data_->ConstructHistogramsRowWise(indices, cnt, gradients_, hessians_, hist,
+                                                /*row_indexed=*/true);
+      } else {
+        train_data_->ConstructHistogramsRowWiseGH(indices, cnt, gh_.data(), hist);
+      }
       return;
     }
   }
@@ -509,6 +532,7 @@ std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& 
 
 Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians,
                                bool /*is_first_tree*/) {
+  tree_const_hess_ = -1;  // re-classify hessians for this tree's gradients
   if (!config_->monotone_constraints.empty() &&
       config_->monotone_constraints_method != "basic" && !warned_mc_method_) {
     Log::Warning("monotone_constraints_method=%s is not implemented; using the basic "
@@ -869,10 +893,23 @@ void SerialTreeLearner::CalculateLinear(Tree* tree) {
 }
 
 void SerialTreeLearner::AddPredictionToScore(const Tree* tree, double* out_score) {
-  if (bag_indices_ != nullptr) {
-    tree->AddPredictionToScore(train_data_, bag_indices_, bag_cnt_, out_score);
-  } else {
-    tree->AddPredictionToScore(train_data_, train_data_->num_data(), out_score);
+  if (tree->is_linear()) {
+    // linear leaves read raw feature values per row: use the tree-walk path
+    if (bag_indices_ != nullptr)
+      tree->AddPredictionToScore(train_data_, bag_indices_, bag_cnt_, out_score);
+    else
+      tree->AddPredictionToScore(train_data_, train_data_->num_data(), out_score);
+    return;
+  }
+  // the partition already maps every trained row to its leaf: O(n) scatter-add
+  // instead of a per-row tree walk (reference ScoreUpdater::AddScore(tree_learner))
+  const int nl = tree->num_leaves();
+#pragma omp parallel for schedule(dynamic, 1)
+  for (int l = 0; l < nl; ++l) {
+    data_size_t cnt;
+    const data_size_t* idx = partition_.GetIndexOnLeaf(l, &cnt);
+    const double out = tree->LeafOutput(l);
+    for (data_size_t i = 0; i < cnt; ++i) out_score[idx[i]] += out;
   }
 }
 
