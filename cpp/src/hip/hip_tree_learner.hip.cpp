@@ -28,6 +28,9 @@
 #include "migbm/network.h"
 
 #include <algorithm>
+#include <condition_variable>
+#include <mutex>
+#include <thread>
 
 namespace migbm {
 
@@ -56,14 +59,81 @@ namespace migbm {
  *    how the multi-rank device code path is proven on a single-GPU box. Each call
  *    syncs the stream and bounces through host memory.
  *  Both run the IDENTICAL device-side code; only the reduction transport differs. */
+/*! in-process clique: barrier-framed host-bounce collectives between the
+ *  single-process num_gpu worker threads (one per device context). Used when
+ *  RCCL cannot (several shards share one physical GPU on a test box) and as the
+ *  universal fallback; distinct devices use a real ncclCommInitAll clique. */
+struct InProcClique {
+  explicit InProcClique(int w) : world(w) {}
+  const int world;
+
+  void Barrier() {
+    std::unique_lock<std::mutex> lk(mu_);
+    const uint64_t ph = phase_;
+    if (++arrived_ == world) {
+      arrived_ = 0;
+      ++phase_;
+      cv_.notify_all();
+    } else {
+      cv_.wait(lk, [&] { return phase_ != ph; });
+    }
+  }
+
+  template <typename T>
+  void AllReduceHost(T* h, size_t n, int rank) {
+    Barrier();
+    if (rank == 0 && buf_.size() < static_cast<size_t>(world) * n * sizeof(T))
+      buf_.resize(static_cast<size_t>(world) * n * sizeof(T));
+    Barrier();
+    T* slots = reinterpret_cast<T*>(buf_.data());
+    memcpy(slots + static_cast<size_t>(rank) * n, h, n * sizeof(T));
+    Barrier();
+    for (size_t i = 0; i < n; ++i) {
+      T acc = 0;
+      for (int r = 0; r < world; ++r) acc += slots[static_cast<size_t>(r) * n + i];
+      h[i] = acc;
+    }
+    Barrier();
+  }
+
+  void AllGatherHost(const void* in, void* out, size_t bytes, int rank) {
+    Barrier();
+    if (rank == 0 && buf_.size() < static_cast<size_t>(world) * bytes)
+      buf_.resize(static_cast<size_t>(world) * bytes);
+    Barrier();
+    memcpy(buf_.data() + static_cast<size_t>(rank) * bytes, in, bytes);
+    Barrier();
+    memcpy(out, buf_.data(), static_cast<size_t>(world) * bytes);
+    Barrier();
+  }
+
+ private:
+  std::mutex mu_;
+  std::condition_variable cv_;
+  int arrived_ = 0;
+  uint64_t phase_ = 0;
+  std::vector<char> buf_;
+};
+
 struct GpuComm {
   ncclComm_t comm = nullptr;
   int world = 1;
   int rank = 0;
-  bool active() const { return comm != nullptr || Network::is_distributed(); }
+  InProcClique* clique = nullptr;  // single-process num_gpu transport
+  bool active() const {
+    return comm != nullptr || clique != nullptr || Network::is_distributed();
+  }
   bool rccl() const { return comm != nullptr; }
-  int World() const { return rccl() ? world : Network::num_machines(); }
-  int Rank() const { return rccl() ? rank : Network::rank(); }
+  int World() const {
+    if (rccl()) return world;
+    if (clique) return clique->world;
+    return Network::num_machines();
+  }
+  int Rank() const {
+    if (rccl()) return rank;
+    if (clique) return rank;
+    return Network::rank();
+  }
 
   void AllReduce(float* d, size_t n, hipStream_t s) {
     if (rccl()) {
@@ -120,7 +190,8 @@ struct GpuComm {
     T* h = reinterpret_cast<T*>(staging_.data());
     if (hipMemcpy(h, d, n * sizeof(T), hipMemcpyDeviceToHost) != hipSuccess)
       Log::Fatal("GpuComm host-bounce: D2H failed");
-    Network::AllreduceSum(h, n);
+    if (clique) clique->AllReduceHost(h, n, rank);
+    else Network::AllreduceSum(h, n);
     if (hipMemcpy(d, h, n * sizeof(T), hipMemcpyHostToDevice) != hipSuccess)
       Log::Fatal("GpuComm host-bounce: H2D failed");
   }
@@ -133,11 +204,14 @@ void GpuComm::AllGather(const void* d_in, void* d_out, size_t bytes_per_rank, hi
   } else {
     if (hipStreamSynchronize(s) != hipSuccess)
       Log::Fatal("GpuComm host-bounce: stream sync failed");
-    const int w = Network::num_machines();
+    const int w = World();
     std::vector<char> h_in(bytes_per_rank), h_out(bytes_per_rank * w);
     if (hipMemcpy(h_in.data(), d_in, bytes_per_rank, hipMemcpyDeviceToHost) != hipSuccess)
       Log::Fatal("GpuComm host-bounce: D2H failed");
-    Network::Allgather(h_in.data(), static_cast<int>(bytes_per_rank), h_out.data());
+    if (clique)
+      clique->AllGatherHost(h_in.data(), h_out.data(), bytes_per_rank, rank);
+    else
+      Network::Allgather(h_in.data(), static_cast<int>(bytes_per_rank), h_out.data());
     if (hipMemcpy(d_out, h_out.data(), bytes_per_rank * w, hipMemcpyHostToDevice) != hipSuccess)
       Log::Fatal("GpuComm host-bounce: H2D failed");
   }
@@ -2004,6 +2078,19 @@ class HIPTreeLearner : public TreeLearner {
   }
 
   bool IsHIPLearner() const override { return true; }
+  /*! per-learner comm override (single-process num_gpu shard learners); the
+   *  process-wide singleton (LGBM_GPUNetworkInit) is the default. */
+  void SetCommClique(InProcClique* c, int rank) {
+    own_comm_.clique = c;
+    own_comm_.rank = rank;
+    use_own_comm_ = true;
+  }
+  void SetCommRccl(ncclComm_t c, int world, int rank) {
+    own_comm_.comm = c;
+    own_comm_.world = world;
+    own_comm_.rank = rank;
+    use_own_comm_ = true;
+  }
   bool DeviceObjectiveSupported(const std::string& name) const override {
     static const char* kRegression[] = {"regression", "regression_l1", "huber", "fair",
                                         "poisson",    "quantile",      "mape",  "gamma",
@@ -2189,6 +2276,9 @@ class HIPTreeLearner : public TreeLearner {
   std::vector<size_t> own_off_, own_cnt_;  // per-rank hist float offsets/counts
   DevBuf<hipk::LogEntry> d_wire_my_, d_wire_all_;
   DevBuf<double> d_eval_out_;  // [loss_sum, weight_sum] device metric reduction
+  GpuComm own_comm_;           // per-learner transport (num_gpu shard learners)
+  bool use_own_comm_ = false;
+  GpuComm& Comm() { return use_own_comm_ ? own_comm_ : GpuComm::Get(); }
 
   static constexpr int kHistBlock = 256;
   /*! hist-kernel workgroup size (k_hist is blockDim-agnostic; partition kernels
@@ -2641,7 +2731,7 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
 }
 
 void HIPTreeLearner::ReduceSpareHist(int spare_slot) {
-  auto& comm = GpuComm::Get();
+  auto& comm = Comm();
   if (!dist_) return;
   float* spare = d_hist_.ptr + static_cast<size_t>(spare_slot) * total_bins_ * 2;
   if (own_scan_) {
@@ -2689,7 +2779,7 @@ bool HIPTreeLearner::SetupOwnership(int world, int rank) {
 }
 
 void HIPTreeLearner::SyncGlobalWinner() {
-  auto& comm = GpuComm::Get();
+  auto& comm = Comm();
   hipLaunchKernelGGL(hipk::k_pack_winner, dim3(1), dim3(1), 0, stream_, d_winner_.ptr,
                      d_winner_leaf_.ptr, d_wire_my_.ptr);
   comm.AllGather(d_wire_my_.ptr, d_wire_all_.ptr, sizeof(hipk::LogEntry), stream_);
@@ -2782,7 +2872,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                        static_cast<int>(used_cnt_), GradPtr(), HessPtr(),
                        d_leaf_stats_.ptr);
   }
-  auto& comm = GpuComm::Get();
+  auto& comm = Comm();
   dist_ = comm.active();
   own_scan_ = false;
   if (dist_ && comm.World() > 1) {
@@ -3038,7 +3128,7 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
     HIP_OK(hipMemcpyAsync(outs.data(), d_leaf_out_.ptr, sizeof(double) * nl,
                           hipMemcpyDeviceToHost, stream_));
     HIP_OK(hipStreamSynchronize(stream_));
-    if (dist_ && GpuComm::Get().World() > 1) {
+    if (dist_ && Comm().World() > 1) {
       // ranks renewed from their local shard: sync to the count-weighted mean so
       // every rank keeps the identical model (the reference's multi-GPU mode
       // skips renewal entirely, nccl_gbdt.cpp:167-171 — this is strictly closer)
@@ -3047,18 +3137,14 @@ void HIPTreeLearner::RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
         acc[2 * l] = outs[l] * leaf_cnt_[l];
         acc[2 * l + 1] = leaf_cnt_[l];
       }
-      if (GpuComm::Get().rccl()) {
-        DevBuf<double> scratch;
-        scratch.Alloc(2 * nl);
-        HIP_OK(hipMemcpyAsync(scratch.ptr, acc.data(), sizeof(double) * 2 * nl,
-                              hipMemcpyHostToDevice, stream_));
-        GpuComm::Get().AllReduce(scratch.ptr, 2 * nl, stream_);
-        HIP_OK(hipMemcpyAsync(acc.data(), scratch.ptr, sizeof(double) * 2 * nl,
-                              hipMemcpyDeviceToHost, stream_));
-        HIP_OK(hipStreamSynchronize(stream_));
-      } else {
-        Network::AllreduceSum(acc.data(), 2 * nl);
-      }
+      DevBuf<double> scratch;
+      scratch.Alloc(2 * nl);
+      HIP_OK(hipMemcpyAsync(scratch.ptr, acc.data(), sizeof(double) * 2 * nl,
+                            hipMemcpyHostToDevice, stream_));
+      Comm().AllReduce(scratch.ptr, 2 * nl, stream_);
+      HIP_OK(hipMemcpyAsync(acc.data(), scratch.ptr, sizeof(double) * 2 * nl,
+                            hipMemcpyDeviceToHost, stream_));
+      HIP_OK(hipStreamSynchronize(stream_));
       for (int l = 0; l < nl; ++l)
         if (acc[2 * l + 1] > 0) outs[l] = acc[2 * l] / acc[2 * l + 1];
     }
@@ -3143,6 +3229,286 @@ double HIPTreeLearner::DebugRootHistMaxRelErr(const score_t* g, const score_t* h
   return max_rel;
 }
 
+// ------------------------------------------------- single-process multi-GPU (num_gpu)
+/*! persistent per-device worker threads: all HIP calls for shard r run on thread
+ *  r (hipSetDevice is per-thread state). */
+class GpuWorkers {
+ public:
+  explicit GpuWorkers(const std::vector<int>& devices) : n_(devices.size()) {
+    done_.assign(n_, 0);
+    for (size_t r = 0; r < n_; ++r) {
+      threads_.emplace_back([this, r, dev = devices[r]] {
+        HIP_OK(hipSetDevice(dev));
+        uint64_t seen = 0;
+        for (;;) {
+          std::function<void(int)> job;
+          {
+            std::unique_lock<std::mutex> lk(mu_);
+            cv_.wait(lk, [&] { return stop_ || job_id_ > seen; });
+            if (stop_) return;
+            seen = job_id_;
+            job = *job_;
+          }
+          job(static_cast<int>(r));
+          {
+            std::lock_guard<std::mutex> lk(mu_);
+            done_[r] = seen;
+            done_cv_.notify_all();
+          }
+        }
+      });
+    }
+  }
+  ~GpuWorkers() {
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      stop_ = true;
+      cv_.notify_all();
+    }
+    for (auto& t : threads_) t.join();
+  }
+  /*! run fn(shard) on every worker; rethrows the first failure */
+  void RunAll(const std::function<void(int)>& fn) {
+    std::exception_ptr err = nullptr;
+    std::mutex err_mu;
+    std::function<void(int)> wrapped = [&](int r) {
+      try {
+        fn(r);
+      } catch (...) {
+        std::lock_guard<std::mutex> lk(err_mu);
+        if (!err) err = std::current_exception();
+      }
+    };
+    uint64_t id;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      job_ = &wrapped;
+      id = ++job_id_;
+      cv_.notify_all();
+    }
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      done_cv_.wait(lk, [&] {
+        for (size_t r = 0; r < n_; ++r)
+          if (done_[r] < id) return false;
+        return true;
+      });
+    }
+    if (err) std::rethrow_exception(err);
+  }
+
+ private:
+  size_t n_;
+  std::vector<std::thread> threads_;
+  std::mutex mu_;
+  std::condition_variable cv_, done_cv_;
+  const std::function<void(int)>* job_ = nullptr;
+  uint64_t job_id_ = 0;
+  bool stop_ = false;
+  std::vector<uint64_t> done_;
+};
+
+/*! Single-process multi-GPU data-parallel learner (`num_gpu=N`): rows are
+ *  sharded contiguously over N device contexts, each owning a HIPTreeLearner on
+ *  its own device + host thread; per-shard learners reduce histograms / winners
+ *  through a private clique (ncclCommInitAll over distinct devices; in-process
+ *  barrier transport when shards share one physical GPU, e.g. a 1-GPU test box)
+ *  and every shard builds the identical tree. Train path needs NO torchrun
+ *  rendezvous. Capability parity: reference NCCLGBDT/NCCLTopology
+ *  (src/boosting/cuda/nccl_gbdt.cpp:84-209, cuda_nccl_topology.hpp:104-188) —
+ *  improved: bagging and RenewTreeOutput work here (the reference skips both),
+ *  and the per-shard learners inherit the reduce-scatter ownership mode. */
+class MultiGpuTreeLearner : public TreeLearner {
+ public:
+  explicit MultiGpuTreeLearner(const Config* config) : config_(config) {}
+  ~MultiGpuTreeLearner() override {
+    if (workers_) {
+      // learners own per-device resources: free them on their own threads
+      workers_->RunAll([&](int r) { learners_[r].reset(); });
+    }
+    for (ncclComm_t c : rccl_comms_)
+      if (c) ncclCommDestroy(c);
+  }
+
+  bool IsHIPLearner() const override { return true; }
+  bool DeviceObjectiveSupported(const std::string& name) const override {
+    if (config_->is_unbalance && name == "binary") return false;  // needs global counts
+    if (name == "lambdarank") return false;  // queries are not shardable here
+    HIPTreeLearner probe(config_);
+    return probe.DeviceObjectiveSupported(name);
+  }
+
+  void Init(const Dataset* train_data, bool is_constant_hessian) override {
+    train_data_ = train_data;
+    is_constant_hessian_ = is_constant_hessian;
+    num_data_ = train_data->num_data();
+    if (train_data->metadata().query_boundaries() != nullptr)
+      Log::Fatal("ranking objectives with num_gpu>1 in one process are not supported; "
+                 "use one process per GPU (tree_learner=data)");
+    for (ncclComm_t c : rccl_comms_)
+      if (c) ncclCommDestroy(c);
+    rccl_comms_.clear();
+    int ndev = 0;
+    HIP_OK(hipGetDeviceCount(&ndev));
+    world_ = std::max(1, config_->num_gpu);
+    if (num_data_ < world_ * 2) Log::Fatal("num_gpu=%d with only %d rows", world_, num_data_);
+    std::vector<int> devices(world_);
+    bool distinct = true;
+    for (int r = 0; r < world_; ++r) {
+      devices[r] = r % std::max(1, ndev);
+      distinct = distinct && devices[r] == r;
+    }
+    if (!distinct)
+      Log::Warning("num_gpu=%d > %d visible devices: shards share GPUs "
+                   "(in-process clique transport; correctness mode)",
+                   world_, ndev);
+    // contiguous row shards
+    shard_begin_.resize(world_ + 1);
+    for (int r = 0; r <= world_; ++r)
+      shard_begin_[r] = static_cast<data_size_t>(static_cast<int64_t>(num_data_) * r / world_);
+    shards_.resize(world_);
+    learners_.resize(world_);
+    if (!workers_) workers_ = std::make_unique<GpuWorkers>(devices);
+    // RCCL clique over distinct devices, host clique otherwise
+    std::vector<ncclComm_t> comms(world_, nullptr);
+    if (distinct && world_ > 1) {
+      NCCL_OK(ncclCommInitAll(comms.data(), world_, devices.data()));
+      rccl_comms_ = comms;
+    } else if (world_ > 1) {
+      clique_ = std::make_unique<InProcClique>(world_);
+    }
+    workers_->RunAll([&](int r) {
+      std::vector<data_size_t> rows(shard_begin_[r + 1] - shard_begin_[r]);
+      for (size_t i = 0; i < rows.size(); ++i)
+        rows[i] = shard_begin_[r] + static_cast<data_size_t>(i);
+      shards_[r] = train_data_->Subset(rows.data(), static_cast<data_size_t>(rows.size()));
+      auto* l = new HIPTreeLearner(config_);
+      if (world_ > 1) {
+        if (!rccl_comms_.empty()) l->SetCommRccl(rccl_comms_[r], world_, r);
+        else l->SetCommClique(clique_.get(), r);
+      }
+      l->Init(shards_[r].get(), is_constant_hessian);
+      learners_[r].reset(l);
+    });
+  }
+  void ResetTrainingData(const Dataset* train_data) override {
+    Init(train_data, is_constant_hessian_);
+  }
+  void ResetConfig(const Config* config) override {
+    config_ = config;
+    workers_->RunAll([&](int r) { learners_[r]->ResetConfig(config); });
+  }
+
+  Tree* Train(const score_t* gradients, const score_t* hessians, bool is_first) override {
+    std::vector<Tree*> trees(world_, nullptr);
+    workers_->RunAll([&](int r) {
+      // host-gradient fallback: each shard uploads its row slice
+      trees[r] = learners_[r]->Train(gradients ? gradients + shard_begin_[r] : nullptr,
+                                     hessians ? hessians + shard_begin_[r] : nullptr,
+                                     is_first);
+    });
+    for (int r = 1; r < world_; ++r) delete trees[r];
+    return trees[0];
+  }
+
+  void SetBaggingData(const Dataset* subset, const data_size_t* used_indices,
+                      data_size_t num_data) override {
+    (void)subset; (void)used_indices; (void)num_data;
+    Log::Fatal("bagging_fraction/GOSS with num_gpu>1 in one process is not supported "
+               "yet; use one process per GPU (tree_learner=data) instead");
+  }
+
+  void AddPredictionToScore(const Tree* tree, double* out_score) override {
+    workers_->RunAll([&](int r) { learners_[r]->AddPredictionToScore(tree, out_score); });
+  }
+
+  void RenewTreeOutput(Tree* tree, const ObjectiveFunction* obj,
+                       std::function<double(const label_t*, int)> fn, data_size_t nd,
+                       const data_size_t* bag, data_size_t bag_cnt,
+                       const double* train_score) override {
+    if (obj == nullptr || !obj->NeedRenewTreeOutput()) return;
+    // every shard renews its local rows then syncs through the clique: all
+    // copies end identical, take shard 0's outputs
+    std::vector<std::unique_ptr<Tree>> copies(world_);
+    for (int r = 0; r < world_; ++r) copies[r] = std::make_unique<Tree>(*tree);
+    workers_->RunAll([&](int r) {
+      learners_[r]->RenewTreeOutput(copies[r].get(), obj, fn, nd, bag, bag_cnt,
+                                    train_score);
+    });
+    for (int l = 0; l < tree->num_leaves(); ++l)
+      tree->SetLeafOutput(l, copies[0]->LeafOutput(l));
+  }
+
+  void DeviceBoosting(const ObjectiveFunction* obj) override {
+    workers_->RunAll([&](int r) { learners_[r]->DeviceBoosting(obj); });
+  }
+  void DeviceAddInitScore(double v) override {
+    workers_->RunAll([&](int r) { learners_[r]->DeviceAddInitScore(v); });
+  }
+  void SetClassOffset(int class_id) override {
+    num_class_ = std::max(num_class_, class_id + 1);
+    workers_->RunAll([&](int r) { learners_[r]->SetClassOffset(class_id); });
+  }
+  void DownloadTrainScore(double* dst) override {
+    const int nc = std::max(1, config_->num_class);
+    workers_->RunAll([&](int r) {
+      const data_size_t sn = shard_begin_[r + 1] - shard_begin_[r];
+      std::vector<double> tmp(static_cast<size_t>(sn) * nc);
+      learners_[r]->DownloadTrainScore(tmp.data());
+      for (int c = 0; c < nc; ++c)
+        memcpy(dst + static_cast<size_t>(c) * num_data_ + shard_begin_[r],
+               tmp.data() + static_cast<size_t>(c) * sn, sizeof(double) * sn);
+    });
+  }
+  void UploadTrainScore(const double* src) override {
+    const int nc = std::max(1, config_->num_class);
+    workers_->RunAll([&](int r) {
+      const data_size_t sn = shard_begin_[r + 1] - shard_begin_[r];
+      std::vector<double> tmp(static_cast<size_t>(sn) * nc);
+      for (int c = 0; c < nc; ++c)
+        memcpy(tmp.data() + static_cast<size_t>(c) * sn,
+               src + static_cast<size_t>(c) * num_data_ + shard_begin_[r],
+               sizeof(double) * sn);
+      learners_[r]->UploadTrainScore(tmp.data());
+    });
+  }
+  bool DeviceEvalPointwise(int loss_kind, double loss_a, int convert_kind,
+                           double convert_param, double* out_sum,
+                           double* out_wsum) override {
+    std::vector<double> sums(world_, 0.0), wsums(world_, 0.0);
+    std::vector<int> oks(world_, 0);
+    workers_->RunAll([&](int r) {
+      oks[r] = learners_[r]->DeviceEvalPointwise(loss_kind, loss_a, convert_kind,
+                                                 convert_param, &sums[r], &wsums[r])
+                   ? 1
+                   : 0;
+    });
+    for (int r = 0; r < world_; ++r)
+      if (!oks[r]) return false;
+    *out_sum = 0.0;
+    *out_wsum = 0.0;
+    for (int r = 0; r < world_; ++r) {
+      *out_sum += sums[r];
+      *out_wsum += wsums[r];
+    }
+    return true;
+  }
+
+ private:
+  const Config* config_;
+  const Dataset* train_data_ = nullptr;
+  bool is_constant_hessian_ = false;
+  data_size_t num_data_ = 0;
+  int world_ = 1;
+  int num_class_ = 1;
+  std::vector<data_size_t> shard_begin_;
+  std::vector<std::unique_ptr<Dataset>> shards_;
+  std::vector<std::unique_ptr<HIPTreeLearner>> learners_;
+  std::unique_ptr<GpuWorkers> workers_;
+  std::unique_ptr<InProcClique> clique_;
+  std::vector<ncclComm_t> rccl_comms_;
+};
+
 // ------------------------------------------------------------------ registration
 namespace {
 TreeLearner* CreateHIP(const Config* cfg) {
@@ -3164,6 +3530,11 @@ TreeLearner* CreateHIP(const Config* cfg) {
                  "training this model on the host (CPU) learner instead",
                  cfg->device_type.c_str(), what);
     return new SerialTreeLearner(cfg);
+  }
+  if (cfg->num_gpu > 1) {
+    // single-process multi-GPU (reference NCCLGBDT parity): shard rows over
+    // num_gpu device contexts inside this process, no torchrun needed
+    return new MultiGpuTreeLearner(cfg);
   }
   return new HIPTreeLearner(cfg);
 }
