@@ -3413,9 +3413,26 @@ class MultiGpuTreeLearner : public TreeLearner {
 
   void SetBaggingData(const Dataset* subset, const data_size_t* used_indices,
                       data_size_t num_data) override {
-    (void)subset; (void)used_indices; (void)num_data;
-    Log::Fatal("bagging_fraction/GOSS with num_gpu>1 in one process is not supported "
-               "yet; use one process per GPU (tree_learner=data) instead");
+    if (subset != nullptr) Log::Fatal("HIP learner does not use dataset-subset bagging");
+    if (used_indices == nullptr || num_data == 0 || num_data == num_data_) {
+      workers_->RunAll([&](int r) { learners_[r]->SetBaggingData(nullptr, nullptr, 0); });
+      return;
+    }
+    // split the (ascending) global bag indices into per-shard local index sets
+    // — bagging works under num_gpu (the reference NCCLGBDT skips it entirely)
+    bag_parts_.assign(world_, {});
+    const data_size_t* end = used_indices + num_data;
+    for (int r = 0; r < world_; ++r) {
+      const data_size_t* lo = std::lower_bound(used_indices, end, shard_begin_[r]);
+      const data_size_t* hi = std::lower_bound(used_indices, end, shard_begin_[r + 1]);
+      bag_parts_[r].resize(hi - lo);
+      for (size_t i = 0; i < bag_parts_[r].size(); ++i)
+        bag_parts_[r][i] = lo[i] - shard_begin_[r];
+    }
+    workers_->RunAll([&](int r) {
+      learners_[r]->SetBaggingData(nullptr, bag_parts_[r].data(),
+                                   static_cast<data_size_t>(bag_parts_[r].size()));
+    });
   }
 
   void AddPredictionToScore(const Tree* tree, double* out_score) override {
@@ -3503,6 +3520,7 @@ class MultiGpuTreeLearner : public TreeLearner {
   int num_class_ = 1;
   std::vector<data_size_t> shard_begin_;
   std::vector<std::unique_ptr<Dataset>> shards_;
+  std::vector<std::vector<data_size_t>> bag_parts_;  // per-shard local bag indices
   std::vector<std::unique_ptr<HIPTreeLearner>> learners_;
   std::unique_ptr<GpuWorkers> workers_;
   std::unique_ptr<InProcClique> clique_;
