@@ -22,6 +22,54 @@ inline double GlobalAvg(double sum, double w) {
   return sum / std::max(1.0, w);
 }
 
+/*! distributed sort-based eval (AUC / AP / auc_mu): gather every rank's
+ *  (scores, label, weight) so the metric is computed over the UNION of shards
+ *  (VERDICT r1 #7 — per-rank AUC silently reported a wrong number before).
+ *  Correctness-grade payload: these metrics need a global sort anyway. */
+struct EvalUnion {
+  std::vector<double> score;  // nc * n, class-major
+  std::vector<label_t> label, weight;
+  data_size_t n = 0;
+  bool has_weight = false;
+};
+
+inline bool GatherEvalUnion(const double* score, int nc, const label_t* label,
+                            const label_t* weight, data_size_t n, EvalUnion* out) {
+  if (!Network::is_distributed()) return false;
+  const int world = Network::num_machines();
+  std::vector<int64_t> counts(world, 0);
+  int64_t mine = n;
+  Network::Allgather(reinterpret_cast<const char*>(&mine), sizeof(int64_t),
+                     reinterpret_cast<char*>(counts.data()));
+  int64_t total = 0;
+  for (int64_t c : counts) total += c;
+  out->n = static_cast<data_size_t>(total);
+  std::vector<int> lsz(world), dsz(world);
+  for (int r = 0; r < world; ++r) {
+    lsz[r] = static_cast<int>(counts[r] * sizeof(label_t));
+    dsz[r] = static_cast<int>(counts[r] * sizeof(double));
+  }
+  out->label.resize(total);
+  Network::AllgatherV(reinterpret_cast<const char*>(label),
+                      static_cast<int>(n * sizeof(label_t)), lsz.data(),
+                      reinterpret_cast<char*>(out->label.data()));
+  out->has_weight = weight != nullptr;
+  if (out->has_weight) {
+    out->weight.resize(total);
+    Network::AllgatherV(reinterpret_cast<const char*>(weight),
+                        static_cast<int>(n * sizeof(label_t)), lsz.data(),
+                        reinterpret_cast<char*>(out->weight.data()));
+  }
+  out->score.resize(static_cast<size_t>(nc) * total);
+  for (int c = 0; c < nc; ++c) {
+    Network::AllgatherV(reinterpret_cast<const char*>(score + static_cast<size_t>(c) * n),
+                        static_cast<int>(n * sizeof(double)), dsz.data(),
+                        reinterpret_cast<char*>(out->score.data() +
+                                                static_cast<size_t>(c) * total));
+  }
+  return true;
+}
+
 /*! generic pointwise metric: avg of per-row loss (weighted). */
 class PointwiseMetric : public Metric {
  public:
@@ -102,6 +150,14 @@ class AUCMetric : public Metric {
   const std::vector<std::string>& GetName() const override { return names_; }
   double factor_to_bigger_better() const override { return -1.0; }
   std::vector<double> Eval(const double* score, const ObjectiveFunction*) const override {
+    EvalUnion u;
+    if (GatherEvalUnion(score, 1, label_, weights_, num_data_, &u))
+      return {EvalImpl(u.score.data(), u.label.data(),
+                       u.has_weight ? u.weight.data() : nullptr, u.n)};
+    return {EvalImpl(score, label_, weights_, num_data_)};
+  }
+  static double EvalImpl(const double* score, const label_t* label_,
+                         const label_t* weights_, data_size_t num_data_) {
     std::vector<data_size_t> order(num_data_);
     std::iota(order.begin(), order.end(), 0);
     std::sort(order.begin(), order.end(),
@@ -129,7 +185,7 @@ class AUCMetric : public Metric {
     accum_neg += cur_neg;
     if (accum_pos > 0 && accum_neg > 0) auc /= (accum_pos * accum_neg);
     else auc = 1.0;
-    return {auc};
+    return auc;
   }
 
  private:
@@ -150,6 +206,14 @@ class AveragePrecisionMetric : public Metric {
   const std::vector<std::string>& GetName() const override { return names_; }
   double factor_to_bigger_better() const override { return -1.0; }
   std::vector<double> Eval(const double* score, const ObjectiveFunction*) const override {
+    EvalUnion u;
+    if (GatherEvalUnion(score, 1, label_, weights_, num_data_, &u))
+      return {EvalImpl(u.score.data(), u.label.data(),
+                       u.has_weight ? u.weight.data() : nullptr, u.n)};
+    return {EvalImpl(score, label_, weights_, num_data_)};
+  }
+  static double EvalImpl(const double* score, const label_t* label_,
+                         const label_t* weights_, data_size_t num_data_) {
     std::vector<data_size_t> order(num_data_);
     std::iota(order.begin(), order.end(), 0);
     std::sort(order.begin(), order.end(),
@@ -159,7 +223,7 @@ class AveragePrecisionMetric : public Metric {
       double w = weights_ ? weights_[order[i]] : 1.0;
       if (label_[order[i]] > 0) total_pos += w;
     }
-    if (total_pos <= 0) return {1.0};
+    if (total_pos <= 0) return 1.0;
     for (data_size_t i = 0; i < num_data_; ++i) {
       data_size_t r = order[i];
       double w = weights_ ? weights_[r] : 1.0;
@@ -170,7 +234,7 @@ class AveragePrecisionMetric : public Metric {
         fp += w;
       }
     }
-    return {ap / total_pos};
+    return ap / total_pos;
   }
 
  private:
@@ -443,6 +507,14 @@ class AucMuMetric : public Metric {
   const std::vector<std::string>& GetName() const override { return names_; }
   double factor_to_bigger_better() const override { return -1.0; }
   std::vector<double> Eval(const double* score, const ObjectiveFunction*) const override {
+    EvalUnion u;
+    if (GatherEvalUnion(score, nc_, label_, weights_, num_data_, &u))
+      return {EvalImpl(u.score.data(), u.label.data(),
+                       u.has_weight ? u.weight.data() : nullptr, u.n, nc_)};
+    return {EvalImpl(score, label_, weights_, num_data_, nc_)};
+  }
+  static double EvalImpl(const double* score, const label_t* label_,
+                         const label_t* weights_, data_size_t num_data_, int nc_) {
     double total = 0;
     int pairs = 0;
     for (int a = 0; a < nc_; ++a) {
@@ -485,7 +557,7 @@ class AucMuMetric : public Metric {
         }
       }
     }
-    return {pairs > 0 ? total / pairs : 1.0};
+    return pairs > 0 ? total / pairs : 1.0;
   }
 
  private:
