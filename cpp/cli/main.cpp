@@ -125,7 +125,8 @@ class Application {
     double t0 = Timer::Now();
     boosting_->Train(config_.snapshot_freq, config_.output_model);
     Log::Info("Finished training in %.3f seconds", Timer::Now() - t0);
-    boosting_->SaveModelToFile(0, -1, 0, config_.output_model.c_str());
+    boosting_->SaveModelToFile(0, -1, config_.saved_feature_importance_type,
+                               config_.output_model.c_str());
     Log::Info("Model saved to %s", config_.output_model.c_str());
   }
 
@@ -199,7 +200,8 @@ class Application {
         leaf_preds[static_cast<size_t>(i) * ntrees + t] = static_cast<int32_t>(out[t]);
     }
     boosting_->RefitTree(leaf_preds.data(), n, ntrees);
-    boosting_->SaveModelToFile(0, -1, 0, config_.output_model.c_str());
+    boosting_->SaveModelToFile(0, -1, config_.saved_feature_importance_type,
+                               config_.output_model.c_str());
     Log::Info("Refitted model saved to %s", config_.output_model.c_str());
   }
 

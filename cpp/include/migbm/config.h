@@ -145,6 +145,7 @@ struct Config {
   double tweedie_variance_power = 1.5;
   int lambdarank_truncation_level = 30;
   bool lambdarank_norm = true;
+  double lambdarank_position_bias_regularization = 0.0;
   std::vector<double> label_gain;
   int objective_seed = 5;
 
@@ -168,6 +169,14 @@ struct Config {
   int gpu_device_id = -1;
   bool gpu_use_dp = false;
   int num_gpu = 1;
+  std::string gpu_device_id_list;  // comma-separated device ids for num_gpu>1
+  // model io / loader / prediction knobs (reference parameter_set parity)
+  int saved_feature_importance_type = 0;  // 0 split, 1 gain (CLI model save)
+  bool precise_float_parser = false;      // our parser is always strtod-precise
+  std::string parser_config_file;
+  bool pred_early_stop = false;
+  int pred_early_stop_freq = 10;
+  double pred_early_stop_margin = 10.0;
 
   // raw key->value as given by user (post-alias-resolution), echoed into model file
   std::map<std::string, std::string> raw;

@@ -216,6 +216,14 @@ class SerialTreeLearner : public TreeLearner {
   DataPartition partition_;
   std::vector<hist_t> hist_store_;          // num_leaves slots x 2*num_total_bin
   std::vector<int> leaf_to_slot_;
+  // histogram_pool_size enforcement (reference HistogramPool): when the cap is
+  // below one-slot-per-leaf, slots are LRU-shared; an evicted parent histogram
+  // forces the larger child to be recomputed instead of subtracted
+  int pool_slots_ = 0;
+  std::vector<int> slot_owner_;     // slot -> leaf (-1 free)
+  std::vector<int64_t> slot_used_;  // LRU stamps
+  int64_t slot_clock_ = 0;
+  int AcquireSlot(int leaf, int pin_a = -1, int pin_b = -1);
   std::vector<SplitInfo> best_split_per_leaf_;
   std::vector<LeafContext> leaf_ctx_;
   std::vector<int8_t> is_feature_used_;     // per-tree mask

@@ -287,6 +287,15 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
     else if (k == "gpu_device_id") gpu_device_id = ParseInt(v);
     else if (k == "gpu_use_dp") gpu_use_dp = ParseBool(v);
     else if (k == "num_gpu") num_gpu = ParseInt(v);
+    else if (k == "gpu_device_id_list") gpu_device_id_list = v;
+    else if (k == "saved_feature_importance_type") saved_feature_importance_type = ParseInt(v);
+    else if (k == "precise_float_parser") precise_float_parser = ParseBool(v);
+    else if (k == "parser_config_file") parser_config_file = v;
+    else if (k == "pred_early_stop") pred_early_stop = ParseBool(v);
+    else if (k == "pred_early_stop_freq") pred_early_stop_freq = ParseInt(v);
+    else if (k == "pred_early_stop_margin") pred_early_stop_margin = ParseDouble(v);
+    else if (k == "lambdarank_position_bias_regularization")
+      lambdarank_position_bias_regularization = ParseDouble(v);
     // unknown keys are kept in raw only (tolerated, like the reference's pass-through)
   }
 

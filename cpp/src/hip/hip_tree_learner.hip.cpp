@@ -3354,9 +3354,30 @@ class MultiGpuTreeLearner : public TreeLearner {
     if (num_data_ < world_ * 2) Log::Fatal("num_gpu=%d with only %d rows", world_, num_data_);
     std::vector<int> devices(world_);
     bool distinct = true;
-    for (int r = 0; r < world_; ++r) {
-      devices[r] = r % std::max(1, ndev);
-      distinct = distinct && devices[r] == r;
+    if (!config_->gpu_device_id_list.empty()) {
+      // explicit device selection: "gpu_device_id_list=0,2,4,6"
+      std::vector<int> ids;
+      std::string tok;
+      for (char ch : config_->gpu_device_id_list + ",") {
+        if (ch == ',') {
+          if (!tok.empty()) ids.push_back(atoi(tok.c_str()));
+          tok.clear();
+        } else {
+          tok += ch;
+        }
+      }
+      if (static_cast<int>(ids.size()) != world_)
+        Log::Fatal("gpu_device_id_list has %d entries but num_gpu=%d",
+                   static_cast<int>(ids.size()), world_);
+      for (int r = 0; r < world_; ++r) {
+        devices[r] = ids[r] % std::max(1, ndev);
+        for (int q = 0; q < r; ++q) distinct = distinct && devices[q] != devices[r];
+      }
+    } else {
+      for (int r = 0; r < world_; ++r) {
+        devices[r] = r % std::max(1, ndev);
+        distinct = distinct && devices[r] == r;
+      }
     }
     if (!distinct)
       Log::Warning("num_gpu=%d > %d visible devices: shards share GPUs "

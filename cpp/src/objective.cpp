@@ -569,6 +569,7 @@ class LambdarankNDCG : public ObjectiveFunction {
  public:
   explicit LambdarankNDCG(const Config& cfg)
       : sigmoid_(cfg.sigmoid), norm_(cfg.lambdarank_norm),
+        bias_reg_(cfg.lambdarank_position_bias_regularization),
         truncation_level_(cfg.lambdarank_truncation_level), label_gain_(cfg.label_gain) {
     if (label_gain_.empty()) {
       // default 2^i - 1
@@ -628,7 +629,11 @@ class LambdarankNDCG : public ObjectiveFunction {
     if (base <= 0) return;
     for (size_t k = 0; k < position_bias_.size(); ++k) {
       if (bias_den_[k] > 0) {
-        const double est = (bias_num_[k] / bias_den_[k]) / base;
+        double est = (bias_num_[k] / bias_den_[k]) / base;
+        // L2 regularization pulls the propensity estimate toward 1 (analogue of
+        // the reference's Newton-step regularizer, rank_objective.hpp:333-334)
+        if (bias_reg_ > 0.0)
+          est = 1.0 + (est - 1.0) / (1.0 + bias_reg_ * bias_den_[k]);
         // smoothed multiplicative update, clamped for stability
         position_bias_[k] = std::min(10.0, std::max(0.1, 0.9 * position_bias_[k] +
                                                               0.1 * est));
@@ -735,6 +740,7 @@ class LambdarankNDCG : public ObjectiveFunction {
   data_size_t num_queries_ = 0;
   double sigmoid_;
   bool norm_;
+  double bias_reg_ = 0.0;
   int truncation_level_;
   std::vector<double> label_gain_;
   std::vector<double> inverse_max_dcg_;

