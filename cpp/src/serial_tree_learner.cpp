@@ -335,10 +335,13 @@ void SerialTreeLearner::ComputeHistogram(int leaf, data_size_t cnt,
         for (data_size_t i = 0; i < cnt; ++i) ok = ok && hessians_[indices[i]] == h0;
         tree_const_hess_ = ok ? 1 : 0;
         if (!ok) {
-          gh_.resize(2 * static_cast<size_t>(train_data_->num_data()));
+          // interleave over ALL rows: the first row-wise call of a tree may be
+          // a leaf subset (histogram trial phase), and later leaves read gh_ at
+          // their own row ids
+          const data_size_t n_all = train_data_->num_data();
+          gh_.resize(2 * static_cast<size_t>(n_all));
 #pragma omp parallel for schedule(static)
-          for (data_size_t i = 0; i < cnt; ++i) {
-            const data_size_t r = indices[i];
+          for (data_size_t r = 0; r < n_all; ++r) {
             gh_[2 * static_cast<size_t>(r)] = gradients_[r];
             gh_[2 * static_cast<size_t>(r) + 1] = hessians_[r];
           }
