@@ -8,6 +8,8 @@
 #include "migbm/metric.h"
 #include "migbm/objective.h"
 
+#include "migbm/network.h"
+
 #include <cstdio>
 #include <fstream>
 #include <memory>
@@ -51,6 +53,29 @@ class Application {
   }
 
   void Run() {
+    // distributed CPU training over the standalone TCP mesh (reference
+    // application.cpp Network::Init parity): machines= or machine_list_filename
+    if (config_.num_machines > 1 &&
+        (config_.tree_learner == "data" || config_.tree_learner == "feature" ||
+         config_.tree_learner == "voting")) {
+      std::string machines = config_.machines;
+      if (machines.empty() && !config_.machine_list_filename.empty()) {
+        std::ifstream mf(config_.machine_list_filename);
+        if (!mf.good())
+          Log::Fatal("Cannot open machine list %s", config_.machine_list_filename.c_str());
+        std::string line;
+        while (std::getline(mf, line)) {
+          line = Common::Trim(line);
+          if (line.empty()) continue;
+          if (!machines.empty()) machines += ",";
+          machines += line;
+        }
+      }
+      if (machines.empty())
+        Log::Fatal("num_machines>1 requires machines= or machine_list_filename=");
+      NetworkInitSockets(machines, config_.local_listen_port, config_.time_out,
+                         config_.num_machines);
+    }
     if (config_.task == "train") Train();
     else if (config_.task == "refit" || config_.task == "refit_tree") Refit();
     else if (config_.task == "predict" || config_.task == "prediction" ||
@@ -68,6 +93,10 @@ class Application {
     DatasetLoader loader(config_);
     if (Dataset::IsBinFile(config_.data.c_str())) {
       train_data_ = Dataset::LoadFromBinFile(config_.data.c_str());
+    } else if (Network::is_distributed() && !config_.pre_partition) {
+      // round-robin row sharding by rank (each machine reads the shared file)
+      train_data_ = loader.LoadFromFile(config_.data.c_str(), Network::rank(),
+                                        Network::num_machines());
     } else {
       train_data_ = loader.LoadFromFile(config_.data.c_str());
     }

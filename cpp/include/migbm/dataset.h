@@ -176,6 +176,10 @@ class Dataset {
    *  \param categorical per-original-column flag */
   void ConstructFromMat(const std::function<double(data_size_t, int)>& get, data_size_t nrow,
                         int ncol, const Config& cfg, const std::vector<int8_t>& categorical);
+  /*! single-rank body (no distributed bin-mapper sync) */
+  void ConstructFromMatLocal(const std::function<double(data_size_t, int)>& get,
+                             data_size_t nrow, int ncol, const Config& cfg,
+                             const std::vector<int8_t>& categorical);
 
   /*! Build an aligned valid set re-using this (train) dataset's bin mappers. */
   std::unique_ptr<Dataset> CreateValid(const std::function<double(data_size_t, int)>& get,

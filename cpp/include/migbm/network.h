@@ -47,6 +47,12 @@ class Network {
   static AllgatherFn allgather_;
 };
 
+/*! standalone TCP collectives (socket_linker.cpp): full mesh + ring allgather,
+ *  no external runtime required. machines = "ip:port,ip:port,...". */
+void NetworkInitSockets(const std::string& machines, int local_listen_port,
+                        int timeout_sec, int num_machines);
+void NetworkFreeSockets();
+
 }  // namespace migbm
 
 #endif  // MIGBM_NETWORK_H_

@@ -1566,19 +1566,18 @@ void AllgatherBridge(const char* input, int input_size, char* output) {
 }
 }  // namespace
 
-int LGBM_NetworkInit(const char* /*machines*/, int /*local_listen_port*/,
-                     int /*listen_time_out*/, int /*num_machines*/) {
+int LGBM_NetworkInit(const char* machines, int local_listen_port, int listen_time_out,
+                     int num_machines) {
   API_BEGIN();
-  Log::Fatal(
-      "migbm does not ship the reference's raw TCP socket mesh; distributed CPU training "
-      "uses LGBM_NetworkInitWithFunctions (the Python package wires torch.distributed/gloo "
-      "collectives), and multi-GPU training uses RCCL over xGMI inside the HIP learner.");
+  NetworkInitSockets(machines ? machines : "", local_listen_port, listen_time_out,
+                     num_machines);
   API_END();
 }
 
 int LGBM_NetworkFree() {
   API_BEGIN();
   Network::Free();
+  NetworkFreeSockets();
   g_allgather_ext = nullptr;
   API_END();
 }

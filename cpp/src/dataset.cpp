@@ -3,6 +3,7 @@
  *  Parity target: reference src/io/dataset.cpp (ConstructHistogramsInner, SaveBinaryFile),
  *  src/io/metadata.cpp. Fresh implementation. */
 #include "migbm/dataset.h"
+#include "migbm/network.h"
 
 #include <cstdio>
 #include <numeric>
@@ -71,6 +72,52 @@ void Dataset::FinishBinMappers(const Config&) {
 void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& get,
                                data_size_t nrow, int ncol, const Config& cfg,
                                const std::vector<int8_t>& categorical) {
+  if (Network::is_distributed()) {
+    // distributed in-memory construction: every rank must end with IDENTICAL
+    // bin mappers. Gather a per-rank row sample into a union sample matrix
+    // (same bytes everywhere), find bins on it, then bin the local shard
+    // (reference ConstructBinMappersFromTextData sample-sync parity).
+    const int world = Network::num_machines();
+    const data_size_t cap =
+        std::max<data_size_t>(1, cfg.bin_construct_sample_cnt / std::max(1, world));
+    const data_size_t s = std::min(nrow, cap);
+    std::vector<double> flat(static_cast<size_t>(s) * ncol);
+    const double stride = s > 0 ? static_cast<double>(nrow) / s : 1.0;
+    for (data_size_t i = 0; i < s; ++i) {
+      const data_size_t r =
+          std::min<data_size_t>(nrow - 1, static_cast<data_size_t>(i * stride));
+      for (int c = 0; c < ncol; ++c) flat[static_cast<size_t>(i) * ncol + c] = get(r, c);
+    }
+    std::vector<int64_t> counts(world, 0);
+    int64_t mine = s;
+    Network::Allgather(reinterpret_cast<const char*>(&mine), sizeof(int64_t),
+                       reinterpret_cast<char*>(counts.data()));
+    int64_t total_s = 0;
+    std::vector<int> sizes(world);
+    for (int r = 0; r < world; ++r) {
+      total_s += counts[r];
+      sizes[r] = static_cast<int>(counts[r] * ncol * sizeof(double));
+    }
+    std::vector<double> uni(static_cast<size_t>(total_s) * ncol);
+    Network::AllgatherV(reinterpret_cast<const char*>(flat.data()),
+                        static_cast<int>(flat.size() * sizeof(double)), sizes.data(),
+                        reinterpret_cast<char*>(uni.data()));
+    Dataset holder;
+    auto uni_at = [&uni, ncol](data_size_t r, int c) {
+      return uni[static_cast<size_t>(r) * ncol + c];
+    };
+    holder.ConstructFromMatLocal(uni_at, static_cast<data_size_t>(total_s), ncol, cfg,
+                                 categorical);
+    auto shard = holder.CreateValid(get, nrow);
+    *this = std::move(*shard);
+    return;
+  }
+  ConstructFromMatLocal(get, nrow, ncol, cfg, categorical);
+}
+
+void Dataset::ConstructFromMatLocal(const std::function<double(data_size_t, int)>& get,
+                                    data_size_t nrow, int ncol, const Config& cfg,
+                                    const std::vector<int8_t>& categorical) {
   num_data_ = nrow;
   num_total_features_ = ncol;
   categorical_flags_ = categorical;

@@ -2,6 +2,7 @@
  *  distributed row sharding. Parity target: reference src/io/dataset_loader.cpp +
  *  parser.cpp (autodetect), metadata.cpp (.weight/.query sidecar files). */
 #include "migbm/dataset.h"
+#include "migbm/network.h"
 
 #include <cstdio>
 #include <cstring>

@@ -274,3 +274,55 @@ int main(){double r[3],o[1];
                          check=True, timeout=300)
     gen_pred = np.array([float(t) for t in res.stdout.split()])
     np.testing.assert_allclose(gen_pred, bst.predict(X[:100]), rtol=1e-10)
+
+
+def test_cli_distributed_socket_mesh(tmp_path):
+    """Distributed CPU training over the standalone TCP socket mesh — NO torch,
+    no injected collectives: N CLI processes, localhost machine list (reference
+    tests/distributed/_test_distributed.py mockup). Models must be identical
+    across workers and predict well (VERDICT r1 #9)."""
+    import socket
+    from concurrent.futures import ThreadPoolExecutor
+
+    def free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    n_workers = 2
+    ports = [free_port() for _ in range(n_workers)]
+    machines = ",".join(f"127.0.0.1:{p}" for p in ports)
+
+    rng = np.random.RandomState(3)
+    n = 6000
+    X = rng.randn(n, 6)
+    y = (X[:, 0] + 0.5 * X[:, 1] + 0.2 * rng.randn(n) > 0).astype(int)
+    train = tmp_path / "train.tsv"
+    np.savetxt(train, np.column_stack([y, X]), delimiter="\t", fmt="%.8g")
+
+    def run_worker(i):
+        model = tmp_path / f"model{i}.txt"
+        r = subprocess.run(
+            [str(CLI), "task=train", f"data={train}", "objective=binary",
+             "tree_learner=data", f"num_machines={n_workers}",
+             f"machines={machines}", f"local_listen_port={ports[i]}",
+             "time_out=60", "num_trees=15", "num_leaves=31", "verbosity=-1",
+             f"output_model={model}"],
+            capture_output=True, text=True, timeout=240)
+        return i, r, model
+
+    with ThreadPoolExecutor(n_workers) as ex:
+        results = list(ex.map(run_worker, range(n_workers)))
+    models = []
+    for i, r, model in results:
+        assert r.returncode == 0, f"worker {i}:\n{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
+        models.append(model.read_text())
+    trees = [m[m.index("Tree=0"):m.index("end of trees")] for m in models]
+    assert trees[0] == trees[1]
+
+    # quality: load in the Python package and predict
+    bst = lgb.Booster(model_file=str(results[0][2]))
+    acc = ((bst.predict(X) > 0.5) == y).mean()
+    assert acc > 0.93, acc
