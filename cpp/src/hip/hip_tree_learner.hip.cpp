@@ -471,8 +471,8 @@ __global__ void k_iota(uint32_t* p, int n) {
  *  NCOPIES lane-group LDS copies (lane % NCOPIES) dilute same-bin DS-atomic conflicts:
  *  a fully homogeneous leaf serializes 64-wide without them (measured 92us for a
  *  2048-row leaf), /NCOPIES with. Rows are read as 16B uint4 chunks. */
-template <int NCOPIES>
-__global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
+template <int NCOPIES, typename BIN_T = uint8_t>
+__global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
                        const uint32_t* __restrict__ idx_base,
                        const int* __restrict__ leaf_begin, const int* __restrict__ leaf_cnt,
                        const LeafStat* __restrict__ stats, const int* __restrict__ leaf_slot,
@@ -513,15 +513,28 @@ __global__ void k_hist(const uint8_t* __restrict__ rows, int stride,
     const uint32_t r = idx[i];
     const float gi = g[r];
     const float hi = h[r];
-    const uint8_t* rp = rows + static_cast<size_t>(r) * stride;
-    for (int c = c0; c < feat_end; c += 16) {
-      const uint4 v = *reinterpret_cast<const uint4*>(rp + c);
-      const uint32_t w[4] = {v.x, v.y, v.z, v.w};
+    const BIN_T* rp = rows + static_cast<size_t>(r) * stride;
+    if constexpr (sizeof(BIN_T) == 1) {
+      // uint8 rows: 16B uint4 vector loads (the headline path)
+      for (int c = c0; c < feat_end; c += 16) {
+        const uint4 v = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const uint8_t*>(rp) + c);
+        const uint32_t w[4] = {v.x, v.y, v.z, v.w};
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int f = c + j;
-        if (f < feat_begin || f >= feat_end) continue;
-        const int b = (w[j >> 2] >> ((j & 3) * 8)) & 0xFF;
+        for (int j = 0; j < 16; ++j) {
+          const int f = c + j;
+          if (f < feat_begin || f >= feat_end) continue;
+          const int b = (w[j >> 2] >> ((j & 3) * 8)) & 0xFF;
+          float* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
+          atomicAdd(dst, gi);
+          atomicAdd(dst + 1, hi);
+        }
+      }
+    } else {
+      // uint16 rows (max_bin > 256): plain element loads — capability path,
+      // parity: reference CUDAConstructHistogramDenseKernel 16-bit bins
+      for (int f = feat_begin; f < feat_end; ++f) {
+        const int b = rp[f];
         float* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
         atomicAdd(dst, gi);
         atomicAdd(dst + 1, hi);
@@ -1034,11 +1047,12 @@ __device__ __forceinline__ int part_decide(int b, int thr_bin, int nan_bin, int 
   return b <= thr_bin ? 1 : 0;
 }
 
+template <typename BIN_T = uint8_t>
 __global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
                             const int* __restrict__ leaf_begin,
                             const int* __restrict__ leaf_cnt,
                             const int* __restrict__ Lptr, const SplitRec* __restrict__ win,
-                            const FeatMeta* __restrict__ fm, const uint8_t* __restrict__ cols,
+                            const FeatMeta* __restrict__ fm, const BIN_T* __restrict__ cols,
                             int num_data, uint8_t* __restrict__ marks,
                             int* __restrict__ block_cnt) {
   __shared__ int s_cnt[4];
@@ -1046,7 +1060,7 @@ __global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
   if (L < 0) return;
   const int f = win->feature;
   const FeatMeta m = fm[f];
-  const uint8_t* colbins2 = cols + static_cast<size_t>(f) * num_data;
+  const BIN_T* colbins2 = cols + static_cast<size_t>(f) * num_data;
   const int thr_bin = win->bin;
   const int nan_bin = m.is_cat ? -1 : m.nan_bin;
   const int default_left = win->default_left;
@@ -2018,7 +2032,8 @@ __global__ void k_score_update(const uint32_t* __restrict__ idx,
 }
 
 /*! device tree walk over column bins (out-of-bag score update under bagging). */
-__global__ void k_tree_predict_add(const uint8_t* __restrict__ cols, int num_data,
+template <typename BIN_T = uint8_t>
+__global__ void k_tree_predict_add(const BIN_T* __restrict__ cols, int num_data,
                                    const int* __restrict__ split_feat,
                                    const int* __restrict__ thr_bin,
                                    const int* __restrict__ left_child,
@@ -2247,6 +2262,7 @@ class HIPTreeLearner : public TreeLearner {
   std::vector<hipk::LogEntry> host_log_;
 
   bool grads_on_device_ = false;
+  bool rows16_ = false;   // uint16 bins (max_bin > 256)
   bool quantized_ = false;
   bool coop_launch_ = false;   // fused cooperative partition kernel available
   bool use_mono_ = false;      // monotone constraints active (bounds tracked on device)
@@ -2338,6 +2354,8 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_feat_meta_.Alloc(nf_);
   HIP_OK(hipMemcpy(d_feat_meta_.ptr, feat_meta_host_.data(), sizeof(hipk::FeatMeta) * nf_,
                    hipMemcpyHostToDevice));
+  rows16_ = false;  // must be known before LDS partition planning
+  for (int f = 0; f < nf_; ++f) rows16_ = rows16_ || feat_meta_host_[f].num_bin > 256;
 
   // LDS feature partitioning with privatized copies; partitions are 16-feature aligned
   // so row bytes load as whole uint4 chunks. Shrink the copy count if bins are too many.
@@ -2351,8 +2369,9 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
       int end = begin;
       int bins = 0;
       while (end < nf_ && end - begin < 256) {
-        // advance in 16-feature blocks (last block may be partial)
-        int blk_end = std::min(nf_, end + 16);
+        // advance in 16-feature blocks so uint8 row bytes load as whole uint4
+        // chunks (uint16 rows read per element: single-feature granularity)
+        int blk_end = std::min(nf_, end + (rows16_ ? 1 : 16));
         int blk_bins = 0;
         for (int f = end; f < blk_end; ++f) blk_bins += feat_meta_host_[f].num_bin;
         if (bins + blk_bins > max_bins && bins > 0) break;
@@ -2388,6 +2407,10 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
                             reinterpret_cast<const void*>(&hipk::k_hist<4>),
                             reinterpret_cast<const void*>(&hipk::k_hist<2>),
                             reinterpret_cast<const void*>(&hipk::k_hist<1>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<8, uint16_t>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<4, uint16_t>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<2, uint16_t>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<1, uint16_t>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<4>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<2>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<1>)}) {
@@ -2398,28 +2421,41 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   }
 
   const auto& view = train_data->GetRowMajorView();
-  if (view.is16) Log::Fatal("HIP learner currently supports max_bin<=255 (uint8 bins)");
+  MIGBM_CHECK(rows16_ == view.is16);
   row_stride_ = view.row_stride;
-  d_rows_.Alloc(view.data.size());
-  HIP_OK(hipMemcpy(d_rows_.ptr, view.data.data(), view.data.size(), hipMemcpyHostToDevice));
-  d_cols_.Alloc(static_cast<size_t>(nf_) * num_data_);
+  if (rows16_) {
+    d_rows_.Alloc(view.data16.size() * 2);
+    HIP_OK(hipMemcpy(d_rows_.ptr, view.data16.data(), view.data16.size() * 2,
+                     hipMemcpyHostToDevice));
+  } else {
+    d_rows_.Alloc(view.data.size());
+    HIP_OK(hipMemcpy(d_rows_.ptr, view.data.data(), view.data.size(),
+                     hipMemcpyHostToDevice));
+  }
+  const size_t col_esize = rows16_ ? 2 : 1;
+  d_cols_.Alloc(static_cast<size_t>(nf_) * num_data_ * col_esize);
   {
     // per-feature dense bins on device (EFB bundles are decoded here: the GPU keeps
     // feature-major layouts; 288GB HBM makes the unbundled copy cheap)
-    std::vector<uint8_t> colbuf(num_data_);
+    std::vector<uint8_t> colbuf(num_data_ * col_esize);
     for (int f = 0; f < nf_; ++f) {
       const auto& col = train_data->column(train_data->feature_column(f));
-      if (train_data->feature_bundled(f) || col.is_sparse() || col.is4()) {
+      uint8_t* dst = d_cols_.ptr + static_cast<size_t>(f) * num_data_ * col_esize;
+      if (rows16_) {
+        uint16_t* cb16 = reinterpret_cast<uint16_t*>(colbuf.data());
+#pragma omp parallel for schedule(static)
+        for (int i = 0; i < num_data_; ++i)
+          cb16[i] = static_cast<uint16_t>(train_data->GetBin(i, f));
+        HIP_OK(hipMemcpy(dst, colbuf.data(), num_data_ * 2, hipMemcpyHostToDevice));
+      } else if (train_data->feature_bundled(f) || col.is_sparse() || col.is4()) {
         // bundled features decode; sparse columns densify (the device layout is
         // dense row-major + col-major — sparsity is a host-memory concern)
 #pragma omp parallel for schedule(static)
         for (int i = 0; i < num_data_; ++i)
           colbuf[i] = static_cast<uint8_t>(train_data->GetBin(i, f));
-        HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_, colbuf.data(),
-                         num_data_, hipMemcpyHostToDevice));
+        HIP_OK(hipMemcpy(dst, colbuf.data(), num_data_, hipMemcpyHostToDevice));
       } else {
-        HIP_OK(hipMemcpy(d_cols_.ptr + static_cast<size_t>(f) * num_data_, col.data8(),
-                         num_data_, hipMemcpyHostToDevice));
+        HIP_OK(hipMemcpy(dst, col.data8(), num_data_, hipMemcpyHostToDevice));
       }
     }
   }
@@ -2436,9 +2472,13 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     // Measured on MI355X/ROCm 7.2: the cooperative fused partition is ~1.4x
     // SLOWER than 4 stream launches (grid-sync dispatch overhead dominates at
     // this kernel size) — opt-in only, kept as a documented negative result.
-    coop_launch_ = coop != 0 && getenv("MIGBM_COOP_PARTITION") != nullptr;
+    coop_launch_ = coop != 0 && getenv("MIGBM_COOP_PARTITION") != nullptr && !rows16_;
   }
   quantized_ = config_->use_quantized_grad;
+  if (quantized_ && rows16_) {
+    Log::Warning("use_quantized_grad with max_bin>256 runs unquantized on GPU");
+    quantized_ = false;
+  }
   quant_levels_ = std::max(1, config_->num_grad_quant_bins / 2);
   if (quantized_) {
     d_grad_packed_.Alloc(num_data_);
@@ -2694,39 +2734,45 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
       continue;
     }
     const size_t lds = static_cast<size_t>(bins) * (2 * n_copies_ + 2) * sizeof(float);
-    switch (n_copies_) {
-      case 8:
-        hipLaunchKernelGGL(hipk::k_hist<8>, dim3(blocks), dim3(HistThreads()), lds, stream_,
-                           d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
-                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
-                           slot_stride);
-        break;
-      case 4:
-        hipLaunchKernelGGL(hipk::k_hist<4>, dim3(blocks), dim3(HistThreads()), lds, stream_,
-                           d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
-                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
-                           slot_stride);
-        break;
-      case 2:
-        hipLaunchKernelGGL(hipk::k_hist<2>, dim3(blocks), dim3(HistThreads()), lds, stream_,
-                           d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
-                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
-                           slot_stride);
-        break;
-      default:
-        hipLaunchKernelGGL(hipk::k_hist<1>, dim3(blocks), dim3(HistThreads()), lds, stream_,
-                           d_rows_.ptr, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
-                           d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr, leafA_ptr,
-                           d_counters_.ptr, leafB_from_counters, GradPtr(), HessPtr(),
-                           d_feat_meta_.ptr, fb, fe, bin_base, bins, d_hist_.ptr,
-                           slot_stride);
-    }
+    auto launch_hist = [&](auto bin_tag) {
+      using BIN_T = decltype(bin_tag);
+      const BIN_T* rp = reinterpret_cast<const BIN_T*>(d_rows_.ptr);
+      switch (n_copies_) {
+        case 8:
+          hipLaunchKernelGGL((hipk::k_hist<8, BIN_T>), dim3(blocks), dim3(HistThreads()),
+                             lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
+                             d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
+                             leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
+                             HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
+                             d_hist_.ptr, slot_stride);
+          break;
+        case 4:
+          hipLaunchKernelGGL((hipk::k_hist<4, BIN_T>), dim3(blocks), dim3(HistThreads()),
+                             lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
+                             d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
+                             leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
+                             HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
+                             d_hist_.ptr, slot_stride);
+          break;
+        case 2:
+          hipLaunchKernelGGL((hipk::k_hist<2, BIN_T>), dim3(blocks), dim3(HistThreads()),
+                             lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
+                             d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
+                             leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
+                             HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
+                             d_hist_.ptr, slot_stride);
+          break;
+        default:
+          hipLaunchKernelGGL((hipk::k_hist<1, BIN_T>), dim3(blocks), dim3(HistThreads()),
+                             lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
+                             d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
+                             leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
+                             HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
+                             d_hist_.ptr, slot_stride);
+      }
+    };
+    if (rows16_) launch_hist(uint16_t{});
+    else launch_hist(uint8_t{});
   }
 }
 
@@ -2938,10 +2984,20 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
           reinterpret_cast<void*>(hipk::k_part_fused), dim3(kPartBlocks),
           dim3(kHistBlock), args, 0, stream_));
     } else {
-      hipLaunchKernelGGL(hipk::k_part_mark, dim3(kPartBlocks), dim3(kHistBlock), 0, stream_,
-                         d_idx_.ptr, d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_winner_leaf_.ptr,
-                         d_winner_.ptr, d_feat_meta_.ptr, d_cols_.ptr, num_data_,
-                         d_marks_.ptr, d_block_cnt_.ptr);
+      if (rows16_) {
+        hipLaunchKernelGGL((hipk::k_part_mark<uint16_t>), dim3(kPartBlocks),
+                           dim3(kHistBlock), 0, stream_, d_idx_.ptr, d_leaf_begin_.ptr,
+                           d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_winner_.ptr,
+                           d_feat_meta_.ptr,
+                           reinterpret_cast<const uint16_t*>(d_cols_.ptr), num_data_,
+                           d_marks_.ptr, d_block_cnt_.ptr);
+      } else {
+        hipLaunchKernelGGL((hipk::k_part_mark<uint8_t>), dim3(kPartBlocks),
+                           dim3(kHistBlock), 0, stream_, d_idx_.ptr, d_leaf_begin_.ptr,
+                           d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_winner_.ptr,
+                           d_feat_meta_.ptr, d_cols_.ptr, num_data_, d_marks_.ptr,
+                           d_block_cnt_.ptr);
+      }
       hipLaunchKernelGGL(hipk::k_part_scan, dim3(1), dim3(256), 0, stream_, d_block_cnt_.ptr,
                          kPartBlocks, d_leaf_cnt_.ptr, d_winner_leaf_.ptr, d_block_loff_.ptr,
                          d_block_roff_.ptr, d_ctr_.ptr);
@@ -3092,12 +3148,21 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
                             hipMemcpyHostToDevice, stream_));
       HIP_OK(hipMemcpyAsync(d_oob_.ptr, oob.data(), sizeof(uint32_t) * oob.size(),
                             hipMemcpyHostToDevice, stream_));
-      hipLaunchKernelGGL(hipk::k_tree_predict_add,
-                         dim3((static_cast<int>(oob.size()) + 255) / 256), dim3(256), 0,
-                         stream_, d_cols_.ptr, num_data_, d_tw_feat_.ptr, d_tw_thr_.ptr,
-                         d_tw_left_.ptr, d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr,
-                         d_tw_out_.ptr, d_oob_.ptr, static_cast<int>(oob.size()),
-                         ScorePtr());
+      if (rows16_) {
+        hipLaunchKernelGGL((hipk::k_tree_predict_add<uint16_t>),
+                           dim3((static_cast<int>(oob.size()) + 255) / 256), dim3(256), 0,
+                           stream_, reinterpret_cast<const uint16_t*>(d_cols_.ptr),
+                           num_data_, d_tw_feat_.ptr, d_tw_thr_.ptr, d_tw_left_.ptr,
+                           d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr, d_tw_out_.ptr,
+                           d_oob_.ptr, static_cast<int>(oob.size()), ScorePtr());
+      } else {
+        hipLaunchKernelGGL((hipk::k_tree_predict_add<uint8_t>),
+                           dim3((static_cast<int>(oob.size()) + 255) / 256), dim3(256), 0,
+                           stream_, d_cols_.ptr, num_data_, d_tw_feat_.ptr, d_tw_thr_.ptr,
+                           d_tw_left_.ptr, d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr,
+                           d_tw_out_.ptr, d_oob_.ptr, static_cast<int>(oob.size()),
+                           ScorePtr());
+      }
     }
   }
 }

@@ -575,3 +575,36 @@ def test_gpu_num_gpu_l2_and_metrics():
     # device metric (summed over shards) must equal the host-recomputed value
     assert abs(ev["training"]["l1"][-1] - mae) < 1e-6
     assert mae < 0.5 * float(np.abs(y - y.mean()).mean())
+
+
+def test_gpu_large_bins_uint16():
+    """max_bin > 256 runs on device with uint16 bins (the GPU
+    previously hard-errored at num_bin>256). Parity with the CPU learner at
+    max_bin=1023, plus the kernel-level fp64 hist oracle at 16-bit bins."""
+    import ctypes
+    from lightgbm_amd.basic import _LIB, _c_str
+    rng = np.random.RandomState(8)
+    n = 60_000
+    X = rng.randn(n, 10).astype(np.float32)
+    y = (1.3 * X[:, 0] - 0.7 * X[:, 1] + 0.5 * X[:, 2] * X[:, 3] +
+         0.8 * rng.randn(n) > 0).astype(np.float32)
+    aucs = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "binary", "device_type": dev, "max_bin": 1023,
+                  "num_leaves": 63, "min_data_in_leaf": 20, "verbosity": 0,
+                  "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 20)
+        aucs[dev] = _auc(y[:20000], bst.predict(X[:20000]))
+    assert aucs["gpu"] > 0.8
+    assert abs(aucs["cpu"] - aucs["gpu"]) < 3e-3, aucs
+    # kernel probe: device uint16 histogram vs fp64 host oracle
+    ds = lgb.Dataset(X, label=y, params={"max_bin": 1023}).construct()
+    g = rng.randn(n).astype(np.float32)
+    h = rng.uniform(0.5, 2.0, n).astype(np.float32)
+    err = ctypes.c_double(1e9)
+    rc = _LIB.MIGBM_DebugDeviceRootHist(
+        ds._handle, _c_str("max_bin=1023 num_leaves=31"),
+        g.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        h.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), ctypes.byref(err))
+    assert rc == 0
+    assert err.value < 2e-3, err.value
