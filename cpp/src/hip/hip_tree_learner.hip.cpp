@@ -2341,7 +2341,6 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   feat_meta_host_.resize(nf_);
   for (int f = 0; f < nf_; ++f) {
     const BinMapper* m = train_data->FeatureBinMapper(f);
-    if (m->num_bin() > 256) Log::Fatal("HIP learner currently supports max_bin<=255");
     if (m->bin_type() == BinType::kCategorical && m->num_bin() > 64) {
       Log::Warning("HIP learner evaluates categorical feature %d with one-hot splits "
                    "only (%d categories exceed the 64-bin device sorted-subset scan)",
