@@ -64,9 +64,26 @@ $(ASAN_CLI): $(HOST_SRCS) cpp/cli/main.cpp
 	mkdir -p $(ASAN_BUILD)
 	$(CXX) $(ASAN_FLAGS) $(HOST_SRCS) cpp/cli/main.cpp -o $(ASAN_CLI)
 
+# ---- UBSAN lane (host-only CLI under UndefinedBehaviorSanitizer)
+UBSAN_FLAGS := -O1 -g -std=c++17 -fopenmp -fsanitize=undefined -fno-omit-frame-pointer \
+               -Icpp/include -DMIGBM_NO_HIP
+ubsan: $(HOST_SRCS) cpp/cli/main.cpp
+	mkdir -p build_ubsan
+	$(CXX) $(UBSAN_FLAGS) $(HOST_SRCS) cpp/cli/main.cpp -o build_ubsan/migbm_ubsan
+
+# ---- TSAN lane (host-only CLI under ThreadSanitizer; OpenMP races in the
+# histogram/partition paths surface here). Built with amdclang++/libomp: its
+# runtime carries TSAN annotations, gcc's libgomp does not (false positives).
+TSAN_CXX   := /opt/rocm/lib/llvm/bin/clang++
+TSAN_FLAGS := -O1 -g -std=c++17 -fopenmp -fsanitize=thread -fno-omit-frame-pointer \
+              -Icpp/include -DMIGBM_NO_HIP
+tsan: $(HOST_SRCS) cpp/cli/main.cpp
+	mkdir -p build_tsan
+	$(TSAN_CXX) $(TSAN_FLAGS) $(HOST_SRCS) cpp/cli/main.cpp -o build_tsan/migbm_tsan
+
 clean:
-	rm -rf $(BUILD) $(TARGET) $(CLI) $(ASAN_BUILD)
+	rm -rf $(BUILD) $(TARGET) $(CLI) $(ASAN_BUILD) build_ubsan build_tsan
 
 -include $(BUILD)/*.d $(BUILD)/hip/*.d
 
-.PHONY: all clean asan
+.PHONY: all clean asan ubsan tsan
