@@ -72,6 +72,7 @@ class Tree {
   bool IsCategoricalSplit(int node) const { return (decision_type_[node] & kCategoricalMask) != 0; }
   double shrinkage() const { return shrinkage_; }
   int leaf_depth(int leaf) const { return leaf_depth_[leaf]; }
+  int leaf_parent(int leaf) const { return leaf_parent_[leaf]; }
   double internal_value(int node) const { return internal_value_[node]; }
   double InternalCountSafe(int node) const {
     return node < static_cast<int>(internal_count_.size()) ? internal_count_[node] : 1.0;

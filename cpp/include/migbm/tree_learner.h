@@ -224,6 +224,25 @@ class SerialTreeLearner : public TreeLearner {
   std::vector<int64_t> slot_used_;  // LRU stamps
   int64_t slot_clock_ = 0;
   int AcquireSlot(int leaf, int pin_a = -1, int pin_b = -1);
+  /*! rebuild a leaf's histogram if the pool evicted it */
+  void EnsureLeafHistogram(int leaf);
+  // ---- intermediate monotone constraints (monotone_constraints_method !=
+  // "basic"): after a split inside a monotone subtree, bounds of CONTIGUOUS
+  // leaves elsewhere in the tree are tightened against the new outputs and
+  // their best splits recomputed (reference IntermediateLeafConstraints,
+  // monotone_constraints.hpp:516-858 — algorithm reimplemented fresh)
+  bool mono_intermediate_ = false;
+  std::vector<int> mono_node_parent_;       // internal node -> parent (-1 root)
+  std::vector<uint8_t> mono_leaf_in_subtree_;
+  void MonotoneIntermediateUpdate(const Tree* tree, int left_leaf, int right_leaf,
+                                  const SplitInfo& s, int split_node,
+                                  bool is_numerical);
+  void MonoGoDown(const Tree* tree, int node, const std::vector<int>& up_feats,
+                  const std::vector<uint32_t>& up_thresholds,
+                  const std::vector<uint8_t>& up_was_right, bool update_max,
+                  int split_feature, const SplitInfo& s, bool use_left,
+                  bool use_right, uint32_t split_threshold,
+                  std::vector<int>* leaves_to_update);
   std::vector<SplitInfo> best_split_per_leaf_;
   std::vector<LeafContext> leaf_ctx_;
   std::vector<int8_t> is_feature_used_;     // per-tree mask

@@ -237,6 +237,11 @@ std::vector<int8_t> SerialTreeLearner::SampleFeatures(bool per_node) {
   return used;
 }
 
+static bool MonoDebug() {
+  static const bool v = getenv("MIGBM_MONO_DEBUG") != nullptr;
+  return v;
+}
+
 int SerialTreeLearner::AcquireSlot(int leaf, int pin_a, int pin_b) {
   if (leaf_to_slot_[leaf] >= 0) {
     slot_used_[leaf_to_slot_[leaf]] = ++slot_clock_;
@@ -493,6 +498,10 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
     }
     if (cand[f] > best) best = cand[f];
   }
+  if (MonoDebug() && best.IsValid())
+    fprintf(stderr, "[mono]   fbs leaf=%d -> f=%d thr=%u mono=%d gain=%.4g b=[%.4g,%.4g]\n",
+            leaf, best.feature, best.threshold, (int)best.monotone_type, best.gain,
+            ctx.out_lo, ctx.out_hi);
 }
 
 bool SerialTreeLearner::MakeForcedSplit(int leaf, const LeafContext& ctx,
@@ -583,12 +592,19 @@ std::function<bool(data_size_t)> SerialTreeLearner::MakeGoLeft(const SplitInfo& 
 Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians,
                                bool /*is_first_tree*/) {
   tree_const_hess_ = -1;  // re-classify hessians for this tree's gradients
-  if (!config_->monotone_constraints.empty() &&
-      config_->monotone_constraints_method != "basic" && !warned_mc_method_) {
-    Log::Warning("monotone_constraints_method=%s is not implemented; using the basic "
-                 "method with leaf-bound propagation",
-                 config_->monotone_constraints_method.c_str());
-    warned_mc_method_ = true;
+  mono_intermediate_ = !config_->monotone_constraints.empty() &&
+                       config_->monotone_constraints_method != "basic";
+  if (mono_intermediate_) {
+    // "advanced" maps to the intermediate policy (per-threshold piecewise
+    // constraints are not implemented; intermediate already recomputes
+    // affected splits with output-tight bounds)
+    if (config_->monotone_constraints_method == "advanced" && !warned_mc_method_) {
+      Log::Warning("monotone_constraints_method=advanced runs the intermediate "
+                   "policy (output-tight bounds + contiguous-leaf re-evaluation)");
+      warned_mc_method_ = true;
+    }
+    mono_node_parent_.assign(config_->num_leaves, -1);
+    mono_leaf_in_subtree_.assign(config_->num_leaves, 0);
   }
   gradients_ = gradients;
   hessians_ = hessians;
@@ -730,6 +746,7 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     s.right_count = right_cnt_actual;
 
     // tree structure update
+    int split_parent_node = -1;
     if (!s.cat_bitset_inner.empty()) {
       // map bin-level bitset to category-value bitset for prediction on raw values
       std::vector<uint32_t> cat_bits;
@@ -746,6 +763,7 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
           }
         }
       }
+      split_parent_node = tree->leaf_parent(best_leaf);
       tree->SplitCategorical(best_leaf, f, orig_f, cat_bits.data(),
                              static_cast<int>(cat_bits.size()), s.left_output, s.right_output,
                              s.left_count, s.right_count, s.left_sum_hessian,
@@ -753,6 +771,7 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                              m->missing_type());
       // store bin-level bitset on the tree node for training-time partition? partition uses s directly
     } else {
+      split_parent_node = tree->leaf_parent(best_leaf);
       tree->Split(best_leaf, f, orig_f, s.threshold,
                   train_data_->RealThreshold(f, s.threshold), s.left_output, s.right_output,
                   s.left_count, s.right_count, s.left_sum_hessian, s.right_sum_hessian,
@@ -785,7 +804,7 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
                             parent_out, parent_depth + 1, b_lo, b_hi};
     leaf_ctx_[right_leaf] = {s.right_sum_gradient, s.right_sum_hessian, right_cnt_actual,
                              parent_out, parent_depth + 1, b_lo, b_hi};
-    if (s.monotone_type != 0) {
+    if (!mono_intermediate_ && s.monotone_type != 0) {
       // BasicLeafConstraints: descendants of the low side may not exceed the split
       // midpoint, and vice versa — monotonicity holds for the whole subtree
       const double mid = (s.left_output + s.right_output) / 2.0;
@@ -796,6 +815,10 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
         leaf_ctx_[best_leaf].out_lo = std::max(b_lo, mid);
         leaf_ctx_[right_leaf].out_hi = std::min(b_hi, mid);
       }
+    }
+    if (mono_intermediate_) {
+      const int new_node = tree->leaf_parent(best_leaf);
+      mono_node_parent_[new_node] = split_parent_node;
     }
     ++num_leaves;
 
@@ -833,8 +856,28 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
       SubtractHistogram(large_leaf, leaf_to_slot_[large_leaf], spare_slot);
     }
 
-    FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
-    FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);
+    if (mono_intermediate_) {
+      // NOTE: s references best_split_per_leaf_[best_leaf], which the child
+      // FindBestSplitForLeaf calls below overwrite — snapshot the APPLIED split
+      const SplitInfo applied = s;
+      FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
+      FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);
+      if (MonoDebug())
+        fprintf(stderr,
+                "[mono] split leaf=%d->(%d,%d) inner=%d thr=%u mono=%d "
+                "out=(%.6g,%.6g) bounds=[%.4g,%.4g]\n",
+                best_leaf, best_leaf, right_leaf, applied.feature, applied.threshold,
+                (int)applied.monotone_type, applied.left_output, applied.right_output,
+                b_lo, b_hi);
+      // tighten contiguous leaves' bounds against the new outputs and recompute
+      // their best splits (reference IntermediateLeafConstraints::Update)
+      MonotoneIntermediateUpdate(tree.get(), best_leaf, right_leaf, applied,
+                                 tree->leaf_parent(best_leaf),
+                                 applied.cat_bitset_inner.empty());
+    } else {
+      FindBestSplitForLeaf(small_leaf, leaf_ctx_[small_leaf]);
+      FindBestSplitForLeaf(large_leaf, leaf_ctx_[large_leaf]);
+    }
   }
   if (config_->use_quantized_grad && config_->quant_train_renew_leaf) {
     // renew leaf outputs from the UNquantized gradients (reference
@@ -951,6 +994,180 @@ void SerialTreeLearner::CalculateLinear(Tree* tree) {
     for (int j = 0; j < k; ++j) feats_real[j] = train_data_->RealFeatureIndex(feats[j]);
     std::vector<double> coeffs(beta.begin(), beta.begin() + k);
     tree->SetLeafLinear(l, beta[k], feats_real, feats, coeffs);
+  }
+}
+
+
+void SerialTreeLearner::EnsureLeafHistogram(int leaf) {
+  if (leaf_to_slot_[leaf] >= 0) return;
+  data_size_t cnt;
+  const data_size_t* idx = partition_.GetIndexOnLeaf(leaf, &cnt);
+  ComputeHistogram(leaf, cnt, idx);
+  OnHistogramReady(leaf);
+}
+
+/*! contiguity-aware descent into the subtree OPPOSITE a monotone ancestor
+ *  split: finds the leaves whose regions touch the freshly split leaves and
+ *  tightens their output bounds against the new outputs. */
+void SerialTreeLearner::MonoGoDown(const Tree* tree, int node,
+                                   const std::vector<int>& up_feats,
+                                   const std::vector<uint32_t>& up_thresholds,
+                                   const std::vector<uint8_t>& up_was_right,
+                                   bool update_max, int split_feature,
+                                   const SplitInfo& s, bool use_left, bool use_right,
+                                   uint32_t split_threshold,
+                                   std::vector<int>* leaves_to_update) {
+  if (node < 0) {
+    const int leaf = ~node;
+    // leaves that cannot split anymore are still bound-tracked (their OUTPUT is
+    // already monotone-consistent; only future-split bounds matter)
+    double lo_c, hi_c;
+    if (use_left && use_right) {
+      lo_c = std::min(s.left_output, s.right_output);
+      hi_c = std::max(s.left_output, s.right_output);
+    } else if (use_right) {
+      lo_c = hi_c = s.right_output;
+    } else {
+      lo_c = hi_c = s.left_output;
+    }
+    bool changed = false;
+    if (MonoDebug()) {
+      const double out = tree->LeafOutput(leaf);
+      if (update_max ? (lo_c < out - 1e-12) : (hi_c > out + 1e-12))
+        fprintf(stderr,
+                "[mono] INVARIANT BROKEN leaf=%d out=%.6g update_max=%d c=[%.6g,%.6g]\n",
+                leaf, out, update_max ? 1 : 0, lo_c, hi_c);
+    }
+    if (update_max) {
+      if (lo_c < leaf_ctx_[leaf].out_hi) {
+        leaf_ctx_[leaf].out_hi = lo_c;
+        changed = true;
+      }
+    } else {
+      if (hi_c > leaf_ctx_[leaf].out_lo) {
+        leaf_ctx_[leaf].out_lo = hi_c;
+        changed = true;
+      }
+    }
+    if (changed) {
+      leaves_to_update->push_back(leaf);
+      if (MonoDebug())
+        fprintf(stderr, "[mono]   update leaf=%d out=%.6g -> bounds=[%.4g,%.4g]\n", leaf,
+                tree->LeafOutput(leaf), leaf_ctx_[leaf].out_lo, leaf_ctx_[leaf].out_hi);
+    }
+    return;
+  }
+  const int feat = tree->split_feature_inner(node);
+  const uint32_t thr = tree->threshold_in_bin(node);
+  const bool numerical = !tree->IsCategoricalSplit(node);
+  // prune subtrees that cannot touch the original leaves' region: a split on a
+  // feature already crossed on the way up bounds the region on one side
+  bool go_left = true, go_right = true;
+  if (numerical) {
+    for (size_t i = 0; i < up_feats.size(); ++i) {
+      if (up_feats[i] != feat) continue;
+      if (thr >= up_thresholds[i] && !up_was_right[i]) go_right = false;
+      if (thr <= up_thresholds[i] && up_was_right[i]) go_left = false;
+      if (!go_left && !go_right) break;
+    }
+  }
+  // a same-feature split separates one child's region from one of the two new
+  // leaves (reference use_left_leaf_for_update_right / use_right_..._left)
+  bool right_child_sees_left = true;  // right descent stays contiguous w/ LEFT leaf
+  bool left_child_sees_right = true;  // left descent stays contiguous w/ RIGHT leaf
+  if (numerical && feat == split_feature) {
+    if (thr >= split_threshold) right_child_sees_left = false;
+    if (thr <= split_threshold) left_child_sees_right = false;
+  }
+  if (go_left) {
+    MonoGoDown(tree, tree->left_child(node), up_feats, up_thresholds, up_was_right,
+               update_max, split_feature, s, use_left,
+               use_right && left_child_sees_right, split_threshold, leaves_to_update);
+  }
+  if (go_right) {
+    MonoGoDown(tree, tree->right_child(node), up_feats, up_thresholds, up_was_right,
+               update_max, split_feature, s, use_left && right_child_sees_left,
+               use_right, split_threshold, leaves_to_update);
+  }
+}
+
+void SerialTreeLearner::MonotoneIntermediateUpdate(const Tree* tree, int left_leaf,
+                                                   int right_leaf, const SplitInfo& s,
+                                                   int split_node, bool is_numerical) {
+  const bool was_mono_subtree = mono_leaf_in_subtree_[left_leaf] != 0;
+  if (s.monotone_type != 0 || was_mono_subtree) {
+    mono_leaf_in_subtree_[left_leaf] = 1;
+    mono_leaf_in_subtree_[right_leaf] = 1;
+  }
+  if (!mono_leaf_in_subtree_[left_leaf]) return;
+  // children bound the SIBLING's actual output (tighter than Basic's midpoint
+  // clamp on one side, looser on the other — reference UpdateConstraintsWithOutputs)
+  if (is_numerical && s.monotone_type != 0) {
+    if (s.monotone_type < 0) {
+      leaf_ctx_[left_leaf].out_lo = std::max(leaf_ctx_[left_leaf].out_lo, s.right_output);
+      leaf_ctx_[right_leaf].out_hi =
+          std::min(leaf_ctx_[right_leaf].out_hi, s.left_output);
+    } else {
+      leaf_ctx_[left_leaf].out_hi = std::min(leaf_ctx_[left_leaf].out_hi, s.right_output);
+      leaf_ctx_[right_leaf].out_lo =
+          std::max(leaf_ctx_[right_leaf].out_lo, s.left_output);
+    }
+  }
+  // walk UP from the new split; at every monotone ancestor whose opposite
+  // subtree is contiguous, walk DOWN it to tighten leaf bounds
+  std::vector<int> leaves_to_update;
+  std::vector<int> up_feats;
+  std::vector<uint32_t> up_thresholds;
+  std::vector<uint8_t> up_was_right;
+  const int split_feature = is_numerical ? tree->split_feature_inner(split_node) : -1;
+  const uint32_t split_threshold = is_numerical ? tree->threshold_in_bin(split_node) : 0;
+  int node = split_node;
+  int parent = mono_node_parent_[node];
+  while (parent >= 0) {
+    const int feat = tree->split_feature_inner(parent);
+    const bool p_numerical = !tree->IsCategoricalSplit(parent);
+    const int orig_f = tree->split_feature(parent);
+    int8_t mono = 0;
+    if (orig_f >= 0 && orig_f < static_cast<int>(config_->monotone_constraints.size()))
+      mono = static_cast<int8_t>(config_->monotone_constraints[orig_f]);
+    const bool is_right_child = tree->right_child(parent) == node;
+    // skip descents that cannot contain contiguous leaves: the same feature
+    // crossed in the same direction earlier already fences the region
+    bool should_descend = p_numerical;
+    if (p_numerical) {
+      for (size_t i = 0; i < up_feats.size(); ++i) {
+        if (up_feats[i] == feat &&
+            (up_was_right[i] != 0) == is_right_child) {
+          should_descend = false;
+          break;
+        }
+      }
+    }
+    if (should_descend) {
+      if (mono != 0) {
+        const int opposite = is_right_child ? tree->left_child(parent)
+                                            : tree->right_child(parent);
+        // mono<0 (decreasing): the LEFT subtree must stay ABOVE the right side;
+        // coming up from the left child, the opposite (right) side gets a MAX cap
+        const bool update_max = (mono < 0) == !is_right_child;
+        MonoGoDown(tree, opposite, up_feats, up_thresholds, up_was_right, update_max,
+                   split_feature, s, true, true, split_threshold, &leaves_to_update);
+      }
+      up_was_right.push_back(is_right_child ? 1 : 0);
+      up_thresholds.push_back(tree->threshold_in_bin(parent));
+      up_feats.push_back(feat);
+    }
+    node = parent;
+    parent = mono_node_parent_[node];
+  }
+  // constraints changed: those leaves' cached best splits may now be invalid
+  std::sort(leaves_to_update.begin(), leaves_to_update.end());
+  leaves_to_update.erase(std::unique(leaves_to_update.begin(), leaves_to_update.end()),
+                         leaves_to_update.end());
+  for (int l : leaves_to_update) {
+    if (l == left_leaf || l == right_leaf) continue;
+    EnsureLeafHistogram(l);
+    FindBestSplitForLeaf(l, leaf_ctx_[l]);
   }
 }
 

@@ -909,3 +909,38 @@ def test_new_reference_params_accepted():
               "lambdarank_position_bias_regularization": 0.1}
     bst = lgb.train(params, lgb.Dataset(X, label=y), 5)
     assert bst.num_trees() == 5
+
+
+@pytest.mark.parametrize("method", ["intermediate", "advanced"])
+def test_monotone_intermediate_policy(method):
+    """monotone_constraints_method=intermediate/advanced: output-tight bounds +
+    contiguous-leaf re-evaluation. Predictions must be globally
+    monotone in the constrained features, and quality must not regress vs basic."""
+    rng = np.random.RandomState(0)
+    n = 20000
+    X = rng.rand(n, 4)
+    y = (2.0 * X[:, 0] - 1.5 * X[:, 1] + np.sin(6 * X[:, 2]) +
+         0.1 * rng.randn(n)).astype(np.float32)
+    preds = {}
+    for m in ("basic", method):
+        params = {"objective": "regression", "verbosity": -1, "num_leaves": 63,
+                  "monotone_constraints": [1, -1, 0, 0],
+                  "monotone_constraints_method": m}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 40)
+        preds[m] = bst
+    # global monotonicity on dense grids across many slices
+    xs = np.linspace(0.01, 0.99, 40)
+    for other in (0.15, 0.5, 0.85):
+        for o2 in (0.2, 0.7):
+            g_up = np.column_stack([xs, np.full(40, other), np.full(40, o2),
+                                    np.full(40, other)])
+            p = preds[method].predict(g_up)
+            assert np.all(np.diff(p) >= -1e-9), "increasing constraint violated"
+            g_dn = np.column_stack([np.full(40, other), xs, np.full(40, o2),
+                                    np.full(40, other)])
+            p = preds[method].predict(g_dn)
+            assert np.all(np.diff(p) <= 1e-9), "decreasing constraint violated"
+    # intermediate's looser child bounds must not hurt accuracy vs basic
+    mse_b = float(np.mean((preds["basic"].predict(X) - y) ** 2))
+    mse_i = float(np.mean((preds[method].predict(X) - y) ** 2))
+    assert mse_i < mse_b * 1.05, (mse_b, mse_i)
