@@ -223,7 +223,9 @@ struct SplitRec {
   double gain;
   double left_g, left_h;
   double left_out, right_out;
-  unsigned long long cat_mask;  // bin-level subset (categorical sorted scan); 0 = one-hot/numeric
+  // bin-level subset for categorical sorted-subset splits, 256 bins max
+  // (4x64-bit words); all-zero = one-hot/numeric
+  unsigned long long cat_mask[4];
   int left_cnt, right_cnt;  // hessian-derived approx (host launch-sizing hints)
   int feature;
   int bin;
@@ -661,7 +663,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
   if (lane == 0) {
     rec.valid = 0;
     rec.feature = f;
-    rec.cat_mask = 0;
+    rec.cat_mask[0] = rec.cat_mask[1] = rec.cat_mask[2] = rec.cat_mask[3] = 0;
   }
   if (f < p.own_fb || f >= p.own_fe) return;  // another rank owns this feature's scan
   if (feat_mask != nullptr && !feat_mask[f]) return;
@@ -827,7 +829,8 @@ __global__ void __launch_bounds__(64) k_best_feat(
             rec.feature = f;
             rec.bin = b_k + 1;  // #cats on the left (CPU threshold semantics)
             rec.default_left = 0;
-            rec.cat_mask = mask;
+            rec.cat_mask[0] = mask;
+            rec.cat_mask[1] = rec.cat_mask[2] = rec.cat_mask[3] = 0;
             rec.left_g = b_lg;
             rec.left_h = b_lh;
             rec.left_out = d_leaf_out_l2(b_lg, b_lh, p, l2c);
@@ -835,6 +838,147 @@ __global__ void __launch_bounds__(64) k_best_feat(
             rec.left_cnt = static_cast<int>(b_lh * cnt_factor + 0.5);
             rec.right_cnt = num_data - rec.left_cnt;
           }
+        }
+      }
+      return;
+    }
+    if (m.num_bin > p.max_cat_to_onehot && m.num_bin <= 256) {
+      // 65..256-bin sorted-subset scan: LDS bitonic sort (4 elems/lane) +
+      // serial prefix, winner mask carried as 4x64-bit words in SplitRec.
+      // Same math as the <=64 wave path (CPU-oracle parity).
+      __shared__ double s_ratio[256], s_cg[256], s_ch[256];
+      __shared__ short s_sorted_orig[256];
+      __shared__ double s_pg[256], s_ph[256];
+      __shared__ int s_nelig;
+      int E = 1;
+      while (E < m.num_bin) E <<= 1;
+      for (int i = lane; i < E; i += 64) {
+        double gb = 0, hb = 0, ratio = 1e300;
+        if (i < m.num_bin) {
+          gb = fh[2 * i];
+          hb = fh[2 * i + 1];
+          const bool elig = hb * cnt_factor >= p.cat_smooth;
+          ratio = elig ? gb / (hb + p.cat_smooth) : 1e300;
+        }
+        s_ratio[i] = ratio;
+        s_cg[i] = gb;
+        s_ch[i] = hb;
+        s_sorted_orig[i] = static_cast<short>(i);
+      }
+      __syncthreads();
+      for (int ksz = 2; ksz <= E; ksz <<= 1) {
+        for (int j = ksz >> 1; j > 0; j >>= 1) {
+          for (int i = lane; i < E; i += 64) {
+            const int ixj = i ^ j;
+            if (ixj > i) {
+              const bool asc = (i & ksz) == 0;
+              const double r1 = s_ratio[i], r2 = s_ratio[ixj];
+              const short o1 = s_sorted_orig[i], o2 = s_sorted_orig[ixj];
+              const bool other_lt = r2 < r1 || (r2 == r1 && o2 < o1);
+              if (other_lt == asc) {
+                s_ratio[i] = r2;
+                s_ratio[ixj] = r1;
+                s_sorted_orig[i] = o2;
+                s_sorted_orig[ixj] = o1;
+                const double tg = s_cg[i], th = s_ch[i];
+                s_cg[i] = s_cg[ixj];
+                s_ch[i] = s_ch[ixj];
+                s_cg[ixj] = tg;
+                s_ch[ixj] = th;
+              }
+            }
+          }
+          __syncthreads();
+        }
+      }
+      {
+        int cnt_e = 0;
+        for (int i = lane; i < E; i += 64)
+          if (s_ratio[i] < 1e300) ++cnt_e;
+        for (int d = 32; d > 0; d >>= 1) cnt_e += __shfl_down(cnt_e, d);
+        if (lane == 0) s_nelig = cnt_e;
+      }
+      __syncthreads();
+      const int n_elig = s_nelig;
+      if (n_elig >= 2) {
+        if (lane == 0) {
+          double pg = 0, ph = 0;
+          for (int i = 0; i < n_elig; ++i) {
+            pg += s_cg[i];
+            ph += s_ch[i];
+            s_pg[i] = pg;
+            s_ph[i] = ph;
+          }
+        }
+        __syncthreads();
+        const int limit = min(p.max_cat_threshold, n_elig - 1);
+        const double tot_g = s_pg[n_elig - 1];
+        const double tot_h = s_ph[n_elig - 1];
+        double bg_cat = -1e308;
+        int b_k = -1, b_dir = 0;
+        double b_lg = 0, b_lh = 0;
+        for (int k2 = lane; k2 < limit; k2 += 64) {
+          for (int dir = 0; dir < 2; ++dir) {
+            double sgl, shl;
+            if (dir == 0) {
+              sgl = s_pg[k2];
+              shl = s_ph[k2];
+            } else {
+              const int idx = n_elig - 2 - k2;
+              if (idx < 0) continue;
+              sgl = tot_g - s_pg[idx];
+              shl = tot_h - s_ph[idx];
+            }
+            const double sgr = sum_g - sgl, shr = sum_h - shl;
+            const int lc = static_cast<int>(shl * cnt_factor + 0.5);
+            const int rc = num_data - lc;
+            if (shl < p.min_hess || lc < p.min_data) continue;
+            if (shr < p.min_hess || rc < p.min_data) continue;
+            const double gain = d_split_gain_l2(sgl, shl, sgr, shr, p, l2c);
+            if (gain <= min_gain_shift) continue;
+            if (gain > bg_cat || (gain == bg_cat && k2 < b_k)) {
+              bg_cat = gain;
+              b_k = k2;
+              b_dir = dir;
+              b_lg = sgl;
+              b_lh = shl;
+            }
+          }
+        }
+        for (int d = 32; d > 0; d >>= 1) {
+          const double og = __shfl_xor(bg_cat, d);
+          const int ok = __shfl_xor(b_k, d);
+          const int od = __shfl_xor(b_dir, d);
+          const double olg = __shfl_xor(b_lg, d);
+          const double olh = __shfl_xor(b_lh, d);
+          if ((og > bg_cat) ||
+              (og == bg_cat && ok >= 0 && (b_k < 0 || ok < b_k))) {
+            bg_cat = og;
+            b_k = ok;
+            b_dir = od;
+            b_lg = olg;
+            b_lh = olh;
+          }
+        }
+        if (lane == 0 && bg_cat > -1e307) {
+          rec.valid = 1;
+          rec.gain = bg_cat - min_gain_shift + p.min_gain_to_split;
+          rec.feature = f;
+          rec.bin = b_k + 1;  // #cats on the left (CPU threshold semantics)
+          rec.default_left = 0;
+          rec.cat_mask[0] = rec.cat_mask[1] = rec.cat_mask[2] = rec.cat_mask[3] = 0;
+          const int lo_i = b_dir == 0 ? 0 : n_elig - 1 - b_k;
+          const int hi_i = b_dir == 0 ? b_k : n_elig - 1;
+          for (int i = lo_i; i <= hi_i; ++i) {
+            const int ob = s_sorted_orig[i];
+            rec.cat_mask[ob >> 6] |= 1ull << (ob & 63);
+          }
+          rec.left_g = b_lg;
+          rec.left_h = b_lh;
+          rec.left_out = d_leaf_out_l2(b_lg, b_lh, p, l2c);
+          rec.right_out = d_leaf_out_l2(sum_g - b_lg, sum_h - b_lh, p, l2c);
+          rec.left_cnt = static_cast<int>(b_lh * cnt_factor + 0.5);
+          rec.right_cnt = num_data - rec.left_cnt;
         }
       }
       return;
@@ -1038,10 +1182,15 @@ __global__ void k_best_leaf_overall(const SplitRec* __restrict__ feat_best, int 
  *  blocks round-robined over the grid). Three phases: mark+count per block, block
  *  offset scan, ranked scatter. No same-address global atomics. */
 __device__ __forceinline__ int part_decide(int b, int thr_bin, int nan_bin, int default_left,
-                                           int is_cat, unsigned long long cat_mask) {
+                                           int is_cat, unsigned long long m0,
+                                           unsigned long long m1, unsigned long long m2,
+                                           unsigned long long m3) {
   if (is_cat) {
-    if (cat_mask != 0) return (cat_mask >> b) & 1ull;  // sorted-subset split
-    return b == thr_bin ? 1 : 0;                       // one-hot split
+    if ((m0 | m1 | m2 | m3) != 0ull) {  // sorted-subset split (<=256 bins)
+      const unsigned long long w = b < 64 ? m0 : b < 128 ? m1 : b < 192 ? m2 : m3;
+      return (w >> (b & 63)) & 1ull;
+    }
+    return b == thr_bin ? 1 : 0;  // one-hot split
   }
   if (nan_bin >= 0 && b == nan_bin) return default_left;
   return b <= thr_bin ? 1 : 0;
@@ -1065,7 +1214,8 @@ __global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
   const int nan_bin = m.is_cat ? -1 : m.nan_bin;
   const int default_left = win->default_left;
   const int cat_onehot = m.is_cat;
-  const unsigned long long cmask = win->cat_mask;
+  const unsigned long long cm0 = win->cat_mask[0], cm1 = win->cat_mask[1];
+  const unsigned long long cm2 = win->cat_mask[2], cm3 = win->cat_mask[3];
   const int begin = leaf_begin[L];
   const int cnt = leaf_cnt[L];
   const uint32_t* idx = idx_base + begin;
@@ -1073,7 +1223,8 @@ __global__ void k_part_mark(const uint32_t* __restrict__ idx_base,
   int local = 0;
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt; i += chunk_stride) {
     const int b = colbins2[idx[i]];
-    const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot, cmask);
+    const int go =
+        part_decide(b, thr_bin, nan_bin, default_left, cat_onehot, cm0, cm1, cm2, cm3);
     marks[i] = static_cast<uint8_t>(go);
     local += go;
   }
@@ -1265,12 +1416,14 @@ __global__ void k_part_fused(const uint32_t* __restrict__ idx_base,
     const int nan_bin = m.is_cat ? -1 : m.nan_bin;
     const int default_left = win->default_left;
     const int cat_onehot = m.is_cat;
-    const unsigned long long cmask = win->cat_mask;
+    const unsigned long long cm0 = win->cat_mask[0], cm1 = win->cat_mask[1];
+    const unsigned long long cm2 = win->cat_mask[2], cm3 = win->cat_mask[3];
     const uint32_t* idx = idx_base + begin;
     int local = 0;
     for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt; i += chunk_stride) {
       const int b = colbins2[idx[i]];
-      const int go = part_decide(b, thr_bin, nan_bin, default_left, cat_onehot, cmask);
+      const int go =
+          part_decide(b, thr_bin, nan_bin, default_left, cat_onehot, cm0, cm1, cm2, cm3);
       marks[i] = static_cast<uint8_t>(go);
       local += go;
     }
@@ -2341,9 +2494,9 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   feat_meta_host_.resize(nf_);
   for (int f = 0; f < nf_; ++f) {
     const BinMapper* m = train_data->FeatureBinMapper(f);
-    if (m->bin_type() == BinType::kCategorical && m->num_bin() > 64) {
+    if (m->bin_type() == BinType::kCategorical && m->num_bin() > 256) {
       Log::Warning("HIP learner evaluates categorical feature %d with one-hot splits "
-                   "only (%d categories exceed the 64-bin device sorted-subset scan)",
+                   "only (%d categories exceed the 256-bin device sorted-subset scan)",
                    train_data->RealFeatureIndex(f), m->num_bin());
     }
     feat_meta_host_[f] = {static_cast<int>(train_data->hist_offset(f)), m->num_bin(),
@@ -3050,9 +3203,11 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
       // value-level bitset from the winner: 64-bit bin subset (sorted scan) or a
       // single one-hot bin
       std::vector<int> cats;
-      if (w.cat_mask != 0) {
-        for (int b = 0; b < 64; ++b)
-          if ((w.cat_mask >> b) & 1ull)
+      const bool has_mask =
+          (w.cat_mask[0] | w.cat_mask[1] | w.cat_mask[2] | w.cat_mask[3]) != 0ull;
+      if (has_mask) {
+        for (int b = 0; b < 256; ++b)
+          if ((w.cat_mask[b >> 6] >> (b & 63)) & 1ull)
             cats.push_back(static_cast<int>(mapper->BinToValue(b)));
       } else {
         cats.push_back(static_cast<int>(mapper->BinToValue(w.bin)));

@@ -608,3 +608,28 @@ def test_gpu_large_bins_uint16():
         h.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), ctypes.byref(err))
     assert rc == 0
     assert err.value < 2e-3, err.value
+
+
+def test_gpu_categorical_large_cardinality():
+    """Sorted-subset categorical splits beyond 64 bins on device (VERDICT r1 #6:
+    previously degraded to one-hot): a 200-category feature must reach CPU-parity
+    quality (the CPU oracle runs the same sorted scan)."""
+    rng = np.random.RandomState(1)
+    n = 80_000
+    ncat = 200
+    cat = rng.randint(0, ncat, size=n)
+    cat_effect = rng.randn(ncat) * 1.5
+    X = np.column_stack([cat.astype(np.float32),
+                         rng.randn(n).astype(np.float32)])
+    y = (cat_effect[cat] + 0.5 * X[:, 1] + 0.5 * rng.randn(n) > 0).astype(np.float32)
+    aucs = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "binary", "device_type": dev, "max_bin": 255,
+                  "num_leaves": 31, "min_data_in_leaf": 20, "verbosity": 0,
+                  "metric": "none", "max_cat_threshold": 64,
+                  "categorical_feature": [0]}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 15)
+        aucs[dev] = _auc(y[:20000], bst.predict(X[:20000]))
+    # a one-hot-only device would land far below the CPU's sorted-subset quality
+    assert aucs["gpu"] > 0.85, aucs
+    assert abs(aucs["cpu"] - aucs["gpu"]) < 0.01, aucs
