@@ -160,17 +160,19 @@ struct GpuComm {
    *  [off[r], off[r]+cnt[r]) slice; other slices are left partial (garbage).
    *  RCCL: grouped ncclReduce, one per root. Host seam: plain allreduce (a
    *  correct superset — owned slices are what the kernels read). */
-  void ReduceBlocks(float* d, const std::vector<size_t>& off, const std::vector<size_t>& cnt,
+  template <typename T>
+  void ReduceBlocks(T* d, const std::vector<size_t>& off, const std::vector<size_t>& cnt,
                     hipStream_t s) {
     if (rccl()) {
+      const ncclDataType_t dt = sizeof(T) == 8 ? ncclFloat64 : ncclFloat32;
       NCCL_OK(ncclGroupStart());
       for (int r = 0; r < world; ++r) {
         if (cnt[r] == 0) continue;
-        NCCL_OK(ncclReduce(d + off[r], d + off[r], cnt[r], ncclFloat32, ncclSum, r, comm, s));
+        NCCL_OK(ncclReduce(d + off[r], d + off[r], cnt[r], dt, ncclSum, r, comm, s));
       }
       NCCL_OK(ncclGroupEnd());
     } else {
-      HostBounce<float>(d, off.back() + cnt.back(), s);
+      HostBounce<T>(d, off.back() + cnt.back(), s);
     }
   }
   /*! byte allgather: world * bytes_per_rank into d_out (rank-major). */
@@ -473,7 +475,7 @@ __global__ void k_iota(uint32_t* p, int n) {
  *  NCOPIES lane-group LDS copies (lane % NCOPIES) dilute same-bin DS-atomic conflicts:
  *  a fully homogeneous leaf serializes 64-wide without them (measured 92us for a
  *  2048-row leaf), /NCOPIES with. Rows are read as 16B uint4 chunks. */
-template <int NCOPIES, typename BIN_T = uint8_t>
+template <int NCOPIES, typename BIN_T = uint8_t, typename HIST_T = float>
 __global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
                        const uint32_t* __restrict__ idx_base,
                        const int* __restrict__ leaf_begin, const int* __restrict__ leaf_cnt,
@@ -482,7 +484,7 @@ __global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
                        int leafB_from_counters, const float* __restrict__ g,
                        const float* __restrict__ h, const FeatMeta* __restrict__ fm,
                        int feat_begin, int feat_end, int part_bin_base, int part_bins,
-                       float* __restrict__ hist_base, size_t slot_stride) {
+                       HIST_T* __restrict__ hist_base, size_t slot_stride) {
   const int leafA = *leafA_ptr;
   if (leafA < 0) return;
   const int leafB = leafB_from_counters ? counters[0] - 1 : -1;
@@ -491,20 +493,22 @@ __global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
   const int begin = leaf_begin[leaf];
   const int cnt = leaf_cnt[leaf];
   const uint32_t* idx = idx_base + begin;
-  float* ghist = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
-                 static_cast<size_t>(part_bin_base) * 2;
+  HIST_T* ghist = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
+                  static_cast<size_t>(part_bin_base) * 2;
 
   // LDS layout [bin][NCOPIES][g,h] with a non-power-of-2 bin stride (2*NCOPIES+2):
   // copies of one bin sit in adjacent banks (same-bin leaves drain through NCOPIES
   // banks in parallel) and consecutive bins rotate across all 32 banks.
+  // HIST_T=double is the gpu_use_dp parity mode (fp64 accumulation end to end).
   constexpr int kBinStride = 2 * NCOPIES + 2;
-  extern __shared__ float lh[];  // part_bins * kBinStride
+  extern __shared__ char k_hist_smem[];
+  HIST_T* lh = reinterpret_cast<HIST_T*>(k_hist_smem);  // part_bins * kBinStride
   __shared__ int loff[256];
   const int nfeat = feat_end - feat_begin;
   for (int i = threadIdx.x; i < nfeat; i += blockDim.x)
     loff[i] = fm[feat_begin + i].bin_off - part_bin_base;
   const int nelem = part_bins * kBinStride;
-  for (int i = threadIdx.x; i < nelem; i += blockDim.x) lh[i] = 0.0f;
+  for (int i = threadIdx.x; i < nelem; i += blockDim.x) lh[i] = HIST_T(0);
   __syncthreads();
 
   const int my_copy = (threadIdx.x % NCOPIES) * 2;
@@ -527,9 +531,9 @@ __global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
           const int f = c + j;
           if (f < feat_begin || f >= feat_end) continue;
           const int b = (w[j >> 2] >> ((j & 3) * 8)) & 0xFF;
-          float* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
-          atomicAdd(dst, gi);
-          atomicAdd(dst + 1, hi);
+          HIST_T* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
+          atomicAdd(dst, HIST_T(gi));
+          atomicAdd(dst + 1, HIST_T(hi));
         }
       }
     } else {
@@ -537,9 +541,9 @@ __global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
       // parity: reference CUDAConstructHistogramDenseKernel 16-bit bins
       for (int f = feat_begin; f < feat_end; ++f) {
         const int b = rp[f];
-        float* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
-        atomicAdd(dst, gi);
-        atomicAdd(dst + 1, hi);
+        HIST_T* dst = lh + (loff[f - feat_begin] + b) * kBinStride + my_copy;
+        atomicAdd(dst, HIST_T(gi));
+        atomicAdd(dst + 1, HIST_T(hi));
       }
     }
   }
@@ -547,25 +551,27 @@ __global__ void k_hist(const BIN_T* __restrict__ rows, int stride,
   for (int i = threadIdx.x; i < part_bins * 2; i += blockDim.x) {
     const int bin = i >> 1;
     const int gh = i & 1;
-    float v = 0.0f;
+    HIST_T v = HIST_T(0);
 #pragma unroll
     for (int cpy = 0; cpy < NCOPIES; ++cpy) v += lh[bin * kBinStride + cpy * 2 + gh];
-    if (v != 0.0f) atomicAdd(&ghist[i], v);
+    if (v != HIST_T(0)) atomicAdd(&ghist[i], v);
   }
 }
 
-__global__ void k_hist_zero(float* hist_base, size_t slot_stride,
+template <typename HIST_T = float>
+__global__ void k_hist_zero(HIST_T* hist_base, size_t slot_stride,
                             const int* __restrict__ counters, int slot_from_counters,
                             int literal_slot, const int* __restrict__ leafA_ptr, int n) {
   if (*leafA_ptr < 0) return;
   const int slot = slot_from_counters ? counters[0] - 1 : literal_slot;
-  float* hist = hist_base + static_cast<size_t>(slot) * slot_stride;
+  HIST_T* hist = hist_base + static_cast<size_t>(slot) * slot_stride;
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
-  for (int i = tid; i < n; i += blockDim.x * gridDim.x) hist[i] = 0.0f;
+  for (int i = tid; i < n; i += blockDim.x * gridDim.x) hist[i] = HIST_T(0);
 }
 
 /*! larger-child histogram = parent (in place at the larger leaf's slot) - smaller. */
-__global__ void k_hist_subtract(float* __restrict__ hist_base, size_t slot_stride,
+template <typename HIST_T = float>
+__global__ void k_hist_subtract(HIST_T* __restrict__ hist_base, size_t slot_stride,
                                 const int* __restrict__ leaf_slot,
                                 const LeafStat* __restrict__ stats,
                                 const int* __restrict__ Lptr,
@@ -575,8 +581,8 @@ __global__ void k_hist_subtract(float* __restrict__ hist_base, size_t slot_strid
   const int R = counters[0] - 1;
   const int smaller = stats[L].cnt <= stats[R].cnt ? L : R;
   const int larger = smaller == L ? R : L;
-  float* big = hist_base + static_cast<size_t>(leaf_slot[larger]) * slot_stride;
-  const float* small = hist_base + static_cast<size_t>(leaf_slot[smaller]) * slot_stride;
+  HIST_T* big = hist_base + static_cast<size_t>(leaf_slot[larger]) * slot_stride;
+  const HIST_T* small = hist_base + static_cast<size_t>(leaf_slot[smaller]) * slot_stride;
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
   for (int i = tid; i < n; i += blockDim.x * gridDim.x) big[i] -= small[i];
 }
@@ -642,8 +648,9 @@ __global__ void k_set_root_global_cnt(LeafStat* stats, const int64_t* gbuf,
 
 // ------------------------------------------------------------------ best split
 /*! one wave per (feature, child); blockIdx.y selects leafA/leafB. */
+template <typename HIST_T = float>
 __global__ void __launch_bounds__(64) k_best_feat(
-    const float* __restrict__ hist_base, size_t slot_stride, const int* __restrict__ leaf_slot,
+    const HIST_T* __restrict__ hist_base, size_t slot_stride, const int* __restrict__ leaf_slot,
     const FeatMeta* __restrict__ fm, int nf, const LeafStat* __restrict__ stats,
     const int* __restrict__ leafA_ptr, const int* __restrict__ counters,
     int leafB_from_counters, GainParams p, const int8_t* __restrict__ feat_mask,
@@ -697,8 +704,8 @@ __global__ void __launch_bounds__(64) k_best_feat(
   const double parent_gain = d_leaf_gain_sm(sum_g, sum_h, num_data, leaf_parent_out, p);
   const double min_gain_shift = parent_gain + p.min_gain_to_split;
 
-  const float* fh = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
-                    static_cast<size_t>(m.bin_off) * 2;
+  const HIST_T* fh = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
+                     static_cast<size_t>(m.bin_off) * 2;
   double g_nan = 0, h_nan = 0;
   const bool has_nan = m.nan_bin >= 0;
   if (has_nan) {
@@ -1582,11 +1589,12 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
 /*! device-side split bookkeeping: segments, stats, slot map, split log. Thread 0
  *  does the bookkeeping; the whole block then zeroes the spare histogram slot the
  *  upcoming smaller-child build will accumulate into (fused k_hist_zero). */
+template <typename HIST_T = float>
 __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafStat* stats,
                            const SplitRec* __restrict__ winner,
                            const int* __restrict__ Lptr, int* counters,
                            LogEntry* __restrict__ log, const int* __restrict__ ctr,
-                           const int64_t* __restrict__ gbuf, float* hist_base,
+                           const int64_t* __restrict__ gbuf, HIST_T* hist_base,
                            size_t slot_stride, int n_elem,
                            const int8_t* __restrict__ mono, double* leaf_bounds,
                            unsigned long long* leaf_branch) {
@@ -1604,8 +1612,8 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
   }
   __syncthreads();
   if (s_spare < 0) return;
-  float* hist = hist_base + static_cast<size_t>(s_spare) * slot_stride;
-  for (int i = threadIdx.x; i < n_elem; i += blockDim.x) hist[i] = 0.0f;
+  HIST_T* hist = hist_base + static_cast<size_t>(s_spare) * slot_stride;
+  for (int i = threadIdx.x; i < n_elem; i += blockDim.x) hist[i] = HIST_T(0);
 }
 
 /*! single-thread bookkeeping body (called from k_finalize thread 0). */
@@ -2416,6 +2424,7 @@ class HIPTreeLearner : public TreeLearner {
 
   bool grads_on_device_ = false;
   bool rows16_ = false;   // uint16 bins (max_bin > 256)
+  bool hist_dp_ = false;  // gpu_use_dp: fp64 histogram accumulation end to end
   bool quantized_ = false;
   bool coop_launch_ = false;   // fused cooperative partition kernel available
   bool use_mono_ = false;      // monotone constraints active (bounds tracked on device)
@@ -2508,6 +2517,7 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
                    hipMemcpyHostToDevice));
   rows16_ = false;  // must be known before LDS partition planning
   for (int f = 0; f < nf_; ++f) rows16_ = rows16_ || feat_meta_host_[f].num_bin > 256;
+  hist_dp_ = config_->gpu_use_dp;  // fp64 LDS doubles the partition bin cost
 
   // LDS feature partitioning with privatized copies; partitions are 16-feature aligned
   // so row bytes load as whole uint4 chunks. Shrink the copy count if bins are too many.
@@ -2515,7 +2525,8 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   auto build_partitions = [&](int copies) -> bool {
     feat_partitions_.clear();
     part_bin_range_.clear();
-    const int max_bins = LdsBudget() / ((2 * copies + 2) * sizeof(float));
+    const int max_bins =
+        LdsBudget() / ((2 * copies + 2) * (hist_dp_ ? sizeof(double) : sizeof(float)));
     int begin = 0;
     while (begin < nf_) {
       int end = begin;
@@ -2563,6 +2574,14 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
                             reinterpret_cast<const void*>(&hipk::k_hist<4, uint16_t>),
                             reinterpret_cast<const void*>(&hipk::k_hist<2, uint16_t>),
                             reinterpret_cast<const void*>(&hipk::k_hist<1, uint16_t>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<8, uint8_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<4, uint8_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<2, uint8_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<1, uint8_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<8, uint16_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<4, uint16_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<2, uint16_t, double>),
+                            reinterpret_cast<const void*>(&hipk::k_hist<1, uint16_t, double>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<4>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<2>),
                             reinterpret_cast<const void*>(&hipk::k_hist_q<1>)}) {
@@ -2627,6 +2646,11 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     coop_launch_ = coop != 0 && getenv("MIGBM_COOP_PARTITION") != nullptr && !rows16_;
   }
   quantized_ = config_->use_quantized_grad;
+  if (quantized_ && hist_dp_) {
+    Log::Warning("gpu_use_dp with use_quantized_grad runs unquantized (the packed-int "
+                 "histogram path is fp32-flush only)");
+    quantized_ = false;
+  }
   if (quantized_ && rows16_) {
     Log::Warning("use_quantized_grad with max_bin>256 runs unquantized on GPU");
     quantized_ = false;
@@ -2658,7 +2682,7 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_ctr_.Alloc(2);
   d_gbuf_.Alloc(2);
   const int nl = config_->num_leaves;
-  d_hist_.Alloc(static_cast<size_t>(nl) * total_bins_ * 2);
+  d_hist_.Alloc(static_cast<size_t>(nl) * total_bins_ * 2 * (hist_dp_ ? 2 : 1));
   d_feat_best_.Alloc(static_cast<size_t>(2) * nf_);
   d_leaf_best_.Alloc(nl);
   d_winner_.Alloc(2);  // rec + trailing winner-leaf int
@@ -2846,9 +2870,14 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
   // zero the spare slot (literal slot 0 for the root). In the split loop the spare
   // was already zeroed by the fused k_finalize, so the launch is skipped.
   if (zero_spare) {
-    hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
-                       slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
-                       n_elem);
+    if (hist_dp_)
+      hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_,
+                         reinterpret_cast<double*>(d_hist_.ptr), slot_stride,
+                         d_counters_.ptr, leafB_from_counters, 0, leafA_ptr, n_elem);
+    else
+      hipLaunchKernelGGL(hipk::k_hist_zero, dim3(16), dim3(256), 0, stream_, d_hist_.ptr,
+                         slot_stride, d_counters_.ptr, leafB_from_counters, 0, leafA_ptr,
+                         n_elem);
   }
   for (size_t pr = 0; pr < feat_partitions_.size(); ++pr) {
     const auto [fb, fe] = feat_partitions_[pr];
@@ -2885,60 +2914,73 @@ void HIPTreeLearner::LaunchHist(const int* leafA_ptr, int leafB_from_counters,
       }
       continue;
     }
-    const size_t lds = static_cast<size_t>(bins) * (2 * n_copies_ + 2) * sizeof(float);
-    auto launch_hist = [&](auto bin_tag) {
+    const size_t lds = static_cast<size_t>(bins) * (2 * n_copies_ + 2) *
+                       (hist_dp_ ? sizeof(double) : sizeof(float));
+    auto launch_hist = [&](auto bin_tag, auto hist_tag) {
       using BIN_T = decltype(bin_tag);
+      using HIST_T = decltype(hist_tag);
       const BIN_T* rp = reinterpret_cast<const BIN_T*>(d_rows_.ptr);
+      HIST_T* hb = reinterpret_cast<HIST_T*>(d_hist_.ptr);
       switch (n_copies_) {
         case 8:
-          hipLaunchKernelGGL((hipk::k_hist<8, BIN_T>), dim3(blocks), dim3(HistThreads()),
+          hipLaunchKernelGGL((hipk::k_hist<8, BIN_T, HIST_T>), dim3(blocks), dim3(HistThreads()),
                              lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                              d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
                              leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
                              HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
-                             d_hist_.ptr, slot_stride);
+                             hb, slot_stride);
           break;
         case 4:
-          hipLaunchKernelGGL((hipk::k_hist<4, BIN_T>), dim3(blocks), dim3(HistThreads()),
+          hipLaunchKernelGGL((hipk::k_hist<4, BIN_T, HIST_T>), dim3(blocks), dim3(HistThreads()),
                              lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                              d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
                              leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
                              HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
-                             d_hist_.ptr, slot_stride);
+                             hb, slot_stride);
           break;
         case 2:
-          hipLaunchKernelGGL((hipk::k_hist<2, BIN_T>), dim3(blocks), dim3(HistThreads()),
+          hipLaunchKernelGGL((hipk::k_hist<2, BIN_T, HIST_T>), dim3(blocks), dim3(HistThreads()),
                              lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                              d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
                              leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
                              HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
-                             d_hist_.ptr, slot_stride);
+                             hb, slot_stride);
           break;
         default:
-          hipLaunchKernelGGL((hipk::k_hist<1, BIN_T>), dim3(blocks), dim3(HistThreads()),
+          hipLaunchKernelGGL((hipk::k_hist<1, BIN_T, HIST_T>), dim3(blocks), dim3(HistThreads()),
                              lds, stream_, rp, row_stride_, d_idx_.ptr, d_leaf_begin_.ptr,
                              d_leaf_cnt_.ptr, d_leaf_stats_.ptr, d_leaf_slot_.ptr,
                              leafA_ptr, d_counters_.ptr, leafB_from_counters, GradPtr(),
                              HessPtr(), d_feat_meta_.ptr, fb, fe, bin_base, bins,
-                             d_hist_.ptr, slot_stride);
+                             hb, slot_stride);
       }
     };
-    if (rows16_) launch_hist(uint16_t{});
-    else launch_hist(uint8_t{});
+    if (rows16_ && hist_dp_) launch_hist(uint16_t{}, double{});
+    else if (rows16_) launch_hist(uint16_t{}, float{});
+    else if (hist_dp_) launch_hist(uint8_t{}, double{});
+    else launch_hist(uint8_t{}, float{});
   }
 }
 
 void HIPTreeLearner::ReduceSpareHist(int spare_slot) {
   auto& comm = Comm();
   if (!dist_) return;
-  float* spare = d_hist_.ptr + static_cast<size_t>(spare_slot) * total_bins_ * 2;
+  const size_t n = static_cast<size_t>(total_bins_) * 2;
+  if (hist_dp_) {
+    double* spare = reinterpret_cast<double*>(d_hist_.ptr) +
+                    static_cast<size_t>(spare_slot) * n;
+    if (own_scan_) comm.ReduceBlocks(spare, own_off_, own_cnt_, stream_);
+    else comm.AllReduce(spare, n, stream_);
+    return;
+  }
+  float* spare = d_hist_.ptr + static_cast<size_t>(spare_slot) * n;
   if (own_scan_) {
     // reduce-scatter analogue: rank r receives the global sum of its owned
     // feature block only (matches the reference data-parallel learner's
     // ReduceScatter + owned-feature gain scan, data_parallel_tree_learner.cpp:283-450)
     comm.ReduceBlocks(spare, own_off_, own_cnt_, stream_);
   } else {
-    comm.AllReduce(spare, static_cast<size_t>(total_bins_) * 2, stream_);
+    comm.AllReduce(spare, n, stream_);
   }
 }
 
@@ -3006,13 +3048,25 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.own_fe = own_scan_ ? own_fe_ : nf_;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
   const int ny = leafB_from_counters ? 2 : 1;
-  hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_, d_hist_.ptr,
-                     slot_stride, d_leaf_slot_.ptr, d_feat_meta_.ptr, nf_,
-                     d_leaf_stats_.ptr, leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
-                     feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
-                     use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
-                     n_interaction_groups_ > 0 ? d_group_masks_.ptr : nullptr,
-                     d_leaf_branch_.ptr, d_feat_best_.ptr);
+  if (hist_dp_) {
+    hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_,
+                       reinterpret_cast<const double*>(d_hist_.ptr), slot_stride,
+                       d_leaf_slot_.ptr, d_feat_meta_.ptr, nf_, d_leaf_stats_.ptr,
+                       leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
+                       feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
+                       use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
+                       n_interaction_groups_ > 0 ? d_group_masks_.ptr : nullptr,
+                       d_leaf_branch_.ptr, d_feat_best_.ptr);
+  } else {
+    hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_,
+                       const_cast<const float*>(d_hist_.ptr), slot_stride,
+                       d_leaf_slot_.ptr, d_feat_meta_.ptr, nf_, d_leaf_stats_.ptr,
+                       leafA_ptr, d_counters_.ptr, leafB_from_counters, p,
+                       feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
+                       use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
+                       n_interaction_groups_ > 0 ? d_group_masks_.ptr : nullptr,
+                       d_leaf_branch_.ptr, d_feat_best_.ptr);
+  }
   (void)ny;
   hipLaunchKernelGGL(hipk::k_best_leaf_overall, dim3(1), dim3(256), 0, stream_,
                      d_feat_best_.ptr, nf_, d_leaf_best_.ptr, leafA_ptr, d_counters_.ptr,
@@ -3162,20 +3216,37 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                          d_winner_leaf_.ptr, d_ctr_.ptr, d_gbuf_.ptr);
     }
     if (dist_) comm.AllReduce(d_gbuf_.ptr, 1, stream_);
-    hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(256), 0, stream_, d_leaf_begin_.ptr,
-                       d_leaf_cnt_.ptr, d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_.ptr,
-                       d_winner_leaf_.ptr, d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr,
-                       d_gbuf_.ptr, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
-                       total_bins_ * 2, use_mono_ ? d_mono_.ptr : nullptr,
-                       d_leaf_bounds_.ptr, d_leaf_branch_.ptr);
+    if (hist_dp_)
+      hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(256), 0, stream_,
+                         d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_leaf_slot_.ptr,
+                         d_leaf_stats_.ptr, d_winner_.ptr, d_winner_leaf_.ptr,
+                         d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr, d_gbuf_.ptr,
+                         reinterpret_cast<double*>(d_hist_.ptr),
+                         static_cast<size_t>(total_bins_) * 2, total_bins_ * 2,
+                         use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
+                         d_leaf_branch_.ptr);
+    else
+      hipLaunchKernelGGL(hipk::k_finalize, dim3(1), dim3(256), 0, stream_,
+                         d_leaf_begin_.ptr, d_leaf_cnt_.ptr, d_leaf_slot_.ptr,
+                         d_leaf_stats_.ptr, d_winner_.ptr, d_winner_leaf_.ptr,
+                         d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr, d_gbuf_.ptr,
+                         d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
+                         total_bins_ * 2, use_mono_ ? d_mono_.ptr : nullptr,
+                         d_leaf_bounds_.ptr, d_leaf_branch_.ptr);
     LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks, /*zero_spare=*/false);
     ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
     {
       const int n_elem = total_bins_ * 2;
-      hipLaunchKernelGGL(hipk::k_hist_subtract, dim3((n_elem + 1023) / 1024), dim3(256), 0,
-                         stream_, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
-                         d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_leaf_.ptr,
-                         d_counters_.ptr, n_elem);
+      if (hist_dp_)
+        hipLaunchKernelGGL(hipk::k_hist_subtract, dim3((n_elem + 1023) / 1024), dim3(256),
+                           0, stream_, reinterpret_cast<double*>(d_hist_.ptr),
+                           static_cast<size_t>(total_bins_) * 2, d_leaf_slot_.ptr,
+                           d_leaf_stats_.ptr, d_winner_leaf_.ptr, d_counters_.ptr, n_elem);
+      else
+        hipLaunchKernelGGL(hipk::k_hist_subtract, dim3((n_elem + 1023) / 1024), dim3(256),
+                           0, stream_, d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
+                           d_leaf_slot_.ptr, d_leaf_stats_.ptr, d_winner_leaf_.ptr,
+                           d_counters_.ptr, n_elem);
     }
     LaunchBestSplit(d_winner_leaf_.ptr, 1);
     if (own_scan_) SyncGlobalWinner();
@@ -3429,9 +3500,16 @@ double HIPTreeLearner::DebugRootHistMaxRelErr(const score_t* g, const score_t* h
   }
   LaunchHist(d_root_leaf_.ptr, 0, HistBlocksFor(static_cast<int>(used_cnt_)));
   HIP_OK(hipStreamSynchronize(stream_));
-  std::vector<float> dev(static_cast<size_t>(total_bins_) * 2);
-  HIP_OK(hipMemcpy(dev.data(), d_hist_.ptr, sizeof(float) * dev.size(),
-                   hipMemcpyDeviceToHost));
+  std::vector<double> dev(static_cast<size_t>(total_bins_) * 2);
+  if (hist_dp_) {
+    HIP_OK(hipMemcpy(dev.data(), d_hist_.ptr, sizeof(double) * dev.size(),
+                     hipMemcpyDeviceToHost));
+  } else {
+    std::vector<float> devf(dev.size());
+    HIP_OK(hipMemcpy(devf.data(), d_hist_.ptr, sizeof(float) * devf.size(),
+                     hipMemcpyDeviceToHost));
+    for (size_t i = 0; i < dev.size(); ++i) dev[i] = devf[i];
+  }
   // fp64 host oracle over the identical rows/gradients
   std::vector<double> host(dev.size(), 0.0);
   std::vector<int8_t> used(train_data_->num_features(), 1);

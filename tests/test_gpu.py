@@ -633,3 +633,40 @@ def test_gpu_categorical_large_cardinality():
     # a one-hot-only device would land far below the CPU's sorted-subset quality
     assert aucs["gpu"] > 0.85, aucs
     assert abs(aucs["cpu"] - aucs["gpu"]) < 0.01, aucs
+
+
+def test_gpu_use_dp_fp64_histograms():
+    """gpu_use_dp=true: fp64 histogram accumulation end to end (reference dp
+    parity mode). The device histogram then matches the
+    fp64 host oracle to ~1e-12 instead of the fp32 ~1e-3, and training quality
+    matches the sp mode."""
+    import ctypes
+    from lightgbm_amd.basic import _LIB, _c_str
+    rng = np.random.RandomState(2)
+    n = 100_000
+    X = rng.randn(n, 12).astype(np.float32)
+    y = (1.2 * X[:, 0] - 0.8 * X[:, 1] + X[:, 2] * X[:, 3] +
+         rng.randn(n) > 0).astype(np.float32)
+    ds = lgb.Dataset(X, label=y, params={"max_bin": 63}).construct()
+    g = rng.randn(n).astype(np.float32)
+    h = rng.uniform(0.5, 2.0, n).astype(np.float32)
+    errs = {}
+    for dp in ("false", "true"):
+        err = ctypes.c_double(1e9)
+        rc = _LIB.MIGBM_DebugDeviceRootHist(
+            ds._handle, _c_str(f"max_bin=63 num_leaves=31 gpu_use_dp={dp}"),
+            g.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            h.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), ctypes.byref(err))
+        assert rc == 0
+        errs[dp] = err.value
+    assert errs["true"] < 1e-10, errs   # fp64 = oracle precision
+    assert errs["false"] < 2e-3, errs   # fp32 stays within the sp envelope
+    aucs = {}
+    for dp in (False, True):
+        params = {"objective": "binary", "device_type": "gpu", "max_bin": 63,
+                  "num_leaves": 63, "verbosity": 0, "gpu_use_dp": dp,
+                  "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 15)
+        aucs[dp] = _auc(y[:20000], bst.predict(X[:20000]))
+    assert aucs[True] > 0.8
+    assert abs(aucs[True] - aucs[False]) < 2e-3, aucs
