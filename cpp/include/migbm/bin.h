@@ -31,7 +31,8 @@ class BinMapper {
    */
   void FindBin(double* values, int num_sample_values, size_t total_sample_cnt, int max_bin,
                int min_data_in_bin, int min_split_data, bool pre_filter, BinType bin_type,
-               bool use_missing, bool zero_as_missing);
+               bool use_missing, bool zero_as_missing,
+               const std::vector<double>* forced_bounds = nullptr);
 
   /*! Map a raw value to its bin. */
   inline uint32_t ValueToBin(double value) const {

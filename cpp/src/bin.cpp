@@ -68,7 +68,7 @@ std::vector<double> GreedyFindBin(const std::vector<double>& dv, const std::vect
 void BinMapper::FindBin(double* values, int num_sample_values, size_t total_sample_cnt,
                         int max_bin, int min_data_in_bin, int /*min_split_data*/,
                         bool pre_filter, BinType bin_type, bool use_missing,
-                        bool zero_as_missing) {
+                        bool zero_as_missing, const std::vector<double>* forced_bounds) {
   bin_type_ = bin_type;
   // split NaN out
   int na_cnt = 0;
@@ -159,6 +159,48 @@ void BinMapper::FindBin(double* values, int num_sample_values, size_t total_samp
   size_t eff_total = 0;
   for (int c : cnt) eff_total += c;
   bin_upper_bound_ = GreedyFindBin(dv, cnt, usable_bins, eff_total, min_data_in_bin);
+  if (forced_bounds != nullptr && !forced_bounds->empty()) {
+    // forcedbins_filename (reference DatasetLoader ForceBins parity): the forced
+    // upper bounds are kept exactly; quantile boundaries fill the remaining
+    // budget, dropping the ones with the smallest neighbour gap when over it
+    std::vector<std::pair<double, bool>> merged;  // (bound, is_forced)
+    for (double b : bin_upper_bound_)
+      if (std::isfinite(b)) merged.push_back({b, false});
+    for (double b : *forced_bounds)
+      if (std::isfinite(b)) merged.push_back({b, true});
+    std::sort(merged.begin(), merged.end());
+    // dedup: forced wins
+    std::vector<std::pair<double, bool>> uniq;
+    for (auto& e : merged) {
+      if (!uniq.empty() && uniq.back().first == e.first)
+        uniq.back().second = uniq.back().second || e.second;
+      else
+        uniq.push_back(e);
+    }
+    const int budget = std::max(1, usable_bins - 1);  // +inf terminator re-added below
+    while (static_cast<int>(uniq.size()) > budget) {
+      int victim = -1;
+      double best_gap = std::numeric_limits<double>::infinity();
+      for (size_t i = 0; i < uniq.size(); ++i) {
+        if (uniq[i].second) continue;  // never drop a forced bound
+        const double lo2 = i == 0 ? -std::numeric_limits<double>::infinity()
+                                  : uniq[i - 1].first;
+        const double hi2 = i + 1 < uniq.size()
+                               ? uniq[i + 1].first
+                               : std::numeric_limits<double>::infinity();
+        const double gap = std::min(uniq[i].first - lo2, hi2 - uniq[i].first);
+        if (gap < best_gap) {
+          best_gap = gap;
+          victim = static_cast<int>(i);
+        }
+      }
+      if (victim < 0) break;  // all forced: honor them even over budget
+      uniq.erase(uniq.begin() + victim);
+    }
+    bin_upper_bound_.clear();
+    for (auto& e : uniq) bin_upper_bound_.push_back(e.first);
+    bin_upper_bound_.push_back(std::numeric_limits<double>::infinity());
+  }
   num_numeric_bin_ = static_cast<int>(bin_upper_bound_.size());
   if (num_numeric_bin_ == 0) {
     bin_upper_bound_.push_back(std::numeric_limits<double>::infinity());

@@ -69,6 +69,59 @@ void Dataset::FinishBinMappers(const Config&) {
   num_total_bin_ = static_cast<int>(off);
 }
 
+/*! parse forcedbins_filename JSON: [{"feature": i, "bin_upper_bound": [v, ...]}, ...]
+ *  (reference DatasetLoader forced-bins format). */
+static std::unordered_map<int, std::vector<double>> ParseForcedBins(const std::string& path) {
+  std::unordered_map<int, std::vector<double>> out;
+  if (path.empty()) return out;
+  FILE* fp = fopen(path.c_str(), "rb");
+  if (!fp) {
+    Log::Warning("Cannot open forced bins file %s", path.c_str());
+    return out;
+  }
+  std::string content;
+  char buf[4096];
+  size_t got;
+  while ((got = fread(buf, 1, sizeof(buf), fp)) > 0) content.append(buf, got);
+  fclose(fp);
+  size_t pos = 0;
+  auto skip = [&]() {
+    while (pos < content.size() &&
+           (isspace(static_cast<unsigned char>(content[pos])) || content[pos] == ','))
+      ++pos;
+  };
+  while (true) {
+    size_t obj = content.find('{', pos);
+    if (obj == std::string::npos) break;
+    size_t end = content.find('}', obj);
+    if (end == std::string::npos) break;
+    std::string o = content.substr(obj, end - obj + 1);
+    int feat = -1;
+    std::vector<double> bounds;
+    size_t fpos = o.find("\"feature\"");
+    if (fpos != std::string::npos) {
+      fpos = o.find(':', fpos);
+      if (fpos != std::string::npos) feat = atoi(o.c_str() + fpos + 1);
+    }
+    size_t bpos = o.find("\"bin_upper_bound\"");
+    if (bpos != std::string::npos) {
+      size_t lb = o.find('[', bpos);
+      size_t rb = o.find(']', bpos);
+      if (lb != std::string::npos && rb != std::string::npos) {
+        std::string arr = o.substr(lb + 1, rb - lb - 1);
+        for (auto& tok : Common::Split(arr.c_str(), ',')) {
+          auto t = Common::Trim(tok);
+          if (!t.empty()) bounds.push_back(atof(t.c_str()));
+        }
+      }
+    }
+    if (feat >= 0 && !bounds.empty()) out[feat] = std::move(bounds);
+    pos = end + 1;
+    skip();
+  }
+  return out;
+}
+
 void Dataset::ConstructFromMat(const std::function<double(data_size_t, int)>& get,
                                data_size_t nrow, int ncol, const Config& cfg,
                                const std::vector<int8_t>& categorical) {
@@ -140,6 +193,7 @@ void Dataset::ConstructFromMatLocal(const std::function<double(data_size_t, int)
     }
   }
   // 2. find bins per column (parallel)
+  const auto forced_bins = ParseForcedBins(cfg.forcedbins_filename);
   std::vector<std::unique_ptr<BinMapper>> mappers(ncol);
   const int ns = static_cast<int>(sample_idx.size());
 #pragma omp parallel for schedule(dynamic, 1)
@@ -147,10 +201,12 @@ void Dataset::ConstructFromMatLocal(const std::function<double(data_size_t, int)
     std::vector<double> vals(ns);
     for (int i = 0; i < ns; ++i) vals[i] = get(sample_idx[i], c);
     auto m = std::make_unique<BinMapper>();
+    auto fb_it = forced_bins.find(c);
     m->FindBin(vals.data(), ns, ns, cfg.max_bin, cfg.min_data_in_bin, 0,
                cfg.feature_pre_filter, categorical_flags_[c] ? BinType::kCategorical
                                                              : BinType::kNumerical,
-               cfg.use_missing, cfg.zero_as_missing);
+               cfg.use_missing, cfg.zero_as_missing,
+               fb_it != forced_bins.end() ? &fb_it->second : nullptr);
     mappers[c] = std::move(m);
   }
   // 3. keep non-trivial features
@@ -955,15 +1011,18 @@ void Dataset::ConstructFromSampleData(double** sample_values, int** sample_indic
   categorical_flags_ = categorical;
   if (categorical_flags_.empty()) categorical_flags_.assign(ncol, 0);
   (void)sample_indices;  // dense binning: implied zeros come from total_sample_cnt
+  const auto forced_bins = ParseForcedBins(cfg.forcedbins_filename);
   std::vector<std::unique_ptr<BinMapper>> mappers(ncol);
 #pragma omp parallel for schedule(dynamic, 1)
   for (int c = 0; c < ncol; ++c) {
     std::vector<double> vals(sample_values[c], sample_values[c] + num_per_col[c]);
     auto m = std::make_unique<BinMapper>();
+    auto fb_it = forced_bins.find(c);
     m->FindBin(vals.data(), num_per_col[c], num_sample_row, cfg.max_bin,
                cfg.min_data_in_bin, 0, cfg.feature_pre_filter,
                categorical_flags_[c] ? BinType::kCategorical : BinType::kNumerical,
-               cfg.use_missing, cfg.zero_as_missing);
+               cfg.use_missing, cfg.zero_as_missing,
+               fb_it != forced_bins.end() ? &fb_it->second : nullptr);
     mappers[c] = std::move(m);
   }
   used_feature_map_.assign(ncol, -1);

@@ -944,3 +944,39 @@ def test_monotone_intermediate_policy(method):
     mse_b = float(np.mean((preds["basic"].predict(X) - y) ** 2))
     mse_i = float(np.mean((preds[method].predict(X) - y) ** 2))
     assert mse_i < mse_b * 1.05, (mse_b, mse_i)
+
+
+def test_forced_bins(tmp_path):
+    """forcedbins_filename: forced bin upper bounds are honored exactly
+    (reference test_engine.py test_forced_bins semantics)."""
+    import json as _json
+    rng = np.random.RandomState(0)
+    n = 5000
+    X = rng.rand(n, 2)
+    y = (10 * (X[:, 0] > 0.33) + 5 * (X[:, 1] > 0.66) +
+         0.1 * rng.randn(n)).astype(np.float32)
+    fb = tmp_path / "forced_bins.json"
+    fb.write_text(_json.dumps([
+        {"feature": 0, "bin_upper_bound": [0.33, 0.54]},
+        {"feature": 1, "bin_upper_bound": [0.66]},
+    ]))
+    params = {"objective": "regression", "verbosity": -1, "num_leaves": 8,
+              "max_bin": 6, "forcedbins_filename": str(fb)}
+    bst = lgb.train(params, lgb.Dataset(X, label=y), 30)
+    # with only 6 coarse quantile bins the exact 0.33/0.66 steps would be missed;
+    # the forced boundaries make the model resolve them precisely
+    eps = 1e-4
+    p = bst.predict(np.array([[0.33 - eps, 0.5], [0.33 + eps, 0.5]]))
+    assert abs(p[1] - p[0]) > 5.0, p  # a ~10-unit jump exactly at 0.33
+    q = bst.predict(np.array([[0.5, 0.66 - eps], [0.5, 0.66 + eps]]))
+    assert abs(q[1] - q[0]) > 2.5, q
+    # and the split thresholds in the model sit exactly on the forced bounds
+    d = bst.dump_model()
+    thresholds = []
+    def walk(node):
+        if "split_feature" in node:
+            thresholds.append((node["split_feature"], node["threshold"]))
+            walk(node["left_child"]); walk(node["right_child"])
+    for t in d["tree_info"]:
+        walk(t["tree_structure"])
+    assert any(f == 0 and abs(thr - 0.33) < 1e-9 for f, thr in thresholds)
