@@ -181,6 +181,7 @@ class GBDT {
   std::vector<double> best_metric_;
   int best_iter_ = 0;
   int es_rounds_since_best_ = 0;
+  std::vector<int> es_counts_;
   // loaded-model state
   std::string loaded_parameter_;
   std::unique_ptr<const ObjectiveFunction> loaded_objective_;

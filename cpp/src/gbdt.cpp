@@ -464,23 +464,35 @@ void GBDT::MergeFrom(const GBDT* other) {
 
 bool GBDT::EvalAndCheckEarlyStopping() {
   if (config_->early_stopping_round <= 0 || valid_data_.empty()) return false;
-  // first metric of first valid set
   if (valid_metrics_.empty() || valid_metrics_[0].empty()) return false;
-  const Metric* m = valid_metrics_[0][0];
-  auto r = m->Eval(valid_score_[0].data(), objective_);
-  double v = r[0] * m->factor_to_bigger_better();
-  if (best_metric_.empty() || v < best_metric_[0] - config_->early_stopping_min_delta) {
-    best_metric_ = {v};
-    best_iter_ = iter_;
-    es_rounds_since_best_ = 0;
-  } else {
-    ++es_rounds_since_best_;
-    if (es_rounds_since_best_ >= config_->early_stopping_round) {
-      Log::Info("Early stopping at iteration %d, best iteration %d", iter_, best_iter_);
-      return true;
+  // reference semantics: each (valid set, metric) pair tracks its own best; a
+  // pair that fails to improve for early_stopping_round evals triggers the stop.
+  // first_metric_only restricts tracking to the first metric of each valid set.
+  size_t slot = 0;
+  bool stop = false;
+  for (size_t vd = 0; vd < valid_data_.size(); ++vd) {
+    for (size_t mi = 0; mi < valid_metrics_[vd].size(); ++mi) {
+      if (config_->first_metric_only && mi > 0) continue;
+      const Metric* m = valid_metrics_[vd][mi];
+      auto r = m->Eval(valid_score_[vd].data(), objective_);
+      const double v = r[0] * m->factor_to_bigger_better();
+      if (best_metric_.size() <= slot) {
+        best_metric_.resize(slot + 1, std::numeric_limits<double>::infinity());
+        es_counts_.resize(slot + 1, 0);
+      }
+      if (v < best_metric_[slot] - config_->early_stopping_min_delta) {
+        best_metric_[slot] = v;
+        es_counts_[slot] = 0;
+        if (vd == 0 && mi == 0) best_iter_ = iter_;
+      } else {
+        if (++es_counts_[slot] >= config_->early_stopping_round) stop = true;
+      }
+      ++slot;
     }
   }
-  return false;
+  if (stop)
+    Log::Info("Early stopping at iteration %d, best iteration %d", iter_, best_iter_);
+  return stop;
 }
 
 std::string GBDT::OutputMetric(int iter) {
