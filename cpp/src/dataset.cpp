@@ -202,7 +202,11 @@ void Dataset::ConstructFromMatLocal(const std::function<double(data_size_t, int)
     for (int i = 0; i < ns; ++i) vals[i] = get(sample_idx[i], c);
     auto m = std::make_unique<BinMapper>();
     auto fb_it = forced_bins.find(c);
-    m->FindBin(vals.data(), ns, ns, cfg.max_bin, cfg.min_data_in_bin, 0,
+    const int mb = c < static_cast<int>(cfg.max_bin_by_feature.size()) &&
+                           cfg.max_bin_by_feature[c] > 1
+                       ? cfg.max_bin_by_feature[c]
+                       : cfg.max_bin;
+    m->FindBin(vals.data(), ns, ns, mb, cfg.min_data_in_bin, 0,
                cfg.feature_pre_filter, categorical_flags_[c] ? BinType::kCategorical
                                                              : BinType::kNumerical,
                cfg.use_missing, cfg.zero_as_missing,
@@ -1018,7 +1022,11 @@ void Dataset::ConstructFromSampleData(double** sample_values, int** sample_indic
     std::vector<double> vals(sample_values[c], sample_values[c] + num_per_col[c]);
     auto m = std::make_unique<BinMapper>();
     auto fb_it = forced_bins.find(c);
-    m->FindBin(vals.data(), num_per_col[c], num_sample_row, cfg.max_bin,
+    const int mb = c < static_cast<int>(cfg.max_bin_by_feature.size()) &&
+                           cfg.max_bin_by_feature[c] > 1
+                       ? cfg.max_bin_by_feature[c]
+                       : cfg.max_bin;
+    m->FindBin(vals.data(), num_per_col[c], num_sample_row, mb,
                cfg.min_data_in_bin, 0, cfg.feature_pre_filter,
                categorical_flags_[c] ? BinType::kCategorical : BinType::kNumerical,
                cfg.use_missing, cfg.zero_as_missing,

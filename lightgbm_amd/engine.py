@@ -57,7 +57,7 @@ def train(params, train_set, num_boost_round=100, valid_sets=None, valid_names=N
                        "use_missing", "zero_as_missing", "feature_pre_filter",
                        "linear_tree", "data_random_seed", "enable_bundle",
                        "max_conflict_rate", "categorical_feature",
-                       "forcedbins_filename", "precise_float_parser")
+                       "forcedbins_filename", "precise_float_parser", "max_bin_by_feature")
     if train_set._handle is None:
         for k in _DATASET_PARAMS:
             if k in params and k not in train_set.params:

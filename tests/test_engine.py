@@ -980,3 +980,18 @@ def test_forced_bins(tmp_path):
     for t in d["tree_info"]:
         walk(t["tree_structure"])
     assert any(f == 0 and abs(thr - 0.33) < 1e-9 for f, thr in thresholds)
+
+
+def test_max_bin_by_feature():
+    """max_bin_by_feature caps per-feature bin counts (reference parity)."""
+    rng = np.random.RandomState(0)
+    X = rng.rand(3000, 2)
+    y = (X[:, 0] + X[:, 1] + 0.1 * rng.randn(3000)).astype(np.float32)
+    ds = lgb.Dataset(X, label=y,
+                     params={"max_bin": 255, "max_bin_by_feature": [4, 255],
+                             "min_data_in_bin": 1}).construct()
+    assert ds.num_feature() == 2
+    nb0 = ds.feature_num_bin(0)
+    nb1 = ds.feature_num_bin(1)
+    assert nb0 <= 4, nb0
+    assert nb1 > 100, nb1
