@@ -995,3 +995,76 @@ def test_max_bin_by_feature():
     nb1 = ds.feature_num_bin(1)
     assert nb0 <= 4, nb0
     assert nb1 > 100, nb1
+
+
+def test_constant_features_all_objectives():
+    """Constant columns are pre-filtered but predictions keep working and the
+    model keeps the full feature surface (reference test_constant_features_*)."""
+    rng = np.random.RandomState(0)
+    X = np.column_stack([rng.randn(2000), np.full(2000, 3.0), rng.randn(2000)])
+    for objective, y in [
+        ("binary", (X[:, 0] > 0).astype(np.float32)),
+        ("regression", (X[:, 0] + 0.1 * rng.randn(2000)).astype(np.float32)),
+        ("multiclass", (np.abs(X[:, 0]).astype(int) % 3).astype(np.float32)),
+    ]:
+        params = {"objective": objective, "verbosity": -1, "num_leaves": 7}
+        if objective == "multiclass":
+            params["num_class"] = 3
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 5)
+        assert bst.num_feature() == 3
+        p = bst.predict(X[:11])
+        assert p.shape[0] == 11
+        # the constant column never splits
+        assert all(f != 1 for f, _ in _all_split_features(bst))
+
+
+def _all_split_features(bst):
+    out = []
+    def walk(node):
+        if "split_feature" in node:
+            out.append((node["split_feature"], node["threshold"]))
+            walk(node["left_child"]); walk(node["right_child"])
+    for t in bst.dump_model()["tree_info"]:
+        walk(t["tree_structure"])
+    return out
+
+
+def test_boost_from_average_with_single_leaf_trees():
+    """All-constant data: every tree is a stump and predictions equal the label
+    mean (reference test_boost_from_average_with_single_leaf_trees)."""
+    X = np.full((200, 2), 1.0)
+    y = np.full(200, 5.0, dtype=np.float32)
+    bst = lgb.train({"objective": "regression", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 3)
+    np.testing.assert_allclose(bst.predict(X[:5]), 5.0, rtol=1e-6)
+
+
+def test_predict_output_shapes():
+    """Prediction output shapes for regression/binary/multiclass and pred_leaf /
+    pred_contrib (reference test_predict_*_output_shape)."""
+    rng = np.random.RandomState(1)
+    X = rng.randn(800, 4)
+    yb = (X[:, 0] > 0).astype(np.float32)
+    ym = (np.abs(X[:, 1]).astype(int) % 3).astype(np.float32)
+    b = lgb.train({"objective": "binary", "verbosity": -1, "num_leaves": 7},
+                  lgb.Dataset(X, label=yb), 5)
+    assert b.predict(X[:9]).shape == (9,)
+    assert b.predict(X[:9], pred_leaf=True).shape == (9, 5)
+    assert b.predict(X[:9], pred_contrib=True).shape == (9, 5)  # nf + bias
+    m = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1,
+                   "num_leaves": 7}, lgb.Dataset(X, label=ym), 4)
+    assert m.predict(X[:9]).shape == (9, 3)
+    assert m.predict(X[:9], pred_leaf=True).shape == (9, 12)
+    assert m.predict(X[:9], pred_contrib=True).shape == (9, 15)
+
+
+def test_predict_with_start_iteration():
+    """start_iteration slices the tree range (reference parity): full prediction
+    equals start-slice + head-slice raw scores."""
+    X, y = _regression_data(n=4000)
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "num_leaves": 15},
+                    lgb.Dataset(X, label=y), 20)
+    full = bst.predict(X[:50], raw_score=True)
+    head = bst.predict(X[:50], raw_score=True, num_iteration=8)
+    tail = bst.predict(X[:50], raw_score=True, start_iteration=8, num_iteration=12)
+    np.testing.assert_allclose(head + tail, full, rtol=1e-9)
