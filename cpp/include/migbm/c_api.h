@@ -113,10 +113,14 @@ LIGHTGBM_C_EXPORT int LGBM_DatasetPushRowsByCSR(DatasetHandle dataset, const voi
                                                 int64_t num_col, int64_t start_row);
 LIGHTGBM_C_EXPORT int LGBM_DatasetMarkFinished(DatasetHandle dataset);
 LIGHTGBM_C_EXPORT int LGBM_DatasetSetWaitForManualFinish(DatasetHandle dataset, int wait);
+typedef void* ByteBufferHandle;
+/*! reference ABI: serialize the dataset schema into a library-owned ByteBuffer */
 LIGHTGBM_C_EXPORT int LGBM_DatasetSerializeReferenceToBinary(DatasetHandle handle,
-                                                             int64_t buffer_len,
-                                                             int64_t* out_len,
-                                                             char* out_buffer);
+                                                             ByteBufferHandle* out,
+                                                             int32_t* out_len);
+LIGHTGBM_C_EXPORT int LGBM_ByteBufferGetAt(ByteBufferHandle handle, int32_t index,
+                                           uint8_t* out_val);
+LIGHTGBM_C_EXPORT int LGBM_ByteBufferFree(ByteBufferHandle handle);
 LIGHTGBM_C_EXPORT int LGBM_DatasetCreateFromSerializedReference(
     const void* ref_buffer, int32_t ref_buffer_size, int64_t num_row,
     int32_t num_classes, const char* parameters, DatasetHandle* out);

@@ -609,12 +609,29 @@ int LGBM_DatasetSetWaitForManualFinish(DatasetHandle, int) {
   API_END();
 }
 
-int LGBM_DatasetSerializeReferenceToBinary(DatasetHandle handle, int64_t buffer_len,
-                                           int64_t* out_len, char* out_buffer) {
+int LGBM_DatasetSerializeReferenceToBinary(DatasetHandle handle, ByteBufferHandle* out,
+                                           int32_t* out_len) {
   API_BEGIN();
+  auto* buf = new std::vector<uint8_t>();
   auto s = static_cast<Dataset*>(handle)->SerializeReference();
-  *out_len = static_cast<int64_t>(s.size());
-  if (out_buffer != nullptr && buffer_len >= *out_len) memcpy(out_buffer, s.data(), s.size());
+  buf->assign(s.begin(), s.end());
+  *out = buf;
+  *out_len = static_cast<int32_t>(buf->size());
+  API_END();
+}
+
+int LGBM_ByteBufferGetAt(ByteBufferHandle handle, int32_t index, uint8_t* out_val) {
+  API_BEGIN();
+  auto* buf = static_cast<std::vector<uint8_t>*>(handle);
+  if (index < 0 || index >= static_cast<int32_t>(buf->size()))
+    Log::Fatal("ByteBuffer index %d out of range", index);
+  *out_val = (*buf)[index];
+  API_END();
+}
+
+int LGBM_ByteBufferFree(ByteBufferHandle handle) {
+  API_BEGIN();
+  delete static_cast<std::vector<uint8_t>*>(handle);
   API_END();
 }
 

@@ -57,12 +57,18 @@ def test_serialized_reference_roundtrip():
     X = rng.randn(n, d)
     base = lgb.Dataset(X, label=np.zeros(n, dtype=np.float32),
                        params={"max_bin": 31}).construct()
-    out_len = ctypes.c_int64(0)
+    # reference ABI: library-owned ByteBuffer + per-byte getter
+    bb = ctypes.c_void_p()
+    out_len = ctypes.c_int32(0)
     _safe_call(_LIB.LGBM_DatasetSerializeReferenceToBinary(
-        base._handle, ctypes.c_int64(0), ctypes.byref(out_len), None))
+        base._handle, ctypes.byref(bb), ctypes.byref(out_len)))
+    assert out_len.value > 0
     buf = ctypes.create_string_buffer(out_len.value)
-    _safe_call(_LIB.LGBM_DatasetSerializeReferenceToBinary(
-        base._handle, out_len, ctypes.byref(out_len), buf))
+    v = ctypes.c_uint8(0)
+    for i in range(out_len.value):
+        _safe_call(_LIB.LGBM_ByteBufferGetAt(bb, ctypes.c_int32(i), ctypes.byref(v)))
+        buf[i] = v.value
+    _safe_call(_LIB.LGBM_ByteBufferFree(bb))
     # rebuild an empty dataset from the serialized reference and push the same rows
     out = ctypes.c_void_p()
     _safe_call(_LIB.LGBM_DatasetCreateFromSerializedReference(
