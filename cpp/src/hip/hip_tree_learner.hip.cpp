@@ -3864,7 +3864,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // loud, not silent: features the device split loop does not implement yet fall
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
-    if (cfg->boosting == "rf") return "random forest score bookkeeping";
     if (cfg->linear_tree) return "linear_tree";
     if (cfg->monotone_penalty > 0.0 && !cfg->monotone_constraints.empty())
       return "monotone_penalty (depth-dependent gain scaling)";

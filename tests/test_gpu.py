@@ -670,3 +670,18 @@ def test_gpu_use_dp_fp64_histograms():
         aucs[dp] = _auc(y[:20000], bst.predict(X[:20000]))
     assert aucs[True] > 0.8
     assert abs(aucs[True] - aucs[False]) < 2e-3, aucs
+
+
+def test_gpu_random_forest():
+    """boosting=rf on the device learner (round 1 fell back to CPU): quality
+    parity with the CPU RF at identical config."""
+    X, y = _binary_data(n=80_000, d=10)
+    aucs = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "binary", "boosting": "rf", "device_type": dev,
+                  "bagging_freq": 1, "bagging_fraction": 0.6, "num_leaves": 63,
+                  "verbosity": 0, "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 20)
+        aucs[dev] = _auc(y[:20000], bst.predict(X[:20000]))
+    assert aucs["gpu"] > 0.8, aucs
+    assert abs(aucs["cpu"] - aucs["gpu"]) < 5e-3, aucs
