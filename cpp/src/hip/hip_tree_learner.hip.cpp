@@ -244,8 +244,8 @@ struct LogEntry {
 struct LeafStat {
   double sum_g, sum_h;
   double parent_out;  // hessian-weighted parent output (path smoothing)
-  int cnt;  // exact GLOBAL row count (allreduced in multi-GPU)
-  int pad;
+  int cnt;    // exact GLOBAL row count (allreduced in multi-GPU)
+  int depth;  // leaf depth (max_depth gating + monotone_penalty decay)
 };
 
 struct GainParams {
@@ -264,6 +264,8 @@ struct GainParams {
   // distributed feature ownership: this rank scans only features [own_fb, own_fe)
   // (reduce-scatter mode divides both wire bytes and the gain scan by world)
   int own_fb, own_fe;
+  int max_depth;         // 0/neg = unbounded
+  double mono_penalty;   // monotone_penalty depth-decay factor (0 = off)
 };
 
 __device__ __forceinline__ uint32_t d_hash3(uint32_t a, uint32_t b, uint32_t c) {
@@ -601,6 +603,7 @@ __global__ void k_init_root(int* leaf_begin, int* leaf_cnt, int* leaf_slot, Leaf
   stats[0].sum_g = 0.0;
   stats[0].sum_h = 0.0;
   stats[0].cnt = used_cnt;
+  stats[0].depth = 0;
   gbuf[0] = used_cnt;
   counters[0] = 1;  // num_leaves
   counters[1] = 0;  // split log index
@@ -699,6 +702,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
   const double sum_h = st.sum_h;
   const int num_data = st.cnt;
   if (num_data < 2 * p.min_data) return;
+  if (p.max_depth > 0 && st.depth >= p.max_depth) return;
   const double cnt_factor = (num_data > 0 && sum_h > 0) ? num_data / sum_h : 1.0;
   const double leaf_parent_out = st.parent_out;
   const double parent_gain = d_leaf_gain_sm(sum_g, sum_h, num_data, leaf_parent_out, p);
@@ -1081,6 +1085,17 @@ __global__ void __launch_bounds__(64) k_best_feat(
   if (lane == 0 && best_bin >= 0) {
     rec.valid = 1;
     rec.gain = best_gain - min_gain_shift + p.min_gain_to_split;
+    if (mc != 0 && p.mono_penalty > 0.0) {
+      // depth-decaying multiplicative penalty on monotone splits (CPU-oracle
+      // ComputeMonotoneSplitGainPenalty parity)
+      const double pen = p.mono_penalty;
+      const int dep = st.depth;
+      double factor;
+      if (pen >= dep + 1.0) factor = 1e-15;
+      else if (pen <= 1.0) factor = 1.0 - pen / exp2((double)dep) + 1e-15;
+      else factor = 1.0 - exp2(pen - 1.0 - dep) + 1e-15;
+      rec.gain *= factor;
+    }
     rec.feature = f;
     rec.bin = best_bin;
     rec.default_left = best_dl;
@@ -1645,10 +1660,12 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
   stats[L].sum_h = w.left_h;
   stats[L].cnt = gl;
   stats[L].parent_out = po;
+  stats[L].depth = parent.depth + 1;
   stats[R].sum_g = parent.sum_g - w.left_g;
   stats[R].sum_h = right_h;
   stats[R].cnt = gr;
   stats[R].parent_out = po;
+  stats[R].depth = parent.depth + 1;
   const int old_slot = leaf_slot[L];
   if (gl <= gr) {
     leaf_slot[L] = spare_slot;
@@ -3059,6 +3076,8 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.rng_seed = bynode_seed_;
   p.own_fb = own_scan_ ? own_fb_ : 0;
   p.own_fe = own_scan_ ? own_fe_ : nf_;
+  p.max_depth = config_->max_depth;
+  p.mono_penalty = use_mono_ ? config_->monotone_penalty : 0.0;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
   const int ny = leafB_from_counters ? 2 : 1;
   if (hist_dp_) {
@@ -3865,8 +3884,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
     if (cfg->linear_tree) return "linear_tree";
-    if (cfg->monotone_penalty > 0.0 && !cfg->monotone_constraints.empty())
-      return "monotone_penalty (depth-dependent gain scaling)";
     if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
     if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||
         !cfg->cegb_penalty_feature_lazy.empty())

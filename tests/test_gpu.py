@@ -685,3 +685,44 @@ def test_gpu_random_forest():
         aucs[dev] = _auc(y[:20000], bst.predict(X[:20000]))
     assert aucs["gpu"] > 0.8, aucs
     assert abs(aucs["cpu"] - aucs["gpu"]) < 5e-3, aucs
+
+
+def test_gpu_max_depth_enforced():
+    """max_depth must bound the device-built trees (was silently ignored on GPU
+    before round 2 — the depth now lives in the device LeafStat)."""
+    X, y = _binary_data(n=60_000, d=10)
+    for md in (3, 6):
+        params = {"objective": "binary", "device_type": "gpu", "num_leaves": 255,
+                  "max_depth": md, "verbosity": 0, "metric": "none",
+                  "min_data_in_leaf": 1}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 5)
+        d = bst.dump_model()
+        def depth(node):
+            if "split_feature" not in node:
+                return 0
+            return 1 + max(depth(node["left_child"]), depth(node["right_child"]))
+        for t in d["tree_info"]:
+            assert depth(t["tree_structure"]) <= md, (md, depth(t["tree_structure"]))
+        # and a depth-md tree has at most 2^md leaves
+        for t in d["tree_info"]:
+            assert t["num_leaves"] <= 2 ** md
+
+
+def test_gpu_monotone_penalty():
+    """monotone_penalty now runs in the device gain scan (was a loud CPU
+    fallback): monotonicity holds and penalized models differ from unpenalized."""
+    rng = np.random.RandomState(0)
+    X = rng.rand(30000, 3)
+    y = (2 * X[:, 0] + np.sin(5 * X[:, 1]) + 0.1 * rng.randn(30000)).astype(np.float32)
+    preds = {}
+    for pen in (0.0, 2.0):
+        params = {"objective": "regression", "device_type": "cuda",
+                  "monotone_constraints": [1, 0, 0], "monotone_penalty": pen,
+                  "num_leaves": 63, "verbosity": -1, "metric": "none"}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 30)
+        xs = np.linspace(0.02, 0.98, 30)
+        for other in (0.2, 0.8):
+            grid = np.column_stack([xs, np.full(30, other), np.full(30, other)])
+            assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
+        preds[pen] = bst.predict(X[:3000])
+    assert np.abs(preds[0.0] - preds[2.0]).max() > 1e-6  # penalty changed the model
