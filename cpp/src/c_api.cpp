@@ -1191,7 +1191,13 @@ int LGBM_BoosterPredictForMat(BoosterHandle handle, const void* data, int data_t
                               int64_t* out_len, double* out_result) {
   API_BEGIN();
   auto* b = static_cast<BoosterWrapper*>(handle)->boosting();
-  if (ncol <= b->MaxFeatureIdx())
+  bool disable_shape_check = false;
+  if (parameter != nullptr) {
+    for (auto& kv : Config::Str2Map(parameter))
+      if (kv.first == "predict_disable_shape_check")
+        disable_shape_check = kv.second == "true" || kv.second == "1";
+  }
+  if (!disable_shape_check && ncol <= b->MaxFeatureIdx())
     Log::Fatal("The number of features in data (%d) is fewer than it was in training data "
                "(%d)", ncol, b->MaxFeatureIdx() + 1);
   auto get = MakeGetter(data, data_type);

@@ -146,6 +146,8 @@ void GBDT::Init(const Config* config, const Dataset* train_data,
       int64_t len = train_data_->metadata().num_init_score();
       std::copy(init_sc, init_sc + std::min<int64_t>(len, train_score_.size()),
                 train_score_.begin());
+      // device-resident scores must start from the same per-row init offsets
+      if (tree_learner_->IsHIPLearner()) tree_learner_->UploadTrainScore(train_score_.data());
     }
     gradients_.assign(train_score_.size(), 0);
     hessians_.assign(train_score_.size(), 0);

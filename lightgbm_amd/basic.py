@@ -899,7 +899,16 @@ class Booster:
         pred_param = _param_dict_to_str(
             {k: v for k, v in kwargs.items()
              if k in ("pred_early_stop", "pred_early_stop_freq",
-                      "pred_early_stop_margin")})
+                      "pred_early_stop_margin", "predict_disable_shape_check")})
+        if not kwargs.get("predict_disable_shape_check", False):
+            ncol = getattr(data, "shape", (0, 0))[1] if hasattr(data, "shape") and \
+                len(getattr(data, "shape", ())) == 2 else None
+            if ncol is not None and ncol != self.num_feature():
+                raise LightGBMError(
+                    f"The number of features in data ({ncol}) is not the same as it "
+                    f"was in training data ({self.num_feature()}).\n"
+                    "You can set ``predict_disable_shape_check=true`` to discard "
+                    "this error, but please be aware what you are doing.")
         ptype = _PREDICT_NORMAL
         if raw_score:
             ptype = _PREDICT_RAW

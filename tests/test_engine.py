@@ -1068,3 +1068,15 @@ def test_predict_with_start_iteration():
     head = bst.predict(X[:50], raw_score=True, num_iteration=8)
     tail = bst.predict(X[:50], raw_score=True, start_iteration=8, num_iteration=12)
     np.testing.assert_allclose(head + tail, full, rtol=1e-9)
+
+
+def test_predict_shape_check():
+    """Mismatched feature count raises unless predict_disable_shape_check
+    (reference parity)."""
+    X, y = _binary_data(n=2000)
+    bst = lgb.train({"objective": "binary", "verbosity": -1, "num_leaves": 7},
+                    lgb.Dataset(X, label=y), 3)
+    with pytest.raises(lgb.LightGBMError, match="number of features"):
+        bst.predict(X[:5, :6])
+    p = bst.predict(X[:5, :6], predict_disable_shape_check=True)
+    assert p.shape == (5,)

@@ -726,3 +726,28 @@ def test_gpu_monotone_penalty():
             assert np.all(np.diff(bst.predict(grid)) >= -1e-9)
         preds[pen] = bst.predict(X[:3000])
     assert np.abs(preds[0.0] - preds[2.0]).max() > 1e-6  # penalty changed the model
+
+
+def test_gpu_init_score_honored():
+    """Per-row init_score must seed the device score buffer (was silently
+    dropped on GPU). Parity: GPU model trained on residuals equals the CPU one
+    at quality level."""
+    rng = np.random.RandomState(5)
+    n = 50_000
+    X = rng.randn(n, 8).astype(np.float32)
+    y = (2 * X[:, 0] + X[:, 1] + 0.3 * rng.randn(n)).astype(np.float32)
+    init = (2 * X[:, 0]).astype(np.float64)  # a strong known component
+    preds = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "regression", "device_type": dev, "num_leaves": 31,
+                  "verbosity": 0, "metric": "none"}
+        ds = lgb.Dataset(X, label=y, init_score=init)
+        bst = lgb.train(params, ds, 15)
+        preds[dev] = bst.predict(X[:5000])
+    # with init_score absorbed, the trees model y - init ~= X1: if the GPU
+    # ignored init_score its trees would re-learn 2*X0 and diverge sharply
+    diff = np.abs(preds["cpu"] - preds["gpu"]).mean()
+    assert diff < 0.05, diff
+    resid_target = y[:5000] - init[:5000]
+    mse = float(np.mean((preds["gpu"] - resid_target) ** 2))
+    assert mse < 0.5 * float(np.var(resid_target)), mse
