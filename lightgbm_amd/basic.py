@@ -903,7 +903,7 @@ class Booster:
         if not kwargs.get("predict_disable_shape_check", False):
             ncol = getattr(data, "shape", (0, 0))[1] if hasattr(data, "shape") and \
                 len(getattr(data, "shape", ())) == 2 else None
-            if ncol is not None and ncol != self.num_feature():
+            if ncol is not None and ncol < self.num_feature():
                 raise LightGBMError(
                     f"The number of features in data ({ncol}) is not the same as it "
                     f"was in training data ({self.num_feature()}).\n"
