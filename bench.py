@@ -148,7 +148,11 @@ def main():
             idb = payload[0]
             rc = _LIB.LGBM_GPUNetworkInit(ctypes.c_int(world), ctypes.c_int(rank), idb)
             if rc != 0:
-                raise RuntimeError(f"rank {rank}: RCCL communicator init failed (see stderr)")
+                # e.g. ranks sharing one physical GPU (RCCL refuses duplicate
+                # devices): the learner falls back to the host-seam transport
+                # (gloo) automatically — correct, just slower
+                print(f"rank {rank}: RCCL init failed; using the host-seam "
+                      "collective transport", file=sys.stderr)
 
     # ---- data: bin mappers must be identical on every rank -> all ranks build the
     # same reference dataset from a common-seed sample, then bin their own shard
