@@ -266,6 +266,8 @@ struct GainParams {
   int own_fb, own_fe;
   int max_depth;         // 0/neg = unbounded
   double mono_penalty;   // monotone_penalty depth-decay factor (0 = off)
+  double cegb_tradeoff;  // cost-effective gradient boosting (0 = off)
+  double cegb_split_pen;
 };
 
 __device__ __forceinline__ uint32_t d_hash3(uint32_t a, uint32_t b, uint32_t c) {
@@ -291,6 +293,20 @@ __device__ __forceinline__ double d_thl1(double s, double l1) {
   r = r > 0.0 ? r : 0.0;
   return s >= 0.0 ? r : -r;
 }
+/*! cost-effective gradient boosting gain penalty (CPU-oracle parity:
+ *  SerialTreeLearner CEGB block). coupled[f] is zeroed once f is used. */
+__device__ __forceinline__ void d_apply_cegb(SplitRec& rec, int f, int num_data,
+                                             const GainParams& p,
+                                             const float* __restrict__ coupled,
+                                             const float* __restrict__ lazy) {
+  if (p.cegb_tradeoff <= 0.0 || !rec.valid) return;
+  double pen = p.cegb_split_pen;
+  if (coupled != nullptr) pen += coupled[f];
+  if (lazy != nullptr) pen += static_cast<double>(lazy[f]) * num_data;
+  rec.gain -= p.cegb_tradeoff * pen;
+  if (rec.gain <= 0.0) rec.valid = 0;
+}
+
 __device__ __forceinline__ double d_leaf_out(double g, double h, const GainParams& p) {
   double r = -d_thl1(g, p.l1) / (h + p.l2);
   if (p.mds > 0.0 && fabs(r) > p.mds) r = r > 0 ? p.mds : -p.mds;
@@ -660,6 +676,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
     const int8_t* __restrict__ mono, const double* __restrict__ leaf_bounds,
     const unsigned long long* __restrict__ group_masks,
     const unsigned long long* __restrict__ leaf_branch,
+    const float* __restrict__ cegb_coupled, const float* __restrict__ cegb_lazy,
     SplitRec* __restrict__ out) {
   const int f = blockIdx.x;
   const int which = blockIdx.y;
@@ -848,6 +865,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
             rec.right_out = d_leaf_out_l2(sum_g - b_lg, sum_h - b_lh, p, l2c);
             rec.left_cnt = static_cast<int>(b_lh * cnt_factor + 0.5);
             rec.right_cnt = num_data - rec.left_cnt;
+            d_apply_cegb(rec, f, num_data, p, cegb_coupled, cegb_lazy);
           }
         }
       }
@@ -990,6 +1008,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
           rec.right_out = d_leaf_out_l2(sum_g - b_lg, sum_h - b_lh, p, l2c);
           rec.left_cnt = static_cast<int>(b_lh * cnt_factor + 0.5);
           rec.right_cnt = num_data - rec.left_cnt;
+          d_apply_cegb(rec, f, num_data, p, cegb_coupled, cegb_lazy);
         }
       }
       return;
@@ -1114,6 +1133,7 @@ __global__ void __launch_bounds__(64) k_best_feat(
     }
     rec.left_cnt = static_cast<int>(best_lh * cnt_factor + 0.5);
     rec.right_cnt = num_data - rec.left_cnt;
+    d_apply_cegb(rec, f, num_data, p, cegb_coupled, cegb_lazy);
   }
 }
 
@@ -1610,7 +1630,7 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
                            const int* __restrict__ Lptr, int* counters,
                            LogEntry* __restrict__ log, const int* __restrict__ ctr,
                            const int64_t* __restrict__ gbuf, HIST_T* hist_base,
-                           size_t slot_stride, int n_elem,
+                           size_t slot_stride, int n_elem, float* cegb_coupled,
                            const int8_t* __restrict__ mono, double* leaf_bounds,
                            unsigned long long* leaf_branch) {
   __shared__ int s_spare;
@@ -1621,6 +1641,7 @@ __global__ void k_finalize(int* leaf_begin, int* leaf_cnt, int* leaf_slot, LeafS
       log[counters[1]].leaf = -1;  // terminator for the host replay
     } else {
       s_spare = counters[0];
+      if (cegb_coupled != nullptr) cegb_coupled[winner->feature] = 0.0f;  // feature used
       FinalizeBookkeeping(leaf_begin, leaf_cnt, leaf_slot, stats, winner, L, counters, log,
                           ctr, gbuf, mono, leaf_bounds, leaf_branch);
     }
@@ -2427,6 +2448,8 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<unsigned long long> d_group_masks_;  // interaction groups (inner-feature bits)
   int n_interaction_groups_ = 0;
   DevBuf<int8_t> d_feat_mask_;
+  DevBuf<float> d_cegb_coupled_, d_cegb_lazy_;  // per inner feature (CEGB)
+  bool use_cegb_ = false;
   DevBuf<int> d_leaf_begin_, d_leaf_cnt_, d_leaf_slot_;
   DevBuf<int> d_sorted_begin_;
   DevBuf<double> d_leaf_out_;
@@ -2674,6 +2697,26 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     // SLOWER than 4 stream launches (grid-sync dispatch overhead dominates at
     // this kernel size) — opt-in only, kept as a documented negative result.
     coop_launch_ = coop != 0 && getenv("MIGBM_COOP_PARTITION") != nullptr && !rows16_;
+  }
+  use_cegb_ = config_->cegb_tradeoff > 0.0 &&
+              (config_->cegb_penalty_split > 0.0 ||
+               !config_->cegb_penalty_feature_coupled.empty() ||
+               !config_->cegb_penalty_feature_lazy.empty());
+  if (use_cegb_) {
+    std::vector<float> coupled(nf_, 0.0f), lazy(nf_, 0.0f);
+    for (int f = 0; f < nf_; ++f) {
+      const int orig = train_data->RealFeatureIndex(f);
+      if (orig < static_cast<int>(config_->cegb_penalty_feature_coupled.size()))
+        coupled[f] = static_cast<float>(config_->cegb_penalty_feature_coupled[orig]);
+      if (orig < static_cast<int>(config_->cegb_penalty_feature_lazy.size()))
+        lazy[f] = static_cast<float>(config_->cegb_penalty_feature_lazy[orig]);
+    }
+    d_cegb_coupled_.Alloc(nf_);
+    d_cegb_lazy_.Alloc(nf_);
+    HIP_OK(hipMemcpy(d_cegb_coupled_.ptr, coupled.data(), sizeof(float) * nf_,
+                     hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(d_cegb_lazy_.ptr, lazy.data(), sizeof(float) * nf_,
+                     hipMemcpyHostToDevice));
   }
   quantized_ = config_->use_quantized_grad;
   if (quantized_ && hist_dp_) {
@@ -3078,6 +3121,8 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
   p.own_fe = own_scan_ ? own_fe_ : nf_;
   p.max_depth = config_->max_depth;
   p.mono_penalty = use_mono_ ? config_->monotone_penalty : 0.0;
+  p.cegb_tradeoff = use_cegb_ ? config_->cegb_tradeoff : 0.0;
+  p.cegb_split_pen = config_->cegb_penalty_split;
   const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
   const int ny = leafB_from_counters ? 2 : 1;
   if (hist_dp_) {
@@ -3088,7 +3133,8 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                        feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
                        use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
                        n_interaction_groups_ > 0 ? d_group_masks_.ptr : nullptr,
-                       d_leaf_branch_.ptr, d_feat_best_.ptr);
+                       d_leaf_branch_.ptr, use_cegb_ ? d_cegb_coupled_.ptr : nullptr,
+                       use_cegb_ ? d_cegb_lazy_.ptr : nullptr, d_feat_best_.ptr);
   } else {
     hipLaunchKernelGGL(hipk::k_best_feat, dim3(nf_, ny), dim3(64), 0, stream_,
                        const_cast<const float*>(d_hist_.ptr), slot_stride,
@@ -3097,7 +3143,8 @@ void HIPTreeLearner::LaunchBestSplit(const int* leafA_ptr, int leafB_from_counte
                        feat_mask_host_.empty() ? nullptr : d_feat_mask_.ptr,
                        use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
                        n_interaction_groups_ > 0 ? d_group_masks_.ptr : nullptr,
-                       d_leaf_branch_.ptr, d_feat_best_.ptr);
+                       d_leaf_branch_.ptr, use_cegb_ ? d_cegb_coupled_.ptr : nullptr,
+                       use_cegb_ ? d_cegb_lazy_.ptr : nullptr, d_feat_best_.ptr);
   }
   (void)ny;
   hipLaunchKernelGGL(hipk::k_best_leaf_overall, dim3(1), dim3(256), 0, stream_,
@@ -3255,6 +3302,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                          d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr, d_gbuf_.ptr,
                          reinterpret_cast<double*>(d_hist_.ptr),
                          static_cast<size_t>(total_bins_) * 2, total_bins_ * 2,
+                         use_cegb_ ? d_cegb_coupled_.ptr : nullptr,
                          use_mono_ ? d_mono_.ptr : nullptr, d_leaf_bounds_.ptr,
                          d_leaf_branch_.ptr);
     else
@@ -3263,7 +3311,8 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
                          d_leaf_stats_.ptr, d_winner_.ptr, d_winner_leaf_.ptr,
                          d_counters_.ptr, d_split_log_.ptr, d_ctr_.ptr, d_gbuf_.ptr,
                          d_hist_.ptr, static_cast<size_t>(total_bins_) * 2,
-                         total_bins_ * 2, use_mono_ ? d_mono_.ptr : nullptr,
+                         total_bins_ * 2, use_cegb_ ? d_cegb_coupled_.ptr : nullptr,
+                         use_mono_ ? d_mono_.ptr : nullptr,
                          d_leaf_bounds_.ptr, d_leaf_branch_.ptr);
     LaunchHist(d_winner_leaf_.ptr, 1, kLoopHistBlocks, /*zero_spare=*/false);
     ReduceSpareHist(split_i + 1);  // spare slot for split i is deterministically i+1
@@ -3885,9 +3934,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
   auto unsupported = [&]() -> const char* {
     if (cfg->linear_tree) return "linear_tree";
     if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
-    if (cfg->cegb_penalty_split > 0.0 || !cfg->cegb_penalty_feature_coupled.empty() ||
-        !cfg->cegb_penalty_feature_lazy.empty())
-      return "cost-effective gradient boosting (cegb_*)";
     return nullptr;
   };
   if (const char* what = unsupported()) {

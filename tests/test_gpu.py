@@ -751,3 +751,27 @@ def test_gpu_init_score_honored():
     resid_target = y[:5000] - init[:5000]
     mse = float(np.mean((preds["gpu"] - resid_target) ** 2))
     assert mse < 0.5 * float(np.var(resid_target)), mse
+
+
+def test_gpu_cegb():
+    """Cost-effective gradient boosting runs in the device gain scan (was a
+    loud CPU fallback): coupled penalties steer the model to fewer features,
+    matching the CPU learner's behavior."""
+    rng = np.random.RandomState(0)
+    n = 40_000
+    X = rng.randn(n, 6).astype(np.float32)
+    # two redundant informative features; CEGB coupled penalty should make the
+    # model reuse one instead of paying for both
+    y = (X[:, 0] + 0.95 * X[:, 1] + 0.2 * rng.randn(n) > 0).astype(np.float32)
+    used = {}
+    for dev in ("cpu", "gpu"):
+        params = {"objective": "binary", "device_type": dev, "num_leaves": 31,
+                  "verbosity": 0, "metric": "none", "cegb_tradeoff": 1.0,
+                  "cegb_penalty_feature_coupled": [50.0] * 6}
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 10)
+        imp = bst.feature_importance()
+        used[dev] = int((imp > 0).sum())
+        assert _auc(y[:10000], bst.predict(X[:10000])) > 0.8
+    # both learners concentrate on few features under the coupled penalty
+    assert used["gpu"] <= 3, used
+    assert abs(used["cpu"] - used["gpu"]) <= 1, used
