@@ -676,6 +676,8 @@ double GBDT::GetLowerBoundValue() const {
 void GBDT::RefitTree(const int32_t* leaf_preds, int nrow, int ncol) {
   MIGBM_CHECK_EQ(static_cast<size_t>(ncol), models_.size());
   // re-derive gradients at current scores, refit leaf outputs with decay
+  if (tree_learner_ && tree_learner_->IsHIPLearner())
+    tree_learner_->DownloadTrainScore(train_score_.data());
   objective_->GetGradients(train_score_.data(), gradients_.data(), hessians_.data());
   for (int t = 0; t < ncol; ++t) {
     int tid = t % num_tree_per_iteration_;

@@ -775,3 +775,17 @@ def test_gpu_cegb():
     # both learners concentrate on few features under the coupled penalty
     assert used["gpu"] <= 3, used
     assert abs(used["cpu"] - used["gpu"]) <= 1, used
+
+
+def test_gpu_refit():
+    """Booster.refit on a device-trained model (scores downloaded for the
+    gradient re-derivation)."""
+    X, y = _binary_data(n=30_000, d=8)
+    params = {"objective": "binary", "device_type": "gpu", "num_leaves": 31,
+              "verbosity": 0, "metric": "none"}
+    bst = lgb.train(params, lgb.Dataset(X, label=y), 8)
+    y2 = 1.0 - y  # refit onto flipped labels shifts outputs
+    new_bst = bst.refit(X, y2)
+    p_old = bst.predict(X[:2000])
+    p_new = new_bst.predict(X[:2000])
+    assert np.abs(p_old - p_new).mean() > 0.05
