@@ -17,6 +17,15 @@
 
 namespace migbm {
 
+/*! forced-splits JSON node (forcedsplits_filename; reference format). */
+struct ForcedNode {
+  int feature = -1;
+  double threshold = 0.0;
+  std::unique_ptr<ForcedNode> left, right;
+};
+/*! parse the forced-splits JSON file (empty path / bad file -> nullptr). */
+std::unique_ptr<ForcedNode> ParseForcedSplits(const std::string& path);
+
 class TreeLearner {
  public:
   virtual ~TreeLearner() = default;
@@ -254,11 +263,6 @@ class SerialTreeLearner : public TreeLearner {
   int iter_counter_ = 0;
   bool build_both_children_ = false;  // voting-parallel: no histogram subtraction
   // forced splits (forcedsplits_filename JSON)
-  struct ForcedNode {
-    int feature = -1;
-    double threshold = 0.0;
-    std::unique_ptr<ForcedNode> left, right;
-  };
   std::unique_ptr<ForcedNode> forced_root_;
   std::vector<const ForcedNode*> forced_of_leaf_;
   bool MakeForcedSplit(int leaf, const LeafContext& ctx, const ForcedNode* node,

@@ -1763,6 +1763,60 @@ __global__ void k_pick_global_winner(const LogEntry* __restrict__ wires, int wor
   }
 }
 
+/*! forced split: override the winner with (leaf, feature, bin) from the
+ *  forcedsplits schedule; stats from the leaf's histogram prefix (CPU
+ *  MakeForcedSplit parity incl. monotone clamp/equalize). One wave. */
+template <typename HIST_T = float>
+__global__ void k_force_winner(const HIST_T* __restrict__ hist_base, size_t slot_stride,
+                               const int* __restrict__ leaf_slot,
+                               const FeatMeta* __restrict__ fm,
+                               const LeafStat* __restrict__ stats, int leaf, int f,
+                               int bin, GainParams p, const int8_t* __restrict__ mono,
+                               const double* __restrict__ leaf_bounds,
+                               SplitRec* __restrict__ winner,
+                               int* __restrict__ winner_leaf) {
+  const int lane = threadIdx.x;
+  const FeatMeta m = fm[f];
+  const HIST_T* fh = hist_base + static_cast<size_t>(leaf_slot[leaf]) * slot_stride +
+                     static_cast<size_t>(m.bin_off) * 2;
+  double gl = 0, hl = 0;
+  for (int b = lane; b <= bin; b += 64) {
+    gl += fh[2 * b];
+    hl += fh[2 * b + 1];
+  }
+  for (int d = 32; d > 0; d >>= 1) {
+    gl += __shfl_down(gl, d);
+    hl += __shfl_down(hl, d);
+  }
+  if (lane != 0) return;
+  const LeafStat st = stats[leaf];
+  SplitRec rec;
+  rec.valid = 1;
+  rec.feature = f;
+  rec.bin = bin;
+  rec.default_left = 0;
+  rec.cat_mask[0] = rec.cat_mask[1] = rec.cat_mask[2] = rec.cat_mask[3] = 0;
+  rec.gain = 1e30;  // forced splits take precedence over gain selection
+  rec.left_g = gl;
+  rec.left_h = hl;
+  const double cf = st.cnt > 0 && st.sum_h > 0 ? st.cnt / st.sum_h : 1.0;
+  rec.left_cnt = static_cast<int>(hl * cf + 0.5);
+  rec.right_cnt = st.cnt - rec.left_cnt;
+  double lo = d_leaf_out(gl, hl, p);
+  double ro = d_leaf_out(st.sum_g - gl, st.sum_h - hl, p);
+  if (mono != nullptr) {
+    const double blo = leaf_bounds[2 * leaf], bhi = leaf_bounds[2 * leaf + 1];
+    lo = fmin(fmax(lo, blo), bhi);
+    ro = fmin(fmax(ro, blo), bhi);
+    const int8_t mc = mono[f];
+    if ((mc > 0 && lo > ro) || (mc < 0 && lo < ro)) lo = ro = (lo + ro) / 2.0;
+  }
+  rec.left_out = lo;
+  rec.right_out = ro;
+  *winner = rec;
+  *winner_leaf = leaf;
+}
+
 // ------------------------------------------------------------------ boosting kernels
 __global__ void k_grad_binary(const double* __restrict__ score,
                               const float* __restrict__ label,
@@ -2450,6 +2504,11 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<int8_t> d_feat_mask_;
   DevBuf<float> d_cegb_coupled_, d_cegb_lazy_;  // per inner feature (CEGB)
   bool use_cegb_ = false;
+  // forcedsplits: host-precomputed per-iteration schedule (leaf,-1 = no force)
+  struct ForcedStep { int leaf = -1; int feature = -1; int bin = -1; };
+  std::unique_ptr<ForcedNode> forced_root2_;
+  std::vector<ForcedStep> forced_sched_;
+  void LaunchForcedWinner(const ForcedStep& fs);
   DevBuf<int> d_leaf_begin_, d_leaf_cnt_, d_leaf_slot_;
   DevBuf<int> d_sorted_begin_;
   DevBuf<double> d_leaf_out_;
@@ -2698,6 +2757,7 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     // this kernel size) — opt-in only, kept as a documented negative result.
     coop_launch_ = coop != 0 && getenv("MIGBM_COOP_PARTITION") != nullptr && !rows16_;
   }
+  forced_root2_ = ParseForcedSplits(config_->forcedsplits_filename);
   use_cegb_ = config_->cegb_tradeoff > 0.0 &&
               (config_->cegb_penalty_split > 0.0 ||
                !config_->cegb_penalty_feature_coupled.empty() ||
@@ -3091,6 +3151,27 @@ bool HIPTreeLearner::SetupOwnership(int world, int rank) {
   return true;
 }
 
+void HIPTreeLearner::LaunchForcedWinner(const ForcedStep& fs) {
+  const size_t slot_stride = static_cast<size_t>(total_bins_) * 2;
+  hipk::GainParams p = {};
+  p.l1 = config_->lambda_l1;
+  p.l2 = config_->lambda_l2;
+  p.mds = config_->max_delta_step;
+  if (hist_dp_) {
+    hipLaunchKernelGGL(hipk::k_force_winner, dim3(1), dim3(64), 0, stream_,
+                       reinterpret_cast<const double*>(d_hist_.ptr), slot_stride,
+                       d_leaf_slot_.ptr, d_feat_meta_.ptr, d_leaf_stats_.ptr, fs.leaf,
+                       fs.feature, fs.bin, p, use_mono_ ? d_mono_.ptr : nullptr,
+                       d_leaf_bounds_.ptr, d_winner_.ptr, d_winner_leaf_.ptr);
+  } else {
+    hipLaunchKernelGGL(hipk::k_force_winner, dim3(1), dim3(64), 0, stream_,
+                       const_cast<const float*>(d_hist_.ptr), slot_stride,
+                       d_leaf_slot_.ptr, d_feat_meta_.ptr, d_leaf_stats_.ptr, fs.leaf,
+                       fs.feature, fs.bin, p, use_mono_ ? d_mono_.ptr : nullptr,
+                       d_leaf_bounds_.ptr, d_winner_.ptr, d_winner_leaf_.ptr);
+  }
+}
+
 void HIPTreeLearner::SyncGlobalWinner() {
   auto& comm = Comm();
   hipLaunchKernelGGL(hipk::k_pack_winner, dim3(1), dim3(1), 0, stream_, d_winner_.ptr,
@@ -3218,6 +3299,45 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
         mode == "reduce_scatter" || (mode == "auto" && payload >= (128u << 10));
     if (want_rs) own_scan_ = SetupOwnership(comm.World(), comm.Rank());
   }
+  // forced splits: the schedule is data-independent (validity checks are all
+  // host-side), so simulate the CPU loop's leaf bookkeeping up front
+  forced_sched_.clear();
+  if (forced_root2_) {
+    own_scan_ = false;  // forced features may not be rank-owned in rs mode
+    std::vector<const ForcedNode*> fol(nl, nullptr);
+    fol[0] = forced_root2_.get();
+    int sim_leaves = 1;
+    for (int it = 0; it < nl - 1; ++it) {
+      ForcedStep step;
+      for (int l = 0; l < sim_leaves; ++l) {
+        const ForcedNode* node = fol[l];
+        if (node == nullptr) continue;
+        const int inner = train_data_->InnerFeatureIndex(node->feature);
+        const BinMapper* mp = inner >= 0 ? train_data_->FeatureBinMapper(inner) : nullptr;
+        int bin = -1;
+        if (mp != nullptr && mp->bin_type() == BinType::kNumerical)
+          bin = std::min<int>(static_cast<int>(mp->ValueToBin(node->threshold)),
+                              mp->num_numeric_bin() - 2);
+        if (bin < 0) {
+          fol[l] = nullptr;  // unusable forced node: cleared, scan continues
+          continue;
+        }
+        step = {l, inner, bin};
+        fol[l] = node->left.get();
+        fol[sim_leaves] = node->right.get();
+        break;
+      }
+      forced_sched_.push_back(step);
+      ++sim_leaves;
+      if (step.leaf < 0) {
+        bool any = false;
+        for (int l = 0; l < sim_leaves; ++l) any = any || fol[l] != nullptr;
+        if (!any) break;  // no pending forced nodes: normal loop from here on
+      }
+    }
+    while (!forced_sched_.empty() && forced_sched_.back().leaf < 0)
+      forced_sched_.pop_back();
+  }
   if (dist_) {
     comm.AllReduce(&d_leaf_stats_.ptr[0].sum_g, 2, stream_);
     comm.AllReduce(d_gbuf_.ptr, 1, stream_);
@@ -3246,6 +3366,10 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
   for (int split_i = 0; split_i < nl - 1; ++split_i) {
     // winner for this split was already selected by the fused k_best_leaf_overall
     // at the end of the previous split (or of the root best-split pass)
+    if (split_i < static_cast<int>(forced_sched_.size()) &&
+        forced_sched_[split_i].leaf >= 0) {
+      LaunchForcedWinner(forced_sched_[split_i]);  // forced splits take priority
+    }
     if (coop_launch_) {
       void* args[] = {
           const_cast<void*>(static_cast<const void*>(&d_idx_.ptr)),
@@ -3933,7 +4057,6 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
     if (cfg->linear_tree) return "linear_tree";
-    if (!cfg->forcedsplits_filename.empty()) return "forcedsplits";
     return nullptr;
   };
   if (const char* what = unsupported()) {

