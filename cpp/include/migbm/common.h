@@ -179,6 +179,43 @@ class Timer {
   std::unordered_map<std::string, double> starts_, totals_;
 };
 
+// ---------------------------------------------------------------- small linear algebra
+/*! Cholesky solve of A beta = b (A: dim x dim, upper triangle filled, SPD after
+ *  ridge). Returns false when not positive definite. Shared by the CPU linear
+ *  tree fit and the device Gram post-pass. */
+inline bool CholeskySolve(const std::vector<double>& A, const std::vector<double>& b,
+                          int dim, std::vector<double>* beta_out) {
+  std::vector<double> Lm(static_cast<size_t>(dim) * dim, 0.0);
+  for (int a = 0; a < dim; ++a) {
+    for (int c2 = 0; c2 <= a; ++c2) {
+      double sum = A[static_cast<size_t>(std::min(a, c2)) * dim + std::max(a, c2)];
+      for (int t = 0; t < c2; ++t) sum -= Lm[a * dim + t] * Lm[c2 * dim + t];
+      if (a == c2) {
+        if (sum <= 1e-12) return false;
+        Lm[a * dim + a] = std::sqrt(sum);
+      } else {
+        Lm[a * dim + c2] = sum / Lm[c2 * dim + c2];
+      }
+    }
+  }
+  std::vector<double> y(dim);
+  std::vector<double>& beta = *beta_out;
+  beta.assign(dim, 0.0);
+  for (int a = 0; a < dim; ++a) {
+    double sum = b[a];
+    for (int t = 0; t < a; ++t) sum -= Lm[a * dim + t] * y[t];
+    y[a] = sum / Lm[a * dim + a];
+  }
+  for (int a = dim - 1; a >= 0; --a) {
+    double sum = y[a];
+    for (int t = a + 1; t < dim; ++t) sum -= Lm[t * dim + a] * beta[t];
+    beta[a] = sum / Lm[a * dim + a];
+  }
+  for (double v : beta)
+    if (!std::isfinite(v)) return false;
+  return true;
+}
+
 // ---------------------------------------------------------------- threading
 namespace Threading {
 

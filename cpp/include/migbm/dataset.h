@@ -322,6 +322,7 @@ class Dataset {
   std::vector<int8_t> categorical_flags_;   // per original column
   bool has_raw() const { return !raw_values_.empty(); }
   float raw_value(int inner, data_size_t row) const { return raw_values_[inner][row]; }
+  const float* raw_column(int inner) const { return raw_values_[inner].data(); }
 
  private:
   friend class DatasetLoader;

@@ -141,6 +141,11 @@ class Tree {
     leaf_features_inner_[leaf] = feats_inner;
     leaf_coeff_[leaf] = coeff;
   }
+  const std::vector<int>& leaf_features_inner(int leaf) const {
+    return leaf_features_inner_[leaf];
+  }
+  const std::vector<double>& leaf_coeffs(int leaf) const { return leaf_coeff_[leaf]; }
+  double leaf_const(int leaf) const { return leaf_const_[leaf]; }
   double LeafOutputLinear(int leaf, const double* feature_values) const {
     if (!is_linear_ || leaf_coeff_[leaf].empty()) return leaf_value_[leaf];
     double out = leaf_const_[leaf];

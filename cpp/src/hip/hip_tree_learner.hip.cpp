@@ -2284,6 +2284,82 @@ __global__ void k_score_update(const uint32_t* __restrict__ idx,
   score[idx[i]] += leaf_out[lo];
 }
 
+/*! linear-leaf score update: score[row] += const + coeff . raw[path feats]
+ *  (NaN in any path feature falls back to the piecewise-constant output, like
+ *  Tree::LeafOutputLinear). Arrays are in SORTED-leaf order (sorted_begin). */
+__global__ void k_score_update_linear(
+    const uint32_t* __restrict__ idx, const int* __restrict__ sorted_begin,
+    int num_leaves, int used_cnt, const double* __restrict__ leaf_const,
+    const double* __restrict__ leaf_fallback, const int* __restrict__ coeff_off,
+    const int* __restrict__ coeff_cnt, const int* __restrict__ feat_flat,
+    const double* __restrict__ coeff_flat, const float* __restrict__ raw,
+    int num_data, double* __restrict__ score) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= used_cnt) return;
+  int lo = 0, hi = num_leaves - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi + 1) >> 1;
+    if (i >= sorted_begin[mid]) lo = mid;
+    else hi = mid - 1;
+  }
+  const uint32_t r = idx[i];
+  const int k = coeff_cnt[lo];
+  double out = leaf_const[lo];
+  bool nan = false;
+  const int off = coeff_off[lo];
+  for (int j = 0; j < k; ++j) {
+    const float v = raw[static_cast<size_t>(feat_flat[off + j]) * num_data + r];
+    if (isnan(v)) { nan = true; break; }
+    out += coeff_flat[off + j] * v;
+  }
+  score[r] += (nan || k == 0) ? leaf_fallback[lo] : out;
+}
+
+/*! per-leaf weighted Gram matrix (upper triangle) + rhs for the linear-tree
+ *  post-pass: A += h z z^T, b += -g z over the leaf rows, z = (raw path feats, 1).
+ *  Capability parity: reference LinearTreeLearner::CalculateLinear inner loops,
+ *  computed on device instead of the host row loop. */
+__global__ void k_linear_gram(const uint32_t* __restrict__ idx, int begin, int cnt,
+                              const float* __restrict__ raw, int num_data,
+                              const int* __restrict__ feats, int k,
+                              const float* __restrict__ g, const float* __restrict__ h,
+                              double* __restrict__ A, double* __restrict__ b,
+                              int* __restrict__ nan_flag) {
+  extern __shared__ double s_acc[];  // dim*dim (upper) + dim
+  const int dim = k + 1;
+  const int n_elem = dim * dim + dim;
+  for (int e = threadIdx.x; e < n_elem; e += blockDim.x) s_acc[e] = 0.0;
+  __syncthreads();
+  double z[33];  // dim capped host-side at 33 (32 features + intercept)
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < cnt;
+       i += gridDim.x * blockDim.x) {
+    const uint32_t r = idx[begin + i];
+    bool nan = false;
+    for (int j = 0; j < k; ++j) {
+      const float v = raw[static_cast<size_t>(feats[j]) * num_data + r];
+      if (isnan(v)) { nan = true; break; }
+      z[j] = v;
+    }
+    if (nan) {
+      atomicOr(nan_flag, 1);
+      continue;
+    }
+    z[k] = 1.0;
+    const double hv = h[r];
+    const double gv = g[r];
+    for (int a = 0; a < dim; ++a) {
+      const double hza = hv * z[a];
+      for (int c2 = a; c2 < dim; ++c2) atomicAdd(&s_acc[a * dim + c2], hza * z[c2]);
+      atomicAdd(&s_acc[dim * dim + a], -gv * z[a]);
+    }
+  }
+  __syncthreads();
+  for (int e = threadIdx.x; e < dim * dim; e += blockDim.x)
+    if (s_acc[e] != 0.0) atomicAdd(&A[e], s_acc[e]);
+  for (int e = threadIdx.x; e < dim; e += blockDim.x)
+    if (s_acc[dim * dim + e] != 0.0) atomicAdd(&b[e], s_acc[dim * dim + e]);
+}
+
 /*! device tree walk over column bins (out-of-bag score update under bagging). */
 template <typename BIN_T = uint8_t>
 __global__ void k_tree_predict_add(const BIN_T* __restrict__ cols, int num_data,
@@ -2509,6 +2585,16 @@ class HIPTreeLearner : public TreeLearner {
   std::unique_ptr<ForcedNode> forced_root2_;
   std::vector<ForcedStep> forced_sched_;
   void LaunchForcedWinner(const ForcedStep& fs);
+  // linear trees (device Gram post-pass over raw features on the branch path)
+  bool linear_ = false;
+  DevBuf<float> d_raw_;
+  DevBuf<int> d_lin_feats_;
+  DevBuf<double> d_lin_A_, d_lin_b_;
+  DevBuf<int> d_lin_nan_;
+  DevBuf<double> d_lin_const_;
+  DevBuf<int> d_lin_coeff_off_, d_lin_coeff_cnt_, d_lin_feat_flat_;
+  DevBuf<double> d_lin_coeff_flat_;
+  void CalculateLinearDevice(Tree* tree);
   DevBuf<int> d_leaf_begin_, d_leaf_cnt_, d_leaf_slot_;
   DevBuf<int> d_sorted_begin_;
   DevBuf<double> d_leaf_out_;
@@ -2758,6 +2844,18 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     coop_launch_ = coop != 0 && getenv("MIGBM_COOP_PARTITION") != nullptr && !rows16_;
   }
   forced_root2_ = ParseForcedSplits(config_->forcedsplits_filename);
+  linear_ = config_->linear_tree && train_data->has_raw();
+  if (config_->linear_tree && !train_data->has_raw())
+    Log::Warning("linear_tree requires raw values; dataset was built without them");
+  if (linear_) {
+    // raw feature values, feature-major (1 float per cell; 288GB HBM)
+    d_raw_.Alloc(static_cast<size_t>(nf_) * num_data_);
+    for (int f = 0; f < nf_; ++f) {
+      HIP_OK(hipMemcpy(d_raw_.ptr + static_cast<size_t>(f) * num_data_,
+                       train_data->raw_column(f), sizeof(float) * num_data_,
+                       hipMemcpyHostToDevice));
+    }
+  }
   use_cegb_ = config_->cegb_tradeoff > 0.0 &&
               (config_->cegb_penalty_split > 0.0 ||
                !config_->cegb_penalty_feature_coupled.empty() ||
@@ -3518,6 +3616,7 @@ Tree* HIPTreeLearner::Train(const score_t* gradients, const score_t* hessians, b
     }
     tree->OverrideLeafCounts(counts);
   }
+  if (linear_) CalculateLinearDevice(tree.get());
   return tree.release();
 }
 
@@ -3538,8 +3637,48 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
   HIP_OK(hipMemcpyAsync(d_sorted_begin_.ptr, sorted_begin.data(), sizeof(int) * nl,
                         hipMemcpyHostToDevice, stream_));
   const int n = static_cast<int>(used_cnt_);
+  if (tree->is_linear()) {
+    // linear leaves: score += const + coeff . raw (sorted-leaf-order flattening)
+    std::vector<double> lin_const(nl), coeff_flat;
+    std::vector<int> coeff_off(nl), coeff_cnt(nl), feat_flat;
+    for (int k2 = 0; k2 < nl; ++k2) {
+      const int l = order[k2];
+      coeff_off[k2] = static_cast<int>(feat_flat.size());
+      const auto& fi = tree->leaf_features_inner(l);
+      const auto& co = tree->leaf_coeffs(l);
+      coeff_cnt[k2] = static_cast<int>(co.size());
+      lin_const[k2] = co.empty() ? tree->LeafOutput(l) : tree->leaf_const(l);
+      for (size_t j = 0; j < co.size(); ++j) {
+        feat_flat.push_back(fi[j]);
+        coeff_flat.push_back(co[j]);
+      }
+    }
+    if (feat_flat.empty()) { feat_flat.push_back(0); coeff_flat.push_back(0.0); }
+    d_lin_const_.Alloc(nl);
+    d_lin_coeff_off_.Alloc(nl);
+    d_lin_coeff_cnt_.Alloc(nl);
+    d_lin_feat_flat_.Alloc(feat_flat.size());
+    d_lin_coeff_flat_.Alloc(coeff_flat.size());
+    HIP_OK(hipMemcpyAsync(d_lin_const_.ptr, lin_const.data(), sizeof(double) * nl,
+                          hipMemcpyHostToDevice, stream_));
+    HIP_OK(hipMemcpyAsync(d_lin_coeff_off_.ptr, coeff_off.data(), sizeof(int) * nl,
+                          hipMemcpyHostToDevice, stream_));
+    HIP_OK(hipMemcpyAsync(d_lin_coeff_cnt_.ptr, coeff_cnt.data(), sizeof(int) * nl,
+                          hipMemcpyHostToDevice, stream_));
+    HIP_OK(hipMemcpyAsync(d_lin_feat_flat_.ptr, feat_flat.data(),
+                          sizeof(int) * feat_flat.size(), hipMemcpyHostToDevice, stream_));
+    HIP_OK(hipMemcpyAsync(d_lin_coeff_flat_.ptr, coeff_flat.data(),
+                          sizeof(double) * coeff_flat.size(), hipMemcpyHostToDevice,
+                          stream_));
+    hipLaunchKernelGGL(hipk::k_score_update_linear, dim3((n + 255) / 256), dim3(256), 0,
+                       stream_, d_idx_.ptr, d_sorted_begin_.ptr, nl, n, d_lin_const_.ptr,
+                       d_leaf_out_.ptr, d_lin_coeff_off_.ptr, d_lin_coeff_cnt_.ptr,
+                       d_lin_feat_flat_.ptr, d_lin_coeff_flat_.ptr, d_raw_.ptr, num_data_,
+                       ScorePtr());
+  } else {
   hipLaunchKernelGGL(hipk::k_score_update, dim3((n + 255) / 256), dim3(256), 0, stream_,
                      d_idx_.ptr, d_sorted_begin_.ptr, nl, n, d_leaf_out_.ptr, ScorePtr());
+  }
   if (bag_indices_ != nullptr && bag_cnt_ > 0 &&
       bag_cnt_ < static_cast<data_size_t>(num_data_)) {
     std::vector<uint32_t> oob;
@@ -3594,6 +3733,91 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
                            ScorePtr());
       }
     }
+  }
+}
+
+void HIPTreeLearner::CalculateLinearDevice(Tree* tree) {
+  tree->SetLinear(true);
+  const int nl = tree->num_leaves();
+  if (nl <= 1) return;
+  // branch-path features per leaf (numerical, unique, inner ids) — host walk
+  std::vector<std::vector<int>> leaf_feats(nl);
+  std::function<void(int, std::vector<int>&)> walk = [&](int node, std::vector<int>& path) {
+    if (node < 0) {
+      leaf_feats[~node] = path;
+      return;
+    }
+    const int fi = tree->split_feature_inner(node);
+    bool added = false;
+    if (!tree->IsCategoricalSplit(node) &&
+        std::find(path.begin(), path.end(), fi) == path.end()) {
+      path.push_back(fi);
+      added = true;
+    }
+    walk(tree->left_child(node), path);
+    walk(tree->right_child(node), path);
+    if (added) path.pop_back();
+  };
+  std::vector<int> path;
+  walk(0, path);
+
+  constexpr int kMaxK = 32;  // LDS/register cap; deeper unique paths stay constant
+  // device Gram accumulation per leaf, then tiny host Cholesky solves
+  for (int l = 0; l < nl; ++l) {
+    const auto& feats = leaf_feats[l];
+    const int k = static_cast<int>(feats.size());
+    if (k == 0 || k > kMaxK || tree->leaf_count(l) < k + 2) continue;
+    const int dim = k + 1;
+    d_lin_feats_.Alloc(kMaxK);
+    d_lin_A_.Alloc((kMaxK + 1) * (kMaxK + 1));
+    d_lin_b_.Alloc(kMaxK + 1);
+    if (!d_lin_nan_.ptr) d_lin_nan_.Alloc(1);
+    HIP_OK(hipMemcpyAsync(d_lin_feats_.ptr, feats.data(), sizeof(int) * k,
+                          hipMemcpyHostToDevice, stream_));
+    HIP_OK(hipMemsetAsync(d_lin_A_.ptr, 0, sizeof(double) * dim * dim, stream_));
+    HIP_OK(hipMemsetAsync(d_lin_b_.ptr, 0, sizeof(double) * dim, stream_));
+    HIP_OK(hipMemsetAsync(d_lin_nan_.ptr, 0, sizeof(int), stream_));
+    const int blocks = std::min(64, (leaf_cnt_[l] + 255) / 256);
+    const size_t lds = sizeof(double) * (dim * dim + dim);
+    hipLaunchKernelGGL(hipk::k_linear_gram, dim3(blocks), dim3(256), lds, stream_,
+                       d_idx_.ptr, leaf_begin_[l], leaf_cnt_[l], d_raw_.ptr, num_data_,
+                       d_lin_feats_.ptr, k, GradPtr(), HessPtr(), d_lin_A_.ptr,
+                       d_lin_b_.ptr, d_lin_nan_.ptr);
+    if (dist_ && Comm().World() > 1) {
+      // shard Gram matrices sum to the global fit (all ranks end identical)
+      Comm().AllReduce(d_lin_A_.ptr, static_cast<size_t>(dim) * dim, stream_);
+      Comm().AllReduce(d_lin_b_.ptr, static_cast<size_t>(dim), stream_);
+    }
+    std::vector<double> A(dim * dim), b(dim);
+    int nan_flag = 0;
+    HIP_OK(hipMemcpyAsync(A.data(), d_lin_A_.ptr, sizeof(double) * dim * dim,
+                          hipMemcpyDeviceToHost, stream_));
+    HIP_OK(hipMemcpyAsync(b.data(), d_lin_b_.ptr, sizeof(double) * dim,
+                          hipMemcpyDeviceToHost, stream_));
+    HIP_OK(hipMemcpyAsync(&nan_flag, d_lin_nan_.ptr, sizeof(int), hipMemcpyDeviceToHost,
+                          stream_));
+    HIP_OK(hipStreamSynchronize(stream_));
+    if (dist_ && Comm().World() > 1) {
+      // a NaN on ANY rank must disable the leaf's linear fit everywhere
+      double nf2 = nan_flag;
+      DevBuf<double> scratch;
+      scratch.Alloc(1);
+      HIP_OK(hipMemcpy(scratch.ptr, &nf2, sizeof(double), hipMemcpyHostToDevice));
+      Comm().AllReduce(scratch.ptr, 1, stream_);
+      HIP_OK(hipStreamSynchronize(stream_));
+      HIP_OK(hipMemcpy(&nf2, scratch.ptr, sizeof(double), hipMemcpyDeviceToHost));
+      nan_flag = nf2 > 0 ? 1 : 0;
+    }
+    if (nan_flag) continue;  // leaves with missing path values stay constant
+    for (int a = 0; a < k; ++a)
+      A[a * dim + a] += config_->linear_lambda + config_->lambda_l2;
+    A[k * dim + k] += config_->lambda_l2;
+    std::vector<double> beta;
+    if (!CholeskySolve(A, b, dim, &beta)) continue;
+    std::vector<int> feats_real(k);
+    for (int j = 0; j < k; ++j) feats_real[j] = train_data_->RealFeatureIndex(feats[j]);
+    std::vector<double> coeffs(beta.begin(), beta.begin() + k);
+    tree->SetLeafLinear(l, beta[k], feats_real, feats, coeffs);
   }
 }
 
@@ -4056,7 +4280,8 @@ TreeLearner* CreateHIP(const Config* cfg) {
   // loud, not silent: features the device split loop does not implement yet fall
   // back to the host serial learner (reference CUDA learner errors similarly)
   auto unsupported = [&]() -> const char* {
-    if (cfg->linear_tree) return "linear_tree";
+    if (cfg->linear_tree && cfg->bagging_freq > 0 && cfg->bagging_fraction < 1.0)
+      return "linear_tree with bagging (out-of-bag linear score updates)";
     return nullptr;
   };
   if (const char* what = unsupported()) {
