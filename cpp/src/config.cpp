@@ -299,6 +299,19 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
     // unknown keys are kept in raw only (tolerated, like the reference's pass-through)
   }
 
+  // master `seed` cascades into every sub-seed not explicitly given
+  // (parity: reference src/io/config.cpp Config::Set seed handling)
+  if (params.count("seed")) {
+    Random rand(seed);
+    const int int_max = 32767;
+    if (!params.count("data_random_seed")) data_random_seed = rand.NextInt(0, int_max);
+    if (!params.count("bagging_seed")) bagging_seed = rand.NextInt(0, int_max);
+    if (!params.count("drop_seed")) drop_seed = rand.NextInt(0, int_max);
+    if (!params.count("feature_fraction_seed")) feature_fraction_seed = rand.NextInt(0, int_max);
+    if (!params.count("objective_seed")) objective_seed = rand.NextInt(0, int_max);
+    if (!params.count("extra_seed")) extra_seed = rand.NextInt(0, int_max);
+  }
+
   // objective aliases
   if (objective == "regression_l2" || objective == "mean_squared_error" || objective == "mse" ||
       objective == "l2_root" || objective == "root_mean_squared_error" || objective == "rmse")

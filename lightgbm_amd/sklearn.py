@@ -110,6 +110,10 @@ class LGBMModel(BaseEstimator):
                 continue
             out[ren.get(k, k)] = v
         out["objective"] = obj
+        if isinstance(out.get("seed"), np.random.RandomState):
+            out["seed"] = int(out["seed"].randint(0, 2**31 - 1))
+        elif hasattr(np.random, "Generator") and isinstance(out.get("seed"), np.random.Generator):
+            out["seed"] = int(out["seed"].integers(0, 2**31 - 1))
         if out.get("bagging_fraction", 1.0) < 1.0 and out.get("bagging_freq", 0) == 0:
             out["bagging_freq"] = 1
         out.setdefault("verbosity", -1)
@@ -192,7 +196,11 @@ class LGBMModel(BaseEstimator):
                                  valid_names=names or None, callbacks=cbs,
                                  fobj=fobj, feval=feval, init_model=init_model)
         self._best_iteration = self._Booster.best_iteration
+        self._objective = self.objective if callable(self.objective) else params["objective"]
         return self
+
+    def __sklearn_is_fitted__(self):
+        return self._Booster is not None
 
     # ------------------------------------------------------------ properties
     @property
@@ -244,7 +252,7 @@ class LGBMModel(BaseEstimator):
                                      pred_contrib=pred_contrib)
 
 
-class LGBMRegressor(LGBMModel, RegressorMixin):
+class LGBMRegressor(RegressorMixin, LGBMModel):
     def fit(self, X, y, sample_weight=None, init_score=None, eval_set=None, **kwargs):
         y = np.asarray(y, dtype=np.float32).ravel()
         return self._fit(X, y, "regression", sample_weight=sample_weight,
@@ -258,7 +266,7 @@ class LGBMRegressor(LGBMModel, RegressorMixin):
         return 1.0 - u / v if v > 0 else 0.0
 
 
-class LGBMClassifier(LGBMModel, ClassifierMixin):
+class LGBMClassifier(ClassifierMixin, LGBMModel):
     def fit(self, X, y, sample_weight=None, init_score=None, eval_set=None, **kwargs):
         y = np.asarray(y).ravel()
         self._classes = np.unique(y)
@@ -267,9 +275,10 @@ class LGBMClassifier(LGBMModel, ClassifierMixin):
         y_enc = np.array([class_to_idx[v] for v in y], dtype=np.float32)
         if self._n_classes <= 2:
             obj = "binary"
+            self._other_params.pop("num_class", None)  # classes re-detected each fit
         else:
             obj = "multiclass"
-            self._other_params.setdefault("num_class", self._n_classes)
+            self._other_params["num_class"] = self._n_classes
         if eval_set is not None:
             if isinstance(eval_set, tuple):
                 eval_set = [eval_set]
