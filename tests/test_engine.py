@@ -1080,3 +1080,104 @@ def test_predict_shape_check():
         bst.predict(X[:5, :6])
     p = bst.predict(X[:5, :6], predict_disable_shape_check=True)
     assert p.shape == (5,)
+
+
+def test_binning_same_sign():
+    """bins split correctly when a feature is all-positive or all-negative
+    (ref test_engine.py:3473)."""
+    x = np.empty((99, 2))
+    x[:, 0] = np.arange(0.01, 1, 0.01)
+    x[:, 1] = -np.arange(0.01, 1, 0.01)
+    y = np.arange(0.01, 1, 0.01)
+    params = {"objective": "regression_l1", "max_bin": 5, "num_leaves": 2,
+              "min_data_in_leaf": 1, "verbosity": -1, "seed": 0}
+    est = lgb.train(params, lgb.Dataset(x, label=y), num_boost_round=20)
+    new_x = np.zeros((3, 2))
+    new_x[:, 0] = [-1, 0, 1]
+    p = est.predict(new_x)
+    assert p[0] == pytest.approx(p[1])       # -1 and 0 both below the positive range
+    assert p[1] != pytest.approx(p[2])
+    new_x = np.zeros((3, 2))
+    new_x[:, 1] = [-1, 0, 1]
+    p = est.predict(new_x)
+    assert p[0] != pytest.approx(p[1])
+    assert p[1] == pytest.approx(p[2])       # 0 and 1 both above the negative range
+
+
+def test_sliced_data():
+    """non-contiguous numpy/CSR slices train identically to contiguous copies
+    (ref test_engine.py:2064)."""
+    from scipy.sparse import csr_matrix
+    rng = np.random.RandomState(5)
+
+    def train_pred(features, labels):
+        ds = lgb.Dataset(features, label=labels)
+        gbm = lgb.train({"objective": "binary", "verbosity": -1, "min_data_in_leaf": 5},
+                        ds, num_boost_round=10)
+        return gbm.predict(features)
+
+    n = 100
+    features = rng.uniform(size=(n, 5))
+    labels = np.append(np.ones(25, dtype=np.float32), np.zeros(75, dtype=np.float32))
+    origin = train_pred(features, labels)
+    sliced_labels = np.column_stack((labels, np.ones(n, dtype=np.float32)))[:, 0]
+    np.testing.assert_allclose(origin, train_pred(features, sliced_labels))
+    stacked = np.column_stack((np.ones(n), np.ones(n), features, np.ones(n), np.ones(n)))
+    stacked = np.vstack([np.ones((2, 9)), stacked, np.ones((2, 9))])
+    sliced = stacked[2:102, 2:7]
+    assert np.all(sliced == features)
+    np.testing.assert_allclose(origin, train_pred(sliced, sliced_labels))
+    sliced_csr = csr_matrix(stacked)[2:102, 2:7]
+    np.testing.assert_allclose(origin, train_pred(sliced_csr, sliced_labels))
+
+
+def test_non_ascii_feature_names():
+    rng = np.random.RandomState(6)
+    X = rng.randn(300, 3)
+    y = X[:, 0]
+    names = ["F_零", "F_一", "渋谷"]
+    ds = lgb.Dataset(X, label=y, feature_name=names)
+    bst = lgb.train({"objective": "regression", "verbosity": -1}, ds, num_boost_round=5)
+    assert bst.feature_name() == names
+    s = bst.model_to_string()
+    bst2 = lgb.Booster(model_str=s)
+    assert bst2.feature_name() == names
+    np.testing.assert_allclose(bst2.predict(X[:20]), bst.predict(X[:20]), rtol=1e-12)
+
+
+def test_get_split_value_histogram():
+    """histogram of split thresholds per feature (ref test_engine.py get_split_value_histogram)."""
+    rng = np.random.RandomState(7)
+    X = rng.randn(2000, 4)
+    y = 3 * X[:, 0] + X[:, 1] + 0.1 * rng.randn(2000)
+    bst = lgb.train({"objective": "regression", "verbosity": -1},
+                    lgb.Dataset(X, label=y), num_boost_round=20)
+    counts, edges = bst.get_split_value_histogram(0)
+    assert counts.sum() > 0
+    assert len(edges) == len(counts) + 1
+    # splits on the dominant feature outnumber a weak feature's
+    c3, _ = bst.get_split_value_histogram(3)
+    assert counts.sum() >= c3.sum()
+    # by name
+    cn, _ = bst.get_split_value_histogram("Column_0")
+    assert cn.sum() == counts.sum()
+
+
+def test_dataset_subset_training():
+    """Dataset.subset trains on the row subset; parent raw data retained
+    (ref test_engine.py test_init_with_subset)."""
+    rng = np.random.RandomState(8)
+    X = rng.uniform(size=(50, 2))
+    y = np.array([1] * 25 + [0] * 25, dtype=np.float32)
+    full = lgb.Dataset(X, y, free_raw_data=False)
+    idx1 = np.sort(rng.choice(np.arange(50), 30, replace=False))
+    sub1 = full.subset(idx1)
+    idx2 = np.sort(rng.choice(np.arange(50), 20, replace=False))
+    sub2 = full.subset(idx2)
+    params = {"objective": "binary", "verbosity": -1, "min_data_in_leaf": 3}
+    m1 = lgb.train(params, sub1, num_boost_round=10, keep_training_booster=True)
+    m2 = lgb.train(params, sub2, num_boost_round=10, init_model=m1)
+    assert m2.num_trees() == 20
+    assert full.get_data().shape[0] == 50
+    assert sub1.num_data() == 30
+    assert sub2.num_data() == 20
