@@ -102,14 +102,28 @@ struct LeafContext {
   double out_hi = std::numeric_limits<double>::infinity();
 };
 
+/*! Per-bin monotone output bounds for monotone_constraints_method=advanced
+ *  ("monotone precise" mode; parity: reference AdvancedLeafConstraints /
+ *  CumulativeFeatureConstraint, monotone_constraints.hpp:143-289 — dense per-bin
+ *  representation instead of the reference's threshold-interval arrays).
+ *  lo[b]/hi[b] bound the output of any child leaf whose feature range covers
+ *  bin b; a candidate child spanning bins [a..c] must lie in
+ *  [max(lo[a..c]), min(hi[a..c])]. */
+struct MonoAdvBounds {
+  std::vector<double> lo, hi;  // size num_numeric_bin
+  bool empty() const { return lo.empty(); }
+};
+
 /*! Best numerical threshold for one feature.
  *  hist: (g,h) pairs, num_bin entries; nan_bin: index of the NaN/Zero bin or -1.
  *  num_numeric_bin: bins eligible as thresholds (nan bin excluded).
- *  rand_threshold: if >=0 (extra_trees), only this bin index is evaluated. */
+ *  rand_threshold: if >=0 (extra_trees), only this bin index is evaluated.
+ *  adv: per-threshold monotone bounds (advanced mode) — when non-null they
+ *  REPLACE the scalar leaf bounds for left/right output clamping. */
 void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric_bin,
                                 int nan_bin, const LeafContext& leaf, const Config& cfg,
                                 int8_t monotone_constraint, int rand_threshold,
-                                SplitInfo* out);
+                                SplitInfo* out, const MonoAdvBounds* adv = nullptr);
 
 /*! Best categorical split (one-hot or sorted-subset scan).
  *  Emits a bin-level bitset in out->cat_bitset_inner. */

@@ -216,7 +216,6 @@ class SerialTreeLearner : public TreeLearner {
   // interleaved (g,h) pair array the varying-hessian row-wise loop reads
   int tree_const_hess_ = -1;
   std::vector<score_t> gh_;
-  bool warned_mc_method_ = false;
   // LOCAL (pre-reduce) leaf gradient totals of the last ComputeHistogram call;
   // distributed learners use them to materialize default bins before reducing
   double local_leaf_sum_g_ = 0.0, local_leaf_sum_h_ = 0.0;
@@ -252,6 +251,19 @@ class SerialTreeLearner : public TreeLearner {
                   int split_feature, const SplitInfo& s, bool use_left,
                   bool use_right, uint32_t split_threshold,
                   std::vector<int>* leaves_to_update);
+  // ---- advanced ("monotone precise") mode: per-(leaf,feature) per-bin output
+  // bounds recomputed from the CURRENT tree at scan time (reference
+  // AdvancedLeafConstraints GoUp/GoDownToFindConstrainingLeaves,
+  // monotone_constraints.hpp:858-1178 — fresh dense-per-bin design)
+  bool mono_advanced_ = false;
+  const Tree* mono_tree_ = nullptr;         // current tree during Train
+  void MonoAdvBoundsForFeature(int leaf, int feature_inner, int num_numeric_bin,
+                               MonoAdvBounds* out) const;
+  void MonoAdvGoDown(const Tree* tree, int node, int feature_inner, bool want_min,
+                     int s, int e, const std::vector<int>& up_feats,
+                     const std::vector<uint32_t>& up_thresholds,
+                     const std::vector<uint8_t>& up_was_right,
+                     MonoAdvBounds* out) const;
   std::vector<SplitInfo> best_split_per_leaf_;
   std::vector<LeafContext> leaf_ctx_;
   std::vector<int8_t> is_feature_used_;     // per-tree mask

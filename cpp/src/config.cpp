@@ -353,6 +353,20 @@ void Config::Set(const std::unordered_map<std::string, std::string>& params_in) 
     if (zero_as_missing)
       Log::Fatal("zero_as_missing must be false when fitting linear trees");
   }
+  // intermediate/advanced monotone constraints need full local histograms and
+  // stable per-node feature sets (reference CheckParamConflict, config.cpp:449-458)
+  if (monotone_constraints_method == "intermediate" ||
+      monotone_constraints_method == "advanced") {
+    if (tree_learner != "serial" && num_machines > 1) {
+      Log::Warning("Cannot use intermediate/advanced monotone constraints in "
+                   "distributed learning; falling back to method=basic");
+      monotone_constraints_method = "basic";
+    } else if (feature_fraction_bynode != 1.0) {
+      Log::Warning("feature_fraction_bynode is incompatible with intermediate/"
+                   "advanced monotone constraints; falling back to method=basic");
+      monotone_constraints_method = "basic";
+    }
+  }
   if (num_threads > 0) omp_set_num_threads(num_threads);
   // only an EXPLICIT verbosity touches the global log level: lazily-constructed
   // datasets with default params must not undo a booster's verbosity=-1

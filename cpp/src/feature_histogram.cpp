@@ -10,7 +10,7 @@ namespace migbm {
 void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric_bin,
                                 int nan_bin, const LeafContext& leaf, const Config& cfg,
                                 int8_t monotone_constraint, int rand_threshold,
-                                SplitInfo* out) {
+                                SplitInfo* out, const MonoAdvBounds* adv) {
   const double l1 = cfg.lambda_l1, l2 = cfg.lambda_l2;
   const double mds = cfg.max_delta_step, smooth = cfg.path_smooth;
   const double min_hess = cfg.min_sum_hessian_in_leaf;
@@ -31,6 +31,24 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
   int best_t = -1;
   bool best_default_left = false;
   double best_lg = 0, best_lh = 0;
+  double best_llo = leaf.out_lo, best_lhi = leaf.out_hi;
+  double best_rlo = leaf.out_lo, best_rhi = leaf.out_hi;
+
+  // advanced monotone: suffix extrema so right-child bounds over bins [t+1..end)
+  // are O(1) per threshold; left-child extrema run forward with the scan
+  std::vector<double> rmax_lo, rmin_hi;
+  if (adv != nullptr && !adv->empty()) {
+    rmax_lo.resize(num_numeric_bin + 1);
+    rmin_hi.resize(num_numeric_bin + 1);
+    rmax_lo[num_numeric_bin] = -std::numeric_limits<double>::infinity();
+    rmin_hi[num_numeric_bin] = std::numeric_limits<double>::infinity();
+    for (int b = num_numeric_bin - 1; b >= 0; --b) {
+      rmax_lo[b] = std::max(rmax_lo[b + 1], adv->lo[b]);
+      rmin_hi[b] = std::min(rmin_hi[b + 1], adv->hi[b]);
+    }
+  }
+  double run_lo = -std::numeric_limits<double>::infinity();
+  double run_hi = std::numeric_limits<double>::infinity();
 
   // prefix over numeric bins: for threshold t, left(no-missing) = sum bins 0..t
   double gl = 0.0, hl = 0.0;
@@ -38,6 +56,14 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
   for (int t = 0; t <= t_max; ++t) {
     gl += hist[2 * t];
     hl += hist[2 * t + 1];
+    double left_lo = leaf.out_lo, left_hi = leaf.out_hi;
+    double right_lo = leaf.out_lo, right_hi = leaf.out_hi;
+    if (!rmax_lo.empty()) {
+      run_lo = std::max(run_lo, adv->lo[t]);
+      run_hi = std::min(run_hi, adv->hi[t]);
+      left_lo = run_lo; left_hi = run_hi;
+      right_lo = rmax_lo[t + 1]; right_hi = rmin_hi[t + 1];
+    }
     if (rand_threshold >= 0 && t != rand_threshold) continue;
     // two missing placements (if missing exists); else single evaluation
     const int n_variants = has_nan ? 2 : 1;
@@ -55,9 +81,10 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
                                                         leaf.parent_output);
       double ro = GainMath::CalculateSplittedLeafOutput(sgr, shr, l1, l2, mds, smooth, rc,
                                                         leaf.parent_output);
-      // BasicLeafConstraints: clamp candidate outputs into the leaf's inherited bounds
-      lo = std::min(std::max(lo, leaf.out_lo), leaf.out_hi);
-      ro = std::min(std::max(ro, leaf.out_lo), leaf.out_hi);
+      // clamp candidate outputs into the inherited bounds (basic/intermediate:
+      // one scalar pair per leaf; advanced: per-threshold piecewise bounds)
+      lo = std::min(std::max(lo, left_lo), left_hi);
+      ro = std::min(std::max(ro, right_lo), right_hi);
       if (monotone_constraint != 0) {
         if (monotone_constraint > 0 && lo > ro) continue;
         if (monotone_constraint < 0 && lo < ro) continue;
@@ -70,6 +97,8 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
         best_t = t;
         best_default_left = missing_left;
         best_lg = sgl; best_lh = shl;
+        best_llo = left_lo; best_lhi = left_hi;
+        best_rlo = right_lo; best_rhi = right_hi;
       }
     }
   }
@@ -88,8 +117,8 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
   out->right_output = GainMath::CalculateSplittedLeafOutput(
       out->right_sum_gradient, out->right_sum_hessian, l1, l2, mds, smooth, out->right_count,
       leaf.parent_output);
-  out->left_output = std::min(std::max(out->left_output, leaf.out_lo), leaf.out_hi);
-  out->right_output = std::min(std::max(out->right_output, leaf.out_lo), leaf.out_hi);
+  out->left_output = std::min(std::max(out->left_output, best_llo), best_lhi);
+  out->right_output = std::min(std::max(out->right_output, best_rlo), best_rhi);
   out->monotone_type = monotone_constraint;
 }
 

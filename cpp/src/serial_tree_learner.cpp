@@ -461,8 +461,13 @@ void SerialTreeLearner::FindBestSplitForLeaf(int leaf, const LeafContext& ctx) {
         if (orig < static_cast<int>(config_->monotone_constraints.size()))
           mono = static_cast<int8_t>(config_->monotone_constraints[orig]);
       }
+      MonoAdvBounds adv;
+      if (mono_advanced_ && leaf < static_cast<int>(mono_leaf_in_subtree_.size()) &&
+          mono_leaf_in_subtree_[leaf])
+        MonoAdvBoundsForFeature(leaf, f, m->num_numeric_bin(), &adv);
       FindBestThresholdNumerical(fh, m->num_bin(), m->num_numeric_bin(), m->nan_bin(), ctx,
-                                 *config_, mono, rand_t, &cand[f]);
+                                 *config_, mono, rand_t, &cand[f],
+                                 adv.empty() ? nullptr : &adv);
     }
     cand[f].feature = f;
   }
@@ -597,15 +602,9 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
   tree_const_hess_ = -1;  // re-classify hessians for this tree's gradients
   mono_intermediate_ = !config_->monotone_constraints.empty() &&
                        config_->monotone_constraints_method != "basic";
+  mono_advanced_ = mono_intermediate_ &&
+                   config_->monotone_constraints_method == "advanced";
   if (mono_intermediate_) {
-    // "advanced" maps to the intermediate policy (per-threshold piecewise
-    // constraints are not implemented; intermediate already recomputes
-    // affected splits with output-tight bounds)
-    if (config_->monotone_constraints_method == "advanced" && !warned_mc_method_) {
-      Log::Warning("monotone_constraints_method=advanced runs the intermediate "
-                   "policy (output-tight bounds + contiguous-leaf re-evaluation)");
-      warned_mc_method_ = true;
-    }
     mono_node_parent_.assign(config_->num_leaves, -1);
     mono_leaf_in_subtree_.assign(config_->num_leaves, 0);
   }
@@ -650,6 +649,7 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
   is_feature_used_ = SampleFeatures(false);
 
   auto tree = std::make_unique<Tree>(config_->num_leaves);
+  mono_tree_ = tree.get();  // advanced monotone bounds walk the growing tree
   partition_.ResetToRoot(bag_indices_, bag_cnt_);
   std::fill(leaf_to_slot_.begin(), leaf_to_slot_.end(), -1);
   std::fill(slot_owner_.begin(), slot_owner_.end(), -1);
@@ -900,6 +900,7 @@ Tree* SerialTreeLearner::Train(const score_t* gradients, const score_t* hessians
     }
   }
   if (config_->linear_tree) CalculateLinear(tree.get());
+  mono_tree_ = nullptr;
   return tree.release();
 }
 
@@ -1052,7 +1053,10 @@ void SerialTreeLearner::MonoGoDown(const Tree* tree, int node,
         changed = true;
       }
     }
-    if (changed) {
+    // advanced mode recomputes bounds fresh at scan time, so a visited leaf's
+    // constraints may have changed (even RELAXED) regardless of the scalar
+    // tightening above (reference AdvancedConstraintEntry always reports change)
+    if (changed || mono_advanced_) {
       leaves_to_update->push_back(leaf);
       if (MonoDebug())
         fprintf(stderr, "[mono]   update leaf=%d out=%.6g -> bounds=[%.4g,%.4g]\n", leaf,
@@ -1091,6 +1095,127 @@ void SerialTreeLearner::MonoGoDown(const Tree* tree, int node,
     MonoGoDown(tree, tree->right_child(node), up_feats, up_thresholds, up_was_right,
                update_max, split_feature, s, use_left && right_child_sees_left,
                use_right, split_threshold, leaves_to_update);
+  }
+}
+
+/*! advanced ("monotone precise") mode — descend the subtree OPPOSITE a monotone
+ *  ancestor and fold every contiguous leaf's output into the per-bin bound
+ *  arrays over the bin interval [s, e) of the scanned feature that the leaf can
+ *  actually constrain. Splits on the scanned feature narrow the interval (an
+ *  empty interval prunes the branch — this is what keeps only boundary-adjacent
+ *  regions alive when the monotone feature IS the scanned feature); splits on
+ *  other features use the same contiguity fences as MonoGoDown.
+ *  (reference GoDownToFindConstrainingLeaves, monotone_constraints.hpp:1002) */
+void SerialTreeLearner::MonoAdvGoDown(const Tree* tree, int node, int f, bool want_min,
+                                      int s, int e, const std::vector<int>& up_feats,
+                                      const std::vector<uint32_t>& up_thresholds,
+                                      const std::vector<uint8_t>& up_was_right,
+                                      MonoAdvBounds* out) const {
+  if (s >= e) return;
+  if (node < 0) {
+    const double v = tree->LeafOutput(~node);
+    if (want_min) {
+      for (int b = s; b < e; ++b) out->lo[b] = std::max(out->lo[b], v);
+    } else {
+      for (int b = s; b < e; ++b) out->hi[b] = std::min(out->hi[b], v);
+    }
+    return;
+  }
+  const int feat = tree->split_feature_inner(node);
+  const uint32_t thr = tree->threshold_in_bin(node);
+  const bool numerical = !tree->IsCategoricalSplit(node);
+  if (numerical && feat == f) {
+    MonoAdvGoDown(tree, tree->left_child(node), f, want_min, s,
+                  std::min<int>(static_cast<int>(thr) + 1, e), up_feats, up_thresholds,
+                  up_was_right, out);
+    MonoAdvGoDown(tree, tree->right_child(node), f, want_min,
+                  std::max<int>(static_cast<int>(thr) + 1, s), e, up_feats, up_thresholds,
+                  up_was_right, out);
+    return;
+  }
+  bool go_left = true, go_right = true;
+  if (numerical) {
+    for (size_t i = 0; i < up_feats.size(); ++i) {
+      if (up_feats[i] != feat) continue;
+      if (thr >= up_thresholds[i] && !up_was_right[i]) go_right = false;
+      if (thr <= up_thresholds[i] && up_was_right[i]) go_left = false;
+      if (!go_left && !go_right) break;
+    }
+  }
+  if (go_left)
+    MonoAdvGoDown(tree, tree->left_child(node), f, want_min, s, e, up_feats,
+                  up_thresholds, up_was_right, out);
+  if (go_right)
+    MonoAdvGoDown(tree, tree->right_child(node), f, want_min, s, e, up_feats,
+                  up_thresholds, up_was_right, out);
+}
+
+/*! per-(leaf, feature) bound computation for the advanced monotone mode: walk UP
+ *  from the leaf; at every monotone ancestor whose opposite subtree lies on the
+ *  constraining side, walk DOWN it collecting contiguous leaves' outputs into
+ *  dense per-bin min/max arrays. Recomputed fresh from the current tree at every
+ *  scan, so bounds RELAX when a constraining leaf is split into finer regions —
+ *  the point of the precise mode. (reference AdvancedConstraintEntry::
+ *  RecomputeConstraintsIfNeeded + GoUpToFindConstrainingLeaves) */
+void SerialTreeLearner::MonoAdvBoundsForFeature(int leaf, int f, int num_numeric_bin,
+                                                MonoAdvBounds* out) const {
+  out->lo.clear();
+  out->hi.clear();
+  const Tree* tree = mono_tree_;
+  if (tree == nullptr || tree->leaf_parent(leaf) < 0) return;
+  out->lo.assign(num_numeric_bin, -std::numeric_limits<double>::infinity());
+  out->hi.assign(num_numeric_bin, std::numeric_limits<double>::infinity());
+  for (int pass = 0; pass < 2; ++pass) {
+    const bool want_min = (pass == 0);
+    int it_start = 0, it_end = num_numeric_bin;
+    std::vector<int> up_feats;
+    std::vector<uint32_t> up_thr;
+    std::vector<uint8_t> up_right;
+    int node = ~leaf;  // child handle as stored in the parent's child slots
+    int parent = tree->leaf_parent(leaf);
+    while (parent >= 0) {
+      const int feat = tree->split_feature_inner(parent);
+      const bool p_num = !tree->IsCategoricalSplit(parent);
+      const bool is_right = tree->right_child(parent) == node;
+      const uint32_t thr = tree->threshold_in_bin(parent);
+      if (p_num && feat == f) {
+        // narrow to the leaf's own bin range; the right-child case keeps bin
+        // `thr` itself so boundary-adjacent opposite regions stay non-empty
+        if (is_right) it_start = std::max<int>(it_start, static_cast<int>(thr));
+        else it_end = std::min<int>(it_end, static_cast<int>(thr) + 1);
+      }
+      bool should_descend = p_num;
+      if (p_num) {
+        for (size_t i = 0; i < up_feats.size(); ++i) {
+          if (up_feats[i] == feat && (up_right[i] != 0) == is_right) {
+            should_descend = false;
+            break;
+          }
+        }
+      }
+      if (should_descend) {
+        const int orig_f = tree->split_feature(parent);
+        int8_t mono = 0;
+        if (orig_f >= 0 && orig_f < static_cast<int>(config_->monotone_constraints.size()))
+          mono = static_cast<int8_t>(config_->monotone_constraints[orig_f]);
+        if (mono != 0) {
+          // increasing (mono>0) with the leaf on the right: the opposite (left)
+          // subtree lies BELOW -> it contributes lower bounds (want_min pass)
+          const bool opposite_is_lower = (mono > 0) == is_right;
+          if (opposite_is_lower == want_min) {
+            const int opposite =
+                is_right ? tree->left_child(parent) : tree->right_child(parent);
+            MonoAdvGoDown(tree, opposite, f, want_min, it_start, it_end, up_feats,
+                          up_thr, up_right, out);
+          }
+        }
+        up_right.push_back(is_right ? 1 : 0);
+        up_thr.push_back(thr);
+        up_feats.push_back(feat);
+      }
+      node = parent;
+      parent = mono_node_parent_[parent];
+    }
   }
 }
 

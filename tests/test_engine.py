@@ -1181,3 +1181,39 @@ def test_dataset_subset_training():
     assert full.get_data().shape[0] == 50
     assert sub1.num_data() == 30
     assert sub2.num_data() == 20
+
+
+def test_monotone_advanced_precise_mode():
+    """monotone_constraints_method=advanced (monotone precise): per-threshold
+    piecewise bounds recomputed from the live tree (reference
+    AdvancedLeafConstraints). Must stay globally monotone, must actually differ
+    from the intermediate policy, and the extra split freedom must not hurt fit."""
+    rng = np.random.RandomState(3)
+    n = 30000
+    X = rng.rand(n, 4)
+    y = (3.0 * X[:, 0] ** 2 - 2.0 * X[:, 1] + 1.5 * np.sin(7 * X[:, 2]) * X[:, 0] +
+         0.1 * rng.randn(n)).astype(np.float32)
+    models = {}
+    for m in ("basic", "intermediate", "advanced"):
+        params = {"objective": "regression", "verbosity": -1, "num_leaves": 127,
+                  "learning_rate": 0.15, "min_data_in_leaf": 5,
+                  "monotone_constraints": [1, -1, 0, 0],
+                  "monotone_constraints_method": m}
+        models[m] = lgb.train(params, lgb.Dataset(X, label=y), 50)
+    # dense monotonicity audit over random slices
+    xs = np.linspace(0.01, 0.99, 60)
+    slice_rng = np.random.RandomState(17)
+    for _ in range(12):
+        o = slice_rng.rand(3)
+        g_up = np.column_stack([xs, np.full(60, o[0]), np.full(60, o[1]),
+                                np.full(60, o[2])])
+        assert np.all(np.diff(models["advanced"].predict(g_up)) >= -1e-9)
+        g_dn = np.column_stack([np.full(60, o[0]), xs, np.full(60, o[1]),
+                                np.full(60, o[2])])
+        assert np.all(np.diff(models["advanced"].predict(g_dn)) <= 1e-9)
+    # the precise policy must be a distinct tree-growth policy
+    assert not np.allclose(models["advanced"].predict(X[:2000]),
+                           models["intermediate"].predict(X[:2000]))
+    # and its looser-but-exact bounds must fit at least as well as intermediate
+    mse = {m: float(np.mean((b.predict(X) - y) ** 2)) for m, b in models.items()}
+    assert mse["advanced"] <= mse["intermediate"] * 1.02, mse
