@@ -50,9 +50,12 @@ void FindBestThresholdNumerical(const hist_t* hist, int num_bin, int num_numeric
   double run_lo = -std::numeric_limits<double>::infinity();
   double run_hi = std::numeric_limits<double>::infinity();
 
-  // prefix over numeric bins: for threshold t, left(no-missing) = sum bins 0..t
+  // prefix over numeric bins: for threshold t, left(no-missing) = sum bins 0..t.
+  // Without missing the last numeric bin can't be a threshold; WITH a NaN bin it
+  // can — left = all numeric values, right = missing only (the reference's
+  // everything-vs-NaN split, required e.g. for constant columns with NaNs)
   double gl = 0.0, hl = 0.0;
-  const int t_max = num_numeric_bin - 2;  // last numeric bin can't be a threshold
+  const int t_max = has_nan ? num_numeric_bin - 1 : num_numeric_bin - 2;
   for (int t = 0; t <= t_max; ++t) {
     gl += hist[2 * t];
     hl += hist[2 * t + 1];

@@ -1031,7 +1031,9 @@ __global__ void __launch_bounds__(64) k_best_feat(
     }
   } else {
     const int nb = m.num_numeric_bin;
-    const int t_max = nb - 2;
+    // with a NaN bin the last numeric bin IS a valid threshold: left = all
+    // numeric values, right = missing only (all-numeric-vs-NaN split)
+    const int t_max = has_nan ? nb - 1 : nb - 2;
     double carry_g = 0, carry_h = 0;
     for (int chunk = 0; chunk * 64 < nb; ++chunk) {
       const int b = chunk * 64 + lane;
@@ -2972,6 +2974,11 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
       if (use_mono_) {
         d_mono_.Alloc(nf_);
         HIP_OK(hipMemcpy(d_mono_.ptr, mono_host.data(), nf_, hipMemcpyHostToDevice));
+        if (config_->monotone_constraints_method != "basic")
+          Log::Warning("device learner enforces monotone constraints with the "
+                       "basic policy; use device_type=cpu for "
+                       "monotone_constraints_method=%s",
+                       config_->monotone_constraints_method.c_str());
       }
     }
   }

@@ -1217,3 +1217,130 @@ def test_monotone_advanced_precise_mode():
     # and its looser-but-exact bounds must fit at least as well as intermediate
     mse = {m: float(np.mean((b.predict(X) - y) ** 2)) for m, b in models.items()}
     assert mse["advanced"] <= mse["intermediate"] * 1.02, mse
+
+
+def test_missing_value_handle_dense_default():
+    """NaN rows learn their own default direction (ref test_engine.py:~test_missing_value_handle)."""
+    rng = np.random.RandomState(21)
+    X = np.zeros((100, 1))
+    y = np.zeros(100)
+    trues = rng.choice(100, 20, replace=False)
+    X[trues, 0] = np.nan
+    y[trues] = 1
+    ds = lgb.Dataset(X, label=y)
+    res = {}
+    bst = lgb.train({"objective": "regression", "metric": "l2", "verbosity": -1,
+                     "boost_from_average": False}, ds, 20,
+                    valid_sets=[lgb.Dataset(X, label=y)],
+                    callbacks=[lgb.record_evaluation(res)])
+    mse = float(np.mean((bst.predict(X) - y) ** 2))
+    assert mse < 0.005
+    assert res["valid_0"]["l2"][-1] == pytest.approx(mse)
+
+
+def test_missing_value_handle_more_na():
+    """majority-NaN column: NaN becomes the most-frequent bin yet still separates."""
+    rng = np.random.RandomState(22)
+    X = np.ones((100, 1))
+    y = np.ones(100)
+    trues = rng.choice(100, 80, replace=False)
+    X[trues, 0] = np.nan
+    y[trues] = 0
+    bst = lgb.train({"objective": "regression", "metric": "l2", "verbosity": -1,
+                     "boost_from_average": False}, lgb.Dataset(X, label=y), 20)
+    assert float(np.mean((bst.predict(X) - y) ** 2)) < 0.005
+
+
+def test_missing_value_handle_na():
+    """one split separates {0..3, NaN} from {4..7} with NaN default-left."""
+    x = [0, 1, 2, 3, 4, 5, 6, 7, np.nan]
+    y = [1, 1, 1, 1, 0, 0, 0, 0, 1]
+    X = np.array(x).reshape(-1, 1)
+    params = {"objective": "regression", "verbosity": -1, "boost_from_average": False,
+              "min_data_in_leaf": 1, "num_leaves": 2, "learning_rate": 1,
+              "min_data_in_bin": 1, "zero_as_missing": False}
+    bst = lgb.train(params, lgb.Dataset(X, label=np.array(y, dtype=np.float64)), 1)
+    np.testing.assert_allclose(bst.predict(X), y, atol=1e-9)
+
+
+def test_missing_value_handle_zero():
+    """zero_as_missing=True routes both 0 and NaN through the missing bin."""
+    x = [0, 1, 2, 3, 4, 5, 6, 7, np.nan]
+    y = [0, 1, 1, 1, 0, 0, 0, 0, 0]
+    X = np.array(x).reshape(-1, 1)
+    params = {"objective": "regression", "verbosity": -1, "boost_from_average": False,
+              "min_data_in_leaf": 1, "num_leaves": 2, "learning_rate": 1,
+              "min_data_in_bin": 1, "zero_as_missing": True}
+    bst = lgb.train(params, lgb.Dataset(X, label=np.array(y, dtype=np.float64)), 1)
+    np.testing.assert_allclose(bst.predict(X), y, atol=1e-9)
+
+
+def test_missing_value_handle_none():
+    """use_missing=False: NaN is treated exactly like 0."""
+    x = [0, 1, 2, 3, 4, 5, 6, 7, np.nan]
+    y = [0, 1, 1, 1, 0, 0, 0, 0, 0]
+    X = np.array(x).reshape(-1, 1)
+    params = {"objective": "regression", "verbosity": -1, "boost_from_average": False,
+              "min_data_in_leaf": 1, "num_leaves": 2, "learning_rate": 1,
+              "min_data_in_bin": 1, "use_missing": False}
+    bst = lgb.train(params, lgb.Dataset(X, label=np.array(y, dtype=np.float64)), 1)
+    pred = bst.predict(X)
+    assert pred[-1] == pytest.approx(pred[0])  # NaN row == 0 row
+    from sklearn.metrics import roc_auc_score
+    assert roc_auc_score(y, pred) > 0.8
+
+
+def test_predict_stump():
+    """num_leaves>=2 is unreachable on pure-noise root: stump tree predicts the
+    boosted average everywhere (ref test_predict_stump / boost_from_average)."""
+    rng = np.random.RandomState(23)
+    X = rng.randn(500, 3)
+    y = np.full(500, 3.7)
+    bst = lgb.train({"objective": "regression", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 5)
+    pred = bst.predict(rng.randn(20, 3))
+    np.testing.assert_allclose(pred, 3.7, rtol=1e-6)
+    d = bst.dump_model()
+    assert d["tree_info"][0]["num_leaves"] == 1
+
+
+def test_small_max_bin():
+    """max_bin=2 trains and still separates a two-sided signal (ref test_small_max_bin)."""
+    rng = np.random.RandomState(24)
+    X = rng.randn(2000, 3)
+    y = (X[:, 0] > 0).astype(np.float64)
+    # even bin counts place a quantile boundary at the median (~0 here); an odd
+    # count puts boundaries at the tertiles, capping accuracy near 0.83
+    for mb, floor in ((2, 0.95), (3, 0.80), (4, 0.95)):
+        bst = lgb.train({"objective": "binary", "verbosity": -1, "max_bin": mb},
+                        lgb.Dataset(X, label=y), 20)
+        pred = bst.predict(X)
+        assert ((pred > 0.5) == y).mean() > floor, mb
+
+
+def test_equal_predict_from_row_major_and_col_major():
+    """C-order and F-order float64 matrices produce identical datasets and
+    predictions (ref test_equal_predict_from_row_major_and_col_major_data)."""
+    rng = np.random.RandomState(25)
+    Xc = np.ascontiguousarray(rng.randn(2000, 6))
+    Xf = np.asfortranarray(Xc)
+    y = Xc[:, 0] + 0.1 * rng.randn(2000)
+    pc = lgb.train({"objective": "regression", "verbosity": -1},
+                   lgb.Dataset(Xc, label=y), 10).predict(Xc)
+    pf = lgb.train({"objective": "regression", "verbosity": -1},
+                   lgb.Dataset(Xf, label=y), 10).predict(Xf)
+    np.testing.assert_allclose(pc, pf, rtol=1e-12)
+
+
+def test_model_size_large_roundtrip():
+    """a многи-tree model string round-trips exactly (ref test_model_size)."""
+    rng = np.random.RandomState(26)
+    X = rng.randn(3000, 10)
+    y = X @ rng.randn(10) + 0.1 * rng.randn(3000)
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "num_leaves": 63},
+                    lgb.Dataset(X, label=y), 100)
+    s = bst.model_to_string()
+    assert len(s) > 100_000
+    bst2 = lgb.Booster(model_str=s)
+    np.testing.assert_allclose(bst2.predict(X[:200]), bst.predict(X[:200]), rtol=1e-12)
+    assert bst2.num_trees() == 100
