@@ -221,7 +221,7 @@ def _make_n_folds(full_data, nfold, params, seed, stratified, shuffle):
 def cv(params, train_set, num_boost_round=100, folds=None, nfold=5, stratified=True,
        shuffle=True, metrics=None, feval=None, init_model=None,
        callbacks=None, eval_train_metric=False, return_cvbooster=False, seed=0,
-       fobj=None):
+       fobj=None, fpreproc=None):
     """Cross-validation (parity: reference engine.py:cv)."""
     params = copy.deepcopy(params) if params else {}
     if metrics is not None:
@@ -242,7 +242,12 @@ def cv(params, train_set, num_boost_round=100, folds=None, nfold=5, stratified=T
     for (train_idx, test_idx) in folds:
         tr = train_set.subset(sorted(train_idx))
         te = train_set.subset(sorted(test_idx))
-        bst = Booster(params=params, train_set=tr)
+        fold_params = params
+        if fpreproc is not None:
+            # per-fold preprocessing hook: (train, test, params) -> same triple
+            # (reference engine.py cv fpreproc)
+            tr, te, fold_params = fpreproc(tr, te, copy.deepcopy(params))
+        bst = Booster(params=fold_params, train_set=tr)
         bst.add_valid(te, "valid")
         cvbooster._append(bst)
         fold_data.append((tr, te))

@@ -1344,3 +1344,129 @@ def test_model_size_large_roundtrip():
     bst2 = lgb.Booster(model_str=s)
     np.testing.assert_allclose(bst2.predict(X[:200]), bst.predict(X[:200]), rtol=1e-12)
     assert bst2.num_trees() == 100
+
+
+def test_sparse_train_predict_contrib_consistency():
+    """CSR/CSC train + predict + contribs agree with dense (ref test_contribs_sparse)."""
+    from scipy.sparse import csc_matrix, csr_matrix
+    rng = np.random.RandomState(27)
+    X = rng.randn(2000, 6)
+    X[rng.rand(2000, 6) < 0.6] = 0.0
+    y = (X[:, 0] + X[:, 1] > 0).astype(np.float64)
+    p_dense = lgb.train({"objective": "binary", "verbosity": -1},
+                        lgb.Dataset(X, label=y), 15).predict(X)
+    for sp in (csr_matrix, csc_matrix):
+        bst = lgb.train({"objective": "binary", "verbosity": -1},
+                        lgb.Dataset(sp(X), label=y), 15)
+        np.testing.assert_allclose(bst.predict(sp(X)), p_dense, rtol=1e-10)
+        contrib = bst.predict(sp(X), pred_contrib=True)
+        assert np.asarray(contrib).shape == (2000, 7)
+        raw = bst.predict(sp(X), raw_score=True)
+        np.testing.assert_allclose(np.asarray(contrib).sum(axis=1),
+                                   np.asarray(raw).ravel(), rtol=1e-6)
+
+
+def test_validate_features():
+    """predict(validate_features=True) rejects renamed DataFrame columns
+    (ref test_validate_features)."""
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(28)
+    df = pd.DataFrame(rng.randn(500, 3), columns=["a", "b", "c"])
+    y = df["a"] > 0
+    bst = lgb.train({"objective": "binary", "verbosity": -1},
+                    lgb.Dataset(df, label=y.astype(float)), 5)
+    bst.predict(df, validate_features=True)  # matching names pass
+    bad = df.rename(columns={"b": "bad_name"})
+    with pytest.raises(Exception, match="[Ff]eature|name"):
+        bst.predict(bad, validate_features=True)
+
+
+def test_cv_fpreproc():
+    """cv(fpreproc=...) runs the per-fold hook (ref test_fpreproc)."""
+    rng = np.random.RandomState(29)
+    X = rng.randn(900, 4)
+    y = X[:, 0] + 0.1 * rng.randn(900)
+    seen = []
+
+    def fpreproc(tr, te, params):
+        seen.append(1)
+        params["learning_rate"] = 0.05
+        return tr, te, params
+
+    res = lgb.cv({"objective": "regression", "metric": "l2", "verbosity": -1},
+                 lgb.Dataset(X, label=y), num_boost_round=5, nfold=3,
+                 fpreproc=fpreproc)
+    assert len(seen) == 3
+    assert len(res["valid l2-mean"]) == 5
+
+
+def test_categorical_handle_exact():
+    """pure-categorical signal fits exactly with per-category leaves
+    (ref test_categorical_handle)."""
+    n = 400
+    rng = np.random.RandomState(30)
+    x = rng.randint(0, 8, n).astype(np.float64)
+    lut = rng.randn(8)
+    y = lut[x.astype(int)]
+    ds = lgb.Dataset(x.reshape(-1, 1), label=y, categorical_feature=[0])
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "min_data_in_leaf": 1,
+                     "min_data_per_group": 1, "cat_smooth": 1e-3, "cat_l2": 0.0,
+                     "learning_rate": 1.0, "num_leaves": 16}, ds, 10)
+    pred = bst.predict(x.reshape(-1, 1))
+    np.testing.assert_allclose(pred, y, atol=1e-3)
+
+
+def test_continue_train_dart_and_multiclass():
+    """init_model continuation under dart and multiclass (ref test_continue_train_*)."""
+    rng = np.random.RandomState(31)
+    X = rng.randn(1500, 4)
+    y3 = rng.randint(0, 3, 1500)
+    p = {"objective": "multiclass", "num_class": 3, "verbosity": -1}
+    m1 = lgb.train(p, lgb.Dataset(X, label=y3.astype(float)), 5)
+    m2 = lgb.train(p, lgb.Dataset(X, label=y3.astype(float)), 5, init_model=m1)
+    assert m2.num_trees() == 30  # (5+5) iterations x 3 classes
+    yd = X[:, 0] + 0.1 * rng.randn(1500)
+    pd_ = {"objective": "regression", "boosting": "dart", "verbosity": -1}
+    d1 = lgb.train(pd_, lgb.Dataset(X, label=yd), 5)
+    d2 = lgb.train(pd_, lgb.Dataset(X, label=yd), 5, init_model=d1)
+    assert d2.num_trees() == 10
+
+
+def test_refit_one_tree_variants():
+    """refit keeps structure, renews outputs, for 1-tree reg/binary/multiclass
+    (ref test_refit_with_one_tree_*)."""
+    rng = np.random.RandomState(32)
+    X = rng.randn(1000, 4)
+    cases = [
+        ({"objective": "regression", "verbosity": -1}, X[:, 0]),
+        ({"objective": "binary", "verbosity": -1}, (X[:, 0] > 0).astype(float)),
+        ({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+         rng.randint(0, 3, 1000).astype(float)),
+    ]
+    for params, y in cases:
+        bst = lgb.train(params, lgb.Dataset(X, label=y), 1)
+        y2 = np.roll(y, 137)
+        new = bst.refit(X, y2, decay_rate=0.5)
+        assert new.num_trees() == bst.num_trees()
+        d_old = bst.dump_model()["tree_info"][0]["tree_structure"]
+        d_new = new.dump_model()["tree_info"][0]["tree_structure"]
+        if "split_feature" in d_old:
+            assert d_old["split_feature"] == d_new["split_feature"]
+        assert not np.allclose(bst.predict(X[:50]), new.predict(X[:50]))
+
+
+def test_dataset_reference_chain():
+    """valid sets created from a valid set still bin against the root reference
+    (ref test_reference_chain)."""
+    rng = np.random.RandomState(33)
+    X = rng.randn(1200, 3)
+    y = X[:, 0] + 0.1 * rng.randn(1200)
+    tr = lgb.Dataset(X[:600], label=y[:600])
+    v1 = tr.create_valid(X[600:900], label=y[600:900])
+    v2 = v1.create_valid(X[900:], label=y[900:])
+    res = {}
+    lgb.train({"objective": "regression", "metric": "l2", "verbosity": -1}, tr, 10,
+              valid_sets=[v1, v2], valid_names=["v1", "v2"],
+              callbacks=[lgb.record_evaluation(res)])
+    assert "v1" in res and "v2" in res
+    assert res["v1"]["l2"][-1] < res["v1"]["l2"][0]
