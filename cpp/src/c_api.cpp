@@ -861,6 +861,16 @@ int LGBM_DatasetSetField(DatasetHandle handle, const char* field_name, const voi
   API_BEGIN();
   Dataset* d = static_cast<Dataset*>(handle);
   std::string name(field_name);
+  // a null/empty payload CLEARS an optional field (reference set_field(None))
+  const bool clearing = field_data == nullptr || num_element == 0;
+  if (clearing && name != "label") {
+    if (name == "weight") d->metadata().SetWeights(nullptr, 0);
+    else if (name == "group" || name == "query") d->metadata().SetQuery(nullptr, 0);
+    else if (name == "init_score") d->metadata().SetInitScore(nullptr, 0);
+    else if (name == "position") d->metadata().SetPosition(nullptr, 0);
+    else Log::Fatal("Unknown field %s", field_name);
+    return 0;
+  }
   if ((name == "label" || name == "weight" || name == "position") &&
       num_element != d->num_data()) {
     Log::Fatal("Length of %s (%d) differs from the number of rows (%d)", field_name,

@@ -1003,6 +1003,40 @@ std::unique_ptr<Dataset> Dataset::Subset(const data_size_t* indices, data_size_t
     for (data_size_t i = 0; i < n; ++i) w[i] = metadata_.weights()[indices[i]];
     d->metadata_.SetWeights(w.data(), n);
   }
+  // query boundaries: count subset rows per original query, keep non-empty
+  // queries (reference Metadata::CheckOrPartition subset semantics — indices
+  // are expected to respect query grouping)
+  if (metadata_.query_boundaries() != nullptr && metadata_.num_queries() > 0) {
+    const data_size_t* qb = metadata_.query_boundaries();
+    const data_size_t nq = metadata_.num_queries();
+    std::vector<data_size_t> boundaries{0};
+    data_size_t q = 0, cnt = 0;
+    for (data_size_t i = 0; i < n; ++i) {
+      while (q < nq && indices[i] >= qb[q + 1]) {
+        if (cnt > 0) { boundaries.push_back(boundaries.back() + cnt); cnt = 0; }
+        ++q;
+      }
+      ++cnt;
+    }
+    if (cnt > 0) boundaries.push_back(boundaries.back() + cnt);
+    d->metadata_.SetQueryBoundaries(std::move(boundaries));
+  }
+  if (metadata_.init_score() != nullptr) {
+    const int64_t total = metadata_.num_init_score();
+    const data_size_t N = num_data_;
+    const int k = N > 0 ? static_cast<int>(total / N) : 1;
+    std::vector<double> is(static_cast<size_t>(k) * n);
+    for (int c = 0; c < k; ++c)
+      for (data_size_t i = 0; i < n; ++i)
+        is[static_cast<size_t>(c) * n + i] =
+            metadata_.init_score()[static_cast<int64_t>(c) * N + indices[i]];
+    d->metadata_.SetInitScore(is.data(), static_cast<int64_t>(is.size()));
+  }
+  if (metadata_.positions() != nullptr) {
+    std::vector<int32_t> pos(n);
+    for (data_size_t i = 0; i < n; ++i) pos[i] = metadata_.positions()[indices[i]];
+    d->metadata_.SetPosition(pos.data(), n);
+  }
   return d;
 }
 

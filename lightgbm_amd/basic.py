@@ -410,7 +410,12 @@ class Dataset:
             ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_F32)))
 
     def set_field(self, field_name, data):
-        """Generic field setter (reference parity)."""
+        """Generic field setter (reference parity). data=None clears the field."""
+        if data is None and field_name != "label":
+            _safe_call(_LIB.LGBM_DatasetSetField(
+                self._handle, field_name.encode("utf-8"), None,
+                ctypes.c_int32(0), ctypes.c_int(0)))
+            return self
         if field_name in ("label", "weight"):
             self._set_float_field(field_name, data)
         elif field_name == "group":
@@ -494,7 +499,11 @@ class Dataset:
     def set_init_score(self, init_score):
         self.init_score = init_score
         if self._handle is not None and init_score is not None:
-            arr = np.ascontiguousarray(np.asarray(init_score).ravel(), dtype=np.float64)
+            a = np.asarray(init_score)
+            # 2D (row, class) flattens class-major: the engine stores scores as
+            # num_data*num_class with class as the outer index (reference order="F")
+            arr = np.ascontiguousarray(
+                a.ravel(order="F") if a.ndim == 2 else a.ravel(), dtype=np.float64)
             _safe_call(_LIB.LGBM_DatasetSetField(
                 self._handle, _c_str("init_score"), arr.ctypes.data_as(ctypes.c_void_p),
                 ctypes.c_int(len(arr)), ctypes.c_int(_DTYPE_F64)))
@@ -515,8 +524,14 @@ class Dataset:
             return np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_float)),
                                          shape=(n,)).copy()
         if out_type.value == _DTYPE_F64:
-            return np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_double)),
-                                         shape=(n,)).copy()
+            arr = np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_double)),
+                                        shape=(n,)).copy()
+            if name == "init_score":
+                nrow = self.num_data()
+                if nrow and n > nrow and n % nrow == 0:
+                    # stored class-major; surface as (row, class) like the reference
+                    arr = arr.reshape(n // nrow, nrow).T
+            return arr
         return np.ctypeslib.as_array(ctypes.cast(out_ptr, ctypes.POINTER(ctypes.c_int32)),
                                      shape=(n,)).copy()
 

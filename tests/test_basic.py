@@ -415,3 +415,60 @@ def test_kitchen_sink_mixed_features():
     contrib = bst.predict(X[:50], pred_contrib=True)
     np.testing.assert_allclose(contrib.sum(axis=1),
                                bst.predict(X[:50], raw_score=True), rtol=1e-6, atol=1e-6)
+
+
+def test_subset_group():
+    """Dataset.subset carries query boundaries, init_score and positions
+    (ref test_basic.py test_subset_group)."""
+    rng = np.random.RandomState(40)
+    X = rng.randn(200, 4)
+    y = rng.randint(0, 3, 200).astype(float)
+    ds = lgb.Dataset(X, label=y, group=[50, 50, 50, 50],
+                     init_score=np.arange(200, dtype=float))
+    ds.construct()
+    sub = ds.subset(list(range(100)))
+    sub.construct()
+    np.testing.assert_array_equal(sub.get_field("group"), [0, 50, 100])
+    np.testing.assert_allclose(sub.get_field("init_score"), np.arange(100.0))
+    # the subset trains as a 2-query ranking shard
+    bst = lgb.train({"objective": "lambdarank", "verbosity": -1}, sub, 3)
+    assert bst.num_trees() == 3
+
+
+def test_set_field_none_removes_field():
+    """set_field(name, None) clears weight/group/init_score/position
+    (ref test_set_field_none_removes_field)."""
+    rng = np.random.RandomState(41)
+    X = rng.randn(100, 3)
+    ds = lgb.Dataset(X, label=rng.rand(100))
+    ds.construct()
+    for field, val in (("weight", np.ones(100)),
+                       ("init_score", np.zeros(100)),
+                       ("group", np.array([50, 50], dtype=np.int32))):
+        ds.set_field(field, val)
+        assert ds.get_field(field) is not None, field
+        ds.set_field(field, None)
+        assert ds.get_field(field) is None, field
+
+
+def test_init_score_multiclass_2d():
+    """2D (row, class) init_score seeds multiclass training
+    (ref test_init_score_for_multiclass_classification)."""
+    rng = np.random.RandomState(42)
+    X = rng.randn(1500, 4)
+    y = rng.randint(0, 3, 1500).astype(float)
+    init = np.zeros((1500, 3))
+    init[:, 1] = 5.0  # huge prior for class 1
+    # 2D roundtrip: stored class-major, surfaced back as (row, class)
+    small = np.array([[i * 10 + j for j in range(3)] for i in range(10)], dtype=float)
+    ds_small = lgb.Dataset(rng.rand(10, 2), label=rng.randint(0, 3, 10).astype(float),
+                           init_score=small).construct()
+    np.testing.assert_array_equal(ds_small.get_field("init_score"), small)
+    # training effect: the softmax prior on class 1 makes its first-tree
+    # gradients positive, so its raw scores are pushed DOWN, the others UP
+    ds = lgb.Dataset(X, label=y, init_score=init)
+    bst = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1,
+                     "learning_rate": 0.1, "boost_from_average": False}, ds, 1)
+    raw = bst.predict(X[:500], raw_score=True).reshape(500, 3)
+    assert raw[:, 1].mean() < 0 < raw[:, 0].mean()
+    assert raw[:, 1].mean() < 0 < raw[:, 2].mean()
