@@ -612,7 +612,7 @@ __global__ void k_init_root(int* leaf_begin, int* leaf_cnt, int* leaf_slot, Leaf
                             unsigned long long* leaf_branch) {
   leaf_bounds[0] = -1e308;  // monotone output bounds of the root leaf
   leaf_bounds[1] = 1e308;
-  leaf_branch[0] = 0ull;
+  leaf_branch[0] = leaf_branch[1] = leaf_branch[2] = leaf_branch[3] = 0ull;
   leaf_begin[0] = 0;
   leaf_cnt[0] = used_cnt;
   leaf_slot[0] = 0;
@@ -695,12 +695,20 @@ __global__ void __launch_bounds__(64) k_best_feat(
   if (f < p.own_fb || f >= p.own_fe) return;  // another rank owns this feature's scan
   if (feat_mask != nullptr && !feat_mask[f]) return;
   if (p.n_interaction_groups > 0) {
-    // interaction constraints (inner-feature bitmask form, nf <= 64): feature f
-    // is allowed iff some group covers the leaf's branch features plus f
-    const unsigned long long need = leaf_branch[leaf] | (1ull << f);
+    // interaction constraints (4-word inner-feature bitmasks, nf <= 256):
+    // feature f is allowed iff some group covers the leaf's branch features + f
+    const int fw = f >> 6;
     bool ok = false;
-    for (int g = 0; g < p.n_interaction_groups; ++g)
-      ok = ok || (need & ~group_masks[g]) == 0ull;
+    for (int g = 0; g < p.n_interaction_groups; ++g) {
+      const unsigned long long* gm = group_masks + 4 * g;
+      bool cover = true;
+      for (int w = 0; w < 4; ++w) {
+        const unsigned long long need =
+            leaf_branch[4 * leaf + w] | (w == fw ? (1ull << (f & 63)) : 0ull);
+        cover = cover && (need & ~gm[w]) == 0ull;
+      }
+      ok = ok || cover;
+    }
     if (!ok) return;
   }
   const int split_idx = counters[1];
@@ -1697,9 +1705,12 @@ __device__ void FinalizeBookkeeping(int* leaf_begin, int* leaf_cnt, int* leaf_sl
     leaf_slot[R] = spare_slot;  // L keeps old slot
   }
   if (leaf_branch != nullptr) {
-    const unsigned long long b = leaf_branch[L] | (1ull << w.feature);
-    leaf_branch[L] = b;
-    leaf_branch[R] = b;
+    for (int wd = 0; wd < 4; ++wd) {
+      unsigned long long b = leaf_branch[4 * L + wd];
+      if (wd == (w.feature >> 6)) b |= 1ull << (w.feature & 63);
+      leaf_branch[4 * L + wd] = b;
+      leaf_branch[4 * R + wd] = b;
+    }
   }
   if (mono != nullptr) {
     // BasicLeafConstraints bound propagation (mirrors the host learner): children
@@ -2576,7 +2587,7 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<hipk::LeafStat> d_leaf_stats_;
   DevBuf<double> d_leaf_bounds_;   // [2*num_leaves] monotone output bounds
   DevBuf<int8_t> d_mono_;          // per inner feature, only when constraints set
-  DevBuf<unsigned long long> d_leaf_branch_;  // [num_leaves] branch-feature bitmasks
+  DevBuf<unsigned long long> d_leaf_branch_;  // [4*num_leaves] branch-feature bitmasks (256-bit)
   DevBuf<unsigned long long> d_group_masks_;  // interaction groups (inner-feature bits)
   int n_interaction_groups_ = 0;
   DevBuf<int8_t> d_feat_mask_;
@@ -2926,13 +2937,14 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
   d_split_log_.Alloc(nl);
   d_leaf_stats_.Alloc(nl);
   d_leaf_bounds_.Alloc(2 * static_cast<size_t>(nl));
-  d_leaf_branch_.Alloc(nl);
+  d_leaf_branch_.Alloc(4 * static_cast<size_t>(nl));
   n_interaction_groups_ = 0;
-  if (!config_->interaction_constraints.empty() && nf_ > 64)
-    Log::Fatal("device interaction constraints support up to 64 features (%d present); "
+  if (!config_->interaction_constraints.empty() && nf_ > 256)
+    Log::Fatal("device interaction constraints support up to 256 features (%d present); "
                "use device_type=cpu for this dataset", nf_);
   if (!config_->interaction_constraints.empty()) {
-    // parse "[0,1],[2,3]" original-feature groups into inner-feature bitmasks
+    // parse "[0,1],[2,3]" original-feature groups into 4-word (256-bit)
+    // inner-feature bitmasks (same layout as the categorical cat_mask words)
     std::vector<unsigned long long> masks;
     std::vector<int> orig_to_inner(train_data->num_total_features(), -1);
     for (int f = 0; f < nf_; ++f) orig_to_inner[train_data->RealFeatureIndex(f)] = f;
@@ -2941,20 +2953,23 @@ void HIPTreeLearner::Init(const Dataset* train_data, bool is_constant_hessian) {
     while ((pos = sgrp.find('[', pos)) != std::string::npos) {
       size_t end = sgrp.find(']', pos);
       if (end == std::string::npos) break;
-      unsigned long long msk = 0;
+      unsigned long long msk[4] = {0, 0, 0, 0};
       for (auto& tok : Common::Split(sgrp.substr(pos + 1, end - pos - 1).c_str(), ',')) {
         auto t = Common::Trim(tok);
         if (t.empty()) continue;
         const int orig = atoi(t.c_str());
         if (orig >= 0 && orig < static_cast<int>(orig_to_inner.size()) &&
-            orig_to_inner[orig] >= 0)
-          msk |= 1ull << orig_to_inner[orig];
+            orig_to_inner[orig] >= 0) {
+          const int f = orig_to_inner[orig];
+          msk[f >> 6] |= 1ull << (f & 63);
+        }
       }
-      if (msk) masks.push_back(msk);
+      if (msk[0] | msk[1] | msk[2] | msk[3])
+        for (int w = 0; w < 4; ++w) masks.push_back(msk[w]);
       pos = end + 1;
     }
     if (!masks.empty()) {
-      n_interaction_groups_ = static_cast<int>(masks.size());
+      n_interaction_groups_ = static_cast<int>(masks.size() / 4);
       d_group_masks_.Alloc(masks.size());
       HIP_OK(hipMemcpy(d_group_masks_.ptr, masks.data(),
                        sizeof(unsigned long long) * masks.size(), hipMemcpyHostToDevice));

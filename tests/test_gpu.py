@@ -849,3 +849,53 @@ def test_gpu_linear_tree():
     bst_c = lgb.train(params_c, lgb.Dataset(X, label=y), 30)
     mse_c = float(np.mean((bst_c.predict(X[:10000]) - y[:10000]) ** 2))
     assert mses["gpu"] < 0.5 * mse_c, (mses["gpu"], mse_c)
+
+
+def test_gpu_interaction_constraints_wide():
+    """>64-feature interaction constraints on device (4-word branch masks):
+    every root->leaf path must stay within one feature group."""
+    rng = np.random.RandomState(5)
+    nf = 100
+    X = rng.randn(30000, nf)
+    y = (X[:, 0] * X[:, 1] + X[:, 70] * X[:, 90] > 0).astype(np.float32)
+    g1 = ",".join(str(i) for i in range(50))
+    g2 = ",".join(str(i) for i in range(50, nf))
+    bst = lgb.train({"objective": "binary", "device_type": "cuda", "num_leaves": 31,
+                     "interaction_constraints": f"[{g1}],[{g2}]", "verbosity": -1},
+                    lgb.Dataset(X, label=y), 12)
+    d = bst.dump_model()
+
+    def paths(node, acc, out):
+        if "leaf_index" in node:
+            out.append(set(acc))
+            return
+        paths(node["left_child"], acc + [node["split_feature"]], out)
+        paths(node["right_child"], acc + [node["split_feature"]], out)
+    lo, hi = set(range(50)), set(range(50, nf))
+    n_split_trees = 0
+    for t in d["tree_info"]:
+        out = []
+        paths(t["tree_structure"], [], out)
+        for p in out:
+            if p:
+                n_split_trees += 1
+                assert p <= lo or p <= hi, p
+    assert n_split_trees > 0
+
+
+def test_gpu_missing_only_split():
+    """device scan allows the all-numeric-vs-NaN split: a constant column whose
+    only signal is missingness must still separate (mirrors the CPU fix)."""
+    rng = np.random.RandomState(6)
+    n = 20000
+    X = np.zeros((n, 2))
+    X[:, 1] = rng.randn(n) * 0.01  # noise column so the dataset isn't degenerate
+    y = np.zeros(n, dtype=np.float32)
+    nan_rows = rng.choice(n, n // 5, replace=False)
+    X[nan_rows, 0] = np.nan
+    y[nan_rows] = 1.0
+    bst = lgb.train({"objective": "regression", "device_type": "cuda",
+                     "verbosity": -1, "boost_from_average": False},
+                    lgb.Dataset(X, label=y), 20)
+    mse = float(np.mean((bst.predict(X) - y) ** 2))
+    assert mse < 0.005, mse
