@@ -2373,7 +2373,10 @@ __global__ void k_linear_gram(const uint32_t* __restrict__ idx, int begin, int c
     if (s_acc[dim * dim + e] != 0.0) atomicAdd(&b[e], s_acc[dim * dim + e]);
 }
 
-/*! device tree walk over column bins (out-of-bag score update under bagging). */
+/*! device tree walk over column bins (out-of-bag score update under bagging).
+ *  is_cat + cat_mask (4 words/node, bin-space bitset) route categorical splits;
+ *  the optional linear arrays (TREE-leaf order) evaluate linear leaves from raw
+ *  feature values, falling back to the constant output on NaN path features. */
 template <typename BIN_T = uint8_t>
 __global__ void k_tree_predict_add(const BIN_T* __restrict__ cols, int num_data,
                                    const int* __restrict__ split_feat,
@@ -2382,7 +2385,15 @@ __global__ void k_tree_predict_add(const BIN_T* __restrict__ cols, int num_data,
                                    const int* __restrict__ right_child,
                                    const int* __restrict__ nan_bin,
                                    const uint8_t* __restrict__ default_left,
+                                   const uint8_t* __restrict__ is_cat,
+                                   const unsigned long long* __restrict__ cat_mask,
                                    const double* __restrict__ leaf_out,
+                                   const int* __restrict__ lin_off,
+                                   const int* __restrict__ lin_cnt,
+                                   const int* __restrict__ lin_feat,
+                                   const double* __restrict__ lin_coeff,
+                                   const double* __restrict__ lin_const,
+                                   const float* __restrict__ raw,
                                    const uint32_t* __restrict__ rows, int n,
                                    double* __restrict__ score) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -2392,13 +2403,30 @@ __global__ void k_tree_predict_add(const BIN_T* __restrict__ cols, int num_data,
   while (node >= 0) {
     const int f = split_feat[node];
     const int b = cols[static_cast<size_t>(f) * num_data + r];
-    if (nan_bin[node] >= 0 && b == nan_bin[node]) {
+    if (is_cat != nullptr && is_cat[node]) {
+      const unsigned long long w = cat_mask[4 * node + (b >> 6)];
+      node = ((w >> (b & 63)) & 1ull) ? left_child[node] : right_child[node];
+    } else if (nan_bin[node] >= 0 && b == nan_bin[node]) {
       node = default_left[node] ? left_child[node] : right_child[node];
     } else {
       node = b <= thr_bin[node] ? left_child[node] : right_child[node];
     }
   }
-  score[r] += leaf_out[~node];
+  const int leaf = ~node;
+  if (lin_cnt != nullptr) {
+    const int k = lin_cnt[leaf];
+    double out = lin_const[leaf];
+    bool nan = false;
+    const int off = lin_off[leaf];
+    for (int j = 0; j < k; ++j) {
+      const float v = raw[static_cast<size_t>(lin_feat[off + j]) * num_data + r];
+      if (isnan(v)) { nan = true; break; }
+      out += lin_coeff[off + j] * v;
+    }
+    score[r] += (nan || k == 0) ? leaf_out[leaf] : out;
+  } else {
+    score[r] += leaf_out[leaf];
+  }
 }
 
 }  // namespace hipk
@@ -2616,6 +2644,10 @@ class HIPTreeLearner : public TreeLearner {
   DevBuf<uint8_t> d_tw_dl_;
   DevBuf<double> d_tw_out_;
   DevBuf<uint32_t> d_oob_;
+  DevBuf<uint8_t> d_tw_iscat_;             // per-node categorical flags (OOB walk)
+  DevBuf<unsigned long long> d_tw_cat_;    // [4*ni] bin-space cat masks (OOB walk)
+  DevBuf<int> d_twl_off_, d_twl_cnt_, d_twl_feat_;   // linear arrays, TREE-leaf order
+  DevBuf<double> d_twl_coeff_, d_twl_const_;
 
   hipk::SplitRec* h_winner_ = nullptr;  // pinned staging (legacy; log path reads below)
   int* h_winner_leaf_ = nullptr;
@@ -3713,7 +3745,9 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
     const int ni = nl - 1;
     if (ni > 0 && !oob.empty()) {
       std::vector<int> feat(ni), thr(ni), lc(ni), rc(ni), nb(ni);
-      std::vector<uint8_t> dl(ni);
+      std::vector<uint8_t> dl(ni), iscat(ni, 0);
+      std::vector<unsigned long long> catmask;
+      bool any_cat = false;
       for (int i2 = 0; i2 < ni; ++i2) {
         feat[i2] = tree->split_feature_inner(i2);
         thr[i2] = static_cast<int>(tree->threshold_in_bin(i2));
@@ -3721,6 +3755,47 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
         rc[i2] = tree->right_child(i2);
         nb[i2] = feat_meta_host_[feat[i2]].nan_bin;
         dl[i2] = (tree->decision_type(i2) & Tree::kDefaultLeftMask) ? 1 : 0;
+        iscat[i2] = tree->IsCategoricalSplit(i2) ? 1 : 0;
+        any_cat = any_cat || iscat[i2];
+      }
+      if (any_cat) {
+        // category-space node bitsets -> bin-space 4-word masks for the walk
+        catmask.assign(static_cast<size_t>(4) * ni, 0ull);
+        const auto& cb = tree->cat_boundaries();
+        const auto& ct = tree->cat_threshold();
+        for (int i2 = 0; i2 < ni; ++i2) {
+          if (!iscat[i2]) continue;
+          const int cat_idx = thr[i2];
+          const uint32_t* bits = ct.data() + cb[cat_idx];
+          const int n_words = cb[cat_idx + 1] - cb[cat_idx];
+          const BinMapper* m = train_data_->FeatureBinMapper(feat[i2]);
+          const int nbin = std::min(m->num_bin(), 256);
+          for (int b = 0; b < nbin; ++b) {
+            const int cat = static_cast<int>(m->BinToValue(b));
+            if (cat >= 0 && (cat >> 5) < n_words && ((bits[cat >> 5] >> (cat & 31)) & 1))
+              catmask[4 * i2 + (b >> 6)] |= 1ull << (b & 63);
+          }
+        }
+      }
+      // linear leaves: coefficient arrays in TREE-leaf order for the walk
+      std::vector<double> lc_const, lc_flat;
+      std::vector<int> lc_off, lc_cnt, lf_flat;
+      if (tree->is_linear()) {
+        lc_const.resize(nl);
+        lc_off.resize(nl);
+        lc_cnt.resize(nl);
+        for (int l = 0; l < nl; ++l) {
+          lc_off[l] = static_cast<int>(lf_flat.size());
+          const auto& fi = tree->leaf_features_inner(l);
+          const auto& co = tree->leaf_coeffs(l);
+          lc_cnt[l] = static_cast<int>(co.size());
+          lc_const[l] = co.empty() ? tree->LeafOutput(l) : tree->leaf_const(l);
+          for (size_t j = 0; j < co.size(); ++j) {
+            lf_flat.push_back(fi[j]);
+            lc_flat.push_back(co[j]);
+          }
+        }
+        if (lf_flat.empty()) { lf_flat.push_back(0); lc_flat.push_back(0.0); }
       }
       std::vector<double> out_by_leaf(nl);
       for (int l = 0; l < nl; ++l) out_by_leaf[l] = tree->LeafOutput(l);
@@ -3739,20 +3814,61 @@ void HIPTreeLearner::AddPredictionToScore(const Tree* tree, double* /*out_score*
                             hipMemcpyHostToDevice, stream_));
       HIP_OK(hipMemcpyAsync(d_oob_.ptr, oob.data(), sizeof(uint32_t) * oob.size(),
                             hipMemcpyHostToDevice, stream_));
+      const unsigned long long* catp = nullptr;
+      const uint8_t* iscatp = nullptr;
+      if (any_cat) {
+        d_tw_iscat_.Alloc(ni);
+        d_tw_cat_.Alloc(catmask.size());
+        HIP_OK(hipMemcpyAsync(d_tw_iscat_.ptr, iscat.data(), ni, hipMemcpyHostToDevice,
+                              stream_));
+        HIP_OK(hipMemcpyAsync(d_tw_cat_.ptr, catmask.data(),
+                              sizeof(unsigned long long) * catmask.size(),
+                              hipMemcpyHostToDevice, stream_));
+        catp = d_tw_cat_.ptr;
+        iscatp = d_tw_iscat_.ptr;
+      }
+      const int *loffp = nullptr, *lcntp = nullptr, *lfeatp = nullptr;
+      const double *lcoefp = nullptr, *lconstp = nullptr;
+      if (tree->is_linear()) {
+        d_twl_off_.Alloc(nl);
+        d_twl_cnt_.Alloc(nl);
+        d_twl_const_.Alloc(nl);
+        d_twl_feat_.Alloc(lf_flat.size());
+        d_twl_coeff_.Alloc(lc_flat.size());
+        HIP_OK(hipMemcpyAsync(d_twl_off_.ptr, lc_off.data(), sizeof(int) * nl,
+                              hipMemcpyHostToDevice, stream_));
+        HIP_OK(hipMemcpyAsync(d_twl_cnt_.ptr, lc_cnt.data(), sizeof(int) * nl,
+                              hipMemcpyHostToDevice, stream_));
+        HIP_OK(hipMemcpyAsync(d_twl_const_.ptr, lc_const.data(), sizeof(double) * nl,
+                              hipMemcpyHostToDevice, stream_));
+        HIP_OK(hipMemcpyAsync(d_twl_feat_.ptr, lf_flat.data(), sizeof(int) * lf_flat.size(),
+                              hipMemcpyHostToDevice, stream_));
+        HIP_OK(hipMemcpyAsync(d_twl_coeff_.ptr, lc_flat.data(),
+                              sizeof(double) * lc_flat.size(), hipMemcpyHostToDevice,
+                              stream_));
+        loffp = d_twl_off_.ptr;
+        lcntp = d_twl_cnt_.ptr;
+        lfeatp = d_twl_feat_.ptr;
+        lcoefp = d_twl_coeff_.ptr;
+        lconstp = d_twl_const_.ptr;
+      }
       if (rows16_) {
         hipLaunchKernelGGL((hipk::k_tree_predict_add<uint16_t>),
                            dim3((static_cast<int>(oob.size()) + 255) / 256), dim3(256), 0,
                            stream_, reinterpret_cast<const uint16_t*>(d_cols_.ptr),
                            num_data_, d_tw_feat_.ptr, d_tw_thr_.ptr, d_tw_left_.ptr,
-                           d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr, d_tw_out_.ptr,
-                           d_oob_.ptr, static_cast<int>(oob.size()), ScorePtr());
+                           d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr, iscatp, catp,
+                           d_tw_out_.ptr, loffp, lcntp, lfeatp, lcoefp, lconstp,
+                           linear_ ? d_raw_.ptr : nullptr, d_oob_.ptr,
+                           static_cast<int>(oob.size()), ScorePtr());
       } else {
         hipLaunchKernelGGL((hipk::k_tree_predict_add<uint8_t>),
                            dim3((static_cast<int>(oob.size()) + 255) / 256), dim3(256), 0,
                            stream_, d_cols_.ptr, num_data_, d_tw_feat_.ptr, d_tw_thr_.ptr,
                            d_tw_left_.ptr, d_tw_right_.ptr, d_tw_nan_.ptr, d_tw_dl_.ptr,
-                           d_tw_out_.ptr, d_oob_.ptr, static_cast<int>(oob.size()),
-                           ScorePtr());
+                           iscatp, catp, d_tw_out_.ptr, loffp, lcntp, lfeatp, lcoefp,
+                           lconstp, linear_ ? d_raw_.ptr : nullptr, d_oob_.ptr,
+                           static_cast<int>(oob.size()), ScorePtr());
       }
     }
   }
@@ -4301,11 +4417,7 @@ namespace {
 TreeLearner* CreateHIP(const Config* cfg) {
   // loud, not silent: features the device split loop does not implement yet fall
   // back to the host serial learner (reference CUDA learner errors similarly)
-  auto unsupported = [&]() -> const char* {
-    if (cfg->linear_tree && cfg->bagging_freq > 0 && cfg->bagging_fraction < 1.0)
-      return "linear_tree with bagging (out-of-bag linear score updates)";
-    return nullptr;
-  };
+  auto unsupported = [&]() -> const char* { return nullptr; };
   if (const char* what = unsupported()) {
     Log::Warning("device_type=%s: %s is not implemented in the HIP split loop; "
                  "training this model on the host (CPU) learner instead",

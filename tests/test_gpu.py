@@ -899,3 +899,41 @@ def test_gpu_missing_only_split():
                     lgb.Dataset(X, label=y), 20)
     mse = float(np.mean((bst.predict(X) - y) ** 2))
     assert mse < 0.005, mse
+
+
+def test_gpu_bagging_categorical_oob():
+    """OOB score replay must route categorical splits via bin-space masks:
+    GPU bagging+categorical quality must match the CPU learner's."""
+    rng = np.random.RandomState(7)
+    n = 40000
+    cat = rng.randint(0, 12, n).astype(np.float64)
+    lut = rng.randn(12) * 2
+    X = np.column_stack([cat, rng.randn(n)])
+    y = (lut[cat.astype(int)] + 0.3 * X[:, 1] + 0.3 * rng.randn(n) > 0).astype(np.float32)
+    aucs = {}
+    for dev in ("cpu", "cuda"):
+        bst = lgb.train({"objective": "binary", "device_type": dev, "verbosity": -1,
+                         "bagging_fraction": 0.6, "bagging_freq": 1, "seed": 7,
+                         "categorical_feature": "0"},
+                        lgb.Dataset(X, label=y, categorical_feature=[0]), 30)
+        from sklearn.metrics import roc_auc_score
+        aucs[dev] = roc_auc_score(y, bst.predict(X))
+    assert aucs["cuda"] > aucs["cpu"] - 0.01, aucs
+
+
+def test_gpu_linear_tree_bagging():
+    """linear_tree + bagging on device: OOB rows get linear-leaf score updates
+    (previously a host fallback). Linear fit must beat constant leaves."""
+    rng = np.random.RandomState(8)
+    n = 30000
+    X = rng.rand(n, 3)
+    y = (3.0 * X[:, 0] + np.where(X[:, 1] > 0.5, 2.0 * X[:, 2], -2.0 * X[:, 2]) +
+         0.05 * rng.randn(n)).astype(np.float32)
+    common = {"objective": "regression", "device_type": "cuda", "verbosity": -1,
+              "num_leaves": 8, "learning_rate": 0.5, "bagging_fraction": 0.6,
+              "bagging_freq": 1, "seed": 3}
+    lin = lgb.train({**common, "linear_tree": True}, lgb.Dataset(X, label=y), 30)
+    const = lgb.train(common, lgb.Dataset(X, label=y), 30)
+    mse_lin = float(np.mean((lin.predict(X) - y) ** 2))
+    mse_const = float(np.mean((const.predict(X) - y) ** 2))
+    assert mse_lin < mse_const * 0.8, (mse_lin, mse_const)
