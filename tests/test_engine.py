@@ -1470,3 +1470,23 @@ def test_dataset_reference_chain():
               callbacks=[lgb.record_evaluation(res)])
     assert "v1" in res and "v2" in res
     assert res["v1"]["l2"][-1] < res["v1"]["l2"][0]
+
+
+def test_all_expected_params_written_to_model_text():
+    """the saved model's parameters block reflects the training config and
+    survives reload (ref test_all_expected_params_are_written_out_to_model_text)."""
+    rng = np.random.RandomState(34)
+    X = rng.randn(500, 3)
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "num_leaves": 7,
+                     "learning_rate": 0.2, "lambda_l2": 0.5,
+                     "bagging_fraction": 0.8, "bagging_freq": 2},
+                    lgb.Dataset(X, label=X[:, 0]), 3)
+    s = bst.model_to_string()
+    assert "parameters:" in s
+    for frag in ("[objective: regression]", "[num_leaves: 7]", "[learning_rate: 0.2]",
+                 "[lambda_l2: 0.5]", "[bagging_fraction: 0.8]", "[bagging_freq: 2]",
+                 "[boosting: gbdt]", "[tree_learner: serial]"):
+        assert frag in s, frag
+    # reloaded booster exposes the stored parameter string
+    bst2 = lgb.Booster(model_str=s)
+    np.testing.assert_allclose(bst2.predict(X[:20]), bst.predict(X[:20]), rtol=1e-12)
