@@ -36,15 +36,19 @@ class BinMapper {
 
   /*! Map a raw value to its bin. */
   inline uint32_t ValueToBin(double value) const {
-    if (std::isnan(value)) {
-      if (missing_type_ == MissingType::kNaN) return static_cast<uint32_t>(num_bin_ - 1);
-      value = 0.0;
-    }
     if (bin_type_ == BinType::kCategorical) {
+      // NaN / negative / unseen categories all land in bin 0 (the reference's
+      // dummy NaN bin, bin_2_categorical_[0] == -1) — DISTINCT from category 0
+      if (std::isnan(value)) return 0;
       int cat = static_cast<int>(value);
+      if (cat < 0) return 0;
       auto it = categorical_2_bin_.find(cat);
       if (it == categorical_2_bin_.end()) return 0;
       return it->second;
+    }
+    if (std::isnan(value)) {
+      if (missing_type_ == MissingType::kNaN) return static_cast<uint32_t>(num_bin_ - 1);
+      value = 0.0;
     }
     if (missing_type_ == MissingType::kZero && value == 0.0)
       return static_cast<uint32_t>(num_bin_ - 1);

@@ -115,7 +115,10 @@ void BinMapper::FindBin(double* values, int num_sample_values, size_t total_samp
     num_bin_ = static_cast<int>(bin_2_categorical_.size());
     num_numeric_bin_ = num_bin_;
     missing_type_ = MissingType::kNone;  // unseen categories -> bin 0
-    is_trivial_ = num_bin_ <= 2 && pre_filter && sorted.size() <= 1;
+    // a populated NaN/unseen bin 0 counts as a distinct value: a feature that is
+    // one category plus NaNs is informative (reference test_categorical_handle_na)
+    is_trivial_ = pre_filter &&
+                  static_cast<int>(sorted.size()) + (na_cnt > 0 ? 1 : 0) <= 1;
     most_freq_bin_ = num_bin_ > 1 ? 1 : 0;
     default_bin_ = 0;
     sparse_rate_ = 0.0;

@@ -99,6 +99,13 @@ def _pandas_to_float64(df, pandas_categorical=None):
     recorded = pandas_categorical is not None
     if pandas_categorical is None:
         pandas_categorical = [list(df[c].cat.categories) for c in cat_cols]
+    # nullable extension dtypes (Int64/Float64/boolean): pd.NA -> NaN column-wise
+    ext_cols = [c for c in df.columns
+                if c not in cat_cols and hasattr(df[c].dtype, "na_value")]
+    if ext_cols:
+        df = df.copy()
+        for c in ext_cols:
+            df[c] = df[c].to_numpy(dtype=np.float64, na_value=np.nan)
     if not cat_cols:
         return np.ascontiguousarray(df.to_numpy(), dtype=np.float64), pandas_categorical
     df = df.copy()

@@ -1490,3 +1490,126 @@ def test_all_expected_params_written_to_model_text():
     # reloaded booster exposes the stored parameter string
     bst2 = lgb.Booster(model_str=s)
     np.testing.assert_allclose(bst2.predict(X[:20]), bst.predict(X[:20]), rtol=1e-12)
+
+
+def test_multi_error_top_k():
+    """multi_error@k counts a row correct if the true class is in the top k
+    (ref test_multi_class_error)."""
+    rng = np.random.RandomState(35)
+    X = rng.randn(2000, 4)
+    y = rng.randint(0, 4, 2000).astype(float)
+    res1, res2 = {}, {}
+    ds = lgb.Dataset(X, label=y)
+    lgb.train({"objective": "multiclass", "num_class": 4, "metric": "multi_error",
+               "verbosity": -1}, ds, 5, valid_sets=[lgb.Dataset(X, label=y)],
+              callbacks=[lgb.record_evaluation(res1)])
+    lgb.train({"objective": "multiclass", "num_class": 4, "metric": "multi_error",
+               "multi_error_top_k": 2, "verbosity": -1},
+              lgb.Dataset(X, label=y), 5, valid_sets=[lgb.Dataset(X, label=y)],
+              callbacks=[lgb.record_evaluation(res2)])
+    e1 = res1["valid_0"]["multi_error"][-1]
+    e2 = res2["valid_0"]["multi_error@2"][-1]
+    assert 0.0 <= e2 <= e1 <= 1.0      # top-2 error can only be lower
+
+
+def test_prediction_early_stopping_multiclass():
+    """pred_early_stop on multiclass prediction stays close to the exact result
+    (ref test_multiclass_prediction_early_stopping)."""
+    rng = np.random.RandomState(36)
+    X = rng.randn(3000, 6)
+    y = (X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)
+    bst = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+                    lgb.Dataset(X, label=y.astype(float)), 50)
+    exact = np.argmax(bst.predict(X[:500]), axis=1)
+    fast = np.argmax(bst.predict(X[:500], pred_early_stop=True,
+                                 pred_early_stop_freq=5,
+                                 pred_early_stop_margin=1.5), axis=1)
+    assert (exact == fast).mean() > 0.95
+
+
+def test_categorical_handle_na():
+    """categorical column with NaN rows routes NaN to the non-bitset side
+    (ref test_categorical_handle_na)."""
+    x = np.array([0, 1, 2, 2, 1, 0, np.nan, 1, 2, np.nan]).reshape(-1, 1)
+    lut = {0: 1.0, 1: 2.0, 2: 3.0}
+    y = np.array([lut.get(v, 0.0) if not np.isnan(v) else 0.0 for v in x[:, 0]])
+    ds = lgb.Dataset(x, label=y, categorical_feature=[0])
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "min_data_in_leaf": 1,
+                     "min_data_per_group": 1, "cat_smooth": 1e-3, "cat_l2": 0.0,
+                     "learning_rate": 1.0, "num_leaves": 8, "min_data_in_bin": 1,
+                     "max_cat_to_onehot": 1}, ds, 10)
+    pred = bst.predict(x)
+    np.testing.assert_allclose(pred, y, atol=1e-2)
+
+
+def test_pandas_nullable_dtypes():
+    """pandas nullable Int64/Float64/boolean columns train and predict
+    (ref test_pandas_nullable_dtypes)."""
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(37)
+    n = 1000
+    df = pd.DataFrame({
+        "a": pd.array(rng.randint(0, 10, n), dtype="Int64"),
+        "b": pd.array(rng.randn(n), dtype="Float64"),
+        "c": pd.array(rng.rand(n) > 0.5, dtype="boolean"),
+    })
+    df.loc[:20, "a"] = pd.NA
+    y = (df["b"].astype(float).to_numpy() > 0).astype(float)
+    bst = lgb.train({"objective": "binary", "verbosity": -1},
+                    lgb.Dataset(df, label=y), 10)
+    pred = bst.predict(df)
+    assert ((pred > 0.5) == y).mean() > 0.9
+
+
+def test_cegb_scaling_equalities():
+    """cegb_penalty_split scaled with tradeoff leaves trees unchanged when both
+    scale together (ref test_cegb_scaling_equalities semantics)."""
+    rng = np.random.RandomState(38)
+    X = rng.randn(2000, 4)
+    y = X[:, 0] + 0.2 * rng.randn(2000)
+    def model(tradeoff, split_pen):
+        return lgb.train({"objective": "regression", "verbosity": -1,
+                          "cegb_tradeoff": tradeoff, "cegb_penalty_split": split_pen},
+                         lgb.Dataset(X, label=y), 10).model_to_string()
+    # tradeoff*penalty identical => identical trees
+    assert model(1.0, 0.5) == model(2.0, 0.25)
+    # different effective penalty => different trees
+    assert model(1.0, 0.5) != model(1.0, 2.0)
+
+
+def test_reset_params_metric_boosting():
+    """reset_parameter callback + Booster.reset_parameter change live params
+    (ref test_reset_params_works_with_metric_num_class_and_boosting)."""
+    rng = np.random.RandomState(39)
+    X = rng.randn(1000, 4)
+    y = X[:, 0] + 0.1 * rng.randn(1000)
+    lrs = []
+    def track(env):
+        lrs.append(float(env.params.get("learning_rate", -1)))
+    track.order = 20
+    lgb.train({"objective": "regression", "verbosity": -1, "learning_rate": 0.1},
+              lgb.Dataset(X, label=y), 5,
+              callbacks=[lgb.reset_parameter(learning_rate=lambda i: 0.1 * (0.5 ** i)),
+                         track])
+    assert len(lrs) == 5
+    assert lrs[0] > lrs[-1]
+
+
+def test_predict_output_shapes():
+    """prediction output shapes across tasks (ref test_predict_*_output_shape)."""
+    rng = np.random.RandomState(40)
+    X = rng.randn(600, 5)
+    yr = X[:, 0]
+    br = lgb.train({"objective": "regression", "verbosity": -1}, lgb.Dataset(X, label=yr), 4)
+    assert br.predict(X).shape == (600,)
+    assert br.predict(X, pred_leaf=True).shape == (600, 4)
+    assert br.predict(X, pred_contrib=True).shape == (600, 6)
+    yb = (X[:, 0] > 0).astype(float)
+    bb = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(X, label=yb), 4)
+    assert bb.predict(X).shape == (600,)
+    ym = rng.randint(0, 3, 600).astype(float)
+    bm = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+                   lgb.Dataset(X, label=ym), 4)
+    assert bm.predict(X).shape == (600, 3)
+    assert bm.predict(X, pred_leaf=True).shape == (600, 12)
+    assert bm.predict(X, pred_contrib=True).shape == (600, 3 * 6)
