@@ -222,6 +222,20 @@ class LGBMModel(BaseEstimator):
         return self._best_iteration
 
     @property
+    def feature_names_in_(self):
+        return np.array(self.booster_.feature_name())
+
+    @property
+    def n_estimators_(self):
+        """actual number of fitted iterations (early stopping aware)."""
+        return self._best_iteration if self._best_iteration > 0 \
+            else self.booster_.current_iteration()
+
+    @property
+    def n_iter_(self):
+        return self.n_estimators_
+
+    @property
     def best_score_(self):
         return self.booster_.best_score
 

@@ -371,3 +371,18 @@ def test_stacking_meta_estimators():
                            final_estimator=Ridge(), cv=2)
     sr.fit(X, X[:, 0])
     assert sr.score(X, X[:, 0]) > 0.5
+
+
+def test_post_fit_attributes():
+    """feature_names_in_ / n_estimators_ / n_iter_ (ref sklearn post-fit attrs)."""
+    rng = np.random.RandomState(50)
+    X = rng.randn(1500, 4)
+    y = X[:, 0] + 0.2 * rng.randn(1500)
+    m = lgb.LGBMRegressor(n_estimators=30, verbosity=-1).fit(X, y)
+    assert list(m.feature_names_in_) == [f"Column_{i}" for i in range(4)]
+    assert m.n_estimators_ == 30
+    assert m.n_iter_ == 30
+    es = lgb.LGBMRegressor(n_estimators=500, verbosity=-1)
+    es.fit(X[:1000], y[:1000], eval_set=[(X[1000:], y[1000:])],
+           eval_metric="l2", early_stopping_rounds=3)
+    assert es.n_estimators_ == es.best_iteration_ < 500
