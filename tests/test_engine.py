@@ -1613,3 +1613,30 @@ def test_predict_output_shapes():
     assert bm.predict(X).shape == (600, 3)
     assert bm.predict(X, pred_leaf=True).shape == (600, 12)
     assert bm.predict(X, pred_contrib=True).shape == (600, 3 * 6)
+
+
+def test_default_metric_per_objective():
+    """each objective evaluates with its matching default metric; metric='None'
+    disables evaluation (ref test_metrics / test_default_objective_and_metric)."""
+    rng = np.random.RandomState(41)
+    X = rng.randn(500, 3)
+    cases = [("regression", X[:, 0], "l2"), ("regression_l1", X[:, 0], "l1"),
+             ("binary", (X[:, 0] > 0).astype(float), "binary_logloss"),
+             ("huber", X[:, 0], "huber"), ("quantile", X[:, 0], "quantile"),
+             ("mape", np.abs(X[:, 0]) + 1, "mape"),
+             ("poisson", np.abs(X[:, 0]), "poisson"),
+             ("gamma", np.abs(X[:, 0]) + 0.1, "gamma"),
+             ("tweedie", np.abs(X[:, 0]), "tweedie"),
+             ("cross_entropy", (X[:, 0] > 0).astype(float), "cross_entropy")]
+    for obj, y, want in cases:
+        res = {}
+        lgb.train({"objective": obj, "verbosity": -1}, lgb.Dataset(X, label=y), 2,
+                  valid_sets=[lgb.Dataset(X, label=y)],
+                  callbacks=[lgb.record_evaluation(res)])
+        assert want in res["valid_0"], (obj, list(res["valid_0"]))
+    res = {}
+    yb = (X[:, 0] > 0).astype(float)
+    lgb.train({"objective": "binary", "metric": "None", "verbosity": -1},
+              lgb.Dataset(X, label=yb), 2, valid_sets=[lgb.Dataset(X, label=yb)],
+              callbacks=[lgb.record_evaluation(res)])
+    assert res == {}
