@@ -326,3 +326,34 @@ def test_cli_distributed_socket_mesh(tmp_path):
     bst = lgb.Booster(model_file=str(results[0][2]))
     acc = ((bst.predict(X) > 0.5) == y).mean()
     assert acc > 0.93, acc
+
+
+def test_cli_python_consistency(tmp_path):
+    """CLI config training and Python API training on the same example data
+    produce the same model predictions (ref tests/python_package_test/
+    test_consistency.py semantics)."""
+    import shutil
+    ex = Path(__file__).resolve().parent.parent / "examples" / "binary_classification"
+    work = tmp_path / "w"
+    shutil.copytree(ex, work)
+    subprocess.run([str(CLI), "config=train.conf", "num_trees=20",
+                    "metric_freq=100"], cwd=work, check=True, timeout=300,
+                   capture_output=True)
+    model_file = work / "LightGBM_model.txt"
+    assert model_file.exists()
+    cli_bst = lgb.Booster(model_file=str(model_file))
+
+    # same data through the Python API with the same params
+    data = np.loadtxt(work / "binary.train")
+    y, X = data[:, 0], data[:, 1:]
+    params = {"objective": "binary", "metric": ["binary_logloss", "auc"],
+              "max_bin": 255, "num_leaves": 63, "learning_rate": 0.1,
+              "min_data_in_leaf": 50, "min_sum_hessian_in_leaf": 5.0,
+              "verbosity": -1}
+    py_bst = lgb.train(params, lgb.Dataset(X, label=y), 20)
+
+    test = np.loadtxt(work / "binary.test")
+    Xt, yt = test[:, 1:], test[:, 0]
+    p_cli = cli_bst.predict(Xt)
+    p_py = py_bst.predict(Xt)
+    np.testing.assert_allclose(p_cli, p_py, rtol=1e-9)

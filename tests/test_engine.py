@@ -1640,3 +1640,79 @@ def test_default_metric_per_objective():
               lgb.Dataset(X, label=yb), 2, valid_sets=[lgb.Dataset(X, label=yb)],
               callbacks=[lgb.record_evaluation(res)])
     assert res == {}
+
+
+def test_objective_aliases_equivalent():
+    """objective aliases train identically to the canonical name
+    (ref test_objective_aliases)."""
+    rng = np.random.RandomState(42)
+    X = rng.randn(800, 3)
+    y = X[:, 0] + 0.1 * rng.randn(800)
+    groups = {"regression": ["mean_squared_error", "mse", "l2", "rmse"],
+              "regression_l1": ["mae", "mean_absolute_error", "l1"],
+              "multiclass": ["softmax"]}
+    for canon, aliases in groups.items():
+        yy = rng.randint(0, 3, 800).astype(float) if canon == "multiclass" else y
+        extra = {"num_class": 3} if canon == "multiclass" else {}
+        base = lgb.train({"objective": canon, "verbosity": -1, **extra},
+                         lgb.Dataset(X, label=yy), 5).model_to_string()
+        for a in aliases:
+            m = lgb.train({"objective": a, "verbosity": -1, **extra},
+                          lgb.Dataset(X, label=yy), 5).model_to_string()
+            assert m == base, (canon, a)
+
+
+def test_goss_boosting_and_strategy_equivalent():
+    """boosting=goss (back-compat) == data_sample_strategy=goss
+    (ref test_goss_boosting_and_strategy_equivalent)."""
+    rng = np.random.RandomState(43)
+    X = rng.randn(3000, 4)
+    y = X[:, 0] + 0.2 * rng.randn(3000)
+    a = lgb.train({"objective": "regression", "boosting": "goss", "verbosity": -1,
+                   "seed": 5}, lgb.Dataset(X, label=y), 10).model_to_string()
+    b = lgb.train({"objective": "regression", "boosting": "gbdt",
+                   "data_sample_strategy": "goss", "verbosity": -1, "seed": 5},
+                  lgb.Dataset(X, label=y), 10).model_to_string()
+    assert a == b
+
+
+def test_boost_from_average_with_single_leaf_trees():
+    """stump-only models still boost from the label average
+    (ref test_boost_from_average_with_single_leaf_trees)."""
+    rng = np.random.RandomState(44)
+    X = rng.randn(500, 2)
+    y = np.full(500, 7.5) + 0.001 * rng.randn(500)
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "min_data_in_leaf": 400},
+                    lgb.Dataset(X, label=y), 5)
+    np.testing.assert_allclose(bst.predict(X[:10]), 7.5, rtol=1e-3)
+
+
+def test_mape_with_bagging():
+    """mape objective under rf/bagging keeps positive predictions
+    (ref test_mape_for_specific_boosting_types)."""
+    rng = np.random.RandomState(45)
+    X = rng.randn(3000, 4)
+    y = np.abs(X[:, 0]) + 1.0 + 0.05 * rng.randn(3000)
+    bst = lgb.train({"objective": "mape", "verbosity": -1, "bagging_fraction": 0.8,
+                     "bagging_freq": 1, "seed": 3}, lgb.Dataset(X, label=y), 30)
+    pred = bst.predict(X)
+    assert (pred > 0).all()
+    assert float(np.mean(np.abs(pred - y) / y)) < 0.3
+
+
+def test_forced_split_feature_indices(tmp_path):
+    """forcedsplits_filename: the forced feature is the root split of every tree
+    (ref test_forced_split_feature_indices semantics)."""
+    import json as _json
+    rng = np.random.RandomState(46)
+    X = rng.randn(2000, 4)
+    y = X[:, 0] + X[:, 3] + 0.1 * rng.randn(2000)
+    fs = tmp_path / "forced.json"
+    fs.write_text(_json.dumps({"feature": 3, "threshold": 0.0}))
+    bst = lgb.train({"objective": "regression", "verbosity": -1,
+                     "forcedsplits_filename": str(fs)}, lgb.Dataset(X, label=y), 5)
+    d = bst.dump_model()
+    for t in d["tree_info"]:
+        root = t["tree_structure"]
+        assert root["split_feature"] == 3
+        assert root["threshold"] == pytest.approx(0.0, abs=0.2)
