@@ -1,5 +1,11 @@
-"""Training callbacks (parity target: reference python-package/lightgbm/callback.py)."""
+"""Training callbacks (parity target: reference python-package/lightgbm/callback.py).
+
+Callbacks are CLASSES (factory functions return instances) so they pickle —
+sklearn estimators carrying callbacks must survive joblib/pickle round-trips
+(reference _EarlyStoppingCallback et al.).
+"""
 import collections
+import warnings
 
 __all__ = ["early_stopping", "log_evaluation", "record_evaluation", "reset_parameter",
            "EarlyStopException", "CallbackEnv"]
@@ -18,45 +24,68 @@ class EarlyStopException(Exception):
         self.best_score = best_score
 
 
-def log_evaluation(period=1, show_stdv=True):
-    """Log evaluation results every `period` iterations."""
-    def _callback(env):
-        if period > 0 and env.evaluation_result_list and (env.iteration + 1) % period == 0:
+class _LogEvaluationCallback:
+    order = 10
+
+    def __init__(self, period=1, show_stdv=True):
+        self.period = period
+        self.show_stdv = show_stdv
+
+    def __call__(self, env):
+        if self.period > 0 and env.evaluation_result_list and \
+                (env.iteration + 1) % self.period == 0:
             result = "\t".join(
                 [f"{name}'s {metric}: {value:g}" if len(r) == 4 else str(r)
                  for r in env.evaluation_result_list
                  for (name, metric, value, _) in [r[:4]]])
             print(f"[{env.iteration + 1}]\t{result}")
-    _callback.order = 10
-    return _callback
+
+
+def log_evaluation(period=1, show_stdv=True):
+    """Log evaluation results every `period` iterations."""
+    return _LogEvaluationCallback(period, show_stdv)
+
+
+class _RecordEvaluationCallback:
+    order = 20
+
+    def __init__(self, eval_result):
+        if not isinstance(eval_result, dict):
+            raise TypeError("eval_result must be a dict")
+        self.eval_result = eval_result
+        self._started = False
+
+    def _init(self, env):
+        self.eval_result.clear()
+        for r in env.evaluation_result_list or []:
+            name, metric = r[0], r[1]
+            self.eval_result.setdefault(name, collections.OrderedDict()).setdefault(metric, [])
+
+    def __call__(self, env):
+        if not self._started:
+            self._init(env)
+            self._started = True
+        for r in env.evaluation_result_list or []:
+            name, metric, value = r[0], r[1], r[2]
+            self.eval_result.setdefault(name, collections.OrderedDict()) \
+                .setdefault(metric, []).append(value)
 
 
 def record_evaluation(eval_result):
     """Record evaluation results into the supplied dict."""
-    if not isinstance(eval_result, dict):
-        raise TypeError("eval_result must be a dict")
-
-    def _init(env):
-        eval_result.clear()
-        for r in env.evaluation_result_list or []:
-            name, metric = r[0], r[1]
-            eval_result.setdefault(name, collections.OrderedDict()).setdefault(metric, [])
-
-    def _callback(env):
-        if not eval_result:
-            _init(env)
-        for r in env.evaluation_result_list or []:
-            name, metric, value = r[0], r[1], r[2]
-            eval_result.setdefault(name, collections.OrderedDict()).setdefault(metric, []).append(value)
-    _callback.order = 20
-    return _callback
+    return _RecordEvaluationCallback(eval_result)
 
 
-def reset_parameter(**kwargs):
-    """Reset parameters on a schedule: value is a list (per iteration) or a callable."""
-    def _callback(env):
+class _ResetParameterCallback:
+    order = 10
+    before_iteration = True
+
+    def __init__(self, kwargs):
+        self.kwargs = kwargs
+
+    def __call__(self, env):
         new_params = {}
-        for key, value in kwargs.items():
+        for key, value in self.kwargs.items():
             if isinstance(value, list):
                 if len(value) != env.end_iteration - env.begin_iteration:
                     raise ValueError(f"Length of list {key!r} must match num_boost_round")
@@ -68,62 +97,79 @@ def reset_parameter(**kwargs):
         if new_params:
             env.model.reset_parameter(new_params)
             env.params.update(new_params)
-    _callback.before_iteration = True
-    _callback.order = 10
-    return _callback
 
 
-def early_stopping(stopping_rounds, first_metric_only=False, verbose=True, min_delta=0.0):
-    """Stop training when a validation metric stops improving."""
-    best_score = []
-    best_iter = []
-    best_score_list = []
-    cmp_op = []
-    enabled = [True]
-    first_metric = [""]
+def reset_parameter(**kwargs):
+    """Reset parameters on a schedule: value is a list (per iteration) or a callable."""
+    return _ResetParameterCallback(kwargs)
 
-    def _init(env):
-        enabled[0] = bool(env.evaluation_result_list)
-        if not enabled[0]:
-            import warnings
+
+class _EarlyStoppingCallback:
+    order = 30
+
+    def __init__(self, stopping_rounds, first_metric_only=False, verbose=True,
+                 min_delta=0.0):
+        self.stopping_rounds = stopping_rounds
+        self.first_metric_only = first_metric_only
+        self.verbose = verbose
+        self.min_delta = min_delta
+        self._reset()
+
+    def _reset(self):
+        self.best_score = []
+        self.best_iter = []
+        self.best_score_list = []
+        self.higher_better = []
+        self.deltas = []
+        self.enabled = True
+        self.first_metric = ""
+
+    def _init(self, env):
+        self._reset()
+        self.enabled = bool(env.evaluation_result_list)
+        if not self.enabled:
             warnings.warn("Early stopping requires at least one validation set")
             return
-        best_score.clear(); best_iter.clear(); best_score_list.clear(); cmp_op.clear()
-        first_metric[0] = env.evaluation_result_list[0][1]
+        self.first_metric = env.evaluation_result_list[0][1]
         n_metric = len(env.evaluation_result_list)
-        deltas = [min_delta] * n_metric if not isinstance(min_delta, list) else min_delta
-        for i, r in enumerate(env.evaluation_result_list):
-            best_iter.append(0)
-            best_score_list.append(None)
-            if r[3]:  # higher better
-                best_score.append(float("-inf"))
-                cmp_op.append(lambda cur, best, d=deltas[i]: cur > best + d)
-            else:
-                best_score.append(float("inf"))
-                cmp_op.append(lambda cur, best, d=deltas[i]: cur < best - d)
+        self.deltas = list(self.min_delta) if isinstance(self.min_delta, list) \
+            else [self.min_delta] * n_metric
+        for r in env.evaluation_result_list:
+            self.best_iter.append(0)
+            self.best_score_list.append(None)
+            self.higher_better.append(bool(r[3]))
+            self.best_score.append(float("-inf") if r[3] else float("inf"))
 
-    def _callback(env):
-        if not best_score:
-            _init(env)
-        if not enabled[0]:
+    def _improved(self, i, cur):
+        if self.higher_better[i]:
+            return cur > self.best_score[i] + self.deltas[i]
+        return cur < self.best_score[i] - self.deltas[i]
+
+    def __call__(self, env):
+        if env.iteration == env.begin_iteration or not self.best_score:
+            self._init(env)  # fresh state per training run (instances are reusable)
+        if not self.enabled:
             return
         for i, r in enumerate(env.evaluation_result_list):
             name, metric, score = r[0], r[1], r[2]
             if name == "training":
                 continue
-            if first_metric_only and metric != first_metric[0]:
+            if self.first_metric_only and metric != self.first_metric:
                 continue
-            if best_score_list[i] is None or cmp_op[i](score, best_score[i]):
-                best_score[i] = score
-                best_iter[i] = env.iteration
-                best_score_list[i] = env.evaluation_result_list
-            elif env.iteration - best_iter[i] >= stopping_rounds:
-                if verbose:
-                    print(f"Early stopping, best iteration is: [{best_iter[i] + 1}]")
-                raise EarlyStopException(best_iter[i], best_score_list[i])
+            if self.best_score_list[i] is None or self._improved(i, score):
+                self.best_score[i] = score
+                self.best_iter[i] = env.iteration
+                self.best_score_list[i] = env.evaluation_result_list
+            elif env.iteration - self.best_iter[i] >= self.stopping_rounds:
+                if self.verbose:
+                    print(f"Early stopping, best iteration is: [{self.best_iter[i] + 1}]")
+                raise EarlyStopException(self.best_iter[i], self.best_score_list[i])
         if env.iteration == env.end_iteration - 1:
-            for i in range(len(best_iter)):
-                if best_score_list[i] is not None:
-                    raise EarlyStopException(best_iter[i], best_score_list[i])
-    _callback.order = 30
-    return _callback
+            for i in range(len(self.best_iter)):
+                if self.best_score_list[i] is not None:
+                    raise EarlyStopException(self.best_iter[i], self.best_score_list[i])
+
+
+def early_stopping(stopping_rounds, first_metric_only=False, verbose=True, min_delta=0.0):
+    """Stop training when a validation metric stops improving."""
+    return _EarlyStoppingCallback(stopping_rounds, first_metric_only, verbose, min_delta)

@@ -386,3 +386,26 @@ def test_post_fit_attributes():
     es.fit(X[:1000], y[:1000], eval_set=[(X[1000:], y[1000:])],
            eval_metric="l2", early_stopping_rounds=3)
     assert es.n_estimators_ == es.best_iteration_ < 500
+
+
+def test_callbacks_and_estimators_picklable_with_joblib():
+    """callbacks are class instances and pickle; estimators carrying callbacks
+    survive joblib dump/load (ref test_joblib / test_non_serializable_objects)."""
+    import pickle
+    cbs = [lgb.early_stopping(5), lgb.log_evaluation(1),
+           lgb.record_evaluation({}), lgb.reset_parameter(learning_rate=[0.1] * 5)]
+    for c in cbs:
+        assert pickle.loads(pickle.dumps(c)) is not None
+    import joblib
+    rng = np.random.RandomState(51)
+    X = rng.randn(800, 3)
+    y = X[:, 0] + 0.1 * rng.randn(800)
+    m = lgb.LGBMRegressor(n_estimators=20, verbosity=-1)
+    m.fit(X[:600], y[:600], eval_set=[(X[600:], y[600:])],
+          early_stopping_rounds=5, callbacks=[lgb.log_evaluation(0)])
+    import io
+    buf = io.BytesIO()
+    joblib.dump(m, buf)
+    buf.seek(0)
+    m2 = joblib.load(buf)
+    np.testing.assert_allclose(m2.predict(X[:50]), m.predict(X[:50]), rtol=1e-12)
