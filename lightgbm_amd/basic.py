@@ -679,6 +679,24 @@ class Booster:
             self.model_from_string(model_str)
         else:
             raise TypeError("Need train_set, model_file or model_str")
+        if train_set is None:
+            self._load_params_from_model()
+
+    def _load_params_from_model(self):
+        """Populate self.params from the loaded model's parameters block
+        (reference LGBM_BoosterGetLoadedParam semantics)."""
+        try:
+            out_len = ctypes.c_int64(0)
+            _safe_call(_LIB.LGBM_BoosterGetLoadedParam(
+                self._handle, ctypes.c_int64(0), ctypes.byref(out_len), None))
+            buf = ctypes.create_string_buffer(out_len.value + 1)
+            _safe_call(_LIB.LGBM_BoosterGetLoadedParam(
+                self._handle, ctypes.c_int64(len(buf)), ctypes.byref(out_len), buf))
+            loaded = json.loads(buf.value.decode("utf-8"))
+            for k, v in loaded.items():
+                self.params.setdefault(k, v)
+        except (LightGBMError, ValueError):
+            pass
 
     @property
     def handle(self):

@@ -21,6 +21,10 @@ def train(params, train_set, num_boost_round=100, valid_sets=None, valid_names=N
                   "num_round", "num_rounds", "num_boost_round", "n_estimators", "max_iter"):
         if alias in params:
             num_boost_round = int(params.pop(alias))
+    if num_boost_round <= 0:
+        raise ValueError("num_boost_round must be greater than 0")
+    if not isinstance(train_set, Dataset):
+        raise TypeError(f"train_set must be a Dataset, not {type(train_set).__name__}")
     first_metric_only = params.get("first_metric_only", False)
     if fobj is not None or callable(params.get("objective")):
         if callable(params.get("objective")):
@@ -223,6 +227,10 @@ def cv(params, train_set, num_boost_round=100, folds=None, nfold=5, stratified=T
        callbacks=None, eval_train_metric=False, return_cvbooster=False, seed=0,
        fobj=None, fpreproc=None):
     """Cross-validation (parity: reference engine.py:cv)."""
+    if num_boost_round <= 0:
+        raise ValueError("num_boost_round must be greater than 0")
+    if not isinstance(train_set, Dataset):
+        raise TypeError(f"train_set must be a Dataset, not {type(train_set).__name__}")
     params = copy.deepcopy(params) if params else {}
     if metrics is not None:
         params["metric"] = metrics

@@ -1716,3 +1716,38 @@ def test_forced_split_feature_indices(tmp_path):
         root = t["tree_structure"]
         assert root["split_feature"] == 3
         assert root["threshold"] == pytest.approx(0.0, abs=0.2)
+
+
+def test_parameters_loaded_from_model_file(tmp_path):
+    """Booster(model_file) exposes the saved parameters block via .params
+    (ref test_parameters_are_loaded_from_model_file)."""
+    rng = np.random.RandomState(47)
+    X = rng.randn(400, 3)
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "num_leaves": 9,
+                     "lambda_l1": 0.4, "bagging_freq": 2, "bagging_fraction": 0.9},
+                    lgb.Dataset(X, label=X[:, 0]), 3)
+    f = tmp_path / "m.txt"
+    bst.save_model(str(f))
+    b2 = lgb.Booster(model_file=str(f))
+    assert b2.params["objective"] == "regression"
+    assert b2.params["num_leaves"] == 9
+    assert b2.params["lambda_l1"] == 0.4
+    assert b2.params["bagging_freq"] == 2
+    b3 = lgb.Booster(model_str=bst.model_to_string())
+    assert b3.params["num_leaves"] == 9
+
+
+def test_train_cv_informative_errors():
+    """train/cv raise informative errors for bad inputs
+    (ref test_train_and_cv_raise_informative_error_*)."""
+    rng = np.random.RandomState(48)
+    X = rng.randn(100, 3)
+    ds = lgb.Dataset(X, label=X[:, 0])
+    with pytest.raises(ValueError, match="num_boost_round"):
+        lgb.train({"verbosity": -1}, ds, num_boost_round=0)
+    with pytest.raises(ValueError, match="num_boost_round"):
+        lgb.cv({"verbosity": -1}, ds, num_boost_round=-5)
+    with pytest.raises(TypeError, match="Dataset"):
+        lgb.train({"verbosity": -1}, np.zeros((5, 2)), 5)
+    with pytest.raises(TypeError, match="Dataset"):
+        lgb.cv({"verbosity": -1}, [1, 2, 3], 5)
