@@ -1751,3 +1751,23 @@ def test_train_cv_informative_errors():
         lgb.train({"verbosity": -1}, np.zeros((5, 2)), 5)
     with pytest.raises(TypeError, match="Dataset"):
         lgb.cv({"verbosity": -1}, [1, 2, 3], 5)
+
+
+def test_is_unbalance_and_scale_pos_weight():
+    """is_unbalance / scale_pos_weight raise minority-class recall
+    (ref binary objective weighting behavior)."""
+    rng = np.random.RandomState(49)
+    X = rng.randn(5000, 4)
+    y = (X[:, 0] + 0.8 * rng.randn(5000) > 1.3).astype(float)  # ~10% positives
+    base = lgb.train({"objective": "binary", "verbosity": -1},
+                     lgb.Dataset(X, label=y), 20)
+    unb = lgb.train({"objective": "binary", "is_unbalance": True, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 20)
+    spw = lgb.train({"objective": "binary", "scale_pos_weight": 8.0, "verbosity": -1},
+                    lgb.Dataset(X, label=y), 20)
+    def recall(b):
+        return float(((b.predict(X) > 0.5) & (y == 1)).sum() / max((y == 1).sum(), 1))
+    assert recall(unb) > recall(base)
+    assert recall(spw) > recall(base)
+    # mean predicted probability rises with positive upweighting
+    assert unb.predict(X).mean() > base.predict(X).mean()
