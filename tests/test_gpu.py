@@ -937,3 +937,22 @@ def test_gpu_linear_tree_bagging():
     mse_lin = float(np.mean((lin.predict(X) - y) ** 2))
     mse_const = float(np.mean((const.predict(X) - y) ** 2))
     assert mse_lin < mse_const * 0.8, (mse_lin, mse_const)
+
+
+def test_gpu_categorical_nan_bin():
+    """NaN categorical values occupy the reserved bin 0 on device too: a model
+    must be able to separate missing from category 0."""
+    rng = np.random.RandomState(9)
+    n = 20000
+    cat = rng.randint(0, 5, n).astype(np.float64)
+    nan_rows = rng.choice(n, n // 4, replace=False)
+    cat[nan_rows] = np.nan
+    X = np.column_stack([cat, rng.randn(n) * 0.01])
+    lut = np.array([1.0, 2.0, 3.0, 4.0, 5.0])
+    y = np.where(np.isnan(cat), -2.0, lut[np.nan_to_num(cat).astype(int)]).astype(np.float32)
+    bst = lgb.train({"objective": "regression", "device_type": "cuda",
+                     "verbosity": -1, "learning_rate": 0.5, "num_leaves": 15,
+                     "min_data_in_leaf": 1, "cat_l2": 0.0, "cat_smooth": 1e-3},
+                    lgb.Dataset(X, label=y, categorical_feature=[0]), 30)
+    mse = float(np.mean((bst.predict(X) - y) ** 2))
+    assert mse < 0.01, mse
