@@ -201,6 +201,22 @@ void FindBestThresholdCategorical(const hist_t* hist, int num_bin, const LeafCon
     }
   }
   if (best_cats.empty()) { out->gain = kMinScore; return; }
+  // normalize: the NaN/unseen bin 0 must route RIGHT. The raw-value predictor
+  // sends NaN/unseen right unconditionally (category bitsets cannot express the
+  // dummy category -1), so a left-set containing bin 0 would diverge from
+  // training. A categorical split is symmetric: complement the subset instead.
+  if (std::find(best_cats.begin(), best_cats.end(), 0) != best_cats.end()) {
+    std::vector<int> comp;
+    comp.reserve(num_bin - best_cats.size());
+    std::vector<uint8_t> in(num_bin, 0);
+    for (int b : best_cats) in[b] = 1;
+    for (int b = 0; b < num_bin; ++b)
+      if (!in[b]) comp.push_back(b);
+    if (comp.empty()) { out->gain = kMinScore; return; }
+    best_cats = std::move(comp);
+    best_lg = leaf.sum_gradient - best_lg;
+    best_lh = leaf.sum_hessian - best_lh;
+  }
   out->gain = best_gain - min_gain_shift + cfg.min_gain_to_split;
   out->left_sum_gradient = best_lg;
   out->left_sum_hessian = best_lh;

@@ -860,10 +860,20 @@ __global__ void __launch_bounds__(64) k_best_feat(
           uint64_t mask = member && lane < n_elig ? (1ull << orig) : 0ull;
           for (int d = 32; d > 0; d >>= 1) mask |= __shfl_xor(mask, d);
           if (lane == 0) {
+            // bin 0 (NaN/unseen) must route RIGHT: the raw-value predictor
+            // cannot express the dummy category -1 in a left bitset, so
+            // complement the (symmetric) subset instead (CPU-oracle parity)
+            if (mask & 1ull) {
+              const unsigned long long all =
+                  m.num_bin >= 64 ? ~0ull : ((1ull << m.num_bin) - 1ull);
+              mask = ~mask & all;
+              b_lg = sum_g - b_lg;
+              b_lh = sum_h - b_lh;
+            }
             rec.valid = 1;
             rec.gain = bg_cat - min_gain_shift + p.min_gain_to_split;
             rec.feature = f;
-            rec.bin = b_k + 1;  // #cats on the left (CPU threshold semantics)
+            rec.bin = __popcll(mask);  // #cats on the left (CPU threshold semantics)
             rec.default_left = 0;
             rec.cat_mask[0] = mask;
             rec.cat_mask[1] = rec.cat_mask[2] = rec.cat_mask[3] = 0;
@@ -1001,7 +1011,6 @@ __global__ void __launch_bounds__(64) k_best_feat(
           rec.valid = 1;
           rec.gain = bg_cat - min_gain_shift + p.min_gain_to_split;
           rec.feature = f;
-          rec.bin = b_k + 1;  // #cats on the left (CPU threshold semantics)
           rec.default_left = 0;
           rec.cat_mask[0] = rec.cat_mask[1] = rec.cat_mask[2] = rec.cat_mask[3] = 0;
           const int lo_i = b_dir == 0 ? 0 : n_elig - 1 - b_k;
@@ -1010,6 +1019,22 @@ __global__ void __launch_bounds__(64) k_best_feat(
             const int ob = s_sorted_orig[i];
             rec.cat_mask[ob >> 6] |= 1ull << (ob & 63);
           }
+          // bin 0 (NaN/unseen) must route RIGHT — complement the subset
+          if (rec.cat_mask[0] & 1ull) {
+            for (int w2 = 0; w2 < 4; ++w2) {
+              const int lo2 = w2 * 64;
+              unsigned long long all = 0ull;
+              if (m.num_bin > lo2) {
+                const int nb2 = m.num_bin - lo2;
+                all = nb2 >= 64 ? ~0ull : ((1ull << nb2) - 1ull);
+              }
+              rec.cat_mask[w2] = ~rec.cat_mask[w2] & all;
+            }
+            b_lg = sum_g - b_lg;
+            b_lh = sum_h - b_lh;
+          }
+          rec.bin = __popcll(rec.cat_mask[0]) + __popcll(rec.cat_mask[1]) +
+                    __popcll(rec.cat_mask[2]) + __popcll(rec.cat_mask[3]);
           rec.left_g = b_lg;
           rec.left_h = b_lh;
           rec.left_out = d_leaf_out_l2(b_lg, b_lh, p, l2c);
@@ -1128,6 +1153,23 @@ __global__ void __launch_bounds__(64) k_best_feat(
     rec.feature = f;
     rec.bin = best_bin;
     rec.default_left = best_dl;
+    if (m.is_cat && best_bin == 0) {
+      // one-hot winner on the NaN/unseen bin: missing must route RIGHT at
+      // predict time, so emit the complement subset ("every bin except 0")
+      for (int w2 = 0; w2 < 4; ++w2) {
+        const int lo2 = w2 * 64;
+        unsigned long long all = 0ull;
+        if (m.num_bin > lo2) {
+          const int nb2 = m.num_bin - lo2;
+          all = nb2 >= 64 ? ~0ull : ((1ull << nb2) - 1ull);
+        }
+        rec.cat_mask[w2] = all;
+      }
+      rec.cat_mask[0] &= ~1ull;
+      best_lg = sum_g - best_lg;
+      best_lh = sum_h - best_lh;
+      rec.bin = m.num_bin - 1;
+    }
     rec.left_g = best_lg;
     rec.left_h = best_lh;
     const double rg = sum_g - best_lg, rh = sum_h - best_lh;
