@@ -59,3 +59,47 @@ def test_predict_for_arrow():
     np.testing.assert_allclose(pred_arrow, bst.predict(X), rtol=1e-12)
     contrib = bst.predict(table, pred_contrib=True)
     assert contrib.shape == (500, 5)
+
+
+def test_arrow_metadata_fields():
+    """label/weight/group/init_score given as Arrow arrays (ref
+    test_dataset_construct_{labels,weights,groups,init_scores_array})."""
+    pa = pytest.importorskip("pyarrow")
+    rng = np.random.RandomState(10)
+    n = 400
+    tbl = pa.table({"a": rng.randn(n), "b": rng.randn(n)})
+    y = rng.rand(n)
+    w = 1.0 + rng.rand(n)
+    ds = lgb.Dataset(tbl, label=pa.array(y), weight=pa.array(w),
+                     init_score=pa.array(np.zeros(n)))
+    ds.construct()
+    np.testing.assert_allclose(ds.get_label(), y, rtol=1e-6)
+    np.testing.assert_allclose(ds.get_weight(), w, rtol=1e-6)
+    rel = (tbl["a"].to_numpy() > 0).astype(np.float64)  # integer relevance
+    g = lgb.Dataset(tbl, label=pa.array(rel),
+                    group=pa.array(np.array([100, 100, 100, 100], dtype=np.int32)))
+    g.construct()
+    np.testing.assert_array_equal(g.get_group(), [100, 100, 100, 100])
+    bst = lgb.train({"objective": "lambdarank", "verbosity": -1}, g, 3)
+    assert bst.num_trees() == 3
+
+
+def test_arrow_feature_names_and_predict_tasks():
+    """feature names come from the table schema; predict accepts tables across
+    tasks (ref test_arrow_feature_name_auto / test_predict_*)."""
+    pa = pytest.importorskip("pyarrow")
+    rng = np.random.RandomState(11)
+    n = 600
+    tbl = pa.table({"f_one": rng.randn(n), "f_two": rng.randn(n),
+                    "f_three": rng.randn(n)})
+    Xnp = np.column_stack([tbl[c].to_numpy() for c in tbl.column_names])
+    yb = (Xnp[:, 0] > 0).astype(float)
+    bb = lgb.train({"objective": "binary", "verbosity": -1},
+                   lgb.Dataset(tbl, label=yb), 5)
+    assert bb.feature_name() == ["f_one", "f_two", "f_three"]
+    np.testing.assert_allclose(bb.predict(tbl), bb.predict(Xnp), rtol=1e-12)
+    ym = rng.randint(0, 3, n).astype(float)
+    bm = lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+                   lgb.Dataset(tbl, label=ym), 5)
+    assert bm.predict(tbl).shape == (n, 3)
+    assert bm.predict(tbl, pred_leaf=True).shape == (n, 15)
