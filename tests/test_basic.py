@@ -472,3 +472,50 @@ def test_init_score_multiclass_2d():
     raw = bst.predict(X[:500], raw_score=True).reshape(500, 3)
     assert raw[:, 1].mean() < 0 < raw[:, 0].mean()
     assert raw[:, 1].mean() < 0 < raw[:, 2].mean()
+
+
+def test_save_dataset_subset_and_load_from_file(tmp_path):
+    """a subset Dataset saves to the binary format and loads back with its rows
+    (ref test_save_dataset_subset_and_load_from_file)."""
+    rng = np.random.RandomState(60)
+    X = rng.randn(120, 4)
+    y = rng.rand(120)
+    ds = lgb.Dataset(X, label=y, free_raw_data=False)
+    ds.construct()
+    sub = ds.subset(np.arange(30, 100))
+    sub.construct()
+    f = tmp_path / "sub.bin"
+    sub.save_binary(str(f))
+    d2 = lgb.Dataset(str(f))
+    d2.construct()
+    assert d2.num_data() == 70
+    np.testing.assert_allclose(d2.get_label(), y[30:100], rtol=1e-6)
+    bst = lgb.train({"objective": "regression", "verbosity": -1}, d2, 3)
+    assert bst.num_trees() == 3
+
+
+def test_consistent_state_for_dataset_fields():
+    """float64 label/weight are stored as float32 consistently
+    (ref test_consistent_state_for_dataset_fields)."""
+    rng = np.random.RandomState(61)
+    X = rng.randn(100, 3)
+    y64 = rng.rand(100)
+    w64 = 1 + rng.rand(100)
+    ds = lgb.Dataset(X, label=y64, weight=w64)
+    ds.construct()
+    np.testing.assert_allclose(ds.get_label(), y64.astype(np.float32), rtol=1e-7)
+    np.testing.assert_allclose(ds.get_weight(), w64.astype(np.float32), rtol=1e-7)
+    assert ds.get_label().dtype == np.float32
+
+
+def test_feature_names_default_and_custom():
+    """Column_N default names; custom names flow to the model
+    (ref test_feature_names_are_set_correctly...)."""
+    rng = np.random.RandomState(62)
+    X = rng.randn(200, 3)
+    ds = lgb.Dataset(X, label=X[:, 0])
+    bst = lgb.train({"objective": "regression", "verbosity": -1}, ds, 2)
+    assert bst.feature_name() == ["Column_0", "Column_1", "Column_2"]
+    ds2 = lgb.Dataset(X, label=X[:, 0], feature_name=["x", "y", "z"])
+    bst2 = lgb.train({"objective": "regression", "verbosity": -1}, ds2, 2)
+    assert bst2.feature_name() == ["x", "y", "z"]
