@@ -1771,3 +1771,28 @@ def test_is_unbalance_and_scale_pos_weight():
     assert recall(spw) > recall(base)
     # mean predicted probability rises with positive upweighting
     assert unb.predict(X).mean() > base.predict(X).mean()
+
+
+def test_pandas_sparse_dtype():
+    """pandas SparseArray columns train and predict (ref test_pandas_sparse)."""
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(50)
+    df = pd.DataFrame({
+        "a": pd.arrays.SparseArray(np.where(rng.rand(800) < 0.8, 0.0, rng.randn(800))),
+        "b": rng.randn(800)})
+    y = (df["b"] > 0).astype(float)
+    bst = lgb.train({"objective": "binary", "verbosity": -1}, lgb.Dataset(df, label=y), 5)
+    assert ((bst.predict(df) > 0.5) == y).mean() > 0.95
+
+
+def test_categorical_non_zero_inputs():
+    """categories need not start at 0 nor be contiguous
+    (ref test_categorical_non_zero_inputs)."""
+    rng = np.random.RandomState(51)
+    x = rng.choice([5, 7, 11, 400], 500).astype(float).reshape(-1, 1)
+    lut = {5: 1.0, 7: -2.0, 11: 3.0, 400: 0.5}
+    y = np.array([lut[int(v)] for v in x[:, 0]])
+    bst = lgb.train({"objective": "regression", "verbosity": -1, "min_data_in_leaf": 1,
+                     "learning_rate": 1.0, "cat_l2": 0.0, "cat_smooth": 1e-3},
+                    lgb.Dataset(x, label=y, categorical_feature=[0]), 10)
+    np.testing.assert_allclose(bst.predict(x), y, atol=1e-3)
