@@ -357,3 +357,25 @@ def test_cli_python_consistency(tmp_path):
     p_cli = cli_bst.predict(Xt)
     p_py = py_bst.predict(Xt)
     np.testing.assert_allclose(p_cli, p_py, rtol=1e-9)
+
+
+def test_cli_snapshot_and_continue(tmp_path):
+    """snapshot_freq writes model.txt.snapshot_iter_N checkpoints; input_model
+    continues training from a checkpoint (ref Application/GBDT::Train)."""
+    import shutil
+    ex = Path(__file__).resolve().parent.parent / "examples" / "binary_classification"
+    work = tmp_path / "w"
+    shutil.copytree(ex, work)
+    subprocess.run([str(CLI), "config=train.conf", "num_trees=10", "snapshot_freq=5",
+                    "metric_freq=100"], cwd=work, check=True, timeout=300,
+                   capture_output=True)
+    snap = work / "LightGBM_model.txt.snapshot_iter_5"
+    assert snap.exists()
+    assert lgb.Booster(model_file=str(snap)).num_trees() == 5
+    # continue from the snapshot
+    subprocess.run([str(CLI), "config=train.conf", "num_trees=5",
+                    f"input_model={snap}", "output_model=cont.txt",
+                    "metric_freq=100"], cwd=work, check=True, timeout=300,
+                   capture_output=True)
+    cont = lgb.Booster(model_file=str(work / "cont.txt"))
+    assert cont.num_trees() == 10
