@@ -593,7 +593,10 @@ class Dataset:
 
     def subset(self, used_indices, params=None):
         self.construct()
-        used = np.ascontiguousarray(np.asarray(used_indices).ravel(), dtype=np.int32)
+        # sorted order is required downstream (query-boundary slicing in
+        # Dataset::Subset counts rows per query in index order)
+        used = np.sort(np.asarray(used_indices).ravel()).astype(np.int32)
+        used = np.ascontiguousarray(used)
         out = ctypes.c_void_p()
         _safe_call(_LIB.LGBM_DatasetGetSubset(
             self._handle, used.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
