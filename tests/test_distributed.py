@@ -303,3 +303,81 @@ def test_distributed_auc_is_global(tmp_path):
         capture_output=True, text=True, timeout=300, env=env)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "DIST_OK" in r.stdout
+
+
+WORKER_CAT_MONO = r"""
+import hashlib
+import os, sys
+sys.path.insert(0, sys.argv[1])
+import numpy as np
+import torch.distributed as dist
+import datetime
+
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+dist.init_process_group("gloo", timeout=datetime.timedelta(seconds=120))
+rank, world = dist.get_rank(), dist.get_world_size()
+
+import lightgbm_amd as lgb
+from lightgbm_amd.parallel import init_network_from_torch_distributed
+init_network_from_torch_distributed()
+
+# shared-seed reference dataset with a CATEGORICAL column
+rng = np.random.RandomState(7)
+nref = 5000
+cat = rng.randint(0, 8, nref).astype(np.float64)
+Xref = np.column_stack([cat, rng.randn(nref, 3)])
+lut = np.array([2.0, -1.0, 0.5, 3.0, -2.0, 1.0, 0.0, -0.5])
+yref = (lut[cat.astype(int)] + Xref[:, 1] > 0).astype(np.float32)
+ref = lgb.Dataset(Xref, label=yref, categorical_feature=[0],
+                  params={"max_bin": 63}).construct()
+
+rng = np.random.RandomState(300 + rank)
+n = 6000
+catl = rng.randint(0, 8, n).astype(np.float64)
+X = np.column_stack([catl, rng.randn(n, 3)])
+y = (lut[catl.astype(int)] + X[:, 1] + 0.3 * rng.randn(n) > 0).astype(np.float32)
+train = ref.create_valid(X, label=y)
+
+# monotone constraint on the increasing dense feature (basic policy is the only
+# one allowed in distributed mode — the config downgrade must kick in silently
+# when intermediate is requested)
+params = {"objective": "binary", "tree_learner": "data", "num_leaves": 31,
+          "verbosity": -1, "max_bin": 63,
+          "monotone_constraints": [0, 1, 0, 0],
+          "monotone_constraints_method": "intermediate"}
+bst = lgb.train(params, train, num_boost_round=15)
+digest = hashlib.sha256(bst.model_to_string().encode()).hexdigest()
+payload = [None] * world
+dist.all_gather_object(payload, digest)
+assert len(set(payload)) == 1, f"rank models differ: {payload}"
+
+# monotone property on the constrained feature
+xs = np.linspace(-2, 2, 30)
+for c in (0.0, 3.0):
+    g = np.column_stack([np.full(30, c), xs, np.zeros(30), np.zeros(30)])
+    p = bst.predict(g)
+    assert np.all(np.diff(p) >= -1e-9), "monotone violated"
+# categorical feature is actually used
+imp = bst.feature_importance()
+assert imp[0] > 0
+if rank == 0:
+    print("DIST_CAT_MONO_OK")
+dist.destroy_process_group()
+"""
+
+
+def test_data_parallel_categorical_and_monotone(tmp_path):
+    """distributed training with categorical features + monotone constraints:
+    identical models per rank, constraint respected, intermediate->basic
+    downgrade applied in distributed mode."""
+    script = tmp_path / "worker_cm.py"
+    script.write_text(WORKER_CAT_MONO)
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29547", str(script), str(REPO)],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "DIST_CAT_MONO_OK" in r.stdout
