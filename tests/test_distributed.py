@@ -338,9 +338,9 @@ X = np.column_stack([catl, rng.randn(n, 3)])
 y = (lut[catl.astype(int)] + X[:, 1] + 0.3 * rng.randn(n) > 0).astype(np.float32)
 train = ref.create_valid(X, label=y)
 
-# monotone constraint on the increasing dense feature (basic policy is the only
-# one allowed in distributed mode — the config downgrade must kick in silently
-# when intermediate is requested)
+# monotone constraint on the increasing dense feature; requesting intermediate
+# must stay SAFE in distributed mode (every rank applies identical bound updates
+# and recomputes in lockstep, so collectives stay matched)
 params = {"objective": "binary", "tree_learner": "data", "num_leaves": 31,
           "verbosity": -1, "max_bin": 63,
           "monotone_constraints": [0, 1, 0, 0],
@@ -368,8 +368,8 @@ dist.destroy_process_group()
 
 def test_data_parallel_categorical_and_monotone(tmp_path):
     """distributed training with categorical features + monotone constraints:
-    identical models per rank, constraint respected, intermediate->basic
-    downgrade applied in distributed mode."""
+    identical models per rank, constraint respected with the intermediate
+    policy requested (lockstep recompute keeps collectives matched)."""
     script = tmp_path / "worker_cm.py"
     script.write_text(WORKER_CAT_MONO)
     env = dict(os.environ)
