@@ -1796,3 +1796,26 @@ def test_categorical_non_zero_inputs():
                      "learning_rate": 1.0, "cat_l2": 0.0, "cat_smooth": 1e-3},
                     lgb.Dataset(x, label=y, categorical_feature=[0]), 10)
     np.testing.assert_allclose(bst.predict(x), y, atol=1e-3)
+
+
+def test_register_logger_captures_native_logs():
+    """register_logger redirects native Warning/Info lines into a Python logger
+    (ref test_utilities.py test_register_logger)."""
+    import logging
+    records = []
+    logger = logging.getLogger("migbm_capture_test")
+    logger.setLevel(logging.DEBUG)
+
+    class _H(logging.Handler):
+        def emit(self, r):
+            records.append(r.getMessage())
+    h = _H()
+    logger.addHandler(h)
+    try:
+        lgb.register_logger(logger)
+        X = np.zeros((100, 2))  # all-constant features emit a Warning
+        lgb.train({"objective": "regression", "verbosity": 0},
+                  lgb.Dataset(X, label=np.zeros(100)), 1)
+        assert any("trivial" in m for m in records), records
+    finally:
+        logger.removeHandler(h)
