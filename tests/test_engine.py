@@ -1819,3 +1819,7 @@ def test_register_logger_captures_native_logs():
         assert any("trivial" in m for m in records), records
     finally:
         logger.removeHandler(h)
+        # restore native stderr logging for the rest of the suite
+        import ctypes
+        from lightgbm_amd.basic import _LIB
+        _LIB.LGBM_RegisterLogCallback(ctypes.cast(None, ctypes.CFUNCTYPE(None, ctypes.c_char_p)))
