@@ -53,7 +53,7 @@ class Log {
   static Callback& Cb() { static Callback cb = nullptr; return cb; }
 
   static void Write(LogLevel lv, const char* tag, const char* fmt, va_list ap) {
-    if (lv > Level() && Cb() == nullptr) return;
+    if (lv > Level()) return;  // the registered callback respects the level too
     char buf[2048];
     vsnprintf(buf, sizeof(buf), fmt, ap);
     char out[2112];
