@@ -1814,8 +1814,10 @@ def test_register_logger_captures_native_logs():
     try:
         lgb.register_logger(logger)
         X = np.zeros((100, 2))  # all-constant features emit a Warning
+        # verbosity on the DATASET too: the warning fires at construction time
+        # and the log level must already allow warnings then
         lgb.train({"objective": "regression", "verbosity": 0},
-                  lgb.Dataset(X, label=np.zeros(100)), 1)
+                  lgb.Dataset(X, label=np.zeros(100), params={"verbosity": 0}), 1)
         assert any("trivial" in m for m in records), records
     finally:
         logger.removeHandler(h)
