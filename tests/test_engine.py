@@ -1825,3 +1825,30 @@ def test_register_logger_captures_native_logs():
         import ctypes
         from lightgbm_amd.basic import _LIB
         _LIB.LGBM_RegisterLogCallback(ctypes.cast(None, ctypes.CFUNCTYPE(None, ctypes.c_char_p)))
+
+
+def test_degenerate_inputs_error_gracefully():
+    """degenerate inputs either train or raise informative LightGBMError —
+    never crash (reference error-contract behaviors)."""
+    rng = np.random.RandomState(52)
+    # these succeed
+    lgb.train({"objective": "regression", "verbosity": -1, "min_data_in_leaf": 1,
+               "min_data_in_bin": 1}, lgb.Dataset(np.ones((1, 2)), label=np.ones(1)), 2)
+    lgb.train({"objective": "regression", "verbosity": -1},
+              lgb.Dataset(rng.randn(100, 1), label=rng.rand(100)), 2)
+    b = lgb.train({"objective": "regression", "verbosity": -1},
+                  lgb.Dataset(rng.randn(100, 2), label=rng.rand(100)), 2)
+    assert b.predict(np.zeros((0, 2))).shape == (0,)
+    # these raise with informative messages
+    from lightgbm_amd.basic import LightGBMError
+    with pytest.raises(LightGBMError, match="non-negative"):
+        lgb.train({"objective": "poisson", "verbosity": -1},
+                  lgb.Dataset(rng.randn(100, 2), label=-np.ones(100)), 2)
+    with pytest.raises(LightGBMError, match="Length of label"):
+        lgb.Dataset(rng.randn(100, 2), label=np.zeros(50)).construct()
+    with pytest.raises(LightGBMError, match="finite"):
+        lgb.train({"objective": "regression", "verbosity": -1},
+                  lgb.Dataset(rng.randn(100, 2), label=np.full(100, np.inf)), 2)
+    with pytest.raises(LightGBMError, match="out of range"):
+        lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
+                  lgb.Dataset(rng.randn(100, 2), label=np.full(100, 7.0)), 2)
