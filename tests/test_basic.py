@@ -519,3 +519,28 @@ def test_feature_names_default_and_custom():
     ds2 = lgb.Dataset(X, label=X[:, 0], feature_name=["x", "y", "z"])
     bst2 = lgb.train({"objective": "regression", "verbosity": -1}, ds2, 2)
     assert bst2.feature_name() == ["x", "y", "z"]
+
+
+def test_chunked_sequence_batches_match_numpy():
+    """Sequence ingestion with a small batch_size builds the same dataset as a
+    direct numpy matrix (ref test_chunked_dataset)."""
+    class SmallBatchSeq(lgb.basic.Sequence):
+        batch_size = 37  # deliberately awkward chunking
+
+        def __init__(self, arr):
+            self.arr = arr
+
+        def __getitem__(self, i):
+            return self.arr[i]
+
+        def __len__(self):
+            return len(self.arr)
+
+    rng = np.random.RandomState(63)
+    X = rng.randn(1000, 4)
+    y = (X[:, 0] > 0).astype(np.float32)
+    p = {"objective": "binary", "verbosity": -1, "seed": 3}
+    m_np = lgb.train(p, lgb.Dataset(X, label=y), 8).model_to_string()
+    m_seq = lgb.train(p, lgb.Dataset([SmallBatchSeq(X[:400]), SmallBatchSeq(X[400:])],
+                                     label=y), 8).model_to_string()
+    assert m_np == m_seq
