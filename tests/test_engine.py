@@ -1852,3 +1852,36 @@ def test_degenerate_inputs_error_gracefully():
     with pytest.raises(LightGBMError, match="out of range"):
         lgb.train({"objective": "multiclass", "num_class": 3, "verbosity": -1},
                   lgb.Dataset(rng.randn(100, 2), label=np.full(100, 7.0)), 2)
+
+
+def test_rank_xendcg_quality():
+    """rank_xendcg reaches lambdarank-class NDCG on a clean ranking signal
+    (ref test_xendcg)."""
+    rng = np.random.RandomState(53)
+    qsizes = [20] * 120
+    n = sum(qsizes)
+    X = rng.rand(n, 6)
+    rel = np.clip((3 * X[:, 0] + 0.5 * rng.randn(n)).astype(int), 0, 3)
+    res = {}
+    ds = lgb.Dataset(X, label=rel.astype(float), group=qsizes)
+    vs = lgb.Dataset(X, label=rel.astype(float), group=qsizes, reference=ds) \
+        if hasattr(lgb.Dataset, "reference") else ds.create_valid(X, label=rel.astype(float), group=qsizes)
+    lgb.train({"objective": "rank_xendcg", "metric": "ndcg", "eval_at": [5],
+               "verbosity": -1, "objective_seed": 7}, ds, 40,
+              valid_sets=[vs], callbacks=[lgb.record_evaluation(res)])
+    ndcg = res["valid_0"]["ndcg@5"][-1]
+    assert ndcg > 0.85, ndcg
+
+
+def test_multiclass_rf():
+    """random forest boosting with multiclass softmax (ref test_multiclass_rf)."""
+    rng = np.random.RandomState(54)
+    X = rng.randn(4000, 5)
+    y = (X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)
+    bst = lgb.train({"objective": "multiclass", "num_class": 3, "boosting": "rf",
+                     "bagging_fraction": 0.7, "bagging_freq": 1, "verbosity": -1,
+                     "num_leaves": 31}, lgb.Dataset(X, label=y.astype(float)), 30)
+    pred = bst.predict(X)
+    assert pred.shape == (4000, 3)
+    np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-6)
+    assert (np.argmax(pred, axis=1) == y).mean() > 0.8
