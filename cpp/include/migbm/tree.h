@@ -105,6 +105,9 @@ class Tree {
 
   /*! score[i] += tree(data row i) over an index set (used by score updater on CPU). */
   void AddPredictionToScore(const class Dataset* data, data_size_t num_data, double* score) const;
+  /*! bin->representative-value walk for trees WITHOUT bin thresholds (loaded models). */
+  void AddPredictionToScoreByValue(const class Dataset* data, data_size_t num_data,
+                                   double* score) const;
   void AddPredictionToScore(const class Dataset* data, const data_size_t* used_indices,
                             data_size_t num_data, double* score) const;
 

@@ -1417,6 +1417,58 @@ void Tree::AddPredictionToScore(const Dataset* data, data_size_t num_data, doubl
   }
 }
 
+void Tree::AddPredictionToScoreByValue(const Dataset* data, data_size_t num_data,
+                                       double* score) const {
+  // for LOADED/merged trees: threshold_in_bin_ is not populated (model text
+  // stores only real-valued thresholds), so route every row by representative
+  // raw values reconstructed from its bins via the dataset's bin mappers
+  if (num_leaves_ <= 1) {
+    if (leaf_value_[0] != 0.0) {
+#pragma omp parallel for schedule(static)
+      for (data_size_t i = 0; i < num_data; ++i) score[i] += leaf_value_[0];
+    }
+    return;
+  }
+#pragma omp parallel for schedule(static, 2048)
+  for (data_size_t i = 0; i < num_data; ++i) {
+    int node = 0;
+    while (node >= 0) {
+      const int orig = split_feature_[node];
+      double v = 0.0;
+      const int inner = orig >= 0 && orig < static_cast<int>(data->num_total_features())
+                            ? data->InnerFeatureIndex(orig) : -1;
+      if (inner >= 0) {
+        const BinMapper* m = data->FeatureBinMapper(inner);
+        const uint32_t bin = data->GetBin(i, inner);
+        if (m->bin_type() != BinType::kCategorical && m->nan_bin() >= 0 &&
+            bin == static_cast<uint32_t>(m->nan_bin()))
+          v = std::numeric_limits<double>::quiet_NaN();
+        else
+          v = m->BinToValue(bin);
+      }
+      node = IsCategoricalSplit(node) ? CategoricalDecision(v, node)
+                                      : NumericalDecision(v, node);
+    }
+    if (is_linear_ && data->has_raw()) {
+      const int leaf = ~node;
+      double out = leaf_const_.empty() || leaf_coeff_[leaf].empty()
+                       ? leaf_value_[leaf] : leaf_const_[leaf];
+      if (!leaf_const_.empty() && !leaf_coeff_[leaf].empty()) {
+        bool ok = true;
+        for (size_t k = 0; k < leaf_coeff_[leaf].size(); ++k) {
+          const float rv = data->raw_value(leaf_features_inner_[leaf][k], i);
+          if (std::isnan(rv)) { ok = false; break; }
+          out += leaf_coeff_[leaf][k] * rv;
+        }
+        if (!ok) out = leaf_value_[leaf];
+      }
+      score[i] += out;
+    } else {
+      score[i] += leaf_value_[~node];
+    }
+  }
+}
+
 void Tree::AddPredictionToScore(const Dataset* data, const data_size_t* used_indices,
                                 data_size_t num_data, double* score) const {
   if (num_leaves_ <= 1) {

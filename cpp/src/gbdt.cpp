@@ -149,6 +149,16 @@ void GBDT::Init(const Config* config, const Dataset* train_data,
       // device-resident scores must start from the same per-row init offsets
       if (tree_learner_->IsHIPLearner()) tree_learner_->UploadTrainScore(train_score_.data());
     }
+    // continued training (input_model loaded before Init): fold the existing
+    // trees into the scores so new gradients continue from them
+    for (size_t i = 0; i < models_.size(); ++i) {
+      const int tid = static_cast<int>(i) % num_tree_per_iteration_;
+      models_[i]->AddPredictionToScoreByValue(
+          train_data_, num_data_,
+          train_score_.data() + static_cast<size_t>(tid) * num_data_);
+    }
+    if (!models_.empty() && tree_learner_->IsHIPLearner())
+      tree_learner_->UploadTrainScore(train_score_.data());
     gradients_.assign(train_score_.size(), 0);
     hessians_.assign(train_score_.size(), 0);
     sample_strategy_.reset(SampleStrategy::Create(config_, train_data_, objective_,
@@ -168,11 +178,13 @@ void GBDT::ResetTrainingData(const Dataset* train_data, const ObjectiveFunction*
   // a file-loaded booster has no learner/config until (re)initialized for training
   if (tree_learner_ != nullptr) tree_learner_->ResetTrainingData(train_data_);
   train_score_.assign(static_cast<size_t>(num_data_) * num_tree_per_iteration_, 0.0);
-  // re-apply existing model to scores
+  // re-apply existing model to scores — ByValue walk: file-loaded models carry
+  // only real-valued thresholds (also exact for in-memory trees)
   for (size_t i = 0; i < models_.size(); ++i) {
     int tid = static_cast<int>(i) % num_tree_per_iteration_;
-    models_[i]->AddPredictionToScore(train_data_, num_data_,
-                                     train_score_.data() + static_cast<size_t>(tid) * num_data_);
+    models_[i]->AddPredictionToScoreByValue(
+        train_data_, num_data_,
+        train_score_.data() + static_cast<size_t>(tid) * num_data_);
   }
   gradients_.assign(train_score_.size(), 0);
   hessians_.assign(train_score_.size(), 0);
@@ -203,11 +215,12 @@ void GBDT::AddValidDataset(const Dataset* valid_data,
     int64_t len = valid_data->metadata().num_init_score();
     std::copy(init_sc, init_sc + std::min<int64_t>(len, score.size()), score.begin());
   }
-  // apply existing model
+  // apply existing model (ByValue: merged/loaded trees carry no bin thresholds)
   for (size_t i = 0; i < models_.size(); ++i) {
     int tid = static_cast<int>(i) % num_tree_per_iteration_;
-    models_[i]->AddPredictionToScore(valid_data, valid_data->num_data(),
-                                     score.data() + static_cast<size_t>(tid) * valid_data->num_data());
+    models_[i]->AddPredictionToScoreByValue(
+        valid_data, valid_data->num_data(),
+        score.data() + static_cast<size_t>(tid) * valid_data->num_data());
   }
   valid_score_.push_back(std::move(score));
 }
@@ -458,10 +471,32 @@ std::vector<std::string> GBDT::EvalNames() const {
 }
 
 void GBDT::MergeFrom(const GBDT* other) {
+  const int first_new = NumberOfTotalModel();
   for (int i = 0; i < other->NumberOfTotalModel(); ++i) {
     models_.emplace_back(new Tree(*other->models_[i]));
   }
   iter_ += other->iter_;
+  // continued training must SEE the merged trees: fold their outputs into the
+  // train/valid scores so the next iteration's gradients continue from them
+  // (the reference reaches the same state by loading the model before Init)
+  if (train_data_ != nullptr && !train_score_.empty()) {
+    const bool dev = tree_learner_ != nullptr && tree_learner_->IsHIPLearner();
+    for (int i = first_new; i < NumberOfTotalModel(); ++i) {
+      const int cls = (i - first_new) % num_tree_per_iteration_;
+      // ByValue walk: merged trees come from model text and carry only
+      // real-valued thresholds (no bin-space thresholds)
+      models_[i]->AddPredictionToScoreByValue(
+          train_data_, num_data_,
+          train_score_.data() + static_cast<size_t>(cls) * num_data_);
+      for (size_t v = 0; v < valid_data_.size(); ++v) {
+        models_[i]->AddPredictionToScoreByValue(
+            valid_data_[v], valid_data_[v]->num_data(),
+            valid_score_[v].data() +
+                static_cast<size_t>(cls) * valid_data_[v]->num_data());
+      }
+    }
+    if (dev) tree_learner_->UploadTrainScore(train_score_.data());
+  }
 }
 
 bool GBDT::EvalAndCheckEarlyStopping() {

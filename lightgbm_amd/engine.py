@@ -234,8 +234,11 @@ def cv(params, train_set, num_boost_round=100, folds=None, nfold=5, stratified=T
     params = copy.deepcopy(params) if params else {}
     if metrics is not None:
         params["metric"] = metrics
-    if params.get("objective") in ("binary",) or stratified is None:
-        pass
+    # callable objective in params (sklearn-style) trains with objective=none
+    if fobj is not None or callable(params.get("objective")):
+        if callable(params.get("objective")):
+            fobj = params["objective"]
+        params["objective"] = "none"
     train_set.construct()
     if folds is None:
         obj = params.get("objective", "")
@@ -256,6 +259,13 @@ def cv(params, train_set, num_boost_round=100, folds=None, nfold=5, stratified=T
             # (reference engine.py cv fpreproc)
             tr, te, fold_params = fpreproc(tr, te, copy.deepcopy(params))
         bst = Booster(params=fold_params, train_set=tr)
+        if init_model is not None:
+            # continue every fold booster from the given model (reference
+            # cv(init_model=...) semantics)
+            from .basic import _LIB, _safe_call
+            base = Booster(model_str=init_model.model_to_string()) \
+                if isinstance(init_model, Booster) else Booster(model_file=str(init_model))
+            _safe_call(_LIB.LGBM_BoosterMerge(bst._handle, base._handle))
         bst.add_valid(te, "valid")
         cvbooster._append(bst)
         fold_data.append((tr, te))

@@ -1885,3 +1885,51 @@ def test_multiclass_rf():
     assert pred.shape == (4000, 3)
     np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-6)
     assert (np.argmax(pred, axis=1) == y).mean() > 0.8
+
+
+def test_cv_with_init_model_and_callable_objective():
+    """cv(init_model=...) continues every fold from the model; a callable
+    objective in params drives cv like train (ref test_cv_works_with_init_model
+    / test_objective_callable_cv_*)."""
+    rng = np.random.RandomState(55)
+    X = rng.randn(900, 4)
+    y = X[:, 0] + 0.1 * rng.randn(900)
+    ds = lgb.Dataset(X, label=y, free_raw_data=False)
+    base = lgb.train({"objective": "regression", "verbosity": -1}, ds, 5)
+    res = lgb.cv({"objective": "regression", "metric": "l2", "verbosity": -1},
+                 ds, num_boost_round=5, nfold=3, init_model=base,
+                 return_cvbooster=True)
+    for b in res["cvbooster"].boosters:
+        assert b.num_trees() == 10  # 5 merged + 5 trained
+    # continued folds start from the base model's error level, not from zero
+    assert res["valid l2-mean"][0] < float(np.var(y))
+
+    def l2_obj(preds, dataset):
+        g = (preds - dataset.get_label()).astype(np.float32)
+        return g, np.ones_like(g)
+    yb = (X[:, 0] > 0).astype(float)
+    res2 = lgb.cv({"objective": l2_obj, "metric": "l2", "verbosity": -1},
+                  lgb.Dataset(X, label=y), num_boost_round=5, nfold=3)
+    assert len(res2["valid l2-mean"]) == 5
+    assert res2["valid l2-mean"][-1] < res2["valid l2-mean"][0]
+
+
+def test_continue_training_equals_straight_run():
+    """init_model continuation is numerically identical to uninterrupted
+    training: merged trees' outputs are folded into the training scores
+    (was silently restarting gradients from zero)."""
+    rng = np.random.RandomState(56)
+    X = rng.randn(3000, 4)
+    cases = [
+        ({"objective": "regression", "learning_rate": 0.2}, X[:, 0] + 0.1 * rng.randn(3000)),
+        ({"objective": "binary", "learning_rate": 0.2}, (X[:, 0] > 0).astype(float)),
+        ({"objective": "multiclass", "num_class": 3, "learning_rate": 0.2},
+         rng.randint(0, 3, 3000).astype(float)),
+    ]
+    for extra, y in cases:
+        p = {**extra, "verbosity": -1}
+        straight = lgb.train(p, lgb.Dataset(X, label=y), 10)
+        half = lgb.train(p, lgb.Dataset(X, label=y), 5)
+        cont = lgb.train(p, lgb.Dataset(X, label=y), 5, init_model=half)
+        np.testing.assert_allclose(cont.predict(X[:500]), straight.predict(X[:500]),
+                                   rtol=1e-9, err_msg=str(extra))
