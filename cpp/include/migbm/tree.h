@@ -199,6 +199,9 @@ class Tree {
   int num_leaves_;
   int num_cat_ = 0;
   bool is_linear_ = false;
+  // false for text-loaded trees: threshold_in_bin_ / bin-space cat bitsets are
+  // absent, so dataset walks must route by real values (ByValue)
+  bool bin_thresholds_valid_ = true;
   double shrinkage_ = 1.0;
   // per internal node (num_leaves_-1 entries)
   std::vector<int> left_child_, right_child_;

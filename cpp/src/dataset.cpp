@@ -1365,6 +1365,11 @@ void Dataset::DumpTextFile(const char* filename) const {
 namespace migbm {
 
 void Tree::AddPredictionToScore(const Dataset* data, data_size_t num_data, double* score) const {
+  if (!bin_thresholds_valid_) {
+    // text-loaded tree: no bin-space thresholds — route by real values
+    AddPredictionToScoreByValue(data, num_data, score);
+    return;
+  }
   if (num_leaves_ <= 1) {
     if (leaf_value_[0] != 0.0) {
 #pragma omp parallel for schedule(static)
@@ -1475,6 +1480,33 @@ void Tree::AddPredictionToScore(const Dataset* data, const data_size_t* used_ind
     if (leaf_value_[0] != 0.0) {
 #pragma omp parallel for schedule(static)
       for (data_size_t i = 0; i < num_data; ++i) score[used_indices[i]] += leaf_value_[0];
+    }
+    return;
+  }
+  if (!bin_thresholds_valid_) {
+    // text-loaded tree: route by representative real values (cf. ByValue walk)
+#pragma omp parallel for schedule(static, 2048)
+    for (data_size_t i = 0; i < num_data; ++i) {
+      const data_size_t r = used_indices[i];
+      int node = 0;
+      while (node >= 0) {
+        const int orig = split_feature_[node];
+        double v = 0.0;
+        const int inner = orig >= 0 && orig < static_cast<int>(data->num_total_features())
+                              ? data->InnerFeatureIndex(orig) : -1;
+        if (inner >= 0) {
+          const BinMapper* m = data->FeatureBinMapper(inner);
+          const uint32_t bin = data->GetBin(r, inner);
+          if (m->bin_type() != BinType::kCategorical && m->nan_bin() >= 0 &&
+              bin == static_cast<uint32_t>(m->nan_bin()))
+            v = std::numeric_limits<double>::quiet_NaN();
+          else
+            v = m->BinToValue(bin);
+        }
+        node = IsCategoricalSplit(node) ? CategoricalDecision(v, node)
+                                        : NumericalDecision(v, node);
+      }
+      score[r] += leaf_value_[~node];
     }
     return;
   }

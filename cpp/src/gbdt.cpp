@@ -771,6 +771,12 @@ std::vector<int> DART::DroppingTrees() {
 }
 
 bool DART::TrainOneIter(const score_t* gradients, const score_t* hessians) {
+  // continued training: merged/loaded iterations have no recorded drop weights —
+  // pad them at weight 1.0 so absolute iteration indexing stays in bounds
+  while (tree_weight_.size() < static_cast<size_t>(GetCurrentIteration())) {
+    tree_weight_.push_back(1.0);
+    sum_weight_ += 1.0;
+  }
   auto dropped = DroppingTrees();
   // device-resident scores: DART's drop/renormalize surgery happens on the host
   // copy, synced down before and up after each host-side mutation

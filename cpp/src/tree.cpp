@@ -207,6 +207,7 @@ std::map<std::string, std::string> ParseKV(const char* str, size_t* used_len) {
 }  // namespace
 
 Tree::Tree(const char* str, size_t* used_len) {
+  bin_thresholds_valid_ = false;  // model text carries only real-valued thresholds
   auto kv = ParseKV(str, used_len);
   auto get = [&](const char* k) -> const std::string& {
     static const std::string empty;
