@@ -304,3 +304,35 @@ def test_ranking_quality_parity_with_reference(tmp_path):
         return total / len(groups)
     ours, theirs = ndcg5(our_scores), ndcg5(ref_scores)
     assert ours > theirs - 0.01, (ours, theirs)
+
+
+def test_continue_training_from_reference_model(tmp_path):
+    """a REFERENCE-trained model continues training in migbm: the merged trees'
+    outputs seed the scores, so the continued model strictly improves and the
+    combined model still loads back in the reference."""
+    X, y = _data(n=6000, seed=9)
+    np.save(tmp_path / "x.npy", X)
+    np.save(tmp_path / "y.npy", y)
+    f = tmp_path / "ref_base.txt"
+    _ref({"op": "train", "x": str(tmp_path / "x.npy"), "y": str(tmp_path / "y.npy"),
+          "model": str(f), "ds_params": "max_bin=255",
+          "params": "objective=binary verbosity=-1 num_leaves=31 learning_rate=0.1",
+          "iters": 10}, tmp_path)
+
+    def logloss(p):
+        p = np.clip(p, 1e-12, 1 - 1e-12)
+        return float(-np.mean(y * np.log(p) + (1 - y) * np.log(1 - p)))
+
+    base = lgb.Booster(model_file=str(f))
+    ll_base = logloss(base.predict(X))
+    cont = lgb.train({"objective": "binary", "verbosity": -1, "num_leaves": 31,
+                      "learning_rate": 0.1}, lgb.Dataset(X, label=y), 10,
+                     init_model=str(f))
+    assert cont.num_trees() == 20
+    ll_cont = logloss(cont.predict(X))
+    assert ll_cont < ll_base * 0.9, (ll_base, ll_cont)
+    # the combined model round-trips through the reference byte-compatibly
+    out = tmp_path / "combined.txt"
+    cont.save_model(str(out))
+    theirs = _ref_predict(out, X[:500], tmp_path)
+    np.testing.assert_allclose(cont.predict(X[:500]), theirs, rtol=1e-9, atol=1e-12)
