@@ -1933,3 +1933,41 @@ def test_continue_training_equals_straight_run():
         cont = lgb.train(p, lgb.Dataset(X, label=y), 5, init_model=half)
         np.testing.assert_allclose(cont.predict(X[:500]), straight.predict(X[:500]),
                                    rtol=1e-9, err_msg=str(extra))
+
+
+def test_feature_fraction_bynode_cpu():
+    """per-node column sampling changes trees but keeps quality
+    (ref test_node_level_subcol)."""
+    rng = np.random.RandomState(57)
+    X = rng.randn(4000, 8)
+    y = X[:, 0] + X[:, 1] + 0.1 * rng.randn(4000)
+    full = lgb.train({"objective": "regression", "verbosity": -1, "seed": 1},
+                     lgb.Dataset(X, label=y), 30)
+    sub = lgb.train({"objective": "regression", "verbosity": -1, "seed": 1,
+                     "feature_fraction_bynode": 0.5}, lgb.Dataset(X, label=y), 30)
+    assert sub.model_to_string() != full.model_to_string()
+    mse_sub = float(np.mean((sub.predict(X) - y) ** 2))
+    mse_full = float(np.mean((full.predict(X) - y) ** 2))
+    assert mse_sub < mse_full * 3.0
+    # reproducible under the same seed
+    sub2 = lgb.train({"objective": "regression", "verbosity": -1, "seed": 1,
+                      "feature_fraction_bynode": 0.5}, lgb.Dataset(X, label=y), 30)
+    assert sub2.model_to_string() == sub.model_to_string()
+
+
+def test_sample_strategy_with_boosting_combos():
+    """GOSS composes with dart and with plain gbdt; bagging composes with dart
+    (ref test_sample_strategy_with_boosting)."""
+    rng = np.random.RandomState(58)
+    X = rng.randn(3000, 5)
+    y = X[:, 0] + 0.2 * rng.randn(3000)
+    combos = [
+        {"boosting": "dart", "data_sample_strategy": "goss"},
+        {"boosting": "gbdt", "data_sample_strategy": "goss"},
+        {"boosting": "dart", "bagging_fraction": 0.7, "bagging_freq": 1},
+    ]
+    for extra in combos:
+        bst = lgb.train({"objective": "regression", "verbosity": -1, "seed": 2, **extra},
+                        lgb.Dataset(X, label=y), 20)
+        mse = float(np.mean((bst.predict(X) - y) ** 2))
+        assert mse < float(np.var(y)) * 0.5, (extra, mse)
