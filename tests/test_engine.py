@@ -1971,3 +1971,26 @@ def test_sample_strategy_with_boosting_combos():
                         lgb.Dataset(X, label=y), 20)
         mse = float(np.mean((bst.predict(X) - y) ** 2))
         assert mse < float(np.var(y)) * 0.5, (extra, mse)
+
+
+def test_early_stopping_via_global_params_with_min_delta():
+    """early_stopping_round + early_stopping_min_delta in params drive stopping
+    without an explicit callback (ref test_early_stopping_via_global_params /
+    _min_delta_via_global_params)."""
+    rng = np.random.RandomState(59)
+    X = rng.randn(4000, 5)
+    y = X[:, 0] + 0.3 * rng.randn(4000)
+    tr = lgb.Dataset(X[:3000], label=y[:3000])
+    va = tr.create_valid(X[3000:], label=y[3000:])
+    res = {}
+    bst = lgb.train({"objective": "regression", "metric": "l2", "verbosity": -1,
+                     "early_stopping_round": 5}, tr, 300, valid_sets=[va],
+                    callbacks=[lgb.record_evaluation(res)])
+    assert 0 < bst.best_iteration < 300
+    # a large min_delta stops much earlier: tiny improvements no longer count
+    res2 = {}
+    bst2 = lgb.train({"objective": "regression", "metric": "l2", "verbosity": -1,
+                      "early_stopping_round": 5, "early_stopping_min_delta": 0.05},
+                     tr, 300, valid_sets=[va], callbacks=[lgb.record_evaluation(res2)])
+    assert bst2.best_iteration <= bst.best_iteration
+    assert len(res2["valid_0"]["l2"]) < len(res["valid_0"]["l2"])
